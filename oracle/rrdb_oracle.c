@@ -1,0 +1,1991 @@
+/* rrdb_oracle.c — CPU restatement oracle for the Pegasus hot path.
+ *
+ * TEST INFRASTRUCTURE ONLY.  This library is the parity checker for the HIP
+ * engine: only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg
+ * may load it.  The product path (incubator_pegasus_amd + librrdb_hip.so)
+ * must never route through this code.
+ *
+ * It implements the same C-ABI as the GPU engine (include/rrdb_engine.h) as a
+ * plain single-threaded C restatement of the reference algorithms:
+ *   - crc64:        reference src/utils/crc.cpp:289-481 (table regenerated
+ *                   from the polynomial, bit list at crc.cpp:290-295)
+ *   - key codec:    reference src/base/pegasus_key_schema.h:41-183
+ *   - value codec:  reference src/base/pegasus_value_schema.h:58-125,158-226
+ *                   and value_schema_v2.cpp:53-125
+ *   - TTL:          reference src/base/pegasus_value_schema.h:113-125,
+ *                   src/base/pegasus_utils.h:40-41 (epoch = 2016-01-01)
+ *   - compaction filter: reference src/server/key_ttl_compaction_filter.h:55-121
+ *   - user rules/ops:    reference src/server/compaction_filter_rule.cpp:31-90,
+ *                        compaction_operation.cpp:33-113,160-190
+ *   - read handlers:     reference src/server/pegasus_server_impl.cpp:418-1547
+ *                        and :2350-2504 (validation/append helpers)
+ *   - limits:       reference src/server/range_read_limiter.h:37-103 with the
+ *                   defaults of pegasus_server_impl_init.cpp:456-511
+ *   - LSM boundary semantics (rocksdb v8.5.3 compaction iterator / merging
+ *     iterator): bytewise key order; across runs the highest seqno wins;
+ *     DELETE tombstones suppress older versions and are dropped at the
+ *     bottommost level; the compaction filter sees only the surviving newest
+ *     PUT version.
+ *
+ * Parity pinning: crc64 is checked against the reference's own crc.cpp
+ * compiled standalone (oracle/_ref, see Makefile); codec/rule behavior is
+ * checked against KATs lifted from the reference's tests
+ * (src/base/test/value_schema_test.cpp:73-136,
+ *  src/server/test/compaction_filter_rule_test.cpp:32-135) in tests/.
+ */
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+#include <stdio.h>
+#include <time.h>
+
+#include "../include/rrdb_engine.h"
+
+/* ================= crc64 (reference src/utils/crc.cpp) ================= */
+/* Polynomial: reversed-order bit list from crc.cpp:290-295
+ * (BIT64(n) = 1 << (63-n)); table built exactly as InitializeTables
+ * (crc.cpp:236-247); compute loop as crc.cpp:45-88 (byte-at-a-time form —
+ * the 16-way unroll is an identical recurrence). */
+static uint64_t crc64_table[256];
+static int crc64_init_done = 0;
+
+static uint64_t crc64_poly(void)
+{
+    static const int bits[] = {63, 61, 59, 58, 56, 55, 52, 49, 48, 47, 46, 44, 41, 37, 36,
+                               34, 32, 31, 28, 26, 23, 22, 19, 16, 13, 12, 10, 9,  6,  4,
+                               3,  0};
+    uint64_t p = 0;
+    for (size_t i = 0; i < sizeof(bits) / sizeof(bits[0]); i++)
+        p |= 1ull << (63 - bits[i]);
+    return p;
+}
+
+static void crc64_init(void)
+{
+    if (crc64_init_done)
+        return;
+    uint64_t poly = crc64_poly();
+    for (int i = 0; i < 256; i++) {
+        uint64_t k = (uint64_t)i;
+        for (int j = 0; j < 8; j++)
+            k = (k & 1) ? ((k >> 1) ^ poly) : (k >> 1);
+        crc64_table[i] = k;
+    }
+    crc64_init_done = 1;
+}
+
+static uint64_t crc64_calc(const void *ptr, size_t size, uint64_t init_crc)
+{
+    crc64_init();
+    const uint8_t *p = (const uint8_t *)ptr;
+    uint64_t crc = ~init_crc;
+    for (size_t i = 0; i < size; i++)
+        crc = crc64_table[(uint8_t)(crc ^ p[i])] ^ (crc >> 8);
+    return ~crc;
+}
+
+/* exported for tests (pin against oracle/_ref build of the reference crc) */
+uint64_t orc_crc64(const void *ptr, uint64_t size, uint64_t init_crc)
+{
+    return crc64_calc(ptr, (size_t)size, init_crc);
+}
+
+/* ---- test hooks (tests/ only): direct access to codec/rule primitives so
+ * the reference's KATs (value_schema_test.cpp:73-136,
+ * compaction_filter_rule_test.cpp:32-135) can pin them without going through
+ * a whole engine instance. ---- */
+uint32_t orc_extract_expire_ts(uint32_t ver, const uint8_t *val, uint64_t len);
+void orc_update_expire_ts(uint32_t ver, uint8_t *val, uint32_t ts);
+int orc_ts_expired(uint32_t epoch_now, uint32_t expire_ts);
+uint64_t orc_key_hash(const uint8_t *key, uint64_t len);
+int orc_pattern_match(const uint8_t *v, uint64_t vlen, int type, const uint8_t *pat,
+                      uint64_t plen);
+
+/* ================= key / value codec ================= */
+/* key = [u16 BE hash_key_len][hash_key][sort_key]
+ * (reference pegasus_key_schema.h:35-58) */
+static uint16_t key_hklen(const uint8_t *key) { return (uint16_t)((key[0] << 8) | key[1]); }
+
+/* pegasus_key_hash (pegasus_key_schema.h:148-165) */
+static uint64_t key_hash(const uint8_t *key, uint64_t len)
+{
+    uint16_t hklen = key_hklen(key);
+    if (hklen > 0)
+        return crc64_calc(key + 2, hklen, 0);
+    return crc64_calc(key + 2, len - 2, 0);
+}
+
+/* check_pegasus_key_hash (pegasus_key_schema.h:175-183) */
+static int check_key_hash(const uint8_t *key, uint64_t len, int32_t pidx, int32_t pver)
+{
+    return (int64_t)(key_hash(key, len) & (uint64_t)pver) == (int64_t)pidx;
+}
+
+/* value header length by table data version (pegasus_value_schema.h:158,205-209;
+ * value_schema_v2.cpp:56,86-90) */
+static uint32_t value_hdr_len(uint32_t ver) { return ver == 0 ? 4u : (ver == 1 ? 12u : 13u); }
+
+/* pegasus_extract_expire_ts (pegasus_value_schema.h:58-62 for v0/v1;
+ * value_schema_v2.cpp:101-106 for v2): u32 BE at offset 0 (v0/v1) or 1 (v2). */
+static uint32_t extract_expire_ts(uint32_t ver, const uint8_t *val, uint64_t len)
+{
+    uint64_t off = (ver == 2) ? 1 : 0;
+    (void)len;
+    return ((uint32_t)val[off] << 24) | ((uint32_t)val[off + 1] << 16) |
+           ((uint32_t)val[off + 2] << 8) | (uint32_t)val[off + 3];
+}
+
+/* pegasus_update_expire_ts (pegasus_value_schema.h:99-110; value_schema_v2.cpp:116-125) */
+static void update_expire_ts(uint32_t ver, uint8_t *val, uint32_t ts)
+{
+    uint64_t off = (ver == 2) ? 1 : 0;
+    val[off] = (uint8_t)(ts >> 24);
+    val[off + 1] = (uint8_t)(ts >> 16);
+    val[off + 2] = (uint8_t)(ts >> 8);
+    val[off + 3] = (uint8_t)ts;
+}
+
+/* check_if_ts_expired (pegasus_value_schema.h:113-116) */
+static int ts_expired(uint32_t epoch_now, uint32_t expire_ts)
+{
+    return expire_ts > 0 && expire_ts <= epoch_now;
+}
+
+/* ================= memory arena for results ================= */
+typedef struct arena_block {
+    struct arena_block *next;
+    size_t used, cap;
+    uint8_t data[];
+} arena_block;
+
+typedef struct {
+    arena_block *head;
+} arena;
+
+static void *arena_alloc(arena *a, size_t sz)
+{
+    sz = (sz + 15) & ~(size_t)15;
+    if (!a->head || a->head->used + sz > a->head->cap) {
+        size_t cap = sz > (1 << 20) ? sz : (1 << 20);
+        arena_block *b = (arena_block *)malloc(sizeof(arena_block) + cap);
+        b->next = a->head;
+        b->used = 0;
+        b->cap = cap;
+        a->head = b;
+    }
+    void *p = a->head->data + a->head->used;
+    a->head->used += sz;
+    return p;
+}
+
+static void arena_destroy(arena *a)
+{
+    arena_block *b = a->head;
+    while (b) {
+        arena_block *n = b->next;
+        free(b);
+        b = n;
+    }
+    free(a);
+}
+
+void rrdb_free_result(rrdb_result *r)
+{
+    if (r && r->_arena) {
+        arena_destroy((arena *)r->_arena);
+        r->_arena = NULL;
+    }
+}
+
+static arena *result_init(rrdb_result *r)
+{
+    memset(r, 0, sizeof(*r));
+    arena *a = (arena *)calloc(1, sizeof(arena));
+    r->_arena = a;
+    return a;
+}
+
+/* ================= user compaction rules/ops ================= */
+/* restatement of compaction_filter_rule.{h,cpp} + compaction_operation.{h,cpp} */
+
+enum { SMT_MATCH_ANYWHERE = 0, SMT_MATCH_PREFIX, SMT_MATCH_POSTFIX, SMT_INVALID };
+enum { FRT_HASHKEY_PATTERN = 0, FRT_SORTKEY_PATTERN, FRT_TTL_RANGE, FRT_INVALID };
+enum { COT_UPDATE_TTL = 0, COT_DELETE, COT_INVALID };
+enum { UTOT_FROM_NOW = 0, UTOT_FROM_CURRENT, UTOT_TIMESTAMP, UTOT_INVALID };
+
+typedef struct {
+    int type; /* FRT_* */
+    /* pattern rules */
+    char *pattern;
+    uint64_t pattern_len;
+    int match_type; /* SMT_* */
+    /* ttl range rule */
+    uint32_t start_ttl, stop_ttl;
+} orc_rule;
+
+typedef struct {
+    int type; /* COT_* */
+    /* update_ttl params */
+    int ut_type; /* UTOT_* */
+    uint32_t ut_value;
+    orc_rule *rules;
+    int n_rules;
+} orc_op;
+
+typedef struct {
+    orc_op *ops;
+    int n_ops;
+} orc_ops;
+
+static void free_ops(orc_ops *o)
+{
+    for (int i = 0; i < o->n_ops; i++) {
+        for (int j = 0; j < o->ops[i].n_rules; j++)
+            free(o->ops[i].rules[j].pattern);
+        free(o->ops[i].rules);
+    }
+    free(o->ops);
+    o->ops = NULL;
+    o->n_ops = 0;
+}
+
+/* string_pattern_match (compaction_filter_rule.cpp:31-53): empty pattern or
+ * value shorter than pattern -> false. */
+static int string_pattern_match(const uint8_t *v, uint64_t vlen, int type, const char *pat,
+                                uint64_t plen)
+{
+    if (plen == 0 || vlen < plen)
+        return 0;
+    switch (type) {
+    case SMT_MATCH_ANYWHERE: {
+        for (uint64_t i = 0; i + plen <= vlen; i++)
+            if (memcmp(v + i, pat, plen) == 0)
+                return 1;
+        return 0;
+    }
+    case SMT_MATCH_PREFIX:
+        return memcmp(v, pat, plen) == 0;
+    case SMT_MATCH_POSTFIX:
+        return memcmp(v + vlen - plen, pat, plen) == 0;
+    default:
+        return 0;
+    }
+}
+
+/* rule::match (compaction_filter_rule.cpp:57-90) */
+static int rule_match(const orc_rule *r, uint32_t data_version, uint32_t epoch_now,
+                      const uint8_t *hk, uint64_t hklen, const uint8_t *sk, uint64_t sklen,
+                      const uint8_t *val, uint64_t vlen)
+{
+    switch (r->type) {
+    case FRT_HASHKEY_PATTERN:
+        return string_pattern_match(hk, hklen, r->match_type, r->pattern, r->pattern_len);
+    case FRT_SORTKEY_PATTERN:
+        return string_pattern_match(sk, sklen, r->match_type, r->pattern, r->pattern_len);
+    case FRT_TTL_RANGE: {
+        uint32_t expire_ts = extract_expire_ts(data_version, val, vlen);
+        if (expire_ts == 0 && r->start_ttl == 0 && r->stop_ttl == 0)
+            return 1;
+        /* u32 wrapping arithmetic, as the reference's uint32 expressions */
+        if ((uint32_t)(r->start_ttl + epoch_now) <= expire_ts &&
+            (uint32_t)(r->stop_ttl + epoch_now) >= expire_ts)
+            return 1;
+        return 0;
+    }
+    default:
+        return 0;
+    }
+}
+
+/* all_rules_match (compaction_operation.cpp:35-49): empty rules -> false */
+static int all_rules_match(const orc_op *op, uint32_t dv, uint32_t now, const uint8_t *hk,
+                           uint64_t hklen, const uint8_t *sk, uint64_t sklen, const uint8_t *val,
+                           uint64_t vlen)
+{
+    if (op->n_rules == 0)
+        return 0;
+    for (int i = 0; i < op->n_rules; i++)
+        if (!rule_match(&op->rules[i], dv, now, hk, hklen, sk, sklen, val, vlen))
+            return 0;
+    return 1;
+}
+
+/* ---- minimal JSON parser (for user_specified_compaction envs; format per
+ * compaction_operation.cpp:116-190 internal::json_helper) ---- */
+typedef struct {
+    const char *s;
+    const char *end;
+    int ok;
+} jp;
+
+static void jp_ws(jp *p)
+{
+    while (p->s < p->end && (*p->s == ' ' || *p->s == '\t' || *p->s == '\n' || *p->s == '\r'))
+        p->s++;
+}
+static int jp_lit(jp *p, char c)
+{
+    jp_ws(p);
+    if (p->s < p->end && *p->s == c) {
+        p->s++;
+        return 1;
+    }
+    return 0;
+}
+/* parse a JSON string; returns malloc'd unescaped bytes (caller frees) */
+static char *jp_string(jp *p, uint64_t *out_len)
+{
+    jp_ws(p);
+    if (p->s >= p->end || *p->s != '"') {
+        p->ok = 0;
+        return NULL;
+    }
+    p->s++;
+    size_t cap = 32, n = 0;
+    char *buf = (char *)malloc(cap);
+    while (p->s < p->end && *p->s != '"') {
+        char c = *p->s++;
+        if (c == '\\' && p->s < p->end) {
+            char e = *p->s++;
+            switch (e) {
+            case 'n': c = '\n'; break;
+            case 't': c = '\t'; break;
+            case 'r': c = '\r'; break;
+            case 'b': c = '\b'; break;
+            case 'f': c = '\f'; break;
+            case '"': c = '"'; break;
+            case '\\': c = '\\'; break;
+            case '/': c = '/'; break;
+            case 'u': {
+                if (p->end - p->s < 4) { p->ok = 0; free(buf); return NULL; }
+                unsigned v = 0;
+                for (int i = 0; i < 4; i++) {
+                    char h = p->s[i];
+                    v <<= 4;
+                    if (h >= '0' && h <= '9') v |= (unsigned)(h - '0');
+                    else if (h >= 'a' && h <= 'f') v |= (unsigned)(h - 'a' + 10);
+                    else if (h >= 'A' && h <= 'F') v |= (unsigned)(h - 'A' + 10);
+                    else { p->ok = 0; free(buf); return NULL; }
+                }
+                p->s += 4;
+                c = (char)v; /* rules only use byte-range patterns */
+                break;
+            }
+            default: c = e; break;
+            }
+        }
+        if (n + 1 >= cap) {
+            cap *= 2;
+            buf = (char *)realloc(buf, cap);
+        }
+        buf[n++] = c;
+    }
+    if (p->s >= p->end) {
+        p->ok = 0;
+        free(buf);
+        return NULL;
+    }
+    p->s++; /* closing quote */
+    buf[n] = 0;
+    if (out_len)
+        *out_len = n;
+    return buf;
+}
+static long long jp_number(jp *p)
+{
+    jp_ws(p);
+    int neg = 0;
+    if (p->s < p->end && *p->s == '-') {
+        neg = 1;
+        p->s++;
+    }
+    if (p->s >= p->end || *p->s < '0' || *p->s > '9') {
+        p->ok = 0;
+        return 0;
+    }
+    long long v = 0;
+    while (p->s < p->end && *p->s >= '0' && *p->s <= '9')
+        v = v * 10 + (*p->s++ - '0');
+    return neg ? -v : v;
+}
+/* skip any JSON value */
+static void jp_skip(jp *p)
+{
+    jp_ws(p);
+    if (p->s >= p->end) { p->ok = 0; return; }
+    char c = *p->s;
+    if (c == '"') {
+        uint64_t l;
+        char *s = jp_string(p, &l);
+        free(s);
+    } else if (c == '{') {
+        p->s++;
+        jp_ws(p);
+        if (jp_lit(p, '}')) return;
+        for (;;) {
+            uint64_t l;
+            char *k = jp_string(p, &l);
+            free(k);
+            if (!p->ok || !jp_lit(p, ':')) { p->ok = 0; return; }
+            jp_skip(p);
+            if (!p->ok) return;
+            if (jp_lit(p, '}')) return;
+            if (!jp_lit(p, ',')) { p->ok = 0; return; }
+        }
+    } else if (c == '[') {
+        p->s++;
+        jp_ws(p);
+        if (jp_lit(p, ']')) return;
+        for (;;) {
+            jp_skip(p);
+            if (!p->ok) return;
+            if (jp_lit(p, ']')) return;
+            if (!jp_lit(p, ',')) { p->ok = 0; return; }
+        }
+    } else if (c == 't') { p->s += (p->end - p->s >= 4) ? 4 : 0; }
+    else if (c == 'f') { p->s += (p->end - p->s >= 5) ? 5 : 0; }
+    else if (c == 'n') { p->s += (p->end - p->s >= 4) ? 4 : 0; }
+    else { jp_number(p); }
+}
+
+static int enum_smt(const char *s)
+{
+    if (s && strcmp(s, "SMT_MATCH_ANYWHERE") == 0) return SMT_MATCH_ANYWHERE;
+    if (s && strcmp(s, "SMT_MATCH_PREFIX") == 0) return SMT_MATCH_PREFIX;
+    if (s && strcmp(s, "SMT_MATCH_POSTFIX") == 0) return SMT_MATCH_POSTFIX;
+    return SMT_INVALID;
+}
+static int enum_frt(const char *s)
+{
+    if (s && strcmp(s, "FRT_HASHKEY_PATTERN") == 0) return FRT_HASHKEY_PATTERN;
+    if (s && strcmp(s, "FRT_SORTKEY_PATTERN") == 0) return FRT_SORTKEY_PATTERN;
+    if (s && strcmp(s, "FRT_TTL_RANGE") == 0) return FRT_TTL_RANGE;
+    return FRT_INVALID;
+}
+static int enum_cot(const char *s)
+{
+    if (s && strcmp(s, "COT_UPDATE_TTL") == 0) return COT_UPDATE_TTL;
+    if (s && strcmp(s, "COT_DELETE") == 0) return COT_DELETE;
+    return COT_INVALID;
+}
+static int enum_utot(const char *s)
+{
+    if (s && strcmp(s, "UTOT_FROM_NOW") == 0) return UTOT_FROM_NOW;
+    if (s && strcmp(s, "UTOT_FROM_CURRENT") == 0) return UTOT_FROM_CURRENT;
+    if (s && strcmp(s, "UTOT_TIMESTAMP") == 0) return UTOT_TIMESTAMP;
+    return UTOT_INVALID;
+}
+
+/* parse rule `params` JSON: {"pattern":...,"match_type":...} or
+ * {"start_ttl":N,"stop_ttl":N}.  All listed fields required (json_forwarder
+ * decode fails on missing fields — compaction_operation_test.cpp:243-248). */
+static int parse_rule_params(orc_rule *r, const char *json, uint64_t len)
+{
+    jp p = {json, json + len, 1};
+    int have_pattern = 0, have_mt = 0, have_start = 0, have_stop = 0;
+    if (!jp_lit(&p, '{'))
+        return 0;
+    if (!jp_lit(&p, '}')) {
+        for (;;) {
+            uint64_t kl;
+            char *k = jp_string(&p, &kl);
+            if (!p.ok || !jp_lit(&p, ':')) {
+                free(k);
+                return 0;
+            }
+            if (k && strcmp(k, "pattern") == 0) {
+                r->pattern = jp_string(&p, &r->pattern_len);
+                have_pattern = p.ok;
+            } else if (k && strcmp(k, "match_type") == 0) {
+                uint64_t ml;
+                char *m = jp_string(&p, &ml);
+                if (p.ok) {
+                    r->match_type = enum_smt(m);
+                    have_mt = 1;
+                }
+                free(m);
+            } else if (k && strcmp(k, "start_ttl") == 0) {
+                r->start_ttl = (uint32_t)jp_number(&p);
+                have_start = p.ok;
+            } else if (k && strcmp(k, "stop_ttl") == 0) {
+                r->stop_ttl = (uint32_t)jp_number(&p);
+                have_stop = p.ok;
+            } else {
+                jp_skip(&p);
+            }
+            free(k);
+            if (!p.ok)
+                return 0;
+            if (jp_lit(&p, '}'))
+                break;
+            if (!jp_lit(&p, ','))
+                return 0;
+        }
+    }
+    if (r->type == FRT_TTL_RANGE)
+        return have_start && have_stop;
+    return have_pattern && have_mt;
+}
+
+static int parse_op_params(orc_op *op, const char *json, uint64_t len)
+{
+    if (op->type == COT_DELETE)
+        return 1; /* delete_key::creator ignores params (compaction_operation.h:99-102) */
+    jp p = {json, json + len, 1};
+    int have_type = 0, have_value = 0;
+    if (!jp_lit(&p, '{'))
+        return 0;
+    if (!jp_lit(&p, '}')) {
+        for (;;) {
+            uint64_t kl;
+            char *k = jp_string(&p, &kl);
+            if (!p.ok || !jp_lit(&p, ':')) {
+                free(k);
+                return 0;
+            }
+            if (k && strcmp(k, "type") == 0) {
+                uint64_t tl;
+                char *t = jp_string(&p, &tl);
+                if (p.ok) {
+                    op->ut_type = enum_utot(t);
+                    have_type = 1;
+                }
+                free(t);
+            } else if (k && strcmp(k, "value") == 0) {
+                op->ut_value = (uint32_t)jp_number(&p);
+                have_value = p.ok;
+            } else {
+                jp_skip(&p);
+            }
+            free(k);
+            if (!p.ok)
+                return 0;
+            if (jp_lit(&p, '}'))
+                break;
+            if (!jp_lit(&p, ','))
+                return 0;
+        }
+    }
+    return have_type && have_value;
+}
+
+/* create_compaction_operations (compaction_operation.cpp:160-190): invalid
+ * rules skipped; op kept only when >=1 rule valid; invalid ops skipped. */
+static int parse_user_ops(orc_ops *out, const char *json, uint64_t len)
+{
+    memset(out, 0, sizeof(*out));
+    jp p = {json, json + len, 1};
+    if (!jp_lit(&p, '{'))
+        return 0;
+    int cap = 4;
+    out->ops = (orc_op *)calloc((size_t)cap, sizeof(orc_op));
+    int done = 0;
+    while (!done) {
+        uint64_t kl;
+        char *k = jp_string(&p, &kl);
+        if (!p.ok || !jp_lit(&p, ':')) {
+            free(k);
+            free_ops(out);
+            return 0;
+        }
+        if (k && strcmp(k, "ops") == 0) {
+            if (!jp_lit(&p, '[')) {
+                free(k);
+                free_ops(out);
+                return 0;
+            }
+            if (!jp_lit(&p, ']')) {
+                for (;;) {
+                    /* one op object */
+                    orc_op op;
+                    memset(&op, 0, sizeof(op));
+                    op.type = COT_INVALID;
+                    op.ut_type = UTOT_INVALID;
+                    char *op_params = NULL;
+                    uint64_t op_params_len = 0;
+                    int rcap = 4;
+                    op.rules = (orc_rule *)calloc((size_t)rcap, sizeof(orc_rule));
+                    if (!jp_lit(&p, '{')) {
+                        free(op.rules);
+                        free(k);
+                        free_ops(out);
+                        return 0;
+                    }
+                    for (;;) {
+                        uint64_t fkl;
+                        char *fk = jp_string(&p, &fkl);
+                        if (!p.ok || !jp_lit(&p, ':')) {
+                            free(fk);
+                            break;
+                        }
+                        if (fk && strcmp(fk, "type") == 0) {
+                            uint64_t tl;
+                            char *t = jp_string(&p, &tl);
+                            if (p.ok)
+                                op.type = enum_cot(t);
+                            free(t);
+                        } else if (fk && strcmp(fk, "params") == 0) {
+                            free(op_params);
+                            op_params = jp_string(&p, &op_params_len);
+                        } else if (fk && strcmp(fk, "rules") == 0) {
+                            if (!jp_lit(&p, '[')) {
+                                p.ok = 0;
+                                free(fk);
+                                break;
+                            }
+                            if (!jp_lit(&p, ']')) {
+                                for (;;) {
+                                    orc_rule r;
+                                    memset(&r, 0, sizeof(r));
+                                    r.type = FRT_INVALID;
+                                    r.match_type = SMT_INVALID;
+                                    char *r_params = NULL;
+                                    uint64_t r_params_len = 0;
+                                    if (!jp_lit(&p, '{')) {
+                                        p.ok = 0;
+                                        break;
+                                    }
+                                    for (;;) {
+                                        uint64_t rkl;
+                                        char *rk = jp_string(&p, &rkl);
+                                        if (!p.ok || !jp_lit(&p, ':')) {
+                                            free(rk);
+                                            break;
+                                        }
+                                        if (rk && strcmp(rk, "type") == 0) {
+                                            uint64_t tl;
+                                            char *t = jp_string(&p, &tl);
+                                            if (p.ok)
+                                                r.type = enum_frt(t);
+                                            free(t);
+                                        } else if (rk && strcmp(rk, "params") == 0) {
+                                            free(r_params);
+                                            r_params = jp_string(&p, &r_params_len);
+                                        } else {
+                                            jp_skip(&p);
+                                        }
+                                        free(rk);
+                                        if (!p.ok)
+                                            break;
+                                        if (jp_lit(&p, '}'))
+                                            break;
+                                        if (!jp_lit(&p, ',')) {
+                                            p.ok = 0;
+                                            break;
+                                        }
+                                    }
+                                    /* rule valid? (factory create + decode) */
+                                    if (p.ok && r.type != FRT_INVALID && r_params &&
+                                        parse_rule_params(&r, r_params, r_params_len)) {
+                                        if (op.n_rules == rcap) {
+                                            rcap *= 2;
+                                            op.rules = (orc_rule *)realloc(
+                                                op.rules, (size_t)rcap * sizeof(orc_rule));
+                                        }
+                                        op.rules[op.n_rules++] = r;
+                                    } else {
+                                        free(r.pattern);
+                                    }
+                                    free(r_params);
+                                    if (!p.ok)
+                                        break;
+                                    if (jp_lit(&p, ']'))
+                                        break;
+                                    if (!jp_lit(&p, ',')) {
+                                        p.ok = 0;
+                                        break;
+                                    }
+                                }
+                            }
+                        } else {
+                            jp_skip(&p);
+                        }
+                        free(fk);
+                        if (!p.ok)
+                            break;
+                        if (jp_lit(&p, '}'))
+                            break;
+                        if (!jp_lit(&p, ',')) {
+                            p.ok = 0;
+                            break;
+                        }
+                    }
+                    /* op valid? rules non-empty + known type + params decode */
+                    int op_ok = p.ok && op.n_rules > 0 && op.type != COT_INVALID && op_params &&
+                                parse_op_params(&op, op_params, op_params_len);
+                    if (op_ok) {
+                        if (out->n_ops == cap) {
+                            cap *= 2;
+                            out->ops = (orc_op *)realloc(out->ops, (size_t)cap * sizeof(orc_op));
+                        }
+                        out->ops[out->n_ops++] = op;
+                    } else {
+                        for (int j = 0; j < op.n_rules; j++)
+                            free(op.rules[j].pattern);
+                        free(op.rules);
+                    }
+                    free(op_params);
+                    if (!p.ok) {
+                        free(k);
+                        free_ops(out);
+                        return 0;
+                    }
+                    if (jp_lit(&p, ']'))
+                        break;
+                    if (!jp_lit(&p, ',')) {
+                        free(k);
+                        free_ops(out);
+                        return 0;
+                    }
+                }
+            }
+        } else {
+            jp_skip(&p);
+        }
+        free(k);
+        if (!p.ok) {
+            free_ops(out);
+            return 0;
+        }
+        if (jp_lit(&p, '}'))
+            done = 1;
+        else if (!jp_lit(&p, ',')) {
+            free_ops(out);
+            return 0;
+        }
+    }
+    return 1;
+}
+
+/* ================= engine state ================= */
+typedef struct {
+    uint64_t n;
+    uint8_t *keys;
+    uint64_t *koff;
+    uint8_t *vals;
+    uint64_t *voff;
+    uint64_t *sk; /* (seqno<<1)|kind */
+    uint64_t min_seq, max_seq;
+} Run;
+
+typedef struct ScanCtx ScanCtx;
+
+typedef struct {
+    int32_t app_id, pidx, partition_version;
+    int validate_hash; /* replica.split.validate_partition_hash env */
+    uint32_t data_version;
+    uint32_t default_ttl;
+    int manual_compact_disabled;
+    uint32_t max_iter_count;        /* rocksdb_max_iteration_count, default 1000 */
+    uint32_t mg_max_iter_count;     /* rocksdb_multi_get_max_iteration_count, 3000 */
+    uint64_t mg_max_iter_size;      /* rocksdb_multi_get_max_iteration_size, 30<<20 */
+    uint64_t iter_time_ms;          /* rocksdb_iteration_threshold_time_ms, 30000 */
+    orc_ops user_ops;
+    Run *runs;
+    int n_runs, runs_cap;
+    uint64_t next_seq_floor; /* ingest seqno monotonicity check */
+    ScanCtx **ctxs;
+    int n_ctxs, ctxs_cap;
+    int64_t next_ctx_id;
+} Engine;
+
+/* a parked scanner: materialized visible view of [cursor..stop) */
+struct ScanCtx {
+    int64_t id;
+    uint64_t *view; /* packed (run<<40|idx), key-ordered visible records */
+    uint64_t view_n;
+    uint64_t cursor; /* next view position */
+    uint8_t *stop;
+    uint64_t stop_len;
+    uint8_t stop_inclusive;
+    int32_t batch_size;
+    uint8_t no_value, validate_hash_req, return_expire_ts, only_return_count;
+    int32_t hk_ft, sk_ft;
+    uint8_t *hk_pat, *sk_pat;
+    uint64_t hk_pat_len, sk_pat_len;
+};
+
+const char *rrdb_backend(void) { return "oracle-cpu"; }
+
+void *rrdb_open(int32_t app_id, int32_t pidx, int32_t gpu_id)
+{
+    (void)gpu_id; /* oracle is CPU-only; gpu_id ignored */
+    Engine *e = (Engine *)calloc(1, sizeof(Engine));
+    e->app_id = app_id;
+    e->pidx = pidx;
+    e->partition_version = -1;
+    e->data_version = 1;
+    e->max_iter_count = 1000;
+    e->mg_max_iter_count = 3000;
+    e->mg_max_iter_size = 30ull << 20;
+    e->iter_time_ms = 30000;
+    return e;
+}
+
+static void free_run(Run *r)
+{
+    free(r->keys);
+    free(r->koff);
+    free(r->vals);
+    free(r->voff);
+    free(r->sk);
+}
+
+static void free_ctx(ScanCtx *c)
+{
+    if (!c)
+        return;
+    free(c->view);
+    free(c->stop);
+    free(c->hk_pat);
+    free(c->sk_pat);
+    free(c);
+}
+
+void rrdb_close(void *h)
+{
+    Engine *e = (Engine *)h;
+    if (!e)
+        return;
+    for (int i = 0; i < e->n_runs; i++)
+        free_run(&e->runs[i]);
+    free(e->runs);
+    for (int i = 0; i < e->n_ctxs; i++)
+        free_ctx(e->ctxs[i]);
+    free(e->ctxs);
+    free_ops(&e->user_ops);
+    free(e);
+}
+
+int32_t rrdb_set_partition_version(void *h, int32_t pv)
+{
+    ((Engine *)h)->partition_version = pv;
+    return RRDB_OK;
+}
+
+int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *values, int32_t n)
+{
+    Engine *e = (Engine *)h;
+    for (int32_t i = 0; i < n; i++) {
+        const char *k = keys[i], *v = values[i];
+        if (strcmp(k, "default_ttl") == 0) {
+            long long t = atoll(v);
+            e->default_ttl = t > 0 ? (uint32_t)t : 0;
+        } else if (strcmp(k, "user_specified_compaction") == 0) {
+            orc_ops ops;
+            if (parse_user_ops(&ops, v, strlen(v))) {
+                free_ops(&e->user_ops);
+                e->user_ops = ops;
+            } else {
+                /* invalid json -> empty ops (compaction_operation.cpp:165-169) */
+                free_ops(&e->user_ops);
+            }
+        } else if (strcmp(k, "replica.split.validate_partition_hash") == 0) {
+            e->validate_hash = strcmp(v, "true") == 0;
+        } else if (strcmp(k, "manual_compact.disabled") == 0) {
+            e->manual_compact_disabled = strcmp(v, "true") == 0;
+        } else if (strcmp(k, "pegasus.data_version") == 0) {
+            e->data_version = (uint32_t)atoi(v);
+        } else if (strcmp(k, "replica.rocksdb_iteration_threshold_time_ms") == 0) {
+            e->iter_time_ms = (uint64_t)atoll(v);
+        } else if (strcmp(k, "rocksdb.max_iteration_count") == 0) {
+            e->max_iter_count = (uint32_t)atoll(v);
+        } else if (strcmp(k, "rocksdb.multi_get_max_iteration_count") == 0) {
+            e->mg_max_iter_count = (uint32_t)atoll(v);
+        } else if (strcmp(k, "rocksdb.multi_get_max_iteration_size") == 0) {
+            e->mg_max_iter_size = (uint64_t)atoll(v);
+        } /* unknown env keys ignored (update_app_envs behavior) */
+    }
+    return RRDB_OK;
+}
+
+static int key_cmp(const uint8_t *a, uint64_t alen, const uint8_t *b, uint64_t blen)
+{
+    uint64_t m = alen < blen ? alen : blen;
+    int c = m ? memcmp(a, b, m) : 0;
+    if (c)
+        return c;
+    return alen < blen ? -1 : (alen > blen ? 1 : 0);
+}
+
+static const uint8_t *run_key(const Run *r, uint64_t i, uint64_t *len)
+{
+    *len = r->koff[i + 1] - r->koff[i];
+    return r->keys + r->koff[i];
+}
+static const uint8_t *run_val(const Run *r, uint64_t i, uint64_t *len)
+{
+    *len = r->voff[i + 1] - r->voff[i];
+    return r->vals + r->voff[i];
+}
+
+int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
+                        const uint8_t *values, const uint64_t *val_offs, const uint64_t *seq_kind,
+                        uint64_t n)
+{
+    Engine *e = (Engine *)h;
+    if (n == 0)
+        return RRDB_OK;
+    /* validate: strictly increasing keys, seqnos above everything ingested */
+    for (uint64_t i = 0; i < n; i++) {
+        if ((seq_kind[i] >> 1) < e->next_seq_floor)
+            return RRDB_INVALID_ARGUMENT;
+        if (i > 0) {
+            int c = key_cmp(keys + key_offs[i - 1], key_offs[i] - key_offs[i - 1],
+                            keys + key_offs[i], key_offs[i + 1] - key_offs[i]);
+            if (c >= 0)
+                return RRDB_INVALID_ARGUMENT;
+        }
+    }
+    Run r;
+    r.n = n;
+    uint64_t kb = key_offs[n], vb = val_offs[n];
+    r.keys = (uint8_t *)malloc(kb ? kb : 1);
+    memcpy(r.keys, keys, kb);
+    r.koff = (uint64_t *)malloc((n + 1) * 8);
+    memcpy(r.koff, key_offs, (n + 1) * 8);
+    r.vals = (uint8_t *)malloc(vb ? vb : 1);
+    memcpy(r.vals, values, vb);
+    r.voff = (uint64_t *)malloc((n + 1) * 8);
+    memcpy(r.voff, val_offs, (n + 1) * 8);
+    r.sk = (uint64_t *)malloc(n * 8);
+    memcpy(r.sk, seq_kind, n * 8);
+    r.min_seq = ~0ull;
+    r.max_seq = 0;
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t s = seq_kind[i] >> 1;
+        if (s < r.min_seq)
+            r.min_seq = s;
+        if (s > r.max_seq)
+            r.max_seq = s;
+    }
+    e->next_seq_floor = r.max_seq + 1;
+    if (e->n_runs == e->runs_cap) {
+        e->runs_cap = e->runs_cap ? e->runs_cap * 2 : 4;
+        e->runs = (Run *)realloc(e->runs, (size_t)e->runs_cap * sizeof(Run));
+    }
+    e->runs[e->n_runs++] = r; /* appended: runs[i] older than runs[i+1] */
+    return RRDB_OK;
+}
+
+uint64_t rrdb_num_runs(void *h) { return (uint64_t)((Engine *)h)->n_runs; }
+uint64_t rrdb_num_records(void *h)
+{
+    Engine *e = (Engine *)h;
+    uint64_t t = 0;
+    for (int i = 0; i < e->n_runs; i++)
+        t += e->runs[i].n;
+    return t;
+}
+
+/* lower_bound: first index in run with key >= target */
+static uint64_t run_lower_bound(const Run *r, const uint8_t *key, uint64_t klen)
+{
+    uint64_t lo = 0, hi = r->n;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2, ml;
+        const uint8_t *mk = run_key(r, mid, &ml);
+        if (key_cmp(mk, ml, key, klen) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+__attribute__((unused)) static uint64_t run_upper_bound(const Run *r, const uint8_t *key,
+                                                        uint64_t klen)
+{
+    uint64_t lo = 0, hi = r->n;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2, ml;
+        const uint8_t *mk = run_key(r, mid, &ml);
+        if (key_cmp(mk, ml, key, klen) <= 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+
+/* newest version of a key across runs: returns run idx or -1; *rec_idx out */
+static int find_newest(Engine *e, const uint8_t *key, uint64_t klen, uint64_t *rec_idx)
+{
+    int best = -1;
+    uint64_t best_seq = 0, bi = 0;
+    for (int r = 0; r < e->n_runs; r++) {
+        const Run *run = &e->runs[r];
+        uint64_t i = run_lower_bound(run, key, klen);
+        if (i < run->n) {
+            uint64_t kl;
+            const uint8_t *k = run_key(run, i, &kl);
+            if (key_cmp(k, kl, key, klen) == 0) {
+                uint64_t seq = run->sk[i] >> 1;
+                if (best < 0 || seq > best_seq) {
+                    best = r;
+                    best_seq = seq;
+                    bi = i;
+                }
+            }
+        }
+    }
+    *rec_idx = bi;
+    return best;
+}
+
+/* ---- merged visible view (merging-iterator equivalent) ----
+ * Builds the key-ordered list of visible records (newest version per key,
+ * tombstones excluded) intersected with [start,stop) bounds given as raw
+ * keys; bounds NULL = unbounded.  Simple k-way pointer merge. */
+static uint64_t *build_view(Engine *e, const uint8_t *start, uint64_t start_len,
+                            const uint8_t *stop, uint64_t stop_len, uint64_t *out_n)
+{
+    int R = e->n_runs;
+    uint64_t *pos = (uint64_t *)malloc((size_t)(R ? R : 1) * 8);
+    uint64_t *end = (uint64_t *)malloc((size_t)(R ? R : 1) * 8);
+    uint64_t total = 0;
+    for (int r = 0; r < R; r++) {
+        pos[r] = start ? run_lower_bound(&e->runs[r], start, start_len) : 0;
+        end[r] = stop ? run_lower_bound(&e->runs[r], stop, stop_len) : e->runs[r].n;
+        if (end[r] < pos[r])
+            end[r] = pos[r];
+        total += end[r] - pos[r];
+    }
+    uint64_t *view = (uint64_t *)malloc((size_t)(total ? total : 1) * 8);
+    uint64_t n = 0;
+    for (;;) {
+        /* pick smallest key; among equal keys the highest seqno wins and all
+         * equal-key cursors advance (newest-wins + shadowing) */
+        int best = -1;
+        uint64_t bl = 0;
+        const uint8_t *bk = NULL;
+        for (int r = 0; r < R; r++) {
+            if (pos[r] >= end[r])
+                continue;
+            uint64_t kl;
+            const uint8_t *k = run_key(&e->runs[r], pos[r], &kl);
+            if (best < 0 || key_cmp(k, kl, bk, bl) < 0) {
+                best = r;
+                bk = k;
+                bl = kl;
+            }
+        }
+        if (best < 0)
+            break;
+        uint64_t win_seq = 0;
+        int win_run = -1;
+        uint64_t win_idx = 0;
+        for (int r = 0; r < R; r++) {
+            if (pos[r] >= end[r])
+                continue;
+            uint64_t kl;
+            const uint8_t *k = run_key(&e->runs[r], pos[r], &kl);
+            if (key_cmp(k, kl, bk, bl) == 0) {
+                uint64_t seq = e->runs[r].sk[pos[r]] >> 1;
+                if (win_run < 0 || seq > win_seq) {
+                    win_seq = seq;
+                    win_run = r;
+                    win_idx = pos[r];
+                }
+                pos[r]++;
+            }
+        }
+        if ((e->runs[win_run].sk[win_idx] & 1) == RRDB_KIND_PUT)
+            view[n++] = ((uint64_t)win_run << 40) | win_idx;
+    }
+    free(pos);
+    free(end);
+    *out_n = n;
+    return view;
+}
+
+/* validate_filter (pegasus_server_impl.cpp:2350-2380): NO_FILTER or empty
+ * pattern -> true; shorter value -> false. */
+static int validate_filter(int32_t ft, const uint8_t *pat, uint64_t plen, const uint8_t *v,
+                           uint64_t vlen)
+{
+    if (ft == RRDB_FT_NO_FILTER || plen == 0)
+        return 1;
+    if (vlen < plen)
+        return 0;
+    if (ft == RRDB_FT_MATCH_ANYWHERE) {
+        for (uint64_t i = 0; i + plen <= vlen; i++)
+            if (memcmp(v + i, pat, plen) == 0)
+                return 1;
+        return 0;
+    }
+    if (ft == RRDB_FT_MATCH_PREFIX)
+        return memcmp(v, pat, plen) == 0;
+    return memcmp(v + vlen - plen, pat, plen) == 0; /* POSTFIX */
+}
+
+/* ================= read handlers ================= */
+
+int32_t rrdb_get(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_now,
+                 rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    arena *a = result_init(out);
+    (void)a;
+    uint64_t idx;
+    int r = find_newest(e, key, key_len, &idx);
+    if (r < 0 || (e->runs[r].sk[idx] & 1) == RRDB_KIND_DELETE) {
+        out->error = RRDB_NOT_FOUND;
+        return out->error;
+    }
+    uint64_t vl;
+    const uint8_t *v = run_val(&e->runs[r], idx, &vl);
+    if (ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl))) {
+        out->error = RRDB_NOT_FOUND; /* on_get:444 expired -> NotFound */
+        return out->error;
+    }
+    uint32_t hdr = value_hdr_len(e->data_version);
+    out->count = 1;
+    out->keys = (rrdb_slice *)arena_alloc((arena *)out->_arena, sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)arena_alloc((arena *)out->_arena, sizeof(rrdb_slice));
+    out->keys[0].data = NULL;
+    out->keys[0].len = 0;
+    uint64_t ul = vl - hdr;
+    out->values[0].data = (uint8_t *)arena_alloc((arena *)out->_arena, ul ? ul : 1);
+    memcpy(out->values[0].data, v + hdr, ul);
+    out->values[0].len = ul;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_now,
+                 rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    result_init(out);
+    uint64_t idx;
+    int r = find_newest(e, key, key_len, &idx);
+    if (r < 0 || (e->runs[r].sk[idx] & 1) == RRDB_KIND_DELETE) {
+        out->error = RRDB_NOT_FOUND;
+        return out->error;
+    }
+    uint64_t vl;
+    const uint8_t *v = run_val(&e->runs[r], idx, &vl);
+    uint32_t expire_ts = extract_expire_ts(e->data_version, v, vl);
+    if (ts_expired(epoch_now, expire_ts)) {
+        out->error = RRDB_NOT_FOUND;
+        return out->error;
+    }
+    /* on_ttl:1135-1142 */
+    out->i64 = expire_ts > 0 ? (int64_t)expire_ts - (int64_t)epoch_now : -1;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+int32_t rrdb_batch_get(void *h, uint64_t n_keys, const uint8_t *keys, const uint64_t *key_offs,
+                       uint32_t epoch_now, rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    arena *a = result_init(out);
+    if (n_keys == 0) {
+        out->error = RRDB_INVALID_ARGUMENT; /* on_batch_get:922-928 */
+        return out->error;
+    }
+    uint32_t hdr = value_hdr_len(e->data_version);
+    out->keys = (rrdb_slice *)arena_alloc(a, n_keys * sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)arena_alloc(a, n_keys * sizeof(rrdb_slice));
+    uint64_t m = 0;
+    for (uint64_t i = 0; i < n_keys; i++) {
+        const uint8_t *k = keys + key_offs[i];
+        uint64_t kl = key_offs[i + 1] - key_offs[i];
+        uint64_t idx;
+        int r = find_newest(e, k, kl, &idx);
+        if (r < 0 || (e->runs[r].sk[idx] & 1) == RRDB_KIND_DELETE)
+            continue; /* NotFound skipped (on_batch_get:952-954) */
+        uint64_t vl;
+        const uint8_t *v = run_val(&e->runs[r], idx, &vl);
+        if (ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl)))
+            continue; /* expired skipped (on_batch_get:961-965) */
+        out->keys[m].data = (uint8_t *)arena_alloc(a, kl ? kl : 1);
+        memcpy(out->keys[m].data, k, kl);
+        out->keys[m].len = kl;
+        uint64_t ul = vl - hdr;
+        out->values[m].data = (uint8_t *)arena_alloc(a, ul ? ul : 1);
+        memcpy(out->values[m].data, v + hdr, ul);
+        out->values[m].len = ul;
+        m++;
+    }
+    out->count = m;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uint32_t epoch_now,
+                           rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    result_init(out);
+    /* start = generate_key(hk,""), stop = generate_next_blob(hk)
+     * (on_sortkey_count:1030-1036) */
+    uint64_t sl = 2 + hklen;
+    uint8_t *start = (uint8_t *)malloc(sl);
+    start[0] = (uint8_t)(hklen >> 8);
+    start[1] = (uint8_t)hklen;
+    memcpy(start + 2, hash_key, hklen);
+    uint8_t *stop = (uint8_t *)malloc(sl);
+    memcpy(stop, start, sl);
+    uint64_t stop_len = sl;
+    {
+        /* pegasus_generate_next_blob (pegasus_key_schema.h:64-81) */
+        uint64_t p = stop_len - 1;
+        while (stop[p] == 0xFF)
+            p--;
+        stop[p]++;
+        stop_len = p + 1;
+    }
+    uint64_t n;
+    uint64_t *view = build_view(e, start, sl, stop, stop_len, &n);
+    int64_t count = 0;
+    /* limiter: max_iteration_count(=max_iter_count) is NOT applied in
+     * on_sortkey_count's loop (only time_check); we mirror that: all visible
+     * records iterate (on_sortkey_count:1047-1060) */
+    for (uint64_t i = 0; i < n; i++) {
+        const Run *r = &e->runs[view[i] >> 40];
+        uint64_t idx = view[i] & 0xFFFFFFFFFFull, vl;
+        const uint8_t *v = run_val(r, idx, &vl);
+        if (!ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl)))
+            count++;
+    }
+    free(view);
+    free(start);
+    free(stop);
+    out->i64 = count;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+/* build raw key from hashkey+sortkey into malloc'd buffer */
+static uint8_t *make_key(const uint8_t *hk, uint64_t hklen, const uint8_t *sk, uint64_t sklen,
+                         uint64_t *out_len)
+{
+    uint8_t *k = (uint8_t *)malloc(2 + hklen + sklen + 1);
+    k[0] = (uint8_t)(hklen >> 8);
+    k[1] = (uint8_t)hklen;
+    memcpy(k + 2, hk, hklen);
+    memcpy(k + 2 + hklen, sk, sklen);
+    *out_len = 2 + hklen + sklen;
+    return k;
+}
+
+/* pegasus_generate_next_blob in place: returns new length */
+static uint64_t next_blob(uint8_t *key, uint64_t len)
+{
+    uint64_t p = len - 1;
+    while (key[p] == 0xFF)
+        p--;
+    key[p]++;
+    return p + 1;
+}
+
+int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
+                       rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    arena *a = result_init(out);
+    uint32_t hdr = value_hdr_len(e->data_version);
+
+    if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+        out->error = RRDB_INVALID_ARGUMENT; /* on_multi_get:508-517 */
+        return out->error;
+    }
+
+    /* caps (on_multi_get:521-533) */
+    uint32_t max_kv_count = e->mg_max_iter_count;
+    if (q->max_kv_count > 0 && (uint32_t)q->max_kv_count < max_kv_count)
+        max_kv_count = (uint32_t)q->max_kv_count;
+    int64_t max_kv_size = q->max_kv_size > 0 ? q->max_kv_size : INT32_MAX;
+    int64_t max_iter_size_cfg = e->mg_max_iter_size > 0 ? (int64_t)e->mg_max_iter_size : INT32_MAX;
+    int64_t max_iteration_size = max_kv_size < max_iter_size_cfg ? max_kv_size : max_iter_size_cfg;
+    uint32_t max_iteration_count = e->mg_max_iter_count;
+
+    if (q->n_sort_keys > 0) {
+        /* point-list variant (on_multi_get:779-860) */
+        out->keys = (rrdb_slice *)arena_alloc(a, q->n_sort_keys * sizeof(rrdb_slice));
+        out->values = (rrdb_slice *)arena_alloc(a, q->n_sort_keys * sizeof(rrdb_slice));
+        uint64_t m = 0;
+        int64_t count = 0, size = 0;
+        int exceed = 0;
+        for (uint64_t i = 0; i < q->n_sort_keys; i++) {
+            const uint8_t *sk = q->sort_keys + q->sort_key_offs[i];
+            uint64_t sklen = q->sort_key_offs[i + 1] - q->sort_key_offs[i];
+            uint64_t kl;
+            uint8_t *k = make_key(q->hash_key.data, q->hash_key.len, sk, sklen, &kl);
+            uint64_t idx;
+            int r = find_newest(e, k, kl, &idx);
+            free(k);
+            if (r < 0 || (e->runs[r].sk[idx] & 1) == RRDB_KIND_DELETE)
+                continue; /* NotFound -> continue (:816-818) */
+            uint64_t vl;
+            const uint8_t *v = run_val(&e->runs[r], idx, &vl);
+            if (ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl)))
+                continue; /* :829-834 */
+            if (count >= (int64_t)max_kv_count || size >= max_kv_size) {
+                exceed = 1; /* :837-841 */
+                break;
+            }
+            out->keys[m].data = (uint8_t *)arena_alloc(a, sklen ? sklen : 1);
+            memcpy(out->keys[m].data, sk, sklen);
+            out->keys[m].len = sklen;
+            if (!q->no_value) {
+                uint64_t ul = vl - hdr;
+                out->values[m].data = (uint8_t *)arena_alloc(a, ul ? ul : 1);
+                memcpy(out->values[m].data, v + hdr, ul);
+                out->values[m].len = ul;
+            } else {
+                out->values[m].data = NULL;
+                out->values[m].len = 0;
+            }
+            count++;
+            size += (int64_t)out->keys[m].len + (int64_t)out->values[m].len;
+            m++;
+        }
+        out->count = m;
+        out->error = exceed ? RRDB_INCOMPLETE : RRDB_OK;
+        return out->error;
+    }
+
+    /* range variant (on_multi_get:540-778) */
+    uint64_t start_len, stop_len;
+    uint8_t *start = make_key(q->hash_key.data, q->hash_key.len, q->start_sortkey.data,
+                              q->start_sortkey.len, &start_len);
+    uint8_t *stop;
+    uint8_t start_inclusive = q->start_inclusive, stop_inclusive;
+    if (q->stop_sortkey.len == 0) {
+        uint64_t l;
+        stop = make_key(q->hash_key.data, q->hash_key.len, NULL, 0, &l);
+        stop_len = next_blob(stop, l);
+        stop_inclusive = 0;
+    } else {
+        stop = make_key(q->hash_key.data, q->hash_key.len, q->stop_sortkey.data,
+                        q->stop_sortkey.len, &stop_len);
+        stop_inclusive = q->stop_inclusive;
+    }
+    /* prefix filter clamps range (:558-578) */
+    if (q->sort_key_filter_type == RRDB_FT_MATCH_PREFIX && q->sort_key_filter_pattern.len > 0) {
+        uint64_t ps_len, pe_len;
+        uint8_t *ps = make_key(q->hash_key.data, q->hash_key.len, q->sort_key_filter_pattern.data,
+                               q->sort_key_filter_pattern.len, &ps_len);
+        uint8_t *pe = make_key(q->hash_key.data, q->hash_key.len, q->sort_key_filter_pattern.data,
+                               q->sort_key_filter_pattern.len, &pe_len);
+        pe_len = next_blob(pe, pe_len);
+        if (key_cmp(ps, ps_len, start, start_len) > 0) {
+            free(start);
+            start = ps;
+            start_len = ps_len;
+            start_inclusive = 1;
+        } else
+            free(ps);
+        if (key_cmp(pe, pe_len, stop, stop_len) <= 0) {
+            free(stop);
+            stop = pe;
+            stop_len = pe_len;
+            stop_inclusive = 0;
+        } else
+            free(pe);
+    }
+    /* empty range check (:580-607) */
+    int c = key_cmp(start, start_len, stop, stop_len);
+    if (c > 0 || (c == 0 && (!start_inclusive || !stop_inclusive))) {
+        free(start);
+        free(stop);
+        out->error = RRDB_OK;
+        return RRDB_OK;
+    }
+
+    /* iterate over visible view of [start, stop]; build_view excludes the
+     * stop key itself (lower_bound), so extend by one when stop_inclusive */
+    uint64_t n;
+    uint64_t *view;
+    if (stop_inclusive) {
+        /* use stop+\0 as exclusive bound: smallest key > stop */
+        uint8_t *stop2 = (uint8_t *)malloc(stop_len + 1);
+        memcpy(stop2, stop, stop_len);
+        stop2[stop_len] = 0;
+        view = build_view(e, start, start_len, stop2, stop_len + 1, &n);
+        free(stop2);
+    } else {
+        view = build_view(e, start, start_len, stop, stop_len, &n);
+    }
+
+    rrdb_slice *keys = (rrdb_slice *)arena_alloc(a, (n ? n : 1) * sizeof(rrdb_slice));
+    rrdb_slice *vals = (rrdb_slice *)arena_alloc(a, (n ? n : 1) * sizeof(rrdb_slice));
+    uint64_t m = 0;
+    int64_t count = 0, size = 0;
+    uint64_t iteration_count = 0;
+    int complete = 0;
+
+    /* forward: walk view left->right; reverse: right->left, then reverse
+     * results (on_multi_get:616-676 / :678-778).  first_exclusive skip of the
+     * exact start/stop key mirrors :636-643 / :700-707. */
+    uint64_t steps = n;
+    for (uint64_t s = 0; s < steps; s++) {
+        uint64_t vi = q->reverse ? (steps - 1 - s) : s;
+        const Run *r = &e->runs[view[vi] >> 40];
+        uint64_t idx = view[vi] & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, idx, &kl);
+        const uint8_t *v = run_val(r, idx, &vl);
+        if (count >= (int64_t)max_kv_count || iteration_count >= max_iteration_count ||
+            size >= max_iteration_size)
+            break;
+        if (!q->reverse && !start_inclusive && key_cmp(k, kl, start, start_len) == 0)
+            continue; /* skipped without counting an iteration */
+        if (q->reverse && !stop_inclusive && key_cmp(k, kl, stop, stop_len) == 0)
+            continue;
+        iteration_count++;
+        /* append_key_value_for_multi_get (:2462-2504) */
+        if (ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl)))
+            continue; /* kExpired */
+        uint64_t sklen = kl - 2 - q->hash_key.len;
+        const uint8_t *skp = k + 2 + q->hash_key.len;
+        if (q->sort_key_filter_type != RRDB_FT_NO_FILTER &&
+            !validate_filter(q->sort_key_filter_type, q->sort_key_filter_pattern.data,
+                             q->sort_key_filter_pattern.len, skp, sklen))
+            continue; /* kFiltered */
+        keys[m].data = (uint8_t *)arena_alloc(a, sklen ? sklen : 1);
+        memcpy(keys[m].data, skp, sklen);
+        keys[m].len = sklen;
+        if (!q->no_value) {
+            uint64_t ul = vl - hdr;
+            vals[m].data = (uint8_t *)arena_alloc(a, ul ? ul : 1);
+            memcpy(vals[m].data, v + hdr, ul);
+            vals[m].len = ul;
+        } else {
+            vals[m].data = NULL;
+            vals[m].len = 0;
+        }
+        count++;
+        size += (int64_t)keys[m].len + (int64_t)vals[m].len;
+        m++;
+        if (s + 1 == steps)
+            complete = 1;
+    }
+    /* complete iff every countable view element was iterated: mirrors
+     * `it->Valid() && !complete -> kIncomplete` (on_multi_get:789-799) */
+    {
+        uint64_t total = n;
+        if (n > 0) {
+            uint64_t fi = q->reverse ? n - 1 : 0;
+            const Run *r = &e->runs[view[fi] >> 40];
+            uint64_t idx = view[fi] & 0xFFFFFFFFFFull, kl;
+            const uint8_t *k = run_key(r, idx, &kl);
+            if (!q->reverse && !start_inclusive && key_cmp(k, kl, start, start_len) == 0)
+                total--;
+            if (q->reverse && !stop_inclusive && key_cmp(k, kl, stop, stop_len) == 0)
+                total--;
+        }
+        complete = (iteration_count >= total);
+    }
+    if (q->reverse && m > 1) {
+        /* revert order to ascending (on_multi_get:758-765) */
+        for (uint64_t i = 0; i < m / 2; i++) {
+            rrdb_slice t = keys[i];
+            keys[i] = keys[m - 1 - i];
+            keys[m - 1 - i] = t;
+            t = vals[i];
+            vals[i] = vals[m - 1 - i];
+            vals[m - 1 - i] = t;
+        }
+    }
+    free(view);
+    free(start);
+    free(stop);
+    out->keys = keys;
+    out->values = vals;
+    out->count = m;
+    out->error = complete ? RRDB_OK : RRDB_INCOMPLETE; /* :789-799 */
+    return out->error;
+}
+
+/* ---- scan ---- */
+
+/* validate_key_value_for_scan (pegasus_server_impl.cpp:2382-2432):
+ * returns 0 normal, 1 expired, 2 filtered, 3 hash-invalid */
+static int validate_for_scan(Engine *e, const ScanCtx *c, const uint8_t *k, uint64_t kl,
+                             const uint8_t *v, uint64_t vl, uint32_t epoch_now)
+{
+    if (ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl)))
+        return 1;
+    if (c->validate_hash_req && e->validate_hash) {
+        if (e->partition_version < 0 || e->pidx > e->partition_version ||
+            !check_key_hash(k, kl, e->pidx, e->partition_version))
+            return 3;
+    }
+    if (c->hk_ft != RRDB_FT_NO_FILTER || c->sk_ft != RRDB_FT_NO_FILTER) {
+        uint16_t hklen = key_hklen(k);
+        const uint8_t *hk = k + 2;
+        const uint8_t *sk = k + 2 + hklen;
+        uint64_t sklen = kl - 2 - hklen;
+        if (c->hk_ft != RRDB_FT_NO_FILTER &&
+            !validate_filter(c->hk_ft, c->hk_pat, c->hk_pat_len, hk, hklen))
+            return 2;
+        if (c->sk_ft != RRDB_FT_NO_FILTER &&
+            !validate_filter(c->sk_ft, c->sk_pat, c->sk_pat_len, sk, sklen))
+            return 2;
+    }
+    return 0;
+}
+
+/* one scan batch over a context; fills out and advances c->cursor */
+static void scan_batch(Engine *e, ScanCtx *c, uint32_t epoch_now, rrdb_result *out)
+{
+    arena *a = (arena *)out->_arena;
+    uint32_t hdr = value_hdr_len(e->data_version);
+    uint32_t batch_count = e->max_iter_count;
+    if (c->batch_size > 0 && (uint32_t)c->batch_size < batch_count)
+        batch_count = (uint32_t)c->batch_size;
+    uint32_t max_iter = e->max_iter_count;
+
+    uint64_t cap = batch_count;
+    rrdb_slice *keys = NULL;
+    rrdb_slice *vals = NULL;
+    int32_t *ets = NULL;
+    if (!c->only_return_count) {
+        keys = (rrdb_slice *)arena_alloc(a, cap * sizeof(rrdb_slice));
+        vals = (rrdb_slice *)arena_alloc(a, cap * sizeof(rrdb_slice));
+        if (c->return_expire_ts)
+            ets = (int32_t *)arena_alloc(a, cap * sizeof(int32_t));
+    }
+    int32_t count = 0;
+    uint64_t m = 0;
+    uint32_t iteration_count = 0;
+    int complete = 0;
+    while (c->cursor < c->view_n) {
+        if ((uint32_t)count >= batch_count || iteration_count >= max_iter)
+            break;
+        uint64_t pv = c->view[c->cursor];
+        const Run *r = &e->runs[pv >> 40];
+        uint64_t idx = pv & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, idx, &kl);
+        const uint8_t *v = run_val(r, idx, &vl);
+        c->cursor++;
+        iteration_count++;
+        int state = validate_for_scan(e, c, k, kl, v, vl, epoch_now);
+        if (state == 0) {
+            count++;
+            if (!c->only_return_count) {
+                keys[m].data = (uint8_t *)arena_alloc(a, kl ? kl : 1);
+                memcpy(keys[m].data, k, kl);
+                keys[m].len = kl;
+                if (c->return_expire_ts)
+                    ets[m] = (int32_t)extract_expire_ts(e->data_version, v, vl);
+                if (!c->no_value) {
+                    uint64_t ul = vl - hdr;
+                    vals[m].data = (uint8_t *)arena_alloc(a, ul ? ul : 1);
+                    memcpy(vals[m].data, v + hdr, ul);
+                    vals[m].len = ul;
+                } else {
+                    vals[m].data = NULL;
+                    vals[m].len = 0;
+                }
+                m++;
+            }
+        }
+    }
+    if (c->cursor >= c->view_n)
+        complete = 1;
+    out->keys = keys;
+    out->values = vals;
+    out->expire_ts = ets;
+    out->count = m;
+    if (c->only_return_count)
+        out->i64 = count;
+    out->error = RRDB_OK;
+    out->context_id = complete ? RRDB_SCAN_CONTEXT_ID_COMPLETED : 0 /* caller sets */;
+    if (complete)
+        c->cursor = c->view_n;
+    (void)complete;
+}
+
+static int64_t park_ctx(Engine *e, ScanCtx *c)
+{
+    c->id = ++e->next_ctx_id;
+    if (e->n_ctxs == e->ctxs_cap) {
+        e->ctxs_cap = e->ctxs_cap ? e->ctxs_cap * 2 : 8;
+        e->ctxs = (ScanCtx **)realloc(e->ctxs, (size_t)e->ctxs_cap * sizeof(ScanCtx *));
+    }
+    e->ctxs[e->n_ctxs++] = c;
+    return c->id;
+}
+
+static ScanCtx *fetch_ctx(Engine *e, int64_t id)
+{
+    for (int i = 0; i < e->n_ctxs; i++) {
+        if (e->ctxs[i] && e->ctxs[i]->id == id) {
+            ScanCtx *c = e->ctxs[i];
+            e->ctxs[i] = e->ctxs[--e->n_ctxs];
+            return c;
+        }
+    }
+    return NULL;
+}
+
+int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    result_init(out);
+    if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
+        q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+        out->error = RRDB_INVALID_ARGUMENT; /* on_get_scanner:1168-1186 */
+        return out->error;
+    }
+    const uint8_t *start = q->start_key.data;
+    uint64_t start_len = q->start_key.len;
+    uint8_t start_inclusive = q->start_inclusive;
+    const uint8_t *stop = q->stop_key.data;
+    uint64_t stop_len = q->stop_key.len;
+    uint8_t stop_inclusive = q->stop_inclusive;
+
+    /* hash-key prefix filter clamps start (on_get_scanner:1206-1224) */
+    uint8_t *prefix_start = NULL;
+    if (q->hash_key_filter_type == RRDB_FT_MATCH_PREFIX && q->hash_key_filter_pattern.len > 0) {
+        uint64_t pl;
+        prefix_start = make_key(q->hash_key_filter_pattern.data, q->hash_key_filter_pattern.len,
+                                NULL, 0, &pl);
+        if (key_cmp(prefix_start, pl, start, start_len) > 0) {
+            start = prefix_start;
+            start_len = pl;
+            start_inclusive = 1;
+        }
+    }
+    /* empty range (on_get_scanner:1227-1243) */
+    int c = key_cmp(start, start_len, stop, stop_len);
+    if (c > 0 || (c == 0 && (!start_inclusive || !stop_inclusive))) {
+        free(prefix_start);
+        out->error = RRDB_OK;
+        out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+        return RRDB_OK;
+    }
+
+    ScanCtx *ctx = (ScanCtx *)calloc(1, sizeof(ScanCtx));
+    /* view = visible records in [start(incl), stop(stop_inclusive?incl:excl)];
+     * first_exclusive start handled by advancing past an exact match
+     * (on_get_scanner:1277-1283) */
+    uint64_t n;
+    uint64_t *view;
+    if (stop_inclusive) {
+        uint8_t *stop2 = (uint8_t *)malloc(stop_len + 1);
+        memcpy(stop2, stop, stop_len);
+        stop2[stop_len] = 0;
+        view = build_view(e, start, start_len, stop2, stop_len + 1, &n);
+        free(stop2);
+    } else {
+        view = build_view(e, start, start_len, stop, stop_len, &n);
+    }
+    uint64_t cur = 0;
+    if (!start_inclusive && n > 0) {
+        const Run *r = &e->runs[view[0] >> 40];
+        uint64_t idx = view[0] & 0xFFFFFFFFFFull, kl;
+        const uint8_t *k = run_key(r, idx, &kl);
+        if (key_cmp(k, kl, start, start_len) == 0)
+            cur = 1;
+    }
+    free(prefix_start);
+    ctx->view = view;
+    ctx->view_n = n;
+    ctx->cursor = cur;
+    ctx->stop = (uint8_t *)malloc(stop_len ? stop_len : 1);
+    memcpy(ctx->stop, stop, stop_len);
+    ctx->stop_len = stop_len;
+    ctx->stop_inclusive = stop_inclusive;
+    ctx->batch_size = q->batch_size;
+    ctx->no_value = q->no_value;
+    ctx->validate_hash_req = q->validate_partition_hash;
+    ctx->return_expire_ts = q->return_expire_ts;
+    ctx->only_return_count = q->only_return_count;
+    ctx->hk_ft = q->hash_key_filter_type;
+    ctx->sk_ft = q->sort_key_filter_type;
+    ctx->hk_pat = (uint8_t *)malloc(q->hash_key_filter_pattern.len ? q->hash_key_filter_pattern.len : 1);
+    memcpy(ctx->hk_pat, q->hash_key_filter_pattern.data, q->hash_key_filter_pattern.len);
+    ctx->hk_pat_len = q->hash_key_filter_pattern.len;
+    ctx->sk_pat = (uint8_t *)malloc(q->sort_key_filter_pattern.len ? q->sort_key_filter_pattern.len : 1);
+    memcpy(ctx->sk_pat, q->sort_key_filter_pattern.data, q->sort_key_filter_pattern.len);
+    ctx->sk_pat_len = q->sort_key_filter_pattern.len;
+
+    scan_batch(e, ctx, epoch_now, out);
+    if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
+        free_ctx(ctx);
+    } else {
+        out->context_id = park_ctx(e, ctx);
+    }
+    return out->error;
+}
+
+int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    result_init(out);
+    ScanCtx *c = fetch_ctx(e, context_id);
+    if (!c) {
+        out->error = RRDB_NOT_FOUND; /* on_scan:1539-1541 */
+        return out->error;
+    }
+    scan_batch(e, c, epoch_now, out);
+    if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
+        free_ctx(c);
+    } else {
+        out->context_id = park_ctx(e, c); /* re-put -> new handle (on_scan:1516-1526) */
+    }
+    return out->error;
+}
+
+void rrdb_clear_scanner(void *h, int64_t context_id)
+{
+    ScanCtx *c = fetch_ctx((Engine *)h, context_id);
+    free_ctx(c);
+}
+
+/* ---- manual compaction ---- */
+int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
+                            rrdb_compact_stats *stats)
+{
+    Engine *e = (Engine *)h;
+    (void)opts;
+    rrdb_compact_stats st;
+    memset(&st, 0, sizeof(st));
+    if (e->manual_compact_disabled) {
+        if (stats)
+            *stats = st;
+        return RRDB_INVALID_ARGUMENT; /* check_manual_compact_state refuses */
+    }
+    for (int i = 0; i < e->n_runs; i++)
+        st.input_records += e->runs[i].n;
+
+    /* full k-way merge, newest wins (compaction iterator); tombstones dropped
+     * at bottommost; filter on surviving PUTs
+     * (key_ttl_compaction_filter.h:55-92) */
+    int R = e->n_runs;
+    uint64_t *pos = (uint64_t *)calloc((size_t)(R ? R : 1), 8);
+    uint64_t total = 0;
+    for (int r = 0; r < R; r++)
+        total += e->runs[r].n;
+
+    /* output buffers (grow-as-needed) */
+    uint64_t out_cap_rec = total ? total : 1;
+    uint64_t *okoff = (uint64_t *)malloc((out_cap_rec + 1) * 8);
+    uint64_t *ovoff = (uint64_t *)malloc((out_cap_rec + 1) * 8);
+    uint64_t *osk = (uint64_t *)malloc(out_cap_rec * 8);
+    size_t okeys_cap = 1 << 20, ovals_cap = 1 << 20;
+    uint8_t *okeys = (uint8_t *)malloc(okeys_cap);
+    uint8_t *ovals = (uint8_t *)malloc(ovals_cap);
+    uint64_t on = 0, okb = 0, ovb = 0;
+    okoff[0] = 0;
+    ovoff[0] = 0;
+
+    for (;;) {
+        int bestr = -1;
+        const uint8_t *bk = NULL;
+        uint64_t bl = 0;
+        for (int r = 0; r < R; r++) {
+            if (pos[r] >= e->runs[r].n)
+                continue;
+            uint64_t kl;
+            const uint8_t *k = run_key(&e->runs[r], pos[r], &kl);
+            if (bestr < 0 || key_cmp(k, kl, bk, bl) < 0) {
+                bestr = r;
+                bk = k;
+                bl = kl;
+            }
+        }
+        if (bestr < 0)
+            break;
+        uint64_t win_seq = 0;
+        int win_run = -1;
+        uint64_t win_idx = 0;
+        int dup = -1;
+        for (int r = 0; r < R; r++) {
+            if (pos[r] >= e->runs[r].n)
+                continue;
+            uint64_t kl;
+            const uint8_t *k = run_key(&e->runs[r], pos[r], &kl);
+            if (key_cmp(k, kl, bk, bl) == 0) {
+                uint64_t seq = e->runs[r].sk[pos[r]] >> 1;
+                if (win_run < 0 || seq > win_seq) {
+                    win_seq = seq;
+                    win_run = r;
+                    win_idx = pos[r];
+                }
+                pos[r]++;
+                dup++;
+            }
+        }
+        st.shadowed += (uint64_t)dup;
+        const Run *wr = &e->runs[win_run];
+        if ((wr->sk[win_idx] & 1) == RRDB_KIND_DELETE) {
+            st.tombstones++;
+            continue;
+        }
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(wr, win_idx, &kl);
+        const uint8_t *v = run_val(wr, win_idx, &vl);
+
+        /* === KeyWithTTLCompactionFilter::Filter (:55-92) === */
+        int drop = 0, value_changed = 0;
+        uint8_t *newv = NULL;
+        if (kl >= 2) {
+            uint32_t expire_ts = extract_expire_ts(e->data_version, v, vl);
+            if (e->default_ttl != 0 && expire_ts == 0) {
+                expire_ts = epoch_now + e->default_ttl;
+                newv = (uint8_t *)malloc(vl ? vl : 1);
+                memcpy(newv, v, vl);
+                update_expire_ts(e->data_version, newv, expire_ts);
+                value_changed = 1;
+            }
+            if (e->user_ops.n_ops > 0) {
+                const uint8_t *vv = value_changed ? newv : v;
+                uint16_t hklen = key_hklen(k);
+                const uint8_t *hk = k + 2;
+                const uint8_t *sk = k + 2 + hklen;
+                uint64_t sklen = kl - 2 - hklen;
+                for (int oi = 0; oi < e->user_ops.n_ops; oi++) {
+                    const orc_op *op = &e->user_ops.ops[oi];
+                    if (!all_rules_match(op, e->data_version, epoch_now, hk, hklen, sk, sklen, vv,
+                                         vl))
+                        continue;
+                    if (op->type == COT_DELETE) {
+                        drop = 1;
+                        break;
+                    }
+                    /* update_ttl (compaction_operation.cpp:78-113) */
+                    uint32_t new_ts = 0;
+                    int apply = 1;
+                    switch (op->ut_type) {
+                    case UTOT_FROM_NOW:
+                        new_ts = epoch_now + op->ut_value;
+                        break;
+                    case UTOT_FROM_CURRENT: {
+                        uint32_t cur = extract_expire_ts(e->data_version, vv, vl);
+                        if (cur == 0)
+                            apply = 0;
+                        else
+                            new_ts = op->ut_value + cur;
+                        break;
+                    }
+                    case UTOT_TIMESTAMP:
+                        new_ts = op->ut_value - 1451606400u; /* epoch_begin */
+                        break;
+                    default:
+                        apply = 0;
+                        break;
+                    }
+                    if (apply) {
+                        /* *new_value = existing_value (the value_view fixed at
+                         * entry), then patch (compaction_operation.cpp:106-109) */
+                        uint8_t *nv = (uint8_t *)malloc(vl ? vl : 1);
+                        memcpy(nv, vv, vl);
+                        update_expire_ts(e->data_version, nv, new_ts);
+                        free(newv);
+                        newv = nv;
+                        value_changed = 1;
+                    }
+                }
+            }
+            if (!drop) {
+                /* final keep/drop: local expire_ts (post default-ttl, pre
+                 * user-op) + stale-split check (:91) */
+                int stale = 0;
+                if (e->validate_hash && kl >= 2 && e->partition_version >= 0 &&
+                    e->pidx <= e->partition_version)
+                    stale = !check_key_hash(k, kl, e->pidx, e->partition_version);
+                if (ts_expired(epoch_now, expire_ts)) {
+                    drop = 1;
+                    st.expired++;
+                } else if (stale) {
+                    drop = 1;
+                    st.filtered++;
+                }
+            } else {
+                st.filtered++;
+            }
+        }
+        if (drop) {
+            free(newv);
+            continue;
+        }
+        const uint8_t *vout = value_changed ? newv : v;
+        /* append record to output run */
+        while (okb + kl > okeys_cap) {
+            okeys_cap *= 2;
+            okeys = (uint8_t *)realloc(okeys, okeys_cap);
+        }
+        while (ovb + vl > ovals_cap) {
+            ovals_cap *= 2;
+            ovals = (uint8_t *)realloc(ovals, ovals_cap);
+        }
+        memcpy(okeys + okb, k, kl);
+        okb += kl;
+        memcpy(ovals + ovb, vout, vl);
+        ovb += vl;
+        osk[on] = wr->sk[win_idx];
+        on++;
+        okoff[on] = okb;
+        ovoff[on] = ovb;
+        free(newv);
+    }
+    free(pos);
+
+    st.output_records = on;
+    st.output_bytes = okb + ovb;
+
+    /* swap in the merged run */
+    for (int i = 0; i < e->n_runs; i++)
+        free_run(&e->runs[i]);
+    e->n_runs = 0;
+    if (on > 0) {
+        Run r;
+        r.n = on;
+        r.keys = okeys;
+        r.koff = okoff;
+        r.vals = ovals;
+        r.voff = ovoff;
+        r.sk = osk;
+        r.min_seq = ~0ull;
+        r.max_seq = 0;
+        for (uint64_t i = 0; i < on; i++) {
+            uint64_t s = osk[i] >> 1;
+            if (s < r.min_seq)
+                r.min_seq = s;
+            if (s > r.max_seq)
+                r.max_seq = s;
+        }
+        if (e->runs_cap == 0) {
+            e->runs_cap = 4;
+            e->runs = (Run *)realloc(e->runs, (size_t)e->runs_cap * sizeof(Run));
+        }
+        e->runs[e->n_runs++] = r;
+    } else {
+        free(okeys);
+        free(okoff);
+        free(ovals);
+        free(ovoff);
+        free(osk);
+    }
+    if (stats)
+        *stats = st;
+    return RRDB_OK;
+}
+
+/* ================= test hook definitions ================= */
+uint32_t orc_extract_expire_ts(uint32_t ver, const uint8_t *val, uint64_t len)
+{
+    return extract_expire_ts(ver, val, len);
+}
+void orc_update_expire_ts(uint32_t ver, uint8_t *val, uint32_t ts)
+{
+    update_expire_ts(ver, val, ts);
+}
+int orc_ts_expired(uint32_t epoch_now, uint32_t expire_ts)
+{
+    return ts_expired(epoch_now, expire_ts);
+}
+uint64_t orc_key_hash(const uint8_t *key, uint64_t len) { return key_hash(key, len); }
+int orc_pattern_match(const uint8_t *v, uint64_t vlen, int type, const uint8_t *pat,
+                      uint64_t plen)
+{
+    return string_pattern_match(v, vlen, type, (const char *)pat, plen);
+}

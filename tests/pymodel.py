@@ -1,0 +1,227 @@
+"""Independent pure-Python model of the hot-path semantics, used as a second
+check on the CPU oracle at small sizes (the oracle itself is the parity anchor
+for the HIP engine; this model keeps the oracle honest).
+
+Written directly from the cited reference logic, independently from the C
+oracle's code:
+  - merging-iterator / newest-wins / tombstones: rocksdb v8.5.3 semantics at
+    the engine boundary
+  - handlers: reference src/server/pegasus_server_impl.cpp:418-1547
+  - compaction filter: reference src/server/key_ttl_compaction_filter.h:55-121
+"""
+from __future__ import annotations
+
+import struct
+
+
+def crc64(data: bytes, init: int = 0) -> int:
+    bits = [63, 61, 59, 58, 56, 55, 52, 49, 48, 47, 46, 44, 41, 37, 36, 34, 32, 31, 28, 26,
+            23, 22, 19, 16, 13, 12, 10, 9, 6, 4, 3, 0]
+    poly = 0
+    for b in bits:
+        poly |= 1 << (63 - b)
+    tbl = []
+    for i in range(256):
+        k = i
+        for _ in range(8):
+            k = (k >> 1) ^ poly if k & 1 else k >> 1
+        tbl.append(k)
+    crc = ~init & 0xFFFFFFFFFFFFFFFF
+    for byte in data:
+        crc = tbl[(crc ^ byte) & 0xFF] ^ (crc >> 8)
+    return ~crc & 0xFFFFFFFFFFFFFFFF
+
+
+def hdr_len(ver):
+    return {0: 4, 1: 12, 2: 13}[ver]
+
+
+def expire_of(val: bytes, ver: int) -> int:
+    off = 1 if ver == 2 else 0
+    return struct.unpack(">I", val[off:off + 4])[0]
+
+
+def expired(epoch_now: int, ts: int) -> bool:
+    return ts > 0 and ts <= epoch_now
+
+
+class Model:
+    def __init__(self, pidx=0, data_version=1):
+        self.runs = []  # list of dict key->(value, seq, kind); index = age order
+        self.pidx = pidx
+        self.partition_version = -1
+        self.validate_hash = False
+        self.data_version = data_version
+        self.default_ttl = 0
+        self.user_ops = []  # parsed op dicts
+
+    def ingest(self, records):
+        """records: [(raw_key, raw_value, seq, kind)]"""
+        self.runs.append({k: (v, s, kd) for k, v, s, kd in records})
+
+    # -- visibility --
+    def newest(self, key: bytes):
+        best = None
+        for run in self.runs:
+            if key in run:
+                v, s, kd = run[key]
+                if best is None or s > best[1]:
+                    best = (v, s, kd)
+        return best
+
+    def visible_items(self, lo=None, hi=None, lo_incl=True, hi_incl=False):
+        """key-ordered [(key, value)] of visible (newest, non-tombstone)."""
+        newest = {}
+        for run in self.runs:
+            for k, (v, s, kd) in run.items():
+                if k not in newest or s > newest[k][1]:
+                    newest[k] = (v, s, kd)
+        out = []
+        for k in sorted(newest):
+            if lo is not None and (k < lo or (k == lo and not lo_incl)):
+                continue
+            if hi is not None and (k > hi or (k == hi and not hi_incl)):
+                continue
+            v, s, kd = newest[k]
+            if kd == 0:
+                out.append((k, v))
+        return out
+
+    # -- handlers --
+    def get(self, key: bytes, now: int):
+        r = self.newest(key)
+        if r is None or r[2] == 1:
+            return 1, None
+        if expired(now, expire_of(r[0], self.data_version)):
+            return 1, None
+        return 0, r[0][hdr_len(self.data_version):]
+
+    def ttl(self, key: bytes, now: int):
+        r = self.newest(key)
+        if r is None or r[2] == 1:
+            return 1, None
+        e = expire_of(r[0], self.data_version)
+        if expired(now, e):
+            return 1, None
+        return 0, (e - now) if e > 0 else -1
+
+    def sortkey_count(self, hash_key: bytes, now: int):
+        lo = struct.pack(">H", len(hash_key)) + hash_key
+        hi = bytearray(lo)
+        p = len(hi) - 1
+        while hi[p] == 0xFF:
+            p -= 1
+        hi[p] += 1
+        hi = bytes(hi[:p + 1])
+        items = self.visible_items(lo, hi, True, False)
+        return sum(1 for _, v in items if not expired(now, expire_of(v, self.data_version)))
+
+    def full_scan(self, now: int, validate_hash_req=True):
+        """All visible non-expired rows (hash-valid) — count_data equivalent."""
+        out = []
+        for k, v in self.visible_items():
+            if expired(now, expire_of(v, self.data_version)):
+                continue
+            if validate_hash_req and self.validate_hash:
+                if self.partition_version < 0 or self.pidx > self.partition_version:
+                    continue
+                hklen = struct.unpack(">H", k[:2])[0]
+                h = crc64(k[2:2 + hklen] if hklen else k[2:])
+                if (h & self.partition_version) != self.pidx:
+                    continue
+            out.append((k, v[hdr_len(self.data_version):]))
+        return out
+
+    def compact(self, now: int):
+        """Manual compaction result: surviving {key: value} after newest-wins,
+        tombstone drop, TTL filter, default-ttl rewrite and user ops."""
+        ver = self.data_version
+        surviving = {}
+        newest = {}
+        for run in self.runs:
+            for k, (v, s, kd) in run.items():
+                if k not in newest or s > newest[k][1]:
+                    newest[k] = (v, s, kd)
+        for k in sorted(newest):
+            v, s, kd = newest[k]
+            if kd == 1:
+                continue
+            drop, newv = self._filter(k, v, now)
+            if not drop:
+                surviving[k] = (newv if newv is not None else v, s)
+        self.runs = [{k: (v, s, 0) for k, (v, s) in surviving.items()}]
+        return surviving
+
+    def _filter(self, key: bytes, value: bytes, now: int):
+        """KeyWithTTLCompactionFilter::Filter (:55-92). Returns (drop, new_value)."""
+        ver = self.data_version
+        if len(key) < 2:
+            return False, None
+        expire_ts = expire_of(value, ver)
+        new_value = None
+        if self.default_ttl != 0 and expire_ts == 0:
+            expire_ts = (now + self.default_ttl) & 0xFFFFFFFF
+            off = 1 if ver == 2 else 0
+            new_value = value[:off] + struct.pack(">I", expire_ts) + value[off + 4:]
+        if self.user_ops:
+            vv = new_value if new_value is not None else value
+            hklen = struct.unpack(">H", key[:2])[0]
+            hk, sk = key[2:2 + hklen], key[2 + hklen:]
+            for op in self.user_ops:
+                if not self._all_rules_match(op, hk, sk, vv, now):
+                    continue
+                if op["type"] == "delete":
+                    return True, None
+                # update_ttl
+                cur = expire_of(vv, ver)
+                t = op["ut_type"]
+                if t == "from_now":
+                    new_ts = (now + op["value"]) & 0xFFFFFFFF
+                elif t == "from_current":
+                    if cur == 0:
+                        continue
+                    new_ts = (op["value"] + cur) & 0xFFFFFFFF
+                elif t == "timestamp":
+                    new_ts = (op["value"] - 1451606400) & 0xFFFFFFFF
+                else:
+                    continue
+                off = 1 if ver == 2 else 0
+                new_value = vv[:off] + struct.pack(">I", new_ts) + vv[off + 4:]
+        if expired(now, expire_ts):
+            return True, None
+        if self.validate_hash and self.partition_version >= 0 and self.pidx <= self.partition_version:
+            hklen = struct.unpack(">H", key[:2])[0]
+            h = crc64(key[2:2 + hklen] if hklen else key[2:])
+            if (h & self.partition_version) != self.pidx:
+                return True, None
+        return False, new_value
+
+    def _all_rules_match(self, op, hk, sk, value, now):
+        if not op["rules"]:
+            return False
+        for r in op["rules"]:
+            if not self._rule_match(r, hk, sk, value, now):
+                return False
+        return True
+
+    def _rule_match(self, r, hk, sk, value, now):
+        t = r["type"]
+        if t in ("hashkey", "sortkey"):
+            target = hk if t == "hashkey" else sk
+            pat = r["pattern"]
+            if not pat or len(target) < len(pat):
+                return False
+            mt = r["match_type"]
+            if mt == "anywhere":
+                return pat in target
+            if mt == "prefix":
+                return target.startswith(pat)
+            if mt == "postfix":
+                return target.endswith(pat)
+            return False
+        if t == "ttl_range":
+            e = expire_of(value, self.data_version)
+            if e == 0 and r["start_ttl"] == 0 and r["stop_ttl"] == 0:
+                return True
+            return ((r["start_ttl"] + now) & 0xFFFFFFFF) <= e <= ((r["stop_ttl"] + now) & 0xFFFFFFFF)
+        return False
