@@ -1,0 +1,297 @@
+#!/usr/bin/env python3
+"""bench.py — headline benchmark for the MI355X rrdb engine.
+
+Metric (BASELINE.json): "compaction keys/sec + scan rows/sec per node;
+achieved HBM GB/s vs roofline".  `value` is compaction keys/sec (the first
+named metric; the workload is BASELINE.json configs[1]: 16 partitions, 50M
+keys, 8-sorted-run L0->L1 merge per partition, on one GPU); scan rows/sec and
+get ops/sec are reported alongside in `extra`.
+
+A step = one full merge+filter+write pass of the compaction hot path over
+every partition's 8 ingested runs (keep_inputs=True makes the pass
+repeatable; the kernels execute the complete merge, filter and output write
+each step — nothing is skipped inside the timed region).
+
+Inputs are resident in HBM when the timed region starts (ingested once during
+setup).  The output merged run is written to HBM and freed between steps.
+
+Multi-GPU (--gpus N, launched by torchrun): weak scaling — each rank owns its
+own 16 partitions (partition -> GPU sharding per SURVEY.md §8(e)); the only
+cross-GPU exchange on the data path is the count_data-style full-scan count
+reduce, done with torch.distributed all_reduce (RCCL over xGMI), mirroring
+the client-side fan-out+aggregate of the reference
+(src/client_lib/pegasus_client_impl.cpp:1197-1267).
+
+CPU baseline: the CPU oracle (restatement of the reference path)
+timed on a bounded sample on this box's host cores — reported, not the target.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+import numpy as np  # noqa: E402
+
+HBM_PEAK_GBS = 8000.0  # MI355X spec peak (MI355X_MICROARCH.md; measured ach. ~6300)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def build_partition_data(keys_pp, n_runs, rank, pidx, value_len=100):
+    from incubator_pegasus_amd import data as D
+
+    return D.build_point_table_runs(
+        keys_pp, n_runs, seed=D.DEFAULT_SEED + rank * 1000 + pidx,
+        value_len=value_len, dup_fraction=0.10, delete_fraction=0.02)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--partitions", type=int, default=16)     # per GPU (configs[1])
+    ap.add_argument("--keys", type=int, default=50_000_000)   # per GPU total
+    ap.add_argument("--runs", type=int, default=8)
+    ap.add_argument("--cpu-sample", type=int, default=2_000_000)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    import torch
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        tdist.init_process_group(backend=backend)
+        dist = tdist
+
+    if not torch.cuda.is_available():
+        print(json.dumps({"error": "no GPU: the rrdb engine has no CPU fallback"}))
+        sys.exit(1)
+    torch.cuda.set_device(local_rank)
+
+    import incubator_pegasus_amd as pa
+
+    hip = pa.hip_lib()
+    keys_pp = args.keys // args.partitions
+
+    # ---- setup: generate + ingest (untimed) ----
+    log(f"generating {args.partitions} partitions x {keys_pp} keys x {args.runs} runs ...")
+    t0 = time.time()
+    parts = []
+    total_records = 0
+    for p in range(args.partitions):
+        runs = build_partition_data(keys_pp, args.runs, rank, p)
+        eng = hip.open(1, p, local_rank)
+        for r in runs:
+            eng.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
+                                  np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
+        total_records += int(eng.num_records())
+        parts.append(eng)
+        if p == 0:
+            log(f"partition 0 ready ({time.time()-t0:.1f}s, {eng.num_records()} records)")
+    log(f"ingest done: {total_records} records on GPU {local_rank} "
+        f"({time.time()-t0:.1f}s)")
+
+    epoch_now = 1_000_000
+
+    def barrier_sync():
+        torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    def one_step():
+        out_records = 0
+        for eng in parts:
+            err, st = eng.manual_compact(epoch_now, keep_inputs=True)
+            assert err == 0
+            out_records += st.output_records
+        return out_records, st
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        out_records, last_stats = one_step()
+    if args.warmup == 0:
+        out_records, last_stats = one_step()  # need sizes for roofline math
+
+    # ---- timed region: K compaction passes ----
+    barrier_sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        one_step()
+    barrier_sync()
+    elapsed = time.time() - t0
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    keys_per_step_job = total_records * max(world, 1)  # whole-job aggregate
+    value = keys_per_step_job * args.steps / elapsed
+    ms_per_step = elapsed * 1000.0 / args.steps
+
+    # ---- roofline: dominant compaction kernel, HIP-event timed in-engine ----
+    eng0 = parts[0]
+    n0 = int(eng0.num_records())
+    # per-partition byte tallies (fixed-width records: 18B keys, 112B values)
+    key_bytes = 18 * n0
+    val_bytes = 112 * n0
+    out_bytes = int(last_stats.output_bytes)
+    phases = {
+        # algorithmic bytes per launch: stated in DESIGN.md §roofline
+        "compact_rank": key_bytes,                      # read every key once
+        "compact_flags": key_bytes + 12 * n0,           # neighbor keys + value headers
+        "compact_emit": 2 * out_bytes,                  # read kept records + write merged run
+    }
+    timings = {ph: eng0.phase_ms(ph) for ph in phases}
+    dominant = max(timings, key=lambda k: timings[k])
+    dur_ms = timings[dominant]
+    achieved = phases[dominant] / (dur_ms * 1e-3) / 1e9 if dur_ms > 0 else 0.0
+    roofline = {
+        "bound": "hbm",
+        "kernel": dominant,
+        "achieved": round(achieved, 1),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved / HBM_PEAK_GBS, 4),
+        "traffic": None,  # PMC counters collected in separate rocprofv3 runs (profiles/)
+        "phase_ms": {k: round(v, 3) for k, v in timings.items()},
+        "compact_total_ms": round(eng0.phase_ms("compact_total"), 3),
+    }
+
+    # ---- secondary: count_data-style full scan (count-only) + RCCL reduce ----
+    for eng in parts:
+        eng.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
+    barrier_sync()
+    t0 = time.time()
+    rows = 0
+    for eng in parts:
+        res = eng.scan_open(b"\x00\x00", b"\xff\xff", epoch_now, only_return_count=True,
+                            full_scan=True, validate_partition_hash=False, batch_size=2**31 - 1)
+        assert res.error == 0 and res.context_id == -1, (res.error, res.context_id)
+        rows += res.kv_count
+    torch.cuda.synchronize()
+    if dist:
+        rt = torch.tensor([rows], dtype=torch.int64,
+                          device="cuda" if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(rt)  # the §8(e) cross-partition count aggregate (RCCL)
+        rows = int(rt.item())
+        t = torch.tensor([time.time() - t0], dtype=torch.float64, device=rt.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        scan_elapsed = float(t.item())
+    else:
+        scan_elapsed = time.time() - t0
+    scan_rows_per_s = rows / scan_elapsed
+    scan_state_ms = eng0.phase_ms("scan_state")
+    # scan_state reads every visible record's key + value header once
+    scan_achieved = ((18 + 12) * n0) / (scan_state_ms * 1e-3) / 1e9 if scan_state_ms > 0 else 0.0
+
+    # ---- secondary: zipfian point gets (YCSB C) ----
+    from incubator_pegasus_amd import data as D
+
+    nq = 1_000_000
+    qids = D.zipfian_ids(nq, keys_pp, seed=D.DEFAULT_SEED)
+    qkeys = D.make_raw_keys(qids).reshape(-1)
+    qoffs = D.fixed_offsets(nq, 18)
+    import ctypes
+
+    barrier_sync()
+    t0 = time.time()
+    # raw ABI call (no per-key python packing)
+    L = hip._lib
+    from incubator_pegasus_amd.capi import _Result
+
+    res = _Result()
+    L.rrdb_batch_get(parts[0]._h, nq, np.ascontiguousarray(qkeys).ctypes.data_as(ctypes.c_void_p),
+                     qoffs.ctypes.data_as(ctypes.c_void_p), epoch_now, ctypes.byref(res))
+    found = int(res.count)
+    L.rrdb_free_result(ctypes.byref(res))
+    get_elapsed = time.time() - t0
+    get_ops_per_s = nq / get_elapsed
+
+    # ---- CPU baseline (oracle restatement, rank 0, N==1 only) ----
+    cpu_baseline = None
+    if rank == 0 and world <= 1 and not args.skip_cpu_baseline:
+        from incubator_pegasus_amd.capi import RrdbLib
+
+        sample = min(args.cpu_sample, keys_pp)
+        log(f"cpu baseline: oracle compaction pass over {sample} keys ...")
+        oracle = RrdbLib(os.path.join(REPO, "oracle", "liboracle.so"))
+        op = oracle.open(1, 0, -1)
+        runs = build_partition_data(sample, args.runs, 0, 0)
+        for r in runs:
+            op.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
+                                 np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
+        nrec = int(op.num_records())
+        t0 = time.time()
+        err, _ = op.manual_compact(epoch_now, keep_inputs=True)
+        cpu_el = time.time() - t0
+        op.close()
+        cpu_baseline = {
+            "value": round(nrec / cpu_el, 1),
+            "unit": "keys/s",
+            "cores": 1,
+            "kind": "port",
+            "sample": f"one partition, {nrec} records ({args.runs} runs), single compaction pass"
+        }
+
+    for eng in parts:
+        eng.close()
+
+    if rank == 0:
+        line = {
+            "metric": "compaction keys/sec per node",
+            "value": round(value, 1),
+            "unit": "keys/s",
+            "n_gpus": max(world, 1),
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # no published reference number
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": "compact-50M-16part-8run" if args.keys == 50_000_000 else
+                            f"compact-{args.keys}-{args.partitions}part-{args.runs}run",
+                "partitions_per_gpu": args.partitions,
+                "keys_per_gpu": total_records,
+                "runs_per_partition": args.runs,
+                "record_bytes": 130,
+                "hashkey_bytes": 16,
+                "value_bytes": 100,
+                "value_schema": "v1",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+            "extra": {
+                "scan_rows_per_s": round(scan_rows_per_s, 1),
+                "scan_rows_total": rows,
+                "scan_state_kernel_GBs": round(scan_achieved, 1),
+                "get_ops_per_s": round(get_ops_per_s, 1),
+                "get_found": found,
+                "compact_output_records_per_gpu": int(out_records),
+            },
+        }
+        print(json.dumps(line))
+
+
+if __name__ == "__main__":
+    main()

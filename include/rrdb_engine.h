@@ -141,6 +141,10 @@ typedef struct {
 typedef struct {
     int32_t target_level;          /* -1 = bottommost (informational) */
     uint8_t bottommost_force;      /* 1 = force (default in reference) */
+    uint8_t keep_inputs;           /* engine extension for benchmarking: run the
+                                    * full merge+filter+write but leave the input
+                                    * runs in place (a repeatable pass); 0 =
+                                    * reference semantics (inputs replaced) */
 } rrdb_compact_options;
 
 typedef struct {
@@ -244,6 +248,13 @@ void rrdb_free_result(rrdb_result *r);
 
 /* Library identity: "oracle-cpu" or "hip-gfx950". */
 const char *rrdb_backend(void);
+
+/* Last recorded GPU time (ms) of a named internal phase, measured with HIP
+ * events on the engine's own stream (bench/roofline introspection; the CPU
+ * oracle returns -1).  Phases: "compact_rank", "compact_flags",
+ * "compact_emit", "compact_total", "scan_state", "scan_emit", "get_search",
+ * "view_rank".  Returns -1 if never recorded. */
+double rrdb_phase_ms(void *h, const char *phase);
 
 #ifdef __cplusplus
 }

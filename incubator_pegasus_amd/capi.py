@@ -98,7 +98,8 @@ class _ScanRequest(C.Structure):
 
 
 class _CompactOptions(C.Structure):
-    _fields_ = [("target_level", C.c_int32), ("bottommost_force", C.c_uint8)]
+    _fields_ = [("target_level", C.c_int32), ("bottommost_force", C.c_uint8),
+                ("keep_inputs", C.c_uint8)]
 
 
 class _CompactStats(C.Structure):
@@ -113,13 +114,16 @@ class _CompactStats(C.Structure):
     ]
 
 
-def _cslice(b: bytes) -> _CSlice:
+def _cslice(b: bytes, keep: list) -> _CSlice:
+    """Build a _CSlice over a copy of b; the backing buffer is appended to
+    `keep`, which the caller must hold alive across the C call (struct field
+    assignment copies the slice by value, so the slice itself cannot own the
+    buffer)."""
     if not b:
         return _CSlice(None, 0)
     buf = (C.c_uint8 * len(b)).from_buffer_copy(b)
-    s = _CSlice(C.cast(buf, C.POINTER(C.c_uint8)), len(b))
-    s._keepalive = buf  # noqa: prevent GC
-    return s
+    keep.append(buf)
+    return _CSlice(C.cast(buf, C.POINTER(C.c_uint8)), len(b))
 
 
 def _read_slice(s: _Slice) -> bytes:
@@ -186,6 +190,8 @@ class RrdbLib:
         L.rrdb_manual_compact.restype = C.c_int32
         L.rrdb_manual_compact.argtypes = [
             C.c_void_p, C.POINTER(_CompactOptions), C.c_uint32, C.POINTER(_CompactStats)]
+        L.rrdb_phase_ms.restype = C.c_double
+        L.rrdb_phase_ms.argtypes = [C.c_void_p, C.c_char_p]
         L.rrdb_num_runs.restype = C.c_uint64
         L.rrdb_num_runs.argtypes = [C.c_void_p]
         L.rrdb_num_records.restype = C.c_uint64
@@ -324,10 +330,11 @@ class RrdbPartition:
                   start_inclusive=True, stop_inclusive=False, max_kv_count=-1, max_kv_size=-1,
                   no_value=False, reverse=False, sort_key_filter_type=FT_NO_FILTER,
                   sort_key_filter_pattern=b"", sort_keys=None):
+        keep = []
         req = _MultiGetRequest()
-        req.hash_key = _cslice(hash_key)
-        req.start_sortkey = _cslice(start_sortkey)
-        req.stop_sortkey = _cslice(stop_sortkey)
+        req.hash_key = _cslice(hash_key, keep)
+        req.start_sortkey = _cslice(start_sortkey, keep)
+        req.stop_sortkey = _cslice(stop_sortkey, keep)
         req.start_inclusive = 1 if start_inclusive else 0
         req.stop_inclusive = 1 if stop_inclusive else 0
         req.max_kv_count = max_kv_count
@@ -335,14 +342,13 @@ class RrdbPartition:
         req.no_value = 1 if no_value else 0
         req.reverse = 1 if reverse else 0
         req.sort_key_filter_type = sort_key_filter_type
-        req.sort_key_filter_pattern = _cslice(sort_key_filter_pattern)
-        keep = []
+        req.sort_key_filter_pattern = _cslice(sort_key_filter_pattern, keep)
         if sort_keys:
             sk, offs = _pack(sort_keys)
             req.n_sort_keys = len(sort_keys)
             req.sort_keys = sk.ctypes.data_as(C.POINTER(C.c_uint8))
             req.sort_key_offs = offs.ctypes.data_as(C.POINTER(C.c_uint64))
-            keep = [sk, offs]
+            keep += [sk, offs]
         res = _Result()
         self._L.rrdb_multi_get(self._h, C.byref(req), epoch_now, C.byref(res))
         del keep
@@ -375,17 +381,18 @@ class RrdbPartition:
                   sort_key_filter_type=FT_NO_FILTER, sort_key_filter_pattern=b"",
                   full_scan=False, validate_partition_hash=True, return_expire_ts=False,
                   only_return_count=False, on_device_out=False) -> ScanResult:
+        keep = []
         req = _ScanRequest()
-        req.start_key = _cslice(start_key)
-        req.stop_key = _cslice(stop_key)
+        req.start_key = _cslice(start_key, keep)
+        req.stop_key = _cslice(stop_key, keep)
         req.start_inclusive = 1 if start_inclusive else 0
         req.stop_inclusive = 1 if stop_inclusive else 0
         req.batch_size = batch_size
         req.no_value = 1 if no_value else 0
         req.hash_key_filter_type = hash_key_filter_type
-        req.hash_key_filter_pattern = _cslice(hash_key_filter_pattern)
+        req.hash_key_filter_pattern = _cslice(hash_key_filter_pattern, keep)
         req.sort_key_filter_type = sort_key_filter_type
-        req.sort_key_filter_pattern = _cslice(sort_key_filter_pattern)
+        req.sort_key_filter_pattern = _cslice(sort_key_filter_pattern, keep)
         req.full_scan = 1 if full_scan else 0
         req.validate_partition_hash = 1 if validate_partition_hash else 0
         req.return_expire_ts = 1 if return_expire_ts else 0
@@ -411,8 +418,13 @@ class RrdbPartition:
     def clear_scanner(self, context_id: int):
         self._L.rrdb_clear_scanner(self._h, context_id)
 
-    def manual_compact(self, epoch_now: int, *, target_level=-1, bottommost_force=True):
-        opts = _CompactOptions(target_level, 1 if bottommost_force else 0)
+    def phase_ms(self, phase: str) -> float:
+        return self._L.rrdb_phase_ms(self._h, phase.encode())
+
+    def manual_compact(self, epoch_now: int, *, target_level=-1, bottommost_force=True,
+                       keep_inputs=False):
+        opts = _CompactOptions(target_level, 1 if bottommost_force else 0,
+                               1 if keep_inputs else 0)
         st = _CompactStats()
         err = self._L.rrdb_manual_compact(self._h, C.byref(opts), epoch_now, C.byref(st))
         stats = CompactStats(**{f[0]: getattr(st, f[0]) for f in _CompactStats._fields_})

@@ -1,0 +1,1670 @@
+/* engine.cpp — host side of the MI355X rrdb engine (C-ABI in
+ * include/rrdb_engine.h).  One handle per partition; all record data lives in
+ * HBM; every per-record data-path operation runs in the HIP kernels of
+ * kernels.hip.  There is NO CPU fallback: rrdb_open fails without a GPU.
+ *
+ * Host responsibilities only: run registry, request parsing (incl. the
+ * user_specified_compaction JSON of compaction_operation.cpp:160-190),
+ * range-bound construction (pegasus_key_schema.h:41-122 byte logic), scan
+ * contexts (pegasus_scan_context.h:33-140 equivalent), result marshalling.
+ *
+ * Reference semantics restated per handler: see the comment on each function
+ * and include/rrdb_engine.h.
+ */
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <cstdint>
+#include <cstring>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "../../include/rrdb_engine.h"
+#include "engine_common.h"
+
+/* ---- launchers (kernels.hip) ---- */
+void launch_crc64_table_init(const uint64_t *host_table);
+void launch_bounds(const DevRun *, int, const uint8_t *, uint64_t, uint64_t *, int, hipStream_t);
+void launch_rank(const DevRun *, int, const uint64_t *, const uint64_t *, const uint64_t *,
+                 uint64_t, uint64_t *, hipStream_t);
+void launch_visible(const DevRun *, const uint64_t *, uint64_t, uint64_t *, hipStream_t);
+void launch_gather(const uint64_t *, const uint64_t *, const uint64_t *, uint64_t, uint64_t *,
+                   hipStream_t);
+void launch_psum(const uint64_t *, uint64_t *, uint64_t, hipStream_t);
+void launch_get(const DevRun *, int, const uint8_t *, const uint64_t *, uint64_t, uint32_t,
+                uint32_t, int32_t *, uint64_t *, uint64_t *, uint32_t *, hipStream_t);
+void launch_emit_values(const DevRun *, const uint64_t *, const int32_t *, uint64_t, uint32_t,
+                        const uint64_t *, uint8_t *, hipStream_t);
+void launch_scan_state(const DevRun *, const uint64_t *, uint64_t, const ScanParams &, uint8_t *,
+                       uint64_t *, uint64_t *, hipStream_t);
+void launch_normal_flags(const uint8_t *, uint64_t, uint64_t *, hipStream_t);
+void launch_cutoff(const uint64_t *, const uint64_t *, uint64_t, uint64_t, uint64_t *,
+                   hipStream_t);
+void launch_cut_sizes(const uint8_t *, uint64_t, uint64_t, uint64_t *, uint64_t *, hipStream_t);
+void launch_widen_u8(const uint8_t *, uint64_t, uint64_t *, hipStream_t);
+void launch_emit_scan(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
+                      const uint64_t *, const uint64_t *, const uint64_t *, const ScanParams &,
+                      uint8_t *, uint8_t *, uint64_t *, uint64_t *, int32_t *, uint64_t,
+                      hipStream_t);
+void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64_t,
+                      const uint64_t *, const uint64_t *, const ScanParams &, uint8_t *,
+                      uint8_t *, hipStream_t);
+void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
+                     uint32_t *, hipStream_t);
+void launch_compact_flags(const DevRun *, const uint64_t *, uint64_t, const CompactParams &,
+                          uint8_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
+                          CompactStatsDev *, hipStream_t);
+void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint8_t *,
+                         const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
+                         const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
+                         uint64_t *, uint64_t, hipStream_t);
+
+#define HIP_OK(x)                                                                                  \
+    do {                                                                                           \
+        hipError_t e_ = (x);                                                                       \
+        if (e_ != hipSuccess) {                                                                    \
+            fprintf(stderr, "rrdb-hip fatal: %s at %s:%d\n", hipGetErrorString(e_), __FILE__,      \
+                    __LINE__);                                                                     \
+            abort();                                                                               \
+        }                                                                                          \
+    } while (0)
+
+namespace {
+
+/* ================ crc64 table (host build; same construction as
+ * reference crc.cpp:236-295, restated) ================ */
+uint64_t host_crc64_table[256];
+void build_crc64_table()
+{
+    static const int bits[] = {63, 61, 59, 58, 56, 55, 52, 49, 48, 47, 46, 44, 41, 37, 36, 34,
+                               32, 31, 28, 26, 23, 22, 19, 16, 13, 12, 10, 9,  6,  4,  3,  0};
+    uint64_t poly = 0;
+    for (int b : bits)
+        poly |= 1ull << (63 - b);
+    for (int i = 0; i < 256; i++) {
+        uint64_t k = (uint64_t)i;
+        for (int j = 0; j < 8; j++)
+            k = (k & 1) ? ((k >> 1) ^ poly) : (k >> 1);
+        host_crc64_table[i] = k;
+    }
+}
+
+/* ================ result arena ================ */
+struct Arena {
+    std::vector<void *> host_blocks;
+    std::vector<void *> dev_ptrs;
+    void *alloc(size_t n)
+    {
+        void *p = malloc(n ? n : 1);
+        host_blocks.push_back(p);
+        return p;
+    }
+    ~Arena()
+    {
+        for (void *p : host_blocks)
+            free(p);
+        for (void *p : dev_ptrs)
+            (void)hipFree(p);
+    }
+};
+
+Arena *result_init(rrdb_result *r)
+{
+    memset(r, 0, sizeof(*r));
+    Arena *a = new Arena();
+    r->_arena = a;
+    return a;
+}
+
+/* ================ tiny JSON parser (independent of the oracle's) ======== */
+struct Json {
+    enum Type { NUL, BOOL, NUM, STR, ARR, OBJ } type = NUL;
+    bool b = false;
+    long long num = 0;
+    std::string str;
+    std::vector<Json> arr;
+    std::vector<std::pair<std::string, Json>> obj;
+    const Json *get(const std::string &k) const
+    {
+        for (auto &kv : obj)
+            if (kv.first == k)
+                return &kv.second;
+        return nullptr;
+    }
+};
+
+struct JParse {
+    const char *s, *e;
+    bool ok = true;
+    void ws()
+    {
+        while (s < e && (*s == ' ' || *s == '\t' || *s == '\n' || *s == '\r'))
+            s++;
+    }
+    bool lit(char c)
+    {
+        ws();
+        if (s < e && *s == c) {
+            s++;
+            return true;
+        }
+        return false;
+    }
+    std::string pstr()
+    {
+        ws();
+        std::string out;
+        if (s >= e || *s != '"') {
+            ok = false;
+            return out;
+        }
+        s++;
+        while (s < e && *s != '"') {
+            char c = *s++;
+            if (c == '\\' && s < e) {
+                char esc = *s++;
+                switch (esc) {
+                case 'n': c = '\n'; break;
+                case 't': c = '\t'; break;
+                case 'r': c = '\r'; break;
+                case 'b': c = '\b'; break;
+                case 'f': c = '\f'; break;
+                case 'u': {
+                    if (e - s < 4) { ok = false; return out; }
+                    unsigned v = 0;
+                    for (int i = 0; i < 4; i++) {
+                        char h = s[i];
+                        v <<= 4;
+                        if (h >= '0' && h <= '9') v |= h - '0';
+                        else if (h >= 'a' && h <= 'f') v |= h - 'a' + 10;
+                        else if (h >= 'A' && h <= 'F') v |= h - 'A' + 10;
+                        else { ok = false; return out; }
+                    }
+                    s += 4;
+                    c = (char)v;
+                    break;
+                }
+                default: c = esc; break;
+                }
+            }
+            out.push_back(c);
+        }
+        if (s >= e) {
+            ok = false;
+            return out;
+        }
+        s++;
+        return out;
+    }
+    Json value()
+    {
+        Json j;
+        ws();
+        if (s >= e) {
+            ok = false;
+            return j;
+        }
+        char c = *s;
+        if (c == '"') {
+            j.type = Json::STR;
+            j.str = pstr();
+        } else if (c == '{') {
+            s++;
+            j.type = Json::OBJ;
+            ws();
+            if (lit('}'))
+                return j;
+            for (;;) {
+                std::string k = pstr();
+                if (!ok || !lit(':')) {
+                    ok = false;
+                    return j;
+                }
+                j.obj.emplace_back(std::move(k), value());
+                if (!ok)
+                    return j;
+                if (lit('}'))
+                    return j;
+                if (!lit(',')) {
+                    ok = false;
+                    return j;
+                }
+            }
+        } else if (c == '[') {
+            s++;
+            j.type = Json::ARR;
+            ws();
+            if (lit(']'))
+                return j;
+            for (;;) {
+                j.arr.push_back(value());
+                if (!ok)
+                    return j;
+                if (lit(']'))
+                    return j;
+                if (!lit(',')) {
+                    ok = false;
+                    return j;
+                }
+            }
+        } else if (c == 't' && e - s >= 4 && !strncmp(s, "true", 4)) {
+            j.type = Json::BOOL;
+            j.b = true;
+            s += 4;
+        } else if (c == 'f' && e - s >= 5 && !strncmp(s, "false", 5)) {
+            j.type = Json::BOOL;
+            s += 5;
+        } else if (c == 'n' && e - s >= 4 && !strncmp(s, "null", 4)) {
+            s += 4;
+        } else if (c == '-' || (c >= '0' && c <= '9')) {
+            j.type = Json::NUM;
+            bool neg = c == '-';
+            if (neg)
+                s++;
+            if (s >= e || *s < '0' || *s > '9') {
+                ok = false;
+                return j;
+            }
+            long long v = 0;
+            while (s < e && *s >= '0' && *s <= '9')
+                v = v * 10 + (*s++ - '0');
+            j.num = neg ? -v : v;
+        } else {
+            ok = false;
+        }
+        return j;
+    }
+};
+
+bool json_parse(const std::string &text, Json &out)
+{
+    JParse p{text.c_str(), text.c_str() + text.size()};
+    out = p.value();
+    return p.ok;
+}
+
+/* host-side parsed rules/ops (mirrors compaction_operation.cpp:116-190) */
+struct HostRule {
+    int type = -1;       /* DFR_* */
+    int match_type = DSM_INVALID;
+    uint32_t start_ttl = 0, stop_ttl = 0;
+    std::string pattern;
+};
+struct HostOp {
+    int type = -1; /* DOP_* */
+    int ut_type = DUT_INVALID;
+    uint32_t ut_value = 0;
+    std::vector<HostRule> rules;
+};
+
+int enum_smt(const std::string &s)
+{
+    if (s == "SMT_MATCH_ANYWHERE") return DSM_ANYWHERE;
+    if (s == "SMT_MATCH_PREFIX") return DSM_PREFIX;
+    if (s == "SMT_MATCH_POSTFIX") return DSM_POSTFIX;
+    return DSM_INVALID;
+}
+int enum_frt(const std::string &s)
+{
+    if (s == "FRT_HASHKEY_PATTERN") return DFR_HASHKEY;
+    if (s == "FRT_SORTKEY_PATTERN") return DFR_SORTKEY;
+    if (s == "FRT_TTL_RANGE") return DFR_TTL_RANGE;
+    return -1;
+}
+int enum_cot(const std::string &s)
+{
+    if (s == "COT_UPDATE_TTL") return DOP_UPDATE_TTL;
+    if (s == "COT_DELETE") return DOP_DELETE;
+    return -1;
+}
+int enum_utot(const std::string &s)
+{
+    if (s == "UTOT_FROM_NOW") return DUT_FROM_NOW;
+    if (s == "UTOT_FROM_CURRENT") return DUT_FROM_CURRENT;
+    if (s == "UTOT_TIMESTAMP") return DUT_TIMESTAMP;
+    return DUT_INVALID;
+}
+
+/* invalid rules skipped; op kept only with >=1 valid rule and valid params
+ * (compaction_operation.cpp:160-190; all DEFINE_JSON fields required) */
+bool parse_user_ops(const std::string &text, std::vector<HostOp> &out)
+{
+    out.clear();
+    Json root;
+    if (!json_parse(text, root) || root.type != Json::OBJ)
+        return false;
+    const Json *ops = root.get("ops");
+    if (!ops || ops->type != Json::ARR)
+        return true; /* decodes, but no ops */
+    for (const Json &jop : ops->arr) {
+        if (jop.type != Json::OBJ)
+            return false;
+        const Json *jt = jop.get("type");
+        const Json *jp = jop.get("params");
+        const Json *jr = jop.get("rules");
+        if (!jt || jt->type != Json::STR || !jp || jp->type != Json::STR || !jr ||
+            jr->type != Json::ARR)
+            continue;
+        HostOp op;
+        op.type = enum_cot(jt->str);
+        for (const Json &jrule : jr->arr) {
+            if (jrule.type != Json::OBJ)
+                continue;
+            const Json *rt = jrule.get("type");
+            const Json *rp = jrule.get("params");
+            if (!rt || rt->type != Json::STR || !rp || rp->type != Json::STR)
+                continue;
+            HostRule rule;
+            rule.type = enum_frt(rt->str);
+            if (rule.type < 0)
+                continue;
+            Json rparams;
+            if (!json_parse(rp->str, rparams) || rparams.type != Json::OBJ)
+                continue;
+            if (rule.type == DFR_TTL_RANGE) {
+                const Json *a = rparams.get("start_ttl");
+                const Json *b = rparams.get("stop_ttl");
+                if (!a || a->type != Json::NUM || !b || b->type != Json::NUM)
+                    continue;
+                rule.start_ttl = (uint32_t)a->num;
+                rule.stop_ttl = (uint32_t)b->num;
+            } else {
+                const Json *pat = rparams.get("pattern");
+                const Json *mt = rparams.get("match_type");
+                if (!pat || pat->type != Json::STR || !mt || mt->type != Json::STR)
+                    continue;
+                rule.pattern = pat->str;
+                rule.match_type = enum_smt(mt->str);
+            }
+            op.rules.push_back(std::move(rule));
+        }
+        if (op.rules.empty() || op.type < 0)
+            continue;
+        if (op.type == DOP_UPDATE_TTL) {
+            Json oparams;
+            if (!json_parse(jp->str, oparams) || oparams.type != Json::OBJ)
+                continue;
+            const Json *t = oparams.get("type");
+            const Json *v = oparams.get("value");
+            if (!t || t->type != Json::STR || !v || v->type != Json::NUM)
+                continue;
+            op.ut_type = enum_utot(t->str);
+            op.ut_value = (uint32_t)v->num;
+        }
+        out.push_back(std::move(op));
+    }
+    return true;
+}
+
+/* ================ key byte helpers (pegasus_key_schema.h) ================ */
+std::string make_key(const uint8_t *hk, uint64_t hklen, const uint8_t *sk, uint64_t sklen)
+{
+    std::string k;
+    k.reserve(2 + hklen + sklen);
+    k.push_back((char)(hklen >> 8));
+    k.push_back((char)(hklen & 0xFF));
+    k.append((const char *)hk, hklen);
+    if (sk)
+        k.append((const char *)sk, sklen);
+    return k;
+}
+std::string next_blob(std::string k)
+{
+    size_t p = k.size() - 1;
+    while ((uint8_t)k[p] == 0xFF)
+        p--;
+    k[p] = (char)((uint8_t)k[p] + 1);
+    k.resize(p + 1);
+    return k;
+}
+int key_cmp(const std::string &a, const std::string &b)
+{
+    int c = memcmp(a.data(), b.data(), std::min(a.size(), b.size()));
+    if (c)
+        return c;
+    return a.size() < b.size() ? -1 : (a.size() > b.size() ? 1 : 0);
+}
+
+uint32_t hdr_len(uint32_t ver) { return ver == 0 ? 4u : (ver == 1 ? 12u : 13u); }
+
+/* ================ engine ================ */
+struct RunBuf {
+    uint8_t *keys = nullptr;
+    uint64_t *koff = nullptr;
+    uint8_t *vals = nullptr;
+    uint64_t *voff = nullptr;
+    uint64_t *sk = nullptr;
+    uint64_t n = 0;
+};
+
+struct HipScanCtx {
+    int64_t id = 0;
+    uint64_t *d_view = nullptr;
+    uint64_t view_n = 0;
+    uint64_t cursor = 0;
+    int32_t batch_size = -1;
+    uint8_t no_value = 0, validate_hash_req = 1, return_expire_ts = 0, only_return_count = 0,
+            on_device_out = 0;
+    int32_t hk_ft = 0, sk_ft = 0;
+    uint8_t *d_hk_pat = nullptr;
+    uint64_t hk_pat_len = 0;
+    uint8_t *d_sk_pat = nullptr;
+    uint64_t sk_pat_len = 0;
+    ~HipScanCtx()
+    {
+        if (d_view)
+            (void)hipFree(d_view);
+        if (d_hk_pat)
+            (void)hipFree(d_hk_pat);
+        if (d_sk_pat)
+            (void)hipFree(d_sk_pat);
+    }
+};
+
+struct HipEngine {
+    int32_t app_id = 0, pidx = 0, partition_version = -1;
+    int device = 0;
+    hipStream_t stream = nullptr;
+    uint32_t data_version = 1, default_ttl = 0;
+    bool validate_hash = false, manual_compact_disabled = false;
+    uint32_t max_iter_count = 1000, mg_max_iter_count = 3000;
+    uint64_t mg_max_iter_size = 30ull << 20, iter_time_ms = 30000;
+    std::vector<RunBuf> runs;
+    DevRun *d_runs = nullptr;
+    bool d_runs_dirty = true;
+    uint64_t next_seq_floor = 0;
+    /* user ops: host + device */
+    std::vector<HostOp> host_ops;
+    DevOp *d_ops = nullptr;
+    DevRule *d_rules = nullptr;
+    uint8_t *d_pats = nullptr;
+    int n_ops = 0;
+    std::unordered_map<int64_t, HipScanCtx *> ctxs;
+    int64_t next_ctx_id = 0;
+    std::mutex mu;
+    std::unordered_map<std::string, double> phase_ms;
+
+    void activate() { HIP_OK(hipSetDevice(device)); }
+
+    void free_run(RunBuf &r)
+    {
+        (void)hipFree(r.keys);
+        (void)hipFree(r.koff);
+        (void)hipFree(r.vals);
+        (void)hipFree(r.voff);
+        (void)hipFree(r.sk);
+        r = RunBuf();
+    }
+
+    DevRun *dev_runs()
+    {
+        if (d_runs_dirty) {
+            if (d_runs)
+                (void)hipFree(d_runs);
+            std::vector<DevRun> h(runs.size() ? runs.size() : 1);
+            for (size_t i = 0; i < runs.size(); i++)
+                h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
+                              runs[i].n};
+            HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
+            HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
+            d_runs_dirty = false;
+        }
+        return d_runs;
+    }
+
+    void upload_ops()
+    {
+        if (d_ops)
+            (void)hipFree(d_ops), d_ops = nullptr;
+        if (d_rules)
+            (void)hipFree(d_rules), d_rules = nullptr;
+        if (d_pats)
+            (void)hipFree(d_pats), d_pats = nullptr;
+        n_ops = (int)host_ops.size();
+        if (n_ops == 0)
+            return;
+        std::vector<DevOp> ops;
+        std::vector<DevRule> rules;
+        std::string pats;
+        for (auto &op : host_ops) {
+            DevOp d;
+            d.type = op.type;
+            d.ut_type = op.ut_type;
+            d.ut_value = op.ut_value;
+            d.rule_off = (int32_t)rules.size();
+            d.n_rules = (int32_t)op.rules.size();
+            for (auto &r : op.rules) {
+                DevRule dr;
+                dr.type = r.type;
+                dr.match_type = r.match_type;
+                dr.start_ttl = r.start_ttl;
+                dr.stop_ttl = r.stop_ttl;
+                dr.pat_off = (uint32_t)pats.size();
+                dr.pat_len = (uint32_t)r.pattern.size();
+                pats += r.pattern;
+                rules.push_back(dr);
+            }
+            ops.push_back(d);
+        }
+        HIP_OK(hipMalloc(&d_ops, ops.size() * sizeof(DevOp)));
+        HIP_OK(hipMemcpy(d_ops, ops.data(), ops.size() * sizeof(DevOp), hipMemcpyHostToDevice));
+        HIP_OK(hipMalloc(&d_rules, rules.size() * sizeof(DevRule)));
+        HIP_OK(hipMemcpy(d_rules, rules.data(), rules.size() * sizeof(DevRule),
+                         hipMemcpyHostToDevice));
+        HIP_OK(hipMalloc(&d_pats, pats.size() ? pats.size() : 1));
+        if (!pats.empty())
+            HIP_OK(hipMemcpy(d_pats, pats.data(), pats.size(), hipMemcpyHostToDevice));
+    }
+
+    uint8_t *upload_bytes(const void *p, uint64_t n)
+    {
+        uint8_t *d = nullptr;
+        HIP_OK(hipMalloc(&d, n ? n : 1));
+        if (n)
+            HIP_OK(hipMemcpy(d, p, n, hipMemcpyHostToDevice));
+        return d;
+    }
+
+    /* build the visible view for [start, stop_excl); returns device array
+     * (caller frees) + count */
+    void build_view(const std::string *start, const std::string *stop_excl, uint64_t **out_view,
+                    uint64_t *out_n)
+    {
+        int R = (int)runs.size();
+        *out_view = nullptr;
+        *out_n = 0;
+        if (R == 0)
+            return;
+        DevRun *dr = dev_runs();
+        uint64_t *d_lo, *d_hi;
+        HIP_OK(hipMalloc(&d_lo, R * 8));
+        HIP_OK(hipMalloc(&d_hi, R * 8));
+        uint8_t *d_start = start ? upload_bytes(start->data(), start->size()) : nullptr;
+        uint8_t *d_stop = stop_excl ? upload_bytes(stop_excl->data(), stop_excl->size()) : nullptr;
+        launch_bounds(dr, R, d_start, start ? start->size() : 0, d_lo, 0, stream);
+        launch_bounds(dr, R, d_stop, stop_excl ? stop_excl->size() : 0, d_hi, 0, stream);
+        std::vector<uint64_t> lo(R), hi(R), wprefix(R + 1);
+        HIP_OK(hipMemcpyAsync(lo.data(), d_lo, R * 8, hipMemcpyDeviceToHost, stream));
+        HIP_OK(hipMemcpyAsync(hi.data(), d_hi, R * 8, hipMemcpyDeviceToHost, stream));
+        HIP_OK(hipStreamSynchronize(stream));
+        uint64_t total = 0;
+        for (int r = 0; r < R; r++) {
+            if (hi[r] < lo[r])
+                hi[r] = lo[r];
+            wprefix[r] = total;
+            total += hi[r] - lo[r];
+        }
+        wprefix[R] = total;
+        if (d_start)
+            (void)hipFree(d_start);
+        if (d_stop)
+            (void)hipFree(d_stop);
+        if (total == 0) {
+            (void)hipFree(d_lo);
+            (void)hipFree(d_hi);
+            return;
+        }
+        /* re-upload corrected hi + wprefix */
+        HIP_OK(hipMemcpy(d_hi, hi.data(), R * 8, hipMemcpyHostToDevice));
+        uint64_t *d_wp = (uint64_t *)upload_bytes(wprefix.data(), (R + 1) * 8);
+        uint64_t *d_order, *d_flags, *d_pos;
+        HIP_OK(hipMalloc(&d_order, total * 8));
+        HIP_OK(hipMalloc(&d_flags, total * 8));
+        HIP_OK(hipMalloc(&d_pos, total * 8));
+        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, stream);
+        launch_visible(dr, d_order, total, d_flags, stream);
+        launch_psum(d_flags, d_pos, total, stream);
+        uint64_t lastp = 0, lastf = 0;
+        HIP_OK(hipMemcpyAsync(&lastp, d_pos + total - 1, 8, hipMemcpyDeviceToHost, stream));
+        HIP_OK(hipMemcpyAsync(&lastf, d_flags + total - 1, 8, hipMemcpyDeviceToHost, stream));
+        HIP_OK(hipStreamSynchronize(stream));
+        uint64_t nv = lastp + lastf;
+        uint64_t *d_view = nullptr;
+        if (nv) {
+            HIP_OK(hipMalloc(&d_view, nv * 8));
+            launch_gather(d_order, d_flags, d_pos, total, d_view, stream);
+            HIP_OK(hipStreamSynchronize(stream));
+        }
+        (void)hipFree(d_lo);
+        (void)hipFree(d_hi);
+        (void)hipFree(d_wp);
+        (void)hipFree(d_order);
+        (void)hipFree(d_flags);
+        (void)hipFree(d_pos);
+        *out_view = d_view;
+        *out_n = nv;
+    }
+};
+
+std::once_flag g_init_once;
+
+} // namespace
+
+extern "C" {
+
+const char *rrdb_backend(void) { return "hip-gfx950"; }
+
+double rrdb_phase_ms(void *h, const char *phase)
+{
+    auto *e = (HipEngine *)h;
+    auto it = e->phase_ms.find(phase);
+    return it == e->phase_ms.end() ? -1.0 : it->second;
+}
+
+void *rrdb_open(int32_t app_id, int32_t pidx, int32_t gpu_id)
+{
+    if (gpu_id < 0)
+        return nullptr; /* GPU engine requires a device */
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || gpu_id >= ndev) {
+        fprintf(stderr, "rrdb-hip: no HIP device %d (count=%d) — the product engine has no CPU "
+                        "fallback\n",
+                gpu_id, ndev);
+        return nullptr;
+    }
+    auto *e = new HipEngine();
+    e->app_id = app_id;
+    e->pidx = pidx;
+    e->device = gpu_id;
+    e->activate();
+    std::call_once(g_init_once, [] {
+        build_crc64_table();
+        launch_crc64_table_init(host_crc64_table);
+    });
+    HIP_OK(hipStreamCreate(&e->stream));
+    return e;
+}
+
+void rrdb_close(void *h)
+{
+    auto *e = (HipEngine *)h;
+    if (!e)
+        return;
+    e->activate();
+    for (auto &kv : e->ctxs)
+        delete kv.second;
+    e->ctxs.clear();
+    for (auto &r : e->runs)
+        e->free_run(r);
+    if (e->d_runs)
+        (void)hipFree(e->d_runs);
+    if (e->d_ops)
+        (void)hipFree(e->d_ops);
+    if (e->d_rules)
+        (void)hipFree(e->d_rules);
+    if (e->d_pats)
+        (void)hipFree(e->d_pats);
+    (void)hipStreamDestroy(e->stream);
+    delete e;
+}
+
+int32_t rrdb_set_partition_version(void *h, int32_t pv)
+{
+    ((HipEngine *)h)->partition_version = pv;
+    return RRDB_OK;
+}
+
+int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *values, int32_t n)
+{
+    auto *e = (HipEngine *)h;
+    e->activate();
+    for (int32_t i = 0; i < n; i++) {
+        std::string k = keys[i], v = values[i];
+        if (k == "default_ttl") {
+            long long t = atoll(v.c_str());
+            e->default_ttl = t > 0 ? (uint32_t)t : 0;
+        } else if (k == "user_specified_compaction") {
+            std::vector<HostOp> ops;
+            if (!parse_user_ops(v, ops))
+                ops.clear(); /* invalid json -> no ops (cpp:165-169) */
+            e->host_ops = std::move(ops);
+            e->upload_ops();
+        } else if (k == "replica.split.validate_partition_hash") {
+            e->validate_hash = (v == "true");
+        } else if (k == "manual_compact.disabled") {
+            e->manual_compact_disabled = (v == "true");
+        } else if (k == "pegasus.data_version") {
+            e->data_version = (uint32_t)atoi(v.c_str());
+        } else if (k == "replica.rocksdb_iteration_threshold_time_ms") {
+            e->iter_time_ms = (uint64_t)atoll(v.c_str());
+        } else if (k == "rocksdb.max_iteration_count") {
+            e->max_iter_count = (uint32_t)atoll(v.c_str());
+        } else if (k == "rocksdb.multi_get_max_iteration_count") {
+            e->mg_max_iter_count = (uint32_t)atoll(v.c_str());
+        } else if (k == "rocksdb.multi_get_max_iteration_size") {
+            e->mg_max_iter_size = (uint64_t)atoll(v.c_str());
+        }
+    }
+    return RRDB_OK;
+}
+
+int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
+                        const uint8_t *values, const uint64_t *val_offs, const uint64_t *seq_kind,
+                        uint64_t n)
+{
+    auto *e = (HipEngine *)h;
+    if (n == 0)
+        return RRDB_OK;
+    /* validate on host (data arrives host-side anyway) */
+    for (uint64_t i = 0; i < n; i++) {
+        if ((seq_kind[i] >> 1) < e->next_seq_floor)
+            return RRDB_INVALID_ARGUMENT;
+        if (i > 0) {
+            uint64_t al = key_offs[i] - key_offs[i - 1], bl = key_offs[i + 1] - key_offs[i];
+            const uint8_t *a = keys + key_offs[i - 1];
+            const uint8_t *b = keys + key_offs[i];
+            int c = memcmp(a, b, std::min(al, bl));
+            if (c > 0 || (c == 0 && al >= bl))
+                return RRDB_INVALID_ARGUMENT;
+        }
+    }
+    e->activate();
+    RunBuf r;
+    r.n = n;
+    r.keys = e->upload_bytes(keys, key_offs[n]);
+    r.koff = (uint64_t *)e->upload_bytes(key_offs, (n + 1) * 8);
+    r.vals = e->upload_bytes(values, val_offs[n]);
+    r.voff = (uint64_t *)e->upload_bytes(val_offs, (n + 1) * 8);
+    r.sk = (uint64_t *)e->upload_bytes(seq_kind, n * 8);
+    uint64_t mx = 0;
+    for (uint64_t i = 0; i < n; i++)
+        mx = std::max(mx, seq_kind[i] >> 1);
+    e->next_seq_floor = mx + 1;
+    e->runs.push_back(r);
+    e->d_runs_dirty = true;
+    return RRDB_OK;
+}
+
+uint64_t rrdb_num_runs(void *h) { return ((HipEngine *)h)->runs.size(); }
+uint64_t rrdb_num_records(void *h)
+{
+    auto *e = (HipEngine *)h;
+    uint64_t t = 0;
+    for (auto &r : e->runs)
+        t += r.n;
+    return t;
+}
+
+void rrdb_free_result(rrdb_result *r)
+{
+    if (r && r->_arena) {
+        delete (Arena *)r->_arena;
+        r->_arena = nullptr;
+    }
+}
+
+/* ---- batched point lookup core (on_get / on_batch_get / DB::MultiGet) ---- */
+static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const uint64_t *key_offs,
+                        uint32_t epoch_now, std::vector<int32_t> &status,
+                        std::vector<std::pair<const uint8_t *, uint64_t>> &vals, Arena *a)
+{
+    e->activate();
+    int R = (int)e->runs.size();
+    status.assign(nq, RRDB_NOT_FOUND);
+    vals.assign(nq, {nullptr, 0});
+    if (R == 0)
+        return RRDB_OK;
+    DevRun *dr = e->dev_runs();
+    uint8_t *d_keys = e->upload_bytes(keys, key_offs[nq]);
+    uint64_t *d_offs = (uint64_t *)e->upload_bytes(key_offs, (nq + 1) * 8);
+    int32_t *d_status;
+    uint64_t *d_hit, *d_ulen, *d_voffs;
+    HIP_OK(hipMalloc(&d_status, nq * 4));
+    HIP_OK(hipMalloc(&d_hit, nq * 8));
+    HIP_OK(hipMalloc(&d_ulen, nq * 8));
+    HIP_OK(hipMalloc(&d_voffs, (nq + 1) * 8));
+    launch_get(dr, R, d_keys, d_offs, nq, epoch_now, e->data_version, d_status, d_hit, d_ulen,
+               nullptr, e->stream);
+    launch_psum(d_ulen, d_voffs, nq, e->stream);
+    uint64_t last_off = 0, last_len = 0;
+    HIP_OK(hipMemcpyAsync(&last_off, d_voffs + nq - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&last_len, d_ulen + nq - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(status.data(), d_status, nq * 4, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipStreamSynchronize(e->stream));
+    uint64_t total = last_off + last_len;
+    std::vector<uint64_t> voffs(nq + 1);
+    HIP_OK(hipMemcpy(voffs.data(), d_voffs, nq * 8, hipMemcpyDeviceToHost));
+    voffs[nq] = total;
+    uint8_t *blob = (uint8_t *)a->alloc(total);
+    if (total) {
+        uint8_t *d_blob;
+        HIP_OK(hipMalloc(&d_blob, total));
+        launch_emit_values(dr, d_hit, d_status, nq, e->data_version, d_voffs, d_blob, e->stream);
+        HIP_OK(hipMemcpyAsync(blob, d_blob, total, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        (void)hipFree(d_blob);
+    }
+    for (uint64_t i = 0; i < nq; i++)
+        if (status[i] == RRDB_OK)
+            vals[i] = {blob + voffs[i], voffs[i + 1] - voffs[i]};
+    (void)hipFree(d_keys);
+    (void)hipFree(d_offs);
+    (void)hipFree(d_status);
+    (void)hipFree(d_hit);
+    (void)hipFree(d_ulen);
+    (void)hipFree(d_voffs);
+    return RRDB_OK;
+}
+
+int32_t rrdb_get(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_now,
+                 rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    Arena *a = result_init(out);
+    uint64_t offs[2] = {0, key_len};
+    std::vector<int32_t> status;
+    std::vector<std::pair<const uint8_t *, uint64_t>> vals;
+    get_core(e, 1, key, offs, epoch_now, status, vals, a);
+    if (status[0] != RRDB_OK) {
+        out->error = status[0];
+        return out->error;
+    }
+    out->count = 1;
+    out->keys = (rrdb_slice *)a->alloc(sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)a->alloc(sizeof(rrdb_slice));
+    out->keys[0] = {nullptr, 0};
+    out->values[0] = {(uint8_t *)vals[0].first, vals[0].second};
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+int32_t rrdb_batch_get(void *h, uint64_t n_keys, const uint8_t *keys, const uint64_t *key_offs,
+                       uint32_t epoch_now, rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    Arena *a = result_init(out);
+    if (n_keys == 0) {
+        out->error = RRDB_INVALID_ARGUMENT; /* on_batch_get:922-928 */
+        return out->error;
+    }
+    std::vector<int32_t> status;
+    std::vector<std::pair<const uint8_t *, uint64_t>> vals;
+    get_core(e, n_keys, keys, key_offs, epoch_now, status, vals, a);
+    out->keys = (rrdb_slice *)a->alloc(n_keys * sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)a->alloc(n_keys * sizeof(rrdb_slice));
+    uint64_t m = 0;
+    for (uint64_t i = 0; i < n_keys; i++) {
+        if (status[i] != RRDB_OK)
+            continue; /* NotFound/expired skipped (on_batch_get:952-965) */
+        uint64_t kl = key_offs[i + 1] - key_offs[i];
+        uint8_t *kc = (uint8_t *)a->alloc(kl);
+        memcpy(kc, keys + key_offs[i], kl);
+        out->keys[m] = {kc, kl};
+        out->values[m] = {(uint8_t *)vals[i].first, vals[i].second};
+        m++;
+    }
+    out->count = m;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_now,
+                 rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    Arena *a = result_init(out);
+    e->activate();
+    int R = (int)e->runs.size();
+    if (R == 0) {
+        out->error = RRDB_NOT_FOUND;
+        return out->error;
+    }
+    DevRun *dr = e->dev_runs();
+    uint64_t offs[2] = {0, key_len};
+    uint8_t *d_key = e->upload_bytes(key, key_len);
+    uint64_t *d_offs = (uint64_t *)e->upload_bytes(offs, 16);
+    int32_t *d_status;
+    uint64_t *d_hit, *d_ulen;
+    uint32_t *d_expire;
+    HIP_OK(hipMalloc(&d_status, 4));
+    HIP_OK(hipMalloc(&d_hit, 8));
+    HIP_OK(hipMalloc(&d_ulen, 8));
+    HIP_OK(hipMalloc(&d_expire, 4));
+    launch_get(dr, R, d_key, d_offs, 1, epoch_now, e->data_version, d_status, d_hit, d_ulen,
+               d_expire, e->stream);
+    int32_t st;
+    uint32_t expire;
+    HIP_OK(hipMemcpyAsync(&st, d_status, 4, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&expire, d_expire, 4, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipStreamSynchronize(e->stream));
+    (void)hipFree(d_key);
+    (void)hipFree(d_offs);
+    (void)hipFree(d_status);
+    (void)hipFree(d_hit);
+    (void)hipFree(d_ulen);
+    (void)hipFree(d_expire);
+    (void)a;
+    out->error = st;
+    if (st == RRDB_OK)
+        out->i64 = expire > 0 ? (int64_t)expire - (int64_t)epoch_now : -1; /* on_ttl:1135-1142 */
+    return out->error;
+}
+
+/* ---- scan machinery shared by scan/sortkey_count/multi_get-range ---- */
+
+struct BatchOut {
+    uint64_t consumed = 0;  /* view entries consumed */
+    uint64_t n_out = 0;     /* normal rows */
+};
+
+/* run one forward batch over ctx->d_view[cursor..]; emits into result */
+static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb_result *out,
+                           Arena *a)
+{
+    uint32_t batch_count = e->max_iter_count;
+    if (c->batch_size > 0 && (uint32_t)c->batch_size < batch_count)
+        batch_count = (uint32_t)c->batch_size;
+    uint64_t remaining = c->view_n - c->cursor;
+    uint64_t w = std::min<uint64_t>(remaining, e->max_iter_count);
+    if (w == 0) {
+        out->error = RRDB_OK;
+        out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+        if (c->only_return_count)
+            out->i64 = 0;
+        return;
+    }
+    DevRun *dr = e->dev_runs();
+    const uint64_t *d_win = c->d_view + c->cursor;
+    ScanParams sp{};
+    sp.epoch_now = epoch_now;
+    sp.data_version = e->data_version;
+    sp.pidx = e->pidx;
+    sp.partition_version = e->partition_version;
+    sp.validate_hash = (uint8_t)(c->validate_hash_req && e->validate_hash);
+    sp.hk_ft = c->hk_ft;
+    sp.sk_ft = c->sk_ft;
+    sp.hk_pat = c->d_hk_pat;
+    sp.hk_pat_len = c->hk_pat_len;
+    sp.sk_pat = c->d_sk_pat;
+    sp.sk_pat_len = c->sk_pat_len;
+    sp.no_value = c->no_value;
+    sp.hash_key_skip = 0;
+
+    uint8_t *d_state;
+    uint64_t *d_ksz, *d_vsz, *d_flags, *d_npos, *d_cut;
+    HIP_OK(hipMalloc(&d_state, w));
+    HIP_OK(hipMalloc(&d_ksz, w * 8));
+    HIP_OK(hipMalloc(&d_vsz, w * 8));
+    HIP_OK(hipMalloc(&d_flags, w * 8));
+    HIP_OK(hipMalloc(&d_npos, w * 8));
+    HIP_OK(hipMalloc(&d_cut, 16));
+    hipEvent_t sev[4];
+    for (auto &x : sev)
+        HIP_OK(hipEventCreate(&x));
+    HIP_OK(hipEventRecord(sev[0], e->stream));
+    launch_scan_state(dr, d_win, w, sp, d_state, d_ksz, d_vsz, e->stream);
+    HIP_OK(hipEventRecord(sev[1], e->stream));
+    launch_normal_flags(d_state, w, d_flags, e->stream);
+    launch_psum(d_flags, d_npos, w, e->stream);
+    launch_cutoff(d_npos, d_flags, w, batch_count, d_cut, e->stream);
+    uint64_t cut[2];
+    HIP_OK(hipMemcpyAsync(cut, d_cut, 16, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipStreamSynchronize(e->stream));
+    uint64_t consumed = cut[0], n_out = cut[1];
+
+    if (c->only_return_count) {
+        out->i64 = (int64_t)n_out;
+        out->count = 0;
+    } else if (n_out == 0) {
+        out->count = 0;
+    } else {
+        launch_cut_sizes(d_state, w, consumed, d_ksz, d_vsz, e->stream);
+        uint64_t *d_koffs, *d_voffs;
+        HIP_OK(hipMalloc(&d_koffs, w * 8));
+        HIP_OK(hipMalloc(&d_voffs, w * 8));
+        launch_psum(d_ksz, d_koffs, w, e->stream);
+        launch_psum(d_vsz, d_voffs, w, e->stream);
+        uint64_t t[4] = {0, 0, 0, 0};
+        HIP_OK(hipMemcpyAsync(&t[0], d_koffs + w - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&t[1], d_ksz + w - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&t[2], d_voffs + w - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&t[3], d_vsz + w - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        uint64_t kbytes = t[0] + t[1], vbytes = t[2] + t[3];
+        uint8_t *d_kout, *d_vout;
+        uint64_t *d_kooffs, *d_vooffs;
+        int32_t *d_ets = nullptr;
+        HIP_OK(hipMalloc(&d_kout, kbytes ? kbytes : 1));
+        HIP_OK(hipMalloc(&d_vout, vbytes ? vbytes : 1));
+        HIP_OK(hipMalloc(&d_kooffs, (n_out + 1) * 8));
+        HIP_OK(hipMalloc(&d_vooffs, (n_out + 1) * 8));
+        if (c->return_expire_ts)
+            HIP_OK(hipMalloc(&d_ets, n_out * 4));
+        HIP_OK(hipEventRecord(sev[2], e->stream));
+        launch_emit_scan(dr, d_win, w, d_state, consumed, d_npos, d_koffs, d_voffs, sp, d_kout,
+                         d_vout, d_kooffs, d_vooffs, d_ets, n_out, e->stream);
+        HIP_OK(hipEventRecord(sev[3], e->stream));
+        if (c->on_device_out) {
+            HIP_OK(hipStreamSynchronize(e->stream));
+            out->dev_keys = d_kout;
+            out->dev_key_offs = d_kooffs;
+            out->dev_vals = d_vout;
+            out->dev_val_offs = d_vooffs;
+            a->dev_ptrs.insert(a->dev_ptrs.end(), {d_kout, d_vout, d_kooffs, d_vooffs});
+            if (d_ets)
+                (void)hipFree(d_ets);
+        } else {
+            uint8_t *kb = (uint8_t *)a->alloc(kbytes);
+            uint8_t *vb = (uint8_t *)a->alloc(vbytes);
+            std::vector<uint64_t> kooffs(n_out + 1), vooffs(n_out + 1);
+            HIP_OK(hipMemcpyAsync(kb, d_kout, kbytes, hipMemcpyDeviceToHost, e->stream));
+            HIP_OK(hipMemcpyAsync(vb, d_vout, vbytes, hipMemcpyDeviceToHost, e->stream));
+            HIP_OK(hipMemcpyAsync(kooffs.data(), d_kooffs, (n_out + 1) * 8,
+                                  hipMemcpyDeviceToHost, e->stream));
+            HIP_OK(hipMemcpyAsync(vooffs.data(), d_vooffs, (n_out + 1) * 8,
+                                  hipMemcpyDeviceToHost, e->stream));
+            int32_t *ets = nullptr;
+            if (d_ets) {
+                ets = (int32_t *)a->alloc(n_out * 4);
+                HIP_OK(hipMemcpyAsync(ets, d_ets, n_out * 4, hipMemcpyDeviceToHost, e->stream));
+            }
+            HIP_OK(hipStreamSynchronize(e->stream));
+            out->keys = (rrdb_slice *)a->alloc(n_out * sizeof(rrdb_slice));
+            out->values = (rrdb_slice *)a->alloc(n_out * sizeof(rrdb_slice));
+            for (uint64_t i = 0; i < n_out; i++) {
+                out->keys[i] = {kb + kooffs[i], kooffs[i + 1] - kooffs[i]};
+                out->values[i] = {vb + vooffs[i], vooffs[i + 1] - vooffs[i]};
+            }
+            out->expire_ts = ets;
+            (void)hipFree(d_kout);
+            (void)hipFree(d_vout);
+            (void)hipFree(d_kooffs);
+            (void)hipFree(d_vooffs);
+            if (d_ets)
+                (void)hipFree(d_ets);
+        }
+        out->count = n_out;
+        HIP_OK(hipFree(d_koffs));
+        HIP_OK(hipFree(d_voffs));
+    }
+    {
+        float ms;
+        HIP_OK(hipEventSynchronize(sev[1]));
+        HIP_OK(hipEventElapsedTime(&ms, sev[0], sev[1]));
+        e->phase_ms["scan_state"] = ms;
+        if (hipEventQuery(sev[3]) == hipSuccess &&
+            hipEventElapsedTime(&ms, sev[2], sev[3]) == hipSuccess)
+            e->phase_ms["scan_emit"] = ms;
+    }
+    for (auto &x : sev)
+        (void)hipEventDestroy(x);
+    (void)hipFree(d_state);
+    (void)hipFree(d_ksz);
+    (void)hipFree(d_vsz);
+    (void)hipFree(d_flags);
+    (void)hipFree(d_npos);
+    (void)hipFree(d_cut);
+    c->cursor += consumed;
+    out->error = RRDB_OK;
+    out->context_id =
+        (c->cursor >= c->view_n) ? RRDB_SCAN_CONTEXT_ID_COMPLETED : 0 /* caller parks */;
+}
+
+int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    Arena *a = result_init(out);
+    if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
+        q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+        out->error = RRDB_INVALID_ARGUMENT; /* on_get_scanner:1168-1186 */
+        return out->error;
+    }
+    e->activate();
+    std::string start((const char *)q->start_key.data, q->start_key.len);
+    std::string stop((const char *)q->stop_key.data, q->stop_key.len);
+    bool start_inclusive = q->start_inclusive, stop_inclusive = q->stop_inclusive;
+    /* hash-key prefix filter clamps start (on_get_scanner:1206-1224) */
+    if (q->hash_key_filter_type == RRDB_FT_MATCH_PREFIX && q->hash_key_filter_pattern.len > 0) {
+        std::string ps =
+            make_key(q->hash_key_filter_pattern.data, q->hash_key_filter_pattern.len, nullptr, 0);
+        if (key_cmp(ps, start) > 0) {
+            start = ps;
+            start_inclusive = true;
+        }
+    }
+    int c = key_cmp(start, stop);
+    if (c > 0 || (c == 0 && (!start_inclusive || !stop_inclusive))) {
+        out->error = RRDB_OK; /* empty range (on_get_scanner:1227-1243) */
+        out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+        return RRDB_OK;
+    }
+    auto *ctx = new HipScanCtx();
+    std::string stop_excl = stop;
+    if (stop_inclusive)
+        stop_excl.push_back('\0'); /* smallest key > stop */
+    e->build_view(&start, &stop_excl, &ctx->d_view, &ctx->view_n);
+    /* first_exclusive: skip an exact start match (on_get_scanner:1277-1283) */
+    if (!start_inclusive && ctx->view_n > 0) {
+        uint32_t *d_eq;
+        HIP_OK(hipMalloc(&d_eq, 4));
+        uint8_t *d_sk = e->upload_bytes(start.data(), start.size());
+        launch_first_eq(e->dev_runs(), ctx->d_view, ctx->view_n, d_sk, start.size(), d_eq,
+                        e->stream);
+        uint32_t eq = 0;
+        HIP_OK(hipMemcpyAsync(&eq, d_eq, 4, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        (void)hipFree(d_eq);
+        (void)hipFree(d_sk);
+        if (eq)
+            ctx->cursor = 1;
+    }
+    ctx->batch_size = q->batch_size;
+    ctx->no_value = q->no_value;
+    ctx->validate_hash_req = q->validate_partition_hash;
+    ctx->return_expire_ts = q->return_expire_ts;
+    ctx->only_return_count = q->only_return_count;
+    ctx->on_device_out = q->on_device_out;
+    ctx->hk_ft = q->hash_key_filter_type;
+    ctx->sk_ft = q->sort_key_filter_type;
+    ctx->hk_pat_len = q->hash_key_filter_pattern.len;
+    ctx->d_hk_pat = e->upload_bytes(q->hash_key_filter_pattern.data, ctx->hk_pat_len);
+    ctx->sk_pat_len = q->sort_key_filter_pattern.len;
+    ctx->d_sk_pat = e->upload_bytes(q->sort_key_filter_pattern.data, ctx->sk_pat_len);
+
+    scan_batch_gpu(e, ctx, epoch_now, out, a);
+    if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
+        delete ctx;
+    } else {
+        std::lock_guard<std::mutex> g(e->mu);
+        ctx->id = ++e->next_ctx_id;
+        e->ctxs[ctx->id] = ctx;
+        out->context_id = ctx->id;
+    }
+    return out->error;
+}
+
+int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    Arena *a = result_init(out);
+    e->activate();
+    HipScanCtx *c = nullptr;
+    {
+        std::lock_guard<std::mutex> g(e->mu);
+        auto it = e->ctxs.find(context_id);
+        if (it != e->ctxs.end()) {
+            c = it->second;
+            e->ctxs.erase(it);
+        }
+    }
+    if (!c) {
+        out->error = RRDB_NOT_FOUND; /* on_scan:1539-1541 */
+        return out->error;
+    }
+    scan_batch_gpu(e, c, epoch_now, out, a);
+    if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
+        delete c;
+    } else {
+        std::lock_guard<std::mutex> g(e->mu);
+        c->id = ++e->next_ctx_id; /* re-park under a fresh handle (on_scan:1516-1526) */
+        e->ctxs[c->id] = c;
+        out->context_id = c->id;
+    }
+    return out->error;
+}
+
+void rrdb_clear_scanner(void *h, int64_t context_id)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    auto it = e->ctxs.find(context_id);
+    if (it != e->ctxs.end()) {
+        e->activate();
+        delete it->second;
+        e->ctxs.erase(it);
+    }
+}
+
+int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uint32_t epoch_now,
+                           rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    result_init(out);
+    e->activate();
+    /* start=(hk,""), stop=next(hk) (on_sortkey_count:1030-1036); count all
+     * visible non-expired rows — no count cap in the reference's loop */
+    std::string start = make_key(hash_key, hklen, nullptr, 0);
+    std::string stop = next_blob(start);
+    uint64_t *d_view = nullptr, n = 0;
+    e->build_view(&start, &stop, &d_view, &n);
+    int64_t count = 0;
+    if (n) {
+        DevRun *dr = e->dev_runs();
+        ScanParams sp{};
+        sp.epoch_now = epoch_now;
+        sp.data_version = e->data_version;
+        sp.pidx = e->pidx;
+        sp.partition_version = e->partition_version;
+        sp.validate_hash = 0;
+        uint8_t *d_state;
+        uint64_t *d_ksz, *d_vsz, *d_flags, *d_npos;
+        HIP_OK(hipMalloc(&d_state, n));
+        HIP_OK(hipMalloc(&d_ksz, n * 8));
+        HIP_OK(hipMalloc(&d_vsz, n * 8));
+        HIP_OK(hipMalloc(&d_flags, n * 8));
+        HIP_OK(hipMalloc(&d_npos, n * 8));
+        launch_scan_state(dr, d_view, n, sp, d_state, d_ksz, d_vsz, e->stream);
+        launch_normal_flags(d_state, n, d_flags, e->stream);
+        launch_psum(d_flags, d_npos, n, e->stream);
+        uint64_t lastp = 0, lastf = 0;
+        HIP_OK(hipMemcpyAsync(&lastp, d_npos + n - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&lastf, d_flags + n - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        count = (int64_t)(lastp + lastf);
+        (void)hipFree(d_state);
+        (void)hipFree(d_ksz);
+        (void)hipFree(d_vsz);
+        (void)hipFree(d_flags);
+        (void)hipFree(d_npos);
+    }
+    if (d_view)
+        (void)hipFree(d_view);
+    out->i64 = count;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}
+
+int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
+                       rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    Arena *a = result_init(out);
+    e->activate();
+    if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+        out->error = RRDB_INVALID_ARGUMENT; /* on_multi_get:508-517 */
+        return out->error;
+    }
+    uint32_t max_kv_count = e->mg_max_iter_count;
+    if (q->max_kv_count > 0 && (uint32_t)q->max_kv_count < max_kv_count)
+        max_kv_count = (uint32_t)q->max_kv_count;
+    int64_t max_kv_size = q->max_kv_size > 0 ? q->max_kv_size : INT32_MAX;
+    int64_t max_iter_size = std::min<int64_t>(
+        max_kv_size, e->mg_max_iter_size > 0 ? (int64_t)e->mg_max_iter_size : INT32_MAX);
+    uint32_t max_iteration_count = e->mg_max_iter_count;
+
+    if (q->n_sort_keys > 0) {
+        /* point-list variant (on_multi_get:779-860) over the batched-get core */
+        std::vector<uint64_t> offs(q->n_sort_keys + 1);
+        std::string keys;
+        for (uint64_t i = 0; i < q->n_sort_keys; i++) {
+            offs[i] = keys.size();
+            keys += make_key(q->hash_key.data, q->hash_key.len,
+                             q->sort_keys + q->sort_key_offs[i],
+                             q->sort_key_offs[i + 1] - q->sort_key_offs[i]);
+        }
+        offs[q->n_sort_keys] = keys.size();
+        std::vector<int32_t> status;
+        std::vector<std::pair<const uint8_t *, uint64_t>> vals;
+        get_core(e, q->n_sort_keys, (const uint8_t *)keys.data(), offs.data(), epoch_now, status,
+                 vals, a);
+        out->keys = (rrdb_slice *)a->alloc(q->n_sort_keys * sizeof(rrdb_slice));
+        out->values = (rrdb_slice *)a->alloc(q->n_sort_keys * sizeof(rrdb_slice));
+        uint64_t m = 0;
+        int64_t count = 0, size = 0;
+        bool exceed = false;
+        for (uint64_t i = 0; i < q->n_sort_keys; i++) {
+            if (status[i] != RRDB_OK)
+                continue;
+            if (count >= (int64_t)max_kv_count || size >= max_kv_size) {
+                exceed = true; /* :837-841 */
+                break;
+            }
+            uint64_t sklen = q->sort_key_offs[i + 1] - q->sort_key_offs[i];
+            uint8_t *kc = (uint8_t *)a->alloc(sklen);
+            memcpy(kc, q->sort_keys + q->sort_key_offs[i], sklen);
+            out->keys[m] = {kc, sklen};
+            if (!q->no_value)
+                out->values[m] = {(uint8_t *)vals[i].first, vals[i].second};
+            else
+                out->values[m] = {nullptr, 0};
+            count++;
+            size += (int64_t)out->keys[m].len + (int64_t)out->values[m].len;
+            m++;
+        }
+        out->count = m;
+        out->error = exceed ? RRDB_INCOMPLETE : RRDB_OK;
+        return out->error;
+    }
+
+    /* range variant (on_multi_get:540-778) */
+    std::string start =
+        make_key(q->hash_key.data, q->hash_key.len, q->start_sortkey.data, q->start_sortkey.len);
+    std::string stop;
+    bool start_inclusive = q->start_inclusive, stop_inclusive;
+    if (q->stop_sortkey.len == 0) {
+        stop = next_blob(make_key(q->hash_key.data, q->hash_key.len, nullptr, 0));
+        stop_inclusive = false;
+    } else {
+        stop = make_key(q->hash_key.data, q->hash_key.len, q->stop_sortkey.data,
+                        q->stop_sortkey.len);
+        stop_inclusive = q->stop_inclusive;
+    }
+    if (q->sort_key_filter_type == RRDB_FT_MATCH_PREFIX && q->sort_key_filter_pattern.len > 0) {
+        std::string ps = make_key(q->hash_key.data, q->hash_key.len,
+                                  q->sort_key_filter_pattern.data, q->sort_key_filter_pattern.len);
+        std::string pe = next_blob(ps);
+        if (key_cmp(ps, start) > 0) {
+            start = ps;
+            start_inclusive = true;
+        }
+        if (key_cmp(pe, stop) <= 0) {
+            stop = pe;
+            stop_inclusive = false;
+        }
+    }
+    int c = key_cmp(start, stop);
+    if (c > 0 || (c == 0 && (!start_inclusive || !stop_inclusive))) {
+        out->error = RRDB_OK; /* empty range (:580-607) */
+        return RRDB_OK;
+    }
+    std::string stop_excl = stop;
+    if (stop_inclusive)
+        stop_excl.push_back('\0');
+    uint64_t *d_view = nullptr, n = 0;
+    e->build_view(&start, &stop_excl, &d_view, &n);
+    if (n == 0) {
+        if (d_view)
+            (void)hipFree(d_view);
+        out->error = RRDB_OK;
+        return RRDB_OK;
+    }
+    /* window: the limiter consumes at most mg_max_iter_count iterations plus
+     * one possible uncounted first/last-exclusive skip */
+    uint64_t wcap = (uint64_t)max_iteration_count + 1;
+    uint64_t wstart = 0, w = std::min<uint64_t>(n, wcap);
+    if (q->reverse)
+        wstart = n - w;
+    DevRun *dr = e->dev_runs();
+    ScanParams sp{};
+    sp.epoch_now = epoch_now;
+    sp.data_version = e->data_version;
+    sp.pidx = e->pidx;
+    sp.partition_version = e->partition_version;
+    sp.validate_hash = 0; /* multi_get does no hash validation */
+    sp.hk_ft = 0;
+    sp.sk_ft = q->sort_key_filter_type;
+    sp.sk_pat = nullptr;
+    sp.sk_pat_len = q->sort_key_filter_pattern.len;
+    uint8_t *d_skpat = e->upload_bytes(q->sort_key_filter_pattern.data, sp.sk_pat_len);
+    sp.sk_pat = d_skpat;
+    sp.no_value = q->no_value;
+    sp.hash_key_skip = 2 + q->hash_key.len; /* emit sortkey only (:2496-2499) */
+
+    uint8_t *d_state;
+    uint64_t *d_ksz, *d_vsz;
+    HIP_OK(hipMalloc(&d_state, w));
+    HIP_OK(hipMalloc(&d_ksz, w * 8));
+    HIP_OK(hipMalloc(&d_vsz, w * 8));
+    /* sortkey filter types: scan_state's sk filter operates on key minus
+     * 2+hklen prefix — same as multi_get's sortkey check */
+    launch_scan_state(dr, d_view + wstart, w, sp, d_state, d_ksz, d_vsz, e->stream);
+    std::vector<uint8_t> state(w);
+    std::vector<uint64_t> ksz(w), vsz(w);
+    HIP_OK(hipMemcpyAsync(state.data(), d_state, w, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(ksz.data(), d_ksz, w * 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(vsz.data(), d_vsz, w * 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipStreamSynchronize(e->stream));
+    (void)hipFree(d_state);
+    (void)hipFree(d_ksz);
+    (void)hipFree(d_vsz);
+
+    /* host limiter loop, mirroring on_multi_get:616-778 exactly */
+    int64_t count = 0, size = 0;
+    uint64_t iteration_count = 0;
+    std::vector<uint64_t> sel; /* view positions to emit, ascending */
+    bool complete = false;
+    bool skipped_first = false;
+    uint64_t steps = w;
+    for (uint64_t s = 0; s < steps; s++) {
+        uint64_t wi = q->reverse ? (steps - 1 - s) : s;
+        if (count >= (int64_t)max_kv_count || iteration_count >= max_iteration_count ||
+            size >= max_iter_size)
+            break;
+        /* first-exclusive skip of the exact boundary key */
+        if (s == 0) {
+            /* check boundary equality via sizes: need the key — use a cheap
+             * device compare */
+            bool check = (!q->reverse && !start_inclusive) || (q->reverse && !stop_inclusive);
+            if (check) {
+                const std::string &bound = q->reverse ? stop : start;
+                uint32_t *d_eq;
+                HIP_OK(hipMalloc(&d_eq, 4));
+                uint8_t *d_b = e->upload_bytes(bound.data(), bound.size());
+                launch_first_eq(dr, d_view + wstart + wi, 1, d_b, bound.size(), d_eq, e->stream);
+                uint32_t eq = 0;
+                HIP_OK(hipMemcpyAsync(&eq, d_eq, 4, hipMemcpyDeviceToHost, e->stream));
+                HIP_OK(hipStreamSynchronize(e->stream));
+                (void)hipFree(d_eq);
+                (void)hipFree(d_b);
+                if (eq) {
+                    skipped_first = true;
+                    continue;
+                }
+            }
+        }
+        iteration_count++;
+        if (state[wi] != ST_NORMAL)
+            continue; /* kExpired / kFiltered */
+        sel.push_back(wstart + wi);
+        count++;
+        size += (int64_t)ksz[wi] + (int64_t)vsz[wi];
+        if (s + 1 == steps)
+            complete = true;
+    }
+    {
+        uint64_t total = n - (skipped_first ? 1 : 0);
+        /* also: entries outside the window were never reachable */
+        if (w < n)
+            complete = false;
+        else
+            complete = (iteration_count >= total);
+    }
+    uint64_t m = sel.size();
+    if (q->reverse)
+        std::reverse(sel.begin(), sel.end()); /* ascending output (:758-765) */
+    out->keys = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
+    if (m > 0) {
+        std::vector<uint64_t> koffs(m + 1), voffs(m + 1);
+        uint64_t kb = 0, vb = 0;
+        for (uint64_t j = 0; j < m; j++) {
+            uint64_t wi = sel[j] - wstart;
+            koffs[j] = kb;
+            voffs[j] = vb;
+            kb += ksz[wi];
+            vb += vsz[wi];
+        }
+        koffs[m] = kb;
+        voffs[m] = vb;
+        uint64_t *d_rows = (uint64_t *)e->upload_bytes(sel.data(), m * 8);
+        uint64_t *d_ko = (uint64_t *)e->upload_bytes(koffs.data(), (m + 1) * 8);
+        uint64_t *d_vo = (uint64_t *)e->upload_bytes(voffs.data(), (m + 1) * 8);
+        uint8_t *d_kout, *d_vout;
+        HIP_OK(hipMalloc(&d_kout, kb ? kb : 1));
+        HIP_OK(hipMalloc(&d_vout, vb ? vb : 1));
+        launch_emit_rows(dr, d_view, d_rows, m, d_ko, d_vo, sp, d_kout, d_vout, e->stream);
+        uint8_t *hkb = (uint8_t *)a->alloc(kb);
+        uint8_t *hvb = (uint8_t *)a->alloc(vb);
+        HIP_OK(hipMemcpyAsync(hkb, d_kout, kb, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(hvb, d_vout, vb, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        for (uint64_t j = 0; j < m; j++) {
+            out->keys[j] = {hkb + koffs[j], koffs[j + 1] - koffs[j]};
+            out->values[j] = {hvb + voffs[j], voffs[j + 1] - voffs[j]};
+        }
+        (void)hipFree(d_rows);
+        (void)hipFree(d_ko);
+        (void)hipFree(d_vo);
+        (void)hipFree(d_kout);
+        (void)hipFree(d_vout);
+    }
+    (void)hipFree(d_skpat);
+    (void)hipFree(d_view);
+    out->count = m;
+    out->error = complete ? RRDB_OK : RRDB_INCOMPLETE; /* :789-799 */
+    return out->error;
+}
+
+int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
+                            rrdb_compact_stats *stats)
+{
+    auto *e = (HipEngine *)h;
+    rrdb_compact_stats st{};
+    if (e->manual_compact_disabled) {
+        if (stats)
+            *stats = st;
+        return RRDB_INVALID_ARGUMENT;
+    }
+    e->activate();
+    int R = (int)e->runs.size();
+    uint64_t total = 0;
+    for (auto &r : e->runs)
+        total += r.n;
+    st.input_records = total;
+    if (total == 0) {
+        if (stats)
+            *stats = st;
+        return RRDB_OK;
+    }
+    DevRun *dr = e->dev_runs();
+    /* full-range rank merge */
+    std::vector<uint64_t> lo(R, 0), hi(R), wprefix(R + 1);
+    uint64_t acc = 0;
+    for (int r = 0; r < R; r++) {
+        hi[r] = e->runs[r].n;
+        wprefix[r] = acc;
+        acc += e->runs[r].n;
+    }
+    wprefix[R] = acc;
+    uint64_t *d_lo = (uint64_t *)e->upload_bytes(lo.data(), R * 8);
+    uint64_t *d_hi = (uint64_t *)e->upload_bytes(hi.data(), R * 8);
+    uint64_t *d_wp = (uint64_t *)e->upload_bytes(wprefix.data(), (R + 1) * 8);
+    uint64_t *d_order;
+    HIP_OK(hipMalloc(&d_order, total * 8));
+    hipEvent_t ev[6];
+    for (auto &x : ev)
+        HIP_OK(hipEventCreate(&x));
+    HIP_OK(hipEventRecord(ev[0], e->stream));
+    launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, e->stream);
+    HIP_OK(hipEventRecord(ev[1], e->stream));
+
+    CompactParams cp{};
+    cp.epoch_now = epoch_now;
+    cp.data_version = e->data_version;
+    cp.default_ttl = e->default_ttl;
+    cp.pidx = e->pidx;
+    cp.partition_version = e->partition_version;
+    cp.validate_hash = e->validate_hash ? 1 : 0;
+    cp.ops = e->d_ops;
+    cp.n_ops = e->n_ops;
+    cp.rules = e->d_rules;
+    cp.pats = e->d_pats;
+
+    uint8_t *d_keep, *d_changed;
+    uint32_t *d_new_expire;
+    uint64_t *d_ksz, *d_vsz;
+    CompactStatsDev *d_stats;
+    HIP_OK(hipMalloc(&d_keep, total));
+    HIP_OK(hipMalloc(&d_changed, total));
+    HIP_OK(hipMalloc(&d_new_expire, total * 4));
+    HIP_OK(hipMalloc(&d_ksz, total * 8));
+    HIP_OK(hipMalloc(&d_vsz, total * 8));
+    HIP_OK(hipMalloc(&d_stats, sizeof(CompactStatsDev)));
+    HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
+    launch_compact_flags(dr, d_order, total, cp, d_keep, d_changed, d_new_expire, d_ksz, d_vsz,
+                         d_stats, e->stream);
+    HIP_OK(hipEventRecord(ev[2], e->stream));
+    /* positions + offsets */
+    uint64_t *d_keepw, *d_kpos, *d_koffs, *d_voffs;
+    HIP_OK(hipMalloc(&d_keepw, total * 8));
+    HIP_OK(hipMalloc(&d_kpos, total * 8));
+    HIP_OK(hipMalloc(&d_koffs, total * 8));
+    HIP_OK(hipMalloc(&d_voffs, total * 8));
+    launch_widen_u8(d_keep, total, d_keepw, e->stream);
+    launch_psum(d_keepw, d_kpos, total, e->stream);
+    launch_psum(d_ksz, d_koffs, total, e->stream);
+    launch_psum(d_vsz, d_voffs, total, e->stream);
+    uint64_t t[6] = {0};
+    HIP_OK(hipMemcpyAsync(&t[0], d_kpos + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&t[1], d_keepw + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&t[2], d_koffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&t[3], d_ksz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&t[4], d_voffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipMemcpyAsync(&t[5], d_vsz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    HIP_OK(hipStreamSynchronize(e->stream));
+    uint64_t n_out = t[0] + t[1], kbytes = t[2] + t[3], vbytes = t[4] + t[5];
+
+    CompactStatsDev hs{};
+    HIP_OK(hipMemcpy(&hs, d_stats, sizeof(hs), hipMemcpyDeviceToHost));
+    st.output_records = n_out;
+    st.expired = hs.expired;
+    st.filtered = hs.filtered;
+    st.tombstones = hs.tombstones;
+    st.shadowed = hs.shadowed;
+    st.output_bytes = kbytes + vbytes;
+
+    RunBuf nr;
+    if (n_out > 0) {
+        nr.n = n_out;
+        HIP_OK(hipMalloc(&nr.keys, kbytes ? kbytes : 1));
+        HIP_OK(hipMalloc(&nr.vals, vbytes ? vbytes : 1));
+        HIP_OK(hipMalloc(&nr.koff, (n_out + 1) * 8));
+        HIP_OK(hipMalloc(&nr.voff, (n_out + 1) * 8));
+        HIP_OK(hipMalloc(&nr.sk, n_out * 8));
+        HIP_OK(hipEventRecord(ev[3], e->stream));
+        launch_emit_compact(dr, d_order, total, d_keep, d_changed, d_new_expire, d_kpos, d_koffs,
+                            d_voffs, e->data_version, nr.keys, nr.vals, nr.koff, nr.voff, nr.sk,
+                            n_out, e->stream);
+        HIP_OK(hipEventRecord(ev[4], e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        float ms;
+        HIP_OK(hipEventElapsedTime(&ms, ev[3], ev[4]));
+        e->phase_ms["compact_emit"] = ms;
+        HIP_OK(hipEventElapsedTime(&ms, ev[0], ev[4]));
+        e->phase_ms["compact_total"] = ms;
+    }
+    {
+        float ms;
+        HIP_OK(hipEventSynchronize(ev[2]));
+        HIP_OK(hipEventElapsedTime(&ms, ev[0], ev[1]));
+        e->phase_ms["compact_rank"] = ms;
+        HIP_OK(hipEventElapsedTime(&ms, ev[1], ev[2]));
+        e->phase_ms["compact_flags"] = ms;
+    }
+    for (auto &x : ev)
+        (void)hipEventDestroy(x);
+    (void)hipFree(d_lo);
+    (void)hipFree(d_hi);
+    (void)hipFree(d_wp);
+    (void)hipFree(d_order);
+    (void)hipFree(d_keep);
+    (void)hipFree(d_changed);
+    (void)hipFree(d_new_expire);
+    (void)hipFree(d_ksz);
+    (void)hipFree(d_vsz);
+    (void)hipFree(d_stats);
+    (void)hipFree(d_keepw);
+    (void)hipFree(d_kpos);
+    (void)hipFree(d_koffs);
+    (void)hipFree(d_voffs);
+    if (opts && opts->keep_inputs) {
+        /* benchmarking: the pass ran in full; drop the output, keep inputs */
+        if (n_out > 0)
+            e->free_run(nr);
+    } else {
+        for (auto &r : e->runs)
+            e->free_run(r);
+        e->runs.clear();
+        if (n_out > 0)
+            e->runs.push_back(nr);
+        e->d_runs_dirty = true;
+    }
+    if (stats)
+        *stats = st;
+    return RRDB_OK;
+}
+
+} /* extern "C" */

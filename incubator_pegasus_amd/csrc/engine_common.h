@@ -1,0 +1,95 @@
+/* engine_common.h — shared host/device types for the MI355X rrdb engine.
+ *
+ * The engine holds each partition's sorted runs resident in HBM:
+ *   keys blob + u64 offsets, values blob + u64 offsets, and
+ *   seq_kind[i] = (seqno<<1)|kind per record (kind: 0=PUT, 1=DELETE).
+ * Runs are ordered oldest -> newest (ingest order); every seqno in run r+1
+ * is greater than every seqno in run r (enforced at ingest), which makes
+ * "newest version wins" a run-index comparison for equal keys.
+ *
+ * Merge strategy (MI355X-native, no sequential merging iterator):
+ *   rank(record) = own index + sum over other runs of bounded binary-search
+ *   counts with (key asc, run desc) tie-break; scatter by rank; a record is
+ *   shadowed iff its predecessor in rank order carries the same key.  This is
+ *   fully data-parallel and replaces rocksdb's merging-iterator heap
+ *   (reference behavior described in SURVEY.md §3.2/§8(c)).
+ */
+#pragma once
+#include <stdint.h>
+
+#define RRDB_MAX_RUNS 64
+
+/* device-visible descriptor of one sorted run */
+struct DevRun {
+    const uint8_t *keys;
+    const uint64_t *koff; /* n+1 */
+    const uint8_t *vals;
+    const uint64_t *voff; /* n+1 */
+    const uint64_t *sk;   /* (seq<<1)|kind */
+    uint64_t n;
+};
+
+/* flattened user compaction rules/ops (device-resident)
+ * mirrors compaction_filter_rule.{h,cpp} / compaction_operation.{h,cpp} */
+enum { DFR_HASHKEY = 0, DFR_SORTKEY = 1, DFR_TTL_RANGE = 2 };
+enum { DSM_ANYWHERE = 0, DSM_PREFIX = 1, DSM_POSTFIX = 2, DSM_INVALID = 3 };
+enum { DOP_UPDATE_TTL = 0, DOP_DELETE = 1 };
+enum { DUT_FROM_NOW = 0, DUT_FROM_CURRENT = 1, DUT_TIMESTAMP = 2, DUT_INVALID = 3 };
+
+struct DevRule {
+    int32_t type;       /* DFR_* */
+    int32_t match_type; /* DSM_* (pattern rules) */
+    uint32_t start_ttl, stop_ttl;
+    uint32_t pat_off, pat_len; /* into pattern blob */
+};
+
+struct DevOp {
+    int32_t type;    /* DOP_* */
+    int32_t ut_type; /* DUT_* */
+    uint32_t ut_value;
+    int32_t rule_off, n_rules;
+};
+
+/* per-record scan/merge states (match the reference's range_iteration_state,
+ * pegasus_server_impl.h) */
+enum {
+    ST_NORMAL = 0,
+    ST_EXPIRED = 1,
+    ST_FILTERED = 2,
+    ST_HASH_INVALID = 3,
+};
+
+/* scan/filter parameter block passed to kernels by value */
+struct ScanParams {
+    uint32_t epoch_now;
+    uint32_t data_version; /* value header: v0=4B, v1=12B, v2=13B */
+    int32_t pidx;
+    int32_t partition_version;
+    uint8_t validate_hash;      /* engine-level env && request flag */
+    int32_t hk_ft, sk_ft;       /* filter types (0..3) */
+    const uint8_t *hk_pat;      /* device */
+    uint64_t hk_pat_len;
+    const uint8_t *sk_pat;
+    uint64_t sk_pat_len;
+    uint8_t no_value;
+    uint64_t hash_key_skip;     /* multi_get: bytes of [len][hashkey] prefix to
+                                   strip from emitted keys (0 for scan) */
+};
+
+struct CompactParams {
+    uint32_t epoch_now;
+    uint32_t data_version;
+    uint32_t default_ttl;
+    int32_t pidx;
+    int32_t partition_version;
+    uint8_t validate_hash;
+    const DevOp *ops;     /* device */
+    int32_t n_ops;
+    const DevRule *rules; /* device */
+    const uint8_t *pats;  /* device pattern blob */
+};
+
+/* compact per-record disposition written by the filter kernel */
+struct CompactStatsDev {
+    unsigned long long expired, filtered, tombstones, shadowed, output_records;
+};

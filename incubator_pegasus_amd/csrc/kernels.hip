@@ -1,0 +1,934 @@
+/* kernels.hip — gfx950 (CDNA4) device kernels for the rrdb engine.
+ *
+ * Replaces, from scratch (no ported code): rocksdb's point-Get / MultiGet
+ * block path, the merging iterator (k-way heap) and the compaction
+ * iterator + KeyWithTTLCompactionFilter evaluation that the reference's
+ * pegasus_server_impl drives (call sites: pegasus_server_impl.cpp:441,618,
+ * 804,948,1243,3389).
+ *
+ * Design notes (MI355X):
+ *  - wave64; blocks of 256 threads; grids capped ~2048 blocks + grid-stride
+ *  - merge is rank-based and fully data-parallel (engine_common.h header
+ *    comment); no sequential heap anywhere
+ *  - emit copies are wave-per-record with lane-strided bytes: consecutive
+ *    lanes touch consecutive bytes -> coalesced 64B segments
+ *  - crc64 table lives in __device__ memory (L1/L2-resident after first use)
+ *  - integer/byte work only: HBM-bound, no MFMA by design
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <stdio.h>
+
+#include "engine_common.h"
+
+#define WAVE 64
+#define BLOCK 256
+#define MAX_GRID 2048
+#define PSUM_ITEMS_PER_THREAD 16
+#define PSUM_BLOCK_ITEMS (BLOCK * PSUM_ITEMS_PER_THREAD) /* 4096 */
+
+static inline int grid_for(uint64_t n, int per_block)
+{
+    uint64_t b = (n + per_block - 1) / per_block;
+    if (b == 0)
+        b = 1;
+    if (b > MAX_GRID)
+        b = MAX_GRID;
+    return (int)b;
+}
+
+#define HIP_CHECK(x)                                                                               \
+    do {                                                                                           \
+        hipError_t err_ = (x);                                                                     \
+        if (err_ != hipSuccess) {                                                                  \
+            fprintf(stderr, "rrdb-hip fatal: %s at %s:%d\n", hipGetErrorString(err_), __FILE__,    \
+                    __LINE__);                                                                     \
+            abort();                                                                               \
+        }                                                                                          \
+    } while (0)
+
+/* ================= crc64 (device) =================
+ * Same polynomial/table construction as the reference crc.cpp:236-295 —
+ * restated; table built on host (launchers.cpp) and copied here. */
+__device__ uint64_t d_crc64_table[256];
+
+__device__ static inline uint64_t dev_crc64(const uint8_t *p, uint64_t n)
+{
+    uint64_t crc = ~0ull;
+    for (uint64_t i = 0; i < n; i++)
+        crc = d_crc64_table[(uint8_t)(crc ^ p[i])] ^ (crc >> 8);
+    return ~crc;
+}
+
+/* pegasus_key_hash (pegasus_key_schema.h:148-165) */
+__device__ static inline uint64_t dev_key_hash(const uint8_t *key, uint64_t len)
+{
+    uint32_t hklen = ((uint32_t)key[0] << 8) | key[1];
+    if (hklen > 0)
+        return dev_crc64(key + 2, hklen);
+    return dev_crc64(key + 2, len - 2);
+}
+
+/* ================= byte compare ================= */
+__device__ static inline int dev_key_cmp(const uint8_t *a, uint64_t alen, const uint8_t *b,
+                                         uint64_t blen)
+{
+    uint64_t m = alen < blen ? alen : blen;
+    uint64_t i = 0;
+    for (; i + 8 <= m; i += 8) {
+        uint64_t wa, wb;
+        __builtin_memcpy(&wa, a + i, 8);
+        __builtin_memcpy(&wb, b + i, 8);
+        if (wa != wb) {
+            wa = __builtin_bswap64(wa);
+            wb = __builtin_bswap64(wb);
+            return wa < wb ? -1 : 1;
+        }
+    }
+    for (; i < m; i++)
+        if (a[i] != b[i])
+            return a[i] < b[i] ? -1 : 1;
+    return alen < blen ? -1 : (alen > blen ? 1 : 0);
+}
+
+__device__ static inline const uint8_t *run_key(const DevRun &r, uint64_t i, uint64_t *len)
+{
+    uint64_t o = r.koff[i];
+    *len = r.koff[i + 1] - o;
+    return r.keys + o;
+}
+__device__ static inline const uint8_t *run_val(const DevRun &r, uint64_t i, uint64_t *len)
+{
+    uint64_t o = r.voff[i];
+    *len = r.voff[i + 1] - o;
+    return r.vals + o;
+}
+
+/* first index in [lo,hi) with key >= target */
+__device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
+                                           uint64_t lo, uint64_t hi)
+{
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1, ml;
+        const uint8_t *mk = run_key(r, mid, &ml);
+        if (dev_key_cmp(mk, ml, key, klen) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+/* first index in [lo,hi) with key > target */
+__device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
+                                           uint64_t lo, uint64_t hi)
+{
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1, ml;
+        const uint8_t *mk = run_key(r, mid, &ml);
+        if (dev_key_cmp(mk, ml, key, klen) <= 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+
+/* ================= value codec (device) =================
+ * pegasus_value_schema.h:58-125 / value_schema_v2.cpp:101-125 */
+__device__ static inline uint32_t dev_hdr_len(uint32_t ver)
+{
+    return ver == 0 ? 4u : (ver == 1 ? 12u : 13u);
+}
+__device__ static inline uint32_t dev_expire_ts(uint32_t ver, const uint8_t *v)
+{
+    uint32_t off = (ver == 2) ? 1 : 0;
+    return ((uint32_t)v[off] << 24) | ((uint32_t)v[off + 1] << 16) | ((uint32_t)v[off + 2] << 8) |
+           (uint32_t)v[off + 3];
+}
+__device__ static inline int dev_ts_expired(uint32_t now, uint32_t ts)
+{
+    return ts > 0 && ts <= now;
+}
+
+/* ================= pattern / rules (device) =================
+ * compaction_filter_rule.cpp:31-90, compaction_operation.cpp:33-113 */
+__device__ static int dev_mem_eq(const uint8_t *a, const uint8_t *b, uint64_t n)
+{
+    for (uint64_t i = 0; i < n; i++)
+        if (a[i] != b[i])
+            return 0;
+    return 1;
+}
+
+__device__ static int dev_pattern_match(const uint8_t *v, uint64_t vlen, int type,
+                                        const uint8_t *pat, uint64_t plen)
+{
+    if (plen == 0 || vlen < plen)
+        return 0;
+    switch (type) {
+    case DSM_ANYWHERE:
+        for (uint64_t i = 0; i + plen <= vlen; i++)
+            if (dev_mem_eq(v + i, pat, plen))
+                return 1;
+        return 0;
+    case DSM_PREFIX:
+        return dev_mem_eq(v, pat, plen);
+    case DSM_POSTFIX:
+        return dev_mem_eq(v + vlen - plen, pat, plen);
+    default:
+        return 0;
+    }
+}
+
+/* validate_filter (pegasus_server_impl.cpp:2350-2380): NO_FILTER / empty
+ * pattern -> pass */
+__device__ static int dev_validate_filter(int ft, const uint8_t *pat, uint64_t plen,
+                                          const uint8_t *v, uint64_t vlen)
+{
+    if (ft == 0 || plen == 0)
+        return 1;
+    if (vlen < plen)
+        return 0;
+    if (ft == 1) { /* anywhere */
+        for (uint64_t i = 0; i + plen <= vlen; i++)
+            if (dev_mem_eq(v + i, pat, plen))
+                return 1;
+        return 0;
+    }
+    if (ft == 2)
+        return dev_mem_eq(v, pat, plen);
+    return dev_mem_eq(v + vlen - plen, pat, plen);
+}
+
+__device__ static int dev_rule_match(const DevRule &r, const uint8_t *pats, uint32_t dv,
+                                     uint32_t now, const uint8_t *hk, uint64_t hklen,
+                                     const uint8_t *sk, uint64_t sklen, const uint8_t *val)
+{
+    switch (r.type) {
+    case DFR_HASHKEY:
+        return dev_pattern_match(hk, hklen, r.match_type, pats + r.pat_off, r.pat_len);
+    case DFR_SORTKEY:
+        return dev_pattern_match(sk, sklen, r.match_type, pats + r.pat_off, r.pat_len);
+    case DFR_TTL_RANGE: {
+        uint32_t e = dev_expire_ts(dv, val);
+        if (e == 0 && r.start_ttl == 0 && r.stop_ttl == 0)
+            return 1;
+        return (uint32_t)(r.start_ttl + now) <= e && (uint32_t)(r.stop_ttl + now) >= e;
+    }
+    default:
+        return 0;
+    }
+}
+
+/* ================= bounds ================= */
+__global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t klen,
+                         uint64_t *out /* [R] */, int upper)
+{
+    int r = blockIdx.x * blockDim.x + threadIdx.x;
+    if (r >= R)
+        return;
+    if (key == nullptr) {
+        out[r] = upper ? runs[r].n : 0;
+        return;
+    }
+    out[r] = upper ? dev_upper_bound(runs[r], key, klen, 0, runs[r].n)
+                   : dev_lower_bound(runs[r], key, klen, 0, runs[r].n);
+}
+
+/* ================= rank merge =================
+ * For windowed record t -> (r,i):
+ * rank = (i - lo[r]) + sum_{r'>r} upper_bound(r',key) + sum_{r'<r} lower_bound(r',key)
+ * (run index order == age order: higher r == newer == first among equal keys)
+ * Scatter order[rank] = (r<<40)|i.  wprefix[r] = running sum of window sizes. */
+__global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
+                       const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order)
+{
+    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
+         t += gridDim.x * (uint64_t)blockDim.x) {
+        int r = 0;
+        while (wprefix[r + 1] <= t)
+            r++;
+        uint64_t i = lo[r] + (t - wprefix[r]);
+        uint64_t kl;
+        const uint8_t *k = run_key(runs[r], i, &kl);
+        uint64_t rank = i - lo[r];
+        for (int q = 0; q < R; q++) {
+            if (q == r)
+                continue;
+            uint64_t pos = (q > r) ? dev_upper_bound(runs[q], k, kl, lo[q], hi[q])
+                                   : dev_lower_bound(runs[q], k, kl, lo[q], hi[q]);
+            rank += pos - lo[q];
+        }
+        order[rank] = ((uint64_t)r << 40) | i;
+    }
+}
+
+/* visible flag: newest version of its key (first of the equal-key group in
+ * (key asc, run desc) order) and not a tombstone */
+__global__ void k_visible(const DevRun *runs, const uint64_t *order, uint64_t m, uint64_t *flags)
+{
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < m;
+         p += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t id = order[p];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl;
+        const uint8_t *k = run_key(r, i, &kl);
+        int newest = 1;
+        if (p > 0) {
+            uint64_t pid = order[p - 1];
+            const DevRun &pr = runs[pid >> 40];
+            uint64_t pi = pid & 0xFFFFFFFFFFull, pkl;
+            const uint8_t *pk = run_key(pr, pi, &pkl);
+            if (dev_key_cmp(k, kl, pk, pkl) == 0)
+                newest = 0;
+        }
+        int tomb = (int)(r.sk[i] & 1);
+        flags[p] = (newest && !tomb) ? 1 : 0;
+    }
+}
+
+/* gather order entries whose flag is set, by exclusive prefix positions */
+__global__ void k_gather(const uint64_t *order, const uint64_t *flags, const uint64_t *pos,
+                         uint64_t m, uint64_t *out)
+{
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < m;
+         p += gridDim.x * (uint64_t)blockDim.x) {
+        if (flags[p])
+            out[pos[p]] = order[p];
+    }
+}
+
+/* ================= prefix sum (u64, exclusive) ================= */
+__global__ void k_psum1(const uint64_t *in, uint64_t *out, uint64_t *blocksums, uint64_t n)
+{
+    __shared__ uint64_t lds[BLOCK];
+    uint64_t base = (uint64_t)blockIdx.x * PSUM_BLOCK_ITEMS;
+    uint64_t tbase = base + (uint64_t)threadIdx.x * PSUM_ITEMS_PER_THREAD;
+    uint64_t loc[PSUM_ITEMS_PER_THREAD];
+    uint64_t s = 0;
+    for (int j = 0; j < PSUM_ITEMS_PER_THREAD; j++) {
+        uint64_t idx = tbase + j;
+        loc[j] = s;
+        s += (idx < n) ? in[idx] : 0;
+    }
+    lds[threadIdx.x] = s;
+    __syncthreads();
+    /* Hillis-Steele inclusive scan over 256 thread sums */
+    for (int off = 1; off < BLOCK; off <<= 1) {
+        uint64_t v = (threadIdx.x >= (unsigned)off) ? lds[threadIdx.x - off] : 0;
+        __syncthreads();
+        lds[threadIdx.x] += v;
+        __syncthreads();
+    }
+    uint64_t texcl = (threadIdx.x == 0) ? 0 : lds[threadIdx.x - 1];
+    for (int j = 0; j < PSUM_ITEMS_PER_THREAD; j++) {
+        uint64_t idx = tbase + j;
+        if (idx < n)
+            out[idx] = texcl + loc[j];
+    }
+    if (threadIdx.x == BLOCK - 1 && blocksums)
+        blocksums[blockIdx.x] = lds[BLOCK - 1];
+}
+
+__global__ void k_psum_add(uint64_t *out, const uint64_t *blockoffs, uint64_t n)
+{
+    uint64_t b = blockIdx.x;
+    uint64_t base = b * PSUM_BLOCK_ITEMS;
+    uint64_t add = blockoffs[b];
+    for (uint64_t i = base + threadIdx.x; i < base + PSUM_BLOCK_ITEMS && i < n; i += blockDim.x)
+        out[i] += add;
+}
+
+/* ================= point get =================
+ * on_get / DB::Get equivalent (pegasus_server_impl.cpp:441): search newest
+ * run first; first hit wins (run seqno ranges are ordered). */
+__global__ void k_get(const DevRun *runs, int R, const uint8_t *qkeys, const uint64_t *qoffs,
+                      uint64_t nq, uint32_t epoch_now, uint32_t dv, int32_t *status,
+                      uint64_t *hit, uint64_t *ulen /* user-data len */, uint32_t *expire_out)
+{
+    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < nq;
+         t += gridDim.x * (uint64_t)blockDim.x) {
+        const uint8_t *k = qkeys + qoffs[t];
+        uint64_t kl = qoffs[t + 1] - qoffs[t];
+        int32_t st = 1; /* NotFound */
+        uint64_t h = 0, ul = 0;
+        uint32_t eo = 0;
+        for (int r = R - 1; r >= 0; r--) {
+            uint64_t i = dev_lower_bound(runs[r], k, kl, 0, runs[r].n);
+            if (i >= runs[r].n)
+                continue;
+            uint64_t ml;
+            const uint8_t *mk = run_key(runs[r], i, &ml);
+            if (dev_key_cmp(mk, ml, k, kl) != 0)
+                continue;
+            /* newest version */
+            if (runs[r].sk[i] & 1)
+                break; /* tombstone -> NotFound */
+            uint64_t vl;
+            const uint8_t *v = run_val(runs[r], i, &vl);
+            uint32_t e = dev_expire_ts(dv, v);
+            if (dev_ts_expired(epoch_now, e))
+                break; /* expired -> NotFound (on_get:444) */
+            st = 0;
+            h = ((uint64_t)r << 40) | i;
+            ul = vl - dev_hdr_len(dv);
+            eo = e;
+            break;
+        }
+        status[t] = st;
+        hit[t] = h;
+        ulen[t] = ul;
+        if (expire_out)
+            expire_out[t] = eo;
+    }
+}
+
+/* emit user data for selected hits (wave per record, lane-strided bytes) */
+__global__ void k_emit_values(const DevRun *runs, const uint64_t *hit, const int32_t *status,
+                              uint64_t nq, uint32_t dv, const uint64_t *voffs, uint8_t *vout)
+{
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    uint32_t hdr = dev_hdr_len(dv);
+    for (uint64_t t = wave; t < nq; t += nwaves) {
+        if (status[t] != 0)
+            continue;
+        uint64_t id = hit[t];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull, vl;
+        const uint8_t *v = run_val(r, i, &vl);
+        const uint8_t *src = v + hdr;
+        uint64_t n = vl - hdr;
+        uint8_t *dst = vout + voffs[t];
+        for (uint64_t b = lane; b < n; b += WAVE)
+            dst[b] = src[b];
+    }
+}
+
+/* ================= scan state =================
+ * validate_key_value_for_scan (pegasus_server_impl.cpp:2382-2432) applied to
+ * a window of the visible view; also writes out key/value sizes. */
+__global__ void k_scan_state(const DevRun *runs, const uint64_t *view, uint64_t w, ScanParams sp,
+                             uint8_t *state, uint64_t *ksz, uint64_t *vsz)
+{
+    uint32_t hdr = dev_hdr_len(sp.data_version);
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < w;
+         p += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t id = view[p];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        const uint8_t *v = run_val(r, i, &vl);
+        int st = ST_NORMAL;
+        if (dev_ts_expired(sp.epoch_now, dev_expire_ts(sp.data_version, v))) {
+            st = ST_EXPIRED;
+        } else if (sp.validate_hash &&
+                   (sp.partition_version < 0 || sp.pidx > sp.partition_version ||
+                    (int64_t)(dev_key_hash(k, kl) & (uint64_t)sp.partition_version) !=
+                        (int64_t)sp.pidx)) {
+            st = ST_HASH_INVALID;
+        } else if (sp.hk_ft != 0 || sp.sk_ft != 0) {
+            uint32_t hklen = ((uint32_t)k[0] << 8) | k[1];
+            const uint8_t *hk = k + 2;
+            const uint8_t *skp = k + 2 + hklen;
+            uint64_t sklen = kl - 2 - hklen;
+            if (sp.hk_ft != 0 &&
+                !dev_validate_filter(sp.hk_ft, sp.hk_pat, sp.hk_pat_len, hk, hklen))
+                st = ST_FILTERED;
+            else if (sp.sk_ft != 0 &&
+                     !dev_validate_filter(sp.sk_ft, sp.sk_pat, sp.sk_pat_len, skp, sklen))
+                st = ST_FILTERED;
+        }
+        state[p] = (uint8_t)st;
+        ksz[p] = (st == ST_NORMAL) ? (kl - sp.hash_key_skip) : 0;
+        vsz[p] = (st == ST_NORMAL && !sp.no_value) ? (vl - hdr) : 0;
+    }
+}
+
+/* normal-flag (1/0) from state, zeroed at/after the cutoff */
+__global__ void k_normal_flags(const uint8_t *state, uint64_t w, uint64_t *flags)
+{
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < w;
+         p += gridDim.x * (uint64_t)blockDim.x)
+        flags[p] = (state[p] == ST_NORMAL) ? 1 : 0;
+}
+
+/* widen u8 0/1 flags to u64 (for the prefix-sum) */
+__global__ void k_widen_u8(const uint8_t *in, uint64_t n, uint64_t *out)
+{
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < n;
+         p += gridDim.x * (uint64_t)blockDim.x)
+        out[p] = in[p];
+}
+
+/* zero sizes for entries at/after consumed cutoff or non-normal */
+__global__ void k_cut_sizes(const uint8_t *state, uint64_t w, uint64_t consumed, uint64_t *ksz,
+                            uint64_t *vsz)
+{
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < w;
+         p += gridDim.x * (uint64_t)blockDim.x) {
+        if (p >= consumed || state[p] != ST_NORMAL) {
+            ksz[p] = 0;
+            vsz[p] = 0;
+        }
+    }
+}
+
+/* first p with nprefix[p] >= batch_count (inclusive-scan semantics handled by
+ * caller passing exclusive prefix + flags) — single thread, tiny */
+__global__ void k_cutoff(const uint64_t *nprefix_excl, const uint64_t *flags, uint64_t w,
+                         uint64_t batch_count, uint64_t *out)
+{
+    if (blockIdx.x != 0 || threadIdx.x != 0)
+        return;
+    /* consumed = index just after the batch_count-th normal record, or w */
+    uint64_t lo = 0, hi = w;
+    /* nprefix_excl[p] + flags[p] = inclusive count at p */
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1;
+        uint64_t incl = nprefix_excl[mid] + flags[mid];
+        if (incl < batch_count)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    out[0] = (lo < w) ? lo + 1 : w;
+    /* also export the number of normals consumed */
+    out[1] = (lo < w) ? batch_count : (w ? nprefix_excl[w - 1] + flags[w - 1] : 0);
+}
+
+/* emit scan kvs: for each consumed normal record, copy (key minus
+ * hash_key_skip) and user-data value into packed buffers, and write offsets +
+ * optional expire_ts.  Wave per record. */
+__global__ void k_emit_scan(const DevRun *runs, const uint64_t *view, uint64_t w,
+                            const uint8_t *state, uint64_t consumed, const uint64_t *npos,
+                            const uint64_t *koffs, const uint64_t *voffs, ScanParams sp,
+                            uint8_t *kout, uint8_t *vout, uint64_t *kout_offs,
+                            uint64_t *vout_offs, int32_t *ets_out, uint64_t n_out)
+{
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    uint32_t hdr = dev_hdr_len(sp.data_version);
+    for (uint64_t p = wave; p < consumed; p += nwaves) {
+        if (state[p] != ST_NORMAL)
+            continue;
+        uint64_t o = npos[p]; /* output row index */
+        uint64_t id = view[p];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        const uint8_t *v = run_val(r, i, &vl);
+        const uint8_t *ksrc = k + sp.hash_key_skip;
+        uint64_t kn = kl - sp.hash_key_skip;
+        uint8_t *kdst = kout + koffs[p];
+        for (uint64_t b = lane; b < kn; b += WAVE)
+            kdst[b] = ksrc[b];
+        if (!sp.no_value) {
+            const uint8_t *vsrc = v + hdr;
+            uint64_t vn = vl - hdr;
+            uint8_t *vdst = vout + voffs[p];
+            for (uint64_t b = lane; b < vn; b += WAVE)
+                vdst[b] = vsrc[b];
+        }
+        if (lane == 0) {
+            kout_offs[o] = koffs[p];
+            vout_offs[o] = voffs[p];
+            if (ets_out)
+                ets_out[o] = (int32_t)dev_expire_ts(sp.data_version, v);
+            if (o == n_out - 1) {
+                kout_offs[n_out] = koffs[p] + kn;
+                vout_offs[n_out] = voffs[p] + ((!sp.no_value) ? (vl - hdr) : 0);
+            }
+        }
+    }
+}
+
+/* emit an explicit host-selected list of view rows (multi_get caps are
+ * applied host-side over a <=3000-entry window, mirroring the reference's
+ * limiter loop exactly).  rows[j] = view position; koffs/voffs per j. */
+__global__ void k_emit_rows(const DevRun *runs, const uint64_t *view, const uint64_t *rows,
+                            uint64_t n_rows, const uint64_t *koffs, const uint64_t *voffs,
+                            ScanParams sp, uint8_t *kout, uint8_t *vout)
+{
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    uint32_t hdr = dev_hdr_len(sp.data_version);
+    for (uint64_t j = wave; j < n_rows; j += nwaves) {
+        uint64_t id = view[rows[j]];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        const uint8_t *v = run_val(r, i, &vl);
+        const uint8_t *ksrc = k + sp.hash_key_skip;
+        uint64_t kn = kl - sp.hash_key_skip;
+        uint8_t *kdst = kout + koffs[j];
+        for (uint64_t b = lane; b < kn; b += WAVE)
+            kdst[b] = ksrc[b];
+        if (!sp.no_value) {
+            const uint8_t *vsrc = v + hdr;
+            uint64_t vn = vl - hdr;
+            uint8_t *vdst = vout + voffs[j];
+            for (uint64_t b = lane; b < vn; b += WAVE)
+                vdst[b] = vsrc[b];
+        }
+    }
+}
+
+/* compare the key of view[0] with a given key; out[0]=1 if equal */
+__global__ void k_first_eq(const DevRun *runs, const uint64_t *view, uint64_t n,
+                           const uint8_t *key, uint64_t klen, uint32_t *out)
+{
+    if (blockIdx.x != 0 || threadIdx.x != 0)
+        return;
+    out[0] = 0;
+    if (n == 0)
+        return;
+    uint64_t id = view[0];
+    const DevRun &r = runs[id >> 40];
+    uint64_t i = id & 0xFFFFFFFFFFull, kl;
+    const uint8_t *k = run_key(r, i, &kl);
+    out[0] = dev_key_cmp(k, kl, key, klen) == 0;
+}
+
+/* ================= compaction =================
+ * Disposition of every record in the full-range order[] array:
+ * KeyWithTTLCompactionFilter::Filter (key_ttl_compaction_filter.h:55-92) on
+ * the surviving newest PUT; tombstones and shadowed versions dropped
+ * (bottommost CompactRange, do_manual_compact:3389). */
+__global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint64_t m,
+                                CompactParams cp, uint8_t *keep, uint8_t *changed,
+                                uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
+                                CompactStatsDev *stats)
+{
+    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < m;
+         p += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t id = order[p];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        const uint8_t *v = run_val(r, i, &vl);
+        keep[p] = 0;
+        changed[p] = 0;
+        ksz[p] = 0;
+        vsz[p] = 0;
+        /* newest-of-key? */
+        if (p > 0) {
+            uint64_t pid = order[p - 1];
+            const DevRun &pr = runs[pid >> 40];
+            uint64_t pi = pid & 0xFFFFFFFFFFull, pkl;
+            const uint8_t *pk = run_key(pr, pi, &pkl);
+            if (dev_key_cmp(k, kl, pk, pkl) == 0) {
+                atomicAdd(&stats->shadowed, 1ull);
+                continue;
+            }
+        }
+        if (r.sk[i] & 1) {
+            atomicAdd(&stats->tombstones, 1ull);
+            continue;
+        }
+        /* --- Filter --- */
+        int drop = 0, value_changed = 0;
+        uint32_t new_ts_val = 0;
+        uint32_t expire_ts = dev_expire_ts(cp.data_version, v);
+        uint32_t eff_expire = expire_ts; /* value_view's expire after default-ttl */
+        if (kl >= 2) {
+            if (cp.default_ttl != 0 && expire_ts == 0) {
+                expire_ts = cp.epoch_now + cp.default_ttl;
+                eff_expire = expire_ts;
+                value_changed = 1;
+                new_ts_val = expire_ts;
+            }
+            if (cp.n_ops > 0) {
+                uint32_t hklen = ((uint32_t)k[0] << 8) | k[1];
+                const uint8_t *hk = k + 2;
+                const uint8_t *skp = k + 2 + hklen;
+                uint64_t sklen = kl - 2 - hklen;
+                /* value_view is fixed at loop entry: rules see eff_expire
+                 * (post default-ttl), not later op rewrites
+                 * (key_ttl_compaction_filter.h:82-89) */
+                for (int oi = 0; oi < cp.n_ops && !drop; oi++) {
+                    const DevOp &op = cp.ops[oi];
+                    int all = (op.n_rules > 0);
+                    for (int ri = 0; ri < op.n_rules && all; ri++) {
+                        const DevRule &rule = cp.rules[op.rule_off + ri];
+                        if (rule.type == DFR_TTL_RANGE) {
+                            uint32_t e = eff_expire;
+                            int ok = (e == 0 && rule.start_ttl == 0 && rule.stop_ttl == 0) ||
+                                     ((uint32_t)(rule.start_ttl + cp.epoch_now) <= e &&
+                                      (uint32_t)(rule.stop_ttl + cp.epoch_now) >= e);
+                            all = ok;
+                        } else {
+                            all = dev_rule_match(rule, cp.pats, cp.data_version, cp.epoch_now, hk,
+                                                 hklen, skp, sklen, v);
+                        }
+                    }
+                    if (!all)
+                        continue;
+                    if (op.type == DOP_DELETE) {
+                        drop = 1;
+                        break;
+                    }
+                    /* update_ttl (compaction_operation.cpp:78-113); current
+                     * ttl read from the value_view (eff_expire) */
+                    uint32_t nts = 0;
+                    int apply = 1;
+                    switch (op.ut_type) {
+                    case DUT_FROM_NOW:
+                        nts = cp.epoch_now + op.ut_value;
+                        break;
+                    case DUT_FROM_CURRENT:
+                        if (eff_expire == 0)
+                            apply = 0;
+                        else
+                            nts = op.ut_value + eff_expire;
+                        break;
+                    case DUT_TIMESTAMP:
+                        nts = op.ut_value - 1451606400u;
+                        break;
+                    default:
+                        apply = 0;
+                        break;
+                    }
+                    if (apply) {
+                        value_changed = 1;
+                        new_ts_val = nts;
+                    }
+                }
+            }
+            if (drop) {
+                atomicAdd(&stats->filtered, 1ull);
+                continue;
+            }
+            /* final keep/drop on local expire_ts (post default-ttl, pre
+             * user-op) + stale split hash (:91,114-121) */
+            if (dev_ts_expired(cp.epoch_now, expire_ts)) {
+                atomicAdd(&stats->expired, 1ull);
+                continue;
+            }
+            if (cp.validate_hash && cp.partition_version >= 0 &&
+                cp.pidx <= cp.partition_version &&
+                (int64_t)(dev_key_hash(k, kl) & (uint64_t)cp.partition_version) !=
+                    (int64_t)cp.pidx) {
+                atomicAdd(&stats->filtered, 1ull);
+                continue;
+            }
+        }
+        keep[p] = 1;
+        changed[p] = (uint8_t)value_changed;
+        new_expire[p] = new_ts_val;
+        ksz[p] = kl;
+        vsz[p] = vl;
+        atomicAdd(&stats->output_records, 1ull);
+    }
+}
+
+/* emit the merged run (wave per record); values copied whole, expire header
+ * patched in-place for changed records */
+__global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64_t m,
+                               const uint8_t *keep, const uint8_t *changed,
+                               const uint32_t *new_expire, const uint64_t *kpos,
+                               const uint64_t *koffs, const uint64_t *voffs, uint32_t dv,
+                               uint8_t *kout, uint8_t *vout, uint64_t *okoff, uint64_t *ovoff,
+                               uint64_t *osk, uint64_t n_out)
+{
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    for (uint64_t p = wave; p < m; p += nwaves) {
+        if (!keep[p])
+            continue;
+        uint64_t o = kpos[p];
+        uint64_t id = order[p];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        const uint8_t *v = run_val(r, i, &vl);
+        uint8_t *kdst = kout + koffs[p];
+        for (uint64_t b = lane; b < kl; b += WAVE)
+            kdst[b] = k[b];
+        uint8_t *vdst = vout + voffs[p];
+        for (uint64_t b = lane; b < vl; b += WAVE)
+            vdst[b] = v[b];
+        if (lane == 0) {
+            if (changed[p]) {
+                uint32_t ts = new_expire[p];
+                uint32_t off = (dv == 2) ? 1 : 0;
+                vdst[off] = (uint8_t)(ts >> 24);
+                vdst[off + 1] = (uint8_t)(ts >> 16);
+                vdst[off + 2] = (uint8_t)(ts >> 8);
+                vdst[off + 3] = (uint8_t)ts;
+            }
+            okoff[o] = koffs[p];
+            ovoff[o] = voffs[p];
+            osk[o] = r.sk[i];
+            if (o == n_out - 1) {
+                okoff[n_out] = koffs[p] + kl;
+                ovoff[n_out] = voffs[p] + vl;
+            }
+        }
+    }
+}
+
+/* ================= launch wrappers (host) ================= */
+extern "C++" {
+
+void launch_crc64_table_init(const uint64_t *host_table)
+{
+    HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(d_crc64_table), host_table, 256 * sizeof(uint64_t)));
+}
+
+void launch_bounds(const DevRun *d_runs, int R, const uint8_t *d_key, uint64_t klen,
+                   uint64_t *d_out, int upper, hipStream_t s)
+{
+    k_bounds<<<(R + 63) / 64, 64, 0, s>>>(d_runs, R, d_key, klen, d_out, upper);
+}
+
+void launch_rank(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
+                 const uint64_t *d_wprefix, uint64_t total, uint64_t *d_order, hipStream_t s)
+{
+    k_rank<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix, total,
+                                                    d_order);
+}
+
+void launch_visible(const DevRun *d_runs, const uint64_t *d_order, uint64_t m, uint64_t *d_flags,
+                    hipStream_t s)
+{
+    k_visible<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, m, d_flags);
+}
+
+void launch_gather(const uint64_t *d_order, const uint64_t *d_flags, const uint64_t *d_pos,
+                   uint64_t m, uint64_t *d_out, hipStream_t s)
+{
+    k_gather<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_order, d_flags, d_pos, m, d_out);
+}
+
+/* exclusive scan; returns total via d_total (device u64, may be null) */
+void launch_psum(const uint64_t *d_in, uint64_t *d_out, uint64_t n, hipStream_t s);
+
+static void psum_rec(const uint64_t *d_in, uint64_t *d_out, uint64_t n, hipStream_t s)
+{
+    uint64_t nb = (n + PSUM_BLOCK_ITEMS - 1) / PSUM_BLOCK_ITEMS;
+    if (nb == 0)
+        nb = 1;
+    uint64_t *d_bs = nullptr;
+    HIP_CHECK(hipMallocAsync(&d_bs, (nb + 1) * sizeof(uint64_t), s));
+    k_psum1<<<(uint32_t)nb, BLOCK, 0, s>>>(d_in, d_out, d_bs, n);
+    if (nb > 1) {
+        uint64_t *d_bo = nullptr;
+        HIP_CHECK(hipMallocAsync(&d_bo, (nb + 1) * sizeof(uint64_t), s));
+        psum_rec(d_bs, d_bo, nb, s);
+        k_psum_add<<<(uint32_t)nb, BLOCK, 0, s>>>(d_out, d_bo, n);
+        HIP_CHECK(hipFreeAsync(d_bo, s));
+    }
+    HIP_CHECK(hipFreeAsync(d_bs, s));
+}
+
+void launch_psum(const uint64_t *d_in, uint64_t *d_out, uint64_t n, hipStream_t s)
+{
+    psum_rec(d_in, d_out, n, s);
+}
+
+void launch_get(const DevRun *d_runs, int R, const uint8_t *d_qkeys, const uint64_t *d_qoffs,
+                uint64_t nq, uint32_t epoch_now, uint32_t dv, int32_t *d_status, uint64_t *d_hit,
+                uint64_t *d_ulen, uint32_t *d_expire, hipStream_t s)
+{
+    k_get<<<grid_for(nq, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_qkeys, d_qoffs, nq, epoch_now, dv,
+                                                d_status, d_hit, d_ulen, d_expire);
+}
+
+void launch_emit_values(const DevRun *d_runs, const uint64_t *d_hit, const int32_t *d_status,
+                        uint64_t nq, uint32_t dv, const uint64_t *d_voffs, uint8_t *d_vout,
+                        hipStream_t s)
+{
+    k_emit_values<<<grid_for(nq * WAVE, BLOCK), BLOCK, 0, s>>>(d_runs, d_hit, d_status, nq, dv,
+                                                               d_voffs, d_vout);
+}
+
+void launch_scan_state(const DevRun *d_runs, const uint64_t *d_view, uint64_t w,
+                       const ScanParams &sp, uint8_t *d_state, uint64_t *d_ksz, uint64_t *d_vsz,
+                       hipStream_t s)
+{
+    k_scan_state<<<grid_for(w, BLOCK), BLOCK, 0, s>>>(d_runs, d_view, w, sp, d_state, d_ksz,
+                                                      d_vsz);
+}
+
+void launch_normal_flags(const uint8_t *d_state, uint64_t w, uint64_t *d_flags, hipStream_t s)
+{
+    k_normal_flags<<<grid_for(w, BLOCK), BLOCK, 0, s>>>(d_state, w, d_flags);
+}
+
+void launch_cutoff(const uint64_t *d_nprefix, const uint64_t *d_flags, uint64_t w,
+                   uint64_t batch_count, uint64_t *d_out2, hipStream_t s)
+{
+    k_cutoff<<<1, 1, 0, s>>>(d_nprefix, d_flags, w, batch_count, d_out2);
+}
+
+void launch_widen_u8(const uint8_t *d_in, uint64_t n, uint64_t *d_out, hipStream_t s)
+{
+    k_widen_u8<<<grid_for(n, BLOCK), BLOCK, 0, s>>>(d_in, n, d_out);
+}
+
+void launch_cut_sizes(const uint8_t *d_state, uint64_t w, uint64_t consumed, uint64_t *d_ksz,
+                      uint64_t *d_vsz, hipStream_t s)
+{
+    k_cut_sizes<<<grid_for(w, BLOCK), BLOCK, 0, s>>>(d_state, w, consumed, d_ksz, d_vsz);
+}
+
+void launch_emit_scan(const DevRun *d_runs, const uint64_t *d_view, uint64_t w,
+                      const uint8_t *d_state, uint64_t consumed, const uint64_t *d_npos,
+                      const uint64_t *d_koffs, const uint64_t *d_voffs, const ScanParams &sp,
+                      uint8_t *d_kout, uint8_t *d_vout, uint64_t *d_kout_offs,
+                      uint64_t *d_vout_offs, int32_t *d_ets, uint64_t n_out, hipStream_t s)
+{
+    k_emit_scan<<<grid_for(consumed * WAVE, BLOCK), BLOCK, 0, s>>>(
+        d_runs, d_view, w, d_state, consumed, d_npos, d_koffs, d_voffs, sp, d_kout, d_vout,
+        d_kout_offs, d_vout_offs, d_ets, n_out);
+}
+
+void launch_emit_rows(const DevRun *d_runs, const uint64_t *d_view, const uint64_t *d_rows,
+                      uint64_t n_rows, const uint64_t *d_koffs, const uint64_t *d_voffs,
+                      const ScanParams &sp, uint8_t *d_kout, uint8_t *d_vout, hipStream_t s)
+{
+    k_emit_rows<<<grid_for(n_rows * WAVE, BLOCK), BLOCK, 0, s>>>(d_runs, d_view, d_rows, n_rows,
+                                                                 d_koffs, d_voffs, sp, d_kout,
+                                                                 d_vout);
+}
+
+void launch_first_eq(const DevRun *d_runs, const uint64_t *d_view, uint64_t n,
+                     const uint8_t *d_key, uint64_t klen, uint32_t *d_out, hipStream_t s)
+{
+    k_first_eq<<<1, 1, 0, s>>>(d_runs, d_view, n, d_key, klen, d_out);
+}
+
+void launch_compact_flags(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
+                          const CompactParams &cp, uint8_t *d_keep, uint8_t *d_changed,
+                          uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
+                          CompactStatsDev *d_stats, hipStream_t s)
+{
+    k_compact_flags<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, m, cp, d_keep,
+                                                         d_changed, d_new_expire, d_ksz, d_vsz,
+                                                         d_stats);
+}
+
+void launch_emit_compact(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
+                         const uint8_t *d_keep, const uint8_t *d_changed,
+                         const uint32_t *d_new_expire, const uint64_t *d_kpos,
+                         const uint64_t *d_koffs, const uint64_t *d_voffs, uint32_t dv,
+                         uint8_t *d_kout, uint8_t *d_vout, uint64_t *d_okoff, uint64_t *d_ovoff,
+                         uint64_t *d_osk, uint64_t n_out, hipStream_t s)
+{
+    k_emit_compact<<<grid_for(m * WAVE, BLOCK), BLOCK, 0, s>>>(
+        d_runs, d_order, m, d_keep, d_changed, d_new_expire, d_kpos, d_koffs, d_voffs, dv, d_kout,
+        d_vout, d_okoff, d_ovoff, d_osk, n_out);
+}
+
+} /* extern C++ */

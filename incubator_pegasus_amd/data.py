@@ -142,13 +142,11 @@ def make_values(ids: np.ndarray, value_len: int = 100, expire_ts: np.ndarray = N
         t = timetag.astype(np.uint64)
         for b in range(8):
             out[:, off + 4 + b] = (t >> np.uint64(8 * (7 - b))).astype(np.uint8)
-    # value body: 8 bytes per splitmix word
+    # value body: 8 bytes per splitmix word (little-endian u64 -> u8 view)
     words = (value_len + 7) // 8
-    body = np.empty((n, words * 8), dtype=np.uint8)
-    for w in range(words):
-        v = splitmix64(ids * np.uint64(words) + np.uint64(w) + np.uint64(salt))
-        for b in range(8):
-            body[:, w * 8 + b] = (v >> np.uint64(8 * b)).astype(np.uint8)
+    idx = (ids.astype(np.uint64)[:, None] * np.uint64(words) +
+           np.arange(words, dtype=np.uint64)[None, :] + np.uint64(salt))
+    body = np.ascontiguousarray(splitmix64(idx.reshape(-1))).view(np.uint8).reshape(n, words * 8)
     out[:, hdr:] = body[:, :value_len]
     return out
 
@@ -198,26 +196,15 @@ def build_point_table_runs(n_keys: int, n_runs: int, *, seed: int = DEFAULT_SEED
         oids = ids[overlay]
         n = len(oids)
         raw = make_raw_keys(oids)
+        # fixed-size encoded values for every overlay record, tombstones
+        # included — a tombstone's value bytes are never read (engine and
+        # reference both branch on kind before touching the value)
         vals = make_values(oids, value_len, version=version, salt=1)
         kinds = np.where(overlay_del[overlay], np.uint64(1), np.uint64(0))
-        # tombstones carry empty values
-        voff = fixed_offsets(n, vals.shape[1]).copy()
-        # keep fixed-size values also for tombstones (engine ignores them);
-        # reference tombstones have no value — model with zero-length value
-        lens = np.where(kinds == 1, 0, vals.shape[1]).astype(np.uint64)
-        voff = np.zeros(n + 1, dtype=np.uint64)
-        np.cumsum(lens, out=voff[1:])
-        packed_vals = np.zeros(int(voff[-1]), dtype=np.uint64)  # placeholder
-        packed = np.zeros(int(voff[-1]), dtype=np.uint8)
-        flat = vals.reshape(-1)
-        src_off = fixed_offsets(n, vals.shape[1])
-        for i in np.nonzero(kinds == 0)[0]:
-            packed[int(voff[i]):int(voff[i + 1])] = flat[int(src_off[i]):int(src_off[i]) + vals.shape[1]]
-        del packed_vals
         sk = (((np.arange(n, dtype=np.uint64)) + np.uint64(seq_base)) << np.uint64(1)) | kinds
         seq_base += n
         runs.append(dict(keys=raw.reshape(-1), koff=fixed_offsets(n, raw.shape[1]),
-                         vals=packed, voff=voff, sk=sk))
+                         vals=vals.reshape(-1), voff=fixed_offsets(n, vals.shape[1]), sk=sk))
     return runs
 
 

@@ -1932,6 +1932,18 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     st.output_records = on;
     st.output_bytes = okb + ovb;
 
+    if (opts && opts->keep_inputs) {
+        /* benchmarking mode: full pass done, inputs untouched */
+        free(okeys);
+        free(okoff);
+        free(ovals);
+        free(ovoff);
+        free(osk);
+        if (stats)
+            *stats = st;
+        return RRDB_OK;
+    }
+
     /* swap in the merged run */
     for (int i = 0; i < e->n_runs; i++)
         free_run(&e->runs[i]);
@@ -1988,4 +2000,12 @@ int orc_pattern_match(const uint8_t *v, uint64_t vlen, int type, const uint8_t *
                       uint64_t plen)
 {
     return string_pattern_match(v, vlen, type, (const char *)pat, plen);
+}
+
+/* phase timings: CPU oracle does not record GPU phases */
+double rrdb_phase_ms(void *h, const char *phase)
+{
+    (void)h;
+    (void)phase;
+    return -1.0;
 }
