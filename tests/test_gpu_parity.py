@@ -275,3 +275,49 @@ def test_device_resident_scan_output(pair):
     assert res.error == OK and res.dev is not None
     assert res.dev["count"] == 100
     assert res.dev["dev_keys"] and res.dev["dev_vals"]
+
+
+@pytest.mark.parametrize("version", [0, 1, 2])
+def test_value_schema_versions_parity(oracle_lib, hip_lib, version):
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        now = 1000
+        for p in (o, g):
+            p.set_envs({"pegasus.data_version": str(version)})
+        recs = [(D.generate_key(b"vk", f"s{i}".encode()),
+                 D.encode_value(f"val{i}".encode(), now + 50 if i == 0 else 0, 7, version),
+                 i + 1, 0) for i in range(5)]
+        o.ingest_run(recs)
+        g.ingest_run(recs)
+        for k, _, _, _ in recs:
+            assert o.get(k, now) == g.get(k, now)
+            assert o.ttl(k, now) == g.ttl(k, now)
+        ro = o.scan_open(b"\x00\x00", b"\xff\xff", now, validate_partition_hash=False,
+                         return_expire_ts=True)
+        rg = g.scan_open(b"\x00\x00", b"\xff\xff", now, validate_partition_hash=False,
+                         return_expire_ts=True)
+        assert (ro.kvs, ro.expire_ts) == (rg.kvs, rg.expire_ts)
+        assert o.manual_compact(now)[1] == g.manual_compact(now)[1]
+    finally:
+        o.close()
+        g.close()
+
+
+def test_env_caps_parity(oracle_lib, hip_lib):
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        now = 100
+        recs = [(D.generate_key(b"caps", f"s{i:04d}".encode()),
+                 D.encode_value(b"v", 0, i + 1, 1), i + 1, 0) for i in range(60)]
+        for p in (o, g):
+            p.ingest_run(recs)
+            p.set_envs({"rocksdb.max_iteration_count": "13",
+                        "rocksdb.multi_get_max_iteration_count": "9"})
+        assert o.multi_get(b"caps", now) == g.multi_get(b"caps", now)
+        assert _drain(o, now, batch_size=1000, validate_partition_hash=False) == \
+               _drain(g, now, batch_size=1000, validate_partition_hash=False)
+    finally:
+        o.close()
+        g.close()
