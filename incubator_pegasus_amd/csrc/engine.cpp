@@ -428,7 +428,7 @@ int key_cmp(const std::string &a, const std::string &b)
     return a.size() < b.size() ? -1 : (a.size() > b.size() ? 1 : 0);
 }
 
-uint32_t hdr_len(uint32_t ver) { return ver == 0 ? 4u : (ver == 1 ? 12u : 13u); }
+[[maybe_unused]] uint32_t hdr_len(uint32_t ver) { return ver == 0 ? 4u : (ver == 1 ? 12u : 13u); }
 
 /* ================ engine ================ */
 struct RunBuf {

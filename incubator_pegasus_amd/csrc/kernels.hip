@@ -602,13 +602,58 @@ __global__ void k_first_eq(const DevRun *runs, const uint64_t *view, uint64_t n,
  * KeyWithTTLCompactionFilter::Filter (key_ttl_compaction_filter.h:55-92) on
  * the surviving newest PUT; tombstones and shadowed versions dropped
  * (bottommost CompactRange, do_manual_compact:3389). */
+/* per-record disposition (stats aggregated per wave via ballot — a single
+ * shared atomic per record serializes at ~88 adds/us and was 85% of the
+ * compaction pass before this) */
+enum { D_NONE = 0, D_KEEP, D_SHADOWED, D_TOMBSTONE, D_EXPIRED, D_FILTERED };
+
+__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order, uint64_t m,
+                                          const CompactParams &cp, uint8_t *keep,
+                                          uint8_t *changed, uint32_t *new_expire, uint64_t *ksz,
+                                          uint64_t *vsz, uint64_t p);
+
 __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint64_t m,
                                 CompactParams cp, uint8_t *keep, uint8_t *changed,
                                 uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
                                 CompactStatsDev *stats)
 {
-    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < m;
-         p += gridDim.x * (uint64_t)blockDim.x) {
+    uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+    uint64_t stride = gridDim.x * (uint64_t)blockDim.x;
+    uint64_t iters = (m + stride - 1) / stride;
+    int lane = threadIdx.x % WAVE;
+    for (uint64_t it = 0; it < iters; it++) {
+        uint64_t p = tid + it * stride;
+        int disp = D_NONE;
+        if (p < m) {
+            disp = compact_disposition(runs, order, m, cp, keep, changed, new_expire, ksz, vsz,
+                                       p);
+        }
+        /* wave-aggregated stats */
+        unsigned long long b;
+        b = __ballot(disp == D_SHADOWED);
+        if (lane == 0 && b)
+            atomicAdd(&stats->shadowed, (unsigned long long)__popcll(b));
+        b = __ballot(disp == D_TOMBSTONE);
+        if (lane == 0 && b)
+            atomicAdd(&stats->tombstones, (unsigned long long)__popcll(b));
+        b = __ballot(disp == D_EXPIRED);
+        if (lane == 0 && b)
+            atomicAdd(&stats->expired, (unsigned long long)__popcll(b));
+        b = __ballot(disp == D_FILTERED);
+        if (lane == 0 && b)
+            atomicAdd(&stats->filtered, (unsigned long long)__popcll(b));
+        b = __ballot(disp == D_KEEP);
+        if (lane == 0 && b)
+            atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+    }
+}
+
+__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order, uint64_t m,
+                                          const CompactParams &cp, uint8_t *keep,
+                                          uint8_t *changed, uint32_t *new_expire, uint64_t *ksz,
+                                          uint64_t *vsz, uint64_t p)
+{
+    {
         uint64_t id = order[p];
         const DevRun &r = runs[id >> 40];
         uint64_t i = id & 0xFFFFFFFFFFull;
@@ -626,13 +671,11 @@ __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint6
             uint64_t pi = pid & 0xFFFFFFFFFFull, pkl;
             const uint8_t *pk = run_key(pr, pi, &pkl);
             if (dev_key_cmp(k, kl, pk, pkl) == 0) {
-                atomicAdd(&stats->shadowed, 1ull);
-                continue;
+                return D_SHADOWED;
             }
         }
         if (r.sk[i] & 1) {
-            atomicAdd(&stats->tombstones, 1ull);
-            continue;
+            return D_TOMBSTONE;
         }
         /* --- Filter --- */
         int drop = 0, value_changed = 0;
@@ -704,21 +747,18 @@ __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint6
                 }
             }
             if (drop) {
-                atomicAdd(&stats->filtered, 1ull);
-                continue;
+                return D_FILTERED;
             }
             /* final keep/drop on local expire_ts (post default-ttl, pre
              * user-op) + stale split hash (:91,114-121) */
             if (dev_ts_expired(cp.epoch_now, expire_ts)) {
-                atomicAdd(&stats->expired, 1ull);
-                continue;
+                return D_EXPIRED;
             }
             if (cp.validate_hash && cp.partition_version >= 0 &&
                 cp.pidx <= cp.partition_version &&
                 (int64_t)(dev_key_hash(k, kl) & (uint64_t)cp.partition_version) !=
                     (int64_t)cp.pidx) {
-                atomicAdd(&stats->filtered, 1ull);
-                continue;
+                return D_FILTERED;
             }
         }
         keep[p] = 1;
@@ -726,7 +766,7 @@ __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint6
         new_expire[p] = new_ts_val;
         ksz[p] = kl;
         vsz[p] = vl;
-        atomicAdd(&stats->output_records, 1ull);
+        return D_KEEP;
     }
 }
 
