@@ -568,6 +568,27 @@ struct HipEngine {
         return d;
     }
 
+    /* stream-ordered (pooled) temporaries — avoids a plain hipMalloc/hipFree
+     * round trip per operation; pool release threshold raised at open */
+    template <typename T> T *talloc(uint64_t n_bytes)
+    {
+        void *d = nullptr;
+        HIP_OK(hipMallocAsync(&d, n_bytes ? n_bytes : 1, stream));
+        return (T *)d;
+    }
+    void tfree(void *p)
+    {
+        if (p)
+            HIP_OK(hipFreeAsync(p, stream));
+    }
+    uint8_t *upload_tmp(const void *p, uint64_t n)
+    {
+        uint8_t *d = talloc<uint8_t>(n);
+        if (n)
+            HIP_OK(hipMemcpyAsync(d, p, n, hipMemcpyHostToDevice, stream));
+        return d;
+    }
+
     /* build the visible view for [start, stop_excl); returns device array
      * (caller frees) + count */
     void build_view(const std::string *start, const std::string *stop_excl, uint64_t **out_view,
@@ -579,11 +600,10 @@ struct HipEngine {
         if (R == 0)
             return;
         DevRun *dr = dev_runs();
-        uint64_t *d_lo, *d_hi;
-        HIP_OK(hipMalloc(&d_lo, R * 8));
-        HIP_OK(hipMalloc(&d_hi, R * 8));
-        uint8_t *d_start = start ? upload_bytes(start->data(), start->size()) : nullptr;
-        uint8_t *d_stop = stop_excl ? upload_bytes(stop_excl->data(), stop_excl->size()) : nullptr;
+        uint64_t *d_lo = talloc<uint64_t>(R * 8);
+        uint64_t *d_hi = talloc<uint64_t>(R * 8);
+        uint8_t *d_start = start ? upload_tmp(start->data(), start->size()) : nullptr;
+        uint8_t *d_stop = stop_excl ? upload_tmp(stop_excl->data(), stop_excl->size()) : nullptr;
         launch_bounds(dr, R, d_start, start ? start->size() : 0, d_lo, 0, stream);
         launch_bounds(dr, R, d_stop, stop_excl ? stop_excl->size() : 0, d_hi, 0, stream);
         std::vector<uint64_t> lo(R), hi(R), wprefix(R + 1);
@@ -598,22 +618,19 @@ struct HipEngine {
             total += hi[r] - lo[r];
         }
         wprefix[R] = total;
-        if (d_start)
-            (void)hipFree(d_start);
-        if (d_stop)
-            (void)hipFree(d_stop);
+        tfree(d_start);
+        tfree(d_stop);
         if (total == 0) {
-            (void)hipFree(d_lo);
-            (void)hipFree(d_hi);
+            tfree(d_lo);
+            tfree(d_hi);
             return;
         }
         /* re-upload corrected hi + wprefix */
         HIP_OK(hipMemcpy(d_hi, hi.data(), R * 8, hipMemcpyHostToDevice));
-        uint64_t *d_wp = (uint64_t *)upload_bytes(wprefix.data(), (R + 1) * 8);
-        uint64_t *d_order, *d_flags, *d_pos;
-        HIP_OK(hipMalloc(&d_order, total * 8));
-        HIP_OK(hipMalloc(&d_flags, total * 8));
-        HIP_OK(hipMalloc(&d_pos, total * 8));
+        uint64_t *d_wp = (uint64_t *)upload_tmp(wprefix.data(), (R + 1) * 8);
+        uint64_t *d_order = talloc<uint64_t>(total * 8);
+        uint64_t *d_flags = talloc<uint64_t>(total * 8);
+        uint64_t *d_pos = talloc<uint64_t>(total * 8);
         launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, stream);
         launch_visible(dr, d_order, total, d_flags, stream);
         launch_psum(d_flags, d_pos, total, stream);
@@ -628,12 +645,13 @@ struct HipEngine {
             launch_gather(d_order, d_flags, d_pos, total, d_view, stream);
             HIP_OK(hipStreamSynchronize(stream));
         }
-        (void)hipFree(d_lo);
-        (void)hipFree(d_hi);
-        (void)hipFree(d_wp);
-        (void)hipFree(d_order);
-        (void)hipFree(d_flags);
-        (void)hipFree(d_pos);
+        tfree(d_lo);
+        tfree(d_hi);
+        tfree(d_wp);
+        tfree(d_order);
+        tfree(d_flags);
+        tfree(d_pos);
+        HIP_OK(hipStreamSynchronize(stream));
         *out_view = d_view;
         *out_n = nv;
     }
@@ -675,6 +693,13 @@ void *rrdb_open(int32_t app_id, int32_t pidx, int32_t gpu_id)
         launch_crc64_table_init(host_crc64_table);
     });
     HIP_OK(hipStreamCreate(&e->stream));
+    {
+        hipMemPool_t pool;
+        if (hipDeviceGetDefaultMemPool(&pool, gpu_id) == hipSuccess) {
+            uint64_t thr = UINT64_MAX;
+            (void)hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &thr);
+        }
+    }
     return e;
 }
 
@@ -808,14 +833,12 @@ static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const ui
     if (R == 0)
         return RRDB_OK;
     DevRun *dr = e->dev_runs();
-    uint8_t *d_keys = e->upload_bytes(keys, key_offs[nq]);
-    uint64_t *d_offs = (uint64_t *)e->upload_bytes(key_offs, (nq + 1) * 8);
-    int32_t *d_status;
-    uint64_t *d_hit, *d_ulen, *d_voffs;
-    HIP_OK(hipMalloc(&d_status, nq * 4));
-    HIP_OK(hipMalloc(&d_hit, nq * 8));
-    HIP_OK(hipMalloc(&d_ulen, nq * 8));
-    HIP_OK(hipMalloc(&d_voffs, (nq + 1) * 8));
+    uint8_t *d_keys = e->upload_tmp(keys, key_offs[nq]);
+    uint64_t *d_offs = (uint64_t *)e->upload_tmp(key_offs, (nq + 1) * 8);
+    int32_t *d_status = e->talloc<int32_t>(nq * 4);
+    uint64_t *d_hit = e->talloc<uint64_t>(nq * 8);
+    uint64_t *d_ulen = e->talloc<uint64_t>(nq * 8);
+    uint64_t *d_voffs = e->talloc<uint64_t>((nq + 1) * 8);
     launch_get(dr, R, d_keys, d_offs, nq, epoch_now, e->data_version, d_status, d_hit, d_ulen,
                nullptr, e->stream);
     launch_psum(d_ulen, d_voffs, nq, e->stream);
@@ -830,22 +853,21 @@ static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const ui
     voffs[nq] = total;
     uint8_t *blob = (uint8_t *)a->alloc(total);
     if (total) {
-        uint8_t *d_blob;
-        HIP_OK(hipMalloc(&d_blob, total));
+        uint8_t *d_blob = e->talloc<uint8_t>(total);
         launch_emit_values(dr, d_hit, d_status, nq, e->data_version, d_voffs, d_blob, e->stream);
         HIP_OK(hipMemcpyAsync(blob, d_blob, total, hipMemcpyDeviceToHost, e->stream));
         HIP_OK(hipStreamSynchronize(e->stream));
-        (void)hipFree(d_blob);
+        e->tfree(d_blob);
     }
     for (uint64_t i = 0; i < nq; i++)
         if (status[i] == RRDB_OK)
             vals[i] = {blob + voffs[i], voffs[i + 1] - voffs[i]};
-    (void)hipFree(d_keys);
-    (void)hipFree(d_offs);
-    (void)hipFree(d_status);
-    (void)hipFree(d_hit);
-    (void)hipFree(d_ulen);
-    (void)hipFree(d_voffs);
+    e->tfree(d_keys);
+    e->tfree(d_offs);
+    e->tfree(d_status);
+    e->tfree(d_hit);
+    e->tfree(d_ulen);
+    e->tfree(d_voffs);
     return RRDB_OK;
 }
 
@@ -914,15 +936,12 @@ int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
     }
     DevRun *dr = e->dev_runs();
     uint64_t offs[2] = {0, key_len};
-    uint8_t *d_key = e->upload_bytes(key, key_len);
-    uint64_t *d_offs = (uint64_t *)e->upload_bytes(offs, 16);
-    int32_t *d_status;
-    uint64_t *d_hit, *d_ulen;
-    uint32_t *d_expire;
-    HIP_OK(hipMalloc(&d_status, 4));
-    HIP_OK(hipMalloc(&d_hit, 8));
-    HIP_OK(hipMalloc(&d_ulen, 8));
-    HIP_OK(hipMalloc(&d_expire, 4));
+    uint8_t *d_key = e->upload_tmp(key, key_len);
+    uint64_t *d_offs = (uint64_t *)e->upload_tmp(offs, 16);
+    int32_t *d_status = e->talloc<int32_t>(4);
+    uint64_t *d_hit = e->talloc<uint64_t>(8);
+    uint64_t *d_ulen = e->talloc<uint64_t>(8);
+    uint32_t *d_expire = e->talloc<uint32_t>(4);
     launch_get(dr, R, d_key, d_offs, 1, epoch_now, e->data_version, d_status, d_hit, d_ulen,
                d_expire, e->stream);
     int32_t st;
@@ -930,12 +949,12 @@ int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
     HIP_OK(hipMemcpyAsync(&st, d_status, 4, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&expire, d_expire, 4, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipStreamSynchronize(e->stream));
-    (void)hipFree(d_key);
-    (void)hipFree(d_offs);
-    (void)hipFree(d_status);
-    (void)hipFree(d_hit);
-    (void)hipFree(d_ulen);
-    (void)hipFree(d_expire);
+    e->tfree(d_key);
+    e->tfree(d_offs);
+    e->tfree(d_status);
+    e->tfree(d_hit);
+    e->tfree(d_ulen);
+    e->tfree(d_expire);
     (void)a;
     out->error = st;
     if (st == RRDB_OK)
@@ -983,14 +1002,12 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
     sp.no_value = c->no_value;
     sp.hash_key_skip = 0;
 
-    uint8_t *d_state;
-    uint64_t *d_ksz, *d_vsz, *d_flags, *d_npos, *d_cut;
-    HIP_OK(hipMalloc(&d_state, w));
-    HIP_OK(hipMalloc(&d_ksz, w * 8));
-    HIP_OK(hipMalloc(&d_vsz, w * 8));
-    HIP_OK(hipMalloc(&d_flags, w * 8));
-    HIP_OK(hipMalloc(&d_npos, w * 8));
-    HIP_OK(hipMalloc(&d_cut, 16));
+    uint8_t *d_state = e->talloc<uint8_t>(w);
+    uint64_t *d_ksz = e->talloc<uint64_t>(w * 8);
+    uint64_t *d_vsz = e->talloc<uint64_t>(w * 8);
+    uint64_t *d_flags = e->talloc<uint64_t>(w * 8);
+    uint64_t *d_npos = e->talloc<uint64_t>(w * 8);
+    uint64_t *d_cut = e->talloc<uint64_t>(16);
     hipEvent_t sev[4];
     for (auto &x : sev)
         HIP_OK(hipEventCreate(&x));
@@ -1012,9 +1029,8 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
         out->count = 0;
     } else {
         launch_cut_sizes(d_state, w, consumed, d_ksz, d_vsz, e->stream);
-        uint64_t *d_koffs, *d_voffs;
-        HIP_OK(hipMalloc(&d_koffs, w * 8));
-        HIP_OK(hipMalloc(&d_voffs, w * 8));
+        uint64_t *d_koffs = e->talloc<uint64_t>(w * 8);
+        uint64_t *d_voffs = e->talloc<uint64_t>(w * 8);
         launch_psum(d_ksz, d_koffs, w, e->stream);
         launch_psum(d_vsz, d_voffs, w, e->stream);
         uint64_t t[4] = {0, 0, 0, 0};
@@ -1027,12 +1043,20 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
         uint8_t *d_kout, *d_vout;
         uint64_t *d_kooffs, *d_vooffs;
         int32_t *d_ets = nullptr;
-        HIP_OK(hipMalloc(&d_kout, kbytes ? kbytes : 1));
-        HIP_OK(hipMalloc(&d_vout, vbytes ? vbytes : 1));
-        HIP_OK(hipMalloc(&d_kooffs, (n_out + 1) * 8));
-        HIP_OK(hipMalloc(&d_vooffs, (n_out + 1) * 8));
+        if (c->on_device_out) {
+            /* handed to the caller: plain allocations, freed by rrdb_free_result */
+            HIP_OK(hipMalloc(&d_kout, kbytes ? kbytes : 1));
+            HIP_OK(hipMalloc(&d_vout, vbytes ? vbytes : 1));
+            HIP_OK(hipMalloc(&d_kooffs, (n_out + 1) * 8));
+            HIP_OK(hipMalloc(&d_vooffs, (n_out + 1) * 8));
+        } else {
+            d_kout = e->talloc<uint8_t>(kbytes);
+            d_vout = e->talloc<uint8_t>(vbytes);
+            d_kooffs = e->talloc<uint64_t>((n_out + 1) * 8);
+            d_vooffs = e->talloc<uint64_t>((n_out + 1) * 8);
+        }
         if (c->return_expire_ts)
-            HIP_OK(hipMalloc(&d_ets, n_out * 4));
+            d_ets = e->talloc<int32_t>(n_out * 4);
         HIP_OK(hipEventRecord(sev[2], e->stream));
         launch_emit_scan(dr, d_win, w, d_state, consumed, d_npos, d_koffs, d_voffs, sp, d_kout,
                          d_vout, d_kooffs, d_vooffs, d_ets, n_out, e->stream);
@@ -1044,8 +1068,6 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
             out->dev_vals = d_vout;
             out->dev_val_offs = d_vooffs;
             a->dev_ptrs.insert(a->dev_ptrs.end(), {d_kout, d_vout, d_kooffs, d_vooffs});
-            if (d_ets)
-                (void)hipFree(d_ets);
         } else {
             uint8_t *kb = (uint8_t *)a->alloc(kbytes);
             uint8_t *vb = (uint8_t *)a->alloc(vbytes);
@@ -1069,16 +1091,16 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
                 out->values[i] = {vb + vooffs[i], vooffs[i + 1] - vooffs[i]};
             }
             out->expire_ts = ets;
-            (void)hipFree(d_kout);
-            (void)hipFree(d_vout);
-            (void)hipFree(d_kooffs);
-            (void)hipFree(d_vooffs);
-            if (d_ets)
-                (void)hipFree(d_ets);
+            e->tfree(d_kout);
+            e->tfree(d_vout);
+            e->tfree(d_kooffs);
+            e->tfree(d_vooffs);
         }
+        if (d_ets)
+            e->tfree(d_ets); /* always pool-allocated; never handed out */
         out->count = n_out;
-        HIP_OK(hipFree(d_koffs));
-        HIP_OK(hipFree(d_voffs));
+        e->tfree(d_koffs);
+        e->tfree(d_voffs);
     }
     {
         float ms;
@@ -1091,12 +1113,12 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
     }
     for (auto &x : sev)
         (void)hipEventDestroy(x);
-    (void)hipFree(d_state);
-    (void)hipFree(d_ksz);
-    (void)hipFree(d_vsz);
-    (void)hipFree(d_flags);
-    (void)hipFree(d_npos);
-    (void)hipFree(d_cut);
+    e->tfree(d_state);
+    e->tfree(d_ksz);
+    e->tfree(d_vsz);
+    e->tfree(d_flags);
+    e->tfree(d_npos);
+    e->tfree(d_cut);
     c->cursor += consumed;
     out->error = RRDB_OK;
     out->context_id =
@@ -1138,16 +1160,15 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
     e->build_view(&start, &stop_excl, &ctx->d_view, &ctx->view_n);
     /* first_exclusive: skip an exact start match (on_get_scanner:1277-1283) */
     if (!start_inclusive && ctx->view_n > 0) {
-        uint32_t *d_eq;
-        HIP_OK(hipMalloc(&d_eq, 4));
-        uint8_t *d_sk = e->upload_bytes(start.data(), start.size());
+        uint32_t *d_eq = e->talloc<uint32_t>(4);
+        uint8_t *d_sk = e->upload_tmp(start.data(), start.size());
         launch_first_eq(e->dev_runs(), ctx->d_view, ctx->view_n, d_sk, start.size(), d_eq,
                         e->stream);
         uint32_t eq = 0;
         HIP_OK(hipMemcpyAsync(&eq, d_eq, 4, hipMemcpyDeviceToHost, e->stream));
         HIP_OK(hipStreamSynchronize(e->stream));
-        (void)hipFree(d_eq);
-        (void)hipFree(d_sk);
+        e->tfree(d_eq);
+        e->tfree(d_sk);
         if (eq)
             ctx->cursor = 1;
     }
@@ -1239,13 +1260,11 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
         sp.pidx = e->pidx;
         sp.partition_version = e->partition_version;
         sp.validate_hash = 0;
-        uint8_t *d_state;
-        uint64_t *d_ksz, *d_vsz, *d_flags, *d_npos;
-        HIP_OK(hipMalloc(&d_state, n));
-        HIP_OK(hipMalloc(&d_ksz, n * 8));
-        HIP_OK(hipMalloc(&d_vsz, n * 8));
-        HIP_OK(hipMalloc(&d_flags, n * 8));
-        HIP_OK(hipMalloc(&d_npos, n * 8));
+        uint8_t *d_state = e->talloc<uint8_t>(n);
+        uint64_t *d_ksz = e->talloc<uint64_t>(n * 8);
+        uint64_t *d_vsz = e->talloc<uint64_t>(n * 8);
+        uint64_t *d_flags = e->talloc<uint64_t>(n * 8);
+        uint64_t *d_npos = e->talloc<uint64_t>(n * 8);
         launch_scan_state(dr, d_view, n, sp, d_state, d_ksz, d_vsz, e->stream);
         launch_normal_flags(d_state, n, d_flags, e->stream);
         launch_psum(d_flags, d_npos, n, e->stream);
@@ -1254,11 +1273,11 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
         HIP_OK(hipMemcpyAsync(&lastf, d_flags + n - 1, 8, hipMemcpyDeviceToHost, e->stream));
         HIP_OK(hipStreamSynchronize(e->stream));
         count = (int64_t)(lastp + lastf);
-        (void)hipFree(d_state);
-        (void)hipFree(d_ksz);
-        (void)hipFree(d_vsz);
-        (void)hipFree(d_flags);
-        (void)hipFree(d_npos);
+        e->tfree(d_state);
+        e->tfree(d_ksz);
+        e->tfree(d_vsz);
+        e->tfree(d_flags);
+        e->tfree(d_npos);
     }
     if (d_view)
         (void)hipFree(d_view);
@@ -1388,16 +1407,14 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     sp.sk_ft = q->sort_key_filter_type;
     sp.sk_pat = nullptr;
     sp.sk_pat_len = q->sort_key_filter_pattern.len;
-    uint8_t *d_skpat = e->upload_bytes(q->sort_key_filter_pattern.data, sp.sk_pat_len);
+    uint8_t *d_skpat = e->upload_tmp(q->sort_key_filter_pattern.data, sp.sk_pat_len);
     sp.sk_pat = d_skpat;
     sp.no_value = q->no_value;
     sp.hash_key_skip = 2 + q->hash_key.len; /* emit sortkey only (:2496-2499) */
 
-    uint8_t *d_state;
-    uint64_t *d_ksz, *d_vsz;
-    HIP_OK(hipMalloc(&d_state, w));
-    HIP_OK(hipMalloc(&d_ksz, w * 8));
-    HIP_OK(hipMalloc(&d_vsz, w * 8));
+    uint8_t *d_state = e->talloc<uint8_t>(w);
+    uint64_t *d_ksz = e->talloc<uint64_t>(w * 8);
+    uint64_t *d_vsz = e->talloc<uint64_t>(w * 8);
     /* sortkey filter types: scan_state's sk filter operates on key minus
      * 2+hklen prefix — same as multi_get's sortkey check */
     launch_scan_state(dr, d_view + wstart, w, sp, d_state, d_ksz, d_vsz, e->stream);
@@ -1407,9 +1424,9 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     HIP_OK(hipMemcpyAsync(ksz.data(), d_ksz, w * 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(vsz.data(), d_vsz, w * 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipStreamSynchronize(e->stream));
-    (void)hipFree(d_state);
-    (void)hipFree(d_ksz);
-    (void)hipFree(d_vsz);
+    e->tfree(d_state);
+    e->tfree(d_ksz);
+    e->tfree(d_vsz);
 
     /* host limiter loop, mirroring on_multi_get:616-778 exactly */
     int64_t count = 0, size = 0;
@@ -1430,15 +1447,14 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
             bool check = (!q->reverse && !start_inclusive) || (q->reverse && !stop_inclusive);
             if (check) {
                 const std::string &bound = q->reverse ? stop : start;
-                uint32_t *d_eq;
-                HIP_OK(hipMalloc(&d_eq, 4));
-                uint8_t *d_b = e->upload_bytes(bound.data(), bound.size());
+                uint32_t *d_eq = e->talloc<uint32_t>(4);
+                uint8_t *d_b = e->upload_tmp(bound.data(), bound.size());
                 launch_first_eq(dr, d_view + wstart + wi, 1, d_b, bound.size(), d_eq, e->stream);
                 uint32_t eq = 0;
                 HIP_OK(hipMemcpyAsync(&eq, d_eq, 4, hipMemcpyDeviceToHost, e->stream));
                 HIP_OK(hipStreamSynchronize(e->stream));
-                (void)hipFree(d_eq);
-                (void)hipFree(d_b);
+                e->tfree(d_eq);
+                e->tfree(d_b);
                 if (eq) {
                     skipped_first = true;
                     continue;
@@ -1479,12 +1495,11 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
         }
         koffs[m] = kb;
         voffs[m] = vb;
-        uint64_t *d_rows = (uint64_t *)e->upload_bytes(sel.data(), m * 8);
-        uint64_t *d_ko = (uint64_t *)e->upload_bytes(koffs.data(), (m + 1) * 8);
-        uint64_t *d_vo = (uint64_t *)e->upload_bytes(voffs.data(), (m + 1) * 8);
-        uint8_t *d_kout, *d_vout;
-        HIP_OK(hipMalloc(&d_kout, kb ? kb : 1));
-        HIP_OK(hipMalloc(&d_vout, vb ? vb : 1));
+        uint64_t *d_rows = (uint64_t *)e->upload_tmp(sel.data(), m * 8);
+        uint64_t *d_ko = (uint64_t *)e->upload_tmp(koffs.data(), (m + 1) * 8);
+        uint64_t *d_vo = (uint64_t *)e->upload_tmp(voffs.data(), (m + 1) * 8);
+        uint8_t *d_kout = e->talloc<uint8_t>(kb);
+        uint8_t *d_vout = e->talloc<uint8_t>(vb);
         launch_emit_rows(dr, d_view, d_rows, m, d_ko, d_vo, sp, d_kout, d_vout, e->stream);
         uint8_t *hkb = (uint8_t *)a->alloc(kb);
         uint8_t *hvb = (uint8_t *)a->alloc(vb);
@@ -1495,13 +1510,13 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
             out->keys[j] = {hkb + koffs[j], koffs[j + 1] - koffs[j]};
             out->values[j] = {hvb + voffs[j], voffs[j + 1] - voffs[j]};
         }
-        (void)hipFree(d_rows);
-        (void)hipFree(d_ko);
-        (void)hipFree(d_vo);
-        (void)hipFree(d_kout);
-        (void)hipFree(d_vout);
+        e->tfree(d_rows);
+        e->tfree(d_ko);
+        e->tfree(d_vo);
+        e->tfree(d_kout);
+        e->tfree(d_vout);
     }
-    (void)hipFree(d_skpat);
+    e->tfree(d_skpat);
     (void)hipFree(d_view);
     out->count = m;
     out->error = complete ? RRDB_OK : RRDB_INCOMPLETE; /* :789-799 */
@@ -1539,11 +1554,10 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         acc += e->runs[r].n;
     }
     wprefix[R] = acc;
-    uint64_t *d_lo = (uint64_t *)e->upload_bytes(lo.data(), R * 8);
-    uint64_t *d_hi = (uint64_t *)e->upload_bytes(hi.data(), R * 8);
-    uint64_t *d_wp = (uint64_t *)e->upload_bytes(wprefix.data(), (R + 1) * 8);
-    uint64_t *d_order;
-    HIP_OK(hipMalloc(&d_order, total * 8));
+    uint64_t *d_lo = (uint64_t *)e->upload_tmp(lo.data(), R * 8);
+    uint64_t *d_hi = (uint64_t *)e->upload_tmp(hi.data(), R * 8);
+    uint64_t *d_wp = (uint64_t *)e->upload_tmp(wprefix.data(), (R + 1) * 8);
+    uint64_t *d_order = e->talloc<uint64_t>(total * 8);
     hipEvent_t ev[6];
     for (auto &x : ev)
         HIP_OK(hipEventCreate(&x));
@@ -1563,26 +1577,21 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     cp.rules = e->d_rules;
     cp.pats = e->d_pats;
 
-    uint8_t *d_keep, *d_changed;
-    uint32_t *d_new_expire;
-    uint64_t *d_ksz, *d_vsz;
-    CompactStatsDev *d_stats;
-    HIP_OK(hipMalloc(&d_keep, total));
-    HIP_OK(hipMalloc(&d_changed, total));
-    HIP_OK(hipMalloc(&d_new_expire, total * 4));
-    HIP_OK(hipMalloc(&d_ksz, total * 8));
-    HIP_OK(hipMalloc(&d_vsz, total * 8));
-    HIP_OK(hipMalloc(&d_stats, sizeof(CompactStatsDev)));
+    uint8_t *d_keep = e->talloc<uint8_t>(total);
+    uint8_t *d_changed = e->talloc<uint8_t>(total);
+    uint32_t *d_new_expire = e->talloc<uint32_t>(total * 4);
+    uint64_t *d_ksz = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_vsz = e->talloc<uint64_t>(total * 8);
+    CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
     HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
     launch_compact_flags(dr, d_order, total, cp, d_keep, d_changed, d_new_expire, d_ksz, d_vsz,
                          d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[2], e->stream));
     /* positions + offsets */
-    uint64_t *d_keepw, *d_kpos, *d_koffs, *d_voffs;
-    HIP_OK(hipMalloc(&d_keepw, total * 8));
-    HIP_OK(hipMalloc(&d_kpos, total * 8));
-    HIP_OK(hipMalloc(&d_koffs, total * 8));
-    HIP_OK(hipMalloc(&d_voffs, total * 8));
+    uint64_t *d_keepw = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_kpos = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_koffs = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_voffs = e->talloc<uint64_t>(total * 8);
     launch_widen_u8(d_keep, total, d_keepw, e->stream);
     launch_psum(d_keepw, d_kpos, total, e->stream);
     launch_psum(d_ksz, d_koffs, total, e->stream);
@@ -1636,20 +1645,20 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     }
     for (auto &x : ev)
         (void)hipEventDestroy(x);
-    (void)hipFree(d_lo);
-    (void)hipFree(d_hi);
-    (void)hipFree(d_wp);
-    (void)hipFree(d_order);
-    (void)hipFree(d_keep);
-    (void)hipFree(d_changed);
-    (void)hipFree(d_new_expire);
-    (void)hipFree(d_ksz);
-    (void)hipFree(d_vsz);
-    (void)hipFree(d_stats);
-    (void)hipFree(d_keepw);
-    (void)hipFree(d_kpos);
-    (void)hipFree(d_koffs);
-    (void)hipFree(d_voffs);
+    e->tfree(d_lo);
+    e->tfree(d_hi);
+    e->tfree(d_wp);
+    e->tfree(d_order);
+    e->tfree(d_keep);
+    e->tfree(d_changed);
+    e->tfree(d_new_expire);
+    e->tfree(d_ksz);
+    e->tfree(d_vsz);
+    e->tfree(d_stats);
+    e->tfree(d_keepw);
+    e->tfree(d_kpos);
+    e->tfree(d_koffs);
+    e->tfree(d_voffs);
     if (opts && opts->keep_inputs) {
         /* benchmarking: the pass ran in full; drop the output, keep inputs */
         if (n_out > 0)

@@ -91,6 +91,20 @@ __device__ static inline int dev_key_cmp(const uint8_t *a, uint64_t alen, const 
     return alen < blen ? -1 : (alen > blen ? 1 : 0);
 }
 
+/* lane-strided record copy: 4B unaligned chunks (consecutive lanes touch
+ * consecutive addresses -> coalesced), byte tail */
+__device__ static inline void wave_copy(uint8_t *dst, const uint8_t *src, uint64_t n, int lane)
+{
+    uint64_t n4 = n >> 2;
+    for (uint64_t c = lane; c < n4; c += WAVE) {
+        uint32_t w;
+        __builtin_memcpy(&w, src + 4 * c, 4);
+        __builtin_memcpy(dst + 4 * c, &w, 4);
+    }
+    for (uint64_t b = (n4 << 2) + lane; b < n; b += WAVE)
+        dst[b] = src[b];
+}
+
 __device__ static inline const uint8_t *run_key(const DevRun &r, uint64_t i, uint64_t *len)
 {
     uint64_t o = r.koff[i];
@@ -399,11 +413,7 @@ __global__ void k_emit_values(const DevRun *runs, const uint64_t *hit, const int
         const DevRun &r = runs[id >> 40];
         uint64_t i = id & 0xFFFFFFFFFFull, vl;
         const uint8_t *v = run_val(r, i, &vl);
-        const uint8_t *src = v + hdr;
-        uint64_t n = vl - hdr;
-        uint8_t *dst = vout + voffs[t];
-        for (uint64_t b = lane; b < n; b += WAVE)
-            dst[b] = src[b];
+        wave_copy(vout + voffs[t], v + hdr, vl - hdr, lane);
     }
 }
 
@@ -525,15 +535,9 @@ __global__ void k_emit_scan(const DevRun *runs, const uint64_t *view, uint64_t w
         const uint8_t *v = run_val(r, i, &vl);
         const uint8_t *ksrc = k + sp.hash_key_skip;
         uint64_t kn = kl - sp.hash_key_skip;
-        uint8_t *kdst = kout + koffs[p];
-        for (uint64_t b = lane; b < kn; b += WAVE)
-            kdst[b] = ksrc[b];
+        wave_copy(kout + koffs[p], ksrc, kn, lane);
         if (!sp.no_value) {
-            const uint8_t *vsrc = v + hdr;
-            uint64_t vn = vl - hdr;
-            uint8_t *vdst = vout + voffs[p];
-            for (uint64_t b = lane; b < vn; b += WAVE)
-                vdst[b] = vsrc[b];
+            wave_copy(vout + voffs[p], v + hdr, vl - hdr, lane);
         }
         if (lane == 0) {
             kout_offs[o] = koffs[p];
@@ -568,15 +572,9 @@ __global__ void k_emit_rows(const DevRun *runs, const uint64_t *view, const uint
         const uint8_t *v = run_val(r, i, &vl);
         const uint8_t *ksrc = k + sp.hash_key_skip;
         uint64_t kn = kl - sp.hash_key_skip;
-        uint8_t *kdst = kout + koffs[j];
-        for (uint64_t b = lane; b < kn; b += WAVE)
-            kdst[b] = ksrc[b];
+        wave_copy(kout + koffs[j], ksrc, kn, lane);
         if (!sp.no_value) {
-            const uint8_t *vsrc = v + hdr;
-            uint64_t vn = vl - hdr;
-            uint8_t *vdst = vout + voffs[j];
-            for (uint64_t b = lane; b < vn; b += WAVE)
-                vdst[b] = vsrc[b];
+            wave_copy(vout + voffs[j], v + hdr, vl - hdr, lane);
         }
     }
 }
@@ -792,12 +790,10 @@ __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64
         uint64_t kl, vl;
         const uint8_t *k = run_key(r, i, &kl);
         const uint8_t *v = run_val(r, i, &vl);
-        uint8_t *kdst = kout + koffs[p];
-        for (uint64_t b = lane; b < kl; b += WAVE)
-            kdst[b] = k[b];
+        wave_copy(kout + koffs[p], k, kl, lane);
         uint8_t *vdst = vout + voffs[p];
-        for (uint64_t b = lane; b < vl; b += WAVE)
-            vdst[b] = v[b];
+        wave_copy(vdst, v, vl, lane);
+        __builtin_amdgcn_wave_barrier(); /* patch below must see the copy */
         if (lane == 0) {
             if (changed[p]) {
                 uint32_t ts = new_expire[p];
