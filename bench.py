@@ -44,6 +44,27 @@ def log(msg):
         print(f"[bench] {msg}", file=sys.stderr, flush=True)
 
 
+def _cpu_compact_worker(params):
+    """Oracle (CPU restatement, checker) compaction pass over one partition.
+    Returns (n_records, compact_seconds).  bench cpu_baseline leg only."""
+    sample, n_runs, pidx, epoch_now = params
+    from incubator_pegasus_amd.capi import RrdbLib
+
+    oracle = RrdbLib(os.path.join(REPO, "oracle", "liboracle.so"))
+    op = oracle.open(1, pidx, -1)
+    try:
+        for r in build_partition_data(sample, n_runs, 0, pidx):
+            op.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
+                                 np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
+        nrec = int(op.num_records())
+        t0 = time.time()
+        err, _ = op.manual_compact(epoch_now, keep_inputs=True)
+        assert err == 0
+        return nrec, time.time() - t0
+    finally:
+        op.close()
+
+
 def build_partition_data(keys_pp, n_runs, rank, pidx, value_len=100):
     from incubator_pegasus_amd import data as D
 
@@ -226,29 +247,36 @@ def main():
     get_ops_per_s = nq / get_elapsed
 
     # ---- CPU baseline (oracle restatement, rank 0, N==1 only) ----
+    # single-core pass + a partition-parallel pass on multiple host cores
+    # (partitions are independent, exactly how the reference parallelizes)
     cpu_baseline = None
     if rank == 0 and world <= 1 and not args.skip_cpu_baseline:
-        from incubator_pegasus_amd.capi import RrdbLib
-
         sample = min(args.cpu_sample, keys_pp)
         log(f"cpu baseline: oracle compaction pass over {sample} keys ...")
-        oracle = RrdbLib(os.path.join(REPO, "oracle", "liboracle.so"))
-        op = oracle.open(1, 0, -1)
-        runs = build_partition_data(sample, args.runs, 0, 0)
-        for r in runs:
-            op.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
-                                 np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
-        nrec = int(op.num_records())
+        nrec1, el1 = _cpu_compact_worker((sample, args.runs, 0, epoch_now))
+        workers = min(os.cpu_count() or 1, args.partitions)
+        log(f"cpu baseline: {workers} parallel partitions x {sample} keys ...")
+        import concurrent.futures as cf
+
         t0 = time.time()
-        err, _ = op.manual_compact(epoch_now, keep_inputs=True)
-        cpu_el = time.time() - t0
-        op.close()
+        with cf.ProcessPoolExecutor(max_workers=workers) as ex:
+            futs = [ex.submit(_cpu_compact_worker, (sample, args.runs, w, epoch_now))
+                    for w in range(workers)]
+            results = [f.result() for f in futs]
+        # wall time includes per-worker data generation; use the sum of
+        # compact-only times / workers as the effective parallel rate and the
+        # max compact time as the conservative wall
+        total_recs = sum(r[0] for r in results)
+        max_el = max(r[1] for r in results)
+        par_rate = total_recs / max_el
         cpu_baseline = {
-            "value": round(nrec / cpu_el, 1),
+            "value": round(par_rate, 1),
             "unit": "keys/s",
-            "cores": 1,
+            "cores": workers,
             "kind": "port",
-            "sample": f"one partition, {nrec} records ({args.runs} runs), single compaction pass"
+            "sample": (f"{workers} parallel partitions x {nrec1} records ({args.runs} runs), "
+                       f"compaction-pass time max over workers; single-core: "
+                       f"{nrec1 / el1:.0f} keys/s"),
         }
 
     for eng in parts:
