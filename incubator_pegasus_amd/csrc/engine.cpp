@@ -29,8 +29,9 @@
 void launch_crc64_table_init(const uint64_t *host_table);
 void launch_bounds(const DevRun *, int, const uint8_t *, uint64_t, uint64_t *, int, hipStream_t);
 void launch_rank(const DevRun *, int, const uint64_t *, const uint64_t *, const uint64_t *,
-                 uint64_t, uint64_t *, hipStream_t);
-void launch_visible(const DevRun *, const uint64_t *, uint64_t, uint64_t *, hipStream_t);
+                 uint64_t, uint64_t *, uint8_t *, hipStream_t);
+void launch_visible(const DevRun *, const uint64_t *, const uint8_t *, uint64_t, uint64_t *,
+                    hipStream_t);
 void launch_gather(const uint64_t *, const uint64_t *, const uint64_t *, uint64_t, uint64_t *,
                    hipStream_t);
 void launch_psum(const uint64_t *, uint64_t *, uint64_t, hipStream_t);
@@ -54,9 +55,9 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       uint8_t *, hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
-void launch_compact_flags(const DevRun *, const uint64_t *, uint64_t, const CompactParams &,
-                          uint8_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
-                          CompactStatsDev *, hipStream_t);
+void launch_compact_flags(const DevRun *, const uint64_t *, const uint8_t *, uint64_t,
+                          const CompactParams &, uint8_t *, uint8_t *, uint32_t *, uint64_t *,
+                          uint64_t *, CompactStatsDev *, hipStream_t);
 void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint8_t *,
                          const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
@@ -629,10 +630,11 @@ struct HipEngine {
         HIP_OK(hipMemcpy(d_hi, hi.data(), R * 8, hipMemcpyHostToDevice));
         uint64_t *d_wp = (uint64_t *)upload_tmp(wprefix.data(), (R + 1) * 8);
         uint64_t *d_order = talloc<uint64_t>(total * 8);
+        uint8_t *d_shadow = talloc<uint8_t>(total);
         uint64_t *d_flags = talloc<uint64_t>(total * 8);
         uint64_t *d_pos = talloc<uint64_t>(total * 8);
-        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, stream);
-        launch_visible(dr, d_order, total, d_flags, stream);
+        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, stream);
+        launch_visible(dr, d_order, d_shadow, total, d_flags, stream);
         launch_psum(d_flags, d_pos, total, stream);
         uint64_t lastp = 0, lastf = 0;
         HIP_OK(hipMemcpyAsync(&lastp, d_pos + total - 1, 8, hipMemcpyDeviceToHost, stream));
@@ -649,6 +651,7 @@ struct HipEngine {
         tfree(d_hi);
         tfree(d_wp);
         tfree(d_order);
+        tfree(d_shadow);
         tfree(d_flags);
         tfree(d_pos);
         HIP_OK(hipStreamSynchronize(stream));
@@ -1558,11 +1561,12 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     uint64_t *d_hi = (uint64_t *)e->upload_tmp(hi.data(), R * 8);
     uint64_t *d_wp = (uint64_t *)e->upload_tmp(wprefix.data(), (R + 1) * 8);
     uint64_t *d_order = e->talloc<uint64_t>(total * 8);
+    uint8_t *d_shadow = e->talloc<uint8_t>(total);
     hipEvent_t ev[6];
     for (auto &x : ev)
         HIP_OK(hipEventCreate(&x));
     HIP_OK(hipEventRecord(ev[0], e->stream));
-    launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, e->stream);
+    launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
 
     CompactParams cp{};
@@ -1584,8 +1588,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     uint64_t *d_vsz = e->talloc<uint64_t>(total * 8);
     CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
     HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
-    launch_compact_flags(dr, d_order, total, cp, d_keep, d_changed, d_new_expire, d_ksz, d_vsz,
-                         d_stats, e->stream);
+    launch_compact_flags(dr, d_order, d_shadow, total, cp, d_keep, d_changed, d_new_expire,
+                         d_ksz, d_vsz, d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[2], e->stream));
     /* positions + offsets */
     uint64_t *d_keepw = e->talloc<uint64_t>(total * 8);
@@ -1616,13 +1620,23 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     st.output_bytes = kbytes + vbytes;
 
     RunBuf nr;
+    bool keep_inputs = opts && opts->keep_inputs;
     if (n_out > 0) {
         nr.n = n_out;
-        HIP_OK(hipMalloc(&nr.keys, kbytes ? kbytes : 1));
-        HIP_OK(hipMalloc(&nr.vals, vbytes ? vbytes : 1));
-        HIP_OK(hipMalloc(&nr.koff, (n_out + 1) * 8));
-        HIP_OK(hipMalloc(&nr.voff, (n_out + 1) * 8));
-        HIP_OK(hipMalloc(&nr.sk, n_out * 8));
+        if (keep_inputs) {
+            /* output is dropped at the end of the pass: pooled temporaries */
+            nr.keys = e->talloc<uint8_t>(kbytes);
+            nr.vals = e->talloc<uint8_t>(vbytes);
+            nr.koff = e->talloc<uint64_t>((n_out + 1) * 8);
+            nr.voff = e->talloc<uint64_t>((n_out + 1) * 8);
+            nr.sk = e->talloc<uint64_t>(n_out * 8);
+        } else {
+            HIP_OK(hipMalloc(&nr.keys, kbytes ? kbytes : 1));
+            HIP_OK(hipMalloc(&nr.vals, vbytes ? vbytes : 1));
+            HIP_OK(hipMalloc(&nr.koff, (n_out + 1) * 8));
+            HIP_OK(hipMalloc(&nr.voff, (n_out + 1) * 8));
+            HIP_OK(hipMalloc(&nr.sk, n_out * 8));
+        }
         HIP_OK(hipEventRecord(ev[3], e->stream));
         launch_emit_compact(dr, d_order, total, d_keep, d_changed, d_new_expire, d_kpos, d_koffs,
                             d_voffs, e->data_version, nr.keys, nr.vals, nr.koff, nr.voff, nr.sk,
@@ -1649,6 +1663,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     e->tfree(d_hi);
     e->tfree(d_wp);
     e->tfree(d_order);
+    e->tfree(d_shadow);
     e->tfree(d_keep);
     e->tfree(d_changed);
     e->tfree(d_new_expire);
@@ -1659,10 +1674,15 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     e->tfree(d_kpos);
     e->tfree(d_koffs);
     e->tfree(d_voffs);
-    if (opts && opts->keep_inputs) {
+    if (keep_inputs) {
         /* benchmarking: the pass ran in full; drop the output, keep inputs */
-        if (n_out > 0)
-            e->free_run(nr);
+        if (n_out > 0) {
+            e->tfree(nr.keys);
+            e->tfree(nr.vals);
+            e->tfree(nr.koff);
+            e->tfree(nr.voff);
+            e->tfree(nr.sk);
+        }
     } else {
         for (auto &r : e->runs)
             e->free_run(r);

@@ -255,7 +255,8 @@ __global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t
  * (run index order == age order: higher r == newer == first among equal keys)
  * Scatter order[rank] = (r<<40)|i.  wprefix[r] = running sum of window sizes. */
 __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
-                       const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order)
+                       const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order,
+                       uint8_t *shadowed /* [total], by rank position */)
 {
     for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
          t += gridDim.x * (uint64_t)blockDim.x) {
@@ -266,39 +267,41 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
         uint64_t kl;
         const uint8_t *k = run_key(runs[r], i, &kl);
         uint64_t rank = i - lo[r];
+        int shadow = 0;
         for (int q = 0; q < R; q++) {
             if (q == r)
                 continue;
-            uint64_t pos = (q > r) ? dev_upper_bound(runs[q], k, kl, lo[q], hi[q])
-                                   : dev_lower_bound(runs[q], k, kl, lo[q], hi[q]);
-            rank += pos - lo[q];
+            if (q > r) {
+                /* newer run: equal keys precede ours; if one exists at ub-1,
+                 * a newer version shadows this record (newest-wins) */
+                uint64_t ub = dev_upper_bound(runs[q], k, kl, lo[q], hi[q]);
+                if (!shadow && ub > lo[q]) {
+                    uint64_t pl;
+                    const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
+                    if (dev_key_cmp(pk, pl, k, kl) == 0)
+                        shadow = 1;
+                }
+                rank += ub - lo[q];
+            } else {
+                rank += dev_lower_bound(runs[q], k, kl, lo[q], hi[q]) - lo[q];
+            }
         }
         order[rank] = ((uint64_t)r << 40) | i;
+        shadowed[rank] = (uint8_t)shadow;
     }
 }
 
 /* visible flag: newest version of its key (first of the equal-key group in
  * (key asc, run desc) order) and not a tombstone */
-__global__ void k_visible(const DevRun *runs, const uint64_t *order, uint64_t m, uint64_t *flags)
+__global__ void k_visible(const DevRun *runs, const uint64_t *order, const uint8_t *shadowed,
+                          uint64_t m, uint64_t *flags)
 {
     for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < m;
          p += gridDim.x * (uint64_t)blockDim.x) {
         uint64_t id = order[p];
         const DevRun &r = runs[id >> 40];
         uint64_t i = id & 0xFFFFFFFFFFull;
-        uint64_t kl;
-        const uint8_t *k = run_key(r, i, &kl);
-        int newest = 1;
-        if (p > 0) {
-            uint64_t pid = order[p - 1];
-            const DevRun &pr = runs[pid >> 40];
-            uint64_t pi = pid & 0xFFFFFFFFFFull, pkl;
-            const uint8_t *pk = run_key(pr, pi, &pkl);
-            if (dev_key_cmp(k, kl, pk, pkl) == 0)
-                newest = 0;
-        }
-        int tomb = (int)(r.sk[i] & 1);
-        flags[p] = (newest && !tomb) ? 1 : 0;
+        flags[p] = (!shadowed[p] && !(r.sk[i] & 1)) ? 1 : 0;
     }
 }
 
@@ -605,15 +608,16 @@ __global__ void k_first_eq(const DevRun *runs, const uint64_t *view, uint64_t n,
  * compaction pass before this) */
 enum { D_NONE = 0, D_KEEP, D_SHADOWED, D_TOMBSTONE, D_EXPIRED, D_FILTERED };
 
-__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order, uint64_t m,
+__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order,
+                                          const uint8_t *shadowed, uint64_t m,
                                           const CompactParams &cp, uint8_t *keep,
                                           uint8_t *changed, uint32_t *new_expire, uint64_t *ksz,
                                           uint64_t *vsz, uint64_t p);
 
-__global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint64_t m,
-                                CompactParams cp, uint8_t *keep, uint8_t *changed,
-                                uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
-                                CompactStatsDev *stats)
+__global__ void k_compact_flags(const DevRun *runs, const uint64_t *order,
+                                const uint8_t *shadowed, uint64_t m, CompactParams cp,
+                                uint8_t *keep, uint8_t *changed, uint32_t *new_expire,
+                                uint64_t *ksz, uint64_t *vsz, CompactStatsDev *stats)
 {
     uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
     uint64_t stride = gridDim.x * (uint64_t)blockDim.x;
@@ -623,8 +627,8 @@ __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint6
         uint64_t p = tid + it * stride;
         int disp = D_NONE;
         if (p < m) {
-            disp = compact_disposition(runs, order, m, cp, keep, changed, new_expire, ksz, vsz,
-                                       p);
+            disp = compact_disposition(runs, order, shadowed, m, cp, keep, changed, new_expire,
+                                       ksz, vsz, p);
         }
         /* wave-aggregated stats */
         unsigned long long b;
@@ -646,7 +650,8 @@ __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order, uint6
     }
 }
 
-__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order, uint64_t m,
+__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order,
+                                          const uint8_t *shadowed, uint64_t m,
                                           const CompactParams &cp, uint8_t *keep,
                                           uint8_t *changed, uint32_t *new_expire, uint64_t *ksz,
                                           uint64_t *vsz, uint64_t p)
@@ -655,26 +660,19 @@ __device__ static int compact_disposition(const DevRun *runs, const uint64_t *or
         uint64_t id = order[p];
         const DevRun &r = runs[id >> 40];
         uint64_t i = id & 0xFFFFFFFFFFull;
-        uint64_t kl, vl;
-        const uint8_t *k = run_key(r, i, &kl);
-        const uint8_t *v = run_val(r, i, &vl);
         keep[p] = 0;
         changed[p] = 0;
         ksz[p] = 0;
         vsz[p] = 0;
-        /* newest-of-key? */
-        if (p > 0) {
-            uint64_t pid = order[p - 1];
-            const DevRun &pr = runs[pid >> 40];
-            uint64_t pi = pid & 0xFFFFFFFFFFull, pkl;
-            const uint8_t *pk = run_key(pr, pi, &pkl);
-            if (dev_key_cmp(k, kl, pk, pkl) == 0) {
-                return D_SHADOWED;
-            }
+        if (shadowed[p]) {
+            return D_SHADOWED; /* newest-wins, decided in k_rank */
         }
         if (r.sk[i] & 1) {
             return D_TOMBSTONE;
         }
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        const uint8_t *v = run_val(r, i, &vl);
         /* --- Filter --- */
         int drop = 0, value_changed = 0;
         uint32_t new_ts_val = 0;
@@ -829,16 +827,17 @@ void launch_bounds(const DevRun *d_runs, int R, const uint8_t *d_key, uint64_t k
 }
 
 void launch_rank(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
-                 const uint64_t *d_wprefix, uint64_t total, uint64_t *d_order, hipStream_t s)
+                 const uint64_t *d_wprefix, uint64_t total, uint64_t *d_order,
+                 uint8_t *d_shadowed, hipStream_t s)
 {
     k_rank<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix, total,
-                                                    d_order);
+                                                    d_order, d_shadowed);
 }
 
-void launch_visible(const DevRun *d_runs, const uint64_t *d_order, uint64_t m, uint64_t *d_flags,
-                    hipStream_t s)
+void launch_visible(const DevRun *d_runs, const uint64_t *d_order, const uint8_t *d_shadowed,
+                    uint64_t m, uint64_t *d_flags, hipStream_t s)
 {
-    k_visible<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, m, d_flags);
+    k_visible<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, d_shadowed, m, d_flags);
 }
 
 void launch_gather(const uint64_t *d_order, const uint64_t *d_flags, const uint64_t *d_pos,
@@ -945,14 +944,15 @@ void launch_first_eq(const DevRun *d_runs, const uint64_t *d_view, uint64_t n,
     k_first_eq<<<1, 1, 0, s>>>(d_runs, d_view, n, d_key, klen, d_out);
 }
 
-void launch_compact_flags(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
-                          const CompactParams &cp, uint8_t *d_keep, uint8_t *d_changed,
-                          uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
-                          CompactStatsDev *d_stats, hipStream_t s)
+void launch_compact_flags(const DevRun *d_runs, const uint64_t *d_order,
+                          const uint8_t *d_shadowed, uint64_t m, const CompactParams &cp,
+                          uint8_t *d_keep, uint8_t *d_changed, uint32_t *d_new_expire,
+                          uint64_t *d_ksz, uint64_t *d_vsz, CompactStatsDev *d_stats,
+                          hipStream_t s)
 {
-    k_compact_flags<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, m, cp, d_keep,
-                                                         d_changed, d_new_expire, d_ksz, d_vsz,
-                                                         d_stats);
+    k_compact_flags<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, d_shadowed, m, cp,
+                                                         d_keep, d_changed, d_new_expire, d_ksz,
+                                                         d_vsz, d_stats);
 }
 
 void launch_emit_compact(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
