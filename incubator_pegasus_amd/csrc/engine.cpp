@@ -55,10 +55,11 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       uint8_t *, hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
-void launch_compact_flags(const DevRun *, const uint64_t *, const uint8_t *, uint64_t,
-                          const CompactParams &, uint8_t *, uint8_t *, uint32_t *, uint64_t *,
-                          uint64_t *, CompactStatsDev *, hipStream_t);
-void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint8_t *,
+void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
+                         const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
+                         uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
+                         CompactStatsDev *, hipStream_t);
+void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint64_t *,
                          const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
                          uint64_t *, uint64_t, hipStream_t);
@@ -1561,13 +1562,9 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     uint64_t *d_hi = (uint64_t *)e->upload_tmp(hi.data(), R * 8);
     uint64_t *d_wp = (uint64_t *)e->upload_tmp(wprefix.data(), (R + 1) * 8);
     uint64_t *d_order = e->talloc<uint64_t>(total * 8);
-    uint8_t *d_shadow = e->talloc<uint8_t>(total);
     hipEvent_t ev[6];
     for (auto &x : ev)
         HIP_OK(hipEventCreate(&x));
-    HIP_OK(hipEventRecord(ev[0], e->stream));
-    launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, e->stream);
-    HIP_OK(hipEventRecord(ev[1], e->stream));
 
     CompactParams cp{};
     cp.epoch_now = epoch_now;
@@ -1581,22 +1578,21 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     cp.rules = e->d_rules;
     cp.pats = e->d_pats;
 
-    uint8_t *d_keep = e->talloc<uint8_t>(total);
     uint8_t *d_changed = e->talloc<uint8_t>(total);
     uint32_t *d_new_expire = e->talloc<uint32_t>(total * 4);
     uint64_t *d_ksz = e->talloc<uint64_t>(total * 8);
     uint64_t *d_vsz = e->talloc<uint64_t>(total * 8);
-    CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
-    HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
-    launch_compact_flags(dr, d_order, d_shadow, total, cp, d_keep, d_changed, d_new_expire,
-                         d_ksz, d_vsz, d_stats, e->stream);
-    HIP_OK(hipEventRecord(ev[2], e->stream));
-    /* positions + offsets */
     uint64_t *d_keepw = e->talloc<uint64_t>(total * 8);
     uint64_t *d_kpos = e->talloc<uint64_t>(total * 8);
     uint64_t *d_koffs = e->talloc<uint64_t>(total * 8);
     uint64_t *d_voffs = e->talloc<uint64_t>(total * 8);
-    launch_widen_u8(d_keep, total, d_keepw, e->stream);
+    CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
+    HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
+    HIP_OK(hipEventRecord(ev[0], e->stream));
+    launch_rank_compact(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw, d_changed,
+                        d_new_expire, d_ksz, d_vsz, d_stats, e->stream);
+    HIP_OK(hipEventRecord(ev[1], e->stream));
+    HIP_OK(hipEventRecord(ev[2], e->stream));
     launch_psum(d_keepw, d_kpos, total, e->stream);
     launch_psum(d_ksz, d_koffs, total, e->stream);
     launch_psum(d_vsz, d_voffs, total, e->stream);
@@ -1638,9 +1634,9 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             HIP_OK(hipMalloc(&nr.sk, n_out * 8));
         }
         HIP_OK(hipEventRecord(ev[3], e->stream));
-        launch_emit_compact(dr, d_order, total, d_keep, d_changed, d_new_expire, d_kpos, d_koffs,
-                            d_voffs, e->data_version, nr.keys, nr.vals, nr.koff, nr.voff, nr.sk,
-                            n_out, e->stream);
+        launch_emit_compact(dr, d_order, total, d_keepw, d_changed, d_new_expire, d_kpos,
+                            d_koffs, d_voffs, e->data_version, nr.keys, nr.vals, nr.koff,
+                            nr.voff, nr.sk, n_out, e->stream);
         HIP_OK(hipEventRecord(ev[4], e->stream));
         HIP_OK(hipStreamSynchronize(e->stream));
         float ms;
@@ -1653,9 +1649,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         float ms;
         HIP_OK(hipEventSynchronize(ev[2]));
         HIP_OK(hipEventElapsedTime(&ms, ev[0], ev[1]));
-        e->phase_ms["compact_rank"] = ms;
-        HIP_OK(hipEventElapsedTime(&ms, ev[1], ev[2]));
-        e->phase_ms["compact_flags"] = ms;
+        e->phase_ms["compact_rank"] = ms; /* fused rank+filter */
+        e->phase_ms["compact_flags"] = 0.0;
     }
     for (auto &x : ev)
         (void)hipEventDestroy(x);
@@ -1663,8 +1658,6 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     e->tfree(d_hi);
     e->tfree(d_wp);
     e->tfree(d_order);
-    e->tfree(d_shadow);
-    e->tfree(d_keep);
     e->tfree(d_changed);
     e->tfree(d_new_expire);
     e->tfree(d_ksz);

@@ -608,29 +608,158 @@ __global__ void k_first_eq(const DevRun *runs, const uint64_t *view, uint64_t n,
  * compaction pass before this) */
 enum { D_NONE = 0, D_KEEP, D_SHADOWED, D_TOMBSTONE, D_EXPIRED, D_FILTERED };
 
-__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order,
-                                          const uint8_t *shadowed, uint64_t m,
-                                          const CompactParams &cp, uint8_t *keep,
-                                          uint8_t *changed, uint32_t *new_expire, uint64_t *ksz,
-                                          uint64_t *vsz, uint64_t p);
+/* record-level KeyWithTTLCompactionFilter::Filter restatement
+ * (key_ttl_compaction_filter.h:55-92); returns D_*, fills outputs on KEEP */
+__device__ static int dev_disposition(const DevRun &r, uint64_t i, const CompactParams &cp,
+                                      int shadow, uint8_t *changed_out, uint32_t *new_ts_out,
+                                      uint64_t *kl_out, uint64_t *vl_out)
+{
+    *changed_out = 0;
+    *new_ts_out = 0;
+    *kl_out = 0;
+    *vl_out = 0;
+    if (shadow)
+        return D_SHADOWED; /* newest-wins, decided during ranking */
+    if (r.sk[i] & 1)
+        return D_TOMBSTONE;
+    uint64_t kl, vl;
+    const uint8_t *k = run_key(r, i, &kl);
+    const uint8_t *v = run_val(r, i, &vl);
+    int drop = 0, value_changed = 0;
+    uint32_t new_ts_val = 0;
+    uint32_t expire_ts = dev_expire_ts(cp.data_version, v);
+    uint32_t eff_expire = expire_ts; /* value_view's expire after default-ttl */
+    if (kl >= 2) {
+        if (cp.default_ttl != 0 && expire_ts == 0) {
+            expire_ts = cp.epoch_now + cp.default_ttl;
+            eff_expire = expire_ts;
+            value_changed = 1;
+            new_ts_val = expire_ts;
+        }
+        if (cp.n_ops > 0) {
+            uint32_t hklen = ((uint32_t)k[0] << 8) | k[1];
+            const uint8_t *hk = k + 2;
+            const uint8_t *skp = k + 2 + hklen;
+            uint64_t sklen = kl - 2 - hklen;
+            /* value_view fixed at loop entry: rules see eff_expire (post
+             * default-ttl), not later op rewrites (filter :82-89) */
+            for (int oi = 0; oi < cp.n_ops && !drop; oi++) {
+                const DevOp &op = cp.ops[oi];
+                int all = (op.n_rules > 0);
+                for (int ri = 0; ri < op.n_rules && all; ri++) {
+                    const DevRule &rule = cp.rules[op.rule_off + ri];
+                    if (rule.type == DFR_TTL_RANGE) {
+                        uint32_t e = eff_expire;
+                        all = (e == 0 && rule.start_ttl == 0 && rule.stop_ttl == 0) ||
+                              ((uint32_t)(rule.start_ttl + cp.epoch_now) <= e &&
+                               (uint32_t)(rule.stop_ttl + cp.epoch_now) >= e);
+                    } else {
+                        all = dev_rule_match(rule, cp.pats, cp.data_version, cp.epoch_now, hk,
+                                             hklen, skp, sklen, v);
+                    }
+                }
+                if (!all)
+                    continue;
+                if (op.type == DOP_DELETE) {
+                    drop = 1;
+                    break;
+                }
+                /* update_ttl (compaction_operation.cpp:78-113) */
+                uint32_t nts = 0;
+                int apply = 1;
+                switch (op.ut_type) {
+                case DUT_FROM_NOW:
+                    nts = cp.epoch_now + op.ut_value;
+                    break;
+                case DUT_FROM_CURRENT:
+                    if (eff_expire == 0)
+                        apply = 0;
+                    else
+                        nts = op.ut_value + eff_expire;
+                    break;
+                case DUT_TIMESTAMP:
+                    nts = op.ut_value - 1451606400u;
+                    break;
+                default:
+                    apply = 0;
+                    break;
+                }
+                if (apply) {
+                    value_changed = 1;
+                    new_ts_val = nts;
+                }
+            }
+        }
+        if (drop)
+            return D_FILTERED;
+        /* final keep/drop on local expire_ts + stale split hash (:91,114-121) */
+        if (dev_ts_expired(cp.epoch_now, expire_ts))
+            return D_EXPIRED;
+        if (cp.validate_hash && cp.partition_version >= 0 && cp.pidx <= cp.partition_version &&
+            (int64_t)(dev_key_hash(k, kl) & (uint64_t)cp.partition_version) != (int64_t)cp.pidx)
+            return D_FILTERED;
+    }
+    *changed_out = (uint8_t)value_changed;
+    *new_ts_out = new_ts_val;
+    *kl_out = kl;
+    *vl_out = vl;
+    return D_KEEP;
+}
 
-__global__ void k_compact_flags(const DevRun *runs, const uint64_t *order,
-                                const uint8_t *shadowed, uint64_t m, CompactParams cp,
-                                uint8_t *keep, uint8_t *changed, uint32_t *new_expire,
-                                uint64_t *ksz, uint64_t *vsz, CompactStatsDev *stats)
+/* fused rank + newest-wins + compaction filter: one pass over the input
+ * records computes each record's merged position, shadow state and filter
+ * disposition, writing keep/changed/new-expire/sizes scattered by rank.
+ * Replaces the separate flags pass (which re-gathered every record through
+ * the order array). */
+__global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
+                               const uint64_t *wprefix, uint64_t total, CompactParams cp,
+                               uint64_t *order, uint64_t *keepw, uint8_t *changed,
+                               uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
+                               CompactStatsDev *stats)
 {
     uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
     uint64_t stride = gridDim.x * (uint64_t)blockDim.x;
-    uint64_t iters = (m + stride - 1) / stride;
+    uint64_t iters = (total + stride - 1) / stride;
     int lane = threadIdx.x % WAVE;
     for (uint64_t it = 0; it < iters; it++) {
-        uint64_t p = tid + it * stride;
+        uint64_t t = tid + it * stride;
         int disp = D_NONE;
-        if (p < m) {
-            disp = compact_disposition(runs, order, shadowed, m, cp, keep, changed, new_expire,
-                                       ksz, vsz, p);
+        if (t < total) {
+            int r = 0;
+            while (wprefix[r + 1] <= t)
+                r++;
+            uint64_t i = lo[r] + (t - wprefix[r]);
+            uint64_t kl;
+            const uint8_t *k = run_key(runs[r], i, &kl);
+            uint64_t rank = i - lo[r];
+            int shadow = 0;
+            for (int q = 0; q < R; q++) {
+                if (q == r)
+                    continue;
+                if (q > r) {
+                    uint64_t ub = dev_upper_bound(runs[q], k, kl, lo[q], hi[q]);
+                    if (!shadow && ub > lo[q]) {
+                        uint64_t pl;
+                        const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
+                        if (dev_key_cmp(pk, pl, k, kl) == 0)
+                            shadow = 1;
+                    }
+                    rank += ub - lo[q];
+                } else {
+                    rank += dev_lower_bound(runs[q], k, kl, lo[q], hi[q]) - lo[q];
+                }
+            }
+            uint8_t ch;
+            uint32_t nts;
+            uint64_t okl, ovl;
+            disp = dev_disposition(runs[r], i, cp, shadow, &ch, &nts, &okl, &ovl);
+            order[rank] = ((uint64_t)r << 40) | i;
+            keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+            changed[rank] = ch;
+            new_expire[rank] = nts;
+            ksz[rank] = okl;
+            vsz[rank] = ovl;
         }
-        /* wave-aggregated stats */
         unsigned long long b;
         b = __ballot(disp == D_SHADOWED);
         if (lane == 0 && b)
@@ -650,126 +779,10 @@ __global__ void k_compact_flags(const DevRun *runs, const uint64_t *order,
     }
 }
 
-__device__ static int compact_disposition(const DevRun *runs, const uint64_t *order,
-                                          const uint8_t *shadowed, uint64_t m,
-                                          const CompactParams &cp, uint8_t *keep,
-                                          uint8_t *changed, uint32_t *new_expire, uint64_t *ksz,
-                                          uint64_t *vsz, uint64_t p)
-{
-    {
-        uint64_t id = order[p];
-        const DevRun &r = runs[id >> 40];
-        uint64_t i = id & 0xFFFFFFFFFFull;
-        keep[p] = 0;
-        changed[p] = 0;
-        ksz[p] = 0;
-        vsz[p] = 0;
-        if (shadowed[p]) {
-            return D_SHADOWED; /* newest-wins, decided in k_rank */
-        }
-        if (r.sk[i] & 1) {
-            return D_TOMBSTONE;
-        }
-        uint64_t kl, vl;
-        const uint8_t *k = run_key(r, i, &kl);
-        const uint8_t *v = run_val(r, i, &vl);
-        /* --- Filter --- */
-        int drop = 0, value_changed = 0;
-        uint32_t new_ts_val = 0;
-        uint32_t expire_ts = dev_expire_ts(cp.data_version, v);
-        uint32_t eff_expire = expire_ts; /* value_view's expire after default-ttl */
-        if (kl >= 2) {
-            if (cp.default_ttl != 0 && expire_ts == 0) {
-                expire_ts = cp.epoch_now + cp.default_ttl;
-                eff_expire = expire_ts;
-                value_changed = 1;
-                new_ts_val = expire_ts;
-            }
-            if (cp.n_ops > 0) {
-                uint32_t hklen = ((uint32_t)k[0] << 8) | k[1];
-                const uint8_t *hk = k + 2;
-                const uint8_t *skp = k + 2 + hklen;
-                uint64_t sklen = kl - 2 - hklen;
-                /* value_view is fixed at loop entry: rules see eff_expire
-                 * (post default-ttl), not later op rewrites
-                 * (key_ttl_compaction_filter.h:82-89) */
-                for (int oi = 0; oi < cp.n_ops && !drop; oi++) {
-                    const DevOp &op = cp.ops[oi];
-                    int all = (op.n_rules > 0);
-                    for (int ri = 0; ri < op.n_rules && all; ri++) {
-                        const DevRule &rule = cp.rules[op.rule_off + ri];
-                        if (rule.type == DFR_TTL_RANGE) {
-                            uint32_t e = eff_expire;
-                            int ok = (e == 0 && rule.start_ttl == 0 && rule.stop_ttl == 0) ||
-                                     ((uint32_t)(rule.start_ttl + cp.epoch_now) <= e &&
-                                      (uint32_t)(rule.stop_ttl + cp.epoch_now) >= e);
-                            all = ok;
-                        } else {
-                            all = dev_rule_match(rule, cp.pats, cp.data_version, cp.epoch_now, hk,
-                                                 hklen, skp, sklen, v);
-                        }
-                    }
-                    if (!all)
-                        continue;
-                    if (op.type == DOP_DELETE) {
-                        drop = 1;
-                        break;
-                    }
-                    /* update_ttl (compaction_operation.cpp:78-113); current
-                     * ttl read from the value_view (eff_expire) */
-                    uint32_t nts = 0;
-                    int apply = 1;
-                    switch (op.ut_type) {
-                    case DUT_FROM_NOW:
-                        nts = cp.epoch_now + op.ut_value;
-                        break;
-                    case DUT_FROM_CURRENT:
-                        if (eff_expire == 0)
-                            apply = 0;
-                        else
-                            nts = op.ut_value + eff_expire;
-                        break;
-                    case DUT_TIMESTAMP:
-                        nts = op.ut_value - 1451606400u;
-                        break;
-                    default:
-                        apply = 0;
-                        break;
-                    }
-                    if (apply) {
-                        value_changed = 1;
-                        new_ts_val = nts;
-                    }
-                }
-            }
-            if (drop) {
-                return D_FILTERED;
-            }
-            /* final keep/drop on local expire_ts (post default-ttl, pre
-             * user-op) + stale split hash (:91,114-121) */
-            if (dev_ts_expired(cp.epoch_now, expire_ts)) {
-                return D_EXPIRED;
-            }
-            if (cp.validate_hash && cp.partition_version >= 0 &&
-                cp.pidx <= cp.partition_version &&
-                (int64_t)(dev_key_hash(k, kl) & (uint64_t)cp.partition_version) !=
-                    (int64_t)cp.pidx) {
-                return D_FILTERED;
-            }
-        }
-        keep[p] = 1;
-        changed[p] = (uint8_t)value_changed;
-        new_expire[p] = new_ts_val;
-        ksz[p] = kl;
-        vsz[p] = vl;
-        return D_KEEP;
-    }
-}
-
 /* emit the merged run (wave per record); values copied whole, expire header
  * patched in-place for changed records */
 __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64_t m,
-                               const uint8_t *keep, const uint8_t *changed,
+                               const uint64_t *keepw, const uint8_t *changed,
                                const uint32_t *new_expire, const uint64_t *kpos,
                                const uint64_t *koffs, const uint64_t *voffs, uint32_t dv,
                                uint8_t *kout, uint8_t *vout, uint64_t *okoff, uint64_t *ovoff,
@@ -779,7 +792,7 @@ __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64
     uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
     int lane = threadIdx.x % WAVE;
     for (uint64_t p = wave; p < m; p += nwaves) {
-        if (!keep[p])
+        if (!keepw[p])
             continue;
         uint64_t o = kpos[p];
         uint64_t id = order[p];
@@ -944,27 +957,28 @@ void launch_first_eq(const DevRun *d_runs, const uint64_t *d_view, uint64_t n,
     k_first_eq<<<1, 1, 0, s>>>(d_runs, d_view, n, d_key, klen, d_out);
 }
 
-void launch_compact_flags(const DevRun *d_runs, const uint64_t *d_order,
-                          const uint8_t *d_shadowed, uint64_t m, const CompactParams &cp,
-                          uint8_t *d_keep, uint8_t *d_changed, uint32_t *d_new_expire,
-                          uint64_t *d_ksz, uint64_t *d_vsz, CompactStatsDev *d_stats,
-                          hipStream_t s)
+void launch_rank_compact(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
+                         const uint64_t *d_wprefix, uint64_t total, const CompactParams &cp,
+                         uint64_t *d_order, uint64_t *d_keepw, uint8_t *d_changed,
+                         uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
+                         CompactStatsDev *d_stats, hipStream_t s)
 {
-    k_compact_flags<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_runs, d_order, d_shadowed, m, cp,
-                                                         d_keep, d_changed, d_new_expire, d_ksz,
-                                                         d_vsz, d_stats);
+    k_rank_compact<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix,
+                                                            total, cp, d_order, d_keepw,
+                                                            d_changed, d_new_expire, d_ksz,
+                                                            d_vsz, d_stats);
 }
 
 void launch_emit_compact(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
-                         const uint8_t *d_keep, const uint8_t *d_changed,
+                         const uint64_t *d_keepw, const uint8_t *d_changed,
                          const uint32_t *d_new_expire, const uint64_t *d_kpos,
                          const uint64_t *d_koffs, const uint64_t *d_voffs, uint32_t dv,
                          uint8_t *d_kout, uint8_t *d_vout, uint64_t *d_okoff, uint64_t *d_ovoff,
                          uint64_t *d_osk, uint64_t n_out, hipStream_t s)
 {
     k_emit_compact<<<grid_for(m * WAVE, BLOCK), BLOCK, 0, s>>>(
-        d_runs, d_order, m, d_keep, d_changed, d_new_expire, d_kpos, d_koffs, d_voffs, dv, d_kout,
-        d_vout, d_okoff, d_ovoff, d_osk, n_out);
+        d_runs, d_order, m, d_keepw, d_changed, d_new_expire, d_kpos, d_koffs, d_voffs, dv,
+        d_kout, d_vout, d_okoff, d_ovoff, d_osk, n_out);
 }
 
 } /* extern C++ */
