@@ -83,6 +83,7 @@ def main():
     ap.add_argument("--runs", type=int, default=8)
     ap.add_argument("--cpu-sample", type=int, default=2_000_000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--emit-mode", choices=["rank", "input"], default="input")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -118,6 +119,7 @@ def main():
     for p in range(args.partitions):
         runs = build_partition_data(keys_pp, args.runs, rank, p)
         eng = hip.open(1, p, local_rank)
+        eng.set_envs({"engine.emit_mode": args.emit_mode})
         for r in runs:
             eng.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
                                   np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
@@ -176,9 +178,9 @@ def main():
     out_bytes = int(last_stats.output_bytes)
     phases = {
         # algorithmic bytes per launch: stated in DESIGN.md §roofline
-        "compact_rank": key_bytes,                      # read every key once
-        "compact_flags": key_bytes + 12 * n0,           # neighbor keys + value headers
-        "compact_emit": 2 * out_bytes,                  # read kept records + write merged run
+        # rank is fused with the filter: every key + value header read once
+        "compact_rank": key_bytes + 12 * n0,
+        "compact_emit": 2 * out_bytes,  # read kept records + write merged run
     }
     timings = {ph: eng0.phase_ms(ph) for ph in phases}
     dominant = max(timings, key=lambda k: timings[k])

@@ -57,8 +57,13 @@ void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *
                      uint32_t *, hipStream_t);
 void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
-                         uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
+                         uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *, uint64_t *,
                          CompactStatsDev *, hipStream_t);
+void launch_emit_compact_inmajor(const DevRun *, int, const uint64_t *, uint64_t,
+                                 const uint64_t *, const uint64_t *, const uint8_t *,
+                                 const uint32_t *, const uint64_t *, const uint64_t *,
+                                 const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *,
+                                 uint64_t *, uint64_t *, uint64_t, hipStream_t);
 void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint64_t *,
                          const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
@@ -488,6 +493,9 @@ struct HipEngine {
     int64_t next_ctx_id = 0;
     std::mutex mu;
     std::unordered_map<std::string, double> phase_ms;
+    int emit_mode = 1; /* 0 = rank-major, 1 = input-major (A/B via env
+                          "engine.emit_mode"; input-major reads each run
+                          sequentially) */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
@@ -765,6 +773,8 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
             e->mg_max_iter_count = (uint32_t)atoll(v.c_str());
         } else if (k == "rocksdb.multi_get_max_iteration_size") {
             e->mg_max_iter_size = (uint64_t)atoll(v.c_str());
+        } else if (k == "engine.emit_mode") {
+            e->emit_mode = (v == "input") ? 1 : 0;
         }
     }
     return RRDB_OK;
@@ -1586,11 +1596,12 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     uint64_t *d_kpos = e->talloc<uint64_t>(total * 8);
     uint64_t *d_koffs = e->talloc<uint64_t>(total * 8);
     uint64_t *d_voffs = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_rank_of = e->emit_mode == 1 ? e->talloc<uint64_t>(total * 8) : nullptr;
     CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
     HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
     HIP_OK(hipEventRecord(ev[0], e->stream));
     launch_rank_compact(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw, d_changed,
-                        d_new_expire, d_ksz, d_vsz, d_stats, e->stream);
+                        d_new_expire, d_ksz, d_vsz, d_rank_of, d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
     HIP_OK(hipEventRecord(ev[2], e->stream));
     launch_psum(d_keepw, d_kpos, total, e->stream);
@@ -1634,9 +1645,15 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             HIP_OK(hipMalloc(&nr.sk, n_out * 8));
         }
         HIP_OK(hipEventRecord(ev[3], e->stream));
-        launch_emit_compact(dr, d_order, total, d_keepw, d_changed, d_new_expire, d_kpos,
-                            d_koffs, d_voffs, e->data_version, nr.keys, nr.vals, nr.koff,
-                            nr.voff, nr.sk, n_out, e->stream);
+        if (e->emit_mode == 1)
+            launch_emit_compact_inmajor(dr, R, d_wp, total, d_rank_of, d_keepw, d_changed,
+                                        d_new_expire, d_kpos, d_koffs, d_voffs,
+                                        e->data_version, nr.keys, nr.vals, nr.koff, nr.voff,
+                                        nr.sk, n_out, e->stream);
+        else
+            launch_emit_compact(dr, d_order, total, d_keepw, d_changed, d_new_expire, d_kpos,
+                                d_koffs, d_voffs, e->data_version, nr.keys, nr.vals, nr.koff,
+                                nr.voff, nr.sk, n_out, e->stream);
         HIP_OK(hipEventRecord(ev[4], e->stream));
         HIP_OK(hipStreamSynchronize(e->stream));
         float ms;
@@ -1658,6 +1675,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     e->tfree(d_hi);
     e->tfree(d_wp);
     e->tfree(d_order);
+    if (d_rank_of)
+        e->tfree(d_rank_of);
     e->tfree(d_changed);
     e->tfree(d_new_expire);
     e->tfree(d_ksz);

@@ -91,6 +91,28 @@ __device__ static inline int dev_key_cmp(const uint8_t *a, uint64_t alen, const 
     return alen < blen ? -1 : (alen > blen ? 1 : 0);
 }
 
+/* per-wave contiguous work chunks: wave w handles [w*chunk, w*chunk+chunk).
+ * Consecutive ranks stay within one wave, so the 8 interleaved run streams
+ * advance sequentially per wave (L1-local shared lines) and output writes
+ * are sequential — the strided mapping spread rank-adjacent records across
+ * XCDs and cost ~4.5x fetch amplification (PMC, r01 profiles). */
+__device__ static inline void wave_chunk(uint64_t total, uint64_t *wstart, uint64_t *wend,
+                                         int *lane_out)
+{
+    uint64_t nw = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    uint64_t wid = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t chunk = (total + nw - 1) / nw;
+    uint64_t s = wid * chunk;
+    uint64_t e = s + chunk;
+    if (s > total)
+        s = total;
+    if (e > total)
+        e = total;
+    *wstart = s;
+    *wend = e;
+    *lane_out = threadIdx.x % WAVE;
+}
+
 /* lane-strided record copy: 4B unaligned chunks (consecutive lanes touch
  * consecutive addresses -> coalesced), byte tail */
 __device__ static inline void wave_copy(uint8_t *dst, const uint8_t *src, uint64_t n, int lane)
@@ -258,8 +280,10 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
                        const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order,
                        uint8_t *shadowed /* [total], by rank position */)
 {
-    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
-         t += gridDim.x * (uint64_t)blockDim.x) {
+    uint64_t ws, we;
+    int lane;
+    wave_chunk(total, &ws, &we, &lane);
+    for (uint64_t t = ws + lane; t < we; t += WAVE) {
         int r = 0;
         while (wprefix[r + 1] <= t)
             r++;
@@ -522,11 +546,11 @@ __global__ void k_emit_scan(const DevRun *runs, const uint64_t *view, uint64_t w
                             uint8_t *kout, uint8_t *vout, uint64_t *kout_offs,
                             uint64_t *vout_offs, int32_t *ets_out, uint64_t n_out)
 {
-    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
-    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
-    int lane = threadIdx.x % WAVE;
+    uint64_t ws, we;
+    int lane;
+    wave_chunk(consumed, &ws, &we, &lane);
     uint32_t hdr = dev_hdr_len(sp.data_version);
-    for (uint64_t p = wave; p < consumed; p += nwaves) {
+    for (uint64_t p = ws; p < we; p++) {
         if (state[p] != ST_NORMAL)
             continue;
         uint64_t o = npos[p]; /* output row index */
@@ -715,16 +739,17 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                                const uint64_t *wprefix, uint64_t total, CompactParams cp,
                                uint64_t *order, uint64_t *keepw, uint8_t *changed,
                                uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
+                               uint64_t *rank_of /* [total] by input index, may be null */,
                                CompactStatsDev *stats)
 {
-    uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
-    uint64_t stride = gridDim.x * (uint64_t)blockDim.x;
-    uint64_t iters = (total + stride - 1) / stride;
-    int lane = threadIdx.x % WAVE;
+    uint64_t ws, we;
+    int lane;
+    wave_chunk(total, &ws, &we, &lane);
+    uint64_t iters = (we - ws + WAVE - 1) / WAVE;
     for (uint64_t it = 0; it < iters; it++) {
-        uint64_t t = tid + it * stride;
+        uint64_t t = ws + it * WAVE + lane;
         int disp = D_NONE;
-        if (t < total) {
+        if (t < we) {
             int r = 0;
             while (wprefix[r + 1] <= t)
                 r++;
@@ -759,6 +784,8 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
             new_expire[rank] = nts;
             ksz[rank] = okl;
             vsz[rank] = ovl;
+            if (rank_of)
+                rank_of[t] = rank;
         }
         unsigned long long b;
         b = __ballot(disp == D_SHADOWED);
@@ -788,10 +815,10 @@ __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64
                                uint8_t *kout, uint8_t *vout, uint64_t *okoff, uint64_t *ovoff,
                                uint64_t *osk, uint64_t n_out)
 {
-    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
-    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
-    int lane = threadIdx.x % WAVE;
-    for (uint64_t p = wave; p < m; p += nwaves) {
+    uint64_t ws, we;
+    int lane;
+    wave_chunk(m, &ws, &we, &lane);
+    for (uint64_t p = ws; p < we; p++) {
         if (!keepw[p])
             continue;
         uint64_t o = kpos[p];
@@ -820,6 +847,57 @@ __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64
             if (o == n_out - 1) {
                 okoff[n_out] = koffs[p] + kl;
                 ovoff[n_out] = voffs[p] + vl;
+            }
+        }
+    }
+}
+
+/* input-major emit: waves sweep each run's records in storage order
+ * (sequential coalesced reads); the write side follows the monotone output
+ * offsets of the kept records.  A/B alternative to the rank-major emit. */
+__global__ void k_emit_compact_inmajor(const DevRun *runs, int R, const uint64_t *wprefix,
+                                       uint64_t total, const uint64_t *rank_of,
+                                       const uint64_t *keepw, const uint8_t *changed,
+                                       const uint32_t *new_expire, const uint64_t *kpos,
+                                       const uint64_t *koffs, const uint64_t *voffs, uint32_t dv,
+                                       uint8_t *kout, uint8_t *vout, uint64_t *okoff,
+                                       uint64_t *ovoff, uint64_t *osk, uint64_t n_out)
+{
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    for (uint64_t t = wave; t < total; t += nwaves) {
+        uint64_t rank = rank_of[t];
+        if (!keepw[rank])
+            continue;
+        int r = 0;
+        while (wprefix[r + 1] <= t)
+            r++;
+        uint64_t i = t - wprefix[r];
+        const DevRun &run = runs[r];
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(run, i, &kl);
+        const uint8_t *v = run_val(run, i, &vl);
+        uint64_t o = kpos[rank];
+        wave_copy(kout + koffs[rank], k, kl, lane);
+        uint8_t *vdst = vout + voffs[rank];
+        wave_copy(vdst, v, vl, lane);
+        __builtin_amdgcn_wave_barrier();
+        if (lane == 0) {
+            if (changed[rank]) {
+                uint32_t ts = new_expire[rank];
+                uint32_t off = (dv == 2) ? 1 : 0;
+                vdst[off] = (uint8_t)(ts >> 24);
+                vdst[off + 1] = (uint8_t)(ts >> 16);
+                vdst[off + 2] = (uint8_t)(ts >> 8);
+                vdst[off + 3] = (uint8_t)ts;
+            }
+            okoff[o] = koffs[rank];
+            ovoff[o] = voffs[rank];
+            osk[o] = run.sk[i];
+            if (o == n_out - 1) {
+                okoff[n_out] = koffs[rank] + kl;
+                ovoff[n_out] = voffs[rank] + vl;
             }
         }
     }
@@ -961,12 +1039,26 @@ void launch_rank_compact(const DevRun *d_runs, int R, const uint64_t *d_lo, cons
                          const uint64_t *d_wprefix, uint64_t total, const CompactParams &cp,
                          uint64_t *d_order, uint64_t *d_keepw, uint8_t *d_changed,
                          uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
-                         CompactStatsDev *d_stats, hipStream_t s)
+                         uint64_t *d_rank_of, CompactStatsDev *d_stats, hipStream_t s)
 {
     k_rank_compact<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix,
                                                             total, cp, d_order, d_keepw,
                                                             d_changed, d_new_expire, d_ksz,
-                                                            d_vsz, d_stats);
+                                                            d_vsz, d_rank_of, d_stats);
+}
+
+void launch_emit_compact_inmajor(const DevRun *d_runs, int R, const uint64_t *d_wprefix,
+                                 uint64_t total, const uint64_t *d_rank_of,
+                                 const uint64_t *d_keepw, const uint8_t *d_changed,
+                                 const uint32_t *d_new_expire, const uint64_t *d_kpos,
+                                 const uint64_t *d_koffs, const uint64_t *d_voffs, uint32_t dv,
+                                 uint8_t *d_kout, uint8_t *d_vout, uint64_t *d_okoff,
+                                 uint64_t *d_ovoff, uint64_t *d_osk, uint64_t n_out,
+                                 hipStream_t s)
+{
+    k_emit_compact_inmajor<<<grid_for(total * WAVE, BLOCK), BLOCK, 0, s>>>(
+        d_runs, R, d_wprefix, total, d_rank_of, d_keepw, d_changed, d_new_expire, d_kpos,
+        d_koffs, d_voffs, dv, d_kout, d_vout, d_okoff, d_ovoff, d_osk, n_out);
 }
 
 void launch_emit_compact(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
