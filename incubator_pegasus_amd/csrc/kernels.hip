@@ -863,10 +863,10 @@ __global__ void k_emit_compact_inmajor(const DevRun *runs, int R, const uint64_t
                                        uint8_t *kout, uint8_t *vout, uint64_t *okoff,
                                        uint64_t *ovoff, uint64_t *osk, uint64_t n_out)
 {
-    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
-    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
-    int lane = threadIdx.x % WAVE;
-    for (uint64_t t = wave; t < total; t += nwaves) {
+    uint64_t ws, we;
+    int lane;
+    wave_chunk(total, &ws, &we, &lane);
+    for (uint64_t t = ws; t < we; t++) {
         uint64_t rank = rank_of[t];
         if (!keepw[rank])
             continue;
