@@ -83,7 +83,7 @@ def main():
     ap.add_argument("--runs", type=int, default=8)
     ap.add_argument("--cpu-sample", type=int, default=2_000_000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--emit-mode", choices=["rank", "input"], default="input")
+    ap.add_argument("--emit-mode", choices=["rank", "input"], default="rank")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -168,6 +168,20 @@ def main():
     keys_per_step_job = total_records * max(world, 1)  # whole-job aggregate
     value = keys_per_step_job * args.steps / elapsed
     ms_per_step = elapsed * 1000.0 / args.steps
+
+    # one instrumented (untimed) step: per-call wall vs in-kernel time
+    call_ms = []
+    for eng in parts:
+        c0 = time.time()
+        eng.manual_compact(epoch_now, keep_inputs=True)
+        call_ms.append((time.time() - c0) * 1e3)
+    call_ms.sort()
+    compact_call_stats = {
+        "min": round(call_ms[0], 2),
+        "med": round(call_ms[len(call_ms) // 2], 2),
+        "max": round(call_ms[-1], 2),
+        "kernel_total_ms": round(parts[-1].phase_ms("compact_total"), 2),
+    }
 
     # ---- roofline: dominant compaction kernel, HIP-event timed in-engine ----
     eng0 = parts[0]
@@ -318,6 +332,7 @@ def main():
                 "get_ops_per_s": round(get_ops_per_s, 1),
                 "get_found": found,
                 "compact_output_records_per_gpu": int(out_records),
+                "compact_call_ms": compact_call_stats,
             },
         }
         print(json.dumps(line))

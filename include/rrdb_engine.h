@@ -193,6 +193,24 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
                         const uint8_t *values, const uint64_t *val_offs,
                         const uint64_t *seq_kind, uint64_t n_records);
 
+/* ---- write path ----
+ * Host-side memtable feeding sorted runs, mirroring the committed-mutation
+ * apply path: pegasus_server_impl::on_batched_write_requests ->
+ * pegasus_write_service put/remove (reference pegasus_write_service.h:119-207)
+ * -> rocksdb_wrapper::write_batch_put/_delete (rocksdb_wrapper.cpp:121-247):
+ * key = pegasus_generate_key(hash_key, sort_key); value encoded with
+ * expire_ts (0 = no TTL); remove writes a tombstone.  Within the memtable the
+ * newest write to a key wins (memtable upsert).  rrdb_flush turns the
+ * memtable into a sorted run (the memtable-flush -> L0 SST step); reads
+ * flush lazily so committed writes are immediately visible, as they are
+ * through rocksdb's memtable read path. */
+int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
+                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts);
+int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
+                    uint64_t sklen);
+uint64_t rrdb_memtable_entries(void *h);
+int32_t rrdb_flush(void *h); /* no-op when the memtable is empty */
+
 /* ---- read service (pegasus_read_service.h:54-68 semantics) ---- */
 
 /* on_get (pegasus_server_impl.cpp:418-494): newest version of `key`;

@@ -14,6 +14,7 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <map>
 #include <cstdint>
 #include <cstring>
 #include <memory>
@@ -34,7 +35,8 @@ void launch_visible(const DevRun *, const uint64_t *, const uint8_t *, uint64_t,
                     hipStream_t);
 void launch_gather(const uint64_t *, const uint64_t *, const uint64_t *, uint64_t, uint64_t *,
                    hipStream_t);
-void launch_psum(const uint64_t *, uint64_t *, uint64_t, hipStream_t);
+void launch_psum(const uint64_t *, uint64_t *, uint64_t, uint64_t *, hipStream_t);
+uint64_t psum_scratch_elems(uint64_t);
 void launch_get(const DevRun *, int, const uint8_t *, const uint64_t *, uint64_t, uint32_t,
                 uint32_t, int32_t *, uint64_t *, uint64_t *, uint32_t *, hipStream_t);
 void launch_emit_values(const DevRun *, const uint64_t *, const int32_t *, uint64_t, uint32_t,
@@ -489,13 +491,15 @@ struct HipEngine {
     DevRule *d_rules = nullptr;
     uint8_t *d_pats = nullptr;
     int n_ops = 0;
+    /* write path (§8(f)1): host memtable, newest write per key wins */
+    std::map<std::string, std::tuple<std::string /*encoded value*/, uint64_t /*seq*/,
+                                     int /*kind*/>> memtable;
     std::unordered_map<int64_t, HipScanCtx *> ctxs;
     int64_t next_ctx_id = 0;
     std::mutex mu;
     std::unordered_map<std::string, double> phase_ms;
-    int emit_mode = 1; /* 0 = rank-major, 1 = input-major (A/B via env
-                          "engine.emit_mode"; input-major reads each run
-                          sequentially) */
+    int emit_mode = 0; /* 0 = rank-major (default; won the on-GPU A/B),
+                          1 = input-major (env "engine.emit_mode") */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
@@ -578,18 +582,54 @@ struct HipEngine {
         return d;
     }
 
-    /* stream-ordered (pooled) temporaries — avoids a plain hipMalloc/hipFree
-     * round trip per operation; pool release threshold raised at open */
+    /* engine-owned persistent scratch arena for per-operation temporaries.
+     * Bump-allocated; reset at the start of each C-ABI operation (no live
+     * scratch crosses operations — persistent data uses plain hipMalloc).
+     * Grows to the high-water mark once and then never allocates again —
+     * per-op hipMalloc/hipFree (and mempool behavior variance across
+     * driver configs) was a large, box-dependent share of step time. */
+    std::vector<std::pair<uint8_t *, size_t>> sblocks;
+    size_t s_off = 0;
+    void scratch_reset()
+    {
+        if (sblocks.size() > 1) {
+            /* consolidate: one block at the high-water total */
+            HIP_OK(hipStreamSynchronize(stream));
+            size_t tot = 0;
+            for (auto &b : sblocks)
+                tot += b.second;
+            for (auto &b : sblocks)
+                (void)hipFree(b.first);
+            sblocks.clear();
+            uint8_t *pb = nullptr;
+            HIP_OK(hipMalloc(&pb, tot));
+            sblocks.push_back({pb, tot});
+        }
+        s_off = 0;
+    }
     template <typename T> T *talloc(uint64_t n_bytes)
     {
-        void *d = nullptr;
-        HIP_OK(hipMallocAsync(&d, n_bytes ? n_bytes : 1, stream));
-        return (T *)d;
+        uint64_t n = (n_bytes + 255) & ~255ull;
+        if (n == 0)
+            n = 256;
+        if (sblocks.empty() || s_off + n > sblocks.back().second) {
+            size_t tot = 0;
+            for (auto &b : sblocks)
+                tot += b.second;
+            size_t want = std::max<size_t>({(size_t)n, tot, (size_t)(256ull << 20)});
+            uint8_t *pb = nullptr;
+            HIP_OK(hipMalloc(&pb, want));
+            sblocks.push_back({pb, want});
+            s_off = 0;
+        }
+        T *out = (T *)(sblocks.back().first + s_off);
+        s_off += n;
+        return out;
     }
-    void tfree(void *p)
+    void tfree(void *) {} /* arena memory is reclaimed at scratch_reset */
+    uint64_t *psum_scratch(uint64_t n)
     {
-        if (p)
-            HIP_OK(hipFreeAsync(p, stream));
+        return talloc<uint64_t>(psum_scratch_elems(n) * 8);
     }
     uint8_t *upload_tmp(const void *p, uint64_t n)
     {
@@ -644,7 +684,7 @@ struct HipEngine {
         uint64_t *d_pos = talloc<uint64_t>(total * 8);
         launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, stream);
         launch_visible(dr, d_order, d_shadow, total, d_flags, stream);
-        launch_psum(d_flags, d_pos, total, stream);
+        launch_psum(d_flags, d_pos, total, psum_scratch(total), stream);
         uint64_t lastp = 0, lastf = 0;
         HIP_OK(hipMemcpyAsync(&lastp, d_pos + total - 1, 8, hipMemcpyDeviceToHost, stream));
         HIP_OK(hipMemcpyAsync(&lastf, d_flags + total - 1, 8, hipMemcpyDeviceToHost, stream));
@@ -672,6 +712,22 @@ struct HipEngine {
 std::once_flag g_init_once;
 
 } // namespace
+
+/* internal: upload a prepared sorted run (validation already established) */
+static void ingest_prepared(HipEngine *e, const std::string &keys,
+                            const std::vector<uint64_t> &koff, const std::string &vals,
+                            const std::vector<uint64_t> &voff, const std::vector<uint64_t> &sk)
+{
+    RunBuf r;
+    r.n = sk.size();
+    r.keys = e->upload_bytes(keys.data(), keys.size());
+    r.koff = (uint64_t *)e->upload_bytes(koff.data(), koff.size() * 8);
+    r.vals = e->upload_bytes(vals.data(), vals.size());
+    r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size() * 8);
+    r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
+    e->runs.push_back(r);
+    e->d_runs_dirty = true;
+}
 
 extern "C" {
 
@@ -728,6 +784,8 @@ void rrdb_close(void *h)
         e->free_run(r);
     if (e->d_runs)
         (void)hipFree(e->d_runs);
+    for (auto &b : e->sblocks)
+        (void)hipFree(b.first);
     if (e->d_ops)
         (void)hipFree(e->d_ops);
     if (e->d_rules)
@@ -785,6 +843,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
                         uint64_t n)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* buffered writes are older than this run */
     if (n == 0)
         return RRDB_OK;
     /* validate on host (data arrives host-side anyway) */
@@ -841,6 +900,7 @@ static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const ui
                         std::vector<std::pair<const uint8_t *, uint64_t>> &vals, Arena *a)
 {
     e->activate();
+    e->scratch_reset();
     int R = (int)e->runs.size();
     status.assign(nq, RRDB_NOT_FOUND);
     vals.assign(nq, {nullptr, 0});
@@ -855,7 +915,7 @@ static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const ui
     uint64_t *d_voffs = e->talloc<uint64_t>((nq + 1) * 8);
     launch_get(dr, R, d_keys, d_offs, nq, epoch_now, e->data_version, d_status, d_hit, d_ulen,
                nullptr, e->stream);
-    launch_psum(d_ulen, d_voffs, nq, e->stream);
+    launch_psum(d_ulen, d_voffs, nq, e->psum_scratch(nq), e->stream);
     uint64_t last_off = 0, last_len = 0;
     HIP_OK(hipMemcpyAsync(&last_off, d_voffs + nq - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&last_len, d_ulen + nq - 1, 8, hipMemcpyDeviceToHost, e->stream));
@@ -889,6 +949,7 @@ int32_t rrdb_get(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
                  rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
     Arena *a = result_init(out);
     uint64_t offs[2] = {0, key_len};
     std::vector<int32_t> status;
@@ -911,6 +972,7 @@ int32_t rrdb_batch_get(void *h, uint64_t n_keys, const uint8_t *keys, const uint
                        uint32_t epoch_now, rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
     Arena *a = result_init(out);
     if (n_keys == 0) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_batch_get:922-928 */
@@ -941,6 +1003,8 @@ int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
                  rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
+    ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
     int R = (int)e->runs.size();
@@ -1029,7 +1093,7 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
     launch_scan_state(dr, d_win, w, sp, d_state, d_ksz, d_vsz, e->stream);
     HIP_OK(hipEventRecord(sev[1], e->stream));
     launch_normal_flags(d_state, w, d_flags, e->stream);
-    launch_psum(d_flags, d_npos, w, e->stream);
+    launch_psum(d_flags, d_npos, w, e->psum_scratch(w), e->stream);
     launch_cutoff(d_npos, d_flags, w, batch_count, d_cut, e->stream);
     uint64_t cut[2];
     HIP_OK(hipMemcpyAsync(cut, d_cut, 16, hipMemcpyDeviceToHost, e->stream));
@@ -1045,8 +1109,8 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
         launch_cut_sizes(d_state, w, consumed, d_ksz, d_vsz, e->stream);
         uint64_t *d_koffs = e->talloc<uint64_t>(w * 8);
         uint64_t *d_voffs = e->talloc<uint64_t>(w * 8);
-        launch_psum(d_ksz, d_koffs, w, e->stream);
-        launch_psum(d_vsz, d_voffs, w, e->stream);
+        launch_psum(d_ksz, d_koffs, w, e->psum_scratch(w), e->stream);
+        launch_psum(d_vsz, d_voffs, w, e->psum_scratch(w), e->stream);
         uint64_t t[4] = {0, 0, 0, 0};
         HIP_OK(hipMemcpyAsync(&t[0], d_koffs + w - 1, 8, hipMemcpyDeviceToHost, e->stream));
         HIP_OK(hipMemcpyAsync(&t[1], d_ksz + w - 1, 8, hipMemcpyDeviceToHost, e->stream));
@@ -1142,6 +1206,8 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
 int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
+    ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
         q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
@@ -1216,6 +1282,7 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
     auto *e = (HipEngine *)h;
     Arena *a = result_init(out);
     e->activate();
+    e->scratch_reset();
     HipScanCtx *c = nullptr;
     {
         std::lock_guard<std::mutex> g(e->mu);
@@ -1257,6 +1324,8 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
                            rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
+    ((HipEngine *)h)->scratch_reset();
     result_init(out);
     e->activate();
     /* start=(hk,""), stop=next(hk) (on_sortkey_count:1030-1036); count all
@@ -1281,7 +1350,7 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
         uint64_t *d_npos = e->talloc<uint64_t>(n * 8);
         launch_scan_state(dr, d_view, n, sp, d_state, d_ksz, d_vsz, e->stream);
         launch_normal_flags(d_state, n, d_flags, e->stream);
-        launch_psum(d_flags, d_npos, n, e->stream);
+        launch_psum(d_flags, d_npos, n, e->psum_scratch(n), e->stream);
         uint64_t lastp = 0, lastf = 0;
         HIP_OK(hipMemcpyAsync(&lastp, d_npos + n - 1, 8, hipMemcpyDeviceToHost, e->stream));
         HIP_OK(hipMemcpyAsync(&lastf, d_flags + n - 1, 8, hipMemcpyDeviceToHost, e->stream));
@@ -1304,6 +1373,8 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
                        rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
+    ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
     if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
@@ -1541,6 +1612,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
                             rrdb_compact_stats *stats)
 {
     auto *e = (HipEngine *)h;
+    rrdb_flush(h); /* memtable visible to reads */
+    ((HipEngine *)h)->scratch_reset();
     rrdb_compact_stats st{};
     if (e->manual_compact_disabled) {
         if (stats)
@@ -1604,9 +1677,9 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
                         d_new_expire, d_ksz, d_vsz, d_rank_of, d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
     HIP_OK(hipEventRecord(ev[2], e->stream));
-    launch_psum(d_keepw, d_kpos, total, e->stream);
-    launch_psum(d_ksz, d_koffs, total, e->stream);
-    launch_psum(d_vsz, d_voffs, total, e->stream);
+    launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);
+    launch_psum(d_ksz, d_koffs, total, e->psum_scratch(total), e->stream);
+    launch_psum(d_vsz, d_voffs, total, e->psum_scratch(total), e->stream);
     uint64_t t[6] = {0};
     HIP_OK(hipMemcpyAsync(&t[0], d_kpos + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[1], d_keepw + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
@@ -1705,6 +1778,67 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     }
     if (stats)
         *stats = st;
+    return RRDB_OK;
+}
+
+} /* extern "C" */
+
+/* ================= write path (§8(f)1) =================
+ * Host memtable + flush -> sorted run; mirrors pegasus_write_service
+ * put/remove + rocksdb_wrapper::write_batch_put/_delete (reference
+ * pegasus_write_service.h:119-207, rocksdb_wrapper.cpp:121-247).  Reads
+ * flush lazily so committed writes are immediately visible (rocksdb
+ * memtable read-path equivalent). */
+extern "C" {
+
+int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
+                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts)
+{
+    auto *e = (HipEngine *)h;
+    std::string key = make_key(hash_key, hklen, sort_key, sklen);
+    uint32_t hdr = hdr_len(e->data_version);
+    std::string val(hdr + vlen, '\0');
+    uint32_t off = (e->data_version == 2) ? 1 : 0;
+    if (e->data_version == 2)
+        val[0] = (char)0x82;
+    val[off] = (char)(expire_ts >> 24);
+    val[off + 1] = (char)(expire_ts >> 16);
+    val[off + 2] = (char)(expire_ts >> 8);
+    val[off + 3] = (char)expire_ts;
+    if (vlen)
+        memcpy(&val[hdr], value, vlen);
+    e->memtable[key] = {std::move(val), e->next_seq_floor++, RRDB_KIND_PUT};
+    return RRDB_OK;
+}
+
+int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
+                    uint64_t sklen)
+{
+    auto *e = (HipEngine *)h;
+    std::string key = make_key(hash_key, hklen, sort_key, sklen);
+    e->memtable[key] = {std::string(), e->next_seq_floor++, RRDB_KIND_DELETE};
+    return RRDB_OK;
+}
+
+uint64_t rrdb_memtable_entries(void *h) { return ((HipEngine *)h)->memtable.size(); }
+
+int32_t rrdb_flush(void *h)
+{
+    auto *e = (HipEngine *)h;
+    if (e->memtable.empty())
+        return RRDB_OK;
+    e->activate();
+    std::string keys, vals;
+    std::vector<uint64_t> koff{0}, voff{0}, sk;
+    for (auto &kv : e->memtable) {
+        keys += kv.first;
+        koff.push_back(keys.size());
+        vals += std::get<0>(kv.second);
+        voff.push_back(vals.size());
+        sk.push_back((std::get<1>(kv.second) << 1) | (uint64_t)std::get<2>(kv.second));
+    }
+    ingest_prepared(e, keys, koff, vals, voff, sk);
+    e->memtable.clear();
     return RRDB_OK;
 }
 

@@ -91,11 +91,10 @@ __device__ static inline int dev_key_cmp(const uint8_t *a, uint64_t alen, const 
     return alen < blen ? -1 : (alen > blen ? 1 : 0);
 }
 
-/* per-wave contiguous work chunks: wave w handles [w*chunk, w*chunk+chunk).
- * Consecutive ranks stay within one wave, so the 8 interleaved run streams
- * advance sequentially per wave (L1-local shared lines) and output writes
- * are sequential — the strided mapping spread rank-adjacent records across
- * XCDs and cost ~4.5x fetch amplification (PMC, r01 profiles). */
+/* per-wave contiguous work chunks.  Measured on gfx950 (r01 PMC): chunked
+ * mappings RAISED fetch bytes for the search kernels (168 vs 125 MB/launch
+ * rank_compact) and slowed the rank-major emit; strided grid mappings are
+ * the default and this helper is used only by the input-major emit A/B. */
 __device__ static inline void wave_chunk(uint64_t total, uint64_t *wstart, uint64_t *wend,
                                          int *lane_out)
 {
@@ -280,10 +279,8 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
                        const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order,
                        uint8_t *shadowed /* [total], by rank position */)
 {
-    uint64_t ws, we;
-    int lane;
-    wave_chunk(total, &ws, &we, &lane);
-    for (uint64_t t = ws + lane; t < we; t += WAVE) {
+    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
+         t += gridDim.x * (uint64_t)blockDim.x) {
         int r = 0;
         while (wprefix[r + 1] <= t)
             r++;
@@ -546,11 +543,11 @@ __global__ void k_emit_scan(const DevRun *runs, const uint64_t *view, uint64_t w
                             uint8_t *kout, uint8_t *vout, uint64_t *kout_offs,
                             uint64_t *vout_offs, int32_t *ets_out, uint64_t n_out)
 {
-    uint64_t ws, we;
-    int lane;
-    wave_chunk(consumed, &ws, &we, &lane);
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
     uint32_t hdr = dev_hdr_len(sp.data_version);
-    for (uint64_t p = ws; p < we; p++) {
+    for (uint64_t p = wave; p < consumed; p += nwaves) {
         if (state[p] != ST_NORMAL)
             continue;
         uint64_t o = npos[p]; /* output row index */
@@ -742,14 +739,14 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                                uint64_t *rank_of /* [total] by input index, may be null */,
                                CompactStatsDev *stats)
 {
-    uint64_t ws, we;
-    int lane;
-    wave_chunk(total, &ws, &we, &lane);
-    uint64_t iters = (we - ws + WAVE - 1) / WAVE;
+    uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+    uint64_t stride = gridDim.x * (uint64_t)blockDim.x;
+    uint64_t iters = (total + stride - 1) / stride;
+    int lane = threadIdx.x % WAVE;
     for (uint64_t it = 0; it < iters; it++) {
-        uint64_t t = ws + it * WAVE + lane;
+        uint64_t t = tid + it * stride;
         int disp = D_NONE;
-        if (t < we) {
+        if (t < total) {
             int r = 0;
             while (wprefix[r + 1] <= t)
                 r++;
@@ -815,10 +812,10 @@ __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64
                                uint8_t *kout, uint8_t *vout, uint64_t *okoff, uint64_t *ovoff,
                                uint64_t *osk, uint64_t n_out)
 {
-    uint64_t ws, we;
-    int lane;
-    wave_chunk(m, &ws, &we, &lane);
-    for (uint64_t p = ws; p < we; p++) {
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    for (uint64_t p = wave; p < m; p += nwaves) {
         if (!keepw[p])
             continue;
         uint64_t o = kpos[p];
@@ -937,30 +934,38 @@ void launch_gather(const uint64_t *d_order, const uint64_t *d_flags, const uint6
     k_gather<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_order, d_flags, d_pos, m, d_out);
 }
 
-/* exclusive scan; returns total via d_total (device u64, may be null) */
-void launch_psum(const uint64_t *d_in, uint64_t *d_out, uint64_t n, hipStream_t s);
+/* exclusive scan; caller provides scratch of psum_scratch_elems(n) u64s
+ * (engine-owned arena — no per-call allocation) */
+uint64_t psum_scratch_elems(uint64_t n)
+{
+    uint64_t tot = 0;
+    while (n > 1) {
+        uint64_t nb = (n + PSUM_BLOCK_ITEMS - 1) / PSUM_BLOCK_ITEMS;
+        tot += 2 * nb;
+        n = nb;
+    }
+    return tot + 4;
+}
 
-static void psum_rec(const uint64_t *d_in, uint64_t *d_out, uint64_t n, hipStream_t s)
+static void psum_rec(const uint64_t *d_in, uint64_t *d_out, uint64_t n, uint64_t *scratch,
+                     hipStream_t s)
 {
     uint64_t nb = (n + PSUM_BLOCK_ITEMS - 1) / PSUM_BLOCK_ITEMS;
     if (nb == 0)
         nb = 1;
-    uint64_t *d_bs = nullptr;
-    HIP_CHECK(hipMallocAsync(&d_bs, (nb + 1) * sizeof(uint64_t), s));
+    uint64_t *d_bs = scratch;
     k_psum1<<<(uint32_t)nb, BLOCK, 0, s>>>(d_in, d_out, d_bs, n);
     if (nb > 1) {
-        uint64_t *d_bo = nullptr;
-        HIP_CHECK(hipMallocAsync(&d_bo, (nb + 1) * sizeof(uint64_t), s));
-        psum_rec(d_bs, d_bo, nb, s);
+        uint64_t *d_bo = scratch + nb;
+        psum_rec(d_bs, d_bo, nb, scratch + 2 * nb, s);
         k_psum_add<<<(uint32_t)nb, BLOCK, 0, s>>>(d_out, d_bo, n);
-        HIP_CHECK(hipFreeAsync(d_bo, s));
     }
-    HIP_CHECK(hipFreeAsync(d_bs, s));
 }
 
-void launch_psum(const uint64_t *d_in, uint64_t *d_out, uint64_t n, hipStream_t s)
+void launch_psum(const uint64_t *d_in, uint64_t *d_out, uint64_t n, uint64_t *d_scratch,
+                 hipStream_t s)
 {
-    psum_rec(d_in, d_out, n, s);
+    psum_rec(d_in, d_out, n, d_scratch, s);
 }
 
 void launch_get(const DevRun *d_runs, int R, const uint8_t *d_qkeys, const uint64_t *d_qoffs,

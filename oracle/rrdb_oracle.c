@@ -757,6 +757,8 @@ static int parse_user_ops(orc_ops *out, const char *json, uint64_t len)
     return 1;
 }
 
+int32_t rrdb_flush(void *h); /* fwd (write path, defined below) */
+
 /* ================= engine state ================= */
 typedef struct {
     uint64_t n;
@@ -784,6 +786,16 @@ typedef struct {
     Run *runs;
     int n_runs, runs_cap;
     uint64_t next_seq_floor; /* ingest seqno monotonicity check */
+    /* write path (§8(f)1): buffered writes, flushed into a sorted run */
+    struct mem_entry {
+        uint8_t *key;
+        uint64_t klen;
+        uint8_t *val; /* encoded value (kind PUT) */
+        uint64_t vlen;
+        uint64_t seq;
+        int kind;
+    } *mem;
+    uint64_t mem_n, mem_cap;
     ScanCtx **ctxs;
     int n_ctxs, ctxs_cap;
     int64_t next_ctx_id;
@@ -850,6 +862,11 @@ void rrdb_close(void *h)
     for (int i = 0; i < e->n_runs; i++)
         free_run(&e->runs[i]);
     free(e->runs);
+    for (uint64_t i = 0; i < e->mem_n; i++) {
+        free(e->mem[i].key);
+        free(e->mem[i].val);
+    }
+    free(e->mem);
     for (int i = 0; i < e->n_ctxs; i++)
         free_ctx(e->ctxs[i]);
     free(e->ctxs);
@@ -924,6 +941,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
                         uint64_t n)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* buffered writes are older than this run */
     if (n == 0)
         return RRDB_OK;
     /* validate: strictly increasing keys, seqnos above everything ingested */
@@ -1124,6 +1142,7 @@ int32_t rrdb_get(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
                  rrdb_result *out)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     arena *a = result_init(out);
     (void)a;
     uint64_t idx;
@@ -1156,6 +1175,7 @@ int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
                  rrdb_result *out)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     result_init(out);
     uint64_t idx;
     int r = find_newest(e, key, key_len, &idx);
@@ -1180,6 +1200,7 @@ int32_t rrdb_batch_get(void *h, uint64_t n_keys, const uint8_t *keys, const uint
                        uint32_t epoch_now, rrdb_result *out)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     arena *a = result_init(out);
     if (n_keys == 0) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_batch_get:922-928 */
@@ -1218,6 +1239,7 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
                            rrdb_result *out)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     result_init(out);
     /* start = generate_key(hk,""), stop = generate_next_blob(hk)
      * (on_sortkey_count:1030-1036) */
@@ -1285,6 +1307,7 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
                        rrdb_result *out)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     arena *a = result_init(out);
     uint32_t hdr = value_hdr_len(e->data_version);
 
@@ -1628,6 +1651,7 @@ static ScanCtx *fetch_ctx(Engine *e, int64_t id)
 int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, rrdb_result *out)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     result_init(out);
     if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
         q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
@@ -1745,6 +1769,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
                             rrdb_compact_stats *stats)
 {
     Engine *e = (Engine *)h;
+    rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     (void)opts;
     rrdb_compact_stats st;
     memset(&st, 0, sizeof(st));
@@ -2008,4 +2033,122 @@ double rrdb_phase_ms(void *h, const char *phase)
     (void)h;
     (void)phase;
     return -1.0;
+}
+
+/* ================= write path (§8(f)1) =================
+ * put/remove buffer into a memtable log; flush sorts by (key asc, seq desc),
+ * keeps the newest version per key (memtable upsert semantics) and appends
+ * the result as a sorted run — the memtable-flush -> L0 step.  Mirrors
+ * pegasus_write_service put/remove + rocksdb_wrapper::write_batch_put/_delete
+ * (reference pegasus_write_service.h:119-207, rocksdb_wrapper.cpp:121-247). */
+
+static void mem_append(Engine *e, uint8_t *key, uint64_t klen, uint8_t *val, uint64_t vlen,
+                       int kind)
+{
+    if (e->mem_n == e->mem_cap) {
+        e->mem_cap = e->mem_cap ? e->mem_cap * 2 : 64;
+        e->mem = realloc(e->mem, e->mem_cap * sizeof(*e->mem));
+    }
+    struct mem_entry *m = &e->mem[e->mem_n++];
+    m->key = key;
+    m->klen = klen;
+    m->val = val;
+    m->vlen = vlen;
+    m->seq = e->next_seq_floor++;
+    m->kind = kind;
+}
+
+int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
+                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts)
+{
+    Engine *e = (Engine *)h;
+    uint64_t klen;
+    uint8_t *key = make_key(hash_key, hklen, sort_key, sklen, &klen);
+    uint32_t hdr = value_hdr_len(e->data_version);
+    uint8_t *val = (uint8_t *)calloc(1, hdr + vlen);
+    if (e->data_version == 2)
+        val[0] = 0x82;
+    update_expire_ts(e->data_version, val, expire_ts);
+    memcpy(val + hdr, value, vlen);
+    mem_append(e, key, klen, val, hdr + vlen, RRDB_KIND_PUT);
+    return RRDB_OK;
+}
+
+int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
+                    uint64_t sklen)
+{
+    Engine *e = (Engine *)h;
+    uint64_t klen;
+    uint8_t *key = make_key(hash_key, hklen, sort_key, sklen, &klen);
+    mem_append(e, key, klen, NULL, 0, RRDB_KIND_DELETE);
+    return RRDB_OK;
+}
+
+uint64_t rrdb_memtable_entries(void *h) { return ((Engine *)h)->mem_n; }
+
+static int mem_cmp(const void *a, const void *b)
+{
+    const struct mem_entry *x = a, *y = b;
+    int c = key_cmp(x->key, x->klen, y->key, y->klen);
+    if (c)
+        return c;
+    return x->seq > y->seq ? -1 : (x->seq < y->seq ? 1 : 0); /* seq desc */
+}
+
+int32_t rrdb_flush(void *h)
+{
+    Engine *e = (Engine *)h;
+    if (e->mem_n == 0)
+        return RRDB_OK;
+    qsort(e->mem, e->mem_n, sizeof(*e->mem), mem_cmp);
+    uint64_t n = 0, kb = 0, vb = 0;
+    for (uint64_t i = 0; i < e->mem_n; i++) {
+        if (i > 0 && key_cmp(e->mem[i - 1].key, e->mem[i - 1].klen, e->mem[i].key,
+                             e->mem[i].klen) == 0)
+            continue; /* older version of the same key: superseded in-memtable */
+        n++;
+        kb += e->mem[i].klen;
+        vb += e->mem[i].vlen;
+    }
+    Run r;
+    r.n = n;
+    r.keys = malloc(kb ? kb : 1);
+    r.koff = malloc((n + 1) * 8);
+    r.vals = malloc(vb ? vb : 1);
+    r.voff = malloc((n + 1) * 8);
+    r.sk = malloc(n * 8);
+    uint64_t j = 0, ko = 0, vo = 0;
+    r.min_seq = ~0ull;
+    r.max_seq = 0;
+    r.koff[0] = 0;
+    r.voff[0] = 0;
+    for (uint64_t i = 0; i < e->mem_n; i++) {
+        if (i > 0 && key_cmp(e->mem[i - 1].key, e->mem[i - 1].klen, e->mem[i].key,
+                             e->mem[i].klen) == 0)
+            continue;
+        memcpy(r.keys + ko, e->mem[i].key, e->mem[i].klen);
+        ko += e->mem[i].klen;
+        if (e->mem[i].vlen)
+            memcpy(r.vals + vo, e->mem[i].val, e->mem[i].vlen);
+        vo += e->mem[i].vlen;
+        r.sk[j] = (e->mem[i].seq << 1) | (uint64_t)e->mem[i].kind;
+        if (e->mem[i].seq < r.min_seq)
+            r.min_seq = e->mem[i].seq;
+        if (e->mem[i].seq > r.max_seq)
+            r.max_seq = e->mem[i].seq;
+        j++;
+        r.koff[j] = ko;
+        r.voff[j] = vo;
+    }
+    for (uint64_t i = 0; i < e->mem_n; i++) {
+        free(e->mem[i].key);
+        free(e->mem[i].val);
+    }
+    e->mem_n = 0;
+    if (e->n_runs == e->runs_cap) {
+        e->runs_cap = e->runs_cap ? e->runs_cap * 2 : 4;
+        e->runs = realloc(e->runs, (size_t)e->runs_cap * sizeof(Run));
+    }
+    e->runs[e->n_runs++] = r;
+    return RRDB_OK;
 }
