@@ -257,6 +257,17 @@ void rrdb_clear_scanner(void *h, int64_t context_id);
 int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
                             rrdb_compact_stats *stats);
 
+/* ---- checkpoint ----
+ * Serializes the partition's runs to <dir>/checkpoint.<decree>/ in the
+ * engine's own run format (decree-tagged like the reference's
+ * `checkpoint.{decree}` snapshots, pegasus_server_impl.cpp:1951-2137;
+ * the on-disk format is ours — SST-byte parity is not owed).
+ * The memtable is flushed first.  rrdb_restore loads a checkpoint into an
+ * EMPTY handle (learner/backup ingest path); checkpoints written by either
+ * backend restore into either backend. */
+int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree);
+int32_t rrdb_restore(void *h, const char *dir, uint64_t decree);
+
 /* ---- introspection ---- */
 uint64_t rrdb_num_runs(void *h);
 uint64_t rrdb_num_records(void *h); /* total across runs, versions included */

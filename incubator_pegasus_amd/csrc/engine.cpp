@@ -1843,3 +1843,121 @@ int32_t rrdb_flush(void *h)
 }
 
 } /* extern "C" */
+
+/* ================= checkpoint (§8(f)2) ================= */
+#include <sys/stat.h>
+#include <cstdio>
+
+extern "C" {
+
+static bool write_file(const std::string &path, const void *data, uint64_t n)
+{
+    FILE *f = fopen(path.c_str(), "wb");
+    if (!f)
+        return false;
+    bool ok = (n == 0) || fwrite(data, 1, n, f) == n;
+    fclose(f);
+    return ok;
+}
+
+static bool read_file(const std::string &path, std::vector<uint8_t> &out)
+{
+    FILE *f = fopen(path.c_str(), "rb");
+    if (!f)
+        return false;
+    fseek(f, 0, SEEK_END);
+    long n = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    out.resize((size_t)n);
+    bool ok = (n == 0) || fread(out.data(), 1, (size_t)n, f) == (size_t)n;
+    fclose(f);
+    return ok;
+}
+
+int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
+{
+    auto *e = (HipEngine *)h;
+    rrdb_flush(h);
+    e->activate();
+    std::string path = std::string(dir) + "/checkpoint." + std::to_string(decree);
+    mkdir(dir, 0755);
+    if (mkdir(path.c_str(), 0755) != 0)
+        return RRDB_IO_ERROR;
+    FILE *mf = fopen((path + "/MANIFEST").c_str(), "w");
+    if (!mf)
+        return RRDB_IO_ERROR;
+    fprintf(mf, "rrdb-checkpoint 1\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n",
+            e->data_version, (unsigned long long)e->next_seq_floor, (int)e->runs.size());
+    for (size_t i = 0; i < e->runs.size(); i++)
+        fprintf(mf, "run %zu %llu\n", i, (unsigned long long)e->runs[i].n);
+    fclose(mf);
+    std::vector<uint8_t> buf;
+    for (size_t i = 0; i < e->runs.size(); i++) {
+        const RunBuf &r = e->runs[i];
+        /* offsets first (host copies give the blob sizes) */
+        std::vector<uint64_t> koff(r.n + 1), voff(r.n + 1), sk(r.n);
+        HIP_OK(hipMemcpy(koff.data(), r.koff, (r.n + 1) * 8, hipMemcpyDeviceToHost));
+        HIP_OK(hipMemcpy(voff.data(), r.voff, (r.n + 1) * 8, hipMemcpyDeviceToHost));
+        HIP_OK(hipMemcpy(sk.data(), r.sk, r.n * 8, hipMemcpyDeviceToHost));
+        std::string base = path + "/run_" + std::to_string(i);
+        buf.resize(koff[r.n]);
+        HIP_OK(hipMemcpy(buf.data(), r.keys, koff[r.n], hipMemcpyDeviceToHost));
+        if (!write_file(base + ".keys", buf.data(), koff[r.n]))
+            return RRDB_IO_ERROR;
+        if (!write_file(base + ".koff", koff.data(), (r.n + 1) * 8))
+            return RRDB_IO_ERROR;
+        buf.resize(voff[r.n]);
+        HIP_OK(hipMemcpy(buf.data(), r.vals, voff[r.n], hipMemcpyDeviceToHost));
+        if (!write_file(base + ".vals", buf.data(), voff[r.n]))
+            return RRDB_IO_ERROR;
+        if (!write_file(base + ".voff", voff.data(), (r.n + 1) * 8))
+            return RRDB_IO_ERROR;
+        if (!write_file(base + ".sk", sk.data(), r.n * 8))
+            return RRDB_IO_ERROR;
+    }
+    return RRDB_OK;
+}
+
+int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
+{
+    auto *e = (HipEngine *)h;
+    if (!e->runs.empty() || !e->memtable.empty())
+        return RRDB_INVALID_ARGUMENT;
+    e->activate();
+    std::string path = std::string(dir) + "/checkpoint." + std::to_string(decree);
+    FILE *mf = fopen((path + "/MANIFEST").c_str(), "r");
+    if (!mf)
+        return RRDB_IO_ERROR;
+    unsigned dv = 1;
+    unsigned long long floor_ = 0;
+    int n_runs = 0, ver = 0;
+    if (fscanf(mf, "rrdb-checkpoint %d\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n", &ver,
+               &dv, &floor_, &n_runs) != 4 ||
+        ver != 1) {
+        fclose(mf);
+        return RRDB_CORRUPTION;
+    }
+    fclose(mf);
+    e->data_version = dv;
+    for (int i = 0; i < n_runs; i++) {
+        std::string base = path + "/run_" + std::to_string(i);
+        std::vector<uint8_t> keys, koff, vals, voff, sk;
+        if (!read_file(base + ".keys", keys) || !read_file(base + ".koff", koff) ||
+            !read_file(base + ".vals", vals) || !read_file(base + ".voff", voff) ||
+            !read_file(base + ".sk", sk))
+            return RRDB_IO_ERROR;
+        RunBuf r;
+        r.n = koff.size() / 8 - 1;
+        r.keys = e->upload_bytes(keys.data(), keys.size());
+        r.koff = (uint64_t *)e->upload_bytes(koff.data(), koff.size());
+        r.vals = e->upload_bytes(vals.data(), vals.size());
+        r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size());
+        r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size());
+        e->runs.push_back(r);
+    }
+    e->next_seq_floor = floor_;
+    e->d_runs_dirty = true;
+    return RRDB_OK;
+}
+
+} /* extern "C" */

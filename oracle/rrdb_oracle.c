@@ -2152,3 +2152,135 @@ int32_t rrdb_flush(void *h)
     e->runs[e->n_runs++] = r;
     return RRDB_OK;
 }
+
+/* ================= checkpoint (§8(f)2) ================= */
+#include <sys/stat.h>
+
+static int write_blob_file(const char *path, const void *data, uint64_t n)
+{
+    FILE *f = fopen(path, "wb");
+    if (!f)
+        return -1;
+    if (n && fwrite(data, 1, n, f) != n) {
+        fclose(f);
+        return -1;
+    }
+    fclose(f);
+    return 0;
+}
+
+static uint8_t *read_blob_file(const char *path, uint64_t *n_out)
+{
+    FILE *f = fopen(path, "rb");
+    if (!f)
+        return NULL;
+    fseek(f, 0, SEEK_END);
+    long n = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    uint8_t *buf = (uint8_t *)malloc(n ? (size_t)n : 1);
+    if (n && fread(buf, 1, (size_t)n, f) != (size_t)n) {
+        fclose(f);
+        free(buf);
+        return NULL;
+    }
+    fclose(f);
+    *n_out = (uint64_t)n;
+    return buf;
+}
+
+int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
+{
+    Engine *e = (Engine *)h;
+    rrdb_flush(h);
+    char path[4096];
+    snprintf(path, sizeof(path), "%s/checkpoint.%llu", dir, (unsigned long long)decree);
+    mkdir(dir, 0755);
+    if (mkdir(path, 0755) != 0)
+        return RRDB_IO_ERROR;
+    char fp[4096];
+    snprintf(fp, sizeof(fp), "%s/MANIFEST", path);
+    FILE *mf = fopen(fp, "w");
+    if (!mf)
+        return RRDB_IO_ERROR;
+    fprintf(mf, "rrdb-checkpoint 1\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n",
+            e->data_version, (unsigned long long)e->next_seq_floor, e->n_runs);
+    for (int i = 0; i < e->n_runs; i++)
+        fprintf(mf, "run %d %llu\n", i, (unsigned long long)e->runs[i].n);
+    fclose(mf);
+    for (int i = 0; i < e->n_runs; i++) {
+        const Run *r = &e->runs[i];
+        snprintf(fp, sizeof(fp), "%s/run_%d.keys", path, i);
+        if (write_blob_file(fp, r->keys, r->koff[r->n]))
+            return RRDB_IO_ERROR;
+        snprintf(fp, sizeof(fp), "%s/run_%d.koff", path, i);
+        if (write_blob_file(fp, r->koff, (r->n + 1) * 8))
+            return RRDB_IO_ERROR;
+        snprintf(fp, sizeof(fp), "%s/run_%d.vals", path, i);
+        if (write_blob_file(fp, r->vals, r->voff[r->n]))
+            return RRDB_IO_ERROR;
+        snprintf(fp, sizeof(fp), "%s/run_%d.voff", path, i);
+        if (write_blob_file(fp, r->voff, (r->n + 1) * 8))
+            return RRDB_IO_ERROR;
+        snprintf(fp, sizeof(fp), "%s/run_%d.sk", path, i);
+        if (write_blob_file(fp, r->sk, r->n * 8))
+            return RRDB_IO_ERROR;
+    }
+    return RRDB_OK;
+}
+
+int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
+{
+    Engine *e = (Engine *)h;
+    if (e->n_runs != 0 || e->mem_n != 0)
+        return RRDB_INVALID_ARGUMENT; /* restore only into an empty handle */
+    char path[4096], fp[4096];
+    snprintf(path, sizeof(path), "%s/checkpoint.%llu", dir, (unsigned long long)decree);
+    snprintf(fp, sizeof(fp), "%s/MANIFEST", path);
+    FILE *mf = fopen(fp, "r");
+    if (!mf)
+        return RRDB_IO_ERROR;
+    unsigned dv = 1;
+    unsigned long long floor_ = 0;
+    int n_runs = 0, ver = 0;
+    if (fscanf(mf, "rrdb-checkpoint %d\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n",
+               &ver, &dv, &floor_, &n_runs) != 4 ||
+        ver != 1) {
+        fclose(mf);
+        return RRDB_CORRUPTION;
+    }
+    fclose(mf);
+    e->data_version = dv;
+    for (int i = 0; i < n_runs; i++) {
+        Run r;
+        uint64_t nb;
+        snprintf(fp, sizeof(fp), "%s/run_%d.keys", path, i);
+        r.keys = read_blob_file(fp, &nb);
+        snprintf(fp, sizeof(fp), "%s/run_%d.koff", path, i);
+        r.koff = (uint64_t *)read_blob_file(fp, &nb);
+        r.n = nb / 8 - 1;
+        snprintf(fp, sizeof(fp), "%s/run_%d.vals", path, i);
+        r.vals = read_blob_file(fp, &nb);
+        snprintf(fp, sizeof(fp), "%s/run_%d.voff", path, i);
+        r.voff = (uint64_t *)read_blob_file(fp, &nb);
+        snprintf(fp, sizeof(fp), "%s/run_%d.sk", path, i);
+        r.sk = (uint64_t *)read_blob_file(fp, &nb);
+        if (!r.keys || !r.koff || !r.vals || !r.voff || !r.sk)
+            return RRDB_IO_ERROR;
+        r.min_seq = ~0ull;
+        r.max_seq = 0;
+        for (uint64_t j = 0; j < r.n; j++) {
+            uint64_t s = r.sk[j] >> 1;
+            if (s < r.min_seq)
+                r.min_seq = s;
+            if (s > r.max_seq)
+                r.max_seq = s;
+        }
+        if (e->n_runs == e->runs_cap) {
+            e->runs_cap = e->runs_cap ? e->runs_cap * 2 : 4;
+            e->runs = realloc(e->runs, (size_t)e->runs_cap * sizeof(Run));
+        }
+        e->runs[e->n_runs++] = r;
+    }
+    e->next_seq_floor = floor_;
+    return RRDB_OK;
+}
