@@ -66,6 +66,12 @@ void launch_emit_compact_inmajor(const DevRun *, int, const uint64_t *, uint64_t
                                  const uint32_t *, const uint64_t *, const uint64_t *,
                                  const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *,
                                  uint64_t *, uint64_t *, uint64_t, hipStream_t);
+void launch_emit_compact_chunked(const DevRun *, const uint64_t *, uint64_t, const uint64_t *,
+                                 const uint8_t *, const uint32_t *, const uint64_t *,
+                                 const uint64_t *, const uint64_t *, uint32_t, uint64_t,
+                                 uint64_t, uint64_t, uint64_t *, uint64_t *, uint32_t *,
+                                 uint32_t *, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
+                                 uint64_t *, hipStream_t);
 void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint64_t *,
                          const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
@@ -498,8 +504,9 @@ struct HipEngine {
     int64_t next_ctx_id = 0;
     std::mutex mu;
     std::unordered_map<std::string, double> phase_ms;
-    int emit_mode = 0; /* 0 = rank-major (default; won the on-GPU A/B),
-                          1 = input-major (env "engine.emit_mode") */
+    int emit_mode = 2; /* 2 = chunked (default; 2.8 TB/s on the copy probe),
+                          0 = rank-major waves, 1 = input-major waves
+                          (env "engine.emit_mode": chunked|rank|input) */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
@@ -832,7 +839,7 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
         } else if (k == "rocksdb.multi_get_max_iteration_size") {
             e->mg_max_iter_size = (uint64_t)atoll(v.c_str());
         } else if (k == "engine.emit_mode") {
-            e->emit_mode = (v == "input") ? 1 : 0;
+            e->emit_mode = (v == "input") ? 1 : (v == "rank" ? 0 : 2);
         }
     }
     return RRDB_OK;
@@ -1718,7 +1725,17 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             HIP_OK(hipMalloc(&nr.sk, n_out * 8));
         }
         HIP_OK(hipEventRecord(ev[3], e->stream));
-        if (e->emit_mode == 1)
+        if (e->emit_mode == 2) {
+            uint64_t *d_row_ksrc = e->talloc<uint64_t>(n_out * 8);
+            uint64_t *d_row_vsrc = e->talloc<uint64_t>(n_out * 8);
+            uint32_t *d_row_patch = e->talloc<uint32_t>(n_out * 4);
+            uint32_t *d_row_expire = e->talloc<uint32_t>(n_out * 4);
+            launch_emit_compact_chunked(dr, d_order, total, d_keepw, d_changed, d_new_expire,
+                                        d_kpos, d_koffs, d_voffs, e->data_version, n_out, kbytes,
+                                        vbytes, d_row_ksrc, d_row_vsrc, d_row_patch,
+                                        d_row_expire, nr.keys, nr.vals, nr.koff, nr.voff, nr.sk,
+                                        e->stream);
+        } else if (e->emit_mode == 1)
             launch_emit_compact_inmajor(dr, R, d_wp, total, d_rank_of, d_keepw, d_changed,
                                         d_new_expire, d_kpos, d_koffs, d_voffs,
                                         e->data_version, nr.keys, nr.vals, nr.koff, nr.voff,
