@@ -30,7 +30,10 @@
 void launch_crc64_table_init(const uint64_t *host_table);
 void launch_bounds(const DevRun *, int, const uint8_t *, uint64_t, uint64_t *, int, hipStream_t);
 void launch_rank(const DevRun *, int, const uint64_t *, const uint64_t *, const uint64_t *,
-                 uint64_t, uint64_t *, uint8_t *, hipStream_t);
+                 uint64_t, uint64_t *, uint8_t *, const uint64_t *, const uint64_t *,
+                 hipStream_t);
+void launch_bound_table(const DevRun *, int, const uint64_t *, const uint64_t *,
+                        const uint64_t *, uint64_t, uint64_t *, hipStream_t);
 void launch_visible(const DevRun *, const uint64_t *, const uint8_t *, uint64_t, uint64_t *,
                     hipStream_t);
 void launch_gather(const uint64_t *, const uint64_t *, const uint64_t *, uint64_t, uint64_t *,
@@ -60,7 +63,7 @@ void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *
 void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                          uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *, uint64_t *,
-                         CompactStatsDev *, hipStream_t);
+                         const uint64_t *, const uint64_t *, CompactStatsDev *, hipStream_t);
 void launch_emit_compact_inmajor(const DevRun *, int, const uint64_t *, uint64_t,
                                  const uint64_t *, const uint64_t *, const uint8_t *,
                                  const uint32_t *, const uint64_t *, const uint64_t *,
@@ -634,6 +637,27 @@ struct HipEngine {
         return out;
     }
     void tfree(void *) {} /* arena memory is reclaimed at scratch_reset */
+    /* build the search bound table for windows lo/hi (BT_SHIFT=8 blocks);
+     * returns device row-offset + table pointers (arena memory) */
+    void build_bound_table(DevRun *dr, int R, const std::vector<uint64_t> &lo,
+                           const std::vector<uint64_t> &hi, const uint64_t *d_lo,
+                           const uint64_t *d_hi, uint64_t **out_bt_off, uint64_t **out_bt)
+    {
+        std::vector<uint64_t> bt_off(R + 1);
+        uint64_t rows = 0;
+        for (int r = 0; r < R; r++) {
+            bt_off[r] = rows;
+            uint64_t w = hi[r] - lo[r];
+            rows += (w >> 8) + 2; /* P_r + sentinel */
+        }
+        bt_off[R] = rows;
+        uint64_t *d_bt_off = (uint64_t *)upload_tmp(bt_off.data(), (R + 1) * 8);
+        uint64_t *d_bt = talloc<uint64_t>(rows * (uint64_t)R * 8);
+        launch_bound_table(dr, R, d_lo, d_hi, d_bt_off, rows, d_bt, stream);
+        *out_bt_off = d_bt_off;
+        *out_bt = d_bt;
+    }
+
     uint64_t *psum_scratch(uint64_t n)
     {
         return talloc<uint64_t>(psum_scratch_elems(n) * 8);
@@ -689,7 +713,10 @@ struct HipEngine {
         uint8_t *d_shadow = talloc<uint8_t>(total);
         uint64_t *d_flags = talloc<uint64_t>(total * 8);
         uint64_t *d_pos = talloc<uint64_t>(total * 8);
-        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, stream);
+        uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
+        if (R > 1 && total > 100000)
+            build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
+        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt, stream);
         launch_visible(dr, d_order, d_shadow, total, d_flags, stream);
         launch_psum(d_flags, d_pos, total, psum_scratch(total), stream);
         uint64_t lastp = 0, lastf = 0;
@@ -1679,9 +1706,13 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     uint64_t *d_rank_of = e->emit_mode == 1 ? e->talloc<uint64_t>(total * 8) : nullptr;
     CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
     HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
+    uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
+    if (R > 1 && total > 100000)
+        e->build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
     HIP_OK(hipEventRecord(ev[0], e->stream));
     launch_rank_compact(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw, d_changed,
-                        d_new_expire, d_ksz, d_vsz, d_rank_of, d_stats, e->stream);
+                        d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, d_stats,
+                        e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
     HIP_OK(hipEventRecord(ev[2], e->stream));
     launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);

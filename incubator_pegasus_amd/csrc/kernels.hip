@@ -270,6 +270,47 @@ __global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t
                    : dev_lower_bound(runs[r], key, klen, 0, runs[r].n);
 }
 
+/* ---- bound table: positions of every (1<<BT_SHIFT)-th record of each run
+ * in every other run.  Narrows each record's cross-run binary searches from
+ * the full run to a ~block-sized, L2-hot window (correctness never depends
+ * on the narrowing: boundaries come from the same comparator). ---- */
+#define BT_SHIFT 8 /* 256 records per block */
+
+/* layout: for run r, rows j = 0..P_r (P_r = ceil(w_r / 256)); row j holds R
+ * u64s = search bounds of run-r record (lo_r + j*256) in every run; the last
+ * row is hi[].  bt_off[r] = row offset of run r's table. */
+__global__ void k_bound_table(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
+                              const uint64_t *bt_off, uint64_t total_rows, uint64_t *bt)
+{
+    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total_rows;
+         t += gridDim.x * (uint64_t)blockDim.x) {
+        /* find the run this row belongs to */
+        int r = 0;
+        while (bt_off[r + 1] <= t)
+            r++;
+        uint64_t j = t - bt_off[r];
+        uint64_t nrows = bt_off[r + 1] - bt_off[r]; /* P_r + 1 */
+        uint64_t *row = bt + t * (uint64_t)R;
+        uint64_t i = lo[r] + (j << BT_SHIFT);
+        if (j == nrows - 1 || i >= hi[r]) { /* sentinel row: hi */
+            for (int q = 0; q < R; q++)
+                row[q] = hi[q];
+            continue;
+        }
+        uint64_t kl;
+        const uint8_t *k = run_key(runs[r], i, &kl);
+        for (int q = 0; q < R; q++) {
+            if (q == r) {
+                row[q] = i;
+            } else if (q > r) {
+                row[q] = dev_upper_bound(runs[q], k, kl, lo[q], hi[q]);
+            } else {
+                row[q] = dev_lower_bound(runs[q], k, kl, lo[q], hi[q]);
+            }
+        }
+    }
+}
+
 /* ================= rank merge =================
  * For windowed record t -> (r,i):
  * rank = (i - lo[r]) + sum_{r'>r} upper_bound(r',key) + sum_{r'<r} lower_bound(r',key)
@@ -277,7 +318,8 @@ __global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t
  * Scatter order[rank] = (r<<40)|i.  wprefix[r] = running sum of window sizes. */
 __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
                        const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order,
-                       uint8_t *shadowed /* [total], by rank position */)
+                       uint8_t *shadowed /* [total], by rank position */,
+                       const uint64_t *bt_off, const uint64_t *bt)
 {
     for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
          t += gridDim.x * (uint64_t)blockDim.x) {
@@ -289,13 +331,19 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
         const uint8_t *k = run_key(runs[r], i, &kl);
         uint64_t rank = i - lo[r];
         int shadow = 0;
+        const uint64_t *b0 = nullptr, *b1 = nullptr;
+        if (bt) {
+            uint64_t j = (i - lo[r]) >> BT_SHIFT;
+            b0 = bt + (bt_off[r] + j) * (uint64_t)R;
+            b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
+        }
         for (int q = 0; q < R; q++) {
             if (q == r)
                 continue;
+            uint64_t qlo = b0 ? b0[q] : lo[q];
+            uint64_t qhi = b1 ? b1[q] : hi[q];
             if (q > r) {
-                /* newer run: equal keys precede ours; if one exists at ub-1,
-                 * a newer version shadows this record (newest-wins) */
-                uint64_t ub = dev_upper_bound(runs[q], k, kl, lo[q], hi[q]);
+                uint64_t ub = dev_upper_bound(runs[q], k, kl, qlo, qhi);
                 if (!shadow && ub > lo[q]) {
                     uint64_t pl;
                     const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
@@ -304,7 +352,7 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
                 }
                 rank += ub - lo[q];
             } else {
-                rank += dev_lower_bound(runs[q], k, kl, lo[q], hi[q]) - lo[q];
+                rank += dev_lower_bound(runs[q], k, kl, qlo, qhi) - lo[q];
             }
         }
         order[rank] = ((uint64_t)r << 40) | i;
@@ -737,6 +785,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                                uint64_t *order, uint64_t *keepw, uint8_t *changed,
                                uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
                                uint64_t *rank_of /* [total] by input index, may be null */,
+                               const uint64_t *bt_off, const uint64_t *bt,
                                CompactStatsDev *stats)
 {
     uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
@@ -755,11 +804,19 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
             const uint8_t *k = run_key(runs[r], i, &kl);
             uint64_t rank = i - lo[r];
             int shadow = 0;
+            const uint64_t *b0 = nullptr, *b1 = nullptr;
+            if (bt) {
+                uint64_t j = (i - lo[r]) >> BT_SHIFT;
+                b0 = bt + (bt_off[r] + j) * (uint64_t)R;
+                b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
+            }
             for (int q = 0; q < R; q++) {
                 if (q == r)
                     continue;
+                uint64_t qlo = b0 ? b0[q] : lo[q];
+                uint64_t qhi = b1 ? b1[q] : hi[q];
                 if (q > r) {
-                    uint64_t ub = dev_upper_bound(runs[q], k, kl, lo[q], hi[q]);
+                    uint64_t ub = dev_upper_bound(runs[q], k, kl, qlo, qhi);
                     if (!shadow && ub > lo[q]) {
                         uint64_t pl;
                         const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
@@ -768,7 +825,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                     }
                     rank += ub - lo[q];
                 } else {
-                    rank += dev_lower_bound(runs[q], k, kl, lo[q], hi[q]) - lo[q];
+                    rank += dev_lower_bound(runs[q], k, kl, qlo, qhi) - lo[q];
                 }
             }
             uint8_t ch;
@@ -1023,10 +1080,19 @@ void launch_bounds(const DevRun *d_runs, int R, const uint8_t *d_key, uint64_t k
 
 void launch_rank(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
                  const uint64_t *d_wprefix, uint64_t total, uint64_t *d_order,
-                 uint8_t *d_shadowed, hipStream_t s)
+                 uint8_t *d_shadowed, const uint64_t *d_bt_off, const uint64_t *d_bt,
+                 hipStream_t s)
 {
     k_rank<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix, total,
-                                                    d_order, d_shadowed);
+                                                    d_order, d_shadowed, d_bt_off, d_bt);
+}
+
+void launch_bound_table(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
+                        const uint64_t *d_bt_off, uint64_t total_rows, uint64_t *d_bt,
+                        hipStream_t s)
+{
+    k_bound_table<<<grid_for(total_rows, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_bt_off,
+                                                                total_rows, d_bt);
 }
 
 void launch_visible(const DevRun *d_runs, const uint64_t *d_order, const uint8_t *d_shadowed,
@@ -1151,12 +1217,14 @@ void launch_rank_compact(const DevRun *d_runs, int R, const uint64_t *d_lo, cons
                          const uint64_t *d_wprefix, uint64_t total, const CompactParams &cp,
                          uint64_t *d_order, uint64_t *d_keepw, uint8_t *d_changed,
                          uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
-                         uint64_t *d_rank_of, CompactStatsDev *d_stats, hipStream_t s)
+                         uint64_t *d_rank_of, const uint64_t *d_bt_off, const uint64_t *d_bt,
+                         CompactStatsDev *d_stats, hipStream_t s)
 {
     k_rank_compact<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix,
                                                             total, cp, d_order, d_keepw,
                                                             d_changed, d_new_expire, d_ksz,
-                                                            d_vsz, d_rank_of, d_stats);
+                                                            d_vsz, d_rank_of, d_bt_off, d_bt,
+                                                            d_stats);
 }
 
 void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
