@@ -192,6 +192,19 @@ class RrdbLib:
             C.c_void_p, C.POINTER(_CompactOptions), C.c_uint32, C.POINTER(_CompactStats)]
         L.rrdb_phase_ms.restype = C.c_double
         L.rrdb_phase_ms.argtypes = [C.c_void_p, C.c_char_p]
+        L.rrdb_put.restype = C.c_int32
+        L.rrdb_put.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64, C.c_char_p, C.c_uint64,
+                               C.c_char_p, C.c_uint64, C.c_uint32]
+        L.rrdb_remove.restype = C.c_int32
+        L.rrdb_remove.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64, C.c_char_p, C.c_uint64]
+        L.rrdb_flush.restype = C.c_int32
+        L.rrdb_flush.argtypes = [C.c_void_p]
+        L.rrdb_memtable_entries.restype = C.c_uint64
+        L.rrdb_memtable_entries.argtypes = [C.c_void_p]
+        L.rrdb_checkpoint.restype = C.c_int32
+        L.rrdb_checkpoint.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64]
+        L.rrdb_restore.restype = C.c_int32
+        L.rrdb_restore.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64]
         L.rrdb_num_runs.restype = C.c_uint64
         L.rrdb_num_runs.argtypes = [C.c_void_p]
         L.rrdb_num_records.restype = C.c_uint64
@@ -420,6 +433,27 @@ class RrdbPartition:
 
     def phase_ms(self, phase: str) -> float:
         return self._L.rrdb_phase_ms(self._h, phase.encode())
+
+    # ---- write path (§8(f)1) ----
+    def put(self, hash_key: bytes, sort_key: bytes, value: bytes, expire_ts: int = 0):
+        return self._L.rrdb_put(self._h, hash_key, len(hash_key), sort_key, len(sort_key),
+                                value, len(value), expire_ts)
+
+    def remove(self, hash_key: bytes, sort_key: bytes):
+        return self._L.rrdb_remove(self._h, hash_key, len(hash_key), sort_key, len(sort_key))
+
+    def flush(self):
+        return self._L.rrdb_flush(self._h)
+
+    def memtable_entries(self):
+        return self._L.rrdb_memtable_entries(self._h)
+
+    # ---- checkpoint (§8(f)2) ----
+    def checkpoint(self, directory: str, decree: int):
+        return self._L.rrdb_checkpoint(self._h, directory.encode(), decree)
+
+    def restore(self, directory: str, decree: int):
+        return self._L.rrdb_restore(self._h, directory.encode(), decree)
 
     def manual_compact(self, epoch_now: int, *, target_level=-1, bottommost_force=True,
                        keep_inputs=False):
