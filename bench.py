@@ -84,6 +84,7 @@ def main():
     ap.add_argument("--cpu-sample", type=int, default=2_000_000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
+    ap.add_argument("--bt-shift", type=int, default=8)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -119,7 +120,8 @@ def main():
     for p in range(args.partitions):
         runs = build_partition_data(keys_pp, args.runs, rank, p)
         eng = hip.open(1, p, local_rank)
-        eng.set_envs({"engine.emit_mode": args.emit_mode})
+        eng.set_envs({"engine.emit_mode": args.emit_mode,
+                      "engine.bt_shift": str(args.bt_shift)})
         for r in runs:
             eng.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
                                   np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])

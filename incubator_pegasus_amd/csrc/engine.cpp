@@ -30,10 +30,10 @@
 void launch_crc64_table_init(const uint64_t *host_table);
 void launch_bounds(const DevRun *, int, const uint8_t *, uint64_t, uint64_t *, int, hipStream_t);
 void launch_rank(const DevRun *, int, const uint64_t *, const uint64_t *, const uint64_t *,
-                 uint64_t, uint64_t *, uint8_t *, const uint64_t *, const uint64_t *,
+                 uint64_t, uint64_t *, uint8_t *, const uint64_t *, const uint64_t *, int,
                  hipStream_t);
 void launch_bound_table(const DevRun *, int, const uint64_t *, const uint64_t *,
-                        const uint64_t *, uint64_t, uint64_t *, hipStream_t);
+                        const uint64_t *, uint64_t, int, uint64_t *, hipStream_t);
 void launch_visible(const DevRun *, const uint64_t *, const uint8_t *, uint64_t, uint64_t *,
                     hipStream_t);
 void launch_gather(const uint64_t *, const uint64_t *, const uint64_t *, uint64_t, uint64_t *,
@@ -63,7 +63,8 @@ void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *
 void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                          uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *, uint64_t *,
-                         const uint64_t *, const uint64_t *, CompactStatsDev *, hipStream_t);
+                         const uint64_t *, const uint64_t *, int, CompactStatsDev *,
+                         hipStream_t);
 void launch_emit_compact_inmajor(const DevRun *, int, const uint64_t *, uint64_t,
                                  const uint64_t *, const uint64_t *, const uint8_t *,
                                  const uint32_t *, const uint64_t *, const uint64_t *,
@@ -510,6 +511,8 @@ struct HipEngine {
     int emit_mode = 2; /* 2 = chunked (default; 2.8 TB/s on the copy probe),
                           0 = rank-major waves, 1 = input-major waves
                           (env "engine.emit_mode": chunked|rank|input) */
+    int bt_shift = 8;  /* bound-table block = 1<<bt_shift records (env
+                          "engine.bt_shift") */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
@@ -648,12 +651,12 @@ struct HipEngine {
         for (int r = 0; r < R; r++) {
             bt_off[r] = rows;
             uint64_t w = hi[r] - lo[r];
-            rows += (w >> 8) + 2; /* P_r + sentinel */
+            rows += (w >> bt_shift) + 2; /* P_r + sentinel */
         }
         bt_off[R] = rows;
         uint64_t *d_bt_off = (uint64_t *)upload_tmp(bt_off.data(), (R + 1) * 8);
         uint64_t *d_bt = talloc<uint64_t>(rows * (uint64_t)R * 8);
-        launch_bound_table(dr, R, d_lo, d_hi, d_bt_off, rows, d_bt, stream);
+        launch_bound_table(dr, R, d_lo, d_hi, d_bt_off, rows, bt_shift, d_bt, stream);
         *out_bt_off = d_bt_off;
         *out_bt = d_bt;
     }
@@ -716,7 +719,8 @@ struct HipEngine {
         uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
         if (R > 1 && total > 100000)
             build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
-        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt, stream);
+        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
+                    bt_shift, stream);
         launch_visible(dr, d_order, d_shadow, total, d_flags, stream);
         launch_psum(d_flags, d_pos, total, psum_scratch(total), stream);
         uint64_t lastp = 0, lastf = 0;
@@ -865,6 +869,10 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
             e->mg_max_iter_count = (uint32_t)atoll(v.c_str());
         } else if (k == "rocksdb.multi_get_max_iteration_size") {
             e->mg_max_iter_size = (uint64_t)atoll(v.c_str());
+        } else if (k == "engine.bt_shift") {
+            int s_ = atoi(v.c_str());
+            if (s_ >= 4 && s_ <= 16)
+                e->bt_shift = s_;
         } else if (k == "engine.emit_mode") {
             e->emit_mode = (v == "input") ? 1 : (v == "rank" ? 0 : 2);
         }
@@ -1711,8 +1719,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         e->build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
     HIP_OK(hipEventRecord(ev[0], e->stream));
     launch_rank_compact(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw, d_changed,
-                        d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, d_stats,
-                        e->stream);
+                        d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, e->bt_shift,
+                        d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
     HIP_OK(hipEventRecord(ev[2], e->stream));
     launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);

@@ -274,13 +274,14 @@ __global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t
  * in every other run.  Narrows each record's cross-run binary searches from
  * the full run to a ~block-sized, L2-hot window (correctness never depends
  * on the narrowing: boundaries come from the same comparator). ---- */
-#define BT_SHIFT 8 /* 256 records per block */
+#define BT_SHIFT_DEFAULT 8 /* 256 records per block (env engine.bt_shift) */
 
 /* layout: for run r, rows j = 0..P_r (P_r = ceil(w_r / 256)); row j holds R
  * u64s = search bounds of run-r record (lo_r + j*256) in every run; the last
  * row is hi[].  bt_off[r] = row offset of run r's table. */
 __global__ void k_bound_table(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
-                              const uint64_t *bt_off, uint64_t total_rows, uint64_t *bt)
+                              const uint64_t *bt_off, uint64_t total_rows, int bt_shift,
+                              uint64_t *bt)
 {
     for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total_rows;
          t += gridDim.x * (uint64_t)blockDim.x) {
@@ -291,7 +292,7 @@ __global__ void k_bound_table(const DevRun *runs, int R, const uint64_t *lo, con
         uint64_t j = t - bt_off[r];
         uint64_t nrows = bt_off[r + 1] - bt_off[r]; /* P_r + 1 */
         uint64_t *row = bt + t * (uint64_t)R;
-        uint64_t i = lo[r] + (j << BT_SHIFT);
+        uint64_t i = lo[r] + (j << bt_shift);
         if (j == nrows - 1 || i >= hi[r]) { /* sentinel row: hi */
             for (int q = 0; q < R; q++)
                 row[q] = hi[q];
@@ -319,7 +320,7 @@ __global__ void k_bound_table(const DevRun *runs, int R, const uint64_t *lo, con
 __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
                        const uint64_t *wprefix /* [R+1] */, uint64_t total, uint64_t *order,
                        uint8_t *shadowed /* [total], by rank position */,
-                       const uint64_t *bt_off, const uint64_t *bt)
+                       const uint64_t *bt_off, const uint64_t *bt, int bt_shift)
 {
     for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
          t += gridDim.x * (uint64_t)blockDim.x) {
@@ -333,7 +334,7 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
         int shadow = 0;
         const uint64_t *b0 = nullptr, *b1 = nullptr;
         if (bt) {
-            uint64_t j = (i - lo[r]) >> BT_SHIFT;
+            uint64_t j = (i - lo[r]) >> bt_shift;
             b0 = bt + (bt_off[r] + j) * (uint64_t)R;
             b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
         }
@@ -785,7 +786,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                                uint64_t *order, uint64_t *keepw, uint8_t *changed,
                                uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
                                uint64_t *rank_of /* [total] by input index, may be null */,
-                               const uint64_t *bt_off, const uint64_t *bt,
+                               const uint64_t *bt_off, const uint64_t *bt, int bt_shift,
                                CompactStatsDev *stats)
 {
     uint64_t tid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
@@ -806,7 +807,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
             int shadow = 0;
             const uint64_t *b0 = nullptr, *b1 = nullptr;
             if (bt) {
-                uint64_t j = (i - lo[r]) >> BT_SHIFT;
+                uint64_t j = (i - lo[r]) >> bt_shift;
                 b0 = bt + (bt_off[r] + j) * (uint64_t)R;
                 b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
             }
@@ -1081,18 +1082,19 @@ void launch_bounds(const DevRun *d_runs, int R, const uint8_t *d_key, uint64_t k
 void launch_rank(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
                  const uint64_t *d_wprefix, uint64_t total, uint64_t *d_order,
                  uint8_t *d_shadowed, const uint64_t *d_bt_off, const uint64_t *d_bt,
-                 hipStream_t s)
+                 int bt_shift, hipStream_t s)
 {
     k_rank<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix, total,
-                                                    d_order, d_shadowed, d_bt_off, d_bt);
+                                                    d_order, d_shadowed, d_bt_off, d_bt,
+                                                    bt_shift);
 }
 
 void launch_bound_table(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
-                        const uint64_t *d_bt_off, uint64_t total_rows, uint64_t *d_bt,
-                        hipStream_t s)
+                        const uint64_t *d_bt_off, uint64_t total_rows, int bt_shift,
+                        uint64_t *d_bt, hipStream_t s)
 {
     k_bound_table<<<grid_for(total_rows, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_bt_off,
-                                                                total_rows, d_bt);
+                                                                total_rows, bt_shift, d_bt);
 }
 
 void launch_visible(const DevRun *d_runs, const uint64_t *d_order, const uint8_t *d_shadowed,
@@ -1218,13 +1220,13 @@ void launch_rank_compact(const DevRun *d_runs, int R, const uint64_t *d_lo, cons
                          uint64_t *d_order, uint64_t *d_keepw, uint8_t *d_changed,
                          uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
                          uint64_t *d_rank_of, const uint64_t *d_bt_off, const uint64_t *d_bt,
-                         CompactStatsDev *d_stats, hipStream_t s)
+                         int bt_shift, CompactStatsDev *d_stats, hipStream_t s)
 {
     k_rank_compact<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_wprefix,
                                                             total, cp, d_order, d_keepw,
                                                             d_changed, d_new_expire, d_ksz,
                                                             d_vsz, d_rank_of, d_bt_off, d_bt,
-                                                            d_stats);
+                                                            bt_shift, d_stats);
 }
 
 void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, uint64_t m,
