@@ -673,6 +673,36 @@ struct HipEngine {
         return d;
     }
 
+    /* pinned host staging (reused; grown to high-water) — unpinned D2H runs
+     * at a fraction of PCIe rate, so big result copies stage through here */
+    uint8_t *pinned = nullptr;
+    size_t pinned_cap = 0;
+    uint8_t *pinned_buf(size_t n)
+    {
+        if (n > pinned_cap) {
+            if (pinned)
+                (void)hipHostFree(pinned);
+            pinned_cap = std::max<size_t>(n, pinned_cap * 2);
+            HIP_OK(hipHostMalloc((void **)&pinned, pinned_cap));
+        }
+        return pinned;
+    }
+    /* D2H via pinned staging into an arbitrary host destination */
+    void d2h(void *dst, const void *src, size_t n)
+    {
+        if (n == 0)
+            return;
+        if (n <= (64u << 10)) { /* small copies: direct */
+            HIP_OK(hipMemcpyAsync(dst, src, n, hipMemcpyDeviceToHost, stream));
+            HIP_OK(hipStreamSynchronize(stream));
+            return;
+        }
+        uint8_t *pb = pinned_buf(n);
+        HIP_OK(hipMemcpyAsync(pb, src, n, hipMemcpyDeviceToHost, stream));
+        HIP_OK(hipStreamSynchronize(stream));
+        memcpy(dst, pb, n);
+    }
+
     /* build the visible view for [start, stop_excl); returns device array
      * (caller frees) + count */
     void build_view(const std::string *start, const std::string *stop_excl, uint64_t **out_view,
@@ -824,6 +854,8 @@ void rrdb_close(void *h)
         (void)hipFree(e->d_runs);
     for (auto &b : e->sblocks)
         (void)hipFree(b.first);
+    if (e->pinned)
+        (void)hipHostFree(e->pinned);
     if (e->d_ops)
         (void)hipFree(e->d_ops);
     if (e->d_rules)
@@ -971,8 +1003,7 @@ static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const ui
     if (total) {
         uint8_t *d_blob = e->talloc<uint8_t>(total);
         launch_emit_values(dr, d_hit, d_status, nq, e->data_version, d_voffs, d_blob, e->stream);
-        HIP_OK(hipMemcpyAsync(blob, d_blob, total, hipMemcpyDeviceToHost, e->stream));
-        HIP_OK(hipStreamSynchronize(e->stream));
+        e->d2h(blob, d_blob, total);
         e->tfree(d_blob);
     }
     for (uint64_t i = 0; i < nq; i++)
@@ -1025,15 +1056,18 @@ int32_t rrdb_batch_get(void *h, uint64_t n_keys, const uint8_t *keys, const uint
     get_core(e, n_keys, keys, key_offs, epoch_now, status, vals, a);
     out->keys = (rrdb_slice *)a->alloc(n_keys * sizeof(rrdb_slice));
     out->values = (rrdb_slice *)a->alloc(n_keys * sizeof(rrdb_slice));
-    uint64_t m = 0;
+    /* one block for every returned key (a per-key arena alloc was ~30ms at
+     * 1M-key batches) */
+    uint8_t *kblock = (uint8_t *)a->alloc(key_offs[n_keys]);
+    uint64_t m = 0, kb = 0;
     for (uint64_t i = 0; i < n_keys; i++) {
         if (status[i] != RRDB_OK)
             continue; /* NotFound/expired skipped (on_batch_get:952-965) */
         uint64_t kl = key_offs[i + 1] - key_offs[i];
-        uint8_t *kc = (uint8_t *)a->alloc(kl);
-        memcpy(kc, keys + key_offs[i], kl);
-        out->keys[m] = {kc, kl};
+        memcpy(kblock + kb, keys + key_offs[i], kl);
+        out->keys[m] = {kblock + kb, kl};
         out->values[m] = {(uint8_t *)vals[i].first, vals[i].second};
+        kb += kl;
         m++;
     }
     out->count = m;
@@ -1192,8 +1226,8 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
             uint8_t *kb = (uint8_t *)a->alloc(kbytes);
             uint8_t *vb = (uint8_t *)a->alloc(vbytes);
             std::vector<uint64_t> kooffs(n_out + 1), vooffs(n_out + 1);
-            HIP_OK(hipMemcpyAsync(kb, d_kout, kbytes, hipMemcpyDeviceToHost, e->stream));
-            HIP_OK(hipMemcpyAsync(vb, d_vout, vbytes, hipMemcpyDeviceToHost, e->stream));
+            e->d2h(kb, d_kout, kbytes);
+            e->d2h(vb, d_vout, vbytes);
             HIP_OK(hipMemcpyAsync(kooffs.data(), d_kooffs, (n_out + 1) * 8,
                                   hipMemcpyDeviceToHost, e->stream));
             HIP_OK(hipMemcpyAsync(vooffs.data(), d_vooffs, (n_out + 1) * 8,
