@@ -84,7 +84,7 @@ def main():
     ap.add_argument("--cpu-sample", type=int, default=2_000_000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
-    ap.add_argument("--bt-shift", type=int, default=8)
+    ap.add_argument("--bt-shift", type=int, default=6)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))

@@ -511,7 +511,7 @@ struct HipEngine {
     int emit_mode = 2; /* 2 = chunked (default; 2.8 TB/s on the copy probe),
                           0 = rank-major waves, 1 = input-major waves
                           (env "engine.emit_mode": chunked|rank|input) */
-    int bt_shift = 8;  /* bound-table block = 1<<bt_shift records (env
+    int bt_shift = 6;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
