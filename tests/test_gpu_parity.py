@@ -321,3 +321,41 @@ def test_env_caps_parity(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+def test_compact_stale_split_parity(oracle_lib, hip_lib):
+    mask, pidx = 3, 1
+    o = oracle_lib.open(1, pidx, -1)
+    g = hip_lib.open(1, pidx, 0)
+    try:
+        recs = [(D.generate_key(f"sp{i:03d}".encode(), b""),
+                 D.encode_value(b"v", 0, i + 1, 1), i + 1, 0) for i in range(64)]
+        for p in (o, g):
+            p.set_envs({"replica.split.validate_partition_hash": "true"})
+            p.set_partition_version(mask)
+            p.ingest_run(recs)
+        so = o.manual_compact(100)
+        sg = g.manual_compact(100)
+        assert so == sg
+        assert _drain(o, 100, validate_partition_hash=False) == \
+               _drain(g, 100, validate_partition_hash=False)
+    finally:
+        o.close()
+        g.close()
+
+
+def test_default_ttl_v2_parity(oracle_lib, hip_lib):
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        now = 7000
+        raw = D.generate_key(b"v2k", b"")
+        for p in (o, g):
+            p.set_envs({"pegasus.data_version": "2", "default_ttl": "100"})
+            p.ingest_run([(raw, D.encode_value(b"data", 0, 5, 2), 1, 0)])
+            p.manual_compact(now)
+        assert o.ttl(raw, now) == g.ttl(raw, now) == (OK, 100)
+        assert o.get(raw, now) == g.get(raw, now)
+    finally:
+        o.close()
+        g.close()

@@ -133,3 +133,57 @@ def test_scan_start_exclusive_pages_correctly(oracle_lib):
         assert got == [r[0] for r in recs[4:]]
     finally:
         p.close()
+
+
+def test_compact_stale_split_drop(oracle_lib):
+    """stale-split-hash drop during compaction (key_ttl_compaction_filter.h:114-121):
+    with validate_partition_hash on and partition_version >= pidx, keys whose
+    crc64(hashkey) & mask != pidx are filtered out of the merged run."""
+    from pymodel import crc64 as py_crc64
+
+    mask = 3
+    pidx = 1
+    p = oracle_lib.open(1, pidx, -1)
+    try:
+        p.set_envs({"replica.split.validate_partition_hash": "true"})
+        p.set_partition_version(mask)
+        recs = []
+        mine = 0
+        for i in range(64):
+            hk = f"sp{i:03d}".encode()
+            if (py_crc64(hk) & mask) == pidx:
+                mine += 1
+            recs.append((D.generate_key(hk, b""), D.encode_value(b"v", 0, i + 1, 1), i + 1, 0))
+        p.ingest_run(recs)
+        err, stats = p.manual_compact(100)
+        assert err == OK
+        assert stats.output_records == mine
+        assert stats.filtered == 64 - mine
+        # pidx > partition_version -> no drop (filter:116-118)
+        p2 = oracle_lib.open(1, 5, -1)
+        p2.set_envs({"replica.split.validate_partition_hash": "true"})
+        p2.set_partition_version(mask)
+        p2.ingest_run(recs)
+        err, stats = p2.manual_compact(100)
+        assert stats.output_records == 64
+        p2.close()
+    finally:
+        p.close()
+
+
+def test_default_ttl_v2_header_offset(oracle_lib):
+    """default-TTL rewrite patches at offset 1 for schema v2 (value_schema_v2.cpp:116-125)."""
+    p = oracle_lib.open(1, 0, -1)
+    try:
+        now = 7000
+        p.set_envs({"pegasus.data_version": "2", "default_ttl": "100"})
+        raw = D.generate_key(b"v2k", b"")
+        p.ingest_run([(raw, D.encode_value(b"data", 0, 5, 2), 1, 0)])
+        err, stats = p.manual_compact(now)
+        assert err == OK and stats.output_records == 1
+        st, ttl = p.ttl(raw, now)
+        assert (st, ttl) == (OK, 100)
+        st, v = p.get(raw, now)
+        assert (st, v) == (OK, b"data")
+    finally:
+        p.close()
