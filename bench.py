@@ -264,14 +264,82 @@ def main():
     get_elapsed = time.time() - t0
     get_ops_per_s = nq / get_elapsed
 
+    # ---- YCSB-E: prefix multi_get over a sortkey table (config 3 shape) ----
+    # single request stream through the C-ABI (per-call latency path; the
+    # reference serves these concurrently across THREAD_POOL_SCAN threads)
+    from incubator_pegasus_amd import data as D2
+
+    scan_part = hip.open(2, 0, local_rank)
+    st_run = D2.build_scan_table_run(200_000, 10, seed=D2.DEFAULT_SEED + 77)
+    scan_part.ingest_run_arrays(np.ascontiguousarray(st_run["keys"]), st_run["koff"],
+                                np.ascontiguousarray(st_run["vals"]), st_run["voff"],
+                                st_run["sk"])
+    mg_n = 2000
+    mg_ids = D2.zipfian_ids(mg_n, 200_000, seed=D2.DEFAULT_SEED + 5)
+    mg_hks = D2.make_hashkeys(mg_ids)
+    barrier_sync()
+    t0 = time.time()
+    mg_rows = 0
+    for qi in range(mg_n):
+        st, kvs = scan_part.multi_get(bytes(mg_hks[qi]), epoch_now)
+        assert st == 0
+        mg_rows += len(kvs)
+    mg_elapsed = time.time() - t0
+    scan_part.close()
+    ycsb_e = {
+        "multi_get_ops_per_s": round(mg_n / mg_elapsed, 1),
+        "rows_per_s": round(mg_rows / mg_elapsed, 1),
+        "rows_returned": mg_rows,
+        "note": "single request stream (per-call latency); server-side concurrency is round-2",
+    }
+
     # ---- CPU baseline (oracle restatement, rank 0, N==1 only) ----
     # single-core pass + a partition-parallel pass on multiple host cores
     # (partitions are independent, exactly how the reference parallelizes)
     cpu_baseline = None
+    cpu_aux = None
     if rank == 0 and world <= 1 and not args.skip_cpu_baseline:
         sample = min(args.cpu_sample, keys_pp)
         log(f"cpu baseline: oracle compaction pass over {sample} keys ...")
         nrec1, el1 = _cpu_compact_worker((sample, args.runs, 0, epoch_now))
+        # single-core oracle get/scan rates on the same sample (BASELINE.md:
+        # get ops/s + scan rows/s CPU side-by-side; config 1 shape)
+        from incubator_pegasus_amd.capi import RrdbLib as _RL
+
+        _orc = _RL(os.path.join(REPO, "oracle", "liboracle.so"))
+        op = _orc.open(3, 0, -1)
+        op.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
+        for r in build_partition_data(sample, args.runs, 0, 0):
+            op.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
+                                 np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
+        t0 = time.time()
+        res = op.scan_open(b"\x00\x00", b"\xff\xff", epoch_now, only_return_count=True,
+                           full_scan=True, validate_partition_hash=False, batch_size=2**31 - 1)
+        cpu_scan_rows = res.kv_count
+        cpu_scan_el = time.time() - t0
+        nq_cpu = 100_000
+        from incubator_pegasus_amd import data as D3
+
+        q_ids = D3.zipfian_ids(nq_cpu, sample, seed=D3.DEFAULT_SEED)
+        qk = D3.make_raw_keys(q_ids).reshape(-1)
+        qo = D3.fixed_offsets(nq_cpu, 18)
+        import ctypes as _C
+
+        from incubator_pegasus_amd.capi import _Result as _R
+
+        t0 = time.time()
+        r_ = _R()
+        _orc._lib.rrdb_batch_get(op._h, nq_cpu, np.ascontiguousarray(qk).ctypes.data_as(_C.c_void_p),
+                                 qo.ctypes.data_as(_C.c_void_p), epoch_now, _C.byref(r_))
+        cpu_get_el = time.time() - t0
+        _orc._lib.rrdb_free_result(_C.byref(r_))
+        op.close()
+        cpu_aux = {
+            "scan_rows_per_s": round(cpu_scan_rows / cpu_scan_el, 1),
+            "get_ops_per_s": round(nq_cpu / cpu_get_el, 1),
+            "cores": 1,
+            "sample_records": int(cpu_scan_rows),
+        }
         workers = min(os.cpu_count() or 1, args.partitions)
         log(f"cpu baseline: {workers} parallel partitions x {sample} keys ...")
         import concurrent.futures as cf
@@ -335,6 +403,8 @@ def main():
                 "get_found": found,
                 "compact_output_records_per_gpu": int(out_records),
                 "compact_call_ms": compact_call_stats,
+                "ycsb_e_prefix_multi_get": ycsb_e,
+                "cpu_single_core": cpu_aux,
             },
         }
         print(json.dumps(line))
