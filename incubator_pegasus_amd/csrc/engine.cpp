@@ -33,7 +33,8 @@ void launch_rank(const DevRun *, int, const uint64_t *, const uint64_t *, const 
                  uint64_t, uint64_t *, uint8_t *, const uint64_t *, const uint64_t *, int,
                  hipStream_t);
 void launch_bound_table(const DevRun *, int, const uint64_t *, const uint64_t *,
-                        const uint64_t *, uint64_t, int, uint64_t *, hipStream_t);
+                        const uint64_t *, uint64_t, int, uint64_t *, const uint64_t *,
+                        const uint64_t *, int, hipStream_t);
 void launch_visible(const DevRun *, const uint64_t *, const uint8_t *, uint64_t, uint64_t *,
                     hipStream_t);
 void launch_gather(const uint64_t *, const uint64_t *, const uint64_t *, uint64_t, uint64_t *,
@@ -642,23 +643,42 @@ struct HipEngine {
     void tfree(void *) {} /* arena memory is reclaimed at scratch_reset */
     /* build the search bound table for windows lo/hi (BT_SHIFT=8 blocks);
      * returns device row-offset + table pointers (arena memory) */
-    void build_bound_table(DevRun *dr, int R, const std::vector<uint64_t> &lo,
-                           const std::vector<uint64_t> &hi, const uint64_t *d_lo,
-                           const uint64_t *d_hi, uint64_t **out_bt_off, uint64_t **out_bt)
+    void build_bound_table_level(DevRun *dr, int R, const std::vector<uint64_t> &lo,
+                                 const std::vector<uint64_t> &hi, const uint64_t *d_lo,
+                                 const uint64_t *d_hi, int shift, const uint64_t *cbt_off,
+                                 const uint64_t *cbt, int cbt_shift, uint64_t **out_bt_off,
+                                 uint64_t **out_bt)
     {
         std::vector<uint64_t> bt_off(R + 1);
         uint64_t rows = 0;
         for (int r = 0; r < R; r++) {
             bt_off[r] = rows;
             uint64_t w = hi[r] - lo[r];
-            rows += (w >> bt_shift) + 2; /* P_r + sentinel */
+            rows += (w >> shift) + 2; /* P_r + sentinel */
         }
         bt_off[R] = rows;
         uint64_t *d_bt_off = (uint64_t *)upload_tmp(bt_off.data(), (R + 1) * 8);
         uint64_t *d_bt = talloc<uint64_t>(rows * (uint64_t)R * 8);
-        launch_bound_table(dr, R, d_lo, d_hi, d_bt_off, rows, bt_shift, d_bt, stream);
+        launch_bound_table(dr, R, d_lo, d_hi, d_bt_off, rows, shift, d_bt, cbt_off, cbt,
+                           cbt_shift, stream);
         *out_bt_off = d_bt_off;
         *out_bt = d_bt;
+    }
+
+    void build_bound_table(DevRun *dr, int R, const std::vector<uint64_t> &lo,
+                           const std::vector<uint64_t> &hi, const uint64_t *d_lo,
+                           const uint64_t *d_hi, uint64_t **out_bt_off, uint64_t **out_bt)
+    {
+        uint64_t wmax = 0;
+        for (int r = 0; r < R; r++)
+            wmax = std::max(wmax, hi[r] - lo[r]);
+        uint64_t *c_off = nullptr, *c_bt = nullptr;
+        int cshift = bt_shift + 6;
+        if ((wmax >> cshift) > 8) /* coarse level pays only at scale */
+            build_bound_table_level(dr, R, lo, hi, d_lo, d_hi, cshift, nullptr, nullptr, 0,
+                                    &c_off, &c_bt);
+        build_bound_table_level(dr, R, lo, hi, d_lo, d_hi, bt_shift, c_off, c_bt, cshift,
+                                out_bt_off, out_bt);
     }
 
     uint64_t *psum_scratch(uint64_t n)

@@ -281,7 +281,8 @@ __global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t
  * row is hi[].  bt_off[r] = row offset of run r's table. */
 __global__ void k_bound_table(const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
                               const uint64_t *bt_off, uint64_t total_rows, int bt_shift,
-                              uint64_t *bt)
+                              uint64_t *bt, const uint64_t *cbt_off /* coarse table (optional) */,
+                              const uint64_t *cbt, int cbt_shift)
 {
     for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total_rows;
          t += gridDim.x * (uint64_t)blockDim.x) {
@@ -300,14 +301,21 @@ __global__ void k_bound_table(const DevRun *runs, int R, const uint64_t *lo, con
         }
         uint64_t kl;
         const uint8_t *k = run_key(runs[r], i, &kl);
+        const uint64_t *c0 = nullptr, *c1 = nullptr;
+        if (cbt) {
+            uint64_t j2 = (i - lo[r]) >> cbt_shift;
+            c0 = cbt + (cbt_off[r] + j2) * (uint64_t)R;
+            c1 = cbt + (cbt_off[r] + j2 + 1) * (uint64_t)R;
+        }
         for (int q = 0; q < R; q++) {
             if (q == r) {
                 row[q] = i;
-            } else if (q > r) {
-                row[q] = dev_upper_bound(runs[q], k, kl, lo[q], hi[q]);
-            } else {
-                row[q] = dev_lower_bound(runs[q], k, kl, lo[q], hi[q]);
+                continue;
             }
+            uint64_t qlo = c0 ? c0[q] : lo[q];
+            uint64_t qhi = c1 ? c1[q] : hi[q];
+            row[q] = (q > r) ? dev_upper_bound(runs[q], k, kl, qlo, qhi)
+                             : dev_lower_bound(runs[q], k, kl, qlo, qhi);
         }
     }
 }
@@ -1091,10 +1099,12 @@ void launch_rank(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64
 
 void launch_bound_table(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
                         const uint64_t *d_bt_off, uint64_t total_rows, int bt_shift,
-                        uint64_t *d_bt, hipStream_t s)
+                        uint64_t *d_bt, const uint64_t *d_cbt_off, const uint64_t *d_cbt,
+                        int cbt_shift, hipStream_t s)
 {
     k_bound_table<<<grid_for(total_rows, BLOCK), BLOCK, 0, s>>>(d_runs, R, d_lo, d_hi, d_bt_off,
-                                                                total_rows, bt_shift, d_bt);
+                                                                total_rows, bt_shift, d_bt,
+                                                                d_cbt_off, d_cbt, cbt_shift);
 }
 
 void launch_visible(const DevRun *d_runs, const uint64_t *d_order, const uint8_t *d_shadowed,
