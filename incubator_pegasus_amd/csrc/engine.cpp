@@ -801,6 +801,8 @@ std::once_flag g_init_once;
 
 } // namespace
 
+static int32_t engine_flush(HipEngine *e); /* write path, defined below */
+
 /* internal: upload a prepared sorted run (validation already established) */
 static void ingest_prepared(HipEngine *e, const std::string &keys,
                             const std::vector<uint64_t> &koff, const std::string &vals,
@@ -937,7 +939,8 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
                         uint64_t n)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* buffered writes are older than this run */
+    std::lock_guard<std::mutex> g(e->mu);
+    engine_flush(e); /* buffered writes are older than this run */
     if (n == 0)
         return RRDB_OK;
     /* validate on host (data arrives host-side anyway) */
@@ -1042,7 +1045,8 @@ int32_t rrdb_get(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
                  rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     Arena *a = result_init(out);
     uint64_t offs[2] = {0, key_len};
     std::vector<int32_t> status;
@@ -1065,7 +1069,8 @@ int32_t rrdb_batch_get(void *h, uint64_t n_keys, const uint8_t *keys, const uint
                        uint32_t epoch_now, rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     Arena *a = result_init(out);
     if (n_keys == 0) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_batch_get:922-928 */
@@ -1099,7 +1104,8 @@ int32_t rrdb_ttl(void *h, const uint8_t *key, uint64_t key_len, uint32_t epoch_n
                  rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
@@ -1302,7 +1308,8 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
 int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
@@ -1365,7 +1372,6 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
     if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
         delete ctx;
     } else {
-        std::lock_guard<std::mutex> g(e->mu);
         ctx->id = ++e->next_ctx_id;
         e->ctxs[ctx->id] = ctx;
         out->context_id = ctx->id;
@@ -1376,12 +1382,12 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
 int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
     Arena *a = result_init(out);
     e->activate();
     e->scratch_reset();
     HipScanCtx *c = nullptr;
     {
-        std::lock_guard<std::mutex> g(e->mu);
         auto it = e->ctxs.find(context_id);
         if (it != e->ctxs.end()) {
             c = it->second;
@@ -1396,7 +1402,6 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
     if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
         delete c;
     } else {
-        std::lock_guard<std::mutex> g(e->mu);
         c->id = ++e->next_ctx_id; /* re-park under a fresh handle (on_scan:1516-1526) */
         e->ctxs[c->id] = c;
         out->context_id = c->id;
@@ -1420,7 +1425,8 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
                            rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     ((HipEngine *)h)->scratch_reset();
     result_init(out);
     e->activate();
@@ -1469,7 +1475,8 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
                        rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
@@ -1708,7 +1715,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
                             rrdb_compact_stats *stats)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h); /* memtable visible to reads */
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    engine_flush((HipEngine *)h); /* memtable visible to reads */
     ((HipEngine *)h)->scratch_reset();
     rrdb_compact_stats st{};
     if (e->manual_compact_disabled) {
@@ -1905,6 +1913,7 @@ int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t
                  uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts)
 {
     auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
     std::string key = make_key(hash_key, hklen, sort_key, sklen);
     uint32_t hdr = hdr_len(e->data_version);
     std::string val(hdr + vlen, '\0');
@@ -1925,6 +1934,7 @@ int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint
                     uint64_t sklen)
 {
     auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
     std::string key = make_key(hash_key, hklen, sort_key, sklen);
     e->memtable[key] = {std::string(), e->next_seq_floor++, RRDB_KIND_DELETE};
     return RRDB_OK;
@@ -1932,9 +1942,8 @@ int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint
 
 uint64_t rrdb_memtable_entries(void *h) { return ((HipEngine *)h)->memtable.size(); }
 
-int32_t rrdb_flush(void *h)
+static int32_t engine_flush(HipEngine *e)
 {
-    auto *e = (HipEngine *)h;
     if (e->memtable.empty())
         return RRDB_OK;
     e->activate();
@@ -1950,6 +1959,13 @@ int32_t rrdb_flush(void *h)
     ingest_prepared(e, keys, koff, vals, voff, sk);
     e->memtable.clear();
     return RRDB_OK;
+}
+
+int32_t rrdb_flush(void *h)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    return engine_flush(e);
 }
 
 } /* extern "C" */
@@ -1987,7 +2003,8 @@ static bool read_file(const std::string &path, std::vector<uint8_t> &out)
 int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
 {
     auto *e = (HipEngine *)h;
-    rrdb_flush(h);
+    std::lock_guard<std::mutex> g(e->mu);
+    engine_flush(e);
     e->activate();
     std::string path = std::string(dir) + "/checkpoint." + std::to_string(decree);
     mkdir(dir, 0755);
@@ -2031,6 +2048,7 @@ int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
 int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
 {
     auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
     if (!e->runs.empty() || !e->memtable.empty())
         return RRDB_INVALID_ARGUMENT;
     e->activate();
