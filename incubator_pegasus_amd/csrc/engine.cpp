@@ -59,6 +59,7 @@ void launch_emit_scan(const DevRun *, const uint64_t *, uint64_t, const uint8_t 
 void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64_t,
                       const uint64_t *, const uint64_t *, const ScanParams &, uint8_t *,
                       uint8_t *, hipStream_t);
+void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
 void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
@@ -1570,6 +1571,64 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     std::string stop_excl = stop;
     if (stop_inclusive)
         stop_excl.push_back('\0');
+
+    /* ---- fused single-launch fast path (small ranges, the YCSB-E shape) ---- */
+    if (!e->runs.empty()) {
+        MgFusedArgs fa{};
+        fa.start = e->upload_tmp(start.data(), start.size());
+        fa.start_len = start.size();
+        fa.stop = e->upload_tmp(stop_excl.data(), stop_excl.size());
+        fa.stop_len = stop_excl.size();
+        fa.start_inclusive = start_inclusive;
+        fa.stop_inclusive = stop_inclusive;
+        fa.reverse = q->reverse;
+        fa.no_value = q->no_value;
+        fa.max_kv_count = max_kv_count;
+        fa.max_iteration_count = max_iteration_count;
+        fa.max_iteration_size = max_iter_size;
+        fa.sk_ft = q->sort_key_filter_type;
+        fa.sk_pat_len = q->sort_key_filter_pattern.len;
+        fa.sk_pat = e->upload_tmp(q->sort_key_filter_pattern.data, fa.sk_pat_len);
+        fa.epoch_now = epoch_now;
+        fa.data_version = e->data_version;
+        fa.hash_key_skip = 2 + q->hash_key.len;
+        fa.out_hdr = e->talloc<int64_t>(4 * 8);
+        fa.out_koff = e->talloc<uint64_t>((MG_MAX_ROWS + 1) * 8);
+        fa.out_voff = e->talloc<uint64_t>((MG_MAX_ROWS + 1) * 8);
+        fa.out_keys = e->talloc<uint8_t>(MG_SCRATCH_BYTES);
+        fa.out_vals = e->talloc<uint8_t>(MG_SCRATCH_BYTES);
+        launch_multi_get_small(e->dev_runs(), (int)e->runs.size(), fa, e->stream);
+        int64_t hdr4[4];
+        HIP_OK(hipMemcpyAsync(hdr4, fa.out_hdr, 32, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        if (hdr4[0] >= 0) {
+            uint64_t m = (uint64_t)hdr4[0];
+            uint64_t kb = (uint64_t)hdr4[2], vb = (uint64_t)hdr4[3];
+            out->keys = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
+            out->values = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
+            if (m) {
+                std::vector<uint64_t> koffs(m + 1), voffs(m + 1);
+                uint8_t *hk = (uint8_t *)a->alloc(kb);
+                uint8_t *hv = (uint8_t *)a->alloc(vb);
+                HIP_OK(hipMemcpyAsync(koffs.data(), fa.out_koff, (m + 1) * 8,
+                                      hipMemcpyDeviceToHost, e->stream));
+                HIP_OK(hipMemcpyAsync(voffs.data(), fa.out_voff, (m + 1) * 8,
+                                      hipMemcpyDeviceToHost, e->stream));
+                HIP_OK(hipMemcpyAsync(hk, fa.out_keys, kb, hipMemcpyDeviceToHost, e->stream));
+                HIP_OK(hipMemcpyAsync(hv, fa.out_vals, vb, hipMemcpyDeviceToHost, e->stream));
+                HIP_OK(hipStreamSynchronize(e->stream));
+                for (uint64_t j = 0; j < m; j++) {
+                    out->keys[j] = {hk + koffs[j], koffs[j + 1] - koffs[j]};
+                    out->values[j] = {hv + voffs[j], voffs[j + 1] - voffs[j]};
+                }
+            }
+            out->count = m;
+            out->error = hdr4[1] ? RRDB_OK : RRDB_INCOMPLETE;
+            return out->error;
+        }
+        /* fallback: large range or oversize rows — general path below */
+    }
+
     uint64_t *d_view = nullptr, n = 0;
     e->build_view(&start, &stop_excl, &d_view, &n);
     if (n == 0) {

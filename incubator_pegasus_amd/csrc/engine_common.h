@@ -89,6 +89,33 @@ struct CompactParams {
     const uint8_t *pats;  /* device pattern blob */
 };
 
+
+/* fused small-multi_get argument block (one-launch YCSB-E path) */
+#define MG_MAX_ROWS 4096
+#define MG_SCRATCH_BYTES (1 << 20)
+struct MgFusedArgs {
+    const uint8_t *start; /* full rocksdb keys (device) */
+    uint64_t start_len;
+    const uint8_t *stop;
+    uint64_t stop_len;
+    uint8_t start_inclusive, stop_inclusive, reverse, no_value;
+    uint32_t max_kv_count;
+    uint32_t max_iteration_count;
+    int64_t max_iteration_size;
+    int32_t sk_ft;
+    const uint8_t *sk_pat;
+    uint64_t sk_pat_len;
+    uint32_t epoch_now;
+    uint32_t data_version;
+    uint64_t hash_key_skip;
+    /* outputs: [4] hdr = n_rows(-1 fallback), complete, kbytes, vbytes */
+    int64_t *out_hdr;
+    uint64_t *out_koff;
+    uint64_t *out_voff;
+    uint8_t *out_keys;
+    uint8_t *out_vals;
+};
+
 /* compact per-record disposition written by the filter kernel */
 struct CompactStatsDev {
     unsigned long long expired, filtered, tombstones, shadowed, output_records;

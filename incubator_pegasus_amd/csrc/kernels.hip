@@ -1290,3 +1290,218 @@ void launch_emit_compact(const DevRun *d_runs, const uint64_t *d_order, uint64_t
 }
 
 } /* extern C++ */
+
+/* ================= fused small multi_get =================
+ * One launch for the whole on_multi_get range path when the hashkey's row
+ * count is small (the common YCSB-E case): per-run bounds, in-window rank
+ * merge, TTL/sortkey-filter, limiter caps (count/iteration/size,
+ * forward/reverse, first-exclusive) and output packing — one block, one
+ * result blob, no intermediate syncs.  Semantics mirror
+ * on_multi_get:540-778 exactly; ranges larger than the scratch budget set
+ * out_hdr[0]=-1 and the host falls back to the general path. */
+__global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int R,
+                                                         MgFusedArgs a)
+{
+    __shared__ uint64_t s_lo[RRDB_MAX_RUNS], s_hi[RRDB_MAX_RUNS], s_wp[RRDB_MAX_RUNS + 1];
+    __shared__ uint64_t s_total;
+    __shared__ uint32_t s_state[MG_MAX_ROWS];   /* 0 normal,1 skip,2 tomb/shadow */
+    __shared__ uint16_t s_klen[MG_MAX_ROWS];
+    __shared__ uint16_t s_vlen_lo[MG_MAX_ROWS]; /* value len (<=64KB supported in fused path) */
+    __shared__ uint64_t s_id[MG_MAX_ROWS];      /* (run<<40)|idx by rank */
+    int tid = threadIdx.x;
+    /* phase 1: bounds */
+    if (tid < R) {
+        s_lo[tid] = dev_lower_bound(runs[tid], a.start, a.start_len, 0, runs[tid].n);
+        s_hi[tid] = dev_lower_bound(runs[tid], a.stop, a.stop_len, 0, runs[tid].n);
+        if (s_hi[tid] < s_lo[tid])
+            s_hi[tid] = s_lo[tid];
+    }
+    __syncthreads();
+    if (tid == 0) {
+        uint64_t t = 0;
+        for (int r = 0; r < R; r++) {
+            s_wp[r] = t;
+            t += s_hi[r] - s_lo[r];
+        }
+        s_wp[R] = t;
+        s_total = t;
+        if (t > MG_MAX_ROWS)
+            a.out_hdr[0] = -1; /* fallback */
+    }
+    __syncthreads();
+    uint64_t total = s_total;
+    if (total > MG_MAX_ROWS)
+        return;
+    /* phase 2: rank + visibility + per-row state/sizes (merged order in LDS) */
+    for (uint64_t t = tid; t < total; t += blockDim.x) {
+        int r = 0;
+        while (s_wp[r + 1] <= t)
+            r++;
+        uint64_t i = s_lo[r] + (t - s_wp[r]);
+        uint64_t kl;
+        const uint8_t *k = run_key(runs[r], i, &kl);
+        uint64_t rank = i - s_lo[r];
+        int shadow = 0;
+        for (int q = 0; q < R; q++) {
+            if (q == r)
+                continue;
+            if (q > r) {
+                uint64_t ub = dev_upper_bound(runs[q], k, kl, s_lo[q], s_hi[q]);
+                if (!shadow && ub > s_lo[q]) {
+                    uint64_t pl;
+                    const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
+                    if (dev_key_cmp(pk, pl, k, kl) == 0)
+                        shadow = 1;
+                }
+                rank += ub - s_lo[q];
+            } else {
+                rank += dev_lower_bound(runs[q], k, kl, s_lo[q], s_hi[q]) - s_lo[q];
+            }
+        }
+        uint32_t st;
+        uint16_t klen_out = 0, vlen_out = 0;
+        if (shadow || (runs[r].sk[i] & 1)) {
+            st = 2; /* invisible: not iterated at all */
+        } else {
+            uint64_t vl;
+            const uint8_t *v = run_val(runs[r], i, &vl);
+            uint32_t hdr = dev_hdr_len(a.data_version);
+            if (dev_ts_expired(a.epoch_now, dev_expire_ts(a.data_version, v))) {
+                st = 1; /* kExpired: iterated but skipped */
+            } else {
+                const uint8_t *skp = k + a.hash_key_skip;
+                uint64_t sklen = kl - a.hash_key_skip;
+                if (a.sk_ft != 0 &&
+                    !dev_validate_filter(a.sk_ft, a.sk_pat, a.sk_pat_len, skp, sklen)) {
+                    st = 1; /* kFiltered */
+                } else if (kl - a.hash_key_skip > 0xFFFF || vl - hdr > 0xFFFF) {
+                    st = 3; /* oversize for fused path -> fallback */
+                } else {
+                    st = 0;
+                    klen_out = (uint16_t)(kl - a.hash_key_skip);
+                    vlen_out = a.no_value ? 0 : (uint16_t)(vl - hdr);
+                }
+            }
+        }
+        s_state[t] = st;
+        s_klen[t] = klen_out;
+        s_vlen_lo[t] = vlen_out;
+        s_id[t] = ((uint64_t)r << 40) | i;
+    }
+    __syncthreads();
+    /* phase 3 (thread 0): limiter walk, exactly on_multi_get:616-778 */
+    __shared__ uint64_t s_sel[MG_MAX_ROWS]; /* selected merged positions */
+    __shared__ uint64_t s_nsel, s_kbytes, s_vbytes, s_complete;
+    if (tid == 0) {
+        int fallback = 0;
+        uint64_t nsel = 0, kb = 0, vb = 0;
+        uint64_t count = 0, iteration = 0;
+        int64_t size = 0;
+        int complete = 0;
+        uint64_t iterated_total = 0, countable = 0;
+        /* countable = entries the reference iterator would visit (visible) */
+        for (uint64_t t = 0; t < total; t++)
+            if (s_state[t] != 2 && s_state[t] != 3)
+                countable++;
+            else if (s_state[t] == 3)
+                fallback = 1;
+        int skipped_first = 0;
+        uint64_t visited = 0;
+        for (uint64_t s = 0; s < total && !fallback; s++) {
+            uint64_t t = a.reverse ? (total - 1 - s) : s;
+            if (s_state[t] == 2)
+                continue; /* invisible to the iterator */
+            if (count >= a.max_kv_count || iteration >= a.max_iteration_count ||
+                size >= a.max_iteration_size)
+                break;
+            if (visited == 0) {
+                /* first-exclusive boundary skip (:636-643 / :700-707) */
+                int check = (!a.reverse && !a.start_inclusive) ||
+                            (a.reverse && !a.stop_inclusive);
+                if (check) {
+                    uint64_t id = s_id[t];
+                    const DevRun &r = runs[id >> 40];
+                    uint64_t i = id & 0xFFFFFFFFFFull, kl;
+                    const uint8_t *k = run_key(r, i, &kl);
+                    const uint8_t *bound = a.reverse ? a.stop : a.start;
+                    uint64_t blen = a.reverse ? a.stop_len : a.start_len;
+                    visited = 1;
+                    if (dev_key_cmp(k, kl, bound, blen) == 0) {
+                        skipped_first = 1;
+                        continue;
+                    }
+                } else {
+                    visited = 1;
+                }
+            }
+            iteration++;
+            if (s_state[t] == 1)
+                continue;
+            s_sel[nsel++] = t;
+            count++;
+            kb += s_klen[t];
+            vb += s_vlen_lo[t];
+            size += (int64_t)s_klen[t] + (int64_t)s_vlen_lo[t];
+        }
+        complete = (iteration >= countable - (uint64_t)skipped_first);
+        if (kb > MG_SCRATCH_BYTES || vb > MG_SCRATCH_BYTES)
+            fallback = 1;
+        if (fallback) {
+            a.out_hdr[0] = -1;
+            s_nsel = (uint64_t)-1;
+        } else {
+            /* ascending output (reverse selected from the top) */
+            if (a.reverse) {
+                for (uint64_t x = 0; x < nsel / 2; x++) {
+                    uint64_t tmp = s_sel[x];
+                    s_sel[x] = s_sel[nsel - 1 - x];
+                    s_sel[nsel - 1 - x] = tmp;
+                }
+            }
+            uint64_t ko = 0, vo = 0;
+            for (uint64_t x = 0; x < nsel; x++) {
+                a.out_koff[x] = ko;
+                a.out_voff[x] = vo;
+                ko += s_klen[s_sel[x]];
+                vo += s_vlen_lo[s_sel[x]];
+            }
+            a.out_koff[nsel] = ko;
+            a.out_voff[nsel] = vo;
+            a.out_hdr[0] = (int64_t)nsel;
+            a.out_hdr[1] = complete;
+            a.out_hdr[2] = (int64_t)ko;
+            a.out_hdr[3] = (int64_t)vo;
+            s_nsel = nsel;
+            s_kbytes = ko;
+            s_vbytes = vo;
+            s_complete = complete;
+        }
+    }
+    __syncthreads();
+    if (s_nsel == (uint64_t)-1)
+        return;
+    /* phase 4: copy selected rows (wave per row) */
+    uint64_t nsel = s_nsel;
+    int lane = tid % WAVE;
+    int wid = tid / WAVE;
+    int nw = blockDim.x / WAVE;
+    uint32_t hdr = dev_hdr_len(a.data_version);
+    for (uint64_t x = wid; x < nsel; x += nw) {
+        uint64_t t = s_sel[x];
+        uint64_t id = s_id[t];
+        const DevRun &r = runs[id >> 40];
+        uint64_t i = id & 0xFFFFFFFFFFull;
+        uint64_t kl, vl;
+        const uint8_t *k = run_key(r, i, &kl);
+        wave_copy(a.out_keys + a.out_koff[x], k + a.hash_key_skip, s_klen[t], lane);
+        if (!a.no_value && s_vlen_lo[t]) {
+            const uint8_t *v = run_val(r, i, &vl);
+            wave_copy(a.out_vals + a.out_voff[x], v + hdr, s_vlen_lo[t], lane);
+        }
+    }
+}
+
+void launch_multi_get_small(const DevRun *d_runs, int R, const MgFusedArgs &a, hipStream_t s)
+{
+    k_multi_get_small<<<1, 256, 0, s>>>(d_runs, R, a);
+}
