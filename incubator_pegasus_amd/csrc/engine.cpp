@@ -51,7 +51,6 @@ void launch_normal_flags(const uint8_t *, uint64_t, uint64_t *, hipStream_t);
 void launch_cutoff(const uint64_t *, const uint64_t *, uint64_t, uint64_t, uint64_t *,
                    hipStream_t);
 void launch_cut_sizes(const uint8_t *, uint64_t, uint64_t, uint64_t *, uint64_t *, hipStream_t);
-void launch_widen_u8(const uint8_t *, uint64_t, uint64_t *, hipStream_t);
 void launch_emit_scan(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                       const uint64_t *, const uint64_t *, const uint64_t *, const ScanParams &,
                       uint8_t *, uint8_t *, uint64_t *, uint64_t *, int32_t *, uint64_t,

@@ -547,14 +547,6 @@ __global__ void k_normal_flags(const uint8_t *state, uint64_t w, uint64_t *flags
         flags[p] = (state[p] == ST_NORMAL) ? 1 : 0;
 }
 
-/* widen u8 0/1 flags to u64 (for the prefix-sum) */
-__global__ void k_widen_u8(const uint8_t *in, uint64_t n, uint64_t *out)
-{
-    for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < n;
-         p += gridDim.x * (uint64_t)blockDim.x)
-        out[p] = in[p];
-}
-
 /* zero sizes for entries at/after consumed cutoff or non-normal */
 __global__ void k_cut_sizes(const uint8_t *state, uint64_t w, uint64_t consumed, uint64_t *ksz,
                             uint64_t *vsz)
@@ -1186,11 +1178,6 @@ void launch_cutoff(const uint64_t *d_nprefix, const uint64_t *d_flags, uint64_t 
                    uint64_t batch_count, uint64_t *d_out2, hipStream_t s)
 {
     k_cutoff<<<1, 1, 0, s>>>(d_nprefix, d_flags, w, batch_count, d_out2);
-}
-
-void launch_widen_u8(const uint8_t *d_in, uint64_t n, uint64_t *d_out, hipStream_t s)
-{
-    k_widen_u8<<<grid_for(n, BLOCK), BLOCK, 0, s>>>(d_in, n, d_out);
 }
 
 void launch_cut_sizes(const uint8_t *d_state, uint64_t w, uint64_t consumed, uint64_t *d_ksz,
