@@ -1370,10 +1370,11 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
                 }
             }
         }
-        s_state[t] = st;
-        s_klen[t] = klen_out;
-        s_vlen_lo[t] = vlen_out;
-        s_id[t] = ((uint64_t)r << 40) | i;
+        /* indexed by RANK: the limiter walk below runs in merged order */
+        s_state[rank] = st;
+        s_klen[rank] = klen_out;
+        s_vlen_lo[rank] = vlen_out;
+        s_id[rank] = ((uint64_t)r << 40) | i;
     }
     __syncthreads();
     /* phase 3 (thread 0): limiter walk, exactly on_multi_get:616-778 */
