@@ -359,3 +359,32 @@ def test_default_ttl_v2_parity(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+def test_multi_get_fused_fallback_boundary(oracle_lib, hip_lib):
+    """ranges around MG_MAX_ROWS (4096) cross between the fused single-launch
+    path and the general view path — results must be identical (and the
+    3000-iteration limiter cap applies either way)."""
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        now = 100
+        hk = b"bighk"
+        recs = [(D.generate_key(hk, f"s{i:05d}".encode()),
+                 D.encode_value(f"v{i}".encode(), 0, i + 1, 1), i + 1, 0)
+                for i in range(5000)]
+        o.ingest_run(recs)
+        g.ingest_run(recs)
+        for kwargs in [
+            dict(),  # 5000 rows -> fallback, caps at 3000 iterations -> INCOMPLETE
+            dict(start_sortkey=b"s00100", stop_sortkey=b"s04000"),  # ~3900 rows
+            dict(start_sortkey=b"s00100", stop_sortkey=b"s04196"),  # = 4096 rows
+            dict(start_sortkey=b"s00000", stop_sortkey=b"s04000", reverse=True),
+            dict(start_sortkey=b"s01000", stop_sortkey=b"s01100"),  # fused
+        ]:
+            ro = o.multi_get(hk, now, **kwargs)
+            rg = g.multi_get(hk, now, **kwargs)
+            assert ro == rg, kwargs
+    finally:
+        o.close()
+        g.close()
