@@ -85,6 +85,7 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
     ap.add_argument("--bt-shift", type=int, default=6)
+    ap.add_argument("--compact-threads", type=int, default=4)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -140,13 +141,21 @@ def main():
             dist.barrier()
         torch.cuda.synchronize()
 
+    # partitions are independent engines with independent HIP streams; ctypes
+    # releases the GIL during the C call, so a small thread pool overlaps the
+    # per-call host work and the tail of each partition's kernels
+    import concurrent.futures as _cf
+
+    pool = _cf.ThreadPoolExecutor(max_workers=args.compact_threads)
+
     def one_step():
-        out_records = 0
-        for eng in parts:
+        def run(eng):
             err, st = eng.manual_compact(epoch_now, keep_inputs=True)
             assert err == 0
-            out_records += st.output_records
-        return out_records, st
+            return st
+
+        stats = list(pool.map(run, parts))
+        return sum(s.output_records for s in stats), stats[-1]
 
     # ---- warmup ----
     for _ in range(args.warmup):
