@@ -86,6 +86,7 @@ def main():
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
     ap.add_argument("--bt-shift", type=int, default=6)
     ap.add_argument("--compact-threads", type=int, default=4)
+    ap.add_argument("--rank-mode", choices=["global", "lds"], default="global")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -122,7 +123,8 @@ def main():
         runs = build_partition_data(keys_pp, args.runs, rank, p)
         eng = hip.open(1, p, local_rank)
         eng.set_envs({"engine.emit_mode": args.emit_mode,
-                      "engine.bt_shift": str(args.bt_shift)})
+                      "engine.bt_shift": str(args.bt_shift),
+                      "engine.rank_mode": args.rank_mode})
         for r in runs:
             eng.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
                                   np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])

@@ -59,6 +59,11 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       const uint64_t *, const uint64_t *, const ScanParams &, uint8_t *,
                       uint8_t *, hipStream_t);
 void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
+void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64_t *,
+                             const uint64_t *, uint64_t, const CompactParams &,
+                             const uint64_t *, const uint64_t *, uint64_t *, uint64_t *,
+                             uint8_t *, uint32_t *, uint64_t *, uint64_t *, CompactStatsDev *,
+                             hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
 void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
@@ -512,6 +517,8 @@ struct HipEngine {
     int emit_mode = 2; /* 2 = chunked (default; 2.8 TB/s on the copy probe),
                           0 = rank-major waves, 1 = input-major waves
                           (env "engine.emit_mode": chunked|rank|input) */
+    int rank_mode = 0; /* 0 = global searches + bound-table narrowing,
+                          1 = LDS-staged block rank (env "engine.rank_mode") */
     int bt_shift = 6;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
 
@@ -923,6 +930,8 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
             e->mg_max_iter_count = (uint32_t)atoll(v.c_str());
         } else if (k == "rocksdb.multi_get_max_iteration_size") {
             e->mg_max_iter_size = (uint64_t)atoll(v.c_str());
+        } else if (k == "engine.rank_mode") {
+            e->rank_mode = (v == "lds") ? 1 : 0;
         } else if (k == "engine.bt_shift") {
             int s_ = atoi(v.c_str());
             if (s_ >= 4 && s_ <= 16)
@@ -1834,6 +1843,34 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     uint64_t *d_rank_of = e->emit_mode == 1 ? e->talloc<uint64_t>(total * 8) : nullptr;
     CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
     HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
+    if (e->rank_mode == 1 && R > 1 && e->emit_mode == 2) {
+        /* LDS-staged block rank: shift-8 bound table + per-run block counts */
+        uint64_t *d_bt8_off = nullptr, *d_bt8 = nullptr;
+        {
+            uint64_t *c_off = nullptr, *c_bt = nullptr;
+            uint64_t wmax = 0;
+            for (int r = 0; r < R; r++)
+                wmax = std::max(wmax, hi[r] - lo[r]);
+            if ((wmax >> 14) > 8)
+                e->build_bound_table_level(dr, R, lo, hi, d_lo, d_hi, 14, nullptr, nullptr, 0,
+                                           &c_off, &c_bt);
+            e->build_bound_table_level(dr, R, lo, hi, d_lo, d_hi, 8, c_off, c_bt, 14, &d_bt8_off,
+                                       &d_bt8);
+        }
+        std::vector<uint64_t> blkp(R + 1);
+        uint64_t nb = 0;
+        for (int r = 0; r < R; r++) {
+            blkp[r] = nb;
+            nb += (hi[r] - lo[r] + 255) >> 8;
+        }
+        blkp[R] = nb;
+        uint64_t *d_blkp = (uint64_t *)e->upload_tmp(blkp.data(), (R + 1) * 8);
+        HIP_OK(hipEventRecord(ev[0], e->stream));
+        launch_rank_compact_lds(dr, R, d_lo, d_hi, d_blkp, nb, cp, d_bt8_off, d_bt8, d_order,
+                                d_keepw, d_changed, d_new_expire, d_ksz, d_vsz, d_stats,
+                                e->stream);
+        HIP_OK(hipEventRecord(ev[1], e->stream));
+    } else {
     uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
     if (R > 1 && total > 100000)
         e->build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
@@ -1842,6 +1879,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
                         d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, e->bt_shift,
                         d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
+    }
     HIP_OK(hipEventRecord(ev[2], e->stream));
     launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);
     launch_psum(d_ksz, d_koffs, total, e->psum_scratch(total), e->stream);

@@ -1493,3 +1493,206 @@ void launch_multi_get_small(const DevRun *d_runs, int R, const MgFusedArgs &a, h
 {
     k_multi_get_small<<<1, 256, 0, s>>>(d_runs, R, a);
 }
+
+/* ================= LDS-staged rank (compaction) =================
+ * One 256-thread workgroup ranks one 256-record block of one run.  The
+ * block's search windows in every run (from the shift-8 bound table) are
+ * staged into LDS once — packed key bytes + local offsets — and all
+ * cross-run binary searches then run on-chip.  Workgroups whose windows
+ * exceed the LDS budget (heavy skew) fall back to global searches inline;
+ * results are identical either way. */
+#define LRK_BLK 256
+#define LRK_MAX_WIN 12288      /* total staged records across runs */
+#define LRK_KEY_BYTES (96 * 1024)
+
+__device__ static inline uint64_t lds_lower_bound(const uint32_t *skoff, const uint8_t *skeys,
+                                                  uint64_t lo, uint64_t hi, const uint8_t *key,
+                                                  uint64_t klen)
+{
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1;
+        const uint8_t *mk = skeys + skoff[mid];
+        uint64_t ml = skoff[mid + 1] - skoff[mid];
+        if (dev_key_cmp(mk, ml, key, klen) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+__device__ static inline uint64_t lds_upper_bound(const uint32_t *skoff, const uint8_t *skeys,
+                                                  uint64_t lo, uint64_t hi, const uint8_t *key,
+                                                  uint64_t klen)
+{
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) >> 1;
+        const uint8_t *mk = skeys + skoff[mid];
+        uint64_t ml = skoff[mid + 1] - skoff[mid];
+        if (dev_key_cmp(mk, ml, key, klen) <= 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+
+__global__ void __launch_bounds__(LRK_BLK, 1) k_rank_compact_lds(
+    const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi,
+    const uint64_t *blk_prefix /* [R+1] workgroup counts */, CompactParams cp,
+    const uint64_t *bt8_off, const uint64_t *bt8 /* shift-8 bound table */, uint64_t *order,
+    uint64_t *keepw, uint8_t *changed, uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
+    CompactStatsDev *stats)
+{
+    __shared__ uint32_t skoff[LRK_MAX_WIN + RRDB_MAX_RUNS]; /* +1 per staged window */
+    __shared__ uint8_t skeys[LRK_KEY_BYTES];
+    __shared__ uint32_t s_w0[RRDB_MAX_RUNS];  /* window start (global idx) */
+    __shared__ uint32_t s_wn[RRDB_MAX_RUNS];  /* window length */
+    __shared__ uint32_t s_obase[RRDB_MAX_RUNS]; /* offset-array base */
+    __shared__ int s_fallback;
+    uint64_t wg = blockIdx.x;
+    int r = 0;
+    while (blk_prefix[r + 1] <= wg)
+        r++;
+    uint64_t j = wg - blk_prefix[r];
+    uint64_t rec0 = lo[r] + (j << 8);
+    uint64_t rec_end = rec0 + LRK_BLK < hi[r] ? rec0 + LRK_BLK : hi[r];
+    int tid = threadIdx.x;
+    /* window extents from the shift-8 table rows j and j+1 */
+    const uint64_t *b0 = bt8 + (bt8_off[r] + j) * (uint64_t)R;
+    const uint64_t *b1 = bt8 + (bt8_off[r] + j + 1) * (uint64_t)R;
+    if (tid == 0) {
+        uint64_t tot = 0, kb = 0, obase = 0;
+        s_fallback = 0;
+        for (int q = 0; q < R; q++) {
+            uint64_t w0 = (q == r) ? rec0 : b0[q];
+            uint64_t w1 = (q == r) ? rec_end : b1[q];
+            if (w1 < w0)
+                w1 = w0;
+            uint64_t wn = w1 - w0;
+            uint64_t kbq = runs[q].koff[w1] - runs[q].koff[w0];
+            s_w0[q] = (uint32_t)w0;
+            s_wn[q] = (uint32_t)wn;
+            s_obase[q] = (uint32_t)(obase + (uint64_t)q); /* +q: one extra offset per window */
+            obase += wn;
+            tot += wn;
+            kb += kbq;
+            if (wn > 0xFFFFFFF0ull || w1 > 0xFFFFFFF0ull)
+                s_fallback = 1;
+        }
+        if (tot > LRK_MAX_WIN || kb > LRK_KEY_BYTES)
+            s_fallback = 1;
+    }
+    __syncthreads();
+    int fallback = s_fallback;
+    if (!fallback) {
+        /* stage windows: offsets then packed key bytes (coalesced) */
+        uint64_t key_base = 0;
+        for (int q = 0; q < R; q++) {
+            uint64_t w0 = s_w0[q], wn = s_wn[q];
+            uint64_t gk0 = runs[q].koff[w0];
+            uint32_t ob = s_obase[q];
+            for (uint64_t t = tid; t <= wn; t += LRK_BLK)
+                skoff[ob + t] = (uint32_t)(runs[q].koff[w0 + t] - gk0 + key_base);
+            uint64_t kbq = runs[q].koff[w0 + wn] - gk0;
+            const uint8_t *src = runs[q].keys + gk0;
+            for (uint64_t b = tid * 4; b + 4 <= kbq; b += LRK_BLK * 4) {
+                uint32_t w;
+                __builtin_memcpy(&w, src + b, 4);
+                __builtin_memcpy(&skeys[key_base + b], &w, 4);
+            }
+            /* byte tail */
+            uint64_t tail = kbq & ~3ull;
+            for (uint64_t b = tail + tid; b < kbq; b += LRK_BLK)
+                skeys[key_base + b] = src[b];
+            key_base += kbq;
+        }
+        __syncthreads();
+    }
+    /* one record per thread */
+    uint64_t i = rec0 + tid;
+    int disp = D_NONE;
+    if (i < rec_end) {
+        uint64_t kl;
+        const uint8_t *k = run_key(runs[r], i, &kl);
+        uint64_t rank = i - lo[r];
+        int shadow = 0;
+        for (int q = 0; q < R; q++) {
+            if (q == r)
+                continue;
+            uint64_t pos; /* count of window records before our key */
+            if (!fallback) {
+                uint32_t ob = s_obase[q], wn = s_wn[q];
+                uint64_t p = (q > r)
+                                 ? lds_upper_bound(skoff + ob, skeys, 0, wn, k, kl)
+                                 : lds_lower_bound(skoff + ob, skeys, 0, wn, k, kl);
+                if (q > r && !shadow) {
+                    if (p > 0) {
+                        const uint8_t *pk = skeys + skoff[ob + p - 1];
+                        uint64_t pl = skoff[ob + p] - skoff[ob + p - 1];
+                        if (dev_key_cmp(pk, pl, k, kl) == 0)
+                            shadow = 1;
+                    } else if ((uint64_t)s_w0[q] > lo[q]) {
+                        /* equal key may sit just before the window */
+                        uint64_t pl;
+                        const uint8_t *pk = run_key(runs[q], s_w0[q] - 1, &pl);
+                        if (dev_key_cmp(pk, pl, k, kl) == 0)
+                            shadow = 1;
+                    }
+                }
+                pos = (uint64_t)s_w0[q] + p;
+            } else {
+                uint64_t ub = (q > r) ? dev_upper_bound(runs[q], k, kl, lo[q], hi[q])
+                                      : dev_lower_bound(runs[q], k, kl, lo[q], hi[q]);
+                if (q > r && !shadow && ub > lo[q]) {
+                    uint64_t pl;
+                    const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
+                    if (dev_key_cmp(pk, pl, k, kl) == 0)
+                        shadow = 1;
+                }
+                pos = ub;
+            }
+            rank += pos - lo[q];
+        }
+        uint8_t ch;
+        uint32_t nts;
+        uint64_t okl, ovl;
+        disp = dev_disposition(runs[r], i, cp, shadow, &ch, &nts, &okl, &ovl);
+        order[rank] = ((uint64_t)r << 40) | i;
+        keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+        changed[rank] = ch;
+        new_expire[rank] = nts;
+        ksz[rank] = okl;
+        vsz[rank] = ovl;
+    }
+    /* wave-aggregated stats */
+    int lane = tid % WAVE;
+    unsigned long long b;
+    b = __ballot(disp == D_SHADOWED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->shadowed, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_TOMBSTONE);
+    if (lane == 0 && b)
+        atomicAdd(&stats->tombstones, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_EXPIRED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->expired, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_FILTERED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->filtered, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_KEEP);
+    if (lane == 0 && b)
+        atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+}
+
+void launch_rank_compact_lds(const DevRun *d_runs, int R, const uint64_t *d_lo,
+                             const uint64_t *d_hi, const uint64_t *d_blk_prefix,
+                             uint64_t n_blocks, const CompactParams &cp,
+                             const uint64_t *d_bt8_off, const uint64_t *d_bt8,
+                             uint64_t *d_order, uint64_t *d_keepw, uint8_t *d_changed,
+                             uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
+                             CompactStatsDev *d_stats, hipStream_t s)
+{
+    k_rank_compact_lds<<<(uint32_t)n_blocks, LRK_BLK, 0, s>>>(
+        d_runs, R, d_lo, d_hi, d_blk_prefix, cp, d_bt8_off, d_bt8, d_order, d_keepw, d_changed,
+        d_new_expire, d_ksz, d_vsz, d_stats);
+}

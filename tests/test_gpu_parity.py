@@ -388,3 +388,30 @@ def test_multi_get_fused_fallback_boundary(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+@pytest.mark.parametrize("seed", range(2))
+def test_compact_parity_lds_rank(oracle_lib, hip_lib, seed):
+    """LDS-staged block rank (engine.rank_mode=lds) must match the oracle
+    exactly, including heavy-skew fallback windows."""
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        g.set_envs({"engine.rank_mode": "lds"})
+        rnd = random.Random(700 + seed)
+        now = 5000
+        _ingest_pair(o, g, _rand_runs(rnd, 5, 400, with_sortkeys=True, now=now))
+        # skewed extra run: many records under one hashkey (stresses windows)
+        skew = [(D.generate_key(b"zskew", f"s{i:05d}".encode()),
+                 D.encode_value(b"sv", 0, 100_000 + i, 1), 100_000 + i, 0)
+                for i in range(3000)]
+        o.ingest_run(skew)
+        g.ingest_run(skew)
+        so = o.manual_compact(now)
+        sg = g.manual_compact(now)
+        assert so == sg
+        assert _drain(o, now, validate_partition_hash=False) == \
+               _drain(g, now, validate_partition_hash=False)
+    finally:
+        o.close()
+        g.close()
