@@ -15,6 +15,7 @@
 
 #include <algorithm>
 #include <map>
+#include <chrono>
 #include <cstdint>
 #include <cstring>
 #include <memory>
@@ -1162,6 +1163,7 @@ struct BatchOut {
 static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb_result *out,
                            Arena *a)
 {
+    auto t_start = std::chrono::steady_clock::now();
     uint32_t batch_count = e->max_iter_count;
     if (c->batch_size > 0 && (uint32_t)c->batch_size < batch_count)
         batch_count = (uint32_t)c->batch_size;
@@ -1312,6 +1314,17 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
     out->error = RRDB_OK;
     out->context_id =
         (c->cursor >= c->view_n) ? RRDB_SCAN_CONTEXT_ID_COMPLETED : 0 /* caller parks */;
+    /* time budget (range_read_limiter.h:56-79), batch granularity: an
+     * over-budget incomplete batch returns kIncomplete, no re-park */
+    if (out->context_id != RRDB_SCAN_CONTEXT_ID_COMPLETED && e->iter_time_ms > 0) {
+        auto ms = std::chrono::duration_cast<std::chrono::milliseconds>(
+                      std::chrono::steady_clock::now() - t_start)
+                      .count();
+        if ((uint64_t)ms > e->iter_time_ms) {
+            out->error = RRDB_INCOMPLETE;
+            out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+        }
+    }
 }
 
 int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, rrdb_result *out)

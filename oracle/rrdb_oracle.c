@@ -38,6 +38,14 @@
 #include <string.h>
 #include <stdio.h>
 #include <time.h>
+#include <stdint.h>
+
+static uint64_t now_ms(void)
+{
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (uint64_t)ts.tv_sec * 1000 + (uint64_t)(ts.tv_nsec / 1000000);
+}
 
 #include "../include/rrdb_engine.h"
 
@@ -1556,6 +1564,7 @@ static int validate_for_scan(Engine *e, const ScanCtx *c, const uint8_t *k, uint
 /* one scan batch over a context; fills out and advances c->cursor */
 static void scan_batch(Engine *e, ScanCtx *c, uint32_t epoch_now, rrdb_result *out)
 {
+    uint64_t t_start = now_ms();
     arena *a = (arena *)out->_arena;
     uint32_t hdr = value_hdr_len(e->data_version);
     uint32_t batch_count = e->max_iter_count;
@@ -1612,6 +1621,21 @@ static void scan_batch(Engine *e, ScanCtx *c, uint32_t epoch_now, rrdb_result *o
     }
     if (c->cursor >= c->view_n)
         complete = 1;
+    /* time budget (range_read_limiter.h:56-79; 30s default), enforced at
+     * batch granularity: an over-budget incomplete batch returns
+     * kIncomplete and the context is not re-parked (on_get_scanner:
+     * 1345-1354) */
+    if (!complete && e->iter_time_ms > 0 && now_ms() - t_start > e->iter_time_ms) {
+        out->keys = keys;
+        out->values = vals;
+        out->expire_ts = ets;
+        out->count = m;
+        if (c->only_return_count)
+            out->i64 = count;
+        out->error = RRDB_INCOMPLETE;
+        out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED; /* caller frees ctx */
+        return;
+    }
     out->keys = keys;
     out->values = vals;
     out->expire_ts = ets;
