@@ -187,3 +187,29 @@ def test_default_ttl_v2_header_offset(oracle_lib):
         assert (st, v) == (OK, b"data")
     finally:
         p.close()
+
+
+def test_scan_time_budget_incomplete(oracle_lib):
+    """rocksdb_iteration_threshold_time_ms (range_read_limiter.h:56-79):
+    an over-budget incomplete batch returns kIncomplete and the context is
+    not re-parked.  (Batch granularity; the oracle's CPU batch over 200K
+    rows takes well over the 1ms threshold.)"""
+    p = oracle_lib.open(1, 0, -1)
+    try:
+        n = 200_000
+        recs = [(D.generate_key(b"tb", f"s{i:07d}".encode()),
+                 D.encode_value(b"v", 0, i + 1, 1), i + 1, 0) for i in range(n)]
+        p.ingest_run(recs)
+        p.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1),
+                    "replica.rocksdb_iteration_threshold_time_ms": "1"})
+        res = p.scan_open(b"\x00\x00", b"\xfe", 100, batch_size=n - 10,
+                          validate_partition_hash=False, only_return_count=True)
+        assert res.error == INCOMPLETE
+        assert res.context_id == SCAN_COMPLETED  # not re-parked
+        # with the default 30s budget the same scan completes fine
+        p.set_envs({"replica.rocksdb_iteration_threshold_time_ms": "30000"})
+        res = p.scan_open(b"\x00\x00", b"\xfe", 100, batch_size=2**31 - 1,
+                          validate_partition_hash=False, only_return_count=True)
+        assert res.error == OK and res.kv_count == n
+    finally:
+        p.close()
