@@ -414,8 +414,13 @@ class RrdbPartition:
         res = _Result()
         self._L.rrdb_scan_open(self._h, C.byref(req), epoch_now, C.byref(res))
         try:
-            self._last_scan_flags = (only_return_count, return_expire_ts, on_device_out)
-            return self._scan_result(res, only_return_count, return_expire_ts, on_device_out)
+            out = self._scan_result(res, only_return_count, return_expire_ts, on_device_out)
+            if out.context_id != SCAN_COMPLETED:
+                # flags follow the parked context (ids change on every batch)
+                self._scan_flags = getattr(self, "_scan_flags", {})
+                self._scan_flags[out.context_id] = (only_return_count, return_expire_ts,
+                                                    on_device_out)
+            return out
         finally:
             self._L.rrdb_free_result(C.byref(res))
 
@@ -423,8 +428,12 @@ class RrdbPartition:
         res = _Result()
         self._L.rrdb_scan_next(self._h, context_id, epoch_now, C.byref(res))
         try:
-            orc, ret, dev = getattr(self, "_last_scan_flags", (False, False, False))
-            return self._scan_result(res, orc, ret, dev)
+            flags = getattr(self, "_scan_flags", {})
+            orc, ret, dev = flags.pop(context_id, (False, False, False))
+            out = self._scan_result(res, orc, ret, dev)
+            if out.context_id != SCAN_COMPLETED and out.error == OK:
+                flags[out.context_id] = (orc, ret, dev)
+            return out
         finally:
             self._L.rrdb_free_result(C.byref(res))
 

@@ -194,12 +194,19 @@ def test_scan_time_budget_incomplete(oracle_lib):
     an over-budget incomplete batch returns kIncomplete and the context is
     not re-parked.  (Batch granularity; the oracle's CPU batch over 200K
     rows takes well over the 1ms threshold.)"""
+    import numpy as np
+
     p = oracle_lib.open(1, 0, -1)
     try:
-        n = 200_000
-        recs = [(D.generate_key(b"tb", f"s{i:07d}".encode()),
-                 D.encode_value(b"v", 0, i + 1, 1), i + 1, 0) for i in range(n)]
-        p.ingest_run(recs)
+        n = 2_000_000
+        ids = np.arange(n, dtype=np.uint64)
+        raw = D.make_raw_keys(ids)
+        vals = D.make_values(ids, 16)
+        p.ingest_run_arrays(np.ascontiguousarray(raw.reshape(-1)),
+                            D.fixed_offsets(n, raw.shape[1]),
+                            np.ascontiguousarray(vals.reshape(-1)),
+                            D.fixed_offsets(n, vals.shape[1]),
+                            ((ids + 1) << np.uint64(1)))
         p.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1),
                     "replica.rocksdb_iteration_threshold_time_ms": "1"})
         res = p.scan_open(b"\x00\x00", b"\xfe", 100, batch_size=n - 10,
