@@ -415,3 +415,59 @@ def test_compact_parity_lds_rank(oracle_lib, hip_lib, seed):
     finally:
         o.close()
         g.close()
+
+
+def test_scale_parity_1m(oracle_lib, hip_lib):
+    """Parity at bench-generator scale (1M keys, 8 runs + overlay): compaction
+    stats must match the oracle AND the survivor count predicted analytically
+    from the generator's splitmix selections (a size-independent property per
+    the measurement contract — the full-size check that small differential
+    tests can't give)."""
+    import numpy as np
+
+    from incubator_pegasus_amd import data as D2
+
+    n_keys, n_runs, seed = 1_000_000, 8, D2.DEFAULT_SEED + 123
+    runs = D2.build_point_table_runs(n_keys, n_runs, seed=seed, dup_fraction=0.10,
+                                     delete_fraction=0.02)
+    # analytic survivor count: overlay deletes kill their keys; everything
+    # else survives (no TTL in this config)
+    ids = np.arange(n_keys, dtype=np.uint64)
+    h = D2.splitmix64(ids + np.uint64(seed))
+    dup_sel = (D2.splitmix64(h) % np.uint64(1000)).astype(np.float64) / 1000.0
+    n_deleted = int(((dup_sel >= 0.10) & (dup_sel < 0.12)).sum())
+    expected_survivors = n_keys - n_deleted
+
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        for part in (o, g):
+            for r in runs:
+                part.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
+                                       np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
+        now = 1000
+        # spot reads pre-compaction
+        probe_ids = D2.zipfian_ids(200, n_keys, seed=seed)
+        probes = [bytes(k) for k in D2.make_raw_keys(probe_ids)]
+        for k in probes:
+            assert o.get(k, now) == g.get(k, now)
+        # full-scan count parity
+        for part in (o, g):
+            part.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
+        ro = o.scan_open(b"\x00\x00", b"\xff\xff", now, only_return_count=True,
+                         validate_partition_hash=False, batch_size=2**31 - 1)
+        rg = g.scan_open(b"\x00\x00", b"\xff\xff", now, only_return_count=True,
+                         validate_partition_hash=False, batch_size=2**31 - 1)
+        assert ro.kv_count == rg.kv_count == expected_survivors
+        # compaction stats parity + analytic check
+        so = o.manual_compact(now)
+        sg = g.manual_compact(now)
+        assert so == sg
+        assert sg[1].output_records == expected_survivors
+        assert sg[1].tombstones == n_deleted
+        # post-compaction spot reads
+        for k in probes[:50]:
+            assert o.get(k, now) == g.get(k, now)
+    finally:
+        o.close()
+        g.close()
