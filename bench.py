@@ -407,6 +407,8 @@ def main():
                 "scan_rows_total": rows,
                 "scan_state_kernel_GBs": round(scan_achieved, 1),
                 "get_ops_per_s": round(get_ops_per_s, 1),
+                "get_kernel_ops_per_s": (round(nq / (parts[0].phase_ms("get_search") * 1e-3), 1)
+                                         if parts[0].phase_ms("get_search") > 0 else None),
                 "get_found": found,
                 "compact_output_records_per_gpu": int(out_records),
                 "compact_call_ms": compact_call_stats,

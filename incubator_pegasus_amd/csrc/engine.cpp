@@ -1020,14 +1020,26 @@ static int32_t get_core(HipEngine *e, uint64_t nq, const uint8_t *keys, const ui
     uint64_t *d_hit = e->talloc<uint64_t>(nq * 8);
     uint64_t *d_ulen = e->talloc<uint64_t>(nq * 8);
     uint64_t *d_voffs = e->talloc<uint64_t>((nq + 1) * 8);
+    hipEvent_t gev[2];
+    HIP_OK(hipEventCreate(&gev[0]));
+    HIP_OK(hipEventCreate(&gev[1]));
+    HIP_OK(hipEventRecord(gev[0], e->stream));
     launch_get(dr, R, d_keys, d_offs, nq, epoch_now, e->data_version, d_status, d_hit, d_ulen,
                nullptr, e->stream);
+    HIP_OK(hipEventRecord(gev[1], e->stream));
     launch_psum(d_ulen, d_voffs, nq, e->psum_scratch(nq), e->stream);
     uint64_t last_off = 0, last_len = 0;
     HIP_OK(hipMemcpyAsync(&last_off, d_voffs + nq - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&last_len, d_ulen + nq - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(status.data(), d_status, nq * 4, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipStreamSynchronize(e->stream));
+    {
+        float ms;
+        if (hipEventElapsedTime(&ms, gev[0], gev[1]) == hipSuccess)
+            e->phase_ms["get_search"] = ms;
+        (void)hipEventDestroy(gev[0]);
+        (void)hipEventDestroy(gev[1]);
+    }
     uint64_t total = last_off + last_len;
     std::vector<uint64_t> voffs(nq + 1);
     HIP_OK(hipMemcpy(voffs.data(), d_voffs, nq * 8, hipMemcpyDeviceToHost));

@@ -27,6 +27,11 @@ struct DevRun {
     const uint64_t *voff; /* n+1 */
     const uint64_t *sk;   /* (seq<<1)|kind */
     uint64_t n;
+    /* blocked bloom filter over full keys (§8(f)3; the reference's 10-bit
+     * FullFilter equivalent, pegasus_server_impl_init.cpp:816-841): 64B
+     * blocks, 6 probe bits; null = no filter (search every run). */
+    const uint64_t *bloom;
+    uint64_t bloom_blocks;
 };
 
 /* flattened user compaction rules/ops (device-resident)

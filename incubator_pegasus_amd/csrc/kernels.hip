@@ -1696,3 +1696,54 @@ void launch_rank_compact_lds(const DevRun *d_runs, int R, const uint64_t *d_lo,
         d_runs, R, d_lo, d_hi, d_blk_prefix, cp, d_bt8_off, d_bt8, d_order, d_keepw, d_changed,
         d_new_expire, d_ksz, d_vsz, d_stats);
 }
+
+/* ================= per-run blocked bloom (point gets, §8(f)3) =================
+ * 64B blocks (one cache line), 6 probe bits from one 64-bit FNV-1a hash;
+ * ~10 bits/key like the reference's FullFilter profile
+ * (pegasus_server_impl_init.cpp:816-841).  Lossless for correctness: a
+ * negative skips the run's binary search, a positive falls through to it. */
+__device__ __host__ static inline uint64_t bloom_hash(const uint8_t *k, uint64_t n)
+{
+    uint64_t h = 0xcbf29ce484222325ull;
+    for (uint64_t i = 0; i < n; i++)
+        h = (h ^ k[i]) * 0x100000001b3ull;
+    return h;
+}
+
+__global__ void k_bloom_build(const DevRun run, uint64_t *bloom, uint64_t n_blocks)
+{
+    for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < run.n;
+         i += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t kl = run.koff[i + 1] - run.koff[i];
+        uint64_t h = bloom_hash(run.keys + run.koff[i], kl);
+        uint64_t blk = (h >> 32) % n_blocks;
+        uint64_t *base = bloom + blk * 8; /* 8 u64 = 64B block */
+        uint32_t x = (uint32_t)h;
+        for (int j = 0; j < 6; j++) {
+            uint32_t bit = (x >> (j * 5)) & 31; /* bit within word */
+            uint32_t word = ((x >> (j * 5 + 3)) ^ (x >> 27)) & 7;
+            atomicOr((unsigned long long *)&base[word], 1ull << (bit + ((x >> j) & 1) * 32));
+        }
+    }
+}
+
+__device__ static inline int bloom_maybe_has(const DevRun &r, uint64_t h)
+{
+    if (!r.bloom)
+        return 1;
+    uint64_t blk = (h >> 32) % r.bloom_blocks;
+    const uint64_t *base = r.bloom + blk * 8;
+    uint32_t x = (uint32_t)h;
+    for (int j = 0; j < 6; j++) {
+        uint32_t bit = (x >> (j * 5)) & 31;
+        uint32_t word = ((x >> (j * 5 + 3)) ^ (x >> 27)) & 7;
+        if (!(base[word] & (1ull << (bit + ((x >> j) & 1) * 32))))
+            return 0;
+    }
+    return 1;
+}
+
+void launch_bloom_build(const DevRun &run, uint64_t *d_bloom, uint64_t n_blocks, hipStream_t s)
+{
+    k_bloom_build<<<grid_for(run.n, BLOCK), BLOCK, 0, s>>>(run, d_bloom, n_blocks);
+}

@@ -435,7 +435,9 @@ def test_scale_parity_1m(oracle_lib, hip_lib):
     ids = np.arange(n_keys, dtype=np.uint64)
     h = D2.splitmix64(ids + np.uint64(seed))
     dup_sel = (D2.splitmix64(h) % np.uint64(1000)).astype(np.float64) / 1000.0
-    n_deleted = int(((dup_sel >= 0.10) & (dup_sel < 0.12)).sum())
+    # replicate the generator's float expression exactly: 0.10 + 0.02 is
+    # 0.12000000000000001, which INCLUDES dup_sel == 0.120
+    n_deleted = int(((dup_sel >= 0.10) & (dup_sel < 0.10 + 0.02)).sum())
     expected_survivors = n_keys - n_deleted
 
     o = oracle_lib.open(1, 0, -1)
