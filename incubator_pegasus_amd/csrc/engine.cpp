@@ -60,6 +60,7 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       const uint64_t *, const uint64_t *, const ScanParams &, uint8_t *,
                       uint8_t *, hipStream_t);
 void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
+void launch_bloom_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
 void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64_t *,
                              const uint64_t *, uint64_t, const CompactParams &,
                              const uint64_t *, const uint64_t *, uint64_t *, uint64_t *,
@@ -464,6 +465,8 @@ struct RunBuf {
     uint64_t *voff = nullptr;
     uint64_t *sk = nullptr;
     uint64_t n = 0;
+    uint64_t *bloom = nullptr; /* blocked bloom over full keys (may be null) */
+    uint64_t bloom_blocks = 0;
 };
 
 struct HipScanCtx {
@@ -496,6 +499,7 @@ struct HipEngine {
     hipStream_t stream = nullptr;
     uint32_t data_version = 1, default_ttl = 0;
     bool validate_hash = false, manual_compact_disabled = false;
+    bool bloom_enabled = true; /* env "rocksdb.filter_type": common(default)|none */
     uint32_t max_iter_count = 1000, mg_max_iter_count = 3000;
     uint64_t mg_max_iter_size = 30ull << 20, iter_time_ms = 30000;
     std::vector<RunBuf> runs;
@@ -532,7 +536,25 @@ struct HipEngine {
         (void)hipFree(r.vals);
         (void)hipFree(r.voff);
         (void)hipFree(r.sk);
+        if (r.bloom)
+            (void)hipFree(r.bloom);
         r = RunBuf();
+    }
+
+    /* §8(f)3: ~10 bits/key blocked bloom, built once per run on device */
+    void build_bloom(RunBuf &r)
+    {
+        if (!bloom_enabled || r.n == 0)
+            return;
+        uint64_t n_blocks = (r.n * 10 + 511) / 512;
+        if (n_blocks == 0)
+            n_blocks = 1;
+        HIP_OK(hipMalloc(&r.bloom, n_blocks * 64));
+        HIP_OK(hipMemsetAsync(r.bloom, 0, n_blocks * 64, stream));
+        DevRun dr{r.keys, r.koff, r.vals, r.voff, r.sk, r.n, nullptr, 0};
+        launch_bloom_build(dr, r.bloom, n_blocks, stream);
+        HIP_OK(hipStreamSynchronize(stream));
+        r.bloom_blocks = n_blocks;
     }
 
     DevRun *dev_runs()
@@ -543,7 +565,7 @@ struct HipEngine {
             std::vector<DevRun> h(runs.size() ? runs.size() : 1);
             for (size_t i = 0; i < runs.size(); i++)
                 h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
-                              runs[i].n};
+                              runs[i].n,    runs[i].bloom, runs[i].bloom_blocks};
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
@@ -823,6 +845,7 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     r.vals = e->upload_bytes(vals.data(), vals.size());
     r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size() * 8);
     r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
+    e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
 }
@@ -931,6 +954,8 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
             e->mg_max_iter_count = (uint32_t)atoll(v.c_str());
         } else if (k == "rocksdb.multi_get_max_iteration_size") {
             e->mg_max_iter_size = (uint64_t)atoll(v.c_str());
+        } else if (k == "rocksdb.filter_type") {
+            e->bloom_enabled = (v != "none"); /* common/prefix -> full-key bloom */
         } else if (k == "engine.rank_mode") {
             e->rank_mode = (v == "lds") ? 1 : 0;
         } else if (k == "engine.bt_shift") {
@@ -978,6 +1003,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     for (uint64_t i = 0; i < n; i++)
         mx = std::max(mx, seq_kind[i] >> 1);
     e->next_seq_floor = mx + 1;
+    e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
     return RRDB_OK;
@@ -2011,8 +2037,10 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         for (auto &r : e->runs)
             e->free_run(r);
         e->runs.clear();
-        if (n_out > 0)
+        if (n_out > 0) {
+            e->build_bloom(nr);
             e->runs.push_back(nr);
+        }
         e->d_runs_dirty = true;
     }
     if (stats)
@@ -2202,6 +2230,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         r.vals = e->upload_bytes(vals.data(), vals.size());
         r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size());
         r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size());
+        e->build_bloom(r); /* blooms are rebuilt, not serialized */
         e->runs.push_back(r);
     }
     e->next_seq_floor = floor_;

@@ -255,6 +255,9 @@ __device__ static int dev_rule_match(const DevRule &r, const uint8_t *pats, uint
     }
 }
 
+__device__ __host__ static inline uint64_t bloom_hash(const uint8_t *k, uint64_t n);
+__device__ static inline int bloom_maybe_has(const DevRun &r, uint64_t h);
+
 /* ================= bounds ================= */
 __global__ void k_bounds(const DevRun *runs, int R, const uint8_t *key, uint64_t klen,
                          uint64_t *out /* [R] */, int upper)
@@ -449,7 +452,10 @@ __global__ void k_get(const DevRun *runs, int R, const uint8_t *qkeys, const uin
         int32_t st = 1; /* NotFound */
         uint64_t h = 0, ul = 0;
         uint32_t eo = 0;
+        uint64_t bh = bloom_hash(k, kl);
         for (int r = R - 1; r >= 0; r--) {
+            if (!bloom_maybe_has(runs[r], bh))
+                continue; /* bloom negative: run cannot contain the key */
             uint64_t i = dev_lower_bound(runs[r], k, kl, 0, runs[r].n);
             if (i >= runs[r].n)
                 continue;

@@ -473,3 +473,36 @@ def test_scale_parity_1m(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+def test_bloom_filter_parity_and_toggle(oracle_lib, hip_lib):
+    """§8(f)3: per-run bloom filters are lossless — gets agree with the
+    oracle (no blooms) for hits and misses, with the filter on and off."""
+    import random as _r
+
+    rnd = _r.Random(31)
+    now = 1000
+    for filt in ("common", "none"):
+        o = oracle_lib.open(1, 0, -1)
+        g = hip_lib.open(1, 0, 0)
+        try:
+            g.set_envs({"rocksdb.filter_type": filt})
+            keys = _ingest_pair(o, g, _rand_runs(rnd, 5, 80, now=now))
+            probes = keys + [D.generate_key(f"miss{i}".encode(), b"") for i in range(40)]
+            for k in probes:
+                assert o.get(k, now) == g.get(k, now), (filt, k)
+            st_o = o.batch_get(probes, now)
+            st_g = g.batch_get(probes, now)
+            assert st_o == st_g
+            # write path + compact rebuild blooms and stay correct
+            g.put(b"bloomk", b"s", b"v")
+            o.put(b"bloomk", b"s", b"v")
+            assert o.get(D.generate_key(b"bloomk", b"s"), now) == \
+                   g.get(D.generate_key(b"bloomk", b"s"), now)
+            o.manual_compact(now)
+            g.manual_compact(now)
+            for k in probes[:60]:
+                assert o.get(k, now) == g.get(k, now), (filt, "post-compact", k)
+        finally:
+            o.close()
+            g.close()
