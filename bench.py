@@ -270,6 +270,7 @@ def main():
     L.rrdb_free_result(ctypes.byref(res))
     get_elapsed = time.time() - t0
     get_ops_per_s = nq / get_elapsed
+    get_kernel_ms = parts[0].phase_ms("get_search")
 
     # ---- YCSB-E: prefix multi_get over a sortkey table (config 3 shape) ----
     # single request stream through the C-ABI (per-call latency path; the
@@ -407,8 +408,8 @@ def main():
                 "scan_rows_total": rows,
                 "scan_state_kernel_GBs": round(scan_achieved, 1),
                 "get_ops_per_s": round(get_ops_per_s, 1),
-                "get_kernel_ops_per_s": (round(nq / (parts[0].phase_ms("get_search") * 1e-3), 1)
-                                         if parts[0].phase_ms("get_search") > 0 else None),
+                "get_kernel_ops_per_s": (round(nq / (get_kernel_ms * 1e-3), 1)
+                                         if get_kernel_ms > 0 else None),
                 "get_found": found,
                 "compact_output_records_per_gpu": int(out_records),
                 "compact_call_ms": compact_call_stats,
