@@ -506,3 +506,95 @@ def test_bloom_filter_parity_and_toggle(oracle_lib, hip_lib):
         finally:
             o.close()
             g.close()
+
+
+def test_mixed_op_soak(oracle_lib, hip_lib, tmp_path):
+    """Randomized interleaving of every ABI surface (writes, flush, ingest,
+    gets, multi_gets, scans with paging, sortkey_count, compaction,
+    checkpoint/restore, env flips) — engine must track the oracle through
+    400 operations."""
+    import random as _r
+
+    rnd = _r.Random(20260915)
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    live_keys = set()
+    now = 1000
+    seq = [1_000_000]  # ingest seqno space above write seqnos
+
+    def hk(i):
+        return f"soak{i:03d}".encode()
+
+    def sk(i):
+        return f"s{i:02d}".encode()
+
+    try:
+        for step in range(400):
+            op = rnd.randrange(100)
+            if op < 35:  # put
+                h, s = hk(rnd.randrange(40)), sk(rnd.randrange(8))
+                v = f"v{step}".encode() * rnd.randrange(1, 4)
+                exp = 0 if rnd.random() < 0.9 else now + rnd.randrange(1, 50)
+                for p in (o, g):
+                    p.put(h, s, v, exp)
+                live_keys.add((h, s))
+            elif op < 45:  # remove
+                h, s = hk(rnd.randrange(40)), sk(rnd.randrange(8))
+                for p in (o, g):
+                    p.remove(h, s)
+            elif op < 50:  # flush
+                for p in (o, g):
+                    p.flush()
+            elif op < 60:  # point get parity
+                h, s = hk(rnd.randrange(40)), sk(rnd.randrange(8))
+                assert o.get(D.generate_key(h, s), now) == g.get(D.generate_key(h, s), now)
+            elif op < 70:  # multi_get parity
+                h = hk(rnd.randrange(40))
+                kw = {}
+                if rnd.random() < 0.3:
+                    kw = dict(reverse=True)
+                elif rnd.random() < 0.3:
+                    kw = dict(sort_key_filter_type=FT_MATCH_PREFIX,
+                              sort_key_filter_pattern=b"s0")
+                assert o.multi_get(h, now, **kw) == g.multi_get(h, now, **kw)
+            elif op < 78:  # sortkey_count
+                h = hk(rnd.randrange(40))
+                assert o.sortkey_count(h, now) == g.sortkey_count(h, now)
+            elif op < 86:  # paged scan parity
+                bs = rnd.choice([3, 17, 1000])
+                assert _drain(o, now, batch_size=bs, validate_partition_hash=False) == \
+                       _drain(g, now, batch_size=bs, validate_partition_hash=False)
+            elif op < 92:  # compact (occasionally with default_ttl)
+                envs = {"default_ttl": str(rnd.choice([0, 1234]))}
+                for p in (o, g):
+                    p.set_envs(envs)
+                assert o.manual_compact(now) == g.manual_compact(now)
+            elif op < 96:  # ingest a pre-built run
+                n = rnd.randrange(1, 20)
+                ks = sorted({D.generate_key(hk(rnd.randrange(40)), sk(rnd.randrange(8)))
+                             for _ in range(n)})
+                recs = []
+                for k in ks:
+                    seq[0] += 1
+                    recs.append((k, D.encode_value(f"ing{step}".encode(), 0, 0, 1), seq[0], 0))
+                for p in (o, g):
+                    p.ingest_run(recs)
+            else:  # checkpoint round-trip spot check
+                d = str(tmp_path / f"ck{step}")
+                assert o.checkpoint(d, step) == 0
+                o2 = oracle_lib.open(1, 0, -1)
+                g2 = hip_lib.open(1, 0, 0)
+                try:
+                    assert o2.restore(d, step) == 0
+                    assert g2.restore(d, step) == 0
+                    assert _drain(o2, now, validate_partition_hash=False) == \
+                           _drain(g2, now, validate_partition_hash=False)
+                finally:
+                    o2.close()
+                    g2.close()
+        # final full comparison
+        assert _drain(o, now, validate_partition_hash=False, return_expire_ts=True) == \
+               _drain(g, now, validate_partition_hash=False, return_expire_ts=True)
+    finally:
+        o.close()
+        g.close()
