@@ -633,22 +633,10 @@ struct HipEngine {
      * per-op hipMalloc/hipFree (and mempool behavior variance across
      * driver configs) was a large, box-dependent share of step time. */
     std::vector<std::pair<uint8_t *, size_t>> sblocks;
-    size_t s_off = 0;
+    size_t sblock_i = 0, s_off = 0;
     void scratch_reset()
     {
-        if (sblocks.size() > 1) {
-            /* consolidate: one block at the high-water total */
-            HIP_OK(hipStreamSynchronize(stream));
-            size_t tot = 0;
-            for (auto &b : sblocks)
-                tot += b.second;
-            for (auto &b : sblocks)
-                (void)hipFree(b.first);
-            sblocks.clear();
-            uint8_t *pb = nullptr;
-            HIP_OK(hipMalloc(&pb, tot));
-            sblocks.push_back({pb, tot});
-        }
+        sblock_i = 0;
         s_off = 0;
     }
     template <typename T> T *talloc(uint64_t n_bytes)
@@ -656,7 +644,18 @@ struct HipEngine {
         uint64_t n = (n_bytes + 255) & ~255ull;
         if (n == 0)
             n = 256;
-        if (sblocks.empty() || s_off + n > sblocks.back().second) {
+        for (;;) {
+            if (sblock_i < sblocks.size()) {
+                if (s_off + n <= sblocks[sblock_i].second)
+                    break;
+                sblock_i++; /* tail of this block is wasted until reset */
+                s_off = 0;
+                continue;
+            }
+            /* grow: blocks are kept forever (an op-time hipMalloc happens at
+             * most a handful of times per engine lifetime; consolidation-
+             * style free+realloc cost ~100ms per engine at 7M-record
+             * partitions and poisoned the first timed step) */
             size_t tot = 0;
             for (auto &b : sblocks)
                 tot += b.second;
@@ -666,7 +665,7 @@ struct HipEngine {
             sblocks.push_back({pb, want});
             s_off = 0;
         }
-        T *out = (T *)(sblocks.back().first + s_off);
+        T *out = (T *)(sblocks[sblock_i].first + s_off);
         s_off += n;
         return out;
     }

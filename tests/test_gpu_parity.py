@@ -573,6 +573,9 @@ def test_mixed_op_soak(oracle_lib, hip_lib, tmp_path):
                 n = rnd.randrange(1, 20)
                 ks = sorted({D.generate_key(hk(rnd.randrange(40)), sk(rnd.randrange(8)))
                              for _ in range(n)})
+                # seqnos must stay above the write-path floor, which each
+                # put/remove advances; jump well past anything allocated so far
+                seq[0] = max(seq[0], 1_000_000 + step * 10_000)
                 recs = []
                 for k in ks:
                     seq[0] += 1
