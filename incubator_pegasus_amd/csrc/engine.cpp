@@ -1651,10 +1651,7 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
         fa.data_version = e->data_version;
         fa.hash_key_skip = 2 + q->hash_key.len;
         fa.out_hdr = e->talloc<int64_t>(4 * 8);
-        fa.out_koff = e->talloc<uint64_t>((MG_MAX_ROWS + 1) * 8);
-        fa.out_voff = e->talloc<uint64_t>((MG_MAX_ROWS + 1) * 8);
-        fa.out_keys = e->talloc<uint8_t>(MG_SCRATCH_BYTES);
-        fa.out_vals = e->talloc<uint8_t>(MG_SCRATCH_BYTES);
+        fa.out_blob = e->talloc<uint8_t>(MG_BLOB_BYTES);
         launch_multi_get_small(e->dev_runs(), (int)e->runs.size(), fa, e->stream);
         int64_t hdr4[4];
         HIP_OK(hipMemcpyAsync(hdr4, fa.out_hdr, 32, hipMemcpyDeviceToHost, e->stream));
@@ -1665,16 +1662,16 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
             out->keys = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
             out->values = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
             if (m) {
-                std::vector<uint64_t> koffs(m + 1), voffs(m + 1);
-                uint8_t *hk = (uint8_t *)a->alloc(kb);
-                uint8_t *hv = (uint8_t *)a->alloc(vb);
-                HIP_OK(hipMemcpyAsync(koffs.data(), fa.out_koff, (m + 1) * 8,
-                                      hipMemcpyDeviceToHost, e->stream));
-                HIP_OK(hipMemcpyAsync(voffs.data(), fa.out_voff, (m + 1) * 8,
-                                      hipMemcpyDeviceToHost, e->stream));
-                HIP_OK(hipMemcpyAsync(hk, fa.out_keys, kb, hipMemcpyDeviceToHost, e->stream));
-                HIP_OK(hipMemcpyAsync(hv, fa.out_vals, vb, hipMemcpyDeviceToHost, e->stream));
+                /* one D2H of [koff|voff|keys|vals] */
+                uint64_t blob_n = 2 * (m + 1) * 8 + kb + vb;
+                uint8_t *hb = (uint8_t *)a->alloc(blob_n);
+                HIP_OK(hipMemcpyAsync(hb, fa.out_blob, blob_n, hipMemcpyDeviceToHost,
+                                      e->stream));
                 HIP_OK(hipStreamSynchronize(e->stream));
+                const uint64_t *koffs = (const uint64_t *)hb;
+                const uint64_t *voffs = koffs + (m + 1);
+                uint8_t *hk = hb + 2 * (m + 1) * 8;
+                uint8_t *hv = hk + kb;
                 for (uint64_t j = 0; j < m; j++) {
                     out->keys[j] = {hk + koffs[j], koffs[j + 1] - koffs[j]};
                     out->values[j] = {hv + voffs[j], voffs[j + 1] - voffs[j]};

@@ -113,13 +113,12 @@ struct MgFusedArgs {
     uint32_t epoch_now;
     uint32_t data_version;
     uint64_t hash_key_skip;
-    /* outputs: [4] hdr = n_rows(-1 fallback), complete, kbytes, vbytes */
+    /* outputs: [4] hdr = n_rows(-1 fallback), complete, kbytes, vbytes.
+     * out_blob layout (one D2H): [koff (n+1)*8][voff (n+1)*8][keys][vals] */
     int64_t *out_hdr;
-    uint64_t *out_koff;
-    uint64_t *out_voff;
-    uint8_t *out_keys;
-    uint8_t *out_vals;
+    uint8_t *out_blob; /* MG_BLOB_BYTES */
 };
+#define MG_BLOB_BYTES (2 * MG_SCRATCH_BYTES + 2 * (MG_MAX_ROWS + 1) * 8)
 
 /* compact per-record disposition written by the filter kernel */
 struct CompactStatsDev {

@@ -1452,15 +1452,18 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
                     s_sel[nsel - 1 - x] = tmp;
                 }
             }
+            /* blob layout: [koff (nsel+1)*8][voff (nsel+1)*8][keys][vals] */
+            uint64_t *okoff = (uint64_t *)a.out_blob;
+            uint64_t *ovoff = okoff + (nsel + 1);
             uint64_t ko = 0, vo = 0;
             for (uint64_t x = 0; x < nsel; x++) {
-                a.out_koff[x] = ko;
-                a.out_voff[x] = vo;
+                okoff[x] = ko;
+                ovoff[x] = vo;
                 ko += s_klen[s_sel[x]];
                 vo += s_vlen_lo[s_sel[x]];
             }
-            a.out_koff[nsel] = ko;
-            a.out_voff[nsel] = vo;
+            okoff[nsel] = ko;
+            ovoff[nsel] = vo;
             a.out_hdr[0] = (int64_t)nsel;
             a.out_hdr[1] = complete;
             a.out_hdr[2] = (int64_t)ko;
@@ -1474,8 +1477,12 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
     __syncthreads();
     if (s_nsel == (uint64_t)-1)
         return;
-    /* phase 4: copy selected rows (wave per row) */
+    /* phase 4: copy selected rows (wave per row) into the blob */
     uint64_t nsel = s_nsel;
+    const uint64_t *okoff = (const uint64_t *)a.out_blob;
+    const uint64_t *ovoff = okoff + (nsel + 1);
+    uint8_t *keys_out = a.out_blob + 2 * (nsel + 1) * 8;
+    uint8_t *vals_out = keys_out + s_kbytes;
     int lane = tid % WAVE;
     int wid = tid / WAVE;
     int nw = blockDim.x / WAVE;
@@ -1487,10 +1494,10 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
         uint64_t i = id & 0xFFFFFFFFFFull;
         uint64_t kl, vl;
         const uint8_t *k = run_key(r, i, &kl);
-        wave_copy(a.out_keys + a.out_koff[x], k + a.hash_key_skip, s_klen[t], lane);
+        wave_copy(keys_out + okoff[x], k + a.hash_key_skip, s_klen[t], lane);
         if (!a.no_value && s_vlen_lo[t]) {
             const uint8_t *v = run_val(r, i, &vl);
-            wave_copy(a.out_vals + a.out_voff[x], v + hdr, s_vlen_lo[t], lane);
+            wave_copy(vals_out + ovoff[x], v + hdr, s_vlen_lo[t], lane);
         }
     }
 }
