@@ -83,6 +83,9 @@ typedef struct {
     rrdb_slice *keys;    /* [count]; scan: full rocksdb key; multi_get: sort key */
     rrdb_slice *values;  /* [count]; user data (value header stripped) */
     int32_t *expire_ts;  /* [count] when return_expire_ts was set, else NULL */
+    /* batched multi_get only: per-request row counts and status codes */
+    uint64_t *group_counts; /* [n_req] */
+    int32_t *group_errors;  /* [n_req] */
     /* engine extension: when a scan asked for device-resident output the kv
      * bytes stay in HBM and these describe the packed device buffers
      * (keys/values above are then NULL). */
@@ -236,6 +239,16 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hash_key_l
 /* on_multi_get (pegasus_server_impl.cpp:496-904). */
 int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *req, uint32_t epoch_now,
                        rrdb_result *out);
+
+/* Batched multi_get (engine extension): N independent on_multi_get range
+ * requests sharing every field of `shared` except hash_key, answered in one
+ * launch (one workgroup per request) — models the reference's concurrent
+ * THREAD_POOL_SCAN handlers without N round trips.  Results
+ * concatenate in request order; out->group_counts / group_errors give each
+ * request's row count and status.  shared->n_sort_keys must be 0. */
+int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *hash_keys,
+                             const uint64_t *hk_offs, const rrdb_multi_get_request *shared,
+                             uint32_t epoch_now, rrdb_result *out);
 
 /* on_get_scanner (pegasus_server_impl.cpp:1151-1397): first batch + parked
  * continuation in out->context_id (RRDB_SCAN_CONTEXT_ID_COMPLETED when done). */

@@ -50,6 +50,8 @@ class _Result(C.Structure):
         ("keys", C.POINTER(_Slice)),
         ("values", C.POINTER(_Slice)),
         ("expire_ts", C.POINTER(C.c_int32)),
+        ("group_counts", C.POINTER(C.c_uint64)),
+        ("group_errors", C.POINTER(C.c_int32)),
         ("dev_keys", C.c_void_p),
         ("dev_key_offs", C.c_void_p),
         ("dev_vals", C.c_void_p),
@@ -205,6 +207,10 @@ class RrdbLib:
         L.rrdb_checkpoint.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64]
         L.rrdb_restore.restype = C.c_int32
         L.rrdb_restore.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64]
+        L.rrdb_multi_get_batch.restype = C.c_int32
+        L.rrdb_multi_get_batch.argtypes = [C.c_void_p, C.c_uint64, C.c_void_p, C.c_void_p,
+                                           C.POINTER(_MultiGetRequest), C.c_uint32,
+                                           C.POINTER(_Result)]
         L.rrdb_num_runs.restype = C.c_uint64
         L.rrdb_num_runs.argtypes = [C.c_void_p]
         L.rrdb_num_records.restype = C.c_uint64
@@ -369,6 +375,45 @@ class RrdbPartition:
             kvs = [(_read_slice(res.keys[i]), _read_slice(res.values[i]))
                    for i in range(res.count)]
             return res.error, kvs
+        finally:
+            self._L.rrdb_free_result(C.byref(res))
+
+    def multi_get_batch(self, hash_keys, epoch_now: int, **shared_kwargs):
+        """N full-range (or shared-shape) multi_gets in one call; returns
+        (error, [per-request (error, [(sortkey, value), ...])])."""
+        keep = []
+        req = _MultiGetRequest()
+        req.hash_key = _CSlice(None, 0)
+        req.start_sortkey = _cslice(shared_kwargs.get("start_sortkey", b""), keep)
+        req.stop_sortkey = _cslice(shared_kwargs.get("stop_sortkey", b""), keep)
+        req.start_inclusive = 1 if shared_kwargs.get("start_inclusive", True) else 0
+        req.stop_inclusive = 1 if shared_kwargs.get("stop_inclusive", False) else 0
+        req.max_kv_count = shared_kwargs.get("max_kv_count", -1)
+        req.max_kv_size = shared_kwargs.get("max_kv_size", -1)
+        req.no_value = 1 if shared_kwargs.get("no_value", False) else 0
+        req.reverse = 1 if shared_kwargs.get("reverse", False) else 0
+        req.sort_key_filter_type = shared_kwargs.get("sort_key_filter_type", FT_NO_FILTER)
+        req.sort_key_filter_pattern = _cslice(
+            shared_kwargs.get("sort_key_filter_pattern", b""), keep)
+        hks, offs = _pack(hash_keys)
+        res = _Result()
+        self._L.rrdb_multi_get_batch(self._h, len(hash_keys),
+                                     hks.ctypes.data_as(C.c_void_p),
+                                     offs.ctypes.data_as(C.c_void_p), C.byref(req), epoch_now,
+                                     C.byref(res))
+        del keep
+        try:
+            if res.error != OK:
+                return res.error, []
+            groups = []
+            m = 0
+            for i in range(len(hash_keys)):
+                n = res.group_counts[i]
+                kvs = [(_read_slice(res.keys[m + j]), _read_slice(res.values[m + j]))
+                       for j in range(n)]
+                m += n
+                groups.append((res.group_errors[i], kvs))
+            return OK, groups
         finally:
             self._L.rrdb_free_result(C.byref(res))
 

@@ -2235,3 +2235,161 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
 }
 
 } /* extern "C" */
+
+/* ================= batched multi_get =================
+ * One workgroup per request (full-range multi_get per hashkey) — models the
+ * reference's concurrent THREAD_POOL_SCAN handlers in a single launch.
+ * Requests whose ranges exceed the fused budget (or exotic shared params)
+ * fall back to the general per-request path. */
+void launch_multi_get_batch(const DevRun *, int, const MgFusedArgs &, const uint8_t *,
+                            const uint64_t *, uint64_t, int64_t *, uint8_t *, uint64_t,
+                            hipStream_t);
+void launch_pack_blobs(const uint8_t *, uint64_t, uint64_t, const uint64_t *, const uint64_t *,
+                       uint8_t *, hipStream_t);
+
+extern "C" int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *hash_keys,
+                                        const uint64_t *hk_offs,
+                                        const rrdb_multi_get_request *shared,
+                                        uint32_t epoch_now, rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    engine_flush(e);
+    e->activate();
+    e->scratch_reset();
+    Arena *a = result_init(out);
+    if (shared->n_sort_keys != 0 || shared->sort_key_filter_type < 0 ||
+        shared->sort_key_filter_type > 3) {
+        out->error = RRDB_INVALID_ARGUMENT;
+        return out->error;
+    }
+    out->group_counts = (uint64_t *)a->alloc(n_req * 8);
+    out->group_errors = (int32_t *)a->alloc(n_req * 4);
+    /* fused batch only covers the full-range shape (no sortkey bounds, no
+     * range-clamping prefix filter); others run per-request */
+    bool fused_ok = shared->start_sortkey.len == 0 && shared->stop_sortkey.len == 0 &&
+                    shared->start_inclusive &&
+                    !(shared->sort_key_filter_type == RRDB_FT_MATCH_PREFIX &&
+                      shared->sort_key_filter_pattern.len > 0) &&
+                    !e->runs.empty();
+    const uint64_t BLOB_STRIDE = 64 << 10;
+    std::vector<int64_t> hdrs(n_req * 4, -1);
+    uint8_t *d_blobs = nullptr;
+    std::vector<uint64_t> used(n_req, 0), pack_off(n_req + 1, 0);
+    std::vector<uint8_t> packed_host;
+    if (fused_ok && n_req > 0) {
+        uint32_t max_kv_count = e->mg_max_iter_count;
+        if (shared->max_kv_count > 0 && (uint32_t)shared->max_kv_count < max_kv_count)
+            max_kv_count = (uint32_t)shared->max_kv_count;
+        int64_t max_kv_size = shared->max_kv_size > 0 ? shared->max_kv_size : INT32_MAX;
+        MgFusedArgs fa{};
+        fa.start_inclusive = 1;
+        fa.stop_inclusive = 0;
+        fa.reverse = shared->reverse;
+        fa.no_value = shared->no_value;
+        fa.max_kv_count = max_kv_count;
+        fa.max_iteration_count = e->mg_max_iter_count;
+        fa.max_iteration_size = std::min<int64_t>(
+            max_kv_size, e->mg_max_iter_size > 0 ? (int64_t)e->mg_max_iter_size : INT32_MAX);
+        fa.sk_ft = shared->sort_key_filter_type;
+        fa.sk_pat_len = shared->sort_key_filter_pattern.len;
+        fa.sk_pat = e->upload_tmp(shared->sort_key_filter_pattern.data, fa.sk_pat_len);
+        fa.epoch_now = epoch_now;
+        fa.data_version = e->data_version;
+        /* per-request hash_key_skip (= start key length) is computed
+         * in-kernel so size caps count sortkey+value, as the reference does */
+        uint8_t *d_hks = e->upload_tmp(hash_keys, hk_offs[n_req]);
+        uint64_t *d_offs = (uint64_t *)e->upload_tmp(hk_offs, (n_req + 1) * 8);
+        int64_t *d_hdrs = e->talloc<int64_t>(n_req * 4 * 8);
+        HIP_OK(hipMemsetAsync(d_hdrs, 0xFF, n_req * 4 * 8, e->stream)); /* -1 fill */
+        d_blobs = e->talloc<uint8_t>(n_req * BLOB_STRIDE);
+        launch_multi_get_batch(e->dev_runs(), (int)e->runs.size(), fa, d_hks, d_offs, n_req,
+                               d_hdrs, d_blobs, BLOB_STRIDE, e->stream);
+        HIP_OK(hipMemcpyAsync(hdrs.data(), d_hdrs, n_req * 4 * 8, hipMemcpyDeviceToHost,
+                              e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        /* pack used regions, one D2H */
+        uint64_t total = 0;
+        for (uint64_t i = 0; i < n_req; i++) {
+            if (hdrs[i * 4] >= 0) {
+                uint64_t m = (uint64_t)hdrs[i * 4];
+                used[i] = 2 * (m + 1) * 8 + (uint64_t)hdrs[i * 4 + 2] + (uint64_t)hdrs[i * 4 + 3];
+            }
+            pack_off[i] = total;
+            total += used[i];
+        }
+        pack_off[n_req] = total;
+        if (total) {
+            uint64_t *d_used = (uint64_t *)e->upload_tmp(used.data(), n_req * 8);
+            uint64_t *d_poff = (uint64_t *)e->upload_tmp(pack_off.data(), (n_req + 1) * 8);
+            uint8_t *d_packed = e->talloc<uint8_t>(total);
+            launch_pack_blobs(d_blobs, BLOB_STRIDE, n_req, d_used, d_poff, d_packed, e->stream);
+            packed_host.resize(total);
+            e->d2h(packed_host.data(), d_packed, total);
+        }
+    }
+    /* count rows (fused + fallback) */
+    uint64_t total_rows = 0;
+    std::vector<rrdb_result> fb(n_req);
+    std::vector<uint8_t> is_fb(n_req, 0);
+    for (uint64_t i = 0; i < n_req; i++) {
+        if (hdrs[i * 4] >= 0) {
+            total_rows += (uint64_t)hdrs[i * 4];
+        } else {
+            /* general path per request (rare) */
+            is_fb[i] = 1;
+            rrdb_multi_get_request req = *shared;
+            req.hash_key.data = hash_keys + hk_offs[i];
+            req.hash_key.len = hk_offs[i + 1] - hk_offs[i];
+            e->mu.unlock(); /* rrdb_multi_get re-locks */
+            rrdb_multi_get(h, &req, epoch_now, &fb[i]);
+            e->mu.lock();
+            total_rows += fb[i].count;
+        }
+    }
+    out->keys = (rrdb_slice *)a->alloc((total_rows ? total_rows : 1) * sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)a->alloc((total_rows ? total_rows : 1) * sizeof(rrdb_slice));
+    uint8_t *pb = nullptr;
+    if (!packed_host.empty()) {
+        pb = (uint8_t *)a->alloc(packed_host.size());
+        memcpy(pb, packed_host.data(), packed_host.size());
+    }
+    uint64_t m = 0;
+    for (uint64_t i = 0; i < n_req; i++) {
+        if (!is_fb[i]) {
+            uint64_t n = (uint64_t)hdrs[i * 4];
+            uint64_t kb = (uint64_t)hdrs[i * 4 + 2];
+            out->group_counts[i] = n;
+            out->group_errors[i] = hdrs[i * 4 + 1] ? RRDB_OK : RRDB_INCOMPLETE;
+            if (n == 0)
+                continue;
+            uint8_t *base = pb + pack_off[i];
+            const uint64_t *koffs = (const uint64_t *)base;
+            const uint64_t *voffs = koffs + (n + 1);
+            uint8_t *keys_b = base + 2 * (n + 1) * 8;
+            uint8_t *vals_b = keys_b + kb;
+            for (uint64_t j = 0; j < n; j++) {
+                out->keys[m] = {keys_b + koffs[j], koffs[j + 1] - koffs[j]};
+                out->values[m] = {vals_b + voffs[j], voffs[j + 1] - voffs[j]};
+                m++;
+            }
+        } else {
+            out->group_counts[i] = fb[i].count;
+            out->group_errors[i] = fb[i].error;
+            for (uint64_t j = 0; j < fb[i].count; j++) {
+                uint64_t kl = fb[i].keys[j].len, vl = fb[i].values[j].len;
+                uint8_t *kc = (uint8_t *)a->alloc(kl ? kl : 1);
+                memcpy(kc, fb[i].keys[j].data, kl);
+                uint8_t *vc = (uint8_t *)a->alloc(vl ? vl : 1);
+                memcpy(vc, fb[i].values[j].data, vl);
+                out->keys[m] = {kc, kl};
+                out->values[m] = {vc, vl};
+                m++;
+            }
+            rrdb_free_result(&fb[i]);
+        }
+    }
+    out->count = m;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}

@@ -1292,8 +1292,11 @@ void launch_emit_compact(const DevRun *d_runs, const uint64_t *d_order, uint64_t
  * result blob, no intermediate syncs.  Semantics mirror
  * on_multi_get:540-778 exactly; ranges larger than the scratch budget set
  * out_hdr[0]=-1 and the host falls back to the general path. */
-__global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int R,
-                                                         MgFusedArgs a)
+__device__ static void mg_core(const DevRun *runs, int R, const MgFusedArgs &a,
+                               const uint8_t *mg_start, uint64_t mg_start_len,
+                               const uint8_t *mg_stop, uint64_t mg_stop_len,
+                               uint64_t hash_key_skip, int64_t *out_hdr, uint8_t *out_blob,
+                               uint64_t blob_cap)
 {
     __shared__ uint64_t s_lo[RRDB_MAX_RUNS], s_hi[RRDB_MAX_RUNS], s_wp[RRDB_MAX_RUNS + 1];
     __shared__ uint64_t s_total;
@@ -1304,8 +1307,8 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
     int tid = threadIdx.x;
     /* phase 1: bounds */
     if (tid < R) {
-        s_lo[tid] = dev_lower_bound(runs[tid], a.start, a.start_len, 0, runs[tid].n);
-        s_hi[tid] = dev_lower_bound(runs[tid], a.stop, a.stop_len, 0, runs[tid].n);
+        s_lo[tid] = dev_lower_bound(runs[tid], mg_start, mg_start_len, 0, runs[tid].n);
+        s_hi[tid] = dev_lower_bound(runs[tid], mg_stop, mg_stop_len, 0, runs[tid].n);
         if (s_hi[tid] < s_lo[tid])
             s_hi[tid] = s_lo[tid];
     }
@@ -1319,7 +1322,7 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
         s_wp[R] = t;
         s_total = t;
         if (t > MG_MAX_ROWS)
-            a.out_hdr[0] = -1; /* fallback */
+            out_hdr[0] = -1; /* fallback */
     }
     __syncthreads();
     uint64_t total = s_total;
@@ -1362,16 +1365,16 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
             if (dev_ts_expired(a.epoch_now, dev_expire_ts(a.data_version, v))) {
                 st = 1; /* kExpired: iterated but skipped */
             } else {
-                const uint8_t *skp = k + a.hash_key_skip;
-                uint64_t sklen = kl - a.hash_key_skip;
+                const uint8_t *skp = k + hash_key_skip;
+                uint64_t sklen = kl - hash_key_skip;
                 if (a.sk_ft != 0 &&
                     !dev_validate_filter(a.sk_ft, a.sk_pat, a.sk_pat_len, skp, sklen)) {
                     st = 1; /* kFiltered */
-                } else if (kl - a.hash_key_skip > 0xFFFF || vl - hdr > 0xFFFF) {
+                } else if (kl - hash_key_skip > 0xFFFF || vl - hdr > 0xFFFF) {
                     st = 3; /* oversize for fused path -> fallback */
                 } else {
                     st = 0;
-                    klen_out = (uint16_t)(kl - a.hash_key_skip);
+                    klen_out = (uint16_t)(kl - hash_key_skip);
                     vlen_out = a.no_value ? 0 : (uint16_t)(vl - hdr);
                 }
             }
@@ -1417,8 +1420,8 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
                     const DevRun &r = runs[id >> 40];
                     uint64_t i = id & 0xFFFFFFFFFFull, kl;
                     const uint8_t *k = run_key(r, i, &kl);
-                    const uint8_t *bound = a.reverse ? a.stop : a.start;
-                    uint64_t blen = a.reverse ? a.stop_len : a.start_len;
+                    const uint8_t *bound = a.reverse ? mg_stop : mg_start;
+                    uint64_t blen = a.reverse ? mg_stop_len : mg_start_len;
                     visited = 1;
                     if (dev_key_cmp(k, kl, bound, blen) == 0) {
                         skipped_first = 1;
@@ -1438,10 +1441,10 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
             size += (int64_t)s_klen[t] + (int64_t)s_vlen_lo[t];
         }
         complete = (iteration >= countable - (uint64_t)skipped_first);
-        if (kb > MG_SCRATCH_BYTES || vb > MG_SCRATCH_BYTES)
+        if (2 * (nsel + 1) * 8 + kb + vb > blob_cap)
             fallback = 1;
         if (fallback) {
-            a.out_hdr[0] = -1;
+            out_hdr[0] = -1;
             s_nsel = (uint64_t)-1;
         } else {
             /* ascending output (reverse selected from the top) */
@@ -1453,7 +1456,7 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
                 }
             }
             /* blob layout: [koff (nsel+1)*8][voff (nsel+1)*8][keys][vals] */
-            uint64_t *okoff = (uint64_t *)a.out_blob;
+            uint64_t *okoff = (uint64_t *)out_blob;
             uint64_t *ovoff = okoff + (nsel + 1);
             uint64_t ko = 0, vo = 0;
             for (uint64_t x = 0; x < nsel; x++) {
@@ -1464,10 +1467,10 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
             }
             okoff[nsel] = ko;
             ovoff[nsel] = vo;
-            a.out_hdr[0] = (int64_t)nsel;
-            a.out_hdr[1] = complete;
-            a.out_hdr[2] = (int64_t)ko;
-            a.out_hdr[3] = (int64_t)vo;
+            out_hdr[0] = (int64_t)nsel;
+            out_hdr[1] = complete;
+            out_hdr[2] = (int64_t)ko;
+            out_hdr[3] = (int64_t)vo;
             s_nsel = nsel;
             s_kbytes = ko;
             s_vbytes = vo;
@@ -1479,9 +1482,9 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
         return;
     /* phase 4: copy selected rows (wave per row) into the blob */
     uint64_t nsel = s_nsel;
-    const uint64_t *okoff = (const uint64_t *)a.out_blob;
+    const uint64_t *okoff = (const uint64_t *)out_blob;
     const uint64_t *ovoff = okoff + (nsel + 1);
-    uint8_t *keys_out = a.out_blob + 2 * (nsel + 1) * 8;
+    uint8_t *keys_out = out_blob + 2 * (nsel + 1) * 8;
     uint8_t *vals_out = keys_out + s_kbytes;
     int lane = tid % WAVE;
     int wid = tid / WAVE;
@@ -1494,7 +1497,7 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
         uint64_t i = id & 0xFFFFFFFFFFull;
         uint64_t kl, vl;
         const uint8_t *k = run_key(r, i, &kl);
-        wave_copy(keys_out + okoff[x], k + a.hash_key_skip, s_klen[t], lane);
+        wave_copy(keys_out + okoff[x], k + hash_key_skip, s_klen[t], lane);
         if (!a.no_value && s_vlen_lo[t]) {
             const uint8_t *v = run_val(r, i, &vl);
             wave_copy(vals_out + ovoff[x], v + hdr, s_vlen_lo[t], lane);
@@ -1502,9 +1505,87 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
     }
 }
 
+__global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int R,
+                                                         MgFusedArgs a)
+{
+    mg_core(runs, R, a, a.start, a.start_len, a.stop, a.stop_len, a.hash_key_skip, a.out_hdr,
+            a.out_blob, MG_BLOB_BYTES);
+}
+
+/* batched full-range multi_get: one workgroup per request (hashkey); start =
+ * [u16 len][hk], stop = pegasus_generate_next_blob(hk)
+ * (pegasus_key_schema.h:64-81) built in-kernel */
+__global__ void __launch_bounds__(256) k_multi_get_batch(const DevRun *runs, int R,
+                                                         MgFusedArgs a, const uint8_t *hks,
+                                                         const uint64_t *hk_offs,
+                                                         uint64_t n_req, int64_t *hdrs,
+                                                         uint8_t *blobs, uint64_t blob_stride)
+{
+    __shared__ uint8_t s_start[2 + 4096], s_stop[2 + 4096];
+    __shared__ uint64_t s_slen, s_stlen;
+    uint64_t req = blockIdx.x;
+    if (req >= n_req)
+        return;
+    if (threadIdx.x == 0) {
+        uint64_t hl = hk_offs[req + 1] - hk_offs[req];
+        if (hl == 0 || hl > 4096) {
+            hdrs[req * 4] = -1; /* host fallback */
+            s_slen = (uint64_t)-1;
+        } else {
+            const uint8_t *hk = hks + hk_offs[req];
+            s_start[0] = (uint8_t)(hl >> 8);
+            s_start[1] = (uint8_t)hl;
+            for (uint64_t b = 0; b < hl; b++)
+                s_start[2 + b] = hk[b];
+            s_slen = 2 + hl;
+            for (uint64_t b = 0; b < s_slen; b++)
+                s_stop[b] = s_start[b];
+            uint64_t p = s_slen - 1;
+            while (s_stop[p] == 0xFF)
+                p--;
+            s_stop[p]++;
+            s_stlen = p + 1;
+        }
+    }
+    __syncthreads();
+    if (s_slen == (uint64_t)-1)
+        return;
+    mg_core(runs, R, a, s_start, s_slen, s_stop, s_stlen, s_slen /* 2+hklen */,
+            hdrs + req * 4, blobs + req * blob_stride, blob_stride);
+}
+
+/* gather each request's used blob bytes into one packed buffer (one D2H) */
+__global__ void k_pack_blobs(const uint8_t *blobs, uint64_t blob_stride, uint64_t n_req,
+                             const uint64_t *used /* [n_req] */,
+                             const uint64_t *pack_off /* [n_req+1] */, uint8_t *packed)
+{
+    uint64_t wave = (blockIdx.x * (uint64_t)blockDim.x + threadIdx.x) / WAVE;
+    uint64_t nwaves = (gridDim.x * (uint64_t)blockDim.x) / WAVE;
+    int lane = threadIdx.x % WAVE;
+    for (uint64_t i = wave; i < n_req; i += nwaves)
+        wave_copy(packed + pack_off[i], blobs + i * blob_stride, used[i], lane);
+}
+
 void launch_multi_get_small(const DevRun *d_runs, int R, const MgFusedArgs &a, hipStream_t s)
 {
     k_multi_get_small<<<1, 256, 0, s>>>(d_runs, R, a);
+}
+
+void launch_multi_get_batch(const DevRun *d_runs, int R, const MgFusedArgs &a,
+                            const uint8_t *d_hks, const uint64_t *d_hk_offs, uint64_t n_req,
+                            int64_t *d_hdrs, uint8_t *d_blobs, uint64_t blob_stride,
+                            hipStream_t s)
+{
+    k_multi_get_batch<<<(uint32_t)n_req, 256, 0, s>>>(d_runs, R, a, d_hks, d_hk_offs, n_req,
+                                                      d_hdrs, d_blobs, blob_stride);
+}
+
+void launch_pack_blobs(const uint8_t *d_blobs, uint64_t blob_stride, uint64_t n_req,
+                       const uint64_t *d_used, const uint64_t *d_pack_off, uint8_t *d_packed,
+                       hipStream_t s)
+{
+    k_pack_blobs<<<grid_for(n_req * WAVE, BLOCK), BLOCK, 0, s>>>(d_blobs, blob_stride, n_req,
+                                                                 d_used, d_pack_off, d_packed);
 }
 
 /* ================= LDS-staged rank (compaction) =================

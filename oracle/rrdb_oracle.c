@@ -2308,3 +2308,51 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
     e->next_seq_floor = floor_;
     return RRDB_OK;
 }
+
+/* batched multi_get: the oracle answers it as N sequential rrdb_multi_get
+ * calls (semantics by definition — the batch is an engine-side concurrency
+ * vehicle, not a semantic change). */
+int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *hash_keys,
+                             const uint64_t *hk_offs, const rrdb_multi_get_request *shared,
+                             uint32_t epoch_now, rrdb_result *out)
+{
+    arena *a = result_init(out);
+    if (shared->n_sort_keys != 0) {
+        out->error = RRDB_INVALID_ARGUMENT;
+        return out->error;
+    }
+    out->group_counts = (uint64_t *)arena_alloc(a, n_req * 8);
+    out->group_errors = (int32_t *)arena_alloc(a, n_req * 4);
+    /* first pass: run each request, tally rows */
+    rrdb_result *sub = (rrdb_result *)malloc(n_req * sizeof(rrdb_result));
+    uint64_t total = 0;
+    for (uint64_t i = 0; i < n_req; i++) {
+        rrdb_multi_get_request req = *shared;
+        req.hash_key.data = hash_keys + hk_offs[i];
+        req.hash_key.len = hk_offs[i + 1] - hk_offs[i];
+        rrdb_multi_get(h, &req, epoch_now, &sub[i]);
+        out->group_counts[i] = sub[i].count;
+        out->group_errors[i] = sub[i].error;
+        total += sub[i].count;
+    }
+    out->keys = (rrdb_slice *)arena_alloc(a, (total ? total : 1) * sizeof(rrdb_slice));
+    out->values = (rrdb_slice *)arena_alloc(a, (total ? total : 1) * sizeof(rrdb_slice));
+    uint64_t m = 0;
+    for (uint64_t i = 0; i < n_req; i++) {
+        for (uint64_t j = 0; j < sub[i].count; j++) {
+            uint64_t kl = sub[i].keys[j].len, vl = sub[i].values[j].len;
+            out->keys[m].data = (uint8_t *)arena_alloc(a, kl ? kl : 1);
+            memcpy(out->keys[m].data, sub[i].keys[j].data, kl);
+            out->keys[m].len = kl;
+            out->values[m].data = (uint8_t *)arena_alloc(a, vl ? vl : 1);
+            memcpy(out->values[m].data, sub[i].values[j].data, vl);
+            out->values[m].len = vl;
+            m++;
+        }
+        rrdb_free_result(&sub[i]);
+    }
+    free(sub);
+    out->count = m;
+    out->error = RRDB_OK;
+    return RRDB_OK;
+}

@@ -601,3 +601,47 @@ def test_mixed_op_soak(oracle_lib, hip_lib, tmp_path):
     finally:
         o.close()
         g.close()
+
+
+def test_multi_get_batch_parity(oracle_lib, hip_lib):
+    """Batched multi_get == N single multi_gets, on both backends, including
+    shapes that force the engine's per-request fallback."""
+    import random as _r
+
+    rnd = _r.Random(77)
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        now = 1000
+        hks = [f"bmg{i:03d}".encode() for i in range(50)]
+        seq = 1
+        recs = {}
+        for hk in hks:
+            for s in range(rnd.randrange(0, 12)):
+                expire = 0 if rnd.random() < 0.8 else (now - 1 if rnd.random() < 0.5 else now + 50)
+                recs[D.generate_key(hk, f"s{s:02d}".encode())] = \
+                    D.encode_value(f"{hk}{s}".encode(), expire, seq, 1)
+        records = []
+        for k in sorted(recs):
+            records.append((k, recs[k], seq, 0))
+            seq += 1
+        o.ingest_run(records)
+        g.ingest_run(records)
+        ask = hks + [b"bmg-missing", b"bmg0"]  # misses included
+        for kwargs in [
+            dict(),
+            dict(no_value=True),
+            dict(reverse=True),
+            dict(max_kv_count=3),
+            dict(sort_key_filter_type=FT_MATCH_POSTFIX, sort_key_filter_pattern=b"1"),
+            dict(start_sortkey=b"s02"),  # engine falls back per request
+        ]:
+            eo, go_ = o.multi_get_batch(ask, now, **kwargs), g.multi_get_batch(ask, now, **kwargs)
+            assert eo == go_, kwargs
+            # and equals the per-request loop
+            err, groups = go_
+            for hk, (gerr, kvs) in zip(ask, groups):
+                assert (gerr, kvs) == g.multi_get(hk, now, **kwargs), (hk, kwargs)
+    finally:
+        o.close()
+        g.close()
