@@ -1530,13 +1530,13 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
     return RRDB_OK;
 }
 
-int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
-                       rrdb_result *out)
+/* body of on_multi_get; caller holds the engine lock */
+static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
+                                rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
-    engine_flush((HipEngine *)h); /* memtable visible to reads */
-    ((HipEngine *)h)->scratch_reset();
+    engine_flush(e); /* memtable visible to reads */
+    e->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
     if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
@@ -1823,6 +1823,13 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     out->count = m;
     out->error = complete ? RRDB_OK : RRDB_INCOMPLETE; /* :789-799 */
     return out->error;
+}
+
+int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
+                       rrdb_result *out)
+{
+    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
+    return multi_get_locked(h, q, epoch_now, out);
 }
 
 int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
@@ -2341,9 +2348,7 @@ extern "C" int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *
             rrdb_multi_get_request req = *shared;
             req.hash_key.data = hash_keys + hk_offs[i];
             req.hash_key.len = hk_offs[i + 1] - hk_offs[i];
-            e->mu.unlock(); /* rrdb_multi_get re-locks */
-            rrdb_multi_get(h, &req, epoch_now, &fb[i]);
-            e->mu.lock();
+            multi_get_locked(h, &req, epoch_now, &fb[i]);
             total_rows += fb[i].count;
         }
     }
