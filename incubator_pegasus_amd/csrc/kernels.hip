@@ -194,11 +194,11 @@ __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, 
     return lo;
 }
 
-/* interpolated bound inside a small window: one estimated probe, a gallop
- * to a verified bracket, then binary search in the bracket.  The estimate
- * skips the window's common key prefix (bench keys share 14+ leading bytes);
- * every boundary decision is a real key compare, so the result is exact for
- * ANY distribution — non-uniform data just gallops more. */
+/* interpolated bound inside a small window (KEPT AS A MEASURED NEGATIVE
+ * RESULT, unused): parity-correct but 1.9x SLOWER than plain binary search
+ * on the narrowed windows (rank 4.08 vs 2.17 ms) — the per-lane gallop
+ * paths diverge across the 64-wide wave and the prefix arithmetic costs
+ * more than ~6 L1-hot binary probes. */
 template <int UPPER>
 __device__ static uint64_t dev_interp_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                             uint64_t lo, uint64_t hi)
@@ -466,7 +466,7 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
             uint64_t qlo = b0 ? b0[q] : lo[q];
             uint64_t qhi = b1 ? b1[q] : hi[q];
             if (q > r) {
-                uint64_t ub = dev_interp_bound<1>(runs[q], k, kl, qlo, qhi);
+                uint64_t ub = dev_upper_bound(runs[q], k, kl, qlo, qhi);
                 if (!shadow && ub > lo[q]) {
                     uint64_t pl;
                     const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
@@ -475,7 +475,7 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
                 }
                 rank += ub - lo[q];
             } else {
-                rank += dev_interp_bound<0>(runs[q], k, kl, qlo, qhi) - lo[q];
+                rank += dev_lower_bound(runs[q], k, kl, qlo, qhi) - lo[q];
             }
         }
         order[rank] = ((uint64_t)r << 40) | i;
@@ -934,7 +934,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                 uint64_t qlo = b0 ? b0[q] : lo[q];
                 uint64_t qhi = b1 ? b1[q] : hi[q];
                 if (q > r) {
-                    uint64_t ub = dev_interp_bound<1>(runs[q], k, kl, qlo, qhi);
+                    uint64_t ub = dev_upper_bound(runs[q], k, kl, qlo, qhi);
                     if (!shadow && ub > lo[q]) {
                         uint64_t pl;
                         const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
@@ -943,7 +943,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                     }
                     rank += ub - lo[q];
                 } else {
-                    rank += dev_interp_bound<0>(runs[q], k, kl, qlo, qhi) - lo[q];
+                    rank += dev_lower_bound(runs[q], k, kl, qlo, qhi) - lo[q];
                 }
             }
             uint8_t ch;
