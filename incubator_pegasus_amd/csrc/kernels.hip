@@ -1499,14 +1499,14 @@ __device__ static void mg_core(const DevRun *runs, int R, const MgFusedArgs &a,
     __syncthreads();
     /* phase 3 (thread 0): limiter walk, exactly on_multi_get:616-778 */
     __shared__ uint64_t s_sel[MG_MAX_ROWS]; /* selected merged positions */
-    __shared__ uint64_t s_nsel, s_kbytes, s_vbytes, s_complete;
+    __shared__ uint64_t s_nsel, s_kbytes;
     if (tid == 0) {
         int fallback = 0;
         uint64_t nsel = 0, kb = 0, vb = 0;
         uint64_t count = 0, iteration = 0;
         int64_t size = 0;
         int complete = 0;
-        uint64_t iterated_total = 0, countable = 0;
+        uint64_t countable = 0;
         /* countable = entries the reference iterator would visit (visible) */
         for (uint64_t t = 0; t < total; t++)
             if (s_state[t] != 2 && s_state[t] != 3)
@@ -1584,8 +1584,6 @@ __device__ static void mg_core(const DevRun *runs, int R, const MgFusedArgs &a,
             out_hdr[3] = (int64_t)vo;
             s_nsel = nsel;
             s_kbytes = ko;
-            s_vbytes = vo;
-            s_complete = complete;
         }
     }
     __syncthreads();
