@@ -139,32 +139,6 @@ __device__ static inline const uint8_t *run_val(const DevRun &r, uint64_t i, uin
     return r.vals + o;
 }
 
-/* u64 big-endian view of key bytes [off, off+8), zero-padded past the end —
- * monotone with bytewise key order among keys sharing the first `off` bytes */
-__device__ static inline uint64_t key_window_u64(const DevRun &r, uint64_t idx, uint64_t off)
-{
-    uint64_t ko = r.koff[idx], kl = r.koff[idx + 1] - ko;
-    const uint8_t *k = r.keys + ko;
-    uint64_t v = 0;
-    uint64_t n = kl > off ? kl - off : 0;
-    if (n > 8)
-        n = 8;
-    for (uint64_t b = 0; b < n; b++)
-        v |= (uint64_t)k[off + b] << (8 * (7 - b));
-    return v;
-}
-
-__device__ static inline uint64_t bytes_u64(const uint8_t *k, uint64_t kl, uint64_t off)
-{
-    uint64_t v = 0;
-    uint64_t n = kl > off ? kl - off : 0;
-    if (n > 8)
-        n = 8;
-    for (uint64_t b = 0; b < n; b++)
-        v |= (uint64_t)k[off + b] << (8 * (7 - b));
-    return v;
-}
-
 /* first index in [lo,hi) with key >= target */
 __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
@@ -192,91 +166,6 @@ __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, 
             hi = mid;
     }
     return lo;
-}
-
-/* interpolated bound inside a small window (KEPT AS A MEASURED NEGATIVE
- * RESULT, unused): parity-correct but 1.9x SLOWER than plain binary search
- * on the narrowed windows (rank 4.08 vs 2.17 ms) — the per-lane gallop
- * paths diverge across the 64-wide wave and the prefix arithmetic costs
- * more than ~6 L1-hot binary probes. */
-template <int UPPER>
-__device__ static uint64_t dev_interp_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
-                                            uint64_t lo, uint64_t hi)
-{
-    uint64_t n = hi - lo;
-    if (n < 24)
-        return UPPER ? dev_upper_bound(r, key, klen, lo, hi)
-                     : dev_lower_bound(r, key, klen, lo, hi);
-    /* common prefix of the window's first and last key */
-    uint64_t al, bl;
-    const uint8_t *a = run_key(r, lo, &al);
-    const uint8_t *b = run_key(r, hi - 1, &bl);
-    uint64_t m = al < bl ? al : bl;
-    uint64_t cpl = 0;
-    while (cpl < m && a[cpl] == b[cpl])
-        cpl++;
-    uint64_t va = bytes_u64(a, al, cpl);
-    uint64_t vb = bytes_u64(b, bl, cpl);
-    uint64_t vk = bytes_u64(key, klen, cpl);
-    if (vb <= va || vk <= va || vk >= vb)
-        return UPPER ? dev_upper_bound(r, key, klen, lo, hi)
-                     : dev_lower_bound(r, key, klen, lo, hi);
-    uint64_t est = lo + (uint64_t)((double)(vk - va) / (double)(vb - va) * (double)(n - 1));
-    if (est < lo)
-        est = lo;
-    if (est >= hi)
-        est = hi - 1;
-    /* establish a bracket [blo, bhi) around the answer by galloping */
-    uint64_t ml;
-    const uint8_t *mk = run_key(r, est, &ml);
-    int c = dev_key_cmp(mk, ml, key, klen);
-    int go_right = UPPER ? (c <= 0) : (c < 0);
-    uint64_t blo, bhi;
-    if (go_right) {
-        blo = est + 1;
-        uint64_t step = 8;
-        bhi = blo;
-        while (bhi < hi) {
-            uint64_t probe = bhi + step < hi ? bhi + step : hi;
-            if (probe == hi) {
-                bhi = hi;
-                break;
-            }
-            const uint8_t *pk = run_key(r, probe, &ml);
-            c = dev_key_cmp(pk, ml, key, klen);
-            if (UPPER ? (c > 0) : (c >= 0)) {
-                bhi = probe;
-                break;
-            }
-            blo = probe + 1;
-            bhi = probe + 1;
-            step <<= 1;
-        }
-        if (bhi < blo)
-            bhi = blo;
-    } else {
-        bhi = est;
-        uint64_t step = 8;
-        blo = bhi;
-        while (blo > lo) {
-            uint64_t probe = blo > lo + step ? blo - step : lo;
-            if (probe == lo) {
-                blo = lo;
-                break;
-            }
-            const uint8_t *pk = run_key(r, probe, &ml);
-            c = dev_key_cmp(pk, ml, key, klen);
-            if (UPPER ? (c <= 0) : (c < 0)) {
-                blo = probe + 1;
-                break;
-            }
-            bhi = probe;
-            blo = probe;
-            step <<= 1;
-        }
-    }
-    return UPPER ? dev_upper_bound(r, key, klen, blo, bhi)
-                 : dev_lower_bound(r, key, klen, blo, bhi);
 }
 
 /* ================= value codec (device) =================
