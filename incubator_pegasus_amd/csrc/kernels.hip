@@ -153,43 +153,6 @@ __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, 
     }
     return lo;
 }
-/* two independent bounded searches advanced in lockstep: the two probe
- * loads of each round are independent, so their ~L1/L2 latencies overlap
- * (the sequential per-run search loop serialized them).  up = 0 -> lower
- * bound, 1 -> upper bound (go right when cmp < up). */
-__device__ static void dev_dual_bound(const DevRun &r1, uint64_t lo1, uint64_t hi1, int up1,
-                                      const DevRun &r2, uint64_t lo2, uint64_t hi2, int up2,
-                                      const uint8_t *key, uint64_t klen, uint64_t *out1,
-                                      uint64_t *out2)
-{
-    while ((lo1 < hi1) | (lo2 < hi2)) {
-        uint64_t mid1 = 0, mid2 = 0, ml1 = 0, ml2 = 0;
-        const uint8_t *mk1 = nullptr, *mk2 = nullptr;
-        if (lo1 < hi1) {
-            mid1 = (lo1 + hi1) >> 1;
-            mk1 = run_key(r1, mid1, &ml1);
-        }
-        if (lo2 < hi2) {
-            mid2 = (lo2 + hi2) >> 1;
-            mk2 = run_key(r2, mid2, &ml2);
-        }
-        if (mk1) {
-            if (dev_key_cmp(mk1, ml1, key, klen) < up1)
-                lo1 = mid1 + 1;
-            else
-                hi1 = mid1;
-        }
-        if (mk2) {
-            if (dev_key_cmp(mk2, ml2, key, klen) < up2)
-                lo2 = mid2 + 1;
-            else
-                hi2 = mid2;
-        }
-    }
-    *out1 = lo1;
-    *out2 = lo2;
-}
-
 /* first index in [lo,hi) with key > target */
 __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
@@ -854,29 +817,13 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                 b0 = bt + (bt_off[r] + j) * (uint64_t)R;
                 b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
             }
-            int qs[RRDB_MAX_RUNS];
-            int nq = 0;
-            for (int q = 0; q < R; q++)
-                if (q != r)
-                    qs[nq++] = q;
-            uint64_t pos[RRDB_MAX_RUNS];
-            for (int x = 0; x < nq; x += 2) {
-                int qa = qs[x];
-                uint64_t alo = b0 ? b0[qa] : lo[qa], ahi = b1 ? b1[qa] : hi[qa];
-                if (x + 1 < nq) {
-                    int qb = qs[x + 1];
-                    uint64_t blo = b0 ? b0[qb] : lo[qb], bhi = b1 ? b1[qb] : hi[qb];
-                    dev_dual_bound(runs[qa], alo, ahi, qa > r ? 1 : 0, runs[qb], blo, bhi,
-                                   qb > r ? 1 : 0, k, kl, &pos[qa], &pos[qb]);
-                } else {
-                    pos[qa] = (qa > r) ? dev_upper_bound(runs[qa], k, kl, alo, ahi)
-                                       : dev_lower_bound(runs[qa], k, kl, alo, ahi);
-                }
-            }
-            for (int x = 0; x < nq; x++) {
-                int q = qs[x];
+            for (int q = 0; q < R; q++) {
+                if (q == r)
+                    continue;
+                uint64_t qlo = b0 ? b0[q] : lo[q];
+                uint64_t qhi = b1 ? b1[q] : hi[q];
                 if (q > r) {
-                    uint64_t ub = pos[q];
+                    uint64_t ub = dev_upper_bound(runs[q], k, kl, qlo, qhi);
                     if (!shadow && ub > lo[q]) {
                         uint64_t pl;
                         const uint8_t *pk = run_key(runs[q], ub - 1, &pl);
@@ -885,7 +832,7 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
                     }
                     rank += ub - lo[q];
                 } else {
-                    rank += pos[q] - lo[q];
+                    rank += dev_lower_bound(runs[q], k, kl, qlo, qhi) - lo[q];
                 }
             }
             uint8_t ch;
