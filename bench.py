@@ -65,12 +65,32 @@ def _cpu_compact_worker(params):
         op.close()
 
 
-def build_partition_data(keys_pp, n_runs, rank, pidx, value_len=100):
+def build_partition_data(keys_pp, n_runs, rank, pidx, value_len=100, ttl_fraction=0.0,
+                         ttl_expire_ts=0):
     from incubator_pegasus_amd import data as D
 
     return D.build_point_table_runs(
         keys_pp, n_runs, seed=D.DEFAULT_SEED + rank * 1000 + pidx,
-        value_len=value_len, dup_fraction=0.10, delete_fraction=0.02)
+        value_len=value_len, dup_fraction=0.10, delete_fraction=0.02,
+        ttl_fraction=ttl_fraction, ttl_expire_ts=ttl_expire_ts)
+
+
+def rules_profile_envs(epoch_now):
+    """configs[4]: default_ttl + user-specified compaction rules
+    (pegasus_compaction_filter) evaluated on every surviving key."""
+    import json as _json
+
+    ops = {"ops": [
+        {"type": "COT_DELETE", "params": "", "rules": [
+            {"type": "FRT_HASHKEY_PATTERN",
+             "params": _json.dumps({"pattern": "777", "match_type": "SMT_MATCH_ANYWHERE"})}]},
+        {"type": "COT_UPDATE_TTL",
+         "params": _json.dumps({"type": "UTOT_FROM_NOW", "value": 5000}),
+         "rules": [
+            {"type": "FRT_TTL_RANGE",
+             "params": _json.dumps({"start_ttl": 3000, "stop_ttl": 4000})}]},
+    ]}
+    return {"default_ttl": "3600", "user_specified_compaction": _json.dumps(ops)}
 
 
 def main():
@@ -86,6 +106,9 @@ def main():
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
     ap.add_argument("--bt-shift", type=int, default=6)
     ap.add_argument("--rank-mode", choices=["global", "lds"], default="global")
+    ap.add_argument("--rules-profile", action="store_true",
+                    help="configs[4] shape: TTL'd data + default_ttl + user "
+                         "delete/update-TTL compaction rules evaluated per key")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -118,12 +141,18 @@ def main():
     t0 = time.time()
     parts = []
     total_records = 0
+    epoch_now = 1_000_000
     for p in range(args.partitions):
-        runs = build_partition_data(keys_pp, args.runs, rank, p)
+        runs = build_partition_data(
+            keys_pp, args.runs, rank, p,
+            ttl_fraction=0.2 if args.rules_profile else 0.0,
+            ttl_expire_ts=epoch_now - 100 if args.rules_profile else 0)
         eng = hip.open(1, p, local_rank)
         eng.set_envs({"engine.emit_mode": args.emit_mode,
                       "engine.bt_shift": str(args.bt_shift),
                       "engine.rank_mode": args.rank_mode})
+        if args.rules_profile:
+            eng.set_envs(rules_profile_envs(epoch_now))
         for r in runs:
             eng.ingest_run_arrays(np.ascontiguousarray(r["keys"]), r["koff"],
                                   np.ascontiguousarray(r["vals"]), r["voff"], r["sk"])
@@ -133,8 +162,6 @@ def main():
             log(f"partition 0 ready ({time.time()-t0:.1f}s, {eng.num_records()} records)")
     log(f"ingest done: {total_records} records on GPU {local_rank} "
         f"({time.time()-t0:.1f}s)")
-
-    epoch_now = 1_000_000
 
     def barrier_sync():
         torch.cuda.synchronize()
@@ -408,8 +435,9 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": "compact-50M-16part-8run" if args.keys == 50_000_000 else
-                            f"compact-{args.keys}-{args.partitions}part-{args.runs}run",
+                "workload": ("compact-50M-16part-8run" if args.keys == 50_000_000 else
+                             f"compact-{args.keys}-{args.partitions}part-{args.runs}run")
+                            + ("-ttl-rules" if args.rules_profile else ""),
                 "partitions_per_gpu": args.partitions,
                 "keys_per_gpu": total_records,
                 "runs_per_partition": args.runs,
