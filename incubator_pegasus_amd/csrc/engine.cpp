@@ -457,6 +457,20 @@ int key_cmp(const std::string &a, const std::string &b)
 
 [[maybe_unused]] uint32_t hdr_len(uint32_t ver) { return ver == 0 ? 4u : (ver == 1 ? 12u : 13u); }
 
+/* constant key stride lets device searches skip the offset-pair loads */
+static uint32_t detect_fixed_klen(const uint64_t *koff, uint64_t n)
+{
+    if (n == 0)
+        return 0;
+    uint64_t s = koff[1] - koff[0];
+    if (s == 0 || s > UINT32_MAX)
+        return 0;
+    for (uint64_t i = 1; i < n; i++)
+        if (koff[i + 1] - koff[i] != s)
+            return 0;
+    return (uint32_t)s;
+}
+
 /* ================ engine ================ */
 struct RunBuf {
     uint8_t *keys = nullptr;
@@ -467,6 +481,7 @@ struct RunBuf {
     uint64_t n = 0;
     uint64_t *bloom = nullptr; /* blocked bloom over full keys (may be null) */
     uint64_t bloom_blocks = 0;
+    uint32_t fixed_klen = 0; /* nonzero when every key in the run has this length */
 };
 
 struct HipScanCtx {
@@ -565,7 +580,8 @@ struct HipEngine {
             std::vector<DevRun> h(runs.size() ? runs.size() : 1);
             for (size_t i = 0; i < runs.size(); i++)
                 h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
-                              runs[i].n,    runs[i].bloom, runs[i].bloom_blocks};
+                              runs[i].n,    runs[i].bloom, runs[i].bloom_blocks,
+                              runs[i].fixed_klen};
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
@@ -844,6 +860,7 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     r.vals = e->upload_bytes(vals.data(), vals.size());
     r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size() * 8);
     r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
+    r.fixed_klen = detect_fixed_klen(koff.data(), r.n);
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
@@ -998,6 +1015,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     r.vals = e->upload_bytes(values, val_offs[n]);
     r.voff = (uint64_t *)e->upload_bytes(val_offs, (n + 1) * 8);
     r.sk = (uint64_t *)e->upload_bytes(seq_kind, n * 8);
+    r.fixed_klen = detect_fixed_klen(key_offs, n);
     uint64_t mx = 0;
     for (uint64_t i = 0; i < n; i++)
         mx = std::max(mx, seq_kind[i] >> 1);
@@ -1961,6 +1979,12 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     bool keep_inputs = opts && opts->keep_inputs;
     if (n_out > 0) {
         nr.n = n_out;
+        /* output keys are verbatim copies of input keys, so a stride shared
+         * by every input run is preserved */
+        nr.fixed_klen = e->runs.empty() ? 0 : e->runs[0].fixed_klen;
+        for (auto &ir : e->runs)
+            if (ir.fixed_klen != nr.fixed_klen)
+                nr.fixed_klen = 0;
         if (keep_inputs) {
             /* output is dropped at the end of the pass: pooled temporaries */
             nr.keys = e->talloc<uint8_t>(kbytes);
@@ -2233,6 +2257,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         r.vals = e->upload_bytes(vals.data(), vals.size());
         r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size());
         r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size());
+        r.fixed_klen = detect_fixed_klen((const uint64_t *)koff.data(), r.n);
         e->build_bloom(r); /* blooms are rebuilt, not serialized */
         e->runs.push_back(r);
     }

@@ -32,6 +32,9 @@ struct DevRun {
      * blocks, 6 probe bits; null = no filter (search every run). */
     const uint64_t *bloom;
     uint64_t bloom_blocks;
+    /* all keys in the run share this length (0 = variable): searches then
+     * compute key addresses directly instead of loading offset pairs */
+    uint32_t fixed_klen;
 };
 
 /* flattened user compaction rules/ops (device-resident)

@@ -143,6 +143,17 @@ __device__ static inline const uint8_t *run_val(const DevRun &r, uint64_t i, uin
 __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
 {
+    uint64_t fk = r.fixed_klen;
+    if (fk) { /* fixed stride: no offset-pair loads per probe */
+        while (lo < hi) {
+            uint64_t mid = (lo + hi) >> 1;
+            if (dev_key_cmp(r.keys + mid * fk, fk, key, klen) < 0)
+                lo = mid + 1;
+            else
+                hi = mid;
+        }
+        return lo;
+    }
     while (lo < hi) {
         uint64_t mid = (lo + hi) >> 1, ml;
         const uint8_t *mk = run_key(r, mid, &ml);
@@ -157,6 +168,17 @@ __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, 
 __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
 {
+    uint64_t fk = r.fixed_klen;
+    if (fk) {
+        while (lo < hi) {
+            uint64_t mid = (lo + hi) >> 1;
+            if (dev_key_cmp(r.keys + mid * fk, fk, key, klen) <= 0)
+                lo = mid + 1;
+            else
+                hi = mid;
+        }
+        return lo;
+    }
     while (lo < hi) {
         uint64_t mid = (lo + hi) >> 1, ml;
         const uint8_t *mk = run_key(r, mid, &ml);
