@@ -35,6 +35,10 @@ struct DevRun {
     /* all keys in the run share this length (0 = variable): searches then
      * compute key addresses directly instead of loading offset pairs */
     uint32_t fixed_klen;
+    /* every key in a sorted run shares the lcp of its first and last key;
+     * searches skip these leading bytes (multiple of 8, <=16) after one
+     * per-call check that the query shares them too */
+    uint32_t pfx_skip;
 };
 
 /* flattened user compaction rules/ops (device-resident)

@@ -139,15 +139,28 @@ __device__ static inline const uint8_t *run_val(const DevRun &r, uint64_t i, uin
     return r.vals + o;
 }
 
+/* once per search: skip the run's shared key prefix iff the query shares it
+ * too (sorted order guarantees every run key carries it and is >= that long) */
+__device__ static inline uint64_t pfx_engage(const DevRun &r, const uint8_t *key, uint64_t klen)
+{
+    uint64_t ps = r.pfx_skip;
+    if (!ps || klen < ps || dev_key_cmp(key, ps, r.keys, ps) != 0)
+        return 0;
+    return ps;
+}
 /* first index in [lo,hi) with key >= target */
 __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
 {
     uint64_t fk = r.fixed_klen;
+    uint64_t ps = pfx_engage(r, key, klen);
+    const uint8_t *q = key + ps;
+    uint64_t ql = klen - ps;
     if (fk) { /* fixed stride: no offset-pair loads per probe */
+        uint64_t fl = fk - ps;
         while (lo < hi) {
             uint64_t mid = (lo + hi) >> 1;
-            if (dev_key_cmp(r.keys + mid * fk, fk, key, klen) < 0)
+            if (dev_key_cmp(r.keys + mid * fk + ps, fl, q, ql) < 0)
                 lo = mid + 1;
             else
                 hi = mid;
@@ -157,7 +170,7 @@ __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, 
     while (lo < hi) {
         uint64_t mid = (lo + hi) >> 1, ml;
         const uint8_t *mk = run_key(r, mid, &ml);
-        if (dev_key_cmp(mk, ml, key, klen) < 0)
+        if (dev_key_cmp(mk + ps, ml - ps, q, ql) < 0)
             lo = mid + 1;
         else
             hi = mid;
@@ -169,10 +182,14 @@ __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, 
                                            uint64_t lo, uint64_t hi)
 {
     uint64_t fk = r.fixed_klen;
+    uint64_t ps = pfx_engage(r, key, klen);
+    const uint8_t *q = key + ps;
+    uint64_t ql = klen - ps;
     if (fk) {
+        uint64_t fl = fk - ps;
         while (lo < hi) {
             uint64_t mid = (lo + hi) >> 1;
-            if (dev_key_cmp(r.keys + mid * fk, fk, key, klen) <= 0)
+            if (dev_key_cmp(r.keys + mid * fk + ps, fl, q, ql) <= 0)
                 lo = mid + 1;
             else
                 hi = mid;
@@ -182,7 +199,7 @@ __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, 
     while (lo < hi) {
         uint64_t mid = (lo + hi) >> 1, ml;
         const uint8_t *mk = run_key(r, mid, &ml);
-        if (dev_key_cmp(mk, ml, key, klen) <= 0)
+        if (dev_key_cmp(mk + ps, ml - ps, q, ql) <= 0)
             lo = mid + 1;
         else
             hi = mid;
