@@ -645,3 +645,51 @@ def test_multi_get_batch_parity(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+def test_prefix_skip_edge_queries(pair):
+    """Runs whose keys share a long (>=8B) prefix engage the device
+    prefix-skip search path; queries that diverge inside, fall short of, or
+    exactly equal the shared prefix must take the fallback compare and still
+    match the oracle bit for bit."""
+    o, g = pair
+    now = 1000
+    # hashkeys "commonprefix-NNNN" -> every raw key shares >=14 leading bytes
+    hks = [f"commonprefix-{i:04d}".encode() for i in range(64)]
+    seq = 1
+    records = []
+    for i, hk in enumerate(hks):
+        records.append((D.generate_key(hk, b""), D.encode_value(b"v%d" % i, 0, i + 1, 1), seq, 0))
+        seq += 1
+    records.sort()
+    o.ingest_run(records)
+    g.ingest_run(records)
+    # second run: varying sortkey lengths -> variable stride, shared prefix;
+    # exercises the offset-pair search branch with prefix skip
+    records2 = []
+    for i, hk in enumerate(hks[:32]):
+        sk = b"s" * (i % 5)
+        records2.append((D.generate_key(hk, sk), D.encode_value(b"w%d" % i, 0, 0, 1), seq, 0))
+        seq += 1
+    records2.sort()
+    o.ingest_run(records2)
+    g.ingest_run(records2)
+    probes = (
+        [D.generate_key(hk, b"") for hk in hks]       # shares prefix (skip path)
+        + [D.generate_key(b"commonprefiy", b""),      # diverges after 8B
+           D.generate_key(b"aommonprefix-0000", b""), # diverges in first word, below
+           D.generate_key(b"zommonprefix-0000", b""), # diverges in first word, above
+           D.generate_key(b"common", b""),            # shorter than the prefix
+           D.generate_key(b"commonprefix-", b""),     # exactly the shared hashkey stem
+           D.generate_key(b"commonprefix-0031", b"x")]  # longer than a run key
+    )
+    for k in probes:
+        assert o.get(k, now) == g.get(k, now), k
+    assert o.batch_get(probes, now) == g.batch_get(probes, now)
+    for hk in [b"commonprefix-0005", b"common", b"zommonprefix"]:
+        assert o.sortkey_count(hk, now) == g.sortkey_count(hk, now), hk
+    o.manual_compact(now)
+    g.manual_compact(now)
+    assert o.num_records() == g.num_records()
+    for k in probes:
+        assert o.get(k, now) == g.get(k, now), k
