@@ -551,7 +551,7 @@ struct HipEngine {
                           (env "engine.emit_mode": chunked|rank|input) */
     int rank_mode = 0; /* 0 = global searches + bound-table narrowing,
                           1 = LDS-staged block rank (env "engine.rank_mode") */
-    int bt_shift = 6;  /* bound-table block = 1<<bt_shift records (env
+    int bt_shift = 5;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
