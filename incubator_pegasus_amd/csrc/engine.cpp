@@ -459,13 +459,18 @@ int key_cmp(const std::string &a, const std::string &b)
 
 /* sorted order => every key in a run shares lcp(first key, last key); device
  * searches skip that many leading bytes (whole 8B words, capped at 16) */
-static uint32_t lcp_skip16(const uint8_t *a, uint64_t al, const uint8_t *b, uint64_t bl)
+static uint32_t lcp_exact32(const uint8_t *a, uint64_t al, const uint8_t *b, uint64_t bl)
 {
-    uint32_t m = (uint32_t)std::min<uint64_t>(std::min(al, bl), 16);
+    uint32_t m = (uint32_t)std::min<uint64_t>(std::min(al, bl), 32);
     uint32_t i = 0;
     while (i < m && a[i] == b[i])
         i++;
-    return i & ~7u;
+    return i;
+}
+static void set_pfx(uint32_t lcp, uint32_t *pfx_skip, uint32_t *lcp_exact)
+{
+    *lcp_exact = lcp;
+    *pfx_skip = std::min(lcp, 16u) & ~7u;
 }
 
 /* constant key stride lets device searches skip the offset-pair loads */
@@ -494,6 +499,7 @@ struct RunBuf {
     uint64_t bloom_blocks = 0;
     uint32_t fixed_klen = 0; /* nonzero when every key in the run has this length */
     uint32_t pfx_skip = 0;   /* lcp(first,last) floored to 8B words, <=16 */
+    uint32_t lcp_exact = 0;  /* exact lcp(first,last), capped at 32 */
 };
 
 struct HipScanCtx {
@@ -593,7 +599,7 @@ struct HipEngine {
             for (size_t i = 0; i < runs.size(); i++)
                 h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
                               runs[i].n,    runs[i].bloom, runs[i].bloom_blocks,
-                              runs[i].fixed_klen, runs[i].pfx_skip};
+                              runs[i].fixed_klen, runs[i].pfx_skip, runs[i].lcp_exact};
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
@@ -873,9 +879,10 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size() * 8);
     r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
     r.fixed_klen = detect_fixed_klen(koff.data(), r.n);
-    r.pfx_skip = lcp_skip16((const uint8_t *)keys.data(), koff[1] - koff[0],
-                            (const uint8_t *)keys.data() + koff[r.n - 1],
-                            koff[r.n] - koff[r.n - 1]);
+    set_pfx(lcp_exact32((const uint8_t *)keys.data(), koff[1] - koff[0],
+                        (const uint8_t *)keys.data() + koff[r.n - 1],
+                        koff[r.n] - koff[r.n - 1]),
+            &r.pfx_skip, &r.lcp_exact);
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
@@ -1031,8 +1038,9 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     r.voff = (uint64_t *)e->upload_bytes(val_offs, (n + 1) * 8);
     r.sk = (uint64_t *)e->upload_bytes(seq_kind, n * 8);
     r.fixed_klen = detect_fixed_klen(key_offs, n);
-    r.pfx_skip = lcp_skip16(keys + key_offs[0], key_offs[1] - key_offs[0],
-                            keys + key_offs[n - 1], key_offs[n] - key_offs[n - 1]);
+    set_pfx(lcp_exact32(keys + key_offs[0], key_offs[1] - key_offs[0],
+                        keys + key_offs[n - 1], key_offs[n] - key_offs[n - 1]),
+            &r.pfx_skip, &r.lcp_exact);
     uint64_t mx = 0;
     for (uint64_t i = 0; i < n; i++)
         mx = std::max(mx, seq_kind[i] >> 1);
@@ -2083,14 +2091,14 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         e->runs.clear();
         if (n_out > 0) {
             uint64_t fo2[2], lo2[2];
-            uint8_t h16[16], t16[16];
+            uint8_t h32[32], t32[32];
             HIP_OK(hipMemcpy(fo2, nr.koff, 16, hipMemcpyDeviceToHost));
             HIP_OK(hipMemcpy(lo2, nr.koff + (n_out - 1), 16, hipMemcpyDeviceToHost));
-            uint64_t fl = std::min<uint64_t>(fo2[1] - fo2[0], 16);
-            uint64_t ll = std::min<uint64_t>(lo2[1] - lo2[0], 16);
-            HIP_OK(hipMemcpy(h16, nr.keys + fo2[0], fl ? fl : 1, hipMemcpyDeviceToHost));
-            HIP_OK(hipMemcpy(t16, nr.keys + lo2[0], ll ? ll : 1, hipMemcpyDeviceToHost));
-            nr.pfx_skip = lcp_skip16(h16, fl, t16, ll);
+            uint64_t fl = std::min<uint64_t>(fo2[1] - fo2[0], 32);
+            uint64_t ll = std::min<uint64_t>(lo2[1] - lo2[0], 32);
+            HIP_OK(hipMemcpy(h32, nr.keys + fo2[0], fl ? fl : 1, hipMemcpyDeviceToHost));
+            HIP_OK(hipMemcpy(t32, nr.keys + lo2[0], ll ? ll : 1, hipMemcpyDeviceToHost));
+            set_pfx(lcp_exact32(h32, fl, t32, ll), &nr.pfx_skip, &nr.lcp_exact);
             e->build_bloom(nr);
             e->runs.push_back(nr);
         }
@@ -2286,8 +2294,9 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         r.fixed_klen = detect_fixed_klen((const uint64_t *)koff.data(), r.n);
         if (r.n) {
             const uint64_t *ko = (const uint64_t *)koff.data();
-            r.pfx_skip = lcp_skip16(keys.data() + ko[0], ko[1] - ko[0],
-                                    keys.data() + ko[r.n - 1], ko[r.n] - ko[r.n - 1]);
+            set_pfx(lcp_exact32(keys.data() + ko[0], ko[1] - ko[0],
+                                keys.data() + ko[r.n - 1], ko[r.n] - ko[r.n - 1]),
+                    &r.pfx_skip, &r.lcp_exact);
         }
         e->build_bloom(r); /* blooms are rebuilt, not serialized */
         e->runs.push_back(r);

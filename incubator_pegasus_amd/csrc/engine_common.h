@@ -39,6 +39,10 @@ struct DevRun {
      * searches skip these leading bytes (multiple of 8, <=16) after one
      * per-call check that the query shares them too */
     uint32_t pfx_skip;
+    /* exact (unfloored) shared-prefix length, capped at 32.  When the run is
+     * fixed-stride and lcp_exact >= fixed_klen-8, the varying suffix fits one
+     * u64 and probes compare a single big-endian word at key[klen-8..klen) */
+    uint32_t lcp_exact;
 };
 
 /* flattened user compaction rules/ops (device-resident)

@@ -148,11 +148,43 @@ __device__ static inline uint64_t pfx_engage(const DevRun &r, const uint8_t *key
         return 0;
     return ps;
 }
+/* single-word probe mode: fixed stride, the varying suffix fits u64, and the
+ * query is same-length sharing the run's constant first klen-8 bytes.  The
+ * whole bytewise compare then reduces to one big-endian word compare. */
+__device__ static inline bool word_probe_engage(const DevRun &r, const uint8_t *key,
+                                                uint64_t klen, uint64_t *qw)
+{
+    uint64_t fk = r.fixed_klen;
+    if (!fk || fk < 8 || klen != fk || r.lcp_exact + 8 < fk)
+        return false;
+    if (fk > 8 && dev_key_cmp(key, fk - 8, r.keys, fk - 8) != 0)
+        return false;
+    uint64_t w;
+    __builtin_memcpy(&w, key + fk - 8, 8);
+    *qw = __builtin_bswap64(w);
+    return true;
+}
+__device__ static inline uint64_t run_tail_word(const DevRun &r, uint64_t fk, uint64_t mid)
+{
+    uint64_t w;
+    __builtin_memcpy(&w, r.keys + mid * fk + fk - 8, 8);
+    return __builtin_bswap64(w);
+}
 /* first index in [lo,hi) with key >= target */
 __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
 {
-    uint64_t fk = r.fixed_klen;
+    uint64_t fk = r.fixed_klen, qw;
+    if (word_probe_engage(r, key, klen, &qw)) {
+        while (lo < hi) {
+            uint64_t mid = (lo + hi) >> 1;
+            if (run_tail_word(r, fk, mid) < qw)
+                lo = mid + 1;
+            else
+                hi = mid;
+        }
+        return lo;
+    }
     uint64_t ps = pfx_engage(r, key, klen);
     const uint8_t *q = key + ps;
     uint64_t ql = klen - ps;
@@ -181,7 +213,17 @@ __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, 
 __device__ static uint64_t dev_upper_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
                                            uint64_t lo, uint64_t hi)
 {
-    uint64_t fk = r.fixed_klen;
+    uint64_t fk = r.fixed_klen, qw;
+    if (word_probe_engage(r, key, klen, &qw)) {
+        while (lo < hi) {
+            uint64_t mid = (lo + hi) >> 1;
+            if (run_tail_word(r, fk, mid) <= qw)
+                lo = mid + 1;
+            else
+                hi = mid;
+        }
+        return lo;
+    }
     uint64_t ps = pfx_engage(r, key, klen);
     const uint8_t *q = key + ps;
     uint64_t ql = klen - ps;
