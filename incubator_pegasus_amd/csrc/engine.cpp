@@ -61,6 +61,7 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       uint8_t *, hipStream_t);
 void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
 void launch_bloom_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
+void launch_build_tails(const uint8_t *, uint64_t, uint64_t, uint64_t *, hipStream_t);
 void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64_t *,
                              const uint64_t *, uint64_t, const CompactParams &,
                              const uint64_t *, const uint64_t *, uint64_t *, uint64_t *,
@@ -500,6 +501,7 @@ struct RunBuf {
     uint32_t fixed_klen = 0; /* nonzero when every key in the run has this length */
     uint32_t pfx_skip = 0;   /* lcp(first,last) floored to 8B words, <=16 */
     uint32_t lcp_exact = 0;  /* exact lcp(first,last), capped at 32 */
+    uint64_t *tails = nullptr; /* packed BE tail words (word-probe mode) */
 };
 
 struct HipScanCtx {
@@ -571,7 +573,19 @@ struct HipEngine {
         (void)hipFree(r.sk);
         if (r.bloom)
             (void)hipFree(r.bloom);
+        if (r.tails)
+            (void)hipFree(r.tails);
         r = RunBuf();
+    }
+
+    /* packed 8B-strided probe words for the single-word search mode */
+    void build_tails(RunBuf &r)
+    {
+        if (r.n == 0 || r.fixed_klen < 8 || r.lcp_exact + 8 < r.fixed_klen)
+            return;
+        HIP_OK(hipMalloc(&r.tails, r.n * 8));
+        launch_build_tails(r.keys, r.fixed_klen, r.n, r.tails, stream);
+        HIP_OK(hipStreamSynchronize(stream));
     }
 
     /* §8(f)3: ~10 bits/key blocked bloom, built once per run on device */
@@ -599,7 +613,8 @@ struct HipEngine {
             for (size_t i = 0; i < runs.size(); i++)
                 h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
                               runs[i].n,    runs[i].bloom, runs[i].bloom_blocks,
-                              runs[i].fixed_klen, runs[i].pfx_skip, runs[i].lcp_exact};
+                              runs[i].fixed_klen, runs[i].pfx_skip, runs[i].lcp_exact,
+                              runs[i].tails};
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
@@ -883,6 +898,7 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
                         (const uint8_t *)keys.data() + koff[r.n - 1],
                         koff[r.n] - koff[r.n - 1]),
             &r.pfx_skip, &r.lcp_exact);
+    e->build_tails(r);
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
@@ -1045,6 +1061,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     for (uint64_t i = 0; i < n; i++)
         mx = std::max(mx, seq_kind[i] >> 1);
     e->next_seq_floor = mx + 1;
+    e->build_tails(r);
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
@@ -2099,6 +2116,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             HIP_OK(hipMemcpy(h32, nr.keys + fo2[0], fl ? fl : 1, hipMemcpyDeviceToHost));
             HIP_OK(hipMemcpy(t32, nr.keys + lo2[0], ll ? ll : 1, hipMemcpyDeviceToHost));
             set_pfx(lcp_exact32(h32, fl, t32, ll), &nr.pfx_skip, &nr.lcp_exact);
+            e->build_tails(nr);
             e->build_bloom(nr);
             e->runs.push_back(nr);
         }
@@ -2298,7 +2316,8 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
                                 keys.data() + ko[r.n - 1], ko[r.n] - ko[r.n - 1]),
                     &r.pfx_skip, &r.lcp_exact);
         }
-        e->build_bloom(r); /* blooms are rebuilt, not serialized */
+        e->build_tails(r); /* rebuilt, not serialized (like blooms) */
+        e->build_bloom(r);
         e->runs.push_back(r);
     }
     e->next_seq_floor = floor_;

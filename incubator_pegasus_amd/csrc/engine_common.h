@@ -43,6 +43,10 @@ struct DevRun {
      * fixed-stride and lcp_exact >= fixed_klen-8, the varying suffix fits one
      * u64 and probes compare a single big-endian word at key[klen-8..klen) */
     uint32_t lcp_exact;
+    /* packed big-endian tail words (key[klen-8..klen) bswapped), built at run
+     * creation when the single-word probe mode is eligible: 8B-strided probe
+     * loads instead of klen-strided ones (null when ineligible) */
+    const uint64_t *tails;
 };
 
 /* flattened user compaction rules/ops (device-resident)

@@ -155,7 +155,7 @@ __device__ static inline bool word_probe_engage(const DevRun &r, const uint8_t *
                                                 uint64_t klen, uint64_t *qw)
 {
     uint64_t fk = r.fixed_klen;
-    if (!fk || fk < 8 || klen != fk || r.lcp_exact + 8 < fk)
+    if (!r.tails || klen != fk)
         return false;
     if (fk > 8 && dev_key_cmp(key, fk - 8, r.keys, fk - 8) != 0)
         return false;
@@ -166,9 +166,27 @@ __device__ static inline bool word_probe_engage(const DevRun &r, const uint8_t *
 }
 __device__ static inline uint64_t run_tail_word(const DevRun &r, uint64_t fk, uint64_t mid)
 {
-    uint64_t w;
-    __builtin_memcpy(&w, r.keys + mid * fk + fk - 8, 8);
-    return __builtin_bswap64(w);
+    (void)fk;
+    return r.tails[mid];
+}
+
+__global__ void k_build_tails(const uint8_t *keys, uint64_t fk, uint64_t n, uint64_t *tails)
+{
+    for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t w;
+        __builtin_memcpy(&w, keys + i * fk + fk - 8, 8);
+        tails[i] = __builtin_bswap64(w);
+    }
+}
+void launch_build_tails(const uint8_t *keys, uint64_t fk, uint64_t n, uint64_t *tails,
+                        hipStream_t s)
+{
+    uint64_t blocks = (n + 255) / 256;
+    if (blocks > 2048)
+        blocks = 2048;
+    hipLaunchKernelGGL(k_build_tails, dim3((uint32_t)blocks), dim3(256), 0, s, keys, fk, n,
+                       tails);
 }
 /* first index in [lo,hi) with key >= target */
 __device__ static uint64_t dev_lower_bound(const DevRun &r, const uint8_t *key, uint64_t klen,
