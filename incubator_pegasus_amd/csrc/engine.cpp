@@ -84,7 +84,8 @@ void launch_emit_compact_chunked(const DevRun *, const uint64_t *, uint64_t, con
                                  const uint64_t *, const uint64_t *, uint32_t, uint64_t,
                                  uint64_t, uint64_t, uint64_t *, uint64_t *, uint32_t *,
                                  uint32_t *, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
-                                 uint64_t *, hipStream_t);
+                                 uint64_t *, uint64_t *, uint64_t *,
+                                 hipStream_t);
 void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint64_t *,
                          const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
@@ -2047,11 +2048,15 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             uint64_t *d_row_vsrc = e->talloc<uint64_t>(n_out * 8);
             uint32_t *d_row_patch = e->talloc<uint32_t>(n_out * 4);
             uint32_t *d_row_expire = e->talloc<uint32_t>(n_out * 4);
+            uint64_t kanch = (((kbytes + 15) >> 4) >> 6) + 1;
+            uint64_t vanch = (((vbytes + 15) >> 4) >> 6) + 1;
+            uint64_t *d_kanchor = e->talloc<uint64_t>(kanch * 8);
+            uint64_t *d_vanchor = e->talloc<uint64_t>(vanch * 8);
             launch_emit_compact_chunked(dr, d_order, total, d_keepw, d_changed, d_new_expire,
                                         d_kpos, d_koffs, d_voffs, e->data_version, n_out, kbytes,
                                         vbytes, d_row_ksrc, d_row_vsrc, d_row_patch,
                                         d_row_expire, nr.keys, nr.vals, nr.koff, nr.voff, nr.sk,
-                                        e->stream);
+                                        d_kanchor, d_vanchor, e->stream);
         } else if (e->emit_mode == 1)
             launch_emit_compact_inmajor(dr, R, d_wp, total, d_rank_of, d_keepw, d_changed,
                                         d_new_expire, d_kpos, d_koffs, d_voffs,
