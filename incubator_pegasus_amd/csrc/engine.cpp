@@ -69,6 +69,11 @@ void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64
                              hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
+void launch_rank_compact_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
+                              const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
+                              uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
+                              uint64_t *, const uint64_t *, const uint64_t *, int,
+                              CompactStatsDev *, hipStream_t);
 void launch_rank_compact(const DevRun *, int, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                          uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *, uint64_t *,
@@ -559,7 +564,10 @@ struct HipEngine {
                           0 = rank-major waves, 1 = input-major waves
                           (env "engine.emit_mode": chunked|rank|input) */
     int rank_mode = 0; /* 0 = global searches + bound-table narrowing,
-                          1 = LDS-staged block rank (env "engine.rank_mode") */
+                          1 = LDS-staged full-key block rank,
+                          3 = LDS-staged tail-word rank (needs word-probe
+                          eligible runs with one shared cross-run prefix)
+                          (env "engine.rank_mode": global|lds|ldst) */
     int bt_shift = 5;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
 
@@ -1012,7 +1020,7 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
         } else if (k == "rocksdb.filter_type") {
             e->bloom_enabled = (v != "none"); /* common/prefix -> full-key bloom */
         } else if (k == "engine.rank_mode") {
-            e->rank_mode = (v == "lds") ? 1 : 0;
+            e->rank_mode = (v == "lds") ? 1 : (v == "ldst" ? 3 : 0);
         } else if (k == "engine.bt_shift") {
             int s_ = atoi(v.c_str());
             if (s_ >= 4 && s_ <= 16)
@@ -1990,9 +1998,35 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     if (R > 1 && total > 100000)
         e->build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
     HIP_OK(hipEventRecord(ev[0], e->stream));
-    launch_rank_compact(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw, d_changed,
-                        d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, e->bt_shift,
-                        d_stats, e->stream);
+    bool ldst_ok = false;
+    if (e->rank_mode == 3 && R > 1) {
+        /* eligibility: every run in single-word probe mode with the SAME
+         * stride, and all runs share one first-(fk-8)-byte prefix (then a
+         * tail-word compare decides cross-run order exactly) */
+        uint32_t fk = e->runs.empty() ? 0 : e->runs[0].fixed_klen;
+        ldst_ok = fk >= 8;
+        for (auto &rr : e->runs)
+            if (!rr.tails || rr.fixed_klen != fk || rr.n == 0)
+                ldst_ok = false;
+        if (ldst_ok && fk > 8) {
+            uint8_t p0[32], pi[32];
+            uint64_t pl = fk - 8 < 32 ? fk - 8 : 32;
+            HIP_OK(hipMemcpy(p0, e->runs[0].keys, pl, hipMemcpyDeviceToHost));
+            for (size_t ri = 1; ri < e->runs.size() && ldst_ok; ri++) {
+                HIP_OK(hipMemcpy(pi, e->runs[ri].keys, pl, hipMemcpyDeviceToHost));
+                if (memcmp(p0, pi, pl) != 0)
+                    ldst_ok = false;
+            }
+        }
+    }
+    if (ldst_ok)
+        launch_rank_compact_ldst(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw,
+                                 d_changed, d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off,
+                                 d_bt, e->bt_shift, d_stats, e->stream);
+    else
+        launch_rank_compact(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw, d_changed,
+                            d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, e->bt_shift,
+                            d_stats, e->stream);
     HIP_OK(hipEventRecord(ev[1], e->stream));
     }
     HIP_OK(hipEventRecord(ev[2], e->stream));

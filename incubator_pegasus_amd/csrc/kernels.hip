@@ -966,6 +966,189 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
     }
 }
 
+/* ================= LDS-tails cooperative rank (rank_mode=3) =================
+ * Same per-record rank+filter algorithm as k_rank_compact, but each 256-thread
+ * workgroup stages the tail words of all windows its 256-record span probes
+ * into LDS first.  Eligible only when every run is in single-word probe mode
+ * with one shared cross-run prefix (host checks; see engine.cpp), so a probe
+ * is one u64 compare and LDS holds ~2.3K words (~18KB) per workgroup — the
+ * earlier full-key LDS experiment died at 1 wg/CU; this one keeps ~6.  Probes
+ * whose first midpoints coincide across lanes become LDS broadcasts. */
+#define LDST_CAP 3072
+#define LDST_MAXR 16
+
+__global__ void __launch_bounds__(BLOCK) k_rank_compact_ldst(
+    const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi, const uint64_t *wprefix,
+    uint64_t total, CompactParams cp, uint64_t *order, uint64_t *keepw, uint8_t *changed,
+    uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of,
+    const uint64_t *bt_off, const uint64_t *bt, int bt_shift, CompactStatsDev *stats)
+{
+    __shared__ uint64_t s_tails[LDST_CAP];
+    __shared__ uint64_t s_winlo[LDST_MAXR], s_winhi[LDST_MAXR], s_base[LDST_MAXR];
+    __shared__ int s_fallback;
+
+    uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+    int lane = threadIdx.x % WAVE;
+
+    if (threadIdx.x == 0)
+        s_fallback = (bt == nullptr || R > LDST_MAXR) ? 1 : 0;
+    if (threadIdx.x < (unsigned)(R < LDST_MAXR ? R : LDST_MAXR)) {
+        s_winlo[threadIdx.x] = ~0ull;
+        s_winhi[threadIdx.x] = 0;
+    }
+    __syncthreads();
+
+    /* per-record identity + window pointers (kept in registers for phase 2) */
+    int r = 0;
+    uint64_t i = 0, qw = 0;
+    const uint64_t *b0 = nullptr, *b1 = nullptr;
+    if (t < total) {
+        while (wprefix[r + 1] <= t)
+            r++;
+        i = lo[r] + (t - wprefix[r]);
+        qw = runs[r].tails[i];
+        if (bt) {
+            uint64_t j = (i - lo[r]) >> bt_shift;
+            b0 = bt + (bt_off[r] + j) * (uint64_t)R;
+            b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
+        }
+    }
+    if (!s_fallback) { /* wave-reduced window union: one atomic per wave per run */
+        for (int q = 0; q < R; q++) {
+            uint64_t wl = (t < total && q != r && b0) ? b0[q] : ~0ull;
+            uint64_t wh = (t < total && q != r && b1) ? b1[q] : 0;
+            for (int d = WAVE / 2; d; d >>= 1) {
+                uint64_t o1 = __shfl_xor(wl, d);
+                if (o1 < wl)
+                    wl = o1;
+                uint64_t o2 = __shfl_xor(wh, d);
+                if (o2 > wh)
+                    wh = o2;
+            }
+            if (lane == 0 && wh > 0) {
+                atomicMin((unsigned long long *)&s_winlo[q], (unsigned long long)wl);
+                atomicMax((unsigned long long *)&s_winhi[q], (unsigned long long)wh);
+            }
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0 && !s_fallback) {
+        uint64_t tot = 0;
+        for (int q = 0; q < R; q++) {
+            s_base[q] = tot;
+            if (s_winhi[q] > s_winlo[q])
+                tot += s_winhi[q] - s_winlo[q];
+        }
+        if (tot > LDST_CAP)
+            s_fallback = 1;
+    }
+    __syncthreads();
+    if (!s_fallback) { /* stage the windows' tail words, coalesced per run */
+        for (int q = 0; q < R; q++) {
+            uint64_t wl = s_winlo[q], wh = s_winhi[q];
+            if (wh <= wl)
+                continue;
+            const uint64_t *tq = runs[q].tails;
+            uint64_t base = s_base[q];
+            for (uint64_t j = wl + threadIdx.x; j < wh; j += blockDim.x)
+                s_tails[base + (j - wl)] = tq[j];
+        }
+    }
+    __syncthreads();
+
+    int disp = D_NONE;
+    if (t < total) {
+        uint64_t rank = i - lo[r];
+        int shadow = 0;
+        bool fb = s_fallback != 0;
+        for (int q = 0; q < R; q++) {
+            if (q == r)
+                continue;
+            uint64_t qlo = b0 ? b0[q] : lo[q];
+            uint64_t qhi = b1 ? b1[q] : hi[q];
+            uint64_t bnd;
+            if (!fb) {
+                uint64_t off = s_base[q] - s_winlo[q];
+                uint64_t l = qlo, h = qhi;
+                if (q > r) {
+                    while (l < h) {
+                        uint64_t mid = (l + h) >> 1;
+                        if (s_tails[off + mid] <= qw)
+                            l = mid + 1;
+                        else
+                            h = mid;
+                    }
+                } else {
+                    while (l < h) {
+                        uint64_t mid = (l + h) >> 1;
+                        if (s_tails[off + mid] < qw)
+                            l = mid + 1;
+                        else
+                            h = mid;
+                    }
+                }
+                bnd = l;
+            } else {
+                uint64_t kl;
+                const uint8_t *k = run_key(runs[r], i, &kl);
+                bnd = (q > r) ? dev_upper_bound(runs[q], k, kl, qlo, qhi)
+                              : dev_lower_bound(runs[q], k, kl, qlo, qhi);
+            }
+            if (q > r) {
+                if (!shadow && bnd > lo[q] && runs[q].tails[bnd - 1] == qw)
+                    shadow = 1;
+                rank += bnd - lo[q];
+            } else {
+                rank += bnd - lo[q];
+            }
+        }
+        uint8_t ch;
+        uint32_t nts;
+        uint64_t okl, ovl;
+        disp = dev_disposition(runs[r], i, cp, shadow, &ch, &nts, &okl, &ovl);
+        order[rank] = ((uint64_t)r << 40) | i;
+        keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+        changed[rank] = ch;
+        new_expire[rank] = nts;
+        ksz[rank] = okl;
+        vsz[rank] = ovl;
+        if (rank_of)
+            rank_of[t] = rank;
+    }
+    unsigned long long b;
+    b = __ballot(disp == D_SHADOWED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->shadowed, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_TOMBSTONE);
+    if (lane == 0 && b)
+        atomicAdd(&stats->tombstones, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_EXPIRED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->expired, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_FILTERED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->filtered, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_KEEP);
+    if (lane == 0 && b)
+        atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+}
+
+void launch_rank_compact_ldst(const DevRun *d_runs, int R, const uint64_t *d_lo,
+                              const uint64_t *d_hi, const uint64_t *d_wprefix, uint64_t total,
+                              const CompactParams &cp, uint64_t *d_order, uint64_t *d_keepw,
+                              uint8_t *d_changed, uint32_t *d_new_expire, uint64_t *d_ksz,
+                              uint64_t *d_vsz, uint64_t *d_rank_of, const uint64_t *d_bt_off,
+                              const uint64_t *d_bt, int bt_shift, CompactStatsDev *d_stats,
+                              hipStream_t s)
+{
+    uint64_t blocks = (total + BLOCK - 1) / BLOCK;
+    if (blocks == 0)
+        blocks = 1;
+    k_rank_compact_ldst<<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+        d_runs, R, d_lo, d_hi, d_wprefix, total, cp, d_order, d_keepw, d_changed, d_new_expire,
+        d_ksz, d_vsz, d_rank_of, d_bt_off, d_bt, bt_shift, d_stats);
+}
+
 /* emit the merged run (wave per record); values copied whole, expire header
  * patched in-place for changed records */
 __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64_t m,
