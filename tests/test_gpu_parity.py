@@ -709,7 +709,7 @@ def test_compact_parity_ldst_rank(oracle_lib, hip_lib, n_keys):
 
     runs = D2.build_point_table_runs(n_keys, 6, seed=D2.DEFAULT_SEED + 77,
                                      dup_fraction=0.12, delete_fraction=0.03,
-                                     ttl_fraction=0.05, now=1000)
+                                     ttl_fraction=0.05, ttl_expire_ts=500)
     o = oracle_lib.open(1, 0, -1)
     g = hip_lib.open(1, 0, 0)
     try:
