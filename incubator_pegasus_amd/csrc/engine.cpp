@@ -563,11 +563,12 @@ struct HipEngine {
     int emit_mode = 2; /* 2 = chunked (default; 2.8 TB/s on the copy probe),
                           0 = rank-major waves, 1 = input-major waves
                           (env "engine.emit_mode": chunked|rank|input) */
-    int rank_mode = 0; /* 0 = global searches + bound-table narrowing,
-                          1 = LDS-staged full-key block rank,
-                          3 = LDS-staged tail-word rank (needs word-probe
-                          eligible runs with one shared cross-run prefix)
-                          (env "engine.rank_mode": global|lds|ldst) */
+    int rank_mode = 3; /* 3 = LDS-staged tail-word rank (default; falls back
+                          to mode 0 per compact when runs are not word-probe
+                          eligible with one shared cross-run prefix),
+                          0 = global searches + bound-table narrowing,
+                          1 = LDS-staged full-key block rank
+                          (env "engine.rank_mode": ldst|global|lds) */
     int bt_shift = 5;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
 
@@ -1020,7 +1021,7 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
         } else if (k == "rocksdb.filter_type") {
             e->bloom_enabled = (v != "none"); /* common/prefix -> full-key bloom */
         } else if (k == "engine.rank_mode") {
-            e->rank_mode = (v == "lds") ? 1 : (v == "ldst" ? 3 : 0);
+            e->rank_mode = (v == "lds") ? 1 : (v == "ldst" ? 3 : (v == "global" ? 0 : 3));
         } else if (k == "engine.bt_shift") {
             int s_ = atoi(v.c_str());
             if (s_ >= 4 && s_ <= 16)
