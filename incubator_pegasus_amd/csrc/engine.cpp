@@ -69,6 +69,9 @@ void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64
                              hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
+void launch_rank_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
+                      const uint64_t *, uint64_t, uint64_t *, uint8_t *, const uint64_t *,
+                      const uint64_t *, int, hipStream_t);
 void launch_rank_compact_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
                               const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                               uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
@@ -574,6 +577,32 @@ struct HipEngine {
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
+    /* LDS tail-word rank eligibility: every run in single-word probe mode
+     * with one shared stride and one shared cross-run first-(fk-8)-byte
+     * prefix, so tail-word compares decide cross-run order exactly */
+    bool ldst_eligible()
+    {
+        if (rank_mode != 3 || runs.size() < 2)
+            return false;
+        uint32_t fk = runs[0].fixed_klen;
+        if (fk < 8)
+            return false;
+        for (auto &rr : runs)
+            if (!rr.tails || rr.fixed_klen != fk || rr.n == 0)
+                return false;
+        if (fk > 8) {
+            uint8_t p0[32], pi[32];
+            uint64_t pl = fk - 8 < 32 ? fk - 8 : 32;
+            HIP_OK(hipMemcpy(p0, runs[0].keys, pl, hipMemcpyDeviceToHost));
+            for (size_t ri = 1; ri < runs.size(); ri++) {
+                HIP_OK(hipMemcpy(pi, runs[ri].keys, pl, hipMemcpyDeviceToHost));
+                if (memcmp(p0, pi, pl) != 0)
+                    return false;
+            }
+        }
+        return true;
+    }
+
     void free_run(RunBuf &r)
     {
         (void)hipFree(r.keys);
@@ -857,8 +886,12 @@ struct HipEngine {
         uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
         if (R > 1 && total > 100000)
             build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
-        launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
-                    bt_shift, stream);
+        if (ldst_eligible())
+            launch_rank_ldst(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
+                             bt_shift, stream);
+        else
+            launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
+                        bt_shift, stream);
         launch_visible(dr, d_order, d_shadow, total, d_flags, stream);
         launch_psum(d_flags, d_pos, total, psum_scratch(total), stream);
         uint64_t lastp = 0, lastf = 0;
@@ -1999,27 +2032,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     if (R > 1 && total > 100000)
         e->build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
     HIP_OK(hipEventRecord(ev[0], e->stream));
-    bool ldst_ok = false;
-    if (e->rank_mode == 3 && R > 1) {
-        /* eligibility: every run in single-word probe mode with the SAME
-         * stride, and all runs share one first-(fk-8)-byte prefix (then a
-         * tail-word compare decides cross-run order exactly) */
-        uint32_t fk = e->runs.empty() ? 0 : e->runs[0].fixed_klen;
-        ldst_ok = fk >= 8;
-        for (auto &rr : e->runs)
-            if (!rr.tails || rr.fixed_klen != fk || rr.n == 0)
-                ldst_ok = false;
-        if (ldst_ok && fk > 8) {
-            uint8_t p0[32], pi[32];
-            uint64_t pl = fk - 8 < 32 ? fk - 8 : 32;
-            HIP_OK(hipMemcpy(p0, e->runs[0].keys, pl, hipMemcpyDeviceToHost));
-            for (size_t ri = 1; ri < e->runs.size() && ldst_ok; ri++) {
-                HIP_OK(hipMemcpy(pi, e->runs[ri].keys, pl, hipMemcpyDeviceToHost));
-                if (memcmp(p0, pi, pl) != 0)
-                    ldst_ok = false;
-            }
-        }
-    }
+    bool ldst_ok = R > 1 && e->ldst_eligible();
     if (ldst_ok)
         launch_rank_compact_ldst(dr, R, d_lo, d_hi, d_wp, total, cp, d_order, d_keepw,
                                  d_changed, d_new_expire, d_ksz, d_vsz, d_rank_of, d_bt_off,

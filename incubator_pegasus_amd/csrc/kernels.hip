@@ -471,6 +471,140 @@ __global__ void k_rank(const DevRun *runs, int R, const uint64_t *lo, const uint
     }
 }
 
+/* LDS tail-word staged variant of k_rank (view builds for scans) — same
+ * staging and eligibility as k_rank_compact_ldst */
+__global__ void __launch_bounds__(BLOCK) k_rank_ldst(
+    const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi, const uint64_t *wprefix,
+    uint64_t total, uint64_t *order, uint8_t *shadowed, const uint64_t *bt_off,
+    const uint64_t *bt, int bt_shift)
+{
+    __shared__ uint64_t s_tails[LDST_CAP];
+    __shared__ uint64_t s_winlo[LDST_MAXR], s_winhi[LDST_MAXR], s_base[LDST_MAXR];
+    __shared__ int s_fallback;
+
+    uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+    int lane = threadIdx.x % WAVE;
+    if (threadIdx.x == 0)
+        s_fallback = (bt == nullptr || R > LDST_MAXR) ? 1 : 0;
+    if (threadIdx.x < (unsigned)(R < LDST_MAXR ? R : LDST_MAXR)) {
+        s_winlo[threadIdx.x] = ~0ull;
+        s_winhi[threadIdx.x] = 0;
+    }
+    __syncthreads();
+    int r = 0;
+    uint64_t i = 0, qw = 0;
+    const uint64_t *b0 = nullptr, *b1 = nullptr;
+    if (t < total) {
+        while (wprefix[r + 1] <= t)
+            r++;
+        i = lo[r] + (t - wprefix[r]);
+        qw = runs[r].tails[i];
+        if (bt) {
+            uint64_t j = (i - lo[r]) >> bt_shift;
+            b0 = bt + (bt_off[r] + j) * (uint64_t)R;
+            b1 = bt + (bt_off[r] + j + 1) * (uint64_t)R;
+        }
+    }
+    if (!s_fallback) {
+        for (int q = 0; q < R; q++) {
+            uint64_t wl = (t < total && q != r && b0) ? b0[q] : ~0ull;
+            uint64_t wh = (t < total && q != r && b1) ? b1[q] : 0;
+            for (int d = WAVE / 2; d; d >>= 1) {
+                uint64_t o1 = __shfl_xor(wl, d);
+                if (o1 < wl)
+                    wl = o1;
+                uint64_t o2 = __shfl_xor(wh, d);
+                if (o2 > wh)
+                    wh = o2;
+            }
+            if (lane == 0 && wh > 0) {
+                atomicMin((unsigned long long *)&s_winlo[q], (unsigned long long)wl);
+                atomicMax((unsigned long long *)&s_winhi[q], (unsigned long long)wh);
+            }
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0 && !s_fallback) {
+        uint64_t tot = 0;
+        for (int q = 0; q < R; q++) {
+            s_base[q] = tot;
+            if (s_winhi[q] > s_winlo[q])
+                tot += s_winhi[q] - s_winlo[q];
+        }
+        if (tot > LDST_CAP)
+            s_fallback = 1;
+    }
+    __syncthreads();
+    if (!s_fallback) {
+        for (int q = 0; q < R; q++) {
+            uint64_t wl = s_winlo[q], wh = s_winhi[q];
+            if (wh <= wl)
+                continue;
+            const uint64_t *tq = runs[q].tails;
+            uint64_t base = s_base[q];
+            for (uint64_t j = wl + threadIdx.x; j < wh; j += blockDim.x)
+                s_tails[base + (j - wl)] = tq[j];
+        }
+    }
+    __syncthreads();
+    if (t < total) {
+        uint64_t rank = i - lo[r];
+        int shadow = 0;
+        bool fb = s_fallback != 0;
+        for (int q = 0; q < R; q++) {
+            if (q == r)
+                continue;
+            uint64_t qlo = b0 ? b0[q] : lo[q];
+            uint64_t qhi = b1 ? b1[q] : hi[q];
+            uint64_t bnd;
+            if (!fb) {
+                uint64_t off = s_base[q] - s_winlo[q];
+                uint64_t l = qlo, h = qhi;
+                if (q > r) {
+                    while (l < h) {
+                        uint64_t mid = (l + h) >> 1;
+                        if (s_tails[off + mid] <= qw)
+                            l = mid + 1;
+                        else
+                            h = mid;
+                    }
+                } else {
+                    while (l < h) {
+                        uint64_t mid = (l + h) >> 1;
+                        if (s_tails[off + mid] < qw)
+                            l = mid + 1;
+                        else
+                            h = mid;
+                    }
+                }
+                bnd = l;
+            } else {
+                uint64_t kl;
+                const uint8_t *k = run_key(runs[r], i, &kl);
+                bnd = (q > r) ? dev_upper_bound(runs[q], k, kl, qlo, qhi)
+                              : dev_lower_bound(runs[q], k, kl, qlo, qhi);
+            }
+            if (q > r && !shadow && bnd > lo[q] && runs[q].tails[bnd - 1] == qw)
+                shadow = 1;
+            rank += bnd - lo[q];
+        }
+        order[rank] = ((uint64_t)r << 40) | i;
+        shadowed[rank] = (uint8_t)shadow;
+    }
+}
+
+void launch_rank_ldst(const DevRun *d_runs, int R, const uint64_t *d_lo, const uint64_t *d_hi,
+                      const uint64_t *d_wprefix, uint64_t total, uint64_t *d_order,
+                      uint8_t *d_shadow, const uint64_t *d_bt_off, const uint64_t *d_bt,
+                      int bt_shift, hipStream_t s)
+{
+    uint64_t blocks = (total + BLOCK - 1) / BLOCK;
+    if (blocks == 0)
+        blocks = 1;
+    k_rank_ldst<<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+        d_runs, R, d_lo, d_hi, d_wprefix, total, d_order, d_shadow, d_bt_off, d_bt, bt_shift);
+}
+
 /* visible flag: newest version of its key (first of the equal-key group in
  * (key asc, run desc) order) and not a tombstone */
 __global__ void k_visible(const DevRun *runs, const uint64_t *order, const uint8_t *shadowed,
