@@ -2096,8 +2096,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             uint64_t *d_row_vsrc = e->talloc<uint64_t>(n_out * 8);
             uint32_t *d_row_patch = e->talloc<uint32_t>(n_out * 4);
             uint32_t *d_row_expire = e->talloc<uint32_t>(n_out * 4);
-            uint64_t kanch = (((kbytes + 15) >> 4) >> 5) + 1;
-            uint64_t vanch = (((vbytes + 15) >> 4) >> 5) + 1;
+            uint64_t kanch = (((kbytes + 15) >> 4) >> 6) + 1;
+            uint64_t vanch = (((vbytes + 15) >> 4) >> 6) + 1;
             uint64_t *d_kanchor = e->talloc<uint64_t>(kanch * 8);
             uint64_t *d_vanchor = e->talloc<uint64_t>(vanch * 8);
             launch_emit_compact_chunked(dr, d_order, total, d_keepw, d_changed, d_new_expire,

@@ -1380,14 +1380,14 @@ __global__ void k_compact_gather_expire(const uint64_t *keepw, const uint64_t *k
 
 /* copy 16B chunks of the packed output; chunk -> row via binary search on the
  * row offset array (hot in L1/L2: consecutive chunks hit consecutive rows) */
-/* every 32nd chunk's (512 B of output) starting row: narrows the per-chunk
- * row search from log2(n_rows) L2 probes to ~2 probes in a ~4-row window */
+/* every 64th chunk's (1 KB of output) starting row: narrows the per-chunk
+ * row search from log2(n_rows) L2 probes to ~3 probes in an 8-row window */
 __global__ void k_chunk_anchors(const uint64_t *row_off, uint64_t n_rows, uint64_t n_anchors,
                                 uint64_t *anchors)
 {
     for (uint64_t a = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; a < n_anchors;
          a += gridDim.x * (uint64_t)blockDim.x) {
-        uint64_t pos = a << 9; /* 32 chunks x 16B */
+        uint64_t pos = a << 10; /* 64 chunks x 16B */
         uint64_t lo = 0, hi = n_rows;
         while (lo + 1 < hi) {
             uint64_t mid = (lo + hi) >> 1;
@@ -1411,7 +1411,7 @@ __global__ void k_copy_chunks(const uint64_t *row_off /* [n_rows+1] */,
         uint64_t pos = t << 4;
         uint64_t end = pos + 16 <= total_bytes ? pos + 16 : total_bytes;
         /* row = last row with row_off[row] <= pos */
-        uint64_t ab = t >> 5;
+        uint64_t ab = t >> 6;
         uint64_t lo = anchors[ab];
         uint64_t hi = (ab + 1 < n_anchors) ? anchors[ab + 1] + 1 : n_rows;
         while (lo + 1 < hi) {
@@ -1688,7 +1688,7 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
         d_ovoff, d_row_ksrc, d_row_vsrc, d_row_patch, d_osk, n_out, kbytes, vbytes);
     k_compact_gather_expire<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_keepw, d_kpos, d_changed,
                                                                  d_new_expire, m, d_row_expire);
-    uint64_t kanch = ((kbytes + 15) >> 4 >> 5) + 1, vanch = ((vbytes + 15) >> 4 >> 5) + 1;
+    uint64_t kanch = ((kbytes + 15) >> 4 >> 6) + 1, vanch = ((vbytes + 15) >> 4 >> 6) + 1;
     k_chunk_anchors<<<grid_for(kanch, BLOCK), BLOCK, 0, s>>>(d_okoff, n_out, kanch, d_kanchor);
     k_chunk_anchors<<<grid_for(vanch, BLOCK), BLOCK, 0, s>>>(d_ovoff, n_out, vanch, d_vanchor);
     k_copy_chunks<<<grid_for((kbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
