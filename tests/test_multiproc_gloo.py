@@ -6,7 +6,6 @@ RCCL/xGMI on the GPU box."""
 import os
 import sys
 
-import pytest
 import torch.multiprocessing as mp
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))

@@ -6,7 +6,6 @@ src/test/function_test/base_api/test_scan.cpp:140-400 /
 test_range_read.cpp (bounds inclusive/exclusive, one-point, void span,
 count-only, batch paging)."""
 import random
-import struct
 
 import pytest
 
@@ -425,7 +424,6 @@ def test_compact_with_default_ttl_and_rules_differential(oracle_part):
 
 
 def test_ingest_validation(oracle_part):
-    from incubator_pegasus_amd.capi import INVALID_ARGUMENT
     k1 = D.generate_key(b"a", b"")
     k2 = D.generate_key(b"b", b"")
     v = D.encode_value(b"x", 0, 0, 1)
