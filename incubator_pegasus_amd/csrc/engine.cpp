@@ -549,6 +549,7 @@ struct HipEngine {
     std::vector<RunBuf> runs;
     DevRun *d_runs = nullptr;
     bool d_runs_dirty = true;
+    int ldst_elig_cache = -1; /* -1 unknown; recomputed when the run set changes */
     uint64_t next_seq_floor = 0;
     /* user ops: host + device */
     std::vector<HostOp> host_ops;
@@ -584,6 +585,9 @@ struct HipEngine {
     {
         if (rank_mode != 3 || runs.size() < 2)
             return false;
+        if (ldst_elig_cache >= 0)
+            return ldst_elig_cache != 0;
+        ldst_elig_cache = 0;
         uint32_t fk = runs[0].fixed_klen;
         if (fk < 8)
             return false;
@@ -600,6 +604,7 @@ struct HipEngine {
                     return false;
             }
         }
+        ldst_elig_cache = 1;
         return true;
     }
 
@@ -945,6 +950,7 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
+    e->ldst_elig_cache = -1;
 }
 
 extern "C" {
@@ -1108,6 +1114,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
+    e->ldst_elig_cache = -1;
     return RRDB_OK;
 }
 
@@ -2174,6 +2181,8 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
             e->runs.push_back(nr);
         }
         e->d_runs_dirty = true;
+        e->ldst_elig_cache = -1;
+    e->ldst_elig_cache = -1;
     }
     if (stats)
         *stats = st;
@@ -2375,6 +2384,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
     }
     e->next_seq_floor = floor_;
     e->d_runs_dirty = true;
+    e->ldst_elig_cache = -1;
     return RRDB_OK;
 }
 
