@@ -731,3 +731,57 @@ def test_compact_parity_ldst_rank(oracle_lib, hip_lib, n_keys):
     finally:
         o.close()
         g.close()
+
+
+def test_ldst_fallback_branches(oracle_lib, hip_lib):
+    """Two ldst dispatch branches the main suite doesn't reach: more runs
+    than the kernel stages (R > 16 -> in-kernel fallback to global probes)
+    and uniform-width runs whose prefixes differ across runs (host
+    eligibility check fails -> standard kernel)."""
+    now = 1000
+    # (a) 18 eligible runs: kernel launches staged variant, falls back inside
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        g.set_envs({"engine.rank_mode": "ldst"})
+        seq = 1
+        for run in range(18):
+            recs = []
+            for i in range(40):
+                k = D.generate_key(f"fb{(run * 7 + i * 3) % 120:04d}".encode(), b"")
+                recs.append((k, D.encode_value(b"v%d" % run, 0, seq, 1), seq, 0))
+                seq += 1
+            recs = sorted({k: r for k, *r in [(k, k, v, s, kd) for k, v, s, kd in recs]}.items())
+            recs = [(k, v, s, kd) for k, (_, v, s, kd) in recs]
+            o.ingest_run(recs)
+            g.ingest_run(recs)
+        assert o.manual_compact(now) == g.manual_compact(now)
+        assert _drain(o, now, validate_partition_hash=False) == \
+               _drain(g, now, validate_partition_hash=False)
+    finally:
+        o.close()
+        g.close()
+    # (b) same fixed width, different cross-run prefixes: host falls back
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        g.set_envs({"engine.rank_mode": "ldst"})
+        seq = 1
+        for pfx in (b"aaaaaaaaaa", b"bbbbbbbbbb", b"cccccccccc"):
+            recs = []
+            for i in range(50):
+                k = D.generate_key(pfx + b"%04d" % i, b"")
+                recs.append((k, D.encode_value(b"w", 0, seq, 1), seq, 0))
+                seq += 1
+            o.ingest_run(recs)
+            g.ingest_run(recs)
+        for i in (0, 7, 49):
+            for pfx in (b"aaaaaaaaaa", b"bbbbbbbbbb", b"cccccccccc"):
+                k = D.generate_key(pfx + b"%04d" % i, b"")
+                assert o.get(k, now) == g.get(k, now), k
+        assert o.manual_compact(now) == g.manual_compact(now)
+        assert _drain(o, now, validate_partition_hash=False) == \
+               _drain(g, now, validate_partition_hash=False)
+    finally:
+        o.close()
+        g.close()
