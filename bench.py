@@ -169,15 +169,19 @@ def main():
             dist.barrier()
         torch.cuda.synchronize()
 
-    # serial per-partition calls: measured 4x FASTER than a thread pool over
-    # per-partition streams on MI355X (multi-threaded HIP submission + sync
-    # contention: 256 vs 64 ms/step at --compact-threads 4); per-call wall is
-    # within ~0.5ms of in-kernel time, so overlap has little to win here
+    # single-threaded PIPELINED submission: begin() submits every partition's
+    # merge phase back-to-back (no host sync between partitions), finish()
+    # then emits each — the GPU never idles on per-call host gaps.  A thread
+    # pool over per-partition streams was measured 4x SLOWER (multi-threaded
+    # HIP submission + sync contention: 256 vs 64 ms/step); this keeps one
+    # submitting thread.
     def one_step():
         out_records = 0
         st = None
         for eng in parts:
-            err, st = eng.manual_compact(epoch_now, keep_inputs=True)
+            assert eng.manual_compact_begin(epoch_now, keep_inputs=True) == 0
+        for eng in parts:
+            err, st = eng.manual_compact_finish()
             assert err == 0
             out_records += st.output_records
         return out_records, st

@@ -270,6 +270,19 @@ void rrdb_clear_scanner(void *h, int64_t context_id);
 int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
                             rrdb_compact_stats *stats);
 
+/* Split compaction: the pipelined-partitions seam.  The reference runs
+ * per-replica compactions concurrently on THREAD_POOL_COMPACT
+ * (pegasus_manual_compact_service, pegasus_server_impl.cpp:3373-3420);
+ * here a single host thread gets the same overlap by calling _begin on
+ * every partition handle (submits the merge + sizing work, returns
+ * without blocking) and then _finish on each (waits, emits the output
+ * run, returns stats).  _begin fails with kInvalidArgument if a split
+ * compaction is already pending on the handle; _finish without a pending
+ * _begin likewise.  rrdb_manual_compact == begin+finish. */
+int32_t rrdb_manual_compact_begin(void *h, const rrdb_compact_options *opts,
+                                  uint32_t epoch_now);
+int32_t rrdb_manual_compact_finish(void *h, rrdb_compact_stats *stats);
+
 /* ---- checkpoint ----
  * Serializes the partition's runs to <dir>/checkpoint.<decree>/ in the
  * engine's own run format (decree-tagged like the reference's

@@ -192,6 +192,11 @@ class RrdbLib:
         L.rrdb_manual_compact.restype = C.c_int32
         L.rrdb_manual_compact.argtypes = [
             C.c_void_p, C.POINTER(_CompactOptions), C.c_uint32, C.POINTER(_CompactStats)]
+        L.rrdb_manual_compact_begin.restype = C.c_int32
+        L.rrdb_manual_compact_begin.argtypes = [
+            C.c_void_p, C.POINTER(_CompactOptions), C.c_uint32]
+        L.rrdb_manual_compact_finish.restype = C.c_int32
+        L.rrdb_manual_compact_finish.argtypes = [C.c_void_p, C.POINTER(_CompactStats)]
         L.rrdb_phase_ms.restype = C.c_double
         L.rrdb_phase_ms.argtypes = [C.c_void_p, C.c_char_p]
         L.rrdb_put.restype = C.c_int32
@@ -515,5 +520,20 @@ class RrdbPartition:
                                1 if keep_inputs else 0)
         st = _CompactStats()
         err = self._L.rrdb_manual_compact(self._h, C.byref(opts), epoch_now, C.byref(st))
+        stats = CompactStats(**{f[0]: getattr(st, f[0]) for f in _CompactStats._fields_})
+        return err, stats
+
+    def manual_compact_begin(self, epoch_now: int, *, target_level=-1, bottommost_force=True,
+                             keep_inputs=False):
+        """Submit the compaction's merge phase without blocking; pair with
+        manual_compact_finish.  The pipelined-partitions seam
+        (include/rrdb_engine.h)."""
+        opts = _CompactOptions(target_level, 1 if bottommost_force else 0,
+                               1 if keep_inputs else 0)
+        return self._L.rrdb_manual_compact_begin(self._h, C.byref(opts), epoch_now)
+
+    def manual_compact_finish(self, epoch_now: int = 0):
+        st = _CompactStats()
+        err = self._L.rrdb_manual_compact_finish(self._h, C.byref(st))
         stats = CompactStats(**{f[0]: getattr(st, f[0]) for f in _CompactStats._fields_})
         return err, stats

@@ -550,6 +550,23 @@ struct HipEngine {
     DevRun *d_runs = nullptr;
     bool d_runs_dirty = true;
     int ldst_elig_cache = -1; /* -1 unknown; recomputed when the run set changes */
+    /* phase-1 state of a split compaction (rrdb_manual_compact_begin):
+     * device pointers are arena memory, valid until the next scratch_reset */
+    struct PendingCompact {
+        bool active = false, trivial = false, keep_inputs = false;
+        int R = 0;
+        uint64_t total = 0;
+        DevRun *dr = nullptr;
+        uint64_t *d_wp = nullptr, *d_order = nullptr, *d_keepw = nullptr;
+        uint8_t *d_changed = nullptr;
+        uint32_t *d_new_expire = nullptr;
+        uint64_t *d_kpos = nullptr, *d_koffs = nullptr, *d_voffs = nullptr;
+        uint64_t *d_rank_of = nullptr;
+        CompactStatsDev *d_stats = nullptr;
+        hipEvent_t ev[6] = {};
+        rrdb_compact_stats st{};
+    } pend;
+    uint64_t *pend_sizes = nullptr; /* pinned [6]: output sizes d2h target */
     uint64_t next_seq_floor = 0;
     /* user ops: host + device */
     std::vector<HostOp> host_ops;
@@ -1012,6 +1029,8 @@ void rrdb_close(void *h)
         (void)hipFree(b.first);
     if (e->pinned)
         (void)hipHostFree(e->pinned);
+    if (e->pend_sizes)
+        (void)hipHostFree(e->pend_sizes);
     if (e->d_ops)
         (void)hipFree(e->d_ops);
     if (e->d_rules)
@@ -1942,19 +1961,19 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     return multi_get_locked(h, q, epoch_now, out);
 }
 
-int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
-                            rrdb_compact_stats *stats)
+/* phase 1 of the compaction (the pipelined-partitions seam; the reference
+ * runs per-replica compactions concurrently on THREAD_POOL_COMPACT —
+ * pegasus_server_impl.cpp:3373): submit rank + prefix sums + async size
+ * reads, NO host-blocking sync.  Caller holds the handle lock. */
+static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uint32_t epoch_now)
 {
-    auto *e = (HipEngine *)h;
-    std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
-    engine_flush((HipEngine *)h); /* memtable visible to reads */
-    ((HipEngine *)h)->scratch_reset();
-    rrdb_compact_stats st{};
-    if (e->manual_compact_disabled) {
-        if (stats)
-            *stats = st;
+    if (e->pend.active)
         return RRDB_INVALID_ARGUMENT;
-    }
+    engine_flush(e); /* memtable visible to reads */
+    e->scratch_reset();
+    rrdb_compact_stats st{};
+    if (e->manual_compact_disabled)
+        return RRDB_INVALID_ARGUMENT;
     e->activate();
     int R = (int)e->runs.size();
     uint64_t total = 0;
@@ -1962,8 +1981,10 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         total += r.n;
     st.input_records = total;
     if (total == 0) {
-        if (stats)
-            *stats = st;
+        e->pend = {};
+        e->pend.active = true;
+        e->pend.trivial = true;
+        e->pend.st = st;
         return RRDB_OK;
     }
     DevRun *dr = e->dev_runs();
@@ -2054,14 +2075,70 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);
     launch_psum(d_ksz, d_koffs, total, e->psum_scratch(total), e->stream);
     launch_psum(d_vsz, d_voffs, total, e->psum_scratch(total), e->stream);
-    uint64_t t[6] = {0};
+    if (!e->pend_sizes)
+        HIP_OK(hipHostMalloc((void **)&e->pend_sizes, 6 * 8));
+    uint64_t *t = e->pend_sizes;
     HIP_OK(hipMemcpyAsync(&t[0], d_kpos + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[1], d_keepw + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[2], d_koffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[3], d_ksz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[4], d_voffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[5], d_vsz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    e->pend = {};
+    e->pend.active = true;
+    e->pend.keep_inputs = opts && opts->keep_inputs;
+    e->pend.R = R;
+    e->pend.total = total;
+    e->pend.dr = dr;
+    e->pend.d_wp = d_wp;
+    e->pend.d_order = d_order;
+    e->pend.d_keepw = d_keepw;
+    e->pend.d_changed = d_changed;
+    e->pend.d_new_expire = d_new_expire;
+    e->pend.d_kpos = d_kpos;
+    e->pend.d_koffs = d_koffs;
+    e->pend.d_voffs = d_voffs;
+    e->pend.d_rank_of = d_rank_of;
+    e->pend.d_stats = d_stats;
+    for (int i = 0; i < 6; i++)
+        e->pend.ev[i] = ev[i];
+    e->pend.st = st;
+    (void)d_lo;
+    (void)d_hi;
+    (void)d_ksz;
+    (void)d_vsz;
+    return RRDB_OK;
+}
+
+/* phase 2: wait for the submitted work, allocate + emit the output run,
+ * collect stats.  Caller holds the handle lock. */
+static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
+{
+    if (!e->pend.active)
+        return RRDB_INVALID_ARGUMENT;
+    e->activate();
+    e->pend.active = false;
+    rrdb_compact_stats st = e->pend.st;
+    if (e->pend.trivial) {
+        if (stats)
+            *stats = st;
+        return RRDB_OK;
+    }
+    int R = e->pend.R;
+    (void)R;
+    uint64_t total = e->pend.total;
+    DevRun *dr = e->pend.dr;
+    uint64_t *d_wp = e->pend.d_wp, *d_order = e->pend.d_order, *d_keepw = e->pend.d_keepw;
+    uint8_t *d_changed = e->pend.d_changed;
+    uint32_t *d_new_expire = e->pend.d_new_expire;
+    uint64_t *d_kpos = e->pend.d_kpos, *d_koffs = e->pend.d_koffs, *d_voffs = e->pend.d_voffs;
+    uint64_t *d_rank_of = e->pend.d_rank_of;
+    CompactStatsDev *d_stats = e->pend.d_stats;
+    hipEvent_t ev[6];
+    for (int i = 0; i < 6; i++)
+        ev[i] = e->pend.ev[i];
     HIP_OK(hipStreamSynchronize(e->stream));
+    uint64_t *t = e->pend_sizes;
     uint64_t n_out = t[0] + t[1], kbytes = t[2] + t[3], vbytes = t[4] + t[5];
 
     CompactStatsDev hs{};
@@ -2074,7 +2151,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     st.output_bytes = kbytes + vbytes;
 
     RunBuf nr;
-    bool keep_inputs = opts && opts->keep_inputs;
+    bool keep_inputs = e->pend.keep_inputs;
     if (n_out > 0) {
         nr.n = n_out;
         /* output keys are verbatim copies of input keys, so a stride shared
@@ -2138,30 +2215,10 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     }
     for (auto &x : ev)
         (void)hipEventDestroy(x);
-    e->tfree(d_lo);
-    e->tfree(d_hi);
-    e->tfree(d_wp);
-    e->tfree(d_order);
-    if (d_rank_of)
-        e->tfree(d_rank_of);
-    e->tfree(d_changed);
-    e->tfree(d_new_expire);
-    e->tfree(d_ksz);
-    e->tfree(d_vsz);
-    e->tfree(d_stats);
-    e->tfree(d_keepw);
-    e->tfree(d_kpos);
-    e->tfree(d_koffs);
-    e->tfree(d_voffs);
+    /* arena temporaries (d_order, flags, sums, keep_inputs outputs) are
+     * reclaimed at the next scratch_reset */
     if (keep_inputs) {
         /* benchmarking: the pass ran in full; drop the output, keep inputs */
-        if (n_out > 0) {
-            e->tfree(nr.keys);
-            e->tfree(nr.vals);
-            e->tfree(nr.koff);
-            e->tfree(nr.voff);
-            e->tfree(nr.sk);
-        }
     } else {
         for (auto &r : e->runs)
             e->free_run(r);
@@ -2182,11 +2239,38 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
         }
         e->d_runs_dirty = true;
         e->ldst_elig_cache = -1;
-    e->ldst_elig_cache = -1;
     }
     if (stats)
         *stats = st;
     return RRDB_OK;
+}
+
+int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t epoch_now,
+                            rrdb_compact_stats *stats)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    int32_t rc = compact_begin(e, opts, epoch_now);
+    if (rc != RRDB_OK) {
+        if (stats)
+            *stats = rrdb_compact_stats{};
+        return rc;
+    }
+    return compact_finish(e, stats);
+}
+
+int32_t rrdb_manual_compact_begin(void *h, const rrdb_compact_options *opts, uint32_t epoch_now)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    return compact_begin(e, opts, epoch_now);
+}
+
+int32_t rrdb_manual_compact_finish(void *h, rrdb_compact_stats *stats)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    return compact_finish(e, stats);
 }
 
 } /* extern "C" */

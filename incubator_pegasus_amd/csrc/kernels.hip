@@ -23,6 +23,10 @@
 
 #define WAVE 64
 #define BLOCK 256
+/* LDS tail-word staged rank: staged window capacity (u64 words) and the
+ * max run count the staging handles (beyond either -> in-kernel fallback) */
+#define LDST_CAP 3072
+#define LDST_MAXR 16
 #define MAX_GRID 2048
 #define PSUM_ITEMS_PER_THREAD 16
 #define PSUM_BLOCK_ITEMS (BLOCK * PSUM_ITEMS_PER_THREAD) /* 4096 */
@@ -1108,8 +1112,6 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
  * is one u64 compare and LDS holds ~2.3K words (~18KB) per workgroup — the
  * earlier full-key LDS experiment died at 1 wg/CU; this one keeps ~6.  Probes
  * whose first midpoints coincide across lanes become LDS broadcasts. */
-#define LDST_CAP 3072
-#define LDST_MAXR 16
 
 __global__ void __launch_bounds__(BLOCK) k_rank_compact_ldst(
     const DevRun *runs, int R, const uint64_t *lo, const uint64_t *hi, const uint64_t *wprefix,

@@ -807,6 +807,12 @@ typedef struct {
     ScanCtx **ctxs;
     int n_ctxs, ctxs_cap;
     int64_t next_ctx_id;
+    /* split-compaction pending state (rrdb_manual_compact_begin): the CPU
+     * oracle has no async phase, so begin just parks the arguments and
+     * finish runs the whole pass */
+    int pend_active;
+    int pend_keep_inputs;
+    uint32_t pend_epoch;
 } Engine;
 
 /* a parked scanner: materialized visible view of [cursor..stop) */
@@ -2355,4 +2361,30 @@ int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *hash_keys,
     out->count = m;
     out->error = RRDB_OK;
     return RRDB_OK;
+}
+
+/* split compaction seam (include/rrdb_engine.h): synchronous restatement —
+ * begin parks the request, finish executes rrdb_manual_compact's pass */
+int32_t rrdb_manual_compact_begin(void *h, const rrdb_compact_options *opts,
+                                  uint32_t epoch_now)
+{
+    Engine *e = (Engine *)h;
+    if (e->pend_active || e->manual_compact_disabled)
+        return RRDB_INVALID_ARGUMENT;
+    e->pend_active = 1;
+    e->pend_keep_inputs = opts && opts->keep_inputs;
+    e->pend_epoch = epoch_now;
+    return RRDB_OK;
+}
+
+int32_t rrdb_manual_compact_finish(void *h, rrdb_compact_stats *stats)
+{
+    Engine *e = (Engine *)h;
+    if (!e->pend_active)
+        return RRDB_INVALID_ARGUMENT;
+    e->pend_active = 0;
+    rrdb_compact_options o;
+    memset(&o, 0, sizeof(o));
+    o.keep_inputs = e->pend_keep_inputs;
+    return rrdb_manual_compact(h, &o, e->pend_epoch, stats);
 }

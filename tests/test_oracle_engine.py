@@ -447,3 +447,31 @@ def test_empty_engine_reads(oracle_part):
     assert res.error == OK and res.kvs == [] and res.context_id == SCAN_COMPLETED
     err, stats = oracle_part.manual_compact(now)
     assert err == OK and stats.output_records == 0
+
+
+def test_split_compact_matches_plain(oracle_lib):
+    """rrdb_manual_compact_begin+finish must equal the one-shot call."""
+    a = oracle_lib.open(1, 0, -1)
+    b = oracle_lib.open(1, 0, -1)
+    try:
+        recs = []
+        for i in range(200):
+            k = D.generate_key(b"sc%03d" % (i % 50), b"s%d" % (i % 4))
+            recs.append((k, D.encode_value(b"v%d" % i, 0, i + 1, 1), i + 1, 1 if i % 9 == 0 else 0))
+        recs = sorted({k: r for k, *r in [(k, k, v, s, kd) for k, v, s, kd in recs]}.items())
+        recs = [(k, v, s, kd) for k, (_, v, s, kd) in recs]
+        a.ingest_run(recs)
+        b.ingest_run(recs)
+        e1, s1 = a.manual_compact(1000)
+        assert b.manual_compact_begin(1000) == 0
+        # double-begin refused while pending
+        from incubator_pegasus_amd.capi import INVALID_ARGUMENT
+        assert b.manual_compact_begin(1000) == INVALID_ARGUMENT
+        e2, s2 = b.manual_compact_finish()
+        assert (e1, s1) == (e2, s2)
+        # finish with nothing pending refused
+        assert b.manual_compact_finish()[0] == INVALID_ARGUMENT
+        assert a.num_records() == b.num_records()
+    finally:
+        a.close()
+        b.close()
