@@ -1,0 +1,102 @@
+"""Stateful differential fuzz (hypothesis RuleBasedStateMachine): the CPU
+oracle engine vs the independent Python model (tests/pymodel.py) through
+randomized interleavings of ingest / get / ttl / sortkey_count / full scan /
+env flips / compaction.  Complements the fixed-seed differential tests in
+test_oracle_engine.py with hypothesis's generation + shrinking.
+
+Oracle use here is as the system-under-differential-test on CPU; the GPU
+parity suite (test_gpu_parity.py) then pins the HIP engine to this same
+oracle bit for bit.
+"""
+import pytest
+from hypothesis import settings
+from hypothesis.stateful import RuleBasedStateMachine, initialize, invariant, rule
+from hypothesis import strategies as st
+
+from incubator_pegasus_amd import data as D
+from tests import pymodel
+from tests.conftest import _ensure_oracle
+
+NOW = 1000
+
+HK = st.sampled_from([b"a", b"bb", b"hash-7", b"zz" * 8])
+SK = st.sampled_from([b"", b"s1", b"s2", b"sort-key-long"])
+BODY = st.binary(min_size=0, max_size=24)
+TTL = st.sampled_from([0, 1, NOW, NOW + 50])
+
+
+class OracleVsModel(RuleBasedStateMachine):
+    def __init__(self):
+        super().__init__()
+        from incubator_pegasus_amd.capi import RrdbLib
+
+        self.part = RrdbLib(_ensure_oracle()).open(1, 0, -1)
+        self.model = pymodel.Model()
+        self.seq = 1
+        self.pending = {}
+
+    @rule(hk=HK, sk=SK, body=BODY, ttl=TTL, kind=st.sampled_from([0, 0, 0, 1]))
+    def stage_record(self, hk, sk, body, ttl, kind):
+        key = D.generate_key(hk, sk)
+        val = D.encode_value(body, ttl, self.seq, 1) if kind == 0 else b"\x00" * 12
+        self.pending[key] = (val, kind)
+
+    @rule()
+    def flush_run(self):
+        if not self.pending:
+            return
+        recs = []
+        for key in sorted(self.pending):
+            val, kind = self.pending[key]
+            recs.append((key, val, self.seq, kind))
+            self.seq += 1
+        self.pending.clear()
+        self.part.ingest_run(recs)
+        self.model.ingest(recs)
+
+    @rule(ttl=st.sampled_from(["0", "77", "3600"]))
+    def set_default_ttl(self, ttl):
+        self.part.set_envs({"default_ttl": ttl})
+        self.model.default_ttl = int(ttl)
+
+    @rule(hk=HK, sk=SK)
+    def check_get_ttl(self, hk, sk):
+        key = D.generate_key(hk, sk)
+        assert self.part.get(key, NOW) == self.model.get(key, NOW)
+        assert self.part.ttl(key, NOW) == self.model.ttl(key, NOW)
+
+    @rule(hk=HK)
+    def check_sortkey_count(self, hk):
+        st_, cnt = self.part.sortkey_count(hk, NOW)
+        assert st_ == 0
+        assert cnt == self.model.sortkey_count(hk, NOW)
+
+    @rule()
+    def compact(self):
+        err, st_ = self.part.manual_compact(NOW)
+        surviving = self.model.compact(NOW)
+        assert err == 0
+        assert st_.output_records == len(surviving)
+
+    @invariant()
+    def full_scan_matches(self):
+        rows = []
+        res = self.part.scan_open(b"\x00\x00", b"\xff\xff", NOW,
+                                  validate_partition_hash=False, batch_size=1000)
+        from incubator_pegasus_amd.capi import SCAN_COMPLETED
+
+        assert res.error == 0
+        rows.extend(res.kvs)
+        while res.context_id != SCAN_COMPLETED:
+            res = self.part.scan_next(res.context_id, NOW)
+            assert res.error == 0
+            rows.extend(res.kvs)
+        assert rows == self.model.full_scan(NOW, validate_hash_req=False)
+
+    def teardown(self):
+        self.part.close()
+
+
+OracleVsModel = settings(max_examples=40, stateful_step_count=30,
+                         deadline=None)(OracleVsModel)
+TestOracleVsModel = OracleVsModel.TestCase
