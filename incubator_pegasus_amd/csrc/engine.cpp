@@ -67,6 +67,8 @@ void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64
                              const uint64_t *, const uint64_t *, uint64_t *, uint64_t *,
                              uint8_t *, uint32_t *, uint64_t *, uint64_t *, CompactStatsDev *,
                              hipStream_t);
+void launch_valid_beyond(const DevRun *, int, const uint8_t *, uint64_t, int, uint32_t *,
+                         hipStream_t);
 void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *, uint64_t,
                      uint32_t *, hipStream_t);
 void launch_rank_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
@@ -127,6 +129,16 @@ void build_crc64_table()
             k = (k & 1) ? ((k >> 1) ^ poly) : (k >> 1);
         host_crc64_table[i] = k;
     }
+}
+
+/* host crc64 over a buffer (same table); used for checkpoint-file integrity */
+static uint64_t host_crc64(const void *p, uint64_t n)
+{
+    const uint8_t *b = (const uint8_t *)p;
+    uint64_t crc = ~0ull;
+    for (uint64_t i = 0; i < n; i++)
+        crc = host_crc64_table[(uint8_t)(crc ^ b[i])] ^ (crc >> 8);
+    return ~crc;
 }
 
 /* ================ result arena ================ */
@@ -437,6 +449,10 @@ bool parse_user_ops(const std::string &text, std::vector<HostOp> &out)
 }
 
 /* ================ key byte helpers (pegasus_key_schema.h) ================ */
+/* the 2-byte length prefix caps hash keys at 64KiB-1; the reference
+ * CHECK_LTs (pegasus_key_schema.h:43) — as a library we reject with
+ * kInvalidArgument at the C-ABI boundary instead of aborting */
+static inline bool hklen_ok(uint64_t hklen) { return hklen < 0xFFFFull; }
 std::string make_key(const uint8_t *hk, uint64_t hklen, const uint8_t *sk, uint64_t sklen)
 {
     std::string k;
@@ -515,6 +531,7 @@ struct RunBuf {
 
 struct HipScanCtx {
     int64_t id = 0;
+    uint32_t parked_at = 0; /* epoch_now when (re-)parked; 5-min GC */
     uint64_t *d_view = nullptr;
     uint64_t view_n = 0;
     uint64_t cursor = 0;
@@ -744,10 +761,48 @@ struct HipEngine {
      * driver configs) was a large, box-dependent share of step time. */
     std::vector<std::pair<uint8_t *, size_t>> sblocks;
     size_t sblock_i = 0, s_off = 0;
+    /* while a split compaction is pending its phase-1 buffers live at the
+     * arena head; reads between begin/finish reset only to this watermark
+     * (unguarded resets let reads reuse the pending buffers) */
+    bool arena_pinned = false;
+    size_t pin_block = 0, pin_off = 0;
     void scratch_reset()
     {
-        sblock_i = 0;
-        s_off = 0;
+        sblock_i = arena_pinned ? pin_block : 0;
+        s_off = arena_pinned ? pin_off : 0;
+    }
+    void arena_pin()
+    {
+        arena_pinned = true;
+        pin_block = sblock_i;
+        pin_off = s_off;
+    }
+    void arena_unpin() { arena_pinned = false; }
+    /* reclaim parked scanner contexts older than 5 minutes (the reference
+     * drops unused contexts after std::chrono::minutes(5),
+     * pegasus_server_impl.cpp:1381-1387; a used context re-parks under a
+     * fresh handle with a fresh timer, as ours does).  Driven by the
+     * caller-supplied epoch clock so tests are deterministic. */
+    void gc_ctxs(uint32_t epoch_now)
+    {
+        for (auto it = ctxs.begin(); it != ctxs.end();) {
+            if (epoch_now > it->second->parked_at &&
+                epoch_now - it->second->parked_at > 300) {
+                delete it->second;
+                it = ctxs.erase(it);
+            } else {
+                ++it;
+            }
+        }
+    }
+    /* compaction rebuilt the run list: every parked view indexes freed runs.
+     * Reclaim them all; a later scan_next returns kNotFound, the reference's
+     * expired-context behavior (on_scan:1539-1541). */
+    void invalidate_ctxs()
+    {
+        for (auto &p : ctxs)
+            delete p.second;
+        ctxs.clear();
     }
     template <typename T> T *talloc(uint64_t n_bytes)
     {
@@ -878,7 +933,9 @@ struct HipEngine {
         uint8_t *d_start = start ? upload_tmp(start->data(), start->size()) : nullptr;
         uint8_t *d_stop = stop_excl ? upload_tmp(stop_excl->data(), stop_excl->size()) : nullptr;
         launch_bounds(dr, R, d_start, start ? start->size() : 0, d_lo, 0, stream);
-        launch_bounds(dr, R, d_stop, stop_excl ? stop_excl->size() : 0, d_hi, 0, stream);
+        /* null stop = unbounded: upper flag makes the null-key case yield n */
+        launch_bounds(dr, R, d_stop, stop_excl ? stop_excl->size() : 0, d_hi,
+                      stop_excl ? 0 : 1, stream);
         std::vector<uint64_t> lo(R), hi(R), wprefix(R + 1);
         HIP_OK(hipMemcpyAsync(lo.data(), d_lo, R * 8, hipMemcpyDeviceToHost, stream));
         HIP_OK(hipMemcpyAsync(hi.data(), d_hi, R * 8, hipMemcpyDeviceToHost, stream));
@@ -1501,7 +1558,9 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
     ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
-        q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+        q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3 ||
+        (q->hash_key_filter_type == RRDB_FT_MATCH_PREFIX &&
+         !hklen_ok(q->hash_key_filter_pattern.len))) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_get_scanner:1168-1186 */
         return out->error;
     }
@@ -1557,10 +1616,12 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
     ctx->d_sk_pat = e->upload_bytes(q->sort_key_filter_pattern.data, ctx->sk_pat_len);
 
     scan_batch_gpu(e, ctx, epoch_now, out, a);
+    e->gc_ctxs(epoch_now);
     if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
         delete ctx;
     } else {
         ctx->id = ++e->next_ctx_id;
+        ctx->parked_at = epoch_now;
         e->ctxs[ctx->id] = ctx;
         out->context_id = ctx->id;
     }
@@ -1574,6 +1635,7 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
     Arena *a = result_init(out);
     e->activate();
     e->scratch_reset();
+    e->gc_ctxs(epoch_now);
     HipScanCtx *c = nullptr;
     {
         auto it = e->ctxs.find(context_id);
@@ -1583,7 +1645,8 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
         }
     }
     if (!c) {
-        out->error = RRDB_NOT_FOUND; /* on_scan:1539-1541 */
+        out->error = RRDB_NOT_FOUND; /* on_scan:1539-1541; also the 5-min
+                                        expired-context path (:1381-1387) */
         return out->error;
     }
     scan_batch_gpu(e, c, epoch_now, out, a);
@@ -1591,6 +1654,7 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
         delete c;
     } else {
         c->id = ++e->next_ctx_id; /* re-park under a fresh handle (on_scan:1516-1526) */
+        c->parked_at = epoch_now;
         e->ctxs[c->id] = c;
         out->context_id = c->id;
     }
@@ -1618,6 +1682,10 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
     ((HipEngine *)h)->scratch_reset();
     result_init(out);
     e->activate();
+    if (!hklen_ok(hklen)) {
+        out->error = RRDB_INVALID_ARGUMENT;
+        return out->error;
+    }
     /* start=(hk,""), stop=next(hk) (on_sortkey_count:1030-1036); count all
      * visible non-expired rows — no count cap in the reference's loop */
     std::string start = make_key(hash_key, hklen, nullptr, 0);
@@ -1660,6 +1728,37 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
 }
 
 /* body of on_multi_get; caller holds the engine lock */
+/* it->Valid() after a limit exit (on_multi_get:777-788): does any
+ * rocksdb-iterator-visible record exist beyond the range boundary?
+ * forward: key >= bound; reverse: key < bound. */
+static bool valid_beyond(HipEngine *e, const std::string &bound, bool reverse)
+{
+    if (e->runs.empty())
+        return false;
+    if ((int)e->runs.size() <= RRDB_MAX_RUNS) {
+        uint32_t *d_out = e->talloc<uint32_t>(4);
+        uint8_t *d_b = e->upload_tmp(bound.data(), bound.size());
+        launch_valid_beyond(e->dev_runs(), (int)e->runs.size(), d_b, bound.size(),
+                            reverse ? 1 : 0, d_out, e->stream);
+        uint32_t v = 0;
+        HIP_OK(hipMemcpyAsync(&v, d_out, 4, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        e->tfree(d_out);
+        e->tfree(d_b);
+        return v != 0;
+    }
+    /* > RRDB_MAX_RUNS runs: fall back to a visible-view existence check */
+    uint64_t *d_view = nullptr, n = 0;
+    std::string b = bound;
+    if (!reverse)
+        e->build_view(&b, nullptr, &d_view, &n);
+    else
+        e->build_view(nullptr, &b, &d_view, &n);
+    if (d_view)
+        (void)hipFree(d_view);
+    return n != 0;
+}
+
 static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
                                 rrdb_result *out)
 {
@@ -1668,7 +1767,8 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
     e->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
-    if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+    if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3 ||
+        !hklen_ok(q->hash_key.len)) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_multi_get:508-517 */
         return out->error;
     }
@@ -1760,7 +1860,7 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
         stop_excl.push_back('\0');
 
     /* ---- fused single-launch fast path (small ranges, the YCSB-E shape) ---- */
-    if (!e->runs.empty()) {
+    if (!e->runs.empty() && (int)e->runs.size() <= RRDB_MAX_RUNS) {
         MgFusedArgs fa{};
         fa.start = e->upload_tmp(start.data(), start.size());
         fa.start_len = start.size();
@@ -1865,9 +1965,26 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
     std::vector<uint64_t> sel; /* view positions to emit, ascending */
     bool complete = false;
     bool skipped_first = false;
+    /* reverse + start-exclusive: a record == start is OUT OF RANGE for the
+     * reverse walk (on_multi_get:697-700) — excluded, and it keeps the
+     * iterator Valid() on a limit exit */
+    uint64_t lo_skip = 0;
+    if (q->reverse && !start_inclusive && wstart == 0 && n > 0) {
+        uint32_t *d_eq = e->talloc<uint32_t>(4);
+        uint8_t *d_b = e->upload_tmp(start.data(), start.size());
+        launch_first_eq(dr, d_view, 1, d_b, start.size(), d_eq, e->stream);
+        uint32_t eq = 0;
+        HIP_OK(hipMemcpyAsync(&eq, d_eq, 4, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipStreamSynchronize(e->stream));
+        e->tfree(d_eq);
+        e->tfree(d_b);
+        lo_skip = eq ? 1 : 0;
+    }
     uint64_t steps = w;
     for (uint64_t s = 0; s < steps; s++) {
         uint64_t wi = q->reverse ? (steps - 1 - s) : s;
+        if (q->reverse && lo_skip && wi == 0)
+            break; /* reached the excluded == start record: out of range */
         if (count >= (int64_t)max_kv_count || iteration_count >= max_iteration_count ||
             size >= max_iter_size)
             break;
@@ -1898,16 +2015,23 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
         sel.push_back(wstart + wi);
         count++;
         size += (int64_t)ksz[wi] + (int64_t)vsz[wi];
-        if (s + 1 == steps)
-            complete = true;
     }
+    /* completion (on_multi_get:777-788): kIncomplete iff the iterator is
+     * still Valid() after a limit exit — even when the remaining records lie
+     * past the range */
     {
-        uint64_t total = n - (skipped_first ? 1 : 0);
-        /* also: entries outside the window were never reachable */
-        if (w < n)
-            complete = false;
+        bool limit_exit = (count >= (int64_t)max_kv_count ||
+                           iteration_count >= max_iteration_count || size >= max_iter_size);
+        uint64_t countable = w - (skipped_first ? 1 : 0) - lo_skip;
+        bool consumed_all = (iteration_count >= countable);
+        if (w < n || !consumed_all)
+            complete = false; /* in-range records remain: iterator valid */
+        else if (!limit_exit)
+            complete = true; /* walked past the range end */
+        else if (q->reverse)
+            complete = !(lo_skip || valid_beyond(e, start, true));
         else
-            complete = (iteration_count >= total);
+            complete = !valid_beyond(e, stop_excl, false);
     }
     uint64_t m = sel.size();
     if (q->reverse)
@@ -2103,6 +2227,7 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     for (int i = 0; i < 6; i++)
         e->pend.ev[i] = ev[i];
     e->pend.st = st;
+    e->arena_pin(); /* reads between begin/finish must not reuse pend buffers */
     (void)d_lo;
     (void)d_hi;
     (void)d_ksz;
@@ -2118,6 +2243,7 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
         return RRDB_INVALID_ARGUMENT;
     e->activate();
     e->pend.active = false;
+    e->arena_unpin();
     rrdb_compact_stats st = e->pend.st;
     if (e->pend.trivial) {
         if (stats)
@@ -2220,6 +2346,7 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
     if (keep_inputs) {
         /* benchmarking: the pass ran in full; drop the output, keep inputs */
     } else {
+        e->invalidate_ctxs(); /* parked views reference the freed runs */
         for (auto &r : e->runs)
             e->free_run(r);
         e->runs.clear();
@@ -2288,6 +2415,8 @@ int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t
 {
     auto *e = (HipEngine *)h;
     std::lock_guard<std::mutex> g(e->mu);
+    if (!hklen_ok(hklen))
+        return RRDB_INVALID_ARGUMENT;
     std::string key = make_key(hash_key, hklen, sort_key, sklen);
     uint32_t hdr = hdr_len(e->data_version);
     std::string val(hdr + vlen, '\0');
@@ -2309,6 +2438,8 @@ int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint
 {
     auto *e = (HipEngine *)h;
     std::lock_guard<std::mutex> g(e->mu);
+    if (!hklen_ok(hklen))
+        return RRDB_INVALID_ARGUMENT;
     std::string key = make_key(hash_key, hklen, sort_key, sklen);
     e->memtable[key] = {std::string(), e->next_seq_floor++, RRDB_KIND_DELETE};
     return RRDB_OK;
@@ -2384,14 +2515,11 @@ int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
     mkdir(dir, 0755);
     if (mkdir(path.c_str(), 0755) != 0)
         return RRDB_IO_ERROR;
-    FILE *mf = fopen((path + "/MANIFEST").c_str(), "w");
-    if (!mf)
-        return RRDB_IO_ERROR;
-    fprintf(mf, "rrdb-checkpoint 1\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n",
-            e->data_version, (unsigned long long)e->next_seq_floor, (int)e->runs.size());
-    for (size_t i = 0; i < e->runs.size(); i++)
-        fprintf(mf, "run %zu %llu\n", i, (unsigned long long)e->runs[i].n);
-    fclose(mf);
+    /* run files first (crc64 of each goes into the MANIFEST; a restorer
+     * verifies — the reference's checkpoints carry rocksdb's per-block
+     * checksums, restated here at file granularity) */
+    std::string crc_lines;
+    char cl[128];
     std::vector<uint8_t> buf;
     for (size_t i = 0; i < e->runs.size(); i++) {
         const RunBuf &r = e->runs[i];
@@ -2401,21 +2529,38 @@ int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
         HIP_OK(hipMemcpy(voff.data(), r.voff, (r.n + 1) * 8, hipMemcpyDeviceToHost));
         HIP_OK(hipMemcpy(sk.data(), r.sk, r.n * 8, hipMemcpyDeviceToHost));
         std::string base = path + "/run_" + std::to_string(i);
+        auto emit = [&](const char *ext, const void *p, uint64_t n) {
+            if (!write_file(base + "." + ext, p, n))
+                return false;
+            snprintf(cl, sizeof(cl), "crc run_%zu.%s %016llx\n", i, ext,
+                     (unsigned long long)host_crc64(p, n));
+            crc_lines += cl;
+            return true;
+        };
         buf.resize(koff[r.n]);
         HIP_OK(hipMemcpy(buf.data(), r.keys, koff[r.n], hipMemcpyDeviceToHost));
-        if (!write_file(base + ".keys", buf.data(), koff[r.n]))
+        if (!emit("keys", buf.data(), koff[r.n]))
             return RRDB_IO_ERROR;
-        if (!write_file(base + ".koff", koff.data(), (r.n + 1) * 8))
+        if (!emit("koff", koff.data(), (r.n + 1) * 8))
             return RRDB_IO_ERROR;
         buf.resize(voff[r.n]);
         HIP_OK(hipMemcpy(buf.data(), r.vals, voff[r.n], hipMemcpyDeviceToHost));
-        if (!write_file(base + ".vals", buf.data(), voff[r.n]))
+        if (!emit("vals", buf.data(), voff[r.n]))
             return RRDB_IO_ERROR;
-        if (!write_file(base + ".voff", voff.data(), (r.n + 1) * 8))
+        if (!emit("voff", voff.data(), (r.n + 1) * 8))
             return RRDB_IO_ERROR;
-        if (!write_file(base + ".sk", sk.data(), r.n * 8))
+        if (!emit("sk", sk.data(), r.n * 8))
             return RRDB_IO_ERROR;
     }
+    FILE *mf = fopen((path + "/MANIFEST").c_str(), "w");
+    if (!mf)
+        return RRDB_IO_ERROR;
+    fprintf(mf, "rrdb-checkpoint 1\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n",
+            e->data_version, (unsigned long long)e->next_seq_floor, (int)e->runs.size());
+    for (size_t i = 0; i < e->runs.size(); i++)
+        fprintf(mf, "run %zu %llu\n", i, (unsigned long long)e->runs[i].n);
+    fputs(crc_lines.c_str(), mf); /* trailing lines; older restorers ignore */
+    fclose(mf);
     return RRDB_OK;
 }
 
@@ -2435,11 +2580,31 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
     int n_runs = 0, ver = 0;
     if (fscanf(mf, "rrdb-checkpoint %d\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n", &ver,
                &dv, &floor_, &n_runs) != 4 ||
-        ver != 1) {
+        ver != 1 || n_runs < 0) {
         fclose(mf);
         return RRDB_CORRUPTION;
     }
+    /* run record counts + optional per-file crc lines (older checkpoints
+     * carry none; then crc validation is skipped) */
+    std::vector<unsigned long long> mrows((size_t)n_runs, ~0ull);
+    std::unordered_map<std::string, uint64_t> mcrc;
+    {
+        char line[256];
+        while (fgets(line, sizeof(line), mf)) {
+            unsigned long long a, b;
+            char name[128];
+            if (sscanf(line, "run %llu %llu", &a, &b) == 2) {
+                if (a < (unsigned long long)n_runs)
+                    mrows[(size_t)a] = b;
+            } else if (sscanf(line, "crc %127s %llx", name, &b) == 2) {
+                mcrc[name] = b;
+            }
+        }
+    }
     fclose(mf);
+    for (int i = 0; i < n_runs; i++)
+        if (mrows[(size_t)i] == ~0ull)
+            return RRDB_CORRUPTION; /* MANIFEST missing a run line */
     e->data_version = dv;
     for (int i = 0; i < n_runs; i++) {
         std::string base = path + "/run_" + std::to_string(i);
@@ -2448,8 +2613,38 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
             !read_file(base + ".vals", vals) || !read_file(base + ".voff", voff) ||
             !read_file(base + ".sk", sk))
             return RRDB_IO_ERROR;
+        /* integrity validation: crc when present, sizes vs the
+         * MANIFEST count, offset monotonicity, key sortedness — a truncated
+         * or bit-flipped file returns kCorruption instead of corrupting the
+         * engine (the reference restores from checksummed rocksdb
+         * checkpoints; same guarantee at our file granularity) */
+        auto crc_ok = [&](const char *ext, const std::vector<uint8_t> &data) {
+            auto it = mcrc.find("run_" + std::to_string(i) + "." + ext);
+            return it == mcrc.end() || it->second == host_crc64(data.data(), data.size());
+        };
+        if (!crc_ok("keys", keys) || !crc_ok("koff", koff) || !crc_ok("vals", vals) ||
+            !crc_ok("voff", voff) || !crc_ok("sk", sk))
+            return RRDB_CORRUPTION;
+        uint64_t nrec = mrows[(size_t)i];
+        if (koff.size() != (nrec + 1) * 8 || voff.size() != (nrec + 1) * 8 ||
+            sk.size() != nrec * 8)
+            return RRDB_CORRUPTION;
+        const uint64_t *ko = (const uint64_t *)koff.data();
+        const uint64_t *vo = (const uint64_t *)voff.data();
+        if (ko[0] != 0 || vo[0] != 0 || ko[nrec] != keys.size() || vo[nrec] != vals.size())
+            return RRDB_CORRUPTION;
+        for (uint64_t j = 0; j < nrec; j++)
+            if (ko[j + 1] <= ko[j] || vo[j + 1] < vo[j])
+                return RRDB_CORRUPTION; /* keys nonempty+increasing offsets */
+        for (uint64_t j = 0; j + 1 < nrec; j++) {
+            uint64_t la = ko[j + 1] - ko[j], lb = ko[j + 2] - ko[j + 1];
+            uint64_t m = la < lb ? la : lb;
+            int c = memcmp(keys.data() + ko[j], keys.data() + ko[j + 1], m);
+            if (c > 0 || (c == 0 && la >= lb))
+                return RRDB_CORRUPTION; /* keys must be strictly increasing */
+        }
         RunBuf r;
-        r.n = koff.size() / 8 - 1;
+        r.n = nrec;
         r.keys = e->upload_bytes(keys.data(), keys.size());
         r.koff = (uint64_t *)e->upload_bytes(koff.data(), koff.size());
         r.vals = e->upload_bytes(vals.data(), vals.size());

@@ -911,6 +911,83 @@ __global__ void k_first_eq(const DevRun *runs, const uint64_t *view, uint64_t n,
     out[0] = dev_key_cmp(k, kl, key, klen) == 0;
 }
 
+/* it->Valid() restatement for multi_get's limit-exit (on_multi_get:777-788):
+ * does any rocksdb-iterator-visible record exist beyond the range boundary?
+ * Visible = the newest version of its key group is a PUT (tombstone groups
+ * are merged away by the iterator; expiry is app-level and does NOT hide
+ * records here).  Walks merged key groups outward from the boundary —
+ * expected O(1) groups (the first group is usually a live PUT).
+ * forward: first group with key >= bound; reverse: first with key < bound. */
+__device__ static int dev_valid_beyond(const DevRun *runs, int R, const uint8_t *bound,
+                                       uint64_t blen, int reverse)
+{
+    uint64_t cur[RRDB_MAX_RUNS];
+    if (!reverse) {
+        for (int q = 0; q < R; q++)
+            cur[q] = dev_lower_bound(runs[q], bound, blen, 0, runs[q].n);
+        for (;;) {
+            int best = -1;
+            const uint8_t *bk = nullptr;
+            uint64_t bl = 0;
+            for (int q = R - 1; q >= 0; q--) { /* ties: newest (highest q) first */
+                if (cur[q] >= runs[q].n)
+                    continue;
+                uint64_t kl;
+                const uint8_t *k = run_key(runs[q], cur[q], &kl);
+                if (best < 0 || dev_key_cmp(k, kl, bk, bl) < 0) {
+                    best = q;
+                    bk = k;
+                    bl = kl;
+                }
+            }
+            if (best < 0)
+                return 0; /* exhausted the DB: iterator invalid */
+            if (!(runs[best].sk[cur[best]] & 1))
+                return 1; /* newest version is a PUT: iterator lands here */
+            for (int q = 0; q < R; q++) /* tombstone group: skip the key */
+                cur[q] = dev_upper_bound(runs[q], bk, bl, cur[q], runs[q].n);
+        }
+    }
+    for (int q = 0; q < R; q++)
+        cur[q] = dev_lower_bound(runs[q], bound, blen, 0, runs[q].n);
+    for (;;) {
+        int best = -1;
+        const uint8_t *bk = nullptr;
+        uint64_t bl = 0;
+        for (int q = R - 1; q >= 0; q--) {
+            if (cur[q] == 0)
+                continue;
+            uint64_t kl;
+            const uint8_t *k = run_key(runs[q], cur[q] - 1, &kl);
+            if (best < 0 || dev_key_cmp(k, kl, bk, bl) > 0) {
+                best = q;
+                bk = k;
+                bl = kl;
+            }
+        }
+        if (best < 0)
+            return 0;
+        if (!(runs[best].sk[cur[best] - 1] & 1))
+            return 1;
+        for (int q = 0; q < R; q++)
+            cur[q] = dev_lower_bound(runs[q], bk, bl, 0, cur[q]);
+    }
+}
+
+__global__ void k_valid_beyond(const DevRun *runs, int R, const uint8_t *bound, uint64_t blen,
+                               int reverse, uint32_t *out)
+{
+    if (blockIdx.x != 0 || threadIdx.x != 0)
+        return;
+    out[0] = (uint32_t)dev_valid_beyond(runs, R, bound, blen, reverse);
+}
+
+void launch_valid_beyond(const DevRun *runs, int R, const uint8_t *bound, uint64_t blen,
+                         int reverse, uint32_t *out, hipStream_t s)
+{
+    k_valid_beyond<<<1, 1, 0, s>>>(runs, R, bound, blen, reverse, out);
+}
+
 /* ================= compaction =================
  * Disposition of every record in the full-range order[] array:
  * KeyWithTTLCompactionFilter::Filter (key_ttl_compaction_filter.h:55-92) on
@@ -1840,17 +1917,37 @@ __device__ static void mg_core(const DevRun *runs, int R, const MgFusedArgs &a,
         uint64_t count = 0, iteration = 0;
         int64_t size = 0;
         int complete = 0;
+        /* reverse + start-exclusive: records == start are OUT OF RANGE for
+         * the reverse walk (on_multi_get:697-700 — c==0 && !start_inclusive
+         * -> complete); they sit at the lowest merged positions.  oor_exists
+         * notes whether the iterator would land on one (it->Valid()). */
+        uint64_t t_floor = 0;
+        int oor_exists = 0;
+        if (a.reverse && !a.start_inclusive) {
+            for (uint64_t t = 0; t < total; t++) {
+                uint64_t id = s_id[t];
+                const DevRun &r = runs[id >> 40];
+                uint64_t i = id & 0xFFFFFFFFFFull, kl;
+                const uint8_t *k = run_key(r, i, &kl);
+                if (dev_key_cmp(k, kl, mg_start, mg_start_len) != 0)
+                    break;
+                if (s_state[t] != 2)
+                    oor_exists = 1; /* iterator-visible == start */
+                t_floor = t + 1;
+            }
+        }
         uint64_t countable = 0;
         /* countable = entries the reference iterator would visit (visible) */
-        for (uint64_t t = 0; t < total; t++)
+        for (uint64_t t = t_floor; t < total; t++)
             if (s_state[t] != 2 && s_state[t] != 3)
                 countable++;
             else if (s_state[t] == 3)
                 fallback = 1;
         int skipped_first = 0;
         uint64_t visited = 0;
-        for (uint64_t s = 0; s < total && !fallback; s++) {
-            uint64_t t = a.reverse ? (total - 1 - s) : s;
+        uint64_t span = total - t_floor;
+        for (uint64_t s = 0; s < span && !fallback; s++) {
+            uint64_t t = a.reverse ? (total - 1 - s) : t_floor + s;
             if (s_state[t] == 2)
                 continue; /* invisible to the iterator */
             if (count >= a.max_kv_count || iteration >= a.max_iteration_count ||
@@ -1885,7 +1982,23 @@ __device__ static void mg_core(const DevRun *runs, int R, const MgFusedArgs &a,
             vb += s_vlen_lo[t];
             size += (int64_t)s_klen[t] + (int64_t)s_vlen_lo[t];
         }
-        complete = (iteration >= countable - (uint64_t)skipped_first);
+        /* completion (on_multi_get:777-788): kIncomplete iff the iterator is
+         * still Valid() after a limit exit — even when the remaining records
+         * lie past the range */
+        {
+            int limit_exit = (count >= a.max_kv_count || iteration >= a.max_iteration_count ||
+                              size >= a.max_iteration_size);
+            int consumed_all = (iteration >= countable - (uint64_t)skipped_first);
+            if (!consumed_all)
+                complete = 0; /* stopped mid-range */
+            else if (!limit_exit)
+                complete = 1; /* walked past the range end */
+            else if (a.reverse)
+                complete = !(oor_exists ||
+                             dev_valid_beyond(runs, R, mg_start, mg_start_len, 1));
+            else
+                complete = !dev_valid_beyond(runs, R, mg_stop, mg_stop_len, 0);
+        }
         if (2 * (nsel + 1) * 8 + kb + vb > blob_cap)
             fallback = 1;
         if (fallback) {

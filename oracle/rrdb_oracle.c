@@ -818,6 +818,7 @@ typedef struct {
 /* a parked scanner: materialized visible view of [cursor..stop) */
 struct ScanCtx {
     int64_t id;
+    uint32_t parked_at; /* epoch_now when (re-)parked; 5-min GC */
     uint64_t *view; /* packed (run<<40|idx), key-ordered visible records */
     uint64_t view_n;
     uint64_t cursor; /* next view position */
@@ -1255,6 +1256,10 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
     Engine *e = (Engine *)h;
     rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     result_init(out);
+    if (hklen >= 0xFFFFull) {
+        out->error = RRDB_INVALID_ARGUMENT;
+        return out->error;
+    }
     /* start = generate_key(hk,""), stop = generate_next_blob(hk)
      * (on_sortkey_count:1030-1036) */
     uint64_t sl = 2 + hklen;
@@ -1317,6 +1322,71 @@ static uint64_t next_blob(uint8_t *key, uint64_t len)
     return p + 1;
 }
 
+
+/* it->Valid() restatement for multi_get's limit exit (on_multi_get:777-788):
+ * does any rocksdb-iterator-visible record exist beyond the range boundary?
+ * Visible = the newest version of its key group is a PUT (tombstone groups
+ * are merged away; expiry is app-level and does not hide records here).
+ * forward: first group with key >= bound; reverse: first with key < bound. */
+static int oracle_valid_beyond(Engine *e, const uint8_t *bound, uint64_t blen, int reverse)
+{
+    int R = e->n_runs;
+    if (R == 0)
+        return 0;
+    uint64_t *cur = (uint64_t *)malloc((size_t)R * 8);
+    int found = 0;
+    for (int q = 0; q < R; q++)
+        cur[q] = run_lower_bound(&e->runs[q], bound, blen);
+    for (;;) {
+        int best = -1;
+        const uint8_t *bk = NULL;
+        uint64_t bl = 0;
+        for (int q = R - 1; q >= 0; q--) { /* ties: newest (highest q) first */
+            uint64_t kl;
+            const uint8_t *k;
+            if (!reverse) {
+                if (cur[q] >= e->runs[q].n)
+                    continue;
+                k = run_key(&e->runs[q], cur[q], &kl);
+                if (best < 0 || key_cmp(k, kl, bk, bl) < 0) {
+                    best = q;
+                    bk = k;
+                    bl = kl;
+                }
+            } else {
+                if (cur[q] == 0)
+                    continue;
+                k = run_key(&e->runs[q], cur[q] - 1, &kl);
+                if (best < 0 || key_cmp(k, kl, bk, bl) > 0) {
+                    best = q;
+                    bk = k;
+                    bl = kl;
+                }
+            }
+        }
+        if (best < 0)
+            break; /* DB exhausted: iterator invalid */
+        uint64_t pos = reverse ? cur[best] - 1 : cur[best];
+        if (!(e->runs[best].sk[pos] & 1)) {
+            found = 1; /* newest version is a PUT: iterator lands here */
+            break;
+        }
+        for (int q = 0; q < R; q++) { /* tombstone group: skip the key */
+            if (!reverse) {
+                uint64_t u = run_upper_bound(&e->runs[q], bk, bl);
+                if (u > cur[q])
+                    cur[q] = u;
+            } else {
+                uint64_t l = run_lower_bound(&e->runs[q], bk, bl);
+                if (l < cur[q])
+                    cur[q] = l;
+            }
+        }
+    }
+    free(cur);
+    return found;
+}
+
 int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_now,
                        rrdb_result *out)
 {
@@ -1325,7 +1395,8 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     arena *a = result_init(out);
     uint32_t hdr = value_hdr_len(e->data_version);
 
-    if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+    if (q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3 ||
+        q->hash_key.len >= 0xFFFFull) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_multi_get:508-517 */
         return out->error;
     }
@@ -1458,9 +1529,22 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
     /* forward: walk view left->right; reverse: right->left, then reverse
      * results (on_multi_get:616-676 / :678-778).  first_exclusive skip of the
      * exact start/stop key mirrors :636-643 / :700-707. */
+    /* reverse + start-exclusive: a record == start is OUT OF RANGE for the
+     * reverse walk (on_multi_get:697-700) — excluded, and it keeps the
+     * iterator Valid() on a limit exit */
+    uint64_t lo_skip = 0;
+    if (q->reverse && !start_inclusive && n > 0) {
+        const Run *r0 = &e->runs[view[0] >> 40];
+        uint64_t kl0;
+        const uint8_t *k0 = run_key(r0, view[0] & 0xFFFFFFFFFFull, &kl0);
+        if (key_cmp(k0, kl0, start, start_len) == 0)
+            lo_skip = 1;
+    }
     uint64_t steps = n;
     for (uint64_t s = 0; s < steps; s++) {
         uint64_t vi = q->reverse ? (steps - 1 - s) : s;
+        if (q->reverse && lo_skip && vi == 0)
+            break; /* reached the excluded == start record: out of range */
         const Run *r = &e->runs[view[vi] >> 40];
         uint64_t idx = view[vi] & 0xFFFFFFFFFFull;
         uint64_t kl, vl;
@@ -1498,24 +1582,46 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
         count++;
         size += (int64_t)keys[m].len + (int64_t)vals[m].len;
         m++;
-        if (s + 1 == steps)
-            complete = 1;
     }
-    /* complete iff every countable view element was iterated: mirrors
-     * `it->Valid() && !complete -> kIncomplete` (on_multi_get:789-799) */
+    /* completion (on_multi_get:777-788): kIncomplete iff the iterator is
+     * still Valid() after a limit exit — even when the remaining records lie
+     * past the range */
     {
-        uint64_t total = n;
-        if (n > 0) {
-            uint64_t fi = q->reverse ? n - 1 : 0;
+        int limit_exit = (count >= (int64_t)max_kv_count ||
+                          iteration_count >= max_iteration_count ||
+                          size >= max_iteration_size);
+        uint64_t skipped_first = 0;
+        if (n > lo_skip) {
+            uint64_t fi = q->reverse ? n - 1 : lo_skip;
             const Run *r = &e->runs[view[fi] >> 40];
             uint64_t idx = view[fi] & 0xFFFFFFFFFFull, kl;
             const uint8_t *k = run_key(r, idx, &kl);
             if (!q->reverse && !start_inclusive && key_cmp(k, kl, start, start_len) == 0)
-                total--;
+                skipped_first = 1;
             if (q->reverse && !stop_inclusive && key_cmp(k, kl, stop, stop_len) == 0)
-                total--;
+                skipped_first = 1;
         }
-        complete = (iteration_count >= total);
+        uint64_t countable = n - skipped_first - lo_skip;
+        int consumed_all = (iteration_count >= countable);
+        if (!consumed_all)
+            complete = 0; /* stopped mid-range: iterator valid */
+        else if (!limit_exit)
+            complete = 1; /* walked past the range end */
+        else if (q->reverse) {
+            /* bound semantics: exists visible < start (lo_skip covers == start) */
+            complete = !(lo_skip || oracle_valid_beyond(e, start, start_len, 1));
+        } else {
+            /* forward: exists visible >= exclusive stop bound */
+            if (stop_inclusive) {
+                uint8_t *stop2 = (uint8_t *)malloc(stop_len + 1);
+                memcpy(stop2, stop, stop_len);
+                stop2[stop_len] = 0;
+                complete = !oracle_valid_beyond(e, stop2, stop_len + 1, 0);
+                free(stop2);
+            } else {
+                complete = !oracle_valid_beyond(e, stop, stop_len, 0);
+            }
+        }
     }
     if (q->reverse && m > 1) {
         /* revert order to ascending (on_multi_get:758-765) */
@@ -1655,8 +1761,34 @@ static void scan_batch(Engine *e, ScanCtx *c, uint32_t epoch_now, rrdb_result *o
     (void)complete;
 }
 
-static int64_t park_ctx(Engine *e, ScanCtx *c)
+/* the reference drops a parked context unused for 5 minutes
+ * (pegasus_server_impl.cpp:1381-1387); each use re-parks under a fresh
+ * handle with a fresh timer.  Driven by the caller's epoch clock. */
+static void gc_ctxs(Engine *e, uint32_t epoch_now)
 {
+    for (int i = 0; i < e->n_ctxs;) {
+        if (e->ctxs[i] && epoch_now > e->ctxs[i]->parked_at &&
+            epoch_now - e->ctxs[i]->parked_at > 300) {
+            free_ctx(e->ctxs[i]);
+            e->ctxs[i] = e->ctxs[--e->n_ctxs];
+        } else {
+            i++;
+        }
+    }
+}
+
+/* compaction rebuilt the run list: parked views index freed runs; reclaim
+ * them (a later scan_next gets the expired-context kNotFound path) */
+static void invalidate_ctxs(Engine *e)
+{
+    for (int i = 0; i < e->n_ctxs; i++)
+        free_ctx(e->ctxs[i]);
+    e->n_ctxs = 0;
+}
+
+static int64_t park_ctx(Engine *e, ScanCtx *c, uint32_t epoch_now)
+{
+    c->parked_at = epoch_now;
     c->id = ++e->next_ctx_id;
     if (e->n_ctxs == e->ctxs_cap) {
         e->ctxs_cap = e->ctxs_cap ? e->ctxs_cap * 2 : 8;
@@ -1684,7 +1816,9 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
     rrdb_flush(h); /* committed writes visible to reads (memtable read path) */
     result_init(out);
     if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
-        q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3) {
+        q->sort_key_filter_type < 0 || q->sort_key_filter_type > 3 ||
+        (q->hash_key_filter_type == RRDB_FT_MATCH_PREFIX &&
+         q->hash_key_filter_pattern.len >= 0xFFFFull)) {
         out->error = RRDB_INVALID_ARGUMENT; /* on_get_scanner:1168-1186 */
         return out->error;
     }
@@ -1716,6 +1850,7 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
         return RRDB_OK;
     }
 
+    gc_ctxs(e, epoch_now);
     ScanCtx *ctx = (ScanCtx *)calloc(1, sizeof(ScanCtx));
     /* view = visible records in [start(incl), stop(stop_inclusive?incl:excl)];
      * first_exclusive start handled by advancing past an exact match
@@ -1765,7 +1900,7 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
     if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
         free_ctx(ctx);
     } else {
-        out->context_id = park_ctx(e, ctx);
+        out->context_id = park_ctx(e, ctx, epoch_now);
     }
     return out->error;
 }
@@ -1774,6 +1909,7 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
 {
     Engine *e = (Engine *)h;
     result_init(out);
+    gc_ctxs(e, epoch_now);
     ScanCtx *c = fetch_ctx(e, context_id);
     if (!c) {
         out->error = RRDB_NOT_FOUND; /* on_scan:1539-1541 */
@@ -1783,7 +1919,7 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
     if (out->context_id == RRDB_SCAN_CONTEXT_ID_COMPLETED) {
         free_ctx(c);
     } else {
-        out->context_id = park_ctx(e, c); /* re-put -> new handle (on_scan:1516-1526) */
+        out->context_id = park_ctx(e, c, epoch_now); /* re-put -> new handle (on_scan:1516-1526) */
     }
     return out->error;
 }
@@ -2000,6 +2136,7 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
     }
 
     /* swap in the merged run */
+    invalidate_ctxs(e); /* parked views index the freed runs */
     for (int i = 0; i < e->n_runs; i++)
         free_run(&e->runs[i]);
     e->n_runs = 0;
@@ -2093,6 +2230,8 @@ int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t
 {
     Engine *e = (Engine *)h;
     uint64_t klen;
+    if (hklen >= 0xFFFFull)
+        return RRDB_INVALID_ARGUMENT; /* 2-byte length prefix cap (key_schema.h:43) */
     uint8_t *key = make_key(hash_key, hklen, sort_key, sklen, &klen);
     uint32_t hdr = value_hdr_len(e->data_version);
     uint8_t *val = (uint8_t *)calloc(1, hdr + vlen);
@@ -2109,6 +2248,8 @@ int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint
 {
     Engine *e = (Engine *)h;
     uint64_t klen;
+    if (hklen >= 0xFFFFull)
+        return RRDB_INVALID_ARGUMENT;
     uint8_t *key = make_key(hash_key, hklen, sort_key, sklen, &klen);
     mem_append(e, key, klen, NULL, 0, RRDB_KIND_DELETE);
     return RRDB_OK;
@@ -2236,6 +2377,20 @@ int32_t rrdb_checkpoint(void *h, const char *dir, uint64_t decree)
             e->data_version, (unsigned long long)e->next_seq_floor, e->n_runs);
     for (int i = 0; i < e->n_runs; i++)
         fprintf(mf, "run %d %llu\n", i, (unsigned long long)e->runs[i].n);
+    /* per-file crc64 lines (integrity on restore; older restorers ignore) */
+    for (int i = 0; i < e->n_runs; i++) {
+        const Run *r = &e->runs[i];
+        fprintf(mf, "crc run_%d.keys %016llx\n", i,
+                (unsigned long long)crc64_calc(r->keys, (size_t)r->koff[r->n], 0));
+        fprintf(mf, "crc run_%d.koff %016llx\n", i,
+                (unsigned long long)crc64_calc(r->koff, (size_t)(r->n + 1) * 8, 0));
+        fprintf(mf, "crc run_%d.vals %016llx\n", i,
+                (unsigned long long)crc64_calc(r->vals, (size_t)r->voff[r->n], 0));
+        fprintf(mf, "crc run_%d.voff %016llx\n", i,
+                (unsigned long long)crc64_calc(r->voff, (size_t)(r->n + 1) * 8, 0));
+        fprintf(mf, "crc run_%d.sk %016llx\n", i,
+                (unsigned long long)crc64_calc(r->sk, (size_t)r->n * 8, 0));
+    }
     fclose(mf);
     for (int i = 0; i < e->n_runs; i++) {
         const Run *r = &e->runs[i];
@@ -2274,28 +2429,101 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
     int n_runs = 0, ver = 0;
     if (fscanf(mf, "rrdb-checkpoint %d\ndata_version %u\nnext_seq_floor %llu\nn_runs %d\n",
                &ver, &dv, &floor_, &n_runs) != 4 ||
-        ver != 1) {
+        ver != 1 || n_runs < 0 || n_runs > 65536) {
         fclose(mf);
         return RRDB_CORRUPTION;
     }
+    /* run counts + optional crc lines (integrity: ADVICE r01 — a truncated
+     * or bit-flipped checkpoint returns kCorruption, like the reference's
+     * checksummed rocksdb checkpoints) */
+    uint64_t *mrows = (uint64_t *)malloc((size_t)(n_runs ? n_runs : 1) * 8);
+    char (*crc_names)[64] = NULL;
+    uint64_t *crc_vals = NULL;
+    int n_crc = 0, crc_cap = 0;
+    {
+        int i;
+        char line[256];
+        for (i = 0; i < n_runs; i++)
+            mrows[i] = ~0ull;
+        while (fgets(line, sizeof(line), mf)) {
+            unsigned long long a, b;
+            char name[64];
+            if (sscanf(line, "run %llu %llu", &a, &b) == 2) {
+                if (a < (unsigned long long)n_runs)
+                    mrows[a] = b;
+            } else if (sscanf(line, "crc %63s %llx", name, &b) == 2) {
+                if (n_crc == crc_cap) {
+                    crc_cap = crc_cap ? crc_cap * 2 : 16;
+                    crc_names = (char (*)[64])realloc(crc_names, (size_t)crc_cap * 64);
+                    crc_vals = (uint64_t *)realloc(crc_vals, (size_t)crc_cap * 8);
+                }
+                memcpy(crc_names[n_crc], name, 64);
+                crc_vals[n_crc++] = b;
+            }
+        }
+    }
     fclose(mf);
+#define ORC_RESTORE_FAIL(code)                                                                     \
+    do {                                                                                           \
+        free(mrows);                                                                               \
+        free(crc_names);                                                                           \
+        free(crc_vals);                                                                            \
+        return (code);                                                                             \
+    } while (0)
+    for (int i = 0; i < n_runs; i++)
+        if (mrows[i] == ~0ull)
+            ORC_RESTORE_FAIL(RRDB_CORRUPTION);
     e->data_version = dv;
     for (int i = 0; i < n_runs; i++) {
         Run r;
-        uint64_t nb;
+        uint64_t nb, nb_keys, nb_koff, nb_vals, nb_voff, nb_sk;
         snprintf(fp, sizeof(fp), "%s/run_%d.keys", path, i);
-        r.keys = read_blob_file(fp, &nb);
+        r.keys = read_blob_file(fp, &nb_keys);
         snprintf(fp, sizeof(fp), "%s/run_%d.koff", path, i);
-        r.koff = (uint64_t *)read_blob_file(fp, &nb);
-        r.n = nb / 8 - 1;
+        r.koff = (uint64_t *)read_blob_file(fp, &nb_koff);
         snprintf(fp, sizeof(fp), "%s/run_%d.vals", path, i);
-        r.vals = read_blob_file(fp, &nb);
+        r.vals = read_blob_file(fp, &nb_vals);
         snprintf(fp, sizeof(fp), "%s/run_%d.voff", path, i);
-        r.voff = (uint64_t *)read_blob_file(fp, &nb);
+        r.voff = (uint64_t *)read_blob_file(fp, &nb_voff);
         snprintf(fp, sizeof(fp), "%s/run_%d.sk", path, i);
-        r.sk = (uint64_t *)read_blob_file(fp, &nb);
+        r.sk = (uint64_t *)read_blob_file(fp, &nb_sk);
+        nb = nb_koff;
         if (!r.keys || !r.koff || !r.vals || !r.voff || !r.sk)
-            return RRDB_IO_ERROR;
+            ORC_RESTORE_FAIL(RRDB_IO_ERROR);
+        r.n = mrows[i];
+        if (nb_koff != (r.n + 1) * 8 || nb_voff != (r.n + 1) * 8 || nb_sk != r.n * 8 ||
+            r.koff[0] != 0 || r.voff[0] != 0 || r.koff[r.n] != nb_keys ||
+            r.voff[r.n] != nb_vals)
+            ORC_RESTORE_FAIL(RRDB_CORRUPTION);
+        for (uint64_t j = 0; j < r.n; j++)
+            if (r.koff[j + 1] <= r.koff[j] || r.voff[j + 1] < r.voff[j])
+                ORC_RESTORE_FAIL(RRDB_CORRUPTION);
+        for (uint64_t j = 0; j + 1 < r.n; j++)
+            if (key_cmp(r.keys + r.koff[j], r.koff[j + 1] - r.koff[j], r.keys + r.koff[j + 1],
+                        r.koff[j + 2] - r.koff[j + 1]) >= 0)
+                ORC_RESTORE_FAIL(RRDB_CORRUPTION); /* strictly increasing */
+        for (int f = 0; f < 5; f++) {
+            static const char *exts[5] = {"keys", "koff", "vals", "voff", "sk"};
+            const void *bufs[5];
+            uint64_t lens[5];
+            bufs[0] = r.keys;
+            lens[0] = nb_keys;
+            bufs[1] = r.koff;
+            lens[1] = nb_koff;
+            bufs[2] = r.vals;
+            lens[2] = nb_vals;
+            bufs[3] = r.voff;
+            lens[3] = nb_voff;
+            bufs[4] = r.sk;
+            lens[4] = nb_sk;
+            char want[64];
+            snprintf(want, sizeof(want), "run_%d.%s", i, exts[f]);
+            for (int c = 0; c < n_crc; c++)
+                if (strcmp(crc_names[c], want) == 0 &&
+                    crc_vals[c] != crc64_calc(bufs[f], (size_t)lens[f], 0))
+                    ORC_RESTORE_FAIL(RRDB_CORRUPTION);
+        }
+        (void)nb;
         r.min_seq = ~0ull;
         r.max_seq = 0;
         for (uint64_t j = 0; j < r.n; j++) {
@@ -2311,6 +2539,10 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         }
         e->runs[e->n_runs++] = r;
     }
+    free(mrows);
+    free(crc_names);
+    free(crc_vals);
+#undef ORC_RESTORE_FAIL
     e->next_seq_floor = floor_;
     return RRDB_OK;
 }
