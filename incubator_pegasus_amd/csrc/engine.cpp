@@ -74,6 +74,13 @@ void launch_first_eq(const DevRun *, const uint64_t *, uint64_t, const uint8_t *
 void launch_rank_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
                       const uint64_t *, uint64_t, uint64_t *, uint8_t *, const uint64_t *,
                       const uint64_t *, int, hipStream_t);
+void launch_anchor_rows(const DevRun *, int, int, const uint64_t *, const uint64_t *, int,
+                        uint64_t, uint64_t *, hipStream_t);
+void launch_rank_grp_compact(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
+                             const CompactParams &, uint64_t *, uint64_t *, uint8_t *,
+                             uint32_t *, uint64_t *, uint64_t *, CompactStatsDev *, hipStream_t);
+void launch_rank_grp_view(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
+                          uint64_t *, uint8_t *, hipStream_t);
 void launch_rank_compact_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
                               const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                               uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
@@ -601,23 +608,26 @@ struct HipEngine {
     int emit_mode = 2; /* 2 = chunked (default; 2.8 TB/s on the copy probe),
                           0 = rank-major waves, 1 = input-major waves
                           (env "engine.emit_mode": chunked|rank|input) */
-    int rank_mode = 3; /* 3 = LDS-staged tail-word rank (default; falls back
-                          to mode 0 per compact when runs are not word-probe
-                          eligible with one shared cross-run prefix),
+    int rank_mode = 4; /* 4 = group-streaming rank (default; anchor-key
+                          groups partition the runs disjointly, each staged
+                          through LDS exactly once; falls back to mode 0 per
+                          compact when runs are not word-probe eligible),
+                          3 = r01 windowed LDS-staged tail-word rank,
                           0 = global searches + bound-table narrowing,
                           1 = LDS-staged full-key block rank
-                          (env "engine.rank_mode": ldst|global|lds) */
+                          (env "engine.rank_mode": grp|ldst|global|lds) */
     int bt_shift = 5;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
-    /* LDS tail-word rank eligibility: every run in single-word probe mode
-     * with one shared stride and one shared cross-run first-(fk-8)-byte
-     * prefix, so tail-word compares decide cross-run order exactly */
-    bool ldst_eligible()
+    /* tail-word rank eligibility (shared by the grp and ldst modes): every
+     * run in single-word probe mode with one shared stride and one shared
+     * cross-run first-(fk-8)-byte prefix, so tail-word compares decide
+     * cross-run order exactly */
+    bool word_eligible()
     {
-        if (rank_mode != 3 || runs.size() < 2)
+        if (runs.size() < 2)
             return false;
         if (ldst_elig_cache >= 0)
             return ldst_elig_cache != 0;
@@ -640,6 +650,11 @@ struct HipEngine {
         }
         ldst_elig_cache = 1;
         return true;
+    }
+    bool ldst_eligible() { return rank_mode == 3 && word_eligible(); }
+    bool grp_eligible()
+    {
+        return rank_mode == 4 && runs.size() <= LDST_MAXR && word_eligible();
     }
 
     void free_run(RunBuf &r)
@@ -919,6 +934,31 @@ struct HipEngine {
 
     /* build the visible view for [start, stop_excl); returns device array
      * (caller frees) + count */
+    /* group-streaming rank: pick the anchor run (largest window) and stride,
+     * build the anchor table (arena memory).  Returns n_groups. */
+    uint64_t build_anchors(DevRun *dr, int R, const std::vector<uint64_t> &lo,
+                           const std::vector<uint64_t> &hi, const uint64_t *d_lo,
+                           const uint64_t *d_hi, uint64_t **out_anch)
+    {
+        int q0 = 0;
+        uint64_t wmax = 0;
+        for (int r = 0; r < R; r++)
+            if (hi[r] - lo[r] > wmax) {
+                wmax = hi[r] - lo[r];
+                q0 = r;
+            }
+        int gs = 4;
+        while ((1ull << (gs + 1)) * (uint64_t)R <= GRP_TARGET && gs < 8)
+            gs++;
+        uint64_t n_groups = (wmax + (1ull << gs) - 1) >> gs;
+        if (n_groups == 0)
+            n_groups = 1;
+        uint64_t *d_anch = talloc<uint64_t>((n_groups + 1) * R * 8);
+        launch_anchor_rows(dr, R, q0, d_lo, d_hi, gs, n_groups, d_anch, stream);
+        *out_anch = d_anch;
+        return n_groups;
+    }
+
     void build_view(const std::string *start, const std::string *stop_excl, uint64_t **out_view,
                     uint64_t *out_n)
     {
@@ -962,15 +1002,21 @@ struct HipEngine {
         uint8_t *d_shadow = talloc<uint8_t>(total);
         uint64_t *d_flags = talloc<uint64_t>(total * 8);
         uint64_t *d_pos = talloc<uint64_t>(total * 8);
-        uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
-        if (R > 1 && total > 100000)
-            build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
-        if (ldst_eligible())
-            launch_rank_ldst(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
-                             bt_shift, stream);
-        else
-            launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
-                        bt_shift, stream);
+        if (R > 1 && grp_eligible()) {
+            uint64_t *d_anch = nullptr;
+            uint64_t n_groups = build_anchors(dr, R, lo, hi, d_lo, d_hi, &d_anch);
+            launch_rank_grp_view(dr, R, d_lo, d_anch, n_groups, d_order, d_shadow, stream);
+        } else {
+            uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
+            if (R > 1 && total > 100000)
+                build_bound_table(dr, R, lo, hi, d_lo, d_hi, &d_bt_off, &d_bt);
+            if (ldst_eligible())
+                launch_rank_ldst(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off,
+                                 d_bt, bt_shift, stream);
+            else
+                launch_rank(dr, R, d_lo, d_hi, d_wp, total, d_order, d_shadow, d_bt_off, d_bt,
+                            bt_shift, stream);
+        }
         launch_visible(dr, d_order, d_shadow, total, d_flags, stream);
         launch_psum(d_flags, d_pos, total, psum_scratch(total), stream);
         uint64_t lastp = 0, lastf = 0;
@@ -1136,7 +1182,8 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
         } else if (k == "rocksdb.filter_type") {
             e->bloom_enabled = (v != "none"); /* common/prefix -> full-key bloom */
         } else if (k == "engine.rank_mode") {
-            e->rank_mode = (v == "lds") ? 1 : (v == "ldst" ? 3 : (v == "global" ? 0 : 3));
+            e->rank_mode =
+                (v == "lds") ? 1 : (v == "ldst" ? 3 : (v == "global" ? 0 : 4));
         } else if (k == "engine.bt_shift") {
             int s_ = atoi(v.c_str());
             if (s_ >= 4 && s_ <= 16)
@@ -2178,6 +2225,14 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
         launch_rank_compact_lds(dr, R, d_lo, d_hi, d_blkp, nb, cp, d_bt8_off, d_bt8, d_order,
                                 d_keepw, d_changed, d_new_expire, d_ksz, d_vsz, d_stats,
                                 e->stream);
+        HIP_OK(hipEventRecord(ev[1], e->stream));
+    } else if (R > 1 && e->emit_mode != 1 && e->grp_eligible()) {
+        /* group-streaming rank: no bound table; anchors partition the runs */
+        uint64_t *d_anch = nullptr;
+        uint64_t n_groups = e->build_anchors(dr, R, lo, hi, d_lo, d_hi, &d_anch);
+        HIP_OK(hipEventRecord(ev[0], e->stream));
+        launch_rank_grp_compact(dr, R, d_lo, d_anch, n_groups, cp, d_order, d_keepw, d_changed,
+                                d_new_expire, d_ksz, d_vsz, d_stats, e->stream);
         HIP_OK(hipEventRecord(ev[1], e->stream));
     } else {
     uint64_t *d_bt_off = nullptr, *d_bt = nullptr;

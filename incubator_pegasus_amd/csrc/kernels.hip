@@ -26,7 +26,6 @@
 /* LDS tail-word staged rank: staged window capacity (u64 words) and the
  * max run count the staging handles (beyond either -> in-kernel fallback) */
 #define LDST_CAP 3072
-#define LDST_MAXR 16
 #define MAX_GRID 2048
 #define PSUM_ITEMS_PER_THREAD 16
 #define PSUM_BLOCK_ITEMS (BLOCK * PSUM_ITEMS_PER_THREAD) /* 4096 */
@@ -1344,6 +1343,292 @@ __global__ void __launch_bounds__(BLOCK) k_rank_compact_ldst(
     b = __ballot(disp == D_KEEP);
     if (lane == 0 && b)
         atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+}
+
+/* ================= group-streaming rank (rank_mode=4, default) =============
+ * Round-2 replacement for the windowed-staging ldst rank.  The windowed
+ * design staged every run's search windows once per 256-record span of EVERY
+ * OTHER run, fetching each tail word ~R times (PMC r01: ~196 B/record
+ * against a ~30 B algorithmic budget).  Here the merged output is cut into
+ * GROUPS at anchor keys (every (1<<gs)-th key of one anchor run, full
+ * comparator: key asc, run desc), which partitions every run's window into
+ * DISJOINT segments: group g owns [anch[g][q], anch[g+1][q]) of run q.  One
+ * workgroup streams its group's segments through LDS exactly once — global
+ * traffic becomes one 8B tail read per record — and every element's exact
+ * global rank is  base(g) + local rank,  base(g) = sum_q (anch[g][q]-lo[q]).
+ * Local ranks come from in-LDS binary searches of the sibling segments
+ * (expected ~R*log2(group/R) LDS probes, wave-coherent).  Output writes land
+ * in the contiguous rank range [base, base+gsize): coalesced, unlike the
+ * old full-range scatter.
+ * Shadow (newest-wins) stays exact across group splits of an equal-key set:
+ * a newer same-key version is either inside the group (found by the search)
+ * or the LAST element of the previous group's segment (s_btail below).
+ * Oversized groups (duplicate-heavy data) fall back per element to global
+ * tail searches inside the same segment windows — same math, no staging. */
+
+/* anchor table: rows g=0..n_groups; row g holds every run's boundary for
+ * group g (row 0 = lo, row n_groups = hi).  Anchor key = run q0's record
+ * lo[q0] + (g<<gs); searched in the other runs' packed tail words (word-
+ * probe eligibility makes tail order == key order). */
+__global__ void k_anchor_rows(const DevRun *runs, int R, int q0, const uint64_t *lo,
+                              const uint64_t *hi, int gs, uint64_t n_groups, uint64_t *anch)
+{
+    uint64_t total = (n_groups + 1) * (uint64_t)R;
+    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < total;
+         t += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t g = t / R;
+        int q = (int)(t % R);
+        uint64_t *cell = anch + t;
+        if (g == 0) {
+            *cell = lo[q];
+            continue;
+        }
+        if (g == n_groups) {
+            *cell = hi[q];
+            continue;
+        }
+        uint64_t i = lo[q0] + (g << gs);
+        if (i >= hi[q0]) {
+            *cell = hi[q];
+            continue;
+        }
+        if (q == q0) {
+            *cell = i;
+            continue;
+        }
+        uint64_t key_t = runs[q0].tails[i];
+        const uint64_t *tq = runs[q].tails;
+        uint64_t l = lo[q], h = hi[q];
+        if (q > q0) { /* equal keys in newer runs rank BEFORE the anchor */
+            while (l < h) {
+                uint64_t mid = (l + h) >> 1;
+                if (tq[mid] <= key_t)
+                    l = mid + 1;
+                else
+                    h = mid;
+            }
+        } else {
+            while (l < h) {
+                uint64_t mid = (l + h) >> 1;
+                if (tq[mid] < key_t)
+                    l = mid + 1;
+                else
+                    h = mid;
+            }
+        }
+        *cell = l;
+    }
+}
+
+void launch_anchor_rows(const DevRun *runs, int R, int q0, const uint64_t *d_lo,
+                        const uint64_t *d_hi, int gs, uint64_t n_groups, uint64_t *d_anch,
+                        hipStream_t s)
+{
+    uint64_t total = (n_groups + 1) * (uint64_t)R;
+    k_anchor_rows<<<grid_for(total, BLOCK), BLOCK, 0, s>>>(runs, R, q0, d_lo, d_hi, gs,
+                                                           n_groups, d_anch);
+}
+
+/* VIEW=0: compaction (fused KeyWithTTLCompactionFilter disposition outputs);
+ * VIEW=1: scan view build (order + shadowed only) */
+template <int VIEW>
+__global__ void __launch_bounds__(BLOCK) k_rank_grp(
+    const DevRun *runs, int R, const uint64_t *lo, const uint64_t *anch, uint64_t n_groups,
+    CompactParams cp, uint64_t *order, uint64_t *keepw, uint8_t *changed, uint32_t *new_expire,
+    uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of, uint8_t *shadowed, CompactStatsDev *stats)
+{
+    __shared__ uint64_t s_tails[GRP_CAP];
+    __shared__ uint64_t s_a0[LDST_MAXR], s_seglen[LDST_MAXR], s_segoff[LDST_MAXR + 1];
+    __shared__ uint64_t s_btail[LDST_MAXR];
+    __shared__ uint64_t s_base, s_gsize;
+    __shared__ uint32_t s_bmask;
+
+    int lane = threadIdx.x % WAVE;
+    for (uint64_t g = blockIdx.x; g < n_groups; g += gridDim.x) {
+        if (threadIdx.x < (unsigned)R) {
+            int q = (int)threadIdx.x;
+            uint64_t a0 = anch[g * R + q], a1 = anch[(g + 1) * R + q];
+            s_a0[q] = a0;
+            s_seglen[q] = a1 - a0;
+            s_btail[q] = (a0 > lo[q]) ? runs[q].tails[a0 - 1] : 0;
+            if (q == 0)
+                s_bmask = 0;
+        }
+        __syncthreads();
+        if (threadIdx.x < (unsigned)R) {
+            int q = (int)threadIdx.x;
+            if (s_a0[q] > lo[q])
+                atomicOr(&s_bmask, 1u << q);
+        }
+        if (threadIdx.x == 0) {
+            uint64_t t = 0, b = 0;
+            for (int q = 0; q < R; q++) {
+                s_segoff[q] = t;
+                t += s_seglen[q];
+                b += s_a0[q] - lo[q];
+            }
+            s_segoff[R] = t;
+            s_gsize = t;
+            s_base = b;
+        }
+        __syncthreads();
+        uint64_t gsize = s_gsize;
+        bool staged = gsize <= GRP_CAP;
+        if (staged) { /* stream each segment through LDS exactly once */
+            for (int q = 0; q < R; q++) {
+                uint64_t len = s_seglen[q], off = s_segoff[q], a0 = s_a0[q];
+                const uint64_t *tq = runs[q].tails;
+                for (uint64_t j = threadIdx.x; j < len; j += blockDim.x)
+                    s_tails[off + j] = tq[a0 + j];
+            }
+        }
+        __syncthreads();
+        uint64_t iters = (gsize + blockDim.x - 1) / blockDim.x;
+        for (uint64_t it = 0; it < iters; it++) {
+            uint64_t e = it * blockDim.x + threadIdx.x;
+            int disp = D_NONE;
+            if (e < gsize) {
+                int q = 0;
+                while (s_segoff[q + 1] <= e)
+                    q++;
+                uint64_t segpos = e - s_segoff[q];
+                uint64_t i = s_a0[q] + segpos;
+                uint64_t myt = staged ? s_tails[e] : runs[q].tails[i];
+                uint64_t lrank = segpos;
+                int shadow = 0;
+                uint32_t bm = s_bmask;
+                for (int p = 0; p < R; p++) {
+                    if (p == q)
+                        continue;
+                    uint64_t plen = s_seglen[p];
+                    uint64_t b;
+                    if (staged) {
+                        uint64_t off = s_segoff[p];
+                        uint64_t l = 0, h = plen;
+                        if (p > q) {
+                            while (l < h) {
+                                uint64_t mid = (l + h) >> 1;
+                                if (s_tails[off + mid] <= myt)
+                                    l = mid + 1;
+                                else
+                                    h = mid;
+                            }
+                        } else {
+                            while (l < h) {
+                                uint64_t mid = (l + h) >> 1;
+                                if (s_tails[off + mid] < myt)
+                                    l = mid + 1;
+                                else
+                                    h = mid;
+                            }
+                        }
+                        b = l;
+                        if (p > q && !shadow && b > 0 && s_tails[off + b - 1] == myt)
+                            shadow = 1;
+                    } else {
+                        const uint64_t *tp = runs[p].tails;
+                        uint64_t a0 = s_a0[p];
+                        uint64_t l = a0, h = a0 + plen;
+                        if (p > q) {
+                            while (l < h) {
+                                uint64_t mid = (l + h) >> 1;
+                                if (tp[mid] <= myt)
+                                    l = mid + 1;
+                                else
+                                    h = mid;
+                            }
+                        } else {
+                            while (l < h) {
+                                uint64_t mid = (l + h) >> 1;
+                                if (tp[mid] < myt)
+                                    l = mid + 1;
+                                else
+                                    h = mid;
+                            }
+                        }
+                        b = l - a0;
+                        if (p > q && !shadow && b > 0 && tp[a0 + b - 1] == myt)
+                            shadow = 1;
+                    }
+                    lrank += b;
+                    /* split equal-key set: newer version ended the previous
+                     * group's segment of run p */
+                    if (p > q && !shadow && ((bm >> p) & 1) && s_btail[p] == myt)
+                        shadow = 1;
+                }
+                uint64_t rank = s_base + lrank;
+                order[rank] = ((uint64_t)q << 40) | i;
+                if (VIEW) {
+                    shadowed[rank] = (uint8_t)shadow;
+                } else {
+                    uint8_t ch;
+                    uint32_t nts;
+                    uint64_t okl, ovl;
+                    disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
+                    keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+                    changed[rank] = ch;
+                    new_expire[rank] = nts;
+                    ksz[rank] = okl;
+                    vsz[rank] = ovl;
+                    if (rank_of) {
+                        /* input-major index: wprefix equivalent is segment
+                         * order per group — not supported; rank_of callers
+                         * use the non-group path (host guards) */
+                    }
+                }
+            }
+            if (!VIEW) {
+                unsigned long long b;
+                b = __ballot(disp == D_SHADOWED);
+                if (lane == 0 && b)
+                    atomicAdd(&stats->shadowed, (unsigned long long)__popcll(b));
+                b = __ballot(disp == D_TOMBSTONE);
+                if (lane == 0 && b)
+                    atomicAdd(&stats->tombstones, (unsigned long long)__popcll(b));
+                b = __ballot(disp == D_EXPIRED);
+                if (lane == 0 && b)
+                    atomicAdd(&stats->expired, (unsigned long long)__popcll(b));
+                b = __ballot(disp == D_FILTERED);
+                if (lane == 0 && b)
+                    atomicAdd(&stats->filtered, (unsigned long long)__popcll(b));
+                b = __ballot(disp == D_KEEP);
+                if (lane == 0 && b)
+                    atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+            }
+        }
+        __syncthreads(); /* LDS reused by the next group */
+    }
+}
+
+void launch_rank_grp_compact(const DevRun *d_runs, int R, const uint64_t *d_lo,
+                             const uint64_t *d_anch, uint64_t n_groups, const CompactParams &cp,
+                             uint64_t *d_order, uint64_t *d_keepw, uint8_t *d_changed,
+                             uint32_t *d_new_expire, uint64_t *d_ksz, uint64_t *d_vsz,
+                             CompactStatsDev *d_stats, hipStream_t s)
+{
+    uint64_t blocks = n_groups;
+    if (blocks == 0)
+        blocks = 1;
+    if (blocks > 65535)
+        blocks = 65535;
+    k_rank_grp<0><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+        d_runs, R, d_lo, d_anch, n_groups, cp, d_order, d_keepw, d_changed, d_new_expire, d_ksz,
+        d_vsz, nullptr, nullptr, d_stats);
+}
+
+void launch_rank_grp_view(const DevRun *d_runs, int R, const uint64_t *d_lo,
+                          const uint64_t *d_anch, uint64_t n_groups, uint64_t *d_order,
+                          uint8_t *d_shadow, hipStream_t s)
+{
+    uint64_t blocks = n_groups;
+    if (blocks == 0)
+        blocks = 1;
+    if (blocks > 65535)
+        blocks = 65535;
+    CompactParams cp{};
+    k_rank_grp<1><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+        d_runs, R, d_lo, d_anch, n_groups, cp, d_order, nullptr, nullptr, nullptr, nullptr,
+        nullptr, nullptr, d_shadow, nullptr);
 }
 
 void launch_rank_compact_ldst(const DevRun *d_runs, int R, const uint64_t *d_lo,
