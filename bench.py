@@ -105,7 +105,7 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
     ap.add_argument("--bt-shift", type=int, default=5)
-    ap.add_argument("--rank-mode", choices=["global", "lds", "ldst"], default="ldst")
+    ap.add_argument("--rank-mode", choices=["global", "lds", "ldst", "grp"], default="grp")
     ap.add_argument("--rules-profile", action="store_true",
                     help="configs[4] shape: TTL'd data + default_ttl + user "
                          "delete/update-TTL compaction rules evaluated per key")
