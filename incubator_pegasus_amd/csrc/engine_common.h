@@ -21,7 +21,7 @@
 /* tail-word rank modes: max run count; group-streaming rank group budget */
 #define LDST_MAXR 16
 #define GRP_CAP 1024
-#define GRP_TARGET 256
+#define GRP_TARGET 512
 
 /* device-visible descriptor of one sorted run */
 struct DevRun {

@@ -1429,19 +1429,53 @@ void launch_anchor_rows(const DevRun *runs, int R, int q0, const uint64_t *d_lo,
                                                            n_groups, d_anch);
 }
 
+/* (tail asc, run desc) strict order on (tail, org) pairs — tails are unique
+ * within a run, so ties are always cross-run and org's run bits decide */
+__device__ static inline bool grp_lt(uint64_t ta, uint16_t oa, uint64_t tb, uint16_t ob)
+{
+    return ta < tb || (ta == tb && (oa >> 12) > (ob >> 12));
+}
+
+__device__ static inline void grp_stat_ballots(int disp, int lane, CompactStatsDev *stats)
+{
+    unsigned long long b;
+    b = __ballot(disp == D_SHADOWED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->shadowed, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_TOMBSTONE);
+    if (lane == 0 && b)
+        atomicAdd(&stats->tombstones, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_EXPIRED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->expired, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_FILTERED);
+    if (lane == 0 && b)
+        atomicAdd(&stats->filtered, (unsigned long long)__popcll(b));
+    b = __ballot(disp == D_KEEP);
+    if (lane == 0 && b)
+        atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+}
+
 /* VIEW=0: compaction (fused KeyWithTTLCompactionFilter disposition outputs);
- * VIEW=1: scan view build (order + shadowed only) */
+ * VIEW=1: scan view build (order + shadowed only).
+ * Staged groups rank by PAIRWISE MERGE-PATH ROUNDS in LDS (log2(R) rounds;
+ * each round every element moves once, each thread owns a fixed output-slot
+ * range found by one diagonal search) — ~5-10x fewer instructions per
+ * element than per-element binary searches, which measured issue-bound. */
 template <int VIEW>
 __global__ void __launch_bounds__(BLOCK) k_rank_grp(
     const DevRun *runs, int R, const uint64_t *lo, const uint64_t *anch, uint64_t n_groups,
     CompactParams cp, uint64_t *order, uint64_t *keepw, uint8_t *changed, uint32_t *new_expire,
     uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of, uint8_t *shadowed, CompactStatsDev *stats)
 {
-    __shared__ uint64_t s_tails[GRP_CAP];
+    __shared__ uint64_t s_ta[GRP_CAP], s_tb[GRP_CAP]; /* tails ping-pong */
+    __shared__ uint16_t s_oa[GRP_CAP], s_ob[GRP_CAP]; /* (q<<12)|segpos */
     __shared__ uint64_t s_a0[LDST_MAXR], s_seglen[LDST_MAXR], s_segoff[LDST_MAXR + 1];
     __shared__ uint64_t s_btail[LDST_MAXR];
+    __shared__ uint64_t s_loff[LDST_MAXR + 1]; /* current merge-list offsets */
     __shared__ uint64_t s_base, s_gsize;
     __shared__ uint32_t s_bmask;
+    __shared__ int s_nl;
 
     int lane = threadIdx.x % WAVE;
     for (uint64_t g = blockIdx.x; g < n_groups; g += gridDim.x) {
@@ -1464,71 +1498,154 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
             uint64_t t = 0, b = 0;
             for (int q = 0; q < R; q++) {
                 s_segoff[q] = t;
+                s_loff[q] = t;
                 t += s_seglen[q];
                 b += s_a0[q] - lo[q];
             }
             s_segoff[R] = t;
+            s_loff[R] = t;
             s_gsize = t;
             s_base = b;
+            s_nl = R;
         }
         __syncthreads();
         uint64_t gsize = s_gsize;
         bool staged = gsize <= GRP_CAP;
-        if (staged) { /* stream each segment through LDS exactly once */
+        if (staged) {
+            /* stream each segment through LDS exactly once, org alongside */
             for (int q = 0; q < R; q++) {
                 uint64_t len = s_seglen[q], off = s_segoff[q], a0 = s_a0[q];
                 const uint64_t *tq = runs[q].tails;
-                for (uint64_t j = threadIdx.x; j < len; j += blockDim.x)
-                    s_tails[off + j] = tq[a0 + j];
+                for (uint64_t j = threadIdx.x; j < len; j += blockDim.x) {
+                    s_ta[off + j] = tq[a0 + j];
+                    s_oa[off + j] = (uint16_t)((q << 12) | (uint32_t)j);
+                }
             }
-        }
-        __syncthreads();
-        uint64_t iters = (gsize + blockDim.x - 1) / blockDim.x;
-        for (uint64_t it = 0; it < iters; it++) {
-            uint64_t e = it * blockDim.x + threadIdx.x;
-            int disp = D_NONE;
-            if (e < gsize) {
-                int q = 0;
-                while (s_segoff[q + 1] <= e)
-                    q++;
-                uint64_t segpos = e - s_segoff[q];
-                uint64_t i = s_a0[q] + segpos;
-                uint64_t myt = staged ? s_tails[e] : runs[q].tails[i];
-                uint64_t lrank = segpos;
-                int shadow = 0;
-                uint32_t bm = s_bmask;
-                for (int p = 0; p < R; p++) {
-                    if (p == q)
-                        continue;
-                    uint64_t plen = s_seglen[p];
-                    uint64_t b;
-                    if (staged) {
-                        uint64_t off = s_segoff[p];
-                        uint64_t l = 0, h = plen;
-                        if (p > q) {
-                            while (l < h) {
-                                uint64_t mid = (l + h) >> 1;
-                                if (s_tails[off + mid] <= myt)
-                                    l = mid + 1;
-                                else
-                                    h = mid;
-                            }
+            __syncthreads();
+            /* pairwise merge rounds; ping-pong a->b->a->... */
+            uint64_t E = (gsize + blockDim.x - 1) / blockDim.x;
+            int cur = 0;
+            while (s_nl > 1) {
+                const uint64_t *st = cur ? s_tb : s_ta;
+                const uint16_t *so = cur ? s_ob : s_oa;
+                uint64_t *dt = cur ? s_ta : s_tb;
+                uint16_t *do_ = cur ? s_oa : s_ob;
+                int nl = s_nl;
+                uint64_t slot = (uint64_t)threadIdx.x * E;
+                uint64_t send = slot + E;
+                if (send > gsize)
+                    send = gsize;
+                while (slot < send) {
+                    /* pair j covering this slot */
+                    int j = 0;
+                    while (2 * j + 2 < nl + 1 && s_loff[2 * j + 2] <= slot)
+                        j++;
+                    uint64_t ka = s_loff[2 * j];
+                    uint64_t kb = (2 * j + 1 <= nl) ? s_loff[2 * j + 1] : gsize;
+                    uint64_t ke = (2 * j + 2 <= nl) ? s_loff[2 * j + 2] : gsize;
+                    uint64_t la = kb - ka, lb = ke - kb;
+                    uint64_t k = slot - ka;
+                    /* diagonal: i = #A elements among the first k of the pair */
+                    uint64_t l = k > lb ? k - lb : 0, h = k < la ? k : la;
+                    while (l < h) {
+                        uint64_t mid = (l + h) >> 1;
+                        if (grp_lt(st[ka + mid], so[ka + mid], st[kb + k - mid - 1],
+                                   so[kb + k - mid - 1]))
+                            l = mid + 1;
+                        else
+                            h = mid;
+                    }
+                    uint64_t ia = ka + l, ib = kb + (k - l);
+                    uint64_t end = send < ke ? send : ke;
+                    for (; slot < end; slot++) {
+                        bool ta = (ib >= ke) ||
+                                  (ia < kb && grp_lt(st[ia], so[ia], st[ib], so[ib]));
+                        if (ta) {
+                            dt[slot] = st[ia];
+                            do_[slot] = so[ia++];
                         } else {
-                            while (l < h) {
-                                uint64_t mid = (l + h) >> 1;
-                                if (s_tails[off + mid] < myt)
-                                    l = mid + 1;
-                                else
-                                    h = mid;
-                            }
+                            dt[slot] = st[ib];
+                            do_[slot] = so[ib++];
                         }
-                        b = l;
-                        if (p > q && !shadow && b > 0 && s_tails[off + b - 1] == myt)
-                            shadow = 1;
+                    }
+                }
+                __syncthreads();
+                if (threadIdx.x == 0) {
+                    int nn = (nl + 1) / 2;
+                    for (int x = 0; x <= nn; x++)
+                        s_loff[x] = (2 * x <= nl) ? s_loff[2 * x] : gsize;
+                    s_nl = nn;
+                }
+                cur ^= 1;
+                __syncthreads();
+            }
+            const uint64_t *ft = cur ? s_tb : s_ta;
+            const uint16_t *fo = cur ? s_ob : s_oa;
+            /* output phase: position p in the merged group == local rank */
+            uint64_t iters = (gsize + blockDim.x - 1) / blockDim.x;
+            for (uint64_t it = 0; it < iters; it++) {
+                uint64_t p = it * blockDim.x + threadIdx.x;
+                int disp = D_NONE;
+                if (p < gsize) {
+                    uint64_t myt = ft[p];
+                    uint16_t org = fo[p];
+                    int q = org >> 12;
+                    uint64_t i = s_a0[q] + (org & 0xFFF);
+                    int shadow = 0;
+                    if (p > 0) {
+                        shadow = (ft[p - 1] == myt);
                     } else {
+                        /* equal-key set split at the group boundary: a newer
+                         * version ended the previous group's segment */
+                        uint32_t bm = s_bmask;
+                        for (int q2 = q + 1; q2 < R; q2++)
+                            if (((bm >> q2) & 1) && s_btail[q2] == myt) {
+                                shadow = 1;
+                                break;
+                            }
+                    }
+                    uint64_t rank = s_base + p;
+                    order[rank] = ((uint64_t)q << 40) | i;
+                    if (VIEW) {
+                        shadowed[rank] = (uint8_t)shadow;
+                    } else {
+                        uint8_t ch;
+                        uint32_t nts;
+                        uint64_t okl, ovl;
+                        disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
+                        keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+                        changed[rank] = ch;
+                        new_expire[rank] = nts;
+                        ksz[rank] = okl;
+                        vsz[rank] = ovl;
+                    }
+                }
+                if (!VIEW)
+                    grp_stat_ballots(disp, lane, stats);
+            }
+        } else {
+            /* oversized group (duplicate-clustered data): per-element global
+             * tail probes inside the same disjoint segment windows */
+            uint64_t iters = (gsize + blockDim.x - 1) / blockDim.x;
+            for (uint64_t it = 0; it < iters; it++) {
+                uint64_t e = it * blockDim.x + threadIdx.x;
+                int disp = D_NONE;
+                if (e < gsize) {
+                    int q = 0;
+                    while (s_segoff[q + 1] <= e)
+                        q++;
+                    uint64_t segpos = e - s_segoff[q];
+                    uint64_t i = s_a0[q] + segpos;
+                    uint64_t myt = runs[q].tails[i];
+                    uint64_t lrank = segpos;
+                    int shadow = 0;
+                    uint32_t bm = s_bmask;
+                    for (int p = 0; p < R; p++) {
+                        if (p == q)
+                            continue;
                         const uint64_t *tp = runs[p].tails;
                         uint64_t a0 = s_a0[p];
-                        uint64_t l = a0, h = a0 + plen;
+                        uint64_t l = a0, h = a0 + s_seglen[p];
                         if (p > q) {
                             while (l < h) {
                                 uint64_t mid = (l + h) >> 1;
@@ -1546,54 +1663,31 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
                                     h = mid;
                             }
                         }
-                        b = l - a0;
+                        uint64_t b = l - a0;
                         if (p > q && !shadow && b > 0 && tp[a0 + b - 1] == myt)
                             shadow = 1;
+                        lrank += b;
+                        if (p > q && !shadow && ((bm >> p) & 1) && s_btail[p] == myt)
+                            shadow = 1;
                     }
-                    lrank += b;
-                    /* split equal-key set: newer version ended the previous
-                     * group's segment of run p */
-                    if (p > q && !shadow && ((bm >> p) & 1) && s_btail[p] == myt)
-                        shadow = 1;
-                }
-                uint64_t rank = s_base + lrank;
-                order[rank] = ((uint64_t)q << 40) | i;
-                if (VIEW) {
-                    shadowed[rank] = (uint8_t)shadow;
-                } else {
-                    uint8_t ch;
-                    uint32_t nts;
-                    uint64_t okl, ovl;
-                    disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
-                    keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-                    changed[rank] = ch;
-                    new_expire[rank] = nts;
-                    ksz[rank] = okl;
-                    vsz[rank] = ovl;
-                    if (rank_of) {
-                        /* input-major index: wprefix equivalent is segment
-                         * order per group — not supported; rank_of callers
-                         * use the non-group path (host guards) */
+                    uint64_t rank = s_base + lrank;
+                    order[rank] = ((uint64_t)q << 40) | i;
+                    if (VIEW) {
+                        shadowed[rank] = (uint8_t)shadow;
+                    } else {
+                        uint8_t ch;
+                        uint32_t nts;
+                        uint64_t okl, ovl;
+                        disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
+                        keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+                        changed[rank] = ch;
+                        new_expire[rank] = nts;
+                        ksz[rank] = okl;
+                        vsz[rank] = ovl;
                     }
                 }
-            }
-            if (!VIEW) {
-                unsigned long long b;
-                b = __ballot(disp == D_SHADOWED);
-                if (lane == 0 && b)
-                    atomicAdd(&stats->shadowed, (unsigned long long)__popcll(b));
-                b = __ballot(disp == D_TOMBSTONE);
-                if (lane == 0 && b)
-                    atomicAdd(&stats->tombstones, (unsigned long long)__popcll(b));
-                b = __ballot(disp == D_EXPIRED);
-                if (lane == 0 && b)
-                    atomicAdd(&stats->expired, (unsigned long long)__popcll(b));
-                b = __ballot(disp == D_FILTERED);
-                if (lane == 0 && b)
-                    atomicAdd(&stats->filtered, (unsigned long long)__popcll(b));
-                b = __ballot(disp == D_KEEP);
-                if (lane == 0 && b)
-                    atomicAdd(&stats->output_records, (unsigned long long)__popcll(b));
+                if (!VIEW)
+                    grp_stat_ballots(disp, lane, stats);
             }
         }
         __syncthreads(); /* LDS reused by the next group */

@@ -162,7 +162,7 @@ def test_grp_rank_rules_and_ttl(oracle_lib, hip_lib):
     ops_json = json.dumps({"ops": [
         {"type": "COT_DELETE", "params": "",
          "rules": [{"type": "FRT_HASHKEY_PATTERN", "params": json.dumps(
-             {"pattern": "00000000000001", "match_type": "SMT_MATCH_PREFIX"})}]},
+             {"pattern": "u:000000000001", "match_type": "SMT_MATCH_PREFIX"})}]},
     ]})
     envs = {"user_specified_compaction": ops_json, "default_ttl": "1000"}
     runs = D.build_point_table_runs(30_000, 4, seed=D.DEFAULT_SEED + 9,
