@@ -62,6 +62,8 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
 void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
 void launch_bloom_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
 void launch_build_tails(const uint8_t *, uint64_t, uint64_t, uint64_t *, hipStream_t);
+void launch_build_meta(const uint8_t *, const uint64_t *, uint64_t, const uint64_t *, uint64_t,
+                       uint32_t, uint64_t *, hipStream_t);
 void launch_rank_compact_lds(const DevRun *, int, const uint64_t *, const uint64_t *,
                              const uint64_t *, uint64_t, const CompactParams &,
                              const uint64_t *, const uint64_t *, uint64_t *, uint64_t *,
@@ -507,6 +509,20 @@ static void set_pfx(uint32_t lcp, uint32_t *pfx_skip, uint32_t *lcp_exact)
 }
 
 /* constant key stride lets device searches skip the offset-pair loads */
+/* constant encoded-value stride (0 = variable); mirrors detect_fixed_klen */
+static uint32_t detect_fixed_vlen(const uint64_t *voff, uint64_t n)
+{
+    if (n == 0)
+        return 0;
+    uint64_t s = voff[1] - voff[0];
+    if (s == 0 || s > UINT32_MAX)
+        return 0;
+    for (uint64_t i = 1; i < n; i++)
+        if (voff[i + 1] - voff[i] != s)
+            return 0;
+    return (uint32_t)s;
+}
+
 static uint32_t detect_fixed_klen(const uint64_t *koff, uint64_t n)
 {
     if (n == 0)
@@ -534,6 +550,8 @@ struct RunBuf {
     uint32_t pfx_skip = 0;   /* lcp(first,last) floored to 8B words, <=16 */
     uint32_t lcp_exact = 0;  /* exact lcp(first,last), capped at 32 */
     uint64_t *tails = nullptr; /* packed BE tail words (word-probe mode) */
+    uint64_t *meta = nullptr;  /* (expire_ts<<32)|kind disposition column */
+    uint32_t fixed_vlen = 0;   /* constant encoded-value stride (0=variable) */
 };
 
 struct HipScanCtx {
@@ -668,6 +686,8 @@ struct HipEngine {
             (void)hipFree(r.bloom);
         if (r.tails)
             (void)hipFree(r.tails);
+        if (r.meta)
+            (void)hipFree(r.meta);
         r = RunBuf();
     }
 
@@ -681,6 +701,18 @@ struct HipEngine {
         HIP_OK(hipStreamSynchronize(stream));
     }
 
+    /* disposition column (expire<<32|kind): one 8B gather replaces the
+     * sk + voff + value-header chain in filter/scan-state evaluation */
+    void build_meta(RunBuf &r)
+    {
+        if (r.n == 0)
+            return;
+        HIP_OK(hipMalloc(&r.meta, r.n * 8));
+        launch_build_meta(r.vals, r.voff, r.fixed_vlen, r.sk, r.n, data_version, r.meta,
+                          stream);
+        HIP_OK(hipStreamSynchronize(stream));
+    }
+
     /* §8(f)3: ~10 bits/key blocked bloom, built once per run on device */
     void build_bloom(RunBuf &r)
     {
@@ -691,7 +723,8 @@ struct HipEngine {
             n_blocks = 1;
         HIP_OK(hipMalloc(&r.bloom, n_blocks * 64));
         HIP_OK(hipMemsetAsync(r.bloom, 0, n_blocks * 64, stream));
-        DevRun dr{r.keys, r.koff, r.vals, r.voff, r.sk, r.n, nullptr, 0};
+        DevRun dr{r.keys, r.koff, r.vals, r.voff, r.sk, r.n, nullptr, 0, 0, 0, 0,
+                  nullptr, nullptr, 0};
         launch_bloom_build(dr, r.bloom, n_blocks, stream);
         HIP_OK(hipStreamSynchronize(stream));
         r.bloom_blocks = n_blocks;
@@ -707,7 +740,7 @@ struct HipEngine {
                 h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
                               runs[i].n,    runs[i].bloom, runs[i].bloom_blocks,
                               runs[i].fixed_klen, runs[i].pfx_skip, runs[i].lcp_exact,
-                              runs[i].tails};
+                              runs[i].tails, runs[i].meta, runs[i].fixed_vlen};
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
@@ -1062,11 +1095,13 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size() * 8);
     r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
     r.fixed_klen = detect_fixed_klen(koff.data(), r.n);
+    r.fixed_vlen = detect_fixed_vlen(voff.data(), r.n);
     set_pfx(lcp_exact32((const uint8_t *)keys.data(), koff[1] - koff[0],
                         (const uint8_t *)keys.data() + koff[r.n - 1],
                         koff[r.n] - koff[r.n - 1]),
             &r.pfx_skip, &r.lcp_exact);
     e->build_tails(r);
+    e->build_meta(r);
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
@@ -1170,7 +1205,20 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
         } else if (k == "manual_compact.disabled") {
             e->manual_compact_disabled = (v == "true");
         } else if (k == "pegasus.data_version") {
-            e->data_version = (uint32_t)atoi(v.c_str());
+            uint32_t nv = (uint32_t)atoi(v.c_str());
+            if (nv != e->data_version) {
+                e->data_version = nv;
+                /* the meta column bakes in the value-header layout */
+                e->activate();
+                for (auto &r : e->runs) {
+                    if (r.meta) {
+                        (void)hipFree(r.meta);
+                        r.meta = nullptr;
+                    }
+                    e->build_meta(r);
+                }
+                e->d_runs_dirty = true;
+            }
         } else if (k == "replica.rocksdb_iteration_threshold_time_ms") {
             e->iter_time_ms = (uint64_t)atoll(v.c_str());
         } else if (k == "rocksdb.max_iteration_count") {
@@ -1226,6 +1274,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     r.voff = (uint64_t *)e->upload_bytes(val_offs, (n + 1) * 8);
     r.sk = (uint64_t *)e->upload_bytes(seq_kind, n * 8);
     r.fixed_klen = detect_fixed_klen(key_offs, n);
+    r.fixed_vlen = detect_fixed_vlen(val_offs, n);
     set_pfx(lcp_exact32(keys + key_offs[0], key_offs[1] - key_offs[0],
                         keys + key_offs[n - 1], key_offs[n] - key_offs[n - 1]),
             &r.pfx_skip, &r.lcp_exact);
@@ -1234,6 +1283,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
         mx = std::max(mx, seq_kind[i] >> 1);
     e->next_seq_floor = mx + 1;
     e->build_tails(r);
+    e->build_meta(r);
     e->build_bloom(r);
     e->runs.push_back(r);
     e->d_runs_dirty = true;
@@ -2341,6 +2391,12 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
         for (auto &ir : e->runs)
             if (ir.fixed_klen != nr.fixed_klen)
                 nr.fixed_klen = 0;
+        /* output values are verbatim-length copies of input values, so a
+         * stride shared by every input run is preserved too */
+        nr.fixed_vlen = e->runs.empty() ? 0 : e->runs[0].fixed_vlen;
+        for (auto &ir : e->runs)
+            if (ir.fixed_vlen != nr.fixed_vlen)
+                nr.fixed_vlen = 0;
         if (keep_inputs) {
             /* output is dropped at the end of the pass: pooled temporaries */
             nr.keys = e->talloc<uint8_t>(kbytes);
@@ -2416,6 +2472,7 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
             HIP_OK(hipMemcpy(t32, nr.keys + lo2[0], ll ? ll : 1, hipMemcpyDeviceToHost));
             set_pfx(lcp_exact32(h32, fl, t32, ll), &nr.pfx_skip, &nr.lcp_exact);
             e->build_tails(nr);
+            e->build_meta(nr);
             e->build_bloom(nr);
             e->runs.push_back(nr);
         }
@@ -2706,6 +2763,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size());
         r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size());
         r.fixed_klen = detect_fixed_klen((const uint64_t *)koff.data(), r.n);
+        r.fixed_vlen = detect_fixed_vlen((const uint64_t *)voff.data(), r.n);
         if (r.n) {
             const uint64_t *ko = (const uint64_t *)koff.data();
             set_pfx(lcp_exact32(keys.data() + ko[0], ko[1] - ko[0],
@@ -2713,6 +2771,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
                     &r.pfx_skip, &r.lcp_exact);
         }
         e->build_tails(r); /* rebuilt, not serialized (like blooms) */
+        e->build_meta(r);
         e->build_bloom(r);
         e->runs.push_back(r);
     }

@@ -51,6 +51,15 @@ struct DevRun {
      * creation when the single-word probe mode is eligible: 8B-strided probe
      * loads instead of klen-strided ones (null when ineligible) */
     const uint64_t *tails;
+    /* per-record disposition column, built at run creation:
+     * (expire_ts << 32) | kind.  Compaction-filter / scan-state evaluation
+     * reads ONE coalescable 8B word instead of the sk word + voff pair + the
+     * dependent 12B value-header parse (that gather chain dominated the
+     * fused rank kernel).  Null only for transient descriptors that never
+     * reach disposition (e.g. bloom builds). */
+    const uint64_t *meta;
+    /* all values in the run share this encoded length (0 = variable) */
+    uint32_t fixed_vlen;
 };
 
 /* flattened user compaction rules/ops (device-resident)

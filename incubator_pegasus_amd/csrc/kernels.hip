@@ -287,6 +287,29 @@ __device__ static inline int dev_ts_expired(uint32_t now, uint32_t ts)
     return ts > 0 && ts <= now;
 }
 
+/* per-record disposition column: (expire_ts << 32) | kind.  Built once at
+ * run creation so filter/scan-state evaluation reads one 8B word instead of
+ * the sk word + voff pair + dependent value-header parse. */
+__global__ void k_build_meta(const uint8_t *vals, const uint64_t *voff, uint64_t fixed_vlen,
+                             const uint64_t *sk, uint64_t n, uint32_t dv, uint64_t *meta)
+{
+    for (uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t kind = sk[i] & 1;
+        uint64_t off = fixed_vlen ? i * fixed_vlen : voff[i];
+        uint64_t vl = fixed_vlen ? fixed_vlen : (voff[i + 1] - voff[i]);
+        uint32_t ts = (!kind && vl >= dev_hdr_len(dv)) ? dev_expire_ts(dv, vals + off) : 0;
+        meta[i] = ((uint64_t)ts << 32) | kind;
+    }
+}
+void launch_build_meta(const uint8_t *vals, const uint64_t *voff, uint64_t fixed_vlen,
+                       const uint64_t *sk, uint64_t n, uint32_t dv, uint64_t *meta,
+                       hipStream_t s)
+{
+    k_build_meta<<<grid_for(n, BLOCK), BLOCK, 0, s>>>(vals, voff, fixed_vlen, sk, n, dv, meta);
+}
+
+
 /* ================= pattern / rules (device) =================
  * compaction_filter_rule.cpp:31-90, compaction_operation.cpp:33-113 */
 __device__ static int dev_mem_eq(const uint8_t *a, const uint8_t *b, uint64_t n)
@@ -335,26 +358,6 @@ __device__ static int dev_validate_filter(int ft, const uint8_t *pat, uint64_t p
     if (ft == 2)
         return dev_mem_eq(v, pat, plen);
     return dev_mem_eq(v + vlen - plen, pat, plen);
-}
-
-__device__ static int dev_rule_match(const DevRule &r, const uint8_t *pats, uint32_t dv,
-                                     uint32_t now, const uint8_t *hk, uint64_t hklen,
-                                     const uint8_t *sk, uint64_t sklen, const uint8_t *val)
-{
-    switch (r.type) {
-    case DFR_HASHKEY:
-        return dev_pattern_match(hk, hklen, r.match_type, pats + r.pat_off, r.pat_len);
-    case DFR_SORTKEY:
-        return dev_pattern_match(sk, sklen, r.match_type, pats + r.pat_off, r.pat_len);
-    case DFR_TTL_RANGE: {
-        uint32_t e = dev_expire_ts(dv, val);
-        if (e == 0 && r.start_ttl == 0 && r.stop_ttl == 0)
-            return 1;
-        return (uint32_t)(r.start_ttl + now) <= e && (uint32_t)(r.stop_ttl + now) >= e;
-    }
-    default:
-        return 0;
-    }
 }
 
 __device__ __host__ static inline uint64_t bloom_hash(const uint8_t *k, uint64_t n);
@@ -752,11 +755,18 @@ __global__ void k_scan_state(const DevRun *runs, const uint64_t *view, uint64_t 
         uint64_t id = view[p];
         const DevRun &r = runs[id >> 40];
         uint64_t i = id & 0xFFFFFFFFFFull;
-        uint64_t kl, vl;
-        const uint8_t *k = run_key(r, i, &kl);
-        const uint8_t *v = run_val(r, i, &vl);
+        uint64_t kl = r.fixed_klen ? r.fixed_klen : (r.koff[i + 1] - r.koff[i]);
+        uint64_t vl = r.fixed_vlen ? r.fixed_vlen : (r.voff[i + 1] - r.voff[i]);
+        const uint8_t *k = r.keys + (r.fixed_klen ? i * r.fixed_klen : r.koff[i]);
+        uint32_t ets;
+        if (r.meta) {
+            ets = (uint32_t)(r.meta[i] >> 32);
+        } else {
+            uint64_t vtmp;
+            ets = dev_expire_ts(sp.data_version, run_val(r, i, &vtmp));
+        }
         int st = ST_NORMAL;
-        if (dev_ts_expired(sp.epoch_now, dev_expire_ts(sp.data_version, v))) {
+        if (dev_ts_expired(sp.epoch_now, ets)) {
             st = ST_EXPIRED;
         } else if (sp.validate_hash &&
                    (sp.partition_version < 0 || sp.pidx > sp.partition_version ||
@@ -998,7 +1008,10 @@ void launch_valid_beyond(const DevRun *runs, int R, const uint8_t *bound, uint64
 enum { D_NONE = 0, D_KEEP, D_SHADOWED, D_TOMBSTONE, D_EXPIRED, D_FILTERED };
 
 /* record-level KeyWithTTLCompactionFilter::Filter restatement
- * (key_ttl_compaction_filter.h:55-92); returns D_*, fills outputs on KEEP */
+ * (key_ttl_compaction_filter.h:55-92); returns D_*, fills outputs on KEEP.
+ * Hot path reads only the 8B meta column ((expire<<32)|kind, built at run
+ * creation) — key bytes are touched only under user rules / hash
+ * validation, value bytes never (no rule inspects value data). */
 __device__ static int dev_disposition(const DevRun &r, uint64_t i, const CompactParams &cp,
                                       int shadow, uint8_t *changed_out, uint32_t *new_ts_out,
                                       uint64_t *kl_out, uint64_t *vl_out)
@@ -1009,14 +1022,24 @@ __device__ static int dev_disposition(const DevRun &r, uint64_t i, const Compact
     *vl_out = 0;
     if (shadow)
         return D_SHADOWED; /* newest-wins, decided during ranking */
-    if (r.sk[i] & 1)
-        return D_TOMBSTONE;
-    uint64_t kl, vl;
-    const uint8_t *k = run_key(r, i, &kl);
-    const uint8_t *v = run_val(r, i, &vl);
+    uint32_t expire_ts;
+    if (r.meta) {
+        uint64_t m = r.meta[i];
+        if (m & 1)
+            return D_TOMBSTONE;
+        expire_ts = (uint32_t)(m >> 32);
+    } else {
+        if (r.sk[i] & 1)
+            return D_TOMBSTONE;
+        uint64_t vl0;
+        expire_ts = dev_expire_ts(cp.data_version, run_val(r, i, &vl0));
+    }
+    uint64_t kl =
+        r.fixed_klen ? r.fixed_klen : (r.koff[i + 1] - r.koff[i]);
+    uint64_t vl =
+        r.fixed_vlen ? r.fixed_vlen : (r.voff[i + 1] - r.voff[i]);
     int drop = 0, value_changed = 0;
     uint32_t new_ts_val = 0;
-    uint32_t expire_ts = dev_expire_ts(cp.data_version, v);
     uint32_t eff_expire = expire_ts; /* value_view's expire after default-ttl */
     if (kl >= 2) {
         if (cp.default_ttl != 0 && expire_ts == 0) {
@@ -1026,6 +1049,7 @@ __device__ static int dev_disposition(const DevRun &r, uint64_t i, const Compact
             new_ts_val = expire_ts;
         }
         if (cp.n_ops > 0) {
+            const uint8_t *k = r.keys + (r.fixed_klen ? i * r.fixed_klen : r.koff[i]);
             uint32_t hklen = ((uint32_t)k[0] << 8) | k[1];
             const uint8_t *hk = k + 2;
             const uint8_t *skp = k + 2 + hklen;
@@ -1042,9 +1066,12 @@ __device__ static int dev_disposition(const DevRun &r, uint64_t i, const Compact
                         all = (e == 0 && rule.start_ttl == 0 && rule.stop_ttl == 0) ||
                               ((uint32_t)(rule.start_ttl + cp.epoch_now) <= e &&
                                (uint32_t)(rule.stop_ttl + cp.epoch_now) >= e);
+                    } else if (rule.type == DFR_HASHKEY) {
+                        all = dev_pattern_match(hk, hklen, rule.match_type,
+                                                cp.pats + rule.pat_off, rule.pat_len);
                     } else {
-                        all = dev_rule_match(rule, cp.pats, cp.data_version, cp.epoch_now, hk,
-                                             hklen, skp, sklen, v);
+                        all = dev_pattern_match(skp, sklen, rule.match_type,
+                                                cp.pats + rule.pat_off, rule.pat_len);
                     }
                 }
                 if (!all)
@@ -1084,9 +1111,13 @@ __device__ static int dev_disposition(const DevRun &r, uint64_t i, const Compact
         /* final keep/drop on local expire_ts + stale split hash (:91,114-121) */
         if (dev_ts_expired(cp.epoch_now, expire_ts))
             return D_EXPIRED;
-        if (cp.validate_hash && cp.partition_version >= 0 && cp.pidx <= cp.partition_version &&
-            (int64_t)(dev_key_hash(k, kl) & (uint64_t)cp.partition_version) != (int64_t)cp.pidx)
-            return D_FILTERED;
+        if (cp.validate_hash && cp.partition_version >= 0 && cp.pidx <= cp.partition_version) {
+            uint64_t kl2;
+            const uint8_t *k2 = run_key(r, i, &kl2);
+            if ((int64_t)(dev_key_hash(k2, kl2) & (uint64_t)cp.partition_version) !=
+                (int64_t)cp.pidx)
+                return D_FILTERED;
+        }
     }
     *changed_out = (uint8_t)value_changed;
     *new_ts_out = new_ts_val;
