@@ -2247,8 +2247,10 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     uint64_t *d_koffs = e->talloc<uint64_t>(total * 8);
     uint64_t *d_voffs = e->talloc<uint64_t>(total * 8);
     uint64_t *d_rank_of = e->emit_mode == 1 ? e->talloc<uint64_t>(total * 8) : nullptr;
-    CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(sizeof(CompactStatsDev));
-    HIP_OK(hipMemsetAsync(d_stats, 0, sizeof(CompactStatsDev), e->stream));
+    /* 8 stat banks: the group rank flushes per-block tallies to bank
+     * blockIdx&7; legacy kernels add to bank 0; finish sums all banks */
+    CompactStatsDev *d_stats = e->talloc<CompactStatsDev>(8 * sizeof(CompactStatsDev));
+    HIP_OK(hipMemsetAsync(d_stats, 0, 8 * sizeof(CompactStatsDev), e->stream));
     if (e->rank_mode == 1 && R > 1 && e->emit_mode == 2) {
         /* LDS-staged block rank: shift-8 bound table + per-run block counts */
         uint64_t *d_bt8_off = nullptr, *d_bt8 = nullptr;
@@ -2372,8 +2374,16 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
     uint64_t *t = e->pend_sizes;
     uint64_t n_out = t[0] + t[1], kbytes = t[2] + t[3], vbytes = t[4] + t[5];
 
+    CompactStatsDev hsb[8];
+    HIP_OK(hipMemcpy(hsb, d_stats, sizeof(hsb), hipMemcpyDeviceToHost));
     CompactStatsDev hs{};
-    HIP_OK(hipMemcpy(&hs, d_stats, sizeof(hs), hipMemcpyDeviceToHost));
+    for (int b = 0; b < 8; b++) {
+        hs.shadowed += hsb[b].shadowed;
+        hs.tombstones += hsb[b].tombstones;
+        hs.expired += hsb[b].expired;
+        hs.filtered += hsb[b].filtered;
+        hs.output_records += hsb[b].output_records;
+    }
     st.output_records = n_out;
     st.expired = hs.expired;
     st.filtered = hs.filtered;
