@@ -258,11 +258,24 @@ def main():
     barrier_sync()
     t0 = time.time()
     rows = 0
+    # pipelined: submit every partition's fused count kernels, then collect —
+    # partitions' scans co-run on their per-engine streams (rrdb_scan_count_*)
+    pending = []
     for eng in parts:
-        res = eng.scan_open(b"\x00\x00", b"\xff\xff", epoch_now, only_return_count=True,
-                            full_scan=True, validate_partition_hash=False, batch_size=2**31 - 1)
-        assert res.error == 0 and res.context_id == -1, (res.error, res.context_id)
-        rows += res.kv_count
+        rc = eng.scan_count_begin(b"\x00\x00", b"\xff\xff", epoch_now,
+                                  validate_partition_hash=False)
+        pending.append(rc == 0)
+        if rc != 0:  # shape/eligibility fallback: synchronous count
+            res = eng.scan_open(b"\x00\x00", b"\xff\xff", epoch_now, only_return_count=True,
+                                full_scan=True, validate_partition_hash=False,
+                                batch_size=2**31 - 1)
+            assert res.error == 0 and res.context_id == -1, (res.error, res.context_id)
+            rows += res.kv_count
+    for eng, ok in zip(parts, pending):
+        if ok:
+            err, cnt = eng.scan_count_finish()
+            assert err == 0
+            rows += cnt
     torch.cuda.synchronize()
     if dist:
         rt = torch.tensor([rows], dtype=torch.int64,

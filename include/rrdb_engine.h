@@ -279,6 +279,16 @@ int32_t rrdb_manual_compact(void *h, const rrdb_compact_options *opts, uint32_t 
  * run, returns stats).  _begin fails with kInvalidArgument if a split
  * compaction is already pending on the handle; _finish without a pending
  * _begin likewise.  rrdb_manual_compact == begin+finish. */
+/* pipelined count scan (the shell count_data fan-out,
+ * src/shell/commands/data_operations.cpp:2305): begin submits the fused
+ * count kernels without blocking so every partition's scan co-runs; finish
+ * returns the count in out->i64.  Only the full-count shape is supported
+ * (only_return_count, forward, start-inclusive, count/iteration caps >= the
+ * table's record count) — other shapes return kInvalidArgument and the
+ * caller uses rrdb_scan_open. */
+int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoch_now);
+int32_t rrdb_scan_count_finish(void *h, rrdb_result *out);
+
 int32_t rrdb_manual_compact_begin(void *h, const rrdb_compact_options *opts,
                                   uint32_t epoch_now);
 int32_t rrdb_manual_compact_finish(void *h, rrdb_compact_stats *stats);

@@ -186,6 +186,10 @@ class RrdbLib:
         L.rrdb_multi_get.argtypes = [C.c_void_p, C.POINTER(_MultiGetRequest), C.c_uint32, C.POINTER(_Result)]
         L.rrdb_scan_open.restype = C.c_int32
         L.rrdb_scan_open.argtypes = [C.c_void_p, C.POINTER(_ScanRequest), C.c_uint32, C.POINTER(_Result)]
+        L.rrdb_scan_count_begin.restype = C.c_int32
+        L.rrdb_scan_count_begin.argtypes = [C.c_void_p, C.POINTER(_ScanRequest), C.c_uint32]
+        L.rrdb_scan_count_finish.restype = C.c_int32
+        L.rrdb_scan_count_finish.argtypes = [C.c_void_p, C.POINTER(_Result)]
         L.rrdb_scan_next.restype = C.c_int32
         L.rrdb_scan_next.argtypes = [C.c_void_p, C.c_int64, C.c_uint32, C.POINTER(_Result)]
         L.rrdb_clear_scanner.argtypes = [C.c_void_p, C.c_int64]
@@ -471,6 +475,43 @@ class RrdbPartition:
                 self._scan_flags[out.context_id] = (only_return_count, return_expire_ts,
                                                     on_device_out)
             return out
+        finally:
+            self._L.rrdb_free_result(C.byref(res))
+
+    def scan_count_begin(self, start_key: bytes, stop_key: bytes, epoch_now: int, *,
+                         stop_inclusive=False, batch_size=2**31 - 1,
+                         hash_key_filter_type=FT_NO_FILTER, hash_key_filter_pattern=b"",
+                         sort_key_filter_type=FT_NO_FILTER, sort_key_filter_pattern=b"",
+                         validate_partition_hash=True):
+        """Submit a fused pipelined count scan; returns the C status.
+        kInvalidArgument = shape unsupported -> fall back to scan_open."""
+        keep = []
+        req = _ScanRequest()
+        req.start_key = _cslice(start_key, keep)
+        req.stop_key = _cslice(stop_key, keep)
+        req.start_inclusive = 1
+        req.stop_inclusive = 1 if stop_inclusive else 0
+        req.batch_size = batch_size
+        req.no_value = 1
+        req.hash_key_filter_type = hash_key_filter_type
+        req.hash_key_filter_pattern = _cslice(hash_key_filter_pattern, keep)
+        req.sort_key_filter_type = sort_key_filter_type
+        req.sort_key_filter_pattern = _cslice(sort_key_filter_pattern, keep)
+        req.full_scan = 1
+        req.validate_partition_hash = 1 if validate_partition_hash else 0
+        req.return_expire_ts = 0
+        req.only_return_count = 1
+        req.on_device_out = 0
+        rc = self._L.rrdb_scan_count_begin(self._h, C.byref(req), epoch_now)
+        del keep
+        return rc
+
+    def scan_count_finish(self):
+        """(error, count) for a submitted scan_count_begin."""
+        res = _Result()
+        self._L.rrdb_scan_count_finish(self._h, C.byref(res))
+        try:
+            return res.error, res.i64
         finally:
             self._L.rrdb_free_result(C.byref(res))
 

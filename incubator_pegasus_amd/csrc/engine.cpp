@@ -83,6 +83,8 @@ void launch_rank_grp_compact(const DevRun *, int, const uint64_t *, const uint64
                              uint32_t *, uint64_t *, uint64_t *, CompactStatsDev *, hipStream_t);
 void launch_rank_grp_view(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
                           uint64_t *, uint8_t *, hipStream_t);
+void launch_rank_grp_count(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
+                           const ScanParams &, CompactStatsDev *, hipStream_t);
 void launch_rank_compact_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
                               const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                               uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
@@ -609,6 +611,16 @@ struct HipEngine {
         rrdb_compact_stats st{};
     } pend;
     uint64_t *pend_sizes = nullptr; /* pinned [6]: output sizes d2h target */
+    /* pending fused count scan (rrdb_scan_count_begin/finish): buffers are
+     * plain hipMallocs so interleaved reads/compactions cannot reclaim them */
+    struct PendingScanCount {
+        bool active = false, trivial = false;
+        int64_t trivial_count = 0;
+        CompactStatsDev *d_stats = nullptr;
+        uint64_t *d_anch = nullptr, *d_lo = nullptr, *d_hi = nullptr;
+        uint8_t *d_start = nullptr, *d_stop = nullptr, *d_hk_pat = nullptr,
+                *d_sk_pat = nullptr;
+    } pend_scan;
     uint64_t next_seq_floor = 0;
     /* user ops: host + device */
     std::vector<HostOp> host_ops;
@@ -1159,6 +1171,14 @@ void rrdb_close(void *h)
     for (auto &kv : e->ctxs)
         delete kv.second;
     e->ctxs.clear();
+    {
+        auto &ps = e->pend_scan;
+        for (void *p : {(void *)ps.d_stats, (void *)ps.d_anch, (void *)ps.d_lo,
+                        (void *)ps.d_hi, (void *)ps.d_start, (void *)ps.d_stop,
+                        (void *)ps.d_hk_pat, (void *)ps.d_sk_pat})
+            if (p)
+                (void)hipFree(p);
+    }
     for (auto &r : e->runs)
         e->free_run(r);
     if (e->d_runs)
@@ -1756,6 +1776,156 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
         out->context_id = c->id;
     }
     return out->error;
+}
+
+/* fused pipelined count scan (the count_data path,
+ * src/shell/commands/data_operations.cpp:2305 fan-out): begin submits
+ * bounds + anchor rows + the MODE=2 group-rank kernel with NO host sync, so
+ * the caller can begin every partition and then finish each — partitions'
+ * count kernels co-run on their per-engine streams exactly like the
+ * split compaction.  Supports the full-count shape (only_return_count,
+ * forward, start-inclusive, caps >= table size); other shapes return
+ * kInvalidArgument and the caller falls back to rrdb_scan_open.  The 30s
+ * time budget does not apply (single fused operation; documented
+ * deviation like the batch-granular budget). */
+int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoch_now)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    if (e->pend_scan.active)
+        return RRDB_INVALID_ARGUMENT;
+    engine_flush(e);
+    e->activate();
+    if (!q->only_return_count || !q->start_inclusive || q->hash_key_filter_type < 0 ||
+        q->hash_key_filter_type > 3 || q->sort_key_filter_type < 0 ||
+        q->sort_key_filter_type > 3)
+        return RRDB_INVALID_ARGUMENT;
+    uint64_t total = 0;
+    for (auto &r : e->runs)
+        total += r.n;
+    uint64_t batch_cap = q->batch_size > 0 ? (uint64_t)q->batch_size : (uint64_t)INT32_MAX;
+    if (!e->grp_eligible() || batch_cap < total || e->max_iter_count < total)
+        return RRDB_INVALID_ARGUMENT;
+
+    std::string start((const char *)q->start_key.data, q->start_key.len);
+    std::string stop((const char *)q->stop_key.data, q->stop_key.len);
+    bool start_inclusive = true, stop_inclusive = q->stop_inclusive;
+    if (q->hash_key_filter_type == RRDB_FT_MATCH_PREFIX && q->hash_key_filter_pattern.len > 0) {
+        if (!hklen_ok(q->hash_key_filter_pattern.len))
+            return RRDB_INVALID_ARGUMENT;
+        std::string ps =
+            make_key(q->hash_key_filter_pattern.data, q->hash_key_filter_pattern.len, nullptr, 0);
+        if (key_cmp(ps, start) > 0)
+            start = ps;
+    }
+    int c = key_cmp(start, stop);
+    e->pend_scan = {};
+    if (c > 0 || (c == 0 && (!start_inclusive || !stop_inclusive))) {
+        e->pend_scan.active = true;
+        e->pend_scan.trivial = true;
+        e->pend_scan.trivial_count = 0;
+        return RRDB_OK; /* empty range (on_get_scanner:1227-1243) */
+    }
+    std::string stop_excl = stop;
+    if (stop_inclusive)
+        stop_excl.push_back('\0');
+
+    int R = (int)e->runs.size();
+    DevRun *dr = e->dev_runs();
+    auto &ps = e->pend_scan;
+    auto upl = [&](const std::string &b) {
+        uint8_t *d = nullptr;
+        HIP_OK(hipMalloc(&d, b.size() ? b.size() : 1));
+        if (!b.empty())
+            HIP_OK(hipMemcpyAsync(d, b.data(), b.size(), hipMemcpyHostToDevice, e->stream));
+        return d;
+    };
+    HIP_OK(hipMalloc(&ps.d_lo, R * 8));
+    HIP_OK(hipMalloc(&ps.d_hi, R * 8));
+    ps.d_start = upl(start);
+    ps.d_stop = upl(stop_excl);
+    launch_bounds(dr, R, ps.d_start, start.size(), ps.d_lo, 0, e->stream);
+    launch_bounds(dr, R, ps.d_stop, stop_excl.size(), ps.d_hi, 0, e->stream);
+    /* anchors from the largest run's FULL size (the window is device-side
+     * only); rows past the window collapse to sentinels in-kernel */
+    int q0 = 0;
+    uint64_t nmax = 0;
+    for (int r = 0; r < R; r++)
+        if (e->runs[r].n > nmax) {
+            nmax = e->runs[r].n;
+            q0 = r;
+        }
+    int gs = 4;
+    while ((1ull << (gs + 1)) * (uint64_t)R <= GRP_TARGET && gs < 8)
+        gs++;
+    uint64_t n_groups = (nmax + (1ull << gs) - 1) >> gs;
+    if (n_groups == 0)
+        n_groups = 1;
+    HIP_OK(hipMalloc(&ps.d_anch, (n_groups + 1) * R * 8));
+    launch_anchor_rows(dr, R, q0, ps.d_lo, ps.d_hi, gs, n_groups, ps.d_anch, e->stream);
+
+    ScanParams sp{};
+    sp.epoch_now = epoch_now;
+    sp.data_version = e->data_version;
+    sp.pidx = e->pidx;
+    sp.partition_version = e->partition_version;
+    sp.validate_hash = (uint8_t)(q->validate_partition_hash && e->validate_hash);
+    sp.hk_ft = q->hash_key_filter_type;
+    sp.sk_ft = q->sort_key_filter_type;
+    sp.hk_pat_len = q->hash_key_filter_pattern.len;
+    ps.d_hk_pat = upl(std::string((const char *)q->hash_key_filter_pattern.data,
+                                  q->hash_key_filter_pattern.len));
+    sp.hk_pat = ps.d_hk_pat;
+    sp.sk_pat_len = q->sort_key_filter_pattern.len;
+    ps.d_sk_pat = upl(std::string((const char *)q->sort_key_filter_pattern.data,
+                                  q->sort_key_filter_pattern.len));
+    sp.sk_pat = ps.d_sk_pat;
+    sp.no_value = 1;
+    sp.hash_key_skip = 0;
+
+    HIP_OK(hipMalloc(&ps.d_stats, 8 * sizeof(CompactStatsDev)));
+    HIP_OK(hipMemsetAsync(ps.d_stats, 0, 8 * sizeof(CompactStatsDev), e->stream));
+    launch_rank_grp_count(dr, R, ps.d_lo, ps.d_anch, n_groups, sp, ps.d_stats, e->stream);
+    ps.active = true;
+    return RRDB_OK;
+}
+
+int32_t rrdb_scan_count_finish(void *h, rrdb_result *out)
+{
+    auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    result_init(out);
+    auto &ps = e->pend_scan;
+    if (!ps.active) {
+        out->error = RRDB_INVALID_ARGUMENT;
+        return out->error;
+    }
+    ps.active = false;
+    e->activate();
+    if (ps.trivial) {
+        out->i64 = ps.trivial_count;
+        out->count = 0;
+        out->error = RRDB_OK;
+        out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+        return RRDB_OK;
+    }
+    HIP_OK(hipStreamSynchronize(e->stream));
+    CompactStatsDev hsb[8];
+    HIP_OK(hipMemcpy(hsb, ps.d_stats, sizeof(hsb), hipMemcpyDeviceToHost));
+    uint64_t count = 0;
+    for (auto &b : hsb)
+        count += b.output_records;
+    for (void *p : {(void *)ps.d_stats, (void *)ps.d_anch, (void *)ps.d_lo, (void *)ps.d_hi,
+                    (void *)ps.d_start, (void *)ps.d_stop, (void *)ps.d_hk_pat,
+                    (void *)ps.d_sk_pat})
+        if (p)
+            (void)hipFree(p);
+    e->pend_scan = {};
+    out->i64 = (int64_t)count;
+    out->count = 0;
+    out->error = RRDB_OK;
+    out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+    return RRDB_OK;
 }
 
 void rrdb_clear_scanner(void *h, int64_t context_id)

@@ -1508,17 +1508,92 @@ struct GrpTally {
 };
 
 
-/* VIEW=0: compaction (fused KeyWithTTLCompactionFilter disposition outputs);
- * VIEW=1: scan view build (order + shadowed only).
+/* countable record for the fused count scan (validate_key_value_for_scan,
+ * pegasus_server_impl.cpp:2382-2432, count-only flavor): visible PUT, not
+ * expired, passes hash validation + hash/sort-key filters */
+__device__ static inline int dev_count_ok(const DevRun &r, uint64_t i, const ScanParams &sp)
+{
+    uint32_t ets;
+    int kind;
+    if (r.meta) {
+        uint64_t m = r.meta[i];
+        kind = (int)(m & 1);
+        ets = (uint32_t)(m >> 32);
+    } else {
+        kind = (int)(r.sk[i] & 1);
+        uint64_t vtmp;
+        ets = kind ? 0 : dev_expire_ts(sp.data_version, run_val(r, i, &vtmp));
+    }
+    if (kind)
+        return 0; /* tombstone: invisible to the iterator */
+    if (dev_ts_expired(sp.epoch_now, ets))
+        return 0;
+    if (sp.validate_hash || sp.hk_ft != 0 || sp.sk_ft != 0) {
+        uint64_t kl;
+        const uint8_t *k = run_key(r, i, &kl);
+        if (sp.validate_hash &&
+            (sp.partition_version < 0 || sp.pidx > sp.partition_version ||
+             (int64_t)(dev_key_hash(k, kl) & (uint64_t)sp.partition_version) !=
+                 (int64_t)sp.pidx))
+            return 0;
+        if (sp.hk_ft != 0 || sp.sk_ft != 0) {
+            uint32_t hklen = ((uint32_t)k[0] << 8) | k[1];
+            const uint8_t *hk = k + 2;
+            const uint8_t *skp = k + 2 + hklen;
+            uint64_t sklen = kl - 2 - hklen;
+            if (sp.hk_ft != 0 &&
+                !dev_validate_filter(sp.hk_ft, sp.hk_pat, sp.hk_pat_len, hk, hklen))
+                return 0;
+            if (sp.sk_ft != 0 &&
+                !dev_validate_filter(sp.sk_ft, sp.sk_pat, sp.sk_pat_len, skp, sklen))
+                return 0;
+        }
+    }
+    return 1;
+}
+
+/* per-element epilogue of the group rank: MODE 0 = compaction outputs,
+ * MODE 1 = view (order + shadowed), MODE 2 = fused count (no arrays) */
+template <int MODE>
+__device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i, int shadow,
+                                          uint64_t rank, const CompactParams &cp,
+                                          const ScanParams &sp, uint64_t *order,
+                                          uint64_t *keepw, uint8_t *changed,
+                                          uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
+                                          uint8_t *shadowed)
+{
+    if (MODE == 2)
+        return (!shadow && dev_count_ok(runs[q], i, sp)) ? D_KEEP : D_NONE;
+    order[rank] = ((uint64_t)q << 40) | i;
+    if (MODE == 1) {
+        shadowed[rank] = (uint8_t)shadow;
+        return D_NONE;
+    }
+    uint8_t ch;
+    uint32_t nts;
+    uint64_t okl, ovl;
+    int disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
+    keepw[rank] = (disp == D_KEEP) ? 1 : 0;
+    changed[rank] = ch;
+    new_expire[rank] = nts;
+    ksz[rank] = okl;
+    vsz[rank] = ovl;
+    return disp;
+}
+
+/* MODE=0: compaction (fused KeyWithTTLCompactionFilter disposition outputs);
+ * MODE=1: scan view build (order + shadowed only);
+ * MODE=2: fused count scan (tallies only — the count_data hot path).
  * Staged groups rank by PAIRWISE MERGE-PATH ROUNDS in LDS (log2(R) rounds;
  * each round every element moves once, each thread owns a fixed output-slot
  * range found by one diagonal search) — ~5-10x fewer instructions per
  * element than per-element binary searches, which measured issue-bound. */
-template <int VIEW>
+template <int MODE>
 __global__ void __launch_bounds__(BLOCK) k_rank_grp(
     const DevRun *runs, int R, const uint64_t *lo, const uint64_t *anch, uint64_t n_groups,
-    CompactParams cp, uint64_t *order, uint64_t *keepw, uint8_t *changed, uint32_t *new_expire,
-    uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of, uint8_t *shadowed, CompactStatsDev *stats)
+    CompactParams cp, ScanParams sp, uint64_t *order, uint64_t *keepw, uint8_t *changed,
+    uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of, uint8_t *shadowed,
+    CompactStatsDev *stats)
 {
     __shared__ uint64_t s_ta[GRP_CAP], s_tb[GRP_CAP]; /* tails ping-pong */
     __shared__ uint16_t s_oa[GRP_CAP], s_ob[GRP_CAP]; /* (q<<12)|segpos */
@@ -1657,22 +1732,10 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
                             }
                     }
                     uint64_t rank = s_base + p;
-                    order[rank] = ((uint64_t)q << 40) | i;
-                    if (VIEW) {
-                        shadowed[rank] = (uint8_t)shadow;
-                    } else {
-                        uint8_t ch;
-                        uint32_t nts;
-                        uint64_t okl, ovl;
-                        disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
-                        keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-                        changed[rank] = ch;
-                        new_expire[rank] = nts;
-                        ksz[rank] = okl;
-                        vsz[rank] = ovl;
-                    }
+                    disp = grp_epilogue<MODE>(runs, q, i, shadow, rank, cp, sp, order, keepw,
+                                              changed, new_expire, ksz, vsz, shadowed);
                 }
-                if (!VIEW)
+                if (MODE != 1)
                     tally.add(disp);
             }
         } else {
@@ -1723,28 +1786,16 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
                             shadow = 1;
                     }
                     uint64_t rank = s_base + lrank;
-                    order[rank] = ((uint64_t)q << 40) | i;
-                    if (VIEW) {
-                        shadowed[rank] = (uint8_t)shadow;
-                    } else {
-                        uint8_t ch;
-                        uint32_t nts;
-                        uint64_t okl, ovl;
-                        disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
-                        keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-                        changed[rank] = ch;
-                        new_expire[rank] = nts;
-                        ksz[rank] = okl;
-                        vsz[rank] = ovl;
-                    }
+                    disp = grp_epilogue<MODE>(runs, q, i, shadow, rank, cp, sp, order, keepw,
+                                              changed, new_expire, ksz, vsz, shadowed);
                 }
-                if (!VIEW)
+                if (MODE != 1)
                     tally.add(disp);
             }
         }
         __syncthreads(); /* LDS reused by the next group */
     }
-    if (!VIEW)
+    if (MODE != 1)
         tally.flush(stats);
 }
 
@@ -1759,9 +1810,10 @@ void launch_rank_grp_compact(const DevRun *d_runs, int R, const uint64_t *d_lo,
         blocks = 1;
     if (blocks > 8192)
         blocks = 8192;
+    ScanParams sp{};
     k_rank_grp<0><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
-        d_runs, R, d_lo, d_anch, n_groups, cp, d_order, d_keepw, d_changed, d_new_expire, d_ksz,
-        d_vsz, nullptr, nullptr, d_stats);
+        d_runs, R, d_lo, d_anch, n_groups, cp, sp, d_order, d_keepw, d_changed, d_new_expire,
+        d_ksz, d_vsz, nullptr, nullptr, d_stats);
 }
 
 void launch_rank_grp_view(const DevRun *d_runs, int R, const uint64_t *d_lo,
@@ -1773,11 +1825,28 @@ void launch_rank_grp_view(const DevRun *d_runs, int R, const uint64_t *d_lo,
         blocks = 1;
     if (blocks > 65535)
         blocks = 65535;
-    (void)0;
     CompactParams cp{};
+    ScanParams sp{};
     k_rank_grp<1><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
-        d_runs, R, d_lo, d_anch, n_groups, cp, d_order, nullptr, nullptr, nullptr, nullptr,
+        d_runs, R, d_lo, d_anch, n_groups, cp, sp, d_order, nullptr, nullptr, nullptr, nullptr,
         nullptr, nullptr, d_shadow, nullptr);
+}
+
+/* fused count scan: shadow + countability evaluated in-kernel, the count
+ * lands in stats->output_records (banked); zero intermediate arrays */
+void launch_rank_grp_count(const DevRun *d_runs, int R, const uint64_t *d_lo,
+                           const uint64_t *d_anch, uint64_t n_groups, const ScanParams &sp,
+                           CompactStatsDev *d_stats, hipStream_t s)
+{
+    uint64_t blocks = n_groups;
+    if (blocks == 0)
+        blocks = 1;
+    if (blocks > 8192)
+        blocks = 8192;
+    CompactParams cp{};
+    k_rank_grp<2><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+        d_runs, R, d_lo, d_anch, n_groups, cp, sp, nullptr, nullptr, nullptr, nullptr, nullptr,
+        nullptr, nullptr, nullptr, d_stats);
 }
 
 void launch_rank_compact_ldst(const DevRun *d_runs, int R, const uint64_t *d_lo,

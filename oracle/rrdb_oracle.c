@@ -813,6 +813,10 @@ typedef struct {
     int pend_active;
     int pend_keep_inputs;
     uint32_t pend_epoch;
+    /* pending fused count scan (rrdb_scan_count_begin/finish): computed
+     * eagerly (the oracle has no async phase), parked until finish */
+    int scancnt_active;
+    int64_t scancnt_value;
 } Engine;
 
 /* a parked scanner: materialized visible view of [cursor..stop) */
@@ -1922,6 +1926,60 @@ int32_t rrdb_scan_next(void *h, int64_t context_id, uint32_t epoch_now, rrdb_res
         out->context_id = park_ctx(e, c, epoch_now); /* re-put -> new handle (on_scan:1516-1526) */
     }
     return out->error;
+}
+
+/* pipelined count scan restatement (include/rrdb_engine.h): same shape
+ * gate as the engine; computed synchronously via the scan machinery */
+int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoch_now)
+{
+    Engine *e = (Engine *)h;
+    if (e->scancnt_active)
+        return RRDB_INVALID_ARGUMENT;
+    if (!q->only_return_count || !q->start_inclusive || q->hash_key_filter_type < 0 ||
+        q->hash_key_filter_type > 3 || q->sort_key_filter_type < 0 ||
+        q->sort_key_filter_type > 3)
+        return RRDB_INVALID_ARGUMENT;
+    rrdb_flush(h);
+    uint64_t total = 0;
+    for (int r = 0; r < e->n_runs; r++)
+        total += e->runs[r].n;
+    uint64_t batch_cap = q->batch_size > 0 ? (uint64_t)q->batch_size : (uint64_t)INT32_MAX;
+    if (batch_cap < total || e->max_iter_count < total)
+        return RRDB_INVALID_ARGUMENT;
+    rrdb_result tmp;
+    int32_t rc = rrdb_scan_open(h, q, epoch_now, &tmp);
+    int64_t count = tmp.i64;
+    int64_t ctx = tmp.context_id;
+    rrdb_free_result(&tmp);
+    if (rc != RRDB_OK)
+        return rc;
+    while (ctx != RRDB_SCAN_CONTEXT_ID_COMPLETED) {
+        rc = rrdb_scan_next(h, ctx, epoch_now, &tmp);
+        count += tmp.i64;
+        ctx = tmp.context_id;
+        rrdb_free_result(&tmp);
+        if (rc != RRDB_OK)
+            return rc;
+    }
+    e->scancnt_active = 1;
+    e->scancnt_value = count;
+    return RRDB_OK;
+}
+
+int32_t rrdb_scan_count_finish(void *h, rrdb_result *out)
+{
+    Engine *e = (Engine *)h;
+    result_init(out);
+    if (!e->scancnt_active) {
+        out->error = RRDB_INVALID_ARGUMENT;
+        return out->error;
+    }
+    e->scancnt_active = 0;
+    out->i64 = e->scancnt_value;
+    out->count = 0;
+    out->error = RRDB_OK;
+    out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
+    return RRDB_OK;
 }
 
 void rrdb_clear_scanner(void *h, int64_t context_id)

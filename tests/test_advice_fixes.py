@@ -152,6 +152,20 @@ def test_compact_invalidates_scanner_oracle(oracle_part):
     assert res.error == NOT_FOUND
 
 
+def test_scan_count_begin_finish_oracle(oracle_part):
+    _fill(oracle_part, [(b"aa", f"s{i}".encode(), b"v") for i in range(10)] +
+                       [(b"bb", b"t0", None)])
+    oracle_part.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
+    rc = oracle_part.scan_count_begin(b"\x00\x00", b"\xff\xff", NOW,
+                                      validate_partition_hash=False)
+    assert rc == OK
+    err, cnt = oracle_part.scan_count_finish()
+    assert err == OK and cnt == 10  # tombstone excluded
+    # finish without begin
+    err, _ = oracle_part.scan_count_finish()
+    assert err == INVALID_ARGUMENT
+
+
 # ---------------- checkpoint restore validation ----------------
 
 def _corrupt(path, mode):
