@@ -61,6 +61,7 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       uint8_t *, hipStream_t);
 void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
 void launch_bloom_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
+void launch_bloom_pfx_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
 void launch_build_tails(const uint8_t *, uint64_t, uint64_t, uint64_t *, hipStream_t);
 void launch_build_meta(const uint8_t *, const uint64_t *, uint64_t, const uint64_t *, uint64_t,
                        uint32_t, uint64_t *, hipStream_t);
@@ -552,6 +553,8 @@ struct RunBuf {
     uint32_t pfx_skip = 0;   /* lcp(first,last) floored to 8B words, <=16 */
     uint32_t lcp_exact = 0;  /* exact lcp(first,last), capped at 32 */
     uint64_t *tails = nullptr; /* packed BE tail words (word-probe mode) */
+    uint64_t *pfx_bloom = nullptr; /* hashkey-prefix blocked bloom */
+    uint64_t pfx_bloom_blocks = 0;
     uint64_t *meta = nullptr;  /* (expire_ts<<32)|kind disposition column */
     uint32_t fixed_vlen = 0;   /* constant encoded-value stride (0=variable) */
 };
@@ -611,6 +614,20 @@ struct HipEngine {
         rrdb_compact_stats st{};
     } pend;
     uint64_t *pend_sizes = nullptr; /* pinned [6]: output sizes d2h target */
+    /* serving-lane staging for the fused multi_get: one pinned H2D of
+     * [start|stop|pattern], one pinned D2H of hdr + blob prefix (small-op
+     * latency; pageable copies + two syncs dominated the 77us/call path) */
+    static constexpr uint64_t MG_IN_CAP = 64 << 10;
+    static constexpr uint64_t MG_OUT_PREFIX = 16 << 10;
+    uint8_t *mg_hin = nullptr, *mg_din = nullptr, *mg_hout = nullptr;
+    void mg_lane_init()
+    {
+        if (mg_hin)
+            return;
+        HIP_OK(hipHostMalloc((void **)&mg_hin, MG_IN_CAP));
+        HIP_OK(hipMalloc(&mg_din, MG_IN_CAP));
+        HIP_OK(hipHostMalloc((void **)&mg_hout, 32 + MG_OUT_PREFIX));
+    }
     /* pending fused count scan (rrdb_scan_count_begin/finish): buffers are
      * plain hipMallocs so interleaved reads/compactions cannot reclaim them */
     struct PendingScanCount {
@@ -698,6 +715,8 @@ struct HipEngine {
             (void)hipFree(r.bloom);
         if (r.tails)
             (void)hipFree(r.tails);
+        if (r.pfx_bloom)
+            (void)hipFree(r.pfx_bloom);
         if (r.meta)
             (void)hipFree(r.meta);
         r = RunBuf();
@@ -735,11 +754,22 @@ struct HipEngine {
             n_blocks = 1;
         HIP_OK(hipMalloc(&r.bloom, n_blocks * 64));
         HIP_OK(hipMemsetAsync(r.bloom, 0, n_blocks * 64, stream));
-        DevRun dr{r.keys, r.koff, r.vals, r.voff, r.sk, r.n, nullptr, 0, 0, 0, 0,
-                  nullptr, nullptr, 0};
+        DevRun dr{};
+        dr.keys = r.keys;
+        dr.koff = r.koff;
+        dr.vals = r.vals;
+        dr.voff = r.voff;
+        dr.sk = r.sk;
+        dr.n = r.n;
         launch_bloom_build(dr, r.bloom, n_blocks, stream);
+        /* §8(f)3 second half: hashkey-prefix bloom (the reference's
+         * prefix-extractor filter, _init.cpp:816-841, hashkey_transform.h) */
+        HIP_OK(hipMalloc(&r.pfx_bloom, n_blocks * 64));
+        HIP_OK(hipMemsetAsync(r.pfx_bloom, 0, n_blocks * 64, stream));
+        launch_bloom_pfx_build(dr, r.pfx_bloom, n_blocks, stream);
         HIP_OK(hipStreamSynchronize(stream));
         r.bloom_blocks = n_blocks;
+        r.pfx_bloom_blocks = n_blocks;
     }
 
     DevRun *dev_runs()
@@ -748,11 +778,26 @@ struct HipEngine {
             if (d_runs)
                 (void)hipFree(d_runs);
             std::vector<DevRun> h(runs.size() ? runs.size() : 1);
-            for (size_t i = 0; i < runs.size(); i++)
-                h[i] = DevRun{runs[i].keys, runs[i].koff, runs[i].vals, runs[i].voff, runs[i].sk,
-                              runs[i].n,    runs[i].bloom, runs[i].bloom_blocks,
-                              runs[i].fixed_klen, runs[i].pfx_skip, runs[i].lcp_exact,
-                              runs[i].tails, runs[i].meta, runs[i].fixed_vlen};
+            for (size_t i = 0; i < runs.size(); i++) {
+                DevRun d{};
+                d.keys = runs[i].keys;
+                d.koff = runs[i].koff;
+                d.vals = runs[i].vals;
+                d.voff = runs[i].voff;
+                d.sk = runs[i].sk;
+                d.n = runs[i].n;
+                d.bloom = runs[i].bloom;
+                d.bloom_blocks = runs[i].bloom_blocks;
+                d.fixed_klen = runs[i].fixed_klen;
+                d.pfx_skip = runs[i].pfx_skip;
+                d.lcp_exact = runs[i].lcp_exact;
+                d.tails = runs[i].tails;
+                d.pfx_bloom = runs[i].pfx_bloom;
+                d.pfx_bloom_blocks = runs[i].pfx_bloom_blocks;
+                d.meta = runs[i].meta;
+                d.fixed_vlen = runs[i].fixed_vlen;
+                h[i] = d;
+            }
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
@@ -1147,6 +1192,9 @@ void *rrdb_open(int32_t app_id, int32_t pidx, int32_t gpu_id)
     e->pidx = pidx;
     e->device = gpu_id;
     e->activate();
+    /* spin completion: interrupt-driven stream waits cost ~10-20us each on
+     * the small-op serving path (ignored if the runtime refuses) */
+    (void)hipSetDeviceFlags(hipDeviceScheduleSpin);
     std::call_once(g_init_once, [] {
         build_crc64_table();
         launch_crc64_table_init(host_crc64_table);
@@ -1189,6 +1237,11 @@ void rrdb_close(void *h)
         (void)hipHostFree(e->pinned);
     if (e->pend_sizes)
         (void)hipHostFree(e->pend_sizes);
+    if (e->mg_hin) {
+        (void)hipHostFree(e->mg_hin);
+        (void)hipFree(e->mg_din);
+        (void)hipHostFree(e->mg_hout);
+    }
     if (e->d_ops)
         (void)hipFree(e->d_ops);
     if (e->d_rules)
@@ -2126,12 +2179,34 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
     if (stop_inclusive)
         stop_excl.push_back('\0');
 
-    /* ---- fused single-launch fast path (small ranges, the YCSB-E shape) ---- */
+    /* ---- fused single-launch fast path (small ranges, the YCSB-E shape):
+     * one pinned H2D of the request bytes, one launch, one pinned D2H of
+     * hdr + blob prefix, one sync ---- */
     if (!e->runs.empty() && (int)e->runs.size() <= RRDB_MAX_RUNS) {
+        e->mg_lane_init();
         MgFusedArgs fa{};
-        fa.start = e->upload_tmp(start.data(), start.size());
+        uint64_t in_n = start.size() + stop_excl.size() + q->sort_key_filter_pattern.len;
+        if (in_n <= HipEngine::MG_IN_CAP) {
+            uint64_t o = 0;
+            memcpy(e->mg_hin + o, start.data(), start.size());
+            fa.start = e->mg_din + o;
+            o += start.size();
+            memcpy(e->mg_hin + o, stop_excl.data(), stop_excl.size());
+            fa.stop = e->mg_din + o;
+            o += stop_excl.size();
+            if (q->sort_key_filter_pattern.len)
+                memcpy(e->mg_hin + o, q->sort_key_filter_pattern.data,
+                       q->sort_key_filter_pattern.len);
+            fa.sk_pat = e->mg_din + o;
+            HIP_OK(hipMemcpyAsync(e->mg_din, e->mg_hin, in_n ? in_n : 1,
+                                  hipMemcpyHostToDevice, e->stream));
+        } else { /* giant keys: pageable per-piece uploads */
+            fa.start = e->upload_tmp(start.data(), start.size());
+            fa.stop = e->upload_tmp(stop_excl.data(), stop_excl.size());
+            fa.sk_pat = e->upload_tmp(q->sort_key_filter_pattern.data,
+                                      q->sort_key_filter_pattern.len);
+        }
         fa.start_len = start.size();
-        fa.stop = e->upload_tmp(stop_excl.data(), stop_excl.size());
         fa.stop_len = stop_excl.size();
         fa.start_inclusive = start_inclusive;
         fa.stop_inclusive = stop_inclusive;
@@ -2142,28 +2217,33 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
         fa.max_iteration_size = max_iter_size;
         fa.sk_ft = q->sort_key_filter_type;
         fa.sk_pat_len = q->sort_key_filter_pattern.len;
-        fa.sk_pat = e->upload_tmp(q->sort_key_filter_pattern.data, fa.sk_pat_len);
         fa.epoch_now = epoch_now;
         fa.data_version = e->data_version;
         fa.hash_key_skip = 2 + q->hash_key.len;
-        fa.out_hdr = e->talloc<int64_t>(4 * 8);
-        fa.out_blob = e->talloc<uint8_t>(MG_BLOB_BYTES);
+        uint8_t *d_out = e->talloc<uint8_t>(32 + MG_BLOB_BYTES);
+        fa.out_hdr = (int64_t *)d_out;
+        fa.out_blob = d_out + 32;
         launch_multi_get_small(e->dev_runs(), (int)e->runs.size(), fa, e->stream);
-        int64_t hdr4[4];
-        HIP_OK(hipMemcpyAsync(hdr4, fa.out_hdr, 32, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(e->mg_hout, d_out, 32 + HipEngine::MG_OUT_PREFIX,
+                              hipMemcpyDeviceToHost, e->stream));
         HIP_OK(hipStreamSynchronize(e->stream));
+        const int64_t *hdr4 = (const int64_t *)e->mg_hout;
         if (hdr4[0] >= 0) {
             uint64_t m = (uint64_t)hdr4[0];
             uint64_t kb = (uint64_t)hdr4[2], vb = (uint64_t)hdr4[3];
+            int64_t complete_flag = hdr4[1];
             out->keys = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
             out->values = (rrdb_slice *)a->alloc(m * sizeof(rrdb_slice));
             if (m) {
-                /* one D2H of [koff|voff|keys|vals] */
                 uint64_t blob_n = 2 * (m + 1) * 8 + kb + vb;
                 uint8_t *hb = (uint8_t *)a->alloc(blob_n);
-                HIP_OK(hipMemcpyAsync(hb, fa.out_blob, blob_n, hipMemcpyDeviceToHost,
-                                      e->stream));
-                HIP_OK(hipStreamSynchronize(e->stream));
+                if (blob_n <= HipEngine::MG_OUT_PREFIX) {
+                    memcpy(hb, e->mg_hout + 32, blob_n); /* already on host */
+                } else {
+                    HIP_OK(hipMemcpyAsync(hb, fa.out_blob, blob_n, hipMemcpyDeviceToHost,
+                                          e->stream));
+                    HIP_OK(hipStreamSynchronize(e->stream));
+                }
                 const uint64_t *koffs = (const uint64_t *)hb;
                 const uint64_t *voffs = koffs + (m + 1);
                 uint8_t *hk = hb + 2 * (m + 1) * 8;
@@ -2174,7 +2254,7 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
                 }
             }
             out->count = m;
-            out->error = hdr4[1] ? RRDB_OK : RRDB_INCOMPLETE;
+            out->error = complete_flag ? RRDB_OK : RRDB_INCOMPLETE;
             return out->error;
         }
         /* fallback: large range or oversize rows — general path below */

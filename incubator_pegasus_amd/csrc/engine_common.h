@@ -51,6 +51,12 @@ struct DevRun {
      * creation when the single-word probe mode is eligible: 8B-strided probe
      * loads instead of klen-strided ones (null when ineligible) */
     const uint64_t *tails;
+    /* hashkey-PREFIX blocked bloom (the reference's prefix-extractor bloom,
+     * pegasus_server_impl_init.cpp:816-841 + hashkey_transform.h:40-54):
+     * hashed over key[0 .. 2+hklen); prefix-scoped reads (multi_get, prefix
+     * scans, sortkey_count) skip runs that cannot contain the hashkey. */
+    const uint64_t *pfx_bloom;
+    uint64_t pfx_bloom_blocks;
     /* per-record disposition column, built at run creation:
      * (expire_ts << 32) | kind.  Compaction-filter / scan-state evaluation
      * reads ONE coalescable 8B word instead of the sk word + voff pair + the

@@ -155,6 +155,48 @@ def test_grp_rank_many_runs_uses_fallback_mode(oracle_lib, hip_lib):
         g.close()
 
 
+def test_prefix_bloom_parity_and_toggle(oracle_lib, hip_lib):
+    """§8(f)3 second half: the hashkey-prefix bloom is lossless — multi_get
+    and sortkey_count agree with the oracle for present and ABSENT hashkeys,
+    filter on and off (reference prefix-extractor bloom,
+    pegasus_server_impl_init.cpp:816-841, hashkey_transform.h:40-54)."""
+    for filt in ("common", "none"):
+        o = oracle_lib.open(1, 0, -1)
+        g = hip_lib.open(1, 0, 0)
+        try:
+            g.set_envs({"rocksdb.filter_type": filt})
+            seq = 1
+            for run in range(5):
+                recs = []
+                for i in range(run, 300, 5):
+                    for sk in (b"a", b"b", b"c"):
+                        recs.append((D.generate_key(b"phk%04d" % i, sk),
+                                     D.encode_value(b"v%d" % run, 0, seq, 1), seq, 0))
+                        seq += 1
+                recs.sort(key=lambda r: r[0])
+                o.ingest_run(recs)
+                g.ingest_run(recs)
+            # hashkeys present in some runs, absent from others, fully absent
+            for i in list(range(0, 300, 7)) + [9999]:
+                hk = b"phk%04d" % i
+                assert o.multi_get(hk, NOW) == g.multi_get(hk, NOW), (filt, hk)
+                assert o.sortkey_count(hk, NOW) == g.sortkey_count(hk, NOW), (filt, hk)
+            # range-limited multi_get under one hashkey
+            assert o.multi_get(b"phk0010", NOW, start_sortkey=b"a", stop_sortkey=b"b",
+                               stop_inclusive=True) == \
+                   g.multi_get(b"phk0010", NOW, start_sortkey=b"a", stop_sortkey=b"b",
+                               stop_inclusive=True), filt
+            # blooms rebuilt after compaction
+            o.manual_compact(NOW)
+            g.manual_compact(NOW)
+            for i in (0, 5, 9999):
+                hk = b"phk%04d" % i
+                assert o.multi_get(hk, NOW) == g.multi_get(hk, NOW), (filt, "post", hk)
+        finally:
+            o.close()
+            g.close()
+
+
 def _count_fused_or_fallback(part, now, **kw):
     rc = part.scan_count_begin(b"\x00\x00", b"\xff\xff", now, **kw)
     if rc == 0:
