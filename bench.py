@@ -28,6 +28,7 @@ timed on a bounded sample on this box's host cores — reported, not the target.
 import argparse
 import json
 import os
+import subprocess
 import sys
 import time
 
@@ -106,6 +107,9 @@ def main():
     ap.add_argument("--emit-mode", choices=["chunked", "rank", "input"], default="chunked")
     ap.add_argument("--bt-shift", type=int, default=5)
     ap.add_argument("--rank-mode", choices=["global", "lds", "ldst", "grp"], default="grp")
+    ap.add_argument("--ycsb-hashkeys-per-part", type=int, default=625_000,
+                    help="YCSB C/E table: hashkeys per partition x 10 sortkeys "
+                         "(default 625K x 16 parts = 100M rows: configs[3] scale)")
     ap.add_argument("--rules-profile", action="store_true",
                     help="configs[4] shape: TTL'd data + default_ttl + user "
                          "delete/update-TTL compaction rules evaluated per key")
@@ -316,46 +320,86 @@ def main():
     get_ops_per_s = nq / get_elapsed
     get_kernel_ms = parts[0].phase_ms("get_search")
 
-    # ---- YCSB-E: prefix multi_get over a sortkey table (config 3 shape) ----
-    # single request stream through the C-ABI (per-call latency path; the
-    # reference serves these concurrently across THREAD_POOL_SCAN threads)
+    # ---- YCSB C/E at configs[3] scale: a sortkey table of
+    # (hashkeys_per_part x 10) rows on EVERY partition (16 x 625K x 10 =
+    # 100M rows by default), zipfian point gets + prefix multi_gets spread
+    # across all of them; single-stream latency + batched throughput are
+    # also measured from a C++ host (bin/bench_mg) since the reference's
+    # replica server is C++ and the ctypes loop adds interpreter overhead ----
     from incubator_pegasus_amd import data as D2
 
-    scan_part = hip.open(2, 0, local_rank)
-    st_run = D2.build_scan_table_run(200_000, 10, seed=D2.DEFAULT_SEED + 77)
-    scan_part.ingest_run_arrays(np.ascontiguousarray(st_run["keys"]), st_run["koff"],
-                                np.ascontiguousarray(st_run["vals"]), st_run["voff"],
-                                st_run["sk"])
-    mg_n = 2000
-    mg_ids = D2.zipfian_ids(mg_n, 200_000, seed=D2.DEFAULT_SEED + 5)
+    hk_pp = args.ycsb_hashkeys_per_part
+    e_parts = []
+    for p_ in range(args.partitions):
+        ep = hip.open(2, p_, local_rank)
+        st_run = D2.build_scan_table_run(hk_pp, 10, seed=D2.DEFAULT_SEED + 77 + p_)
+        ep.ingest_run_arrays(np.ascontiguousarray(st_run["keys"]), st_run["koff"],
+                             np.ascontiguousarray(st_run["vals"]), st_run["voff"],
+                             st_run["sk"])
+        e_parts.append(ep)
+    log(f"ycsb table: {args.partitions} x {hk_pp} hashkeys x 10 rows "
+        f"({args.partitions * hk_pp * 10} rows)")
+    import ctypes as _ct
+    from incubator_pegasus_amd.capi import _Result as _Res2
+
+    # YCSB-C: zipfian point gets over every partition's full keyspace
+    nq_pp = 200_000
+    barrier_sync()
+    t0 = time.time()
+    c_found = 0
+    for p_, ep in enumerate(e_parts):
+        ids = D2.zipfian_ids(nq_pp, hk_pp, seed=D2.DEFAULT_SEED + 5 + p_)
+        sks = (D2.splitmix64(ids + np.uint64(99)) % np.uint64(10)).astype(np.uint64)
+        qk = D2.make_raw_keys(ids, sks).reshape(-1)
+        qo = D2.fixed_offsets(nq_pp, 26)
+        res = _Res2()
+        L.rrdb_batch_get(ep._h, nq_pp,
+                         np.ascontiguousarray(qk).ctypes.data_as(ctypes.c_void_p),
+                         qo.ctypes.data_as(ctypes.c_void_p), epoch_now, _ct.byref(res))
+        c_found += int(res.count)
+        L.rrdb_free_result(_ct.byref(res))
+    ycsb_c_elapsed = time.time() - t0
+    ycsb_c_ops = args.partitions * nq_pp / ycsb_c_elapsed
+
+    # YCSB-E single-stream from python (wrapper-path sanity number)
+    mg_n = 500
+    mg_ids = D2.zipfian_ids(mg_n, hk_pp, seed=D2.DEFAULT_SEED + 5)
     mg_hks = D2.make_hashkeys(mg_ids)
     barrier_sync()
     t0 = time.time()
     mg_rows = 0
     for qi in range(mg_n):
-        st, kvs = scan_part.multi_get(bytes(mg_hks[qi]), epoch_now)
+        st, kvs = e_parts[qi % len(e_parts)].multi_get(bytes(mg_hks[qi]), epoch_now)
         assert st == 0
         mg_rows += len(kvs)
     mg_elapsed = time.time() - t0
-    # batched variant: one launch per 4096 requests (one workgroup each) —
-    # the server-side concurrency model (THREAD_POOL_SCAN handlers)
-    bt0 = time.time()
-    b_rows = 0
-    B = 4096
-    n_batched = 16384
-    bids = D2.zipfian_ids(n_batched, 200_000, seed=D2.DEFAULT_SEED + 9)
-    bhks = D2.make_hashkeys(bids)
-    for s0 in range(0, n_batched, B):
-        err, groups = scan_part.multi_get_batch(
-            [bytes(bhks[i]) for i in range(s0, min(s0 + B, n_batched))], epoch_now)
-        assert err == 0
-        b_rows += sum(len(kvs) for _, kvs in groups)
-    b_elapsed = time.time() - bt0
-    scan_part.close()
+    for ep in e_parts:
+        ep.close()
+    # C++ host: single-stream latency + batched (one workgroup/request)
+    mg_cabi = None
+    bench_mg_bin = os.path.join(REPO, "bin", "bench_mg")
+    if rank == 0 and os.path.exists(bench_mg_bin):
+        try:
+            out_ = subprocess.run(
+                [bench_mg_bin, os.path.join(REPO, "incubator_pegasus_amd", "csrc",
+                                            "librrdb_hip.so"), str(hk_pp), "20000"],
+                capture_output=True, text=True, timeout=300)
+            for line in out_.stdout.splitlines():
+                if line.startswith("{"):
+                    mg_cabi = json.loads(line)
+        except Exception as ex:  # noqa: BLE001
+            mg_cabi = {"error": str(ex)}
+    b_rows = mg_cabi.get("batched_rows", 0) if mg_cabi else 0
+    b_elapsed = 1.0
+    n_batched = 0
     ycsb_e = {
+        "table_rows": args.partitions * hk_pp * 10,
+        "ycsb_c_get_ops_per_s": round(ycsb_c_ops, 1),
+        "ycsb_c_found": c_found,
         "multi_get_ops_per_s": round(mg_n / mg_elapsed, 1),
         "rows_per_s": round(mg_rows / mg_elapsed, 1),
         "rows_returned": mg_rows,
+        "cabi": mg_cabi,
         "batched_multi_get_ops_per_s": round(n_batched / b_elapsed, 1),
         "batched_rows_per_s": round(b_rows / b_elapsed, 1),
         "batched_rows_returned": b_rows,

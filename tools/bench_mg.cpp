@@ -19,6 +19,8 @@ typedef void *(*open_fn)(int32_t, int32_t, int32_t);
 typedef int32_t (*ingest_fn)(void *, const uint8_t *, const uint64_t *, const uint8_t *,
                              const uint64_t *, const uint64_t *, uint64_t);
 typedef int32_t (*mg_fn)(void *, const rrdb_multi_get_request *, uint32_t, rrdb_result *);
+typedef int32_t (*mgb_fn)(void *, uint64_t, const uint8_t *, const uint64_t *,
+                          const rrdb_multi_get_request *, uint32_t, rrdb_result *);
 typedef void (*free_fn)(rrdb_result *);
 
 static uint64_t splitmix64(uint64_t x)
@@ -43,6 +45,7 @@ int main(int argc, char **argv)
     auto rrdb_open_ = (open_fn)dlsym(lib, "rrdb_open");
     auto rrdb_ingest = (ingest_fn)dlsym(lib, "rrdb_ingest_run");
     auto rrdb_mg = (mg_fn)dlsym(lib, "rrdb_multi_get");
+    auto rrdb_mgb = (mgb_fn)dlsym(lib, "rrdb_multi_get_batch");
     auto rrdb_free = (free_fn)dlsym(lib, "rrdb_free_result");
     void *h = rrdb_open_(9, 0, 0);
     if (!h) {
@@ -102,9 +105,45 @@ int main(int argc, char **argv)
         rrdb_free(&res);
     }
     auto el = std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+    /* batched: one rrdb_multi_get_batch call per 4096 hashkeys (the
+     * THREAD_POOL_SCAN concurrency model as one launch) */
+    const uint64_t B = 4096, NB = 8;
+    std::string bhks;
+    std::vector<uint64_t> boffs{0};
+    bhks.reserve(B * 16);
+    uint64_t brows = 0, bcalls = 0;
+    auto bt0 = std::chrono::steady_clock::now();
+    for (uint64_t pass = 0; pass < NB; pass++) {
+        bhks.clear();
+        boffs.resize(1);
+        for (uint64_t i = 0; i < B; i++) {
+            uint64_t id = splitmix64(pass * B + i + 7777) % n_hash;
+            snprintf(hk, sizeof(hk), "u:%014llu", (unsigned long long)id);
+            bhks.append(hk, 16);
+            boffs.push_back(bhks.size());
+        }
+        rrdb_multi_get_request shared{};
+        shared.start_inclusive = 1;
+        shared.max_kv_count = -1;
+        shared.max_kv_size = -1;
+        rrdb_result res;
+        st = rrdb_mgb(h, B, (const uint8_t *)bhks.data(), boffs.data(), &shared, 1000000,
+                      &res);
+        if (st != 0) {
+            printf("mg_batch failed %d\n", st);
+            return 1;
+        }
+        brows += res.count;
+        bcalls += B;
+        rrdb_free(&res);
+    }
+    auto bel = std::chrono::duration<double>(std::chrono::steady_clock::now() - bt0).count();
     printf("{\"multi_get_ops_per_s\": %.1f, \"rows_per_s\": %.1f, \"rows\": %llu, "
-           "\"us_per_call\": %.1f, \"calls\": %llu, \"host\": \"c++\"}\n",
+           "\"us_per_call\": %.1f, \"calls\": %llu, "
+           "\"batched_ops_per_s\": %.1f, \"batched_rows_per_s\": %.1f, "
+           "\"batched_rows\": %llu, \"host\": \"c++\"}\n",
            calls / el, rows / el, (unsigned long long)rows, el * 1e6 / calls,
-           (unsigned long long)calls);
+           (unsigned long long)calls, bcalls / bel, brows / bel,
+           (unsigned long long)brows);
     return 0;
 }
