@@ -614,6 +614,29 @@ struct HipEngine {
         rrdb_compact_stats st{};
     } pend;
     uint64_t *pend_sizes = nullptr; /* pinned [6]: output sizes d2h target */
+    CompactStatsDev *pend_stats_h = nullptr; /* pinned [8]: stats d2h target */
+    /* events of the last compaction pass, resolved lazily by phase_ms() so
+     * the keep_inputs finish can return without draining the emit */
+    hipEvent_t tev[6] = {};
+    bool tev_live = false, tev_have_emit = false;
+    void resolve_phase_events()
+    {
+        if (!tev_live)
+            return;
+        tev_live = false;
+        float ms;
+        HIP_OK(hipEventSynchronize(tev[1]));
+        HIP_OK(hipEventElapsedTime(&ms, tev[0], tev[1]));
+        phase_ms["compact_rank"] = ms;
+        phase_ms["compact_flags"] = 0.0;
+        if (tev_have_emit) {
+            HIP_OK(hipEventSynchronize(tev[4]));
+            HIP_OK(hipEventElapsedTime(&ms, tev[3], tev[4]));
+            phase_ms["compact_emit"] = ms;
+            HIP_OK(hipEventElapsedTime(&ms, tev[0], tev[4]));
+            phase_ms["compact_total"] = ms;
+        }
+    }
     /* serving-lane staging for the fused multi_get: one pinned H2D of
      * [start|stop|pattern], one pinned D2H of hdr + blob prefix (small-op
      * latency; pageable copies + two syncs dominated the 77us/call path) */
@@ -1172,6 +1195,9 @@ const char *rrdb_backend(void) { return "hip-gfx950"; }
 double rrdb_phase_ms(void *h, const char *phase)
 {
     auto *e = (HipEngine *)h;
+    std::lock_guard<std::mutex> g(e->mu);
+    e->activate();
+    e->resolve_phase_events();
     auto it = e->phase_ms.find(phase);
     return it == e->phase_ms.end() ? -1.0 : it->second;
 }
@@ -1237,6 +1263,11 @@ void rrdb_close(void *h)
         (void)hipHostFree(e->pinned);
     if (e->pend_sizes)
         (void)hipHostFree(e->pend_sizes);
+    if (e->pend_stats_h)
+        (void)hipHostFree(e->pend_stats_h);
+    for (auto &x : e->tev)
+        if (x)
+            (void)hipEventDestroy(x);
     if (e->mg_hin) {
         (void)hipHostFree(e->mg_hin);
         (void)hipFree(e->mg_din);
@@ -2487,9 +2518,12 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     uint64_t *d_hi = (uint64_t *)e->upload_tmp(hi.data(), R * 8);
     uint64_t *d_wp = (uint64_t *)e->upload_tmp(wprefix.data(), (R + 1) * 8);
     uint64_t *d_order = e->talloc<uint64_t>(total * 8);
-    hipEvent_t ev[6];
-    for (auto &x : ev)
-        HIP_OK(hipEventCreate(&x));
+    e->resolve_phase_events(); /* previous pass's timing, before reuse */
+    e->tev_have_emit = false;
+    if (!e->tev[0])
+        for (auto &x : e->tev)
+            HIP_OK(hipEventCreate(&x));
+    hipEvent_t *ev = e->tev;
 
     CompactParams cp{};
     cp.epoch_now = epoch_now;
@@ -2568,6 +2602,10 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     HIP_OK(hipEventRecord(ev[1], e->stream));
     }
     HIP_OK(hipEventRecord(ev[2], e->stream));
+    if (!e->pend_stats_h)
+        HIP_OK(hipHostMalloc((void **)&e->pend_stats_h, 8 * sizeof(CompactStatsDev)));
+    HIP_OK(hipMemcpyAsync(e->pend_stats_h, d_stats, 8 * sizeof(CompactStatsDev),
+                          hipMemcpyDeviceToHost, e->stream));
     launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);
     launch_psum(d_ksz, d_koffs, total, e->psum_scratch(total), e->stream);
     launch_psum(d_vsz, d_voffs, total, e->psum_scratch(total), e->stream);
@@ -2632,15 +2670,15 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
     uint64_t *d_kpos = e->pend.d_kpos, *d_koffs = e->pend.d_koffs, *d_voffs = e->pend.d_voffs;
     uint64_t *d_rank_of = e->pend.d_rank_of;
     CompactStatsDev *d_stats = e->pend.d_stats;
-    hipEvent_t ev[6];
-    for (int i = 0; i < 6; i++)
-        ev[i] = e->pend.ev[i];
+    (void)d_stats;
+    hipEvent_t *ev = e->tev;
+    /* wait for the submitted merge phase only (rank + prefix sums + the
+     * async size/stats reads — all of this stream's pending work) */
     HIP_OK(hipStreamSynchronize(e->stream));
     uint64_t *t = e->pend_sizes;
     uint64_t n_out = t[0] + t[1], kbytes = t[2] + t[3], vbytes = t[4] + t[5];
 
-    CompactStatsDev hsb[8];
-    HIP_OK(hipMemcpy(hsb, d_stats, sizeof(hsb), hipMemcpyDeviceToHost));
+    CompactStatsDev *hsb = e->pend_stats_h;
     CompactStatsDev hs{};
     for (int b = 0; b < 8; b++) {
         hs.shadowed += hsb[b].shadowed;
@@ -2711,22 +2749,17 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
                                 d_koffs, d_voffs, e->data_version, nr.keys, nr.vals, nr.koff,
                                 nr.voff, nr.sk, n_out, e->stream);
         HIP_OK(hipEventRecord(ev[4], e->stream));
-        HIP_OK(hipStreamSynchronize(e->stream));
-        float ms;
-        HIP_OK(hipEventElapsedTime(&ms, ev[3], ev[4]));
-        e->phase_ms["compact_emit"] = ms;
-        HIP_OK(hipEventElapsedTime(&ms, ev[0], ev[4]));
-        e->phase_ms["compact_total"] = ms;
+        e->tev_have_emit = true;
+        if (!keep_inputs) {
+            /* the swap below frees the input runs the emit reads: drain */
+            HIP_OK(hipStreamSynchronize(e->stream));
+        }
+        /* keep_inputs (repeatable-pass) mode returns with the emit still
+         * in flight on this engine's stream — later partitions' finishes
+         * overlap it; stream order protects this engine's arena reuse.
+         * Timing is resolved lazily by phase_ms(). */
     }
-    {
-        float ms;
-        HIP_OK(hipEventSynchronize(ev[2]));
-        HIP_OK(hipEventElapsedTime(&ms, ev[0], ev[1]));
-        e->phase_ms["compact_rank"] = ms; /* fused rank+filter */
-        e->phase_ms["compact_flags"] = 0.0;
-    }
-    for (auto &x : ev)
-        (void)hipEventDestroy(x);
+    e->tev_live = true;
     /* arena temporaries (d_order, flags, sums, keep_inputs outputs) are
      * reclaimed at the next scratch_reset */
     if (keep_inputs) {
