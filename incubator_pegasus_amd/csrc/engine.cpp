@@ -60,6 +60,7 @@ void launch_emit_rows(const DevRun *, const uint64_t *, const uint64_t *, uint64
                       const uint64_t *, const uint64_t *, const ScanParams &, uint8_t *,
                       uint8_t *, hipStream_t);
 void launch_multi_get_small(const DevRun *, int, const MgFusedArgs &, hipStream_t);
+void launch_multi_get_graph(const DevRun *, int, const uint8_t *, uint8_t *, hipStream_t);
 void launch_bloom_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
 void launch_bloom_pfx_build(const DevRun &, uint64_t *, uint64_t, hipStream_t);
 void launch_build_tails(const uint8_t *, uint64_t, uint64_t, uint64_t *, hipStream_t);
@@ -643,6 +644,11 @@ struct HipEngine {
     static constexpr uint64_t MG_IN_CAP = 64 << 10;
     static constexpr uint64_t MG_OUT_PREFIX = 16 << 10;
     uint8_t *mg_hin = nullptr, *mg_din = nullptr, *mg_hout = nullptr;
+    uint8_t *mg_dout = nullptr; /* persistent output buffer (graph-stable) */
+    hipGraphExec_t mg_graph = nullptr;
+    uint64_t mg_graph_gen = ~0ull; /* runs_gen the graph was captured at */
+    uint64_t runs_gen = 0;         /* bumped whenever d_runs is rebuilt */
+    bool mg_graph_enabled = true;  /* env "engine.mg_graph" */
     void mg_lane_init()
     {
         if (mg_hin)
@@ -650,6 +656,45 @@ struct HipEngine {
         HIP_OK(hipHostMalloc((void **)&mg_hin, MG_IN_CAP));
         HIP_OK(hipMalloc(&mg_din, MG_IN_CAP));
         HIP_OK(hipHostMalloc((void **)&mg_hout, 32 + MG_OUT_PREFIX));
+        HIP_OK(hipMalloc(&mg_dout, 32 + MG_BLOB_BYTES));
+    }
+    /* (re)capture the serving graph: H2D request slice -> fused kernel ->
+     * D2H hdr+prefix.  One hipGraphLaunch replaces 3 submissions. */
+    bool mg_graph_ready()
+    {
+        if (!mg_graph_enabled)
+            return false;
+        if (mg_graph && mg_graph_gen == runs_gen)
+            return true;
+        if (mg_graph) {
+            (void)hipGraphExecDestroy(mg_graph);
+            mg_graph = nullptr;
+        }
+        DevRun *dr = dev_runs();
+        int R = (int)runs.size();
+        hipGraph_t g = nullptr;
+        if (hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal) != hipSuccess)
+            return false;
+        bool ok = hipMemcpyAsync(mg_din, mg_hin, MG_GRAPH_IN, hipMemcpyHostToDevice,
+                                 stream) == hipSuccess;
+        launch_multi_get_graph(dr, R, mg_din, mg_dout, stream);
+        ok = ok && hipMemcpyAsync(mg_hout, mg_dout, 32 + MG_OUT_PREFIX,
+                                  hipMemcpyDeviceToHost, stream) == hipSuccess;
+        if (hipStreamEndCapture(stream, &g) != hipSuccess || !ok || !g) {
+            if (g)
+                (void)hipGraphDestroy(g);
+            mg_graph_enabled = false; /* capture unsupported: stay on the
+                                         plain path */
+            return false;
+        }
+        if (hipGraphInstantiate(&mg_graph, g, nullptr, nullptr, 0) != hipSuccess) {
+            (void)hipGraphDestroy(g);
+            mg_graph_enabled = false;
+            return false;
+        }
+        (void)hipGraphDestroy(g);
+        mg_graph_gen = runs_gen;
+        return true;
     }
     /* pending fused count scan (rrdb_scan_count_begin/finish): buffers are
      * plain hipMallocs so interleaved reads/compactions cannot reclaim them */
@@ -824,6 +869,7 @@ struct HipEngine {
             HIP_OK(hipMalloc(&d_runs, h.size() * sizeof(DevRun)));
             HIP_OK(hipMemcpy(d_runs, h.data(), h.size() * sizeof(DevRun), hipMemcpyHostToDevice));
             d_runs_dirty = false;
+            runs_gen++;
         }
         return d_runs;
     }
@@ -1268,10 +1314,13 @@ void rrdb_close(void *h)
     for (auto &x : e->tev)
         if (x)
             (void)hipEventDestroy(x);
+    if (e->mg_graph)
+        (void)hipGraphExecDestroy(e->mg_graph);
     if (e->mg_hin) {
         (void)hipHostFree(e->mg_hin);
         (void)hipFree(e->mg_din);
         (void)hipHostFree(e->mg_hout);
+        (void)hipFree(e->mg_dout);
     }
     if (e->d_ops)
         (void)hipFree(e->d_ops);
@@ -1340,6 +1389,13 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
             int s_ = atoi(v.c_str());
             if (s_ >= 4 && s_ <= 16)
                 e->bt_shift = s_;
+        } else if (k == "engine.mg_graph") {
+            e->mg_graph_enabled = (v != "off");
+            if (!e->mg_graph_enabled && e->mg_graph) {
+                e->activate();
+                (void)hipGraphExecDestroy(e->mg_graph);
+                e->mg_graph = nullptr;
+            }
         } else if (k == "engine.emit_mode") {
             e->emit_mode = (v == "input") ? 1 : (v == "rank" ? 0 : 2);
         }
@@ -2232,7 +2288,38 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
         e->mg_lane_init();
         MgFusedArgs fa{};
         uint64_t in_n = start.size() + stop_excl.size() + q->sort_key_filter_pattern.len;
-        if (in_n <= HipEngine::MG_IN_CAP) {
+        const uint8_t *d_blob = nullptr;
+        bool served = false;
+        if (sizeof(MgGraphHdr) + in_n <= MG_GRAPH_IN && e->mg_graph_ready()) {
+            /* captured-graph lane: fill the request slice, one graph launch */
+            MgGraphHdr *hh = (MgGraphHdr *)e->mg_hin;
+            hh->start_len = (uint32_t)start.size();
+            hh->stop_len = (uint32_t)stop_excl.size();
+            hh->sk_pat_len = (uint32_t)q->sort_key_filter_pattern.len;
+            hh->start_inclusive = start_inclusive;
+            hh->stop_inclusive = stop_inclusive;
+            hh->reverse = q->reverse;
+            hh->no_value = q->no_value;
+            hh->max_kv_count = max_kv_count;
+            hh->max_iteration_count = max_iteration_count;
+            hh->max_iteration_size = max_iter_size;
+            hh->sk_ft = q->sort_key_filter_type;
+            hh->epoch_now = epoch_now;
+            hh->data_version = e->data_version;
+            hh->hash_key_skip = 2 + q->hash_key.len;
+            uint8_t *pp = e->mg_hin + sizeof(MgGraphHdr);
+            memcpy(pp, start.data(), start.size());
+            pp += start.size();
+            memcpy(pp, stop_excl.data(), stop_excl.size());
+            pp += stop_excl.size();
+            if (q->sort_key_filter_pattern.len)
+                memcpy(pp, q->sort_key_filter_pattern.data, q->sort_key_filter_pattern.len);
+            HIP_OK(hipGraphLaunch(e->mg_graph, e->stream));
+            spin_sync(e->stream);
+            d_blob = e->mg_dout + 32;
+            served = true;
+        }
+        if (!served && in_n <= HipEngine::MG_IN_CAP) {
             uint64_t o = 0;
             memcpy(e->mg_hin + o, start.data(), start.size());
             fa.start = e->mg_din + o;
@@ -2246,7 +2333,7 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
             fa.sk_pat = e->mg_din + o;
             HIP_OK(hipMemcpyAsync(e->mg_din, e->mg_hin, in_n ? in_n : 1,
                                   hipMemcpyHostToDevice, e->stream));
-        } else { /* giant keys: pageable per-piece uploads */
+        } else if (!served) { /* giant keys: pageable per-piece uploads */
             fa.start = e->upload_tmp(start.data(), start.size());
             fa.stop = e->upload_tmp(stop_excl.data(), stop_excl.size());
             fa.sk_pat = e->upload_tmp(q->sort_key_filter_pattern.data,
@@ -2266,13 +2353,16 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
         fa.epoch_now = epoch_now;
         fa.data_version = e->data_version;
         fa.hash_key_skip = 2 + q->hash_key.len;
-        uint8_t *d_out = e->talloc<uint8_t>(32 + MG_BLOB_BYTES);
-        fa.out_hdr = (int64_t *)d_out;
-        fa.out_blob = d_out + 32;
-        launch_multi_get_small(e->dev_runs(), (int)e->runs.size(), fa, e->stream);
-        HIP_OK(hipMemcpyAsync(e->mg_hout, d_out, 32 + HipEngine::MG_OUT_PREFIX,
-                              hipMemcpyDeviceToHost, e->stream));
-        spin_sync(e->stream);
+        if (!served) {
+            uint8_t *d_out = e->talloc<uint8_t>(32 + MG_BLOB_BYTES);
+            fa.out_hdr = (int64_t *)d_out;
+            fa.out_blob = d_out + 32;
+            launch_multi_get_small(e->dev_runs(), (int)e->runs.size(), fa, e->stream);
+            HIP_OK(hipMemcpyAsync(e->mg_hout, d_out, 32 + HipEngine::MG_OUT_PREFIX,
+                                  hipMemcpyDeviceToHost, e->stream));
+            spin_sync(e->stream);
+            d_blob = fa.out_blob;
+        }
         const int64_t *hdr4 = (const int64_t *)e->mg_hout;
         if (hdr4[0] >= 0) {
             uint64_t m = (uint64_t)hdr4[0];
@@ -2286,7 +2376,7 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
                 if (blob_n <= HipEngine::MG_OUT_PREFIX) {
                     memcpy(hb, e->mg_hout + 32, blob_n); /* already on host */
                 } else {
-                    HIP_OK(hipMemcpyAsync(hb, fa.out_blob, blob_n, hipMemcpyDeviceToHost,
+                    HIP_OK(hipMemcpyAsync(hb, d_blob, blob_n, hipMemcpyDeviceToHost,
                                           e->stream));
                     HIP_OK(hipStreamSynchronize(e->stream));
                 }

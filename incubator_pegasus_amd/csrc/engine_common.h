@@ -154,6 +154,21 @@ struct MgFusedArgs {
 };
 #define MG_BLOB_BYTES (2 * MG_SCRATCH_BYTES + 2 * (MG_MAX_ROWS + 1) * 8)
 
+/* graph-captured serving lane: the per-call request is ONE pinned H2D of
+ * [MgGraphHdr][start][stop][pattern]; the captured graph replays
+ * H2D -> kernel -> D2H as a single launch (hipGraph), cutting the 3-4
+ * submission round-trips of the small-op path */
+#define MG_GRAPH_IN 4096
+struct MgGraphHdr {
+    uint32_t start_len, stop_len, sk_pat_len;
+    uint8_t start_inclusive, stop_inclusive, reverse, no_value;
+    uint32_t max_kv_count, max_iteration_count;
+    int64_t max_iteration_size;
+    int32_t sk_ft;
+    uint32_t epoch_now, data_version;
+    uint64_t hash_key_skip;
+};
+
 /* compact per-record disposition written by the filter kernel */
 struct CompactStatsDev {
     unsigned long long expired, filtered, tombstones, shadowed, output_records;

@@ -2645,6 +2645,40 @@ __global__ void __launch_bounds__(256) k_multi_get_small(const DevRun *runs, int
             a.out_blob, MG_BLOB_BYTES);
 }
 
+/* graph-capturable variant: all per-call state read from the d_in slice */
+__global__ void __launch_bounds__(256) k_multi_get_graph(const DevRun *runs, int R,
+                                                         const uint8_t *d_in, uint8_t *d_out)
+{
+    const MgGraphHdr *h = (const MgGraphHdr *)d_in;
+    MgFusedArgs a{};
+    const uint8_t *p = d_in + sizeof(MgGraphHdr);
+    a.start = p;
+    a.start_len = h->start_len;
+    a.stop = p + h->start_len;
+    a.stop_len = h->stop_len;
+    a.sk_pat = p + h->start_len + h->stop_len;
+    a.sk_pat_len = h->sk_pat_len;
+    a.start_inclusive = h->start_inclusive;
+    a.stop_inclusive = h->stop_inclusive;
+    a.reverse = h->reverse;
+    a.no_value = h->no_value;
+    a.max_kv_count = h->max_kv_count;
+    a.max_iteration_count = h->max_iteration_count;
+    a.max_iteration_size = h->max_iteration_size;
+    a.sk_ft = h->sk_ft;
+    a.epoch_now = h->epoch_now;
+    a.data_version = h->data_version;
+    a.hash_key_skip = h->hash_key_skip;
+    mg_core(runs, R, a, a.start, a.start_len, a.stop, a.stop_len, a.hash_key_skip,
+            (int64_t *)d_out, d_out + 32, MG_BLOB_BYTES);
+}
+
+void launch_multi_get_graph(const DevRun *runs, int R, const uint8_t *d_in, uint8_t *d_out,
+                            hipStream_t s)
+{
+    k_multi_get_graph<<<1, 256, 0, s>>>(runs, R, d_in, d_out);
+}
+
 /* batched full-range multi_get: one workgroup per request (hashkey); start =
  * [u16 len][hk], stop = pegasus_generate_next_blob(hk)
  * (pegasus_key_schema.h:64-81) built in-kernel */

@@ -155,6 +155,47 @@ def test_grp_rank_many_runs_uses_fallback_mode(oracle_lib, hip_lib):
         g.close()
 
 
+def test_mg_graph_lane_parity(oracle_lib, hip_lib):
+    """The hipGraph-captured serving lane must return byte-identical
+    multi_get results to the plain path and the oracle, across run-set
+    changes (graph recapture) and both toggle states."""
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        seq = 1
+        for run in range(3):
+            recs = []
+            for i in range(run, 120, 3):
+                for sk in (b"s0", b"s1", b"s2"):
+                    recs.append((D.generate_key(b"gk%03d" % i, sk),
+                                 D.encode_value(b"r%dv%d" % (run, i), 0, seq, 1), seq, 0))
+                    seq += 1
+            recs.sort(key=lambda r: r[0])
+            o.ingest_run(recs)
+            g.ingest_run(recs)
+        cases = [dict(), dict(reverse=True), dict(max_kv_count=2),
+                 dict(start_sortkey=b"s1"), dict(no_value=True),
+                 dict(sort_key_filter_type=2, sort_key_filter_pattern=b"s2")]
+        for mode in ("on", "off"):
+            g.set_envs({"engine.mg_graph": mode})
+            for i in (0, 1, 2, 55, 119, 999):
+                hk = b"gk%03d" % i
+                for kw in cases:
+                    assert o.multi_get(hk, NOW, **kw) == g.multi_get(hk, NOW, **kw), \
+                        (mode, hk, kw)
+        # run-set change invalidates and recaptures the graph
+        g.set_envs({"engine.mg_graph": "on"})
+        o.put(b"gk000", b"s9", b"new")
+        g.put(b"gk000", b"s9", b"new")
+        assert o.multi_get(b"gk000", NOW) == g.multi_get(b"gk000", NOW)
+        o.manual_compact(NOW)
+        g.manual_compact(NOW)
+        assert o.multi_get(b"gk055", NOW) == g.multi_get(b"gk055", NOW)
+    finally:
+        o.close()
+        g.close()
+
+
 def test_prefix_bloom_parity_and_toggle(oracle_lib, hip_lib):
     """§8(f)3 second half: the hashkey-prefix bloom is lossless — multi_get
     and sortkey_count agree with the oracle for present and ABSENT hashkeys,
