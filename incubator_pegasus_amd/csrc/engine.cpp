@@ -646,6 +646,9 @@ struct HipEngine {
     uint8_t *mg_hin = nullptr, *mg_din = nullptr, *mg_hout = nullptr;
     uint8_t *mg_dout = nullptr; /* persistent output buffer (graph-stable) */
     hipGraphExec_t mg_graph = nullptr;
+    hipStream_t mg_capture_stream = nullptr; /* idle stream used ONLY for
+        capture: capturing e->stream while earlier async work (e.g. a pending
+        compaction) is in flight aborts inside the ROCm runtime */
     uint64_t mg_graph_gen = ~0ull; /* runs_gen the graph was captured at */
     uint64_t runs_gen = 0;         /* bumped whenever d_runs is rebuilt */
     bool mg_graph_enabled = true;  /* env "engine.mg_graph" */
@@ -672,15 +675,20 @@ struct HipEngine {
         }
         DevRun *dr = dev_runs();
         int R = (int)runs.size();
+        if (!mg_capture_stream && hipStreamCreate(&mg_capture_stream) != hipSuccess) {
+            mg_graph_enabled = false;
+            return false;
+        }
+        hipStream_t cs = mg_capture_stream;
         hipGraph_t g = nullptr;
-        if (hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal) != hipSuccess)
+        if (hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal) != hipSuccess)
             return false;
         bool ok = hipMemcpyAsync(mg_din, mg_hin, MG_GRAPH_IN, hipMemcpyHostToDevice,
-                                 stream) == hipSuccess;
-        launch_multi_get_graph(dr, R, mg_din, mg_dout, stream);
+                                 cs) == hipSuccess;
+        launch_multi_get_graph(dr, R, mg_din, mg_dout, cs);
         ok = ok && hipMemcpyAsync(mg_hout, mg_dout, 32 + MG_OUT_PREFIX,
-                                  hipMemcpyDeviceToHost, stream) == hipSuccess;
-        if (hipStreamEndCapture(stream, &g) != hipSuccess || !ok || !g) {
+                                  hipMemcpyDeviceToHost, cs) == hipSuccess;
+        if (hipStreamEndCapture(cs, &g) != hipSuccess || !ok || !g) {
             if (g)
                 (void)hipGraphDestroy(g);
             mg_graph_enabled = false; /* capture unsupported: stay on the
@@ -1316,6 +1324,8 @@ void rrdb_close(void *h)
             (void)hipEventDestroy(x);
     if (e->mg_graph)
         (void)hipGraphExecDestroy(e->mg_graph);
+    if (e->mg_capture_stream)
+        (void)hipStreamDestroy(e->mg_capture_stream);
     if (e->mg_hin) {
         (void)hipHostFree(e->mg_hin);
         (void)hipFree(e->mg_din);
