@@ -667,13 +667,16 @@ struct HipEngine {
     {
         if (!mg_graph_enabled)
             return false;
+        /* refresh the device run table FIRST: it bumps runs_gen when the
+         * run set changed, and a stale graph would replay against freed
+         * run buffers (GPU aperture fault) */
+        DevRun *dr = dev_runs();
         if (mg_graph && mg_graph_gen == runs_gen)
             return true;
         if (mg_graph) {
             (void)hipGraphExecDestroy(mg_graph);
             mg_graph = nullptr;
         }
-        DevRun *dr = dev_runs();
         int R = (int)runs.size();
         if (!mg_capture_stream && hipStreamCreate(&mg_capture_stream) != hipSuccess) {
             mg_graph_enabled = false;
