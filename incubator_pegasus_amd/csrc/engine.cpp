@@ -108,7 +108,7 @@ void launch_emit_compact_chunked(const DevRun *, const uint64_t *, uint64_t, con
                                  uint64_t, uint64_t, uint64_t *, uint64_t *, uint32_t *,
                                  uint32_t *, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
                                  uint64_t *, uint64_t *, uint64_t *,
-                                 hipStream_t);
+                                 uint64_t, uint64_t, hipStream_t);
 void launch_emit_compact(const DevRun *, const uint64_t *, uint64_t, const uint64_t *,
                          const uint8_t *, const uint32_t *, const uint64_t *, const uint64_t *,
                          const uint64_t *, uint32_t, uint8_t *, uint8_t *, uint64_t *, uint64_t *,
@@ -611,6 +611,7 @@ struct HipEngine {
         uint64_t *d_kpos = nullptr, *d_koffs = nullptr, *d_voffs = nullptr;
         uint64_t *d_rank_of = nullptr;
         CompactStatsDev *d_stats = nullptr;
+        uint64_t fk = 0, fv = 0; /* all-fixed-stride emit (ksz/vsz skipped) */
         hipEvent_t ev[6] = {};
         rrdb_compact_stats st{};
     } pend;
@@ -2640,14 +2641,27 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     cp.rules = e->d_rules;
     cp.pats = e->d_pats;
 
+    /* all-fixed-stride fast mode: every run shares one key and one value
+     * stride (the hashkey-table schema), so output offsets derive from the
+     * keep count — the ksz/vsz arrays and two of the three full-length
+     * prefix sums are skipped entirely */
+    uint64_t ffk = e->runs.empty() ? 0 : e->runs[0].fixed_klen;
+    uint64_t ffv = e->runs.empty() ? 0 : e->runs[0].fixed_vlen;
+    for (auto &rr : e->runs) {
+        if (rr.fixed_klen != ffk)
+            ffk = 0;
+        if (rr.fixed_vlen != ffv)
+            ffv = 0;
+    }
+    bool fixed_emit = ffk > 0 && ffv > 0 && e->emit_mode == 2;
     uint8_t *d_changed = e->talloc<uint8_t>(total);
     uint32_t *d_new_expire = e->talloc<uint32_t>(total * 4);
-    uint64_t *d_ksz = e->talloc<uint64_t>(total * 8);
-    uint64_t *d_vsz = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_ksz = fixed_emit ? nullptr : e->talloc<uint64_t>(total * 8);
+    uint64_t *d_vsz = fixed_emit ? nullptr : e->talloc<uint64_t>(total * 8);
     uint64_t *d_keepw = e->talloc<uint64_t>(total * 8);
     uint64_t *d_kpos = e->talloc<uint64_t>(total * 8);
-    uint64_t *d_koffs = e->talloc<uint64_t>(total * 8);
-    uint64_t *d_voffs = e->talloc<uint64_t>(total * 8);
+    uint64_t *d_koffs = fixed_emit ? nullptr : e->talloc<uint64_t>(total * 8);
+    uint64_t *d_voffs = fixed_emit ? nullptr : e->talloc<uint64_t>(total * 8);
     uint64_t *d_rank_of = e->emit_mode == 1 ? e->talloc<uint64_t>(total * 8) : nullptr;
     /* 8 stat banks: the group rank flushes per-block tallies to bank
      * blockIdx&7; legacy kernels add to bank 0; finish sums all banks */
@@ -2710,17 +2724,22 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     HIP_OK(hipMemcpyAsync(e->pend_stats_h, d_stats, 8 * sizeof(CompactStatsDev),
                           hipMemcpyDeviceToHost, e->stream));
     launch_psum(d_keepw, d_kpos, total, e->psum_scratch(total), e->stream);
-    launch_psum(d_ksz, d_koffs, total, e->psum_scratch(total), e->stream);
-    launch_psum(d_vsz, d_voffs, total, e->psum_scratch(total), e->stream);
+    if (!fixed_emit) {
+        launch_psum(d_ksz, d_koffs, total, e->psum_scratch(total), e->stream);
+        launch_psum(d_vsz, d_voffs, total, e->psum_scratch(total), e->stream);
+    }
     if (!e->pend_sizes)
         HIP_OK(hipHostMalloc((void **)&e->pend_sizes, 6 * 8));
     uint64_t *t = e->pend_sizes;
+    memset(t, 0, 6 * 8);
     HIP_OK(hipMemcpyAsync(&t[0], d_kpos + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
     HIP_OK(hipMemcpyAsync(&t[1], d_keepw + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
-    HIP_OK(hipMemcpyAsync(&t[2], d_koffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
-    HIP_OK(hipMemcpyAsync(&t[3], d_ksz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
-    HIP_OK(hipMemcpyAsync(&t[4], d_voffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
-    HIP_OK(hipMemcpyAsync(&t[5], d_vsz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    if (!fixed_emit) {
+        HIP_OK(hipMemcpyAsync(&t[2], d_koffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&t[3], d_ksz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&t[4], d_voffs + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+        HIP_OK(hipMemcpyAsync(&t[5], d_vsz + total - 1, 8, hipMemcpyDeviceToHost, e->stream));
+    }
     e->pend = {};
     e->pend.active = true;
     e->pend.keep_inputs = opts && opts->keep_inputs;
@@ -2737,6 +2756,8 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     e->pend.d_voffs = d_voffs;
     e->pend.d_rank_of = d_rank_of;
     e->pend.d_stats = d_stats;
+    e->pend.fk = fixed_emit ? ffk : 0;
+    e->pend.fv = fixed_emit ? ffv : 0;
     for (int i = 0; i < 6; i++)
         e->pend.ev[i] = ev[i];
     e->pend.st = st;
@@ -2779,7 +2800,9 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
      * async size/stats reads — all of this stream's pending work) */
     HIP_OK(hipStreamSynchronize(e->stream));
     uint64_t *t = e->pend_sizes;
-    uint64_t n_out = t[0] + t[1], kbytes = t[2] + t[3], vbytes = t[4] + t[5];
+    uint64_t n_out = t[0] + t[1];
+    uint64_t kbytes = e->pend.fk ? n_out * e->pend.fk : t[2] + t[3];
+    uint64_t vbytes = e->pend.fk ? n_out * e->pend.fv : t[4] + t[5];
 
     CompactStatsDev *hsb = e->pend_stats_h;
     CompactStatsDev hs{};
@@ -2828,7 +2851,7 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
             HIP_OK(hipMalloc(&nr.sk, n_out * 8));
         }
         HIP_OK(hipEventRecord(ev[3], e->stream));
-        if (e->emit_mode == 2) {
+        if (e->emit_mode == 2 || e->pend.fk) {
             uint64_t *d_row_ksrc = e->talloc<uint64_t>(n_out * 8);
             uint64_t *d_row_vsrc = e->talloc<uint64_t>(n_out * 8);
             uint32_t *d_row_patch = e->talloc<uint32_t>(n_out * 4);
@@ -2841,7 +2864,8 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
                                         d_kpos, d_koffs, d_voffs, e->data_version, n_out, kbytes,
                                         vbytes, d_row_ksrc, d_row_vsrc, d_row_patch,
                                         d_row_expire, nr.keys, nr.vals, nr.koff, nr.voff, nr.sk,
-                                        d_kanchor, d_vanchor, e->stream);
+                                        d_kanchor, d_vanchor, e->pend.fk, e->pend.fv,
+                                        e->stream);
         } else if (e->emit_mode == 1)
             launch_emit_compact_inmajor(dr, R, d_wp, total, d_rank_of, d_keepw, d_changed,
                                         d_new_expire, d_kpos, d_koffs, d_voffs,

@@ -1188,8 +1188,10 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
             keepw[rank] = (disp == D_KEEP) ? 1 : 0;
             changed[rank] = ch;
             new_expire[rank] = nts;
-            ksz[rank] = okl;
-            vsz[rank] = ovl;
+            if (ksz)
+                ksz[rank] = okl;
+            if (vsz)
+                vsz[rank] = ovl;
             if (rank_of)
                 rank_of[t] = rank;
         }
@@ -1354,8 +1356,10 @@ __global__ void __launch_bounds__(BLOCK) k_rank_compact_ldst(
         keepw[rank] = (disp == D_KEEP) ? 1 : 0;
         changed[rank] = ch;
         new_expire[rank] = nts;
-        ksz[rank] = okl;
-        vsz[rank] = ovl;
+        if (ksz)
+            ksz[rank] = okl;
+        if (vsz)
+            vsz[rank] = ovl;
         if (rank_of)
             rank_of[t] = rank;
     }
@@ -1577,8 +1581,10 @@ __device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i,
     keepw[rank] = (disp == D_KEEP) ? 1 : 0;
     changed[rank] = ch;
     new_expire[rank] = nts;
-    ksz[rank] = okl;
-    vsz[rank] = ovl;
+    if (ksz)
+        ksz[rank] = okl;
+    if (vsz)
+        vsz[rank] = ovl;
     return disp;
 }
 
@@ -1915,13 +1921,17 @@ __global__ void k_emit_compact(const DevRun *runs, const uint64_t *order, uint64
 /* chunked emit (the fast path, measured 2.8 TB/s on the probe vs 1.1 for
  * wave-per-record): one scatter pass builds output-row metadata, then pure
  * 16B-chunk-parallel copies with an offset binary search per chunk. */
+/* fk/fv nonzero = all-fixed-stride mode: row offsets and source addresses
+ * derive from the keep count, so the per-rank ksz/vsz arrays and two of the
+ * three full-length prefix sums are never built */
 __global__ void k_compact_gather_meta(const DevRun *runs, const uint64_t *order, uint64_t m,
                                       const uint64_t *keepw, const uint8_t *changed,
                                       const uint32_t *new_expire, const uint64_t *kpos,
                                       const uint64_t *koffs, const uint64_t *voffs,
                                       uint64_t *row_koff, uint64_t *row_voff, uint64_t *row_ksrc,
                                       uint64_t *row_vsrc, uint32_t *row_patch, uint64_t *osk,
-                                      uint64_t n_out, uint64_t kbytes, uint64_t vbytes)
+                                      uint64_t n_out, uint64_t kbytes, uint64_t vbytes,
+                                      uint64_t fk, uint64_t fv)
 {
     for (uint64_t p = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; p < m;
          p += gridDim.x * (uint64_t)blockDim.x) {
@@ -1931,10 +1941,10 @@ __global__ void k_compact_gather_meta(const DevRun *runs, const uint64_t *order,
         uint64_t id = order[p];
         const DevRun &r = runs[id >> 40];
         uint64_t i = id & 0xFFFFFFFFFFull;
-        row_koff[o] = koffs[p];
-        row_voff[o] = voffs[p];
-        row_ksrc[o] = (uint64_t)(r.keys + r.koff[i]);
-        row_vsrc[o] = (uint64_t)(r.vals + r.voff[i]);
+        row_koff[o] = fk ? o * fk : koffs[p];
+        row_voff[o] = fk ? o * fv : voffs[p];
+        row_ksrc[o] = (uint64_t)(fk ? r.keys + i * fk : r.keys + r.koff[i]);
+        row_vsrc[o] = (uint64_t)(fk ? r.vals + i * fv : r.vals + r.voff[i]);
         row_patch[o] = changed[p] ? 1 : 0;
         if (changed[p])
             row_patch[o] |= 2; /* marker; value in new_expire via p -> copy */
@@ -2264,11 +2274,11 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
                                  uint8_t *d_vout, uint64_t *d_okoff /* also the key row offsets */,
                                  uint64_t *d_ovoff /* also the value row offsets */,
                                  uint64_t *d_osk, uint64_t *d_kanchor, uint64_t *d_vanchor,
-                                 hipStream_t s)
+                                 uint64_t fk, uint64_t fv, hipStream_t s)
 {
     k_compact_gather_meta<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(
         d_runs, d_order, m, d_keepw, d_changed, d_new_expire, d_kpos, d_koffs, d_voffs, d_okoff,
-        d_ovoff, d_row_ksrc, d_row_vsrc, d_row_patch, d_osk, n_out, kbytes, vbytes);
+        d_ovoff, d_row_ksrc, d_row_vsrc, d_row_patch, d_osk, n_out, kbytes, vbytes, fk, fv);
     k_compact_gather_expire<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_keepw, d_kpos, d_changed,
                                                                  d_new_expire, m, d_row_expire);
     uint64_t kanch = ((kbytes + 15) >> 4 >> 6) + 1, vanch = ((vbytes + 15) >> 4 >> 6) + 1;
@@ -2922,8 +2932,10 @@ __global__ void __launch_bounds__(LRK_BLK, 1) k_rank_compact_lds(
         keepw[rank] = (disp == D_KEEP) ? 1 : 0;
         changed[rank] = ch;
         new_expire[rank] = nts;
-        ksz[rank] = okl;
-        vsz[rank] = ovl;
+        if (ksz)
+            ksz[rank] = okl;
+        if (vsz)
+            vsz[rank] = ovl;
     }
     /* wave-aggregated stats */
     int lane = tid % WAVE;
