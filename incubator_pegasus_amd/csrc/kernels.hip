@@ -1015,7 +1015,8 @@ enum { D_NONE = 0, D_KEEP, D_SHADOWED, D_TOMBSTONE, D_EXPIRED, D_FILTERED };
  * validation, value bytes never (no rule inspects value data). */
 __device__ static int dev_disposition(const DevRun &r, uint64_t i, const CompactParams &cp,
                                       int shadow, uint8_t *changed_out, uint32_t *new_ts_out,
-                                      uint64_t *kl_out, uint64_t *vl_out)
+                                      uint64_t *kl_out, uint64_t *vl_out,
+                                      const uint64_t *mword = nullptr)
 {
     *changed_out = 0;
     *new_ts_out = 0;
@@ -1024,8 +1025,8 @@ __device__ static int dev_disposition(const DevRun &r, uint64_t i, const Compact
     if (shadow)
         return D_SHADOWED; /* newest-wins, decided during ranking */
     uint32_t expire_ts;
-    if (r.meta) {
-        uint64_t m = r.meta[i];
+    if (mword || r.meta) {
+        uint64_t m = mword ? *mword : r.meta[i];
         if (m & 1)
             return D_TOMBSTONE;
         expire_ts = (uint32_t)(m >> 32);
@@ -1565,10 +1566,20 @@ __device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i,
                                           const ScanParams &sp, uint64_t *order,
                                           uint64_t *keepw, uint8_t *changed,
                                           uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz,
-                                          uint8_t *shadowed)
+                                          uint8_t *shadowed, const uint64_t *smeta)
 {
-    if (MODE == 2)
-        return (!shadow && dev_count_ok(runs[q], i, sp)) ? D_KEEP : D_NONE;
+    if (MODE == 2) {
+        if (shadow)
+            return D_NONE;
+        if (smeta) { /* staged meta: LDS fast path for the common case */
+            uint64_t m = *smeta;
+            if ((m & 1) || dev_ts_expired(sp.epoch_now, (uint32_t)(m >> 32)))
+                return D_NONE;
+            if (!sp.validate_hash && sp.hk_ft == 0 && sp.sk_ft == 0)
+                return D_KEEP;
+        }
+        return dev_count_ok(runs[q], i, sp) ? D_KEEP : D_NONE;
+    }
     order[rank] = ((uint64_t)q << 40) | i;
     if (MODE == 1) {
         shadowed[rank] = (uint8_t)shadow;
@@ -1577,7 +1588,7 @@ __device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i,
     uint8_t ch;
     uint32_t nts;
     uint64_t okl, ovl;
-    int disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl);
+    int disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl, smeta);
     keepw[rank] = (disp == D_KEEP) ? 1 : 0;
     changed[rank] = ch;
     new_expire[rank] = nts;
@@ -1604,12 +1615,13 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
 {
     __shared__ uint64_t s_ta[GRP_CAP], s_tb[GRP_CAP]; /* tails ping-pong */
     __shared__ uint16_t s_oa[GRP_CAP], s_ob[GRP_CAP]; /* (q<<12)|segpos */
+    __shared__ uint64_t s_meta[GRP_CAP]; /* staged disposition column (MODE!=1) */
     __shared__ uint64_t s_a0[LDST_MAXR], s_seglen[LDST_MAXR], s_segoff[LDST_MAXR + 1];
     __shared__ uint64_t s_btail[LDST_MAXR];
     __shared__ uint64_t s_loff[LDST_MAXR + 1]; /* current merge-list offsets */
     __shared__ uint64_t s_base, s_gsize;
     __shared__ uint32_t s_bmask;
-    __shared__ int s_nl;
+    __shared__ int s_nl, s_allmeta;
 
     GrpTally tally;
     for (uint64_t g = blockIdx.x; g < n_groups; g += gridDim.x) {
@@ -1630,17 +1642,21 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
         }
         if (threadIdx.x == 0) {
             uint64_t t = 0, b = 0;
+            int am = 1;
             for (int q = 0; q < R; q++) {
                 s_segoff[q] = t;
                 s_loff[q] = t;
                 t += s_seglen[q];
                 b += s_a0[q] - lo[q];
+                if (!runs[q].meta)
+                    am = 0;
             }
             s_segoff[R] = t;
             s_loff[R] = t;
             s_gsize = t;
             s_base = b;
             s_nl = R;
+            s_allmeta = am;
         }
         __syncthreads();
         uint64_t gsize = s_gsize;
@@ -1650,9 +1666,13 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
             for (int q = 0; q < R; q++) {
                 uint64_t len = s_seglen[q], off = s_segoff[q], a0 = s_a0[q];
                 const uint64_t *tq = runs[q].tails;
+                const uint64_t *mq =
+                    (MODE != 1 && s_allmeta) ? runs[q].meta : nullptr;
                 for (uint64_t j = threadIdx.x; j < len; j += blockDim.x) {
                     s_ta[off + j] = tq[a0 + j];
                     s_oa[off + j] = (uint16_t)((q << 12) | (uint32_t)j);
+                    if (mq)
+                        s_meta[off + j] = mq[a0 + j];
                 }
             }
             __syncthreads();
@@ -1739,8 +1759,11 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
                             }
                     }
                     uint64_t rank = s_base + p;
+                    const uint64_t *sm = (MODE != 1 && s_allmeta)
+                                             ? &s_meta[s_segoff[q] + (org & 0xFFF)]
+                                             : nullptr;
                     disp = grp_epilogue<MODE>(runs, q, i, shadow, rank, cp, sp, order, keepw,
-                                              changed, new_expire, ksz, vsz, shadowed);
+                                              changed, new_expire, ksz, vsz, shadowed, sm);
                 }
                 if (MODE != 1)
                     tally.add(disp);
@@ -1794,7 +1817,7 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
                     }
                     uint64_t rank = s_base + lrank;
                     disp = grp_epilogue<MODE>(runs, q, i, shadow, rank, cp, sp, order, keepw,
-                                              changed, new_expire, ksz, vsz, shadowed);
+                                              changed, new_expire, ksz, vsz, shadowed, nullptr);
                 }
                 if (MODE != 1)
                     tally.add(disp);
