@@ -1615,7 +1615,9 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
 {
     __shared__ uint64_t s_ta[GRP_CAP], s_tb[GRP_CAP]; /* tails ping-pong */
     __shared__ uint16_t s_oa[GRP_CAP], s_ob[GRP_CAP]; /* (q<<12)|segpos */
-    __shared__ uint64_t s_meta[GRP_CAP]; /* staged disposition column (MODE==2) */
+    /* staged disposition column — only the count mode pays its LDS */
+    constexpr int MCAP = (MODE == 2) ? GRP_CAP : 1;
+    __shared__ uint64_t s_meta[MCAP];
     __shared__ uint64_t s_a0[LDST_MAXR], s_seglen[LDST_MAXR], s_segoff[LDST_MAXR + 1];
     __shared__ uint64_t s_btail[LDST_MAXR];
     __shared__ uint64_t s_loff[LDST_MAXR + 1]; /* current merge-list offsets */
