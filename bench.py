@@ -97,8 +97,8 @@ def rules_profile_envs(epoch_now):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=3)
-    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=25)
+    ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--partitions", type=int, default=16)     # per GPU (configs[1])
     ap.add_argument("--keys", type=int, default=50_000_000)   # per GPU total
     ap.add_argument("--runs", type=int, default=8)

@@ -1,6 +1,16 @@
 """Limiter caps, env knobs and value-schema-version coverage against the
 oracle (reference: range_read_limiter.h:37-103 + pegasus_server_impl_init.cpp
 defaults; value schemas v0/v1/v2)."""
+
+# Documented approximations:
+#  - the 30s scan time budget (range_read_limiter.h:56-79) is enforced at
+#    BATCH granularity: an over-budget batch returns kIncomplete between
+#    batches, but a single long batch is not interrupted mid-kernel.  The
+#    deterministic count/size limits below are exact; the time budget is
+#    not pinned by these tests.
+#  - parked scanner contexts expire after 300s of the caller's epoch clock
+#    (in-engine since round 2; reference uses a wall-clock delayed task).
+
 import pytest
 
 from incubator_pegasus_amd import data as D
