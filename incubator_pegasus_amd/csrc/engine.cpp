@@ -612,6 +612,7 @@ struct HipEngine {
         uint64_t *d_rank_of = nullptr;
         CompactStatsDev *d_stats = nullptr;
         uint64_t fk = 0, fv = 0; /* all-fixed-stride emit (ksz/vsz skipped) */
+        bool no_rewrite = false;  /* changed/new_expire arrays skipped */
         hipEvent_t ev[6] = {};
         rrdb_compact_stats st{};
     } pend;
@@ -2654,8 +2655,11 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
             ffv = 0;
     }
     bool fixed_emit = ffk > 0 && ffv > 0 && e->emit_mode == 2;
-    uint8_t *d_changed = e->talloc<uint8_t>(total);
-    uint32_t *d_new_expire = e->talloc<uint32_t>(total * 4);
+    /* no default-TTL and no user ops => the filter can never rewrite a
+     * value: skip the changed/new_expire arrays and the patch pass */
+    bool no_rewrite = e->default_ttl == 0 && e->n_ops == 0 && e->emit_mode == 2;
+    uint8_t *d_changed = no_rewrite ? nullptr : e->talloc<uint8_t>(total);
+    uint32_t *d_new_expire = no_rewrite ? nullptr : e->talloc<uint32_t>(total * 4);
     uint64_t *d_ksz = fixed_emit ? nullptr : e->talloc<uint64_t>(total * 8);
     uint64_t *d_vsz = fixed_emit ? nullptr : e->talloc<uint64_t>(total * 8);
     uint64_t *d_keepw = e->talloc<uint64_t>(total * 8);
@@ -2758,6 +2762,7 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
     e->pend.d_stats = d_stats;
     e->pend.fk = fixed_emit ? ffk : 0;
     e->pend.fv = fixed_emit ? ffv : 0;
+    e->pend.no_rewrite = no_rewrite;
     for (int i = 0; i < 6; i++)
         e->pend.ev[i] = ev[i];
     e->pend.st = st;
@@ -2851,7 +2856,7 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
             HIP_OK(hipMalloc(&nr.sk, n_out * 8));
         }
         HIP_OK(hipEventRecord(ev[3], e->stream));
-        if (e->emit_mode == 2 || e->pend.fk) {
+        if (e->emit_mode == 2 || e->pend.fk || e->pend.no_rewrite) {
             uint64_t *d_row_ksrc = e->talloc<uint64_t>(n_out * 8);
             uint64_t *d_row_vsrc = e->talloc<uint64_t>(n_out * 8);
             uint32_t *d_row_patch = e->talloc<uint32_t>(n_out * 4);

@@ -1187,8 +1187,10 @@ __global__ void k_rank_compact(const DevRun *runs, int R, const uint64_t *lo, co
             disp = dev_disposition(runs[r], i, cp, shadow, &ch, &nts, &okl, &ovl);
             order[rank] = ((uint64_t)r << 40) | i;
             keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-            changed[rank] = ch;
-            new_expire[rank] = nts;
+            if (changed)
+                changed[rank] = ch;
+            if (new_expire)
+                new_expire[rank] = nts;
             if (ksz)
                 ksz[rank] = okl;
             if (vsz)
@@ -1355,8 +1357,10 @@ __global__ void __launch_bounds__(BLOCK) k_rank_compact_ldst(
         disp = dev_disposition(runs[r], i, cp, shadow, &ch, &nts, &okl, &ovl);
         order[rank] = ((uint64_t)r << 40) | i;
         keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-        changed[rank] = ch;
-        new_expire[rank] = nts;
+        if (changed)
+            changed[rank] = ch;
+        if (new_expire)
+            new_expire[rank] = nts;
         if (ksz)
             ksz[rank] = okl;
         if (vsz)
@@ -1590,8 +1594,10 @@ __device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i,
     uint64_t okl, ovl;
     int disp = dev_disposition(runs[q], i, cp, shadow, &ch, &nts, &okl, &ovl, smeta);
     keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-    changed[rank] = ch;
-    new_expire[rank] = nts;
+    if (changed)
+        changed[rank] = ch;
+    if (new_expire)
+        new_expire[rank] = nts;
     if (ksz)
         ksz[rank] = okl;
     if (vsz)
@@ -1973,9 +1979,7 @@ __global__ void k_compact_gather_meta(const DevRun *runs, const uint64_t *order,
         row_voff[o] = fk ? o * fv : voffs[p];
         row_ksrc[o] = (uint64_t)(fk ? r.keys + i * fk : r.keys + r.koff[i]);
         row_vsrc[o] = (uint64_t)(fk ? r.vals + i * fv : r.vals + r.voff[i]);
-        row_patch[o] = changed[p] ? 1 : 0;
-        if (changed[p])
-            row_patch[o] |= 2; /* marker; value in new_expire via p -> copy */
+        row_patch[o] = (changed && changed[p]) ? 3 : 0;
         osk[o] = r.sk[i];
         if (o == n_out - 1) {
             row_koff[n_out] = kbytes;
@@ -2307,8 +2311,9 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
     k_compact_gather_meta<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(
         d_runs, d_order, m, d_keepw, d_changed, d_new_expire, d_kpos, d_koffs, d_voffs, d_okoff,
         d_ovoff, d_row_ksrc, d_row_vsrc, d_row_patch, d_osk, n_out, kbytes, vbytes, fk, fv);
-    k_compact_gather_expire<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(d_keepw, d_kpos, d_changed,
-                                                                 d_new_expire, m, d_row_expire);
+    if (d_changed)
+        k_compact_gather_expire<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(
+            d_keepw, d_kpos, d_changed, d_new_expire, m, d_row_expire);
     uint64_t kanch = ((kbytes + 15) >> 4 >> 6) + 1, vanch = ((vbytes + 15) >> 4 >> 6) + 1;
     k_chunk_anchors<<<grid_for(kanch, BLOCK), BLOCK, 0, s>>>(d_okoff, n_out, kanch, d_kanchor);
     k_chunk_anchors<<<grid_for(vanch, BLOCK), BLOCK, 0, s>>>(d_ovoff, n_out, vanch, d_vanchor);
@@ -2316,8 +2321,9 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
         d_okoff, d_row_ksrc, n_out, kbytes, d_kanchor, kanch, d_kout);
     k_copy_chunks<<<grid_for((vbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
         d_ovoff, d_row_vsrc, n_out, vbytes, d_vanchor, vanch, d_vout);
-    k_patch_expire<<<grid_for(n_out, BLOCK), BLOCK, 0, s>>>(d_ovoff, d_row_expire, n_out, dv,
-                                                            d_vout);
+    if (d_changed)
+        k_patch_expire<<<grid_for(n_out, BLOCK), BLOCK, 0, s>>>(d_ovoff, d_row_expire, n_out,
+                                                                dv, d_vout);
 }
 
 void launch_emit_compact_inmajor(const DevRun *d_runs, int R, const uint64_t *d_wprefix,
@@ -2958,8 +2964,10 @@ __global__ void __launch_bounds__(LRK_BLK, 1) k_rank_compact_lds(
         disp = dev_disposition(runs[r], i, cp, shadow, &ch, &nts, &okl, &ovl);
         order[rank] = ((uint64_t)r << 40) | i;
         keepw[rank] = (disp == D_KEEP) ? 1 : 0;
-        changed[rank] = ch;
-        new_expire[rank] = nts;
+        if (changed)
+            changed[rank] = ch;
+        if (new_expire)
+            new_expire[rank] = nts;
         if (ksz)
             ksz[rank] = okl;
         if (vsz)

@@ -2546,20 +2546,29 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         snprintf(fp, sizeof(fp), "%s/run_%d.sk", path, i);
         r.sk = (uint64_t *)read_blob_file(fp, &nb_sk);
         nb = nb_koff;
+#define ORC_RUN_FAIL(code)                                                                         \
+    do {                                                                                           \
+        free(r.keys);                                                                              \
+        free(r.koff);                                                                              \
+        free(r.vals);                                                                              \
+        free(r.voff);                                                                              \
+        free(r.sk);                                                                                \
+        ORC_RESTORE_FAIL(code);                                                                    \
+    } while (0)
         if (!r.keys || !r.koff || !r.vals || !r.voff || !r.sk)
-            ORC_RESTORE_FAIL(RRDB_IO_ERROR);
+            ORC_RUN_FAIL(RRDB_IO_ERROR);
         r.n = mrows[i];
         if (nb_koff != (r.n + 1) * 8 || nb_voff != (r.n + 1) * 8 || nb_sk != r.n * 8 ||
             r.koff[0] != 0 || r.voff[0] != 0 || r.koff[r.n] != nb_keys ||
             r.voff[r.n] != nb_vals)
-            ORC_RESTORE_FAIL(RRDB_CORRUPTION);
+            ORC_RUN_FAIL(RRDB_CORRUPTION);
         for (uint64_t j = 0; j < r.n; j++)
             if (r.koff[j + 1] <= r.koff[j] || r.voff[j + 1] < r.voff[j])
-                ORC_RESTORE_FAIL(RRDB_CORRUPTION);
+                ORC_RUN_FAIL(RRDB_CORRUPTION);
         for (uint64_t j = 0; j + 1 < r.n; j++)
             if (key_cmp(r.keys + r.koff[j], r.koff[j + 1] - r.koff[j], r.keys + r.koff[j + 1],
                         r.koff[j + 2] - r.koff[j + 1]) >= 0)
-                ORC_RESTORE_FAIL(RRDB_CORRUPTION); /* strictly increasing */
+                ORC_RUN_FAIL(RRDB_CORRUPTION); /* strictly increasing */
         for (int f = 0; f < 5; f++) {
             static const char *exts[5] = {"keys", "koff", "vals", "voff", "sk"};
             const void *bufs[5];
@@ -2579,7 +2588,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
             for (int c = 0; c < n_crc; c++)
                 if (strcmp(crc_names[c], want) == 0 &&
                     crc_vals[c] != crc64_calc(bufs[f], (size_t)lens[f], 0))
-                    ORC_RESTORE_FAIL(RRDB_CORRUPTION);
+                    ORC_RUN_FAIL(RRDB_CORRUPTION);
         }
         (void)nb;
         r.min_seq = ~0ull;
@@ -2600,6 +2609,7 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
     free(mrows);
     free(crc_names);
     free(crc_vals);
+#undef ORC_RUN_FAIL
 #undef ORC_RESTORE_FAIL
     e->next_seq_floor = floor_;
     return RRDB_OK;
