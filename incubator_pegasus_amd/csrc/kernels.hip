@@ -2025,6 +2025,24 @@ __global__ void k_chunk_anchors(const uint64_t *row_off, uint64_t n_rows, uint64
     }
 }
 
+/* fixed-stride copy: every row is `stride` bytes with stride % 16 == 0, so
+ * chunk -> row is a division — no row-offset reads, no anchor search */
+__global__ void k_copy_chunks_fixed(const uint64_t *row_src, uint64_t n_rows, uint64_t stride,
+                                    uint8_t *dst)
+{
+    uint64_t cpr = stride >> 4; /* chunks per row */
+    uint64_t n_chunks = n_rows * cpr;
+    for (uint64_t t = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; t < n_chunks;
+         t += gridDim.x * (uint64_t)blockDim.x) {
+        uint64_t row = t / cpr;
+        uint64_t off = (t - row * cpr) << 4;
+        const uint8_t *src = (const uint8_t *)row_src[row] + off;
+        uint32_t w[4];
+        __builtin_memcpy(w, src, 16);
+        __builtin_memcpy(dst + row * stride + off, w, 16);
+    }
+}
+
 __global__ void k_copy_chunks(const uint64_t *row_off /* [n_rows+1] */,
                               const uint64_t *row_src /* device ptrs */, uint64_t n_rows,
                               uint64_t total_bytes, const uint64_t *anchors,
@@ -2314,13 +2332,27 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
     if (d_changed)
         k_compact_gather_expire<<<grid_for(m, BLOCK), BLOCK, 0, s>>>(
             d_keepw, d_kpos, d_changed, d_new_expire, m, d_row_expire);
-    uint64_t kanch = ((kbytes + 15) >> 4 >> 6) + 1, vanch = ((vbytes + 15) >> 4 >> 6) + 1;
-    k_chunk_anchors<<<grid_for(kanch, BLOCK), BLOCK, 0, s>>>(d_okoff, n_out, kanch, d_kanchor);
-    k_chunk_anchors<<<grid_for(vanch, BLOCK), BLOCK, 0, s>>>(d_ovoff, n_out, vanch, d_vanchor);
-    k_copy_chunks<<<grid_for((kbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
-        d_okoff, d_row_ksrc, n_out, kbytes, d_kanchor, kanch, d_kout);
-    k_copy_chunks<<<grid_for((vbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
-        d_ovoff, d_row_vsrc, n_out, vbytes, d_vanchor, vanch, d_vout);
+    bool kfix = fk > 0 && (fk & 15) == 0, vfix = fk > 0 && (fv & 15) == 0;
+    if (kfix) {
+        k_copy_chunks_fixed<<<grid_for((kbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
+            d_row_ksrc, n_out, fk, d_kout);
+    } else {
+        uint64_t kanch = ((kbytes + 15) >> 4 >> 6) + 1;
+        k_chunk_anchors<<<grid_for(kanch, BLOCK), BLOCK, 0, s>>>(d_okoff, n_out, kanch,
+                                                                d_kanchor);
+        k_copy_chunks<<<grid_for((kbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
+            d_okoff, d_row_ksrc, n_out, kbytes, d_kanchor, kanch, d_kout);
+    }
+    if (vfix) {
+        k_copy_chunks_fixed<<<grid_for((vbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
+            d_row_vsrc, n_out, fv, d_vout);
+    } else {
+        uint64_t vanch = ((vbytes + 15) >> 4 >> 6) + 1;
+        k_chunk_anchors<<<grid_for(vanch, BLOCK), BLOCK, 0, s>>>(d_ovoff, n_out, vanch,
+                                                                d_vanchor);
+        k_copy_chunks<<<grid_for((vbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
+            d_ovoff, d_row_vsrc, n_out, vbytes, d_vanchor, vanch, d_vout);
+    }
     if (d_changed)
         k_patch_expire<<<grid_for(n_out, BLOCK), BLOCK, 0, s>>>(d_ovoff, d_row_expire, n_out,
                                                                 dv, d_vout);
