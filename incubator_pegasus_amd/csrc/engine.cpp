@@ -527,6 +527,27 @@ static uint32_t detect_fixed_vlen(const uint64_t *voff, uint64_t n)
     return (uint32_t)s;
 }
 
+/* constant encoded-value stride over PUT records only (tombstones carry
+ * empty values and break the plain stride): the compaction OUTPUT contains
+ * only surviving PUTs, so this is the output stride the fixed-emit mode
+ * needs even when inputs hold tombstones */
+static uint32_t detect_fixed_vlen_put(const uint64_t *voff, const uint64_t *sk, uint64_t n)
+{
+    uint64_t s = 0;
+    for (uint64_t i = 0; i < n; i++) {
+        if (sk[i] & 1)
+            continue;
+        uint64_t d = voff[i + 1] - voff[i];
+        if (d == 0 || d > UINT32_MAX)
+            return 0;
+        if (s == 0)
+            s = d;
+        else if (d != s)
+            return 0;
+    }
+    return (uint32_t)s;
+}
+
 static uint32_t detect_fixed_klen(const uint64_t *koff, uint64_t n)
 {
     if (n == 0)
@@ -558,6 +579,7 @@ struct RunBuf {
     uint64_t pfx_bloom_blocks = 0;
     uint64_t *meta = nullptr;  /* (expire_ts<<32)|kind disposition column */
     uint32_t fixed_vlen = 0;   /* constant encoded-value stride (0=variable) */
+    uint32_t fixed_vlen_put = 0; /* constant PUT-value stride (emit output) */
 };
 
 struct HipScanCtx {
@@ -1235,6 +1257,7 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
     r.fixed_klen = detect_fixed_klen(koff.data(), r.n);
     r.fixed_vlen = detect_fixed_vlen(voff.data(), r.n);
+    r.fixed_vlen_put = detect_fixed_vlen_put(voff.data(), sk.data(), r.n);
     set_pfx(lcp_exact32((const uint8_t *)keys.data(), koff[1] - koff[0],
                         (const uint8_t *)keys.data() + koff[r.n - 1],
                         koff[r.n] - koff[r.n - 1]),
@@ -1450,6 +1473,7 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     r.sk = (uint64_t *)e->upload_bytes(seq_kind, n * 8);
     r.fixed_klen = detect_fixed_klen(key_offs, n);
     r.fixed_vlen = detect_fixed_vlen(val_offs, n);
+    r.fixed_vlen_put = detect_fixed_vlen_put(val_offs, seq_kind, n);
     set_pfx(lcp_exact32(keys + key_offs[0], key_offs[1] - key_offs[0],
                         keys + key_offs[n - 1], key_offs[n] - key_offs[n - 1]),
             &r.pfx_skip, &r.lcp_exact);
@@ -2647,11 +2671,11 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
      * keep count — the ksz/vsz arrays and two of the three full-length
      * prefix sums are skipped entirely */
     uint64_t ffk = e->runs.empty() ? 0 : e->runs[0].fixed_klen;
-    uint64_t ffv = e->runs.empty() ? 0 : e->runs[0].fixed_vlen;
+    uint64_t ffv = e->runs.empty() ? 0 : e->runs[0].fixed_vlen_put;
     for (auto &rr : e->runs) {
         if (rr.fixed_klen != ffk)
             ffk = 0;
-        if (rr.fixed_vlen != ffv)
+        if (rr.fixed_vlen_put != ffv)
             ffv = 0;
     }
     bool fixed_emit = ffk > 0 && ffv > 0 && e->emit_mode == 2;
@@ -2841,6 +2865,14 @@ static int32_t compact_finish(HipEngine *e, rrdb_compact_stats *stats)
         for (auto &ir : e->runs)
             if (ir.fixed_vlen != nr.fixed_vlen)
                 nr.fixed_vlen = 0;
+        /* the merged output holds only PUTs: its plain stride equals the
+         * shared put-stride of the inputs (when one exists) */
+        nr.fixed_vlen_put = e->runs.empty() ? 0 : e->runs[0].fixed_vlen_put;
+        for (auto &ir : e->runs)
+            if (ir.fixed_vlen_put != nr.fixed_vlen_put)
+                nr.fixed_vlen_put = 0;
+        if (!nr.fixed_vlen)
+            nr.fixed_vlen = nr.fixed_vlen_put;
         if (keep_inputs) {
             /* output is dropped at the end of the pass: pooled temporaries */
             nr.keys = e->talloc<uint8_t>(kbytes);
@@ -3204,6 +3236,8 @@ int32_t rrdb_restore(void *h, const char *dir, uint64_t decree)
         r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size());
         r.fixed_klen = detect_fixed_klen((const uint64_t *)koff.data(), r.n);
         r.fixed_vlen = detect_fixed_vlen((const uint64_t *)voff.data(), r.n);
+        r.fixed_vlen_put = detect_fixed_vlen_put((const uint64_t *)voff.data(),
+                                                 (const uint64_t *)sk.data(), r.n);
         if (r.n) {
             const uint64_t *ko = (const uint64_t *)koff.data();
             set_pfx(lcp_exact32(keys.data() + ko[0], ko[1] - ko[0],

@@ -1978,7 +1978,9 @@ __global__ void k_compact_gather_meta(const DevRun *runs, const uint64_t *order,
         row_koff[o] = fk ? o * fk : koffs[p];
         row_voff[o] = fk ? o * fv : voffs[p];
         row_ksrc[o] = (uint64_t)(fk ? r.keys + i * fk : r.keys + r.koff[i]);
-        row_vsrc[o] = (uint64_t)(fk ? r.vals + i * fv : r.vals + r.voff[i]);
+        /* input offsets stay voff-addressed: tombstones make input value
+         * strides irregular even when every PUT shares one length */
+        row_vsrc[o] = (uint64_t)(r.vals + r.voff[i]);
         row_patch[o] = (changed && changed[p]) ? 3 : 0;
         osk[o] = r.sk[i];
         if (o == n_out - 1) {
