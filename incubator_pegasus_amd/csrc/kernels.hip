@@ -2045,6 +2045,29 @@ __global__ void k_copy_chunks_fixed(const uint64_t *row_src, uint64_t n_rows, ui
     }
 }
 
+/* fixed-stride copy for strides that are NOT 16B multiples (e.g. 18B
+ * keys): one thread copies one whole row in u64 pieces — consecutive
+ * threads write consecutive stride-length spans, which the coalescer folds
+ * into contiguous segments (the 16B-chunk path degraded to byte loops at
+ * every straddled row boundary) */
+__global__ void k_copy_rows_fixed(const uint64_t *row_src, uint64_t n_rows, uint64_t stride,
+                                  uint8_t *dst)
+{
+    for (uint64_t r = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; r < n_rows;
+         r += gridDim.x * (uint64_t)blockDim.x) {
+        const uint8_t *src = (const uint8_t *)row_src[r];
+        uint8_t *d = dst + r * stride;
+        uint64_t b = 0;
+        for (; b + 8 <= stride; b += 8) {
+            uint64_t w;
+            __builtin_memcpy(&w, src + b, 8);
+            __builtin_memcpy(d + b, &w, 8);
+        }
+        for (; b < stride; b++)
+            d[b] = src[b];
+    }
+}
+
 __global__ void k_copy_chunks(const uint64_t *row_off /* [n_rows+1] */,
                               const uint64_t *row_src /* device ptrs */, uint64_t n_rows,
                               uint64_t total_bytes, const uint64_t *anchors,
@@ -2338,6 +2361,9 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
     if (kfix) {
         k_copy_chunks_fixed<<<grid_for((kbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
             d_row_ksrc, n_out, fk, d_kout);
+    } else if (fk > 0) {
+        k_copy_rows_fixed<<<grid_for(n_out, BLOCK), BLOCK, 0, s>>>(d_row_ksrc, n_out, fk,
+                                                                   d_kout);
     } else {
         uint64_t kanch = ((kbytes + 15) >> 4 >> 6) + 1;
         k_chunk_anchors<<<grid_for(kanch, BLOCK), BLOCK, 0, s>>>(d_okoff, n_out, kanch,
@@ -2348,6 +2374,9 @@ void launch_emit_compact_chunked(const DevRun *d_runs, const uint64_t *d_order, 
     if (vfix) {
         k_copy_chunks_fixed<<<grid_for((vbytes + 15) >> 4, BLOCK), BLOCK, 0, s>>>(
             d_row_vsrc, n_out, fv, d_vout);
+    } else if (fk > 0) {
+        k_copy_rows_fixed<<<grid_for(n_out, BLOCK), BLOCK, 0, s>>>(d_row_vsrc, n_out, fv,
+                                                                   d_vout);
     } else {
         uint64_t vanch = ((vbytes + 15) >> 4 >> 6) + 1;
         k_chunk_anchors<<<grid_for(vanch, BLOCK), BLOCK, 0, s>>>(d_ovoff, n_out, vanch,
