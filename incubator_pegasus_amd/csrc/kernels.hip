@@ -1746,6 +1746,19 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
             }
             const uint64_t *ft = cur ? s_tb : s_ta;
             const uint16_t *fo = cur ? s_ob : s_oa;
+            /* MODE 0: the retired ping-pong buffer stages the meta column
+             * (coalesced segment-major loads instead of per-element gathers
+             * in merged order) — costs no extra LDS */
+            uint64_t *s_mt = cur ? s_ta : s_tb;
+            if (MODE == 0 && s_allmeta) {
+                for (int q = 0; q < R; q++) {
+                    uint64_t len = s_seglen[q], off = s_segoff[q], a0 = s_a0[q];
+                    const uint64_t *mq = runs[q].meta;
+                    for (uint64_t j = threadIdx.x; j < len; j += blockDim.x)
+                        s_mt[off + j] = mq[a0 + j];
+                }
+                __syncthreads();
+            }
             /* output phase: position p in the merged group == local rank */
             uint64_t iters = (gsize + blockDim.x - 1) / blockDim.x;
             for (uint64_t it = 0; it < iters; it++) {
@@ -1770,9 +1783,11 @@ __global__ void __launch_bounds__(BLOCK) k_rank_grp(
                             }
                     }
                     uint64_t rank = s_base + p;
-                    const uint64_t *sm = (MODE == 2 && s_allmeta)
-                                             ? &s_meta[s_segoff[q] + (org & 0xFFF)]
-                                             : nullptr;
+                    const uint64_t *sm = nullptr;
+                    if (MODE == 2 && s_allmeta)
+                        sm = &s_meta[s_segoff[q] + (org & 0xFFF)];
+                    else if (MODE == 0 && s_allmeta)
+                        sm = &s_mt[s_segoff[q] + (org & 0xFFF)];
                     disp = grp_epilogue<MODE>(runs, q, i, shadow, rank, cp, sp, order, keepw,
                                               changed, new_expire, ksz, vsz, shadowed, sm);
                 }
