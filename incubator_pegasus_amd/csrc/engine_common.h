@@ -169,6 +169,22 @@ struct MgGraphHdr {
     uint64_t hash_key_skip;
 };
 
+/* persistent serving kernel mailbox (pinned host memory).  Host writes the
+ * request slice then release-stores req_seq; the resident 1-workgroup
+ * kernel acquire-polls it, runs the fused multi_get and release-stores
+ * done_seq after writing the response prefix.  The kernel ALWAYS exits
+ * within MG_SRV_IDLE_MS of the last request (or on quit), so device-wide
+ * synchronization can stall at most that long and can never hang. */
+#define MG_SRV_IDLE_MS 20
+struct MgMailbox {
+    alignas(64) uint64_t req_seq;
+    alignas(64) uint64_t done_seq;
+    alignas(64) uint32_t quit;
+    alignas(64) uint32_t alive; /* 1 while the kernel loop runs, 0 on exit */
+    alignas(64) uint8_t req[MG_GRAPH_IN];
+    alignas(64) uint8_t resp[32 + (16 << 10)];
+};
+
 /* compact per-record disposition written by the filter kernel */
 struct CompactStatsDev {
     unsigned long long expired, filtered, tombstones, shadowed, output_records;

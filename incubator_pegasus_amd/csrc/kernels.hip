@@ -2843,6 +2843,112 @@ __global__ void __launch_bounds__(256) k_multi_get_batch(const DevRun *runs, int
             hdrs + req * 4, blobs + req * blob_stride, blob_stride);
 }
 
+/* ================= persistent serving kernel =================
+ * One resident workgroup serves single-request multi_gets from a pinned
+ * mailbox: no per-call kernel dispatch (the ~15-20us launch-to-retire floor
+ * of a one-workgroup kernel dominated the 40us serving path).  Exit is
+ * BOUNDED: the poll loop leaves after MG_SRV_IDLE_MS without a request, on
+ * the quit flag, always — a device-wide sync can stall at most the idle
+ * window, never hang.  Memory ordering follows the producer-release /
+ * consumer-acquire recipe at system scope (host <-> device over pinned). */
+__global__ void __launch_bounds__(256) k_mg_server(const DevRun *runs, int R, MgMailbox *mb,
+                                                   uint8_t *d_out)
+{
+    __shared__ uint64_t s_req;
+    __shared__ int s_stop;
+    int tid = threadIdx.x;
+    /* adopt the current sequence so a pre-launch doorbell value is never
+     * misread as a fresh request */
+    uint64_t last =
+        __hip_atomic_load(&mb->req_seq, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
+    /* ~100 MHz constant wall clock; idle budget in ticks */
+    const long long IDLE = (long long)MG_SRV_IDLE_MS * 100000;
+    long long idle_start = (long long)wall_clock64();
+    if (tid == 0)
+        __hip_atomic_store(&mb->alive, 1u, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+    for (;;) {
+        if (tid == 0) {
+            uint64_t r = last;
+            int stop = 0;
+            for (;;) {
+                if (__hip_atomic_load(&mb->quit, __ATOMIC_ACQUIRE,
+                                      __HIP_MEMORY_SCOPE_SYSTEM)) {
+                    stop = 1;
+                    break;
+                }
+                r = __hip_atomic_load(&mb->req_seq, __ATOMIC_ACQUIRE,
+                                      __HIP_MEMORY_SCOPE_SYSTEM);
+                if (r != last)
+                    break;
+                if ((long long)wall_clock64() - idle_start > IDLE) {
+                    stop = 1;
+                    break;
+                }
+            }
+            s_req = r;
+            s_stop = stop;
+        }
+        __syncthreads();
+        if (s_stop)
+            break;
+        uint64_t seq = s_req;
+        /* request slice layout = the graph lane's: [MgGraphHdr][start][stop][pat] */
+        const MgGraphHdr *h = (const MgGraphHdr *)mb->req;
+        MgFusedArgs a{};
+        const uint8_t *p = mb->req + sizeof(MgGraphHdr);
+        a.start = p;
+        a.start_len = h->start_len;
+        a.stop = p + h->start_len;
+        a.stop_len = h->stop_len;
+        a.sk_pat = p + h->start_len + h->stop_len;
+        a.sk_pat_len = h->sk_pat_len;
+        a.start_inclusive = h->start_inclusive;
+        a.stop_inclusive = h->stop_inclusive;
+        a.reverse = h->reverse;
+        a.no_value = h->no_value;
+        a.max_kv_count = h->max_kv_count;
+        a.max_iteration_count = h->max_iteration_count;
+        a.max_iteration_size = h->max_iteration_size;
+        a.sk_ft = h->sk_ft;
+        a.epoch_now = h->epoch_now;
+        a.data_version = h->data_version;
+        a.hash_key_skip = h->hash_key_skip;
+        mg_core(runs, R, a, a.start, a.start_len, a.stop, a.stop_len, a.hash_key_skip,
+                (int64_t *)d_out, d_out + 32, MG_BLOB_BYTES);
+        __syncthreads();
+        /* response prefix -> pinned: hdr + the used blob bytes (caller does
+         * a device read of the tail for oversized results) */
+        const int64_t *hdr = (const int64_t *)d_out;
+        uint64_t blob_n = 0;
+        if (hdr[0] >= 0)
+            blob_n = 2 * ((uint64_t)hdr[0] + 1) * 8 + (uint64_t)hdr[2] + (uint64_t)hdr[3];
+        uint64_t n = 32 + (blob_n < (uint64_t)(16 << 10) ? blob_n : (uint64_t)(16 << 10));
+        for (uint64_t b = tid * 8; b + 8 <= n; b += blockDim.x * 8) {
+            uint64_t w;
+            __builtin_memcpy(&w, d_out + b, 8);
+            __builtin_memcpy(mb->resp + b, &w, 8);
+        }
+        if (tid == 0)
+            for (uint64_t b = n & ~7ull; b < n; b++)
+                mb->resp[b] = d_out[b];
+        __threadfence_system();
+        __syncthreads();
+        if (tid == 0)
+            __hip_atomic_store(&mb->done_seq, seq, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+        last = seq;
+        idle_start = (long long)wall_clock64();
+    }
+    __syncthreads();
+    if (tid == 0)
+        __hip_atomic_store(&mb->alive, 0u, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+void launch_mg_server(const DevRun *runs, int R, MgMailbox *mb, uint8_t *d_out, hipStream_t s)
+{
+    k_mg_server<<<1, 256, 0, s>>>(runs, R, mb, d_out);
+}
+
 /* gather each request's used blob bytes into one packed buffer (one D2H) */
 __global__ void k_pack_blobs(const uint8_t *blobs, uint64_t blob_stride, uint64_t n_req,
                              const uint64_t *used /* [n_req] */,
