@@ -778,7 +778,8 @@ struct HipEngine {
             }
             memset((void *)mg_mb, 0, sizeof(MgMailbox));
         }
-        if (!mg_srv_stream && hipStreamCreate(&mg_srv_stream) != hipSuccess) {
+        if (!mg_srv_stream &&
+            hipStreamCreateWithFlags(&mg_srv_stream, hipStreamNonBlocking) != hipSuccess) {
             mg_persist_enabled = false;
             return false;
         }

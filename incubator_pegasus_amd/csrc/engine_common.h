@@ -175,7 +175,10 @@ struct MgGraphHdr {
  * done_seq after writing the response prefix.  The kernel ALWAYS exits
  * within MG_SRV_IDLE_MS of the last request (or on quit), so device-wide
  * synchronization can stall at most that long and can never hang. */
-#define MG_SRV_IDLE_MS 20
+/* small: a resident kernel occupies one of the few HW queues, and
+ * null-stream / device-wide syncs and co-scheduled streams can stall until
+ * it exits — 2ms bounds that while call rates above ~1kHz keep it warm */
+#define MG_SRV_IDLE_MS 2
 struct MgMailbox {
     alignas(64) uint64_t req_seq;
     alignas(64) uint64_t done_seq;
