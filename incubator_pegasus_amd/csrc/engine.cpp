@@ -3426,6 +3426,9 @@ extern "C" int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *
     std::lock_guard<std::mutex> g(e->mu);
     engine_flush(e);
     e->activate();
+    /* bulk path: retire the resident serving kernel so its poll windows do
+     * not share hardware queues with the 4096-workgroup batch launches */
+    e->server_quit_sync();
     e->scratch_reset();
     Arena *a = result_init(out);
     if (shared->n_sort_keys != 0 || shared->sort_key_filter_type < 0 ||

@@ -22,6 +22,7 @@ typedef int32_t (*mg_fn)(void *, const rrdb_multi_get_request *, uint32_t, rrdb_
 typedef int32_t (*mgb_fn)(void *, uint64_t, const uint8_t *, const uint64_t *,
                           const rrdb_multi_get_request *, uint32_t, rrdb_result *);
 typedef void (*free_fn)(rrdb_result *);
+typedef int32_t (*envs_fn)(void *, const char *const *, const char *const *, int32_t);
 
 static uint64_t splitmix64(uint64_t x)
 {
@@ -47,10 +48,22 @@ int main(int argc, char **argv)
     auto rrdb_mg = (mg_fn)dlsym(lib, "rrdb_multi_get");
     auto rrdb_mgb = (mgb_fn)dlsym(lib, "rrdb_multi_get_batch");
     auto rrdb_free = (free_fn)dlsym(lib, "rrdb_free_result");
+    auto rrdb_envs = (envs_fn)dlsym(lib, "rrdb_set_envs");
     void *h = rrdb_open_(9, 0, 0);
     if (!h) {
         printf("open failed (no GPU?)\n");
         return 1;
+    }
+    if (argc > 4 && rrdb_envs) { /* e.g. "engine.mg_persist=off" */
+        char kv[128];
+        snprintf(kv, sizeof(kv), "%s", argv[4]);
+        char *eq = strchr(kv, '=');
+        if (eq) {
+            *eq = 0;
+            const char *ks[1] = {kv};
+            const char *vs[1] = {eq + 1};
+            rrdb_envs(h, ks, vs, 1);
+        }
     }
     /* build one sorted run: n_hash hashkeys x SKS sortkeys, v1 values */
     uint64_t n = n_hash * SKS;
