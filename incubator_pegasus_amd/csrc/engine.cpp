@@ -741,7 +741,31 @@ struct HipEngine {
     hipStream_t mg_srv_stream = nullptr;
     uint64_t mg_srv_gen = ~0ull; /* runs_gen the resident kernel was launched at */
     uint64_t mg_srv_seq = 0;
-    bool mg_persist_enabled = true;
+    /* 0 = off, 1 = auto (default: engage only after a sustained burst on
+     * THIS handle — a resident kernel helps a hot partition but its poll
+     * windows tax handles that are merely visited round-robin), 2 = on */
+    int mg_persist_mode = 1;
+    bool mg_persist_enabled = true; /* false = runtime failure, stay off */
+    std::chrono::steady_clock::time_point mg_last_call{};
+    uint32_t mg_hot = 0;
+    bool mg_persist_wanted()
+    {
+        if (!mg_persist_enabled || mg_persist_mode == 0)
+            return false;
+        if (mg_persist_mode == 2)
+            return true;
+        auto now = std::chrono::steady_clock::now();
+        auto us = std::chrono::duration_cast<std::chrono::microseconds>(now - mg_last_call)
+                      .count();
+        mg_last_call = now;
+        if (us < 1000) {
+            if (mg_hot < 1000)
+                mg_hot++;
+        } else {
+            mg_hot = 0;
+        }
+        return mg_hot >= 32;
+    }
     void server_quit_sync()
     {
         if (!mg_mb || mg_srv_gen == ~0ull)
@@ -1509,8 +1533,12 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
             if (v == "off") {
                 e->activate();
                 e->server_quit_sync();
-                e->mg_persist_enabled = false;
-            } else {
+                e->mg_persist_mode = 0;
+            } else if (v == "on") {
+                e->mg_persist_mode = 2;
+                e->mg_persist_enabled = true;
+            } else { /* "auto" */
+                e->mg_persist_mode = 1;
                 e->mg_persist_enabled = true;
             }
         } else if (k == "engine.mg_graph") {
@@ -2416,7 +2444,7 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
         const uint8_t *d_blob = nullptr;
         const uint8_t *h_resp = e->mg_hout;
         bool served = false;
-        if (sizeof(MgGraphHdr) + in_n <= MG_GRAPH_IN && e->mg_persist_enabled &&
+        if (sizeof(MgGraphHdr) + in_n <= MG_GRAPH_IN && e->mg_persist_wanted() &&
             e->server_ready()) {
             /* resident-kernel lane: write the request slice, bump the
              * doorbell, spin for completion */
