@@ -196,6 +196,51 @@ def test_mg_graph_lane_parity(oracle_lib, hip_lib):
         g.close()
 
 
+def test_mg_persistent_server_parity(oracle_lib, hip_lib):
+    """engine.mg_persist=on forces the resident serving kernel; results must
+    stay byte-identical to the oracle across bursts, run-set changes
+    (server retire + relaunch), idle expiries and the off toggle."""
+    import time as _t
+
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        seq = 1
+        recs = []
+        for i in range(200):
+            for sk in (b"p", b"q"):
+                recs.append((D.generate_key(b"sv%04d" % i, sk),
+                             D.encode_value(b"v%d" % i, 0, seq, 1), seq, 0))
+                seq += 1
+        recs.sort(key=lambda r: r[0])
+        o.ingest_run(recs)
+        g.ingest_run(recs)
+        g.set_envs({"engine.mg_persist": "on"})
+        # burst: keeps the resident kernel warm across calls
+        for rep in range(3):
+            for i in (0, 7, 50, 199, 4444):
+                hk = b"sv%04d" % i
+                assert o.multi_get(hk, NOW) == g.multi_get(hk, NOW), (rep, hk)
+        # idle past the 2ms window: the kernel self-exits; next call relaunches
+        _t.sleep(0.05)
+        assert o.multi_get(b"sv0007", NOW) == g.multi_get(b"sv0007", NOW)
+        # run-set change retires the server before buffers are touched
+        o.put(b"sv0007", b"r", b"new")
+        g.put(b"sv0007", b"r", b"new")
+        assert o.multi_get(b"sv0007", NOW) == g.multi_get(b"sv0007", NOW)
+        o.manual_compact(NOW)
+        g.manual_compact(NOW)
+        for i in (0, 7, 199):
+            hk = b"sv%04d" % i
+            assert o.multi_get(hk, NOW) == g.multi_get(hk, NOW), hk
+        # off toggle retires it; plain path still agrees
+        g.set_envs({"engine.mg_persist": "off"})
+        assert o.multi_get(b"sv0050", NOW) == g.multi_get(b"sv0050", NOW)
+    finally:
+        o.close()
+        g.close()
+
+
 def test_prefix_bloom_parity_and_toggle(oracle_lib, hip_lib):
     """§8(f)3 second half: the hashkey-prefix bloom is lossless — multi_get
     and sortkey_count agree with the oracle for present and ABSENT hashkeys,
