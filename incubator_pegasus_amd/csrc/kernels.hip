@@ -1613,7 +1613,7 @@ __device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i,
  * range found by one diagonal search) — ~5-10x fewer instructions per
  * element than per-element binary searches, which measured issue-bound. */
 template <int MODE>
-__global__ void __launch_bounds__(BLOCK) k_rank_grp(
+__global__ void __launch_bounds__(BLOCK, 6) k_rank_grp(
     const DevRun *runs, int R, const uint64_t *lo, const uint64_t *anch, uint64_t n_groups,
     CompactParams cp, ScanParams sp, uint64_t *order, uint64_t *keepw, uint8_t *changed,
     uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of, uint8_t *shadowed,
