@@ -1565,6 +1565,8 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     if (n == 0)
         return RRDB_OK;
     /* validate on host (data arrives host-side anyway) */
+    if (key_offs[0] != 0 || key_offs[1] == 0)
+        return RRDB_INVALID_ARGUMENT; /* keys must be nonempty, offsets 0-based */
     for (uint64_t i = 0; i < n; i++) {
         if ((seq_kind[i] >> 1) < e->next_seq_floor)
             return RRDB_INVALID_ARGUMENT;
