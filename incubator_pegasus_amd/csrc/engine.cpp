@@ -83,7 +83,8 @@ void launch_anchor_rows(const DevRun *, int, int, const uint64_t *, const uint64
                         uint64_t, uint64_t *, hipStream_t);
 void launch_rank_grp_compact(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
                              const CompactParams &, uint64_t *, uint64_t *, uint8_t *,
-                             uint32_t *, uint64_t *, uint64_t *, CompactStatsDev *, hipStream_t);
+                             uint32_t *, uint64_t *, uint64_t *, CompactStatsDev *, int,
+                             hipStream_t);
 void launch_rank_grp_view(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
                           uint64_t *, uint8_t *, hipStream_t);
 void launch_rank_grp_count(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
@@ -863,6 +864,9 @@ struct HipEngine {
                           (env "engine.rank_mode": grp|ldst|global|lds) */
     int bt_shift = 5;  /* bound-table block = 1<<bt_shift records (env
                           "engine.bt_shift") */
+    int grp_blocks = 3584; /* group-rank grid cap (env "engine.grp_blocks");
+                              grid-stride beats one-block-per-group by ~24%
+                              (workgroup setup/LDS churn) */
 
     void activate() { HIP_OK(hipSetDevice(device)); }
 
@@ -1525,6 +1529,10 @@ int32_t rrdb_set_envs(void *h, const char *const *keys, const char *const *value
         } else if (k == "engine.rank_mode") {
             e->rank_mode =
                 (v == "lds") ? 1 : (v == "ldst" ? 3 : (v == "global" ? 0 : 4));
+        } else if (k == "engine.grp_blocks") {
+            int b = atoi(v.c_str());
+            if (b >= 256 && b <= 65535)
+                e->grp_blocks = b;
         } else if (k == "engine.bt_shift") {
             int s_ = atoi(v.c_str());
             if (s_ >= 4 && s_ <= 16)
@@ -2906,7 +2914,7 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
         uint64_t n_groups = e->build_anchors(dr, R, lo, hi, d_lo, d_hi, &d_anch);
         HIP_OK(hipEventRecord(ev[0], e->stream));
         launch_rank_grp_compact(dr, R, d_lo, d_anch, n_groups, cp, d_order, d_keepw, d_changed,
-                                d_new_expire, d_ksz, d_vsz, d_stats, e->stream);
+                                d_new_expire, d_ksz, d_vsz, d_stats, e->grp_blocks, e->stream);
         HIP_OK(hipEventRecord(ev[1], e->stream));
     } else {
     uint64_t *d_bt_off = nullptr, *d_bt = nullptr;
