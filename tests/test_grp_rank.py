@@ -384,3 +384,55 @@ def test_grp_rank_rules_and_ttl(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+def test_concurrent_threads_one_handle(hip_lib):
+    """The reference serves reads concurrently across thread pools
+    (replication.codes.h:44-50); the C-ABI takes a per-handle lock, so
+    concurrent callers must serialize safely — including while the resident
+    serving lane and compactions interleave."""
+    import threading
+
+    g = hip_lib.open(1, 0, 0)
+    errs = []
+    try:
+        seq = 1
+        recs = []
+        for i in range(500):
+            recs.append((D.generate_key(b"th%04d" % i, b"s"),
+                         D.encode_value(b"v%d" % i, 0, seq, 1), seq, 0))
+            seq += 1
+        g.ingest_run(recs)
+        g.set_envs({"engine.mg_persist": "on",
+                    "rocksdb.max_iteration_count": str(2**31 - 1)})
+        stop = threading.Event()
+
+        def reader(tid):
+            try:
+                n = 0
+                while not stop.is_set() and n < 300:
+                    hk = b"th%04d" % ((tid * 37 + n) % 500)
+                    st, kvs = g.multi_get(hk, NOW)
+                    assert st == 0 and len(kvs) == 1, (st, hk)
+                    n += 1
+            except Exception as ex:  # noqa: BLE001
+                errs.append(ex)
+
+        def compactor():
+            try:
+                for _ in range(5):
+                    err, _ = g.manual_compact(NOW, keep_inputs=True)
+                    assert err == 0
+            except Exception as ex:  # noqa: BLE001
+                errs.append(ex)
+
+        ts = [threading.Thread(target=reader, args=(i,)) for i in range(4)]
+        ts.append(threading.Thread(target=compactor))
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=120)
+        stop.set()
+        assert not errs, errs
+    finally:
+        g.close()
