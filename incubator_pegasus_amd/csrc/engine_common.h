@@ -20,8 +20,8 @@
 #define RRDB_MAX_RUNS 64
 /* tail-word rank modes: max run count; group-streaming rank group budget */
 #define LDST_MAXR 16
-#define GRP_CAP 1024
-#define GRP_TARGET 512
+#define GRP_CAP 2048
+#define GRP_TARGET 1024
 
 /* device-visible descriptor of one sorted run */
 struct DevRun {
