@@ -836,6 +836,10 @@ struct HipEngine {
         uint64_t *d_anch = nullptr, *d_lo = nullptr, *d_hi = nullptr;
         uint8_t *d_start = nullptr, *d_stop = nullptr, *d_hk_pat = nullptr,
                 *d_sk_pat = nullptr;
+        /* pooled capacities: buffers persist across calls, realloc only on
+         * growth (8 hipMallocs per begin measurably taxed the 16-partition
+         * fan-out on slow-host boxes) */
+        uint64_t anch_cap = 0, lo_cap = 0, bytes_cap = 0;
     } pend_scan;
     uint64_t next_seq_floor = 0;
     /* user ops: host + device */
@@ -2122,13 +2126,13 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
             start = ps;
     }
     int c = key_cmp(start, stop);
-    e->pend_scan = {};
     if (c > 0 || (c == 0 && (!start_inclusive || !stop_inclusive))) {
         e->pend_scan.active = true;
         e->pend_scan.trivial = true;
         e->pend_scan.trivial_count = 0;
         return RRDB_OK; /* empty range (on_get_scanner:1227-1243) */
     }
+    e->pend_scan.trivial = false;
     std::string stop_excl = stop;
     if (stop_inclusive)
         stop_excl.push_back('\0');
@@ -2136,19 +2140,36 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     int R = (int)e->runs.size();
     DevRun *dr = e->dev_runs();
     auto &ps = e->pend_scan;
-    auto upl = [&](const std::string &b) {
-        uint8_t *d = nullptr;
-        HIP_OK(hipMalloc(&d, b.size() ? b.size() : 1));
-        if (!b.empty())
-            HIP_OK(hipMemcpyAsync(d, b.data(), b.size(), hipMemcpyHostToDevice, e->stream));
+    /* pooled allocations: grow-only, freed at close */
+    if ((uint64_t)R * 8 > ps.lo_cap) {
+        if (ps.d_lo)
+            (void)hipFree(ps.d_lo);
+        if (ps.d_hi)
+            (void)hipFree(ps.d_hi);
+        HIP_OK(hipMalloc(&ps.d_lo, R * 8));
+        HIP_OK(hipMalloc(&ps.d_hi, R * 8));
+        ps.lo_cap = (uint64_t)R * 8;
+    }
+    uint64_t need_bytes = start.size() + stop_excl.size() + q->hash_key_filter_pattern.len +
+                          q->sort_key_filter_pattern.len + 4;
+    if (need_bytes > ps.bytes_cap) {
+        if (ps.d_start)
+            (void)hipFree(ps.d_start);
+        HIP_OK(hipMalloc(&ps.d_start, need_bytes));
+        ps.bytes_cap = need_bytes;
+    }
+    uint64_t boff = 0;
+    auto upl = [&](const uint8_t *p, uint64_t n) {
+        uint8_t *d = ps.d_start + boff;
+        if (n)
+            HIP_OK(hipMemcpyAsync(d, p, n, hipMemcpyHostToDevice, e->stream));
+        boff += n + 1;
         return d;
     };
-    HIP_OK(hipMalloc(&ps.d_lo, R * 8));
-    HIP_OK(hipMalloc(&ps.d_hi, R * 8));
-    ps.d_start = upl(start);
-    ps.d_stop = upl(stop_excl);
-    launch_bounds(dr, R, ps.d_start, start.size(), ps.d_lo, 0, e->stream);
-    launch_bounds(dr, R, ps.d_stop, stop_excl.size(), ps.d_hi, 0, e->stream);
+    uint8_t *d_start_b = upl((const uint8_t *)start.data(), start.size());
+    uint8_t *d_stop_b = upl((const uint8_t *)stop_excl.data(), stop_excl.size());
+    launch_bounds(dr, R, d_start_b, start.size(), ps.d_lo, 0, e->stream);
+    launch_bounds(dr, R, d_stop_b, stop_excl.size(), ps.d_hi, 0, e->stream);
     /* anchors from the largest run's FULL size (the window is device-side
      * only); rows past the window collapse to sentinels in-kernel */
     int q0 = 0;
@@ -2164,7 +2185,12 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     uint64_t n_groups = (nmax + (1ull << gs) - 1) >> gs;
     if (n_groups == 0)
         n_groups = 1;
-    HIP_OK(hipMalloc(&ps.d_anch, (n_groups + 1) * R * 8));
+    if ((n_groups + 1) * R * 8 > ps.anch_cap) {
+        if (ps.d_anch)
+            (void)hipFree(ps.d_anch);
+        HIP_OK(hipMalloc(&ps.d_anch, (n_groups + 1) * R * 8));
+        ps.anch_cap = (n_groups + 1) * R * 8;
+    }
     launch_anchor_rows(dr, R, q0, ps.d_lo, ps.d_hi, gs, n_groups, ps.d_anch, e->stream);
 
     ScanParams sp{};
@@ -2176,17 +2202,14 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     sp.hk_ft = q->hash_key_filter_type;
     sp.sk_ft = q->sort_key_filter_type;
     sp.hk_pat_len = q->hash_key_filter_pattern.len;
-    ps.d_hk_pat = upl(std::string((const char *)q->hash_key_filter_pattern.data,
-                                  q->hash_key_filter_pattern.len));
-    sp.hk_pat = ps.d_hk_pat;
+    sp.hk_pat = upl(q->hash_key_filter_pattern.data, q->hash_key_filter_pattern.len);
     sp.sk_pat_len = q->sort_key_filter_pattern.len;
-    ps.d_sk_pat = upl(std::string((const char *)q->sort_key_filter_pattern.data,
-                                  q->sort_key_filter_pattern.len));
-    sp.sk_pat = ps.d_sk_pat;
+    sp.sk_pat = upl(q->sort_key_filter_pattern.data, q->sort_key_filter_pattern.len);
     sp.no_value = 1;
     sp.hash_key_skip = 0;
 
-    HIP_OK(hipMalloc(&ps.d_stats, 8 * sizeof(CompactStatsDev)));
+    if (!ps.d_stats)
+        HIP_OK(hipMalloc(&ps.d_stats, 8 * sizeof(CompactStatsDev)));
     HIP_OK(hipMemsetAsync(ps.d_stats, 0, 8 * sizeof(CompactStatsDev), e->stream));
     launch_rank_grp_count(dr, R, ps.d_lo, ps.d_anch, n_groups, sp, ps.d_stats, e->stream);
     ps.active = true;
@@ -2218,12 +2241,8 @@ int32_t rrdb_scan_count_finish(void *h, rrdb_result *out)
     uint64_t count = 0;
     for (auto &b : hsb)
         count += b.output_records;
-    for (void *p : {(void *)ps.d_stats, (void *)ps.d_anch, (void *)ps.d_lo, (void *)ps.d_hi,
-                    (void *)ps.d_start, (void *)ps.d_stop, (void *)ps.d_hk_pat,
-                    (void *)ps.d_sk_pat})
-        if (p)
-            (void)hipFree(p);
-    e->pend_scan = {};
+    /* pooled device buffers persist for the next count (freed at close) */
+    ps.trivial = false;
     out->i64 = (int64_t)count;
     out->count = 0;
     out->error = RRDB_OK;

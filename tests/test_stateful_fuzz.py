@@ -72,6 +72,18 @@ class OracleVsModel(RuleBasedStateMachine):
         assert cnt == self.model.sortkey_count(hk, NOW)
 
     @rule()
+    def check_scan_count(self):
+        # the pipelined count API must agree with the model's full scan
+        self.part.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
+        rc = self.part.scan_count_begin(b"\x00\x00", b"\xff\xff", NOW,
+                                        validate_partition_hash=False)
+        assert rc == 0
+        err, cnt = self.part.scan_count_finish()
+        assert err == 0
+        assert cnt == len(self.model.full_scan(NOW, validate_hash_req=False))
+        self.part.set_envs({"rocksdb.max_iteration_count": "1000"})
+
+    @rule()
     def compact(self):
         err, st_ = self.part.manual_compact(NOW)
         surviving = self.model.compact(NOW)
