@@ -1612,8 +1612,12 @@ __device__ static inline int grp_epilogue(const DevRun *runs, int q, uint64_t i,
  * each round every element moves once, each thread owns a fixed output-slot
  * range found by one diagonal search) — ~5-10x fewer instructions per
  * element than per-element binary searches, which measured issue-bound. */
+/* 256-thread workgroups: 512 was tried (8 waves/WG, double co-residency at
+ * the same LDS) and REGRESSED (rank 0.32 -> 0.43ms): wider barriers and
+ * the serial thread-0 sections stall twice the waves */
+#define GRP_BLOCK 256
 template <int MODE>
-__global__ void __launch_bounds__(BLOCK, 6) k_rank_grp(
+__global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
     const DevRun *runs, int R, const uint64_t *lo, const uint64_t *anch, uint64_t n_groups,
     CompactParams cp, ScanParams sp, uint64_t *order, uint64_t *keepw, uint8_t *changed,
     uint32_t *new_expire, uint64_t *ksz, uint64_t *vsz, uint64_t *rank_of, uint8_t *shadowed,
@@ -1867,7 +1871,7 @@ void launch_rank_grp_compact(const DevRun *d_runs, int R, const uint64_t *d_lo,
     if (blocks > (uint64_t)block_cap)
         blocks = (uint64_t)block_cap;
     ScanParams sp{};
-    k_rank_grp<0><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+    k_rank_grp<0><<<dim3((uint32_t)blocks), dim3(GRP_BLOCK), 0, s>>>(
         d_runs, R, d_lo, d_anch, n_groups, cp, sp, d_order, d_keepw, d_changed, d_new_expire,
         d_ksz, d_vsz, nullptr, nullptr, d_stats);
 }
@@ -1885,7 +1889,7 @@ void launch_rank_grp_view(const DevRun *d_runs, int R, const uint64_t *d_lo,
         blocks = 3584;
     CompactParams cp{};
     ScanParams sp{};
-    k_rank_grp<1><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+    k_rank_grp<1><<<dim3((uint32_t)blocks), dim3(GRP_BLOCK), 0, s>>>(
         d_runs, R, d_lo, d_anch, n_groups, cp, sp, d_order, nullptr, nullptr, nullptr, nullptr,
         nullptr, nullptr, d_shadow, nullptr);
 }
@@ -1902,7 +1906,7 @@ void launch_rank_grp_count(const DevRun *d_runs, int R, const uint64_t *d_lo,
     if (blocks > 3584)
         blocks = 3584;
     CompactParams cp{};
-    k_rank_grp<2><<<dim3((uint32_t)blocks), dim3(BLOCK), 0, s>>>(
+    k_rank_grp<2><<<dim3((uint32_t)blocks), dim3(GRP_BLOCK), 0, s>>>(
         d_runs, R, d_lo, d_anch, n_groups, cp, sp, nullptr, nullptr, nullptr, nullptr, nullptr,
         nullptr, nullptr, nullptr, d_stats);
 }
