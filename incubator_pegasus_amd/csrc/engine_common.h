@@ -18,8 +18,10 @@
 #include <stdint.h>
 
 #define RRDB_MAX_RUNS 64
-/* tail-word rank modes: max run count; group-streaming rank group budget */
-#define LDST_MAXR 16
+/* tail-word rank modes: max run count (the group rank's org word packs the
+ * run in 16-GRP_ORG_SHIFT bits); group-streaming rank group budget */
+#define LDST_MAXR 32
+#define GRP_ORG_SHIFT 11
 /* group-size sweep (r02, same box): target 512 -> 0.362ms rank, 1024 ->
  * 0.316ms (fewer barriers per element), 2048 -> 0.441ms (80KB LDS drops to
  * 2 workgroups/CU).  1024 is the sweet spot. */

@@ -1470,11 +1470,14 @@ void launch_anchor_rows(const DevRun *runs, int R, int q0, const uint64_t *d_lo,
                                                            n_groups, d_anch);
 }
 
+static_assert(GRP_CAP <= (1 << GRP_ORG_SHIFT), "org segpos bits");
+static_assert(LDST_MAXR <= (1 << (16 - GRP_ORG_SHIFT)), "org run bits");
+
 /* (tail asc, run desc) strict order on (tail, org) pairs — tails are unique
  * within a run, so ties are always cross-run and org's run bits decide */
 __device__ static inline bool grp_lt(uint64_t ta, uint16_t oa, uint64_t tb, uint16_t ob)
 {
-    return ta < tb || (ta == tb && (oa >> 12) > (ob >> 12));
+    return ta < tb || (ta == tb && (oa >> GRP_ORG_SHIFT) > (ob >> GRP_ORG_SHIFT));
 }
 
 /* per-thread disposition tallies, flushed ONCE per block at kernel end into
@@ -1685,7 +1688,7 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
                     (MODE == 2 && s_allmeta) ? runs[q].meta : nullptr;
                 for (uint64_t j = threadIdx.x; j < len; j += blockDim.x) {
                     s_ta[off + j] = tq[a0 + j];
-                    s_oa[off + j] = (uint16_t)((q << 12) | (uint32_t)j);
+                    s_oa[off + j] = (uint16_t)((q << GRP_ORG_SHIFT) | (uint32_t)j);
                     if (mq)
                         s_meta[off + j] = mq[a0 + j];
                 }
@@ -1771,8 +1774,8 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
                 if (p < gsize) {
                     uint64_t myt = ft[p];
                     uint16_t org = fo[p];
-                    int q = org >> 12;
-                    uint64_t i = s_a0[q] + (org & 0xFFF);
+                    int q = org >> GRP_ORG_SHIFT;
+                    uint64_t i = s_a0[q] + (org & ((1u << GRP_ORG_SHIFT) - 1));
                     int shadow = 0;
                     if (p > 0) {
                         shadow = (ft[p - 1] == myt);
@@ -1789,9 +1792,9 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
                     uint64_t rank = s_base + p;
                     const uint64_t *sm = nullptr;
                     if (MODE == 2 && s_allmeta)
-                        sm = &s_meta[s_segoff[q] + (org & 0xFFF)];
+                        sm = &s_meta[s_segoff[q] + (org & ((1u << GRP_ORG_SHIFT) - 1))];
                     else if (MODE == 0 && s_allmeta)
-                        sm = &s_mt[s_segoff[q] + (org & 0xFFF)];
+                        sm = &s_mt[s_segoff[q] + (org & ((1u << GRP_ORG_SHIFT) - 1))];
                     disp = grp_epilogue<MODE>(runs, q, i, shadow, rank, cp, sp, order, keepw,
                                               changed, new_expire, ksz, vsz, shadowed, sm);
                 }

@@ -131,15 +131,17 @@ def test_grp_rank_oversized_group_fallback(oracle_lib, hip_lib):
 
 
 def test_grp_rank_many_runs_uses_fallback_mode(oracle_lib, hip_lib):
-    """R > LDST_MAXR (16): grp ineligible, host dispatches the standard
-    kernel — results must still match."""
+    """R > LDST_MAXR (32): grp ineligible, host dispatches the standard
+    kernel — results must still match.  (A 17-run compact regression also
+    lives here: the overlay run once pushed R one past the old 16-run org
+    limit and silently fell back 6x slower.)"""
     o = oracle_lib.open(1, 0, -1)
     g = hip_lib.open(1, 0, 0)
     try:
         g.set_envs({"engine.rank_mode": "grp"})
         seq = 1
-        for run in range(18):
-            ids = np.arange(run, 2000, 1, dtype=np.uint64)[::18][:80]
+        for run in range(34):
+            ids = np.arange(run, 2000, 1, dtype=np.uint64)[::34][:50]
             keys = D.make_raw_keys(np.sort(ids))
             recs = []
             for j in range(len(keys)):
