@@ -2745,9 +2745,29 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
     }
     /* completion (on_multi_get:777-788): kIncomplete iff the iterator is
      * still Valid() after a limit exit — even when the remaining records lie
-     * past the range */
+     * past the range.  Exception: the post-append c==0 break
+     * (:668-672/:739-744) completes when the LAST iterated record equals the
+     * inclusive far boundary, regardless of limits. */
     {
-        bool limit_exit = (count >= (int64_t)max_kv_count ||
+        bool boundary_hit = false;
+        uint64_t countable_chk = w - (skipped_first ? 1 : 0) - lo_skip;
+        if (iteration_count >= countable_chk && iteration_count > 0 && w == n && w > 0 &&
+            ((!q->reverse && stop_inclusive) || (q->reverse && start_inclusive))) {
+            /* last iterated in-range record = the far end of the view */
+            uint64_t vidx = q->reverse ? (lo_skip) : (n - 1);
+            const std::string &bnd = q->reverse ? start : stop;
+            uint32_t *d_eq = e->talloc<uint32_t>(4);
+            uint8_t *d_b = e->upload_tmp(bnd.data(), bnd.size());
+            launch_first_eq(dr, d_view + vidx, 1, d_b, bnd.size(), d_eq, e->stream);
+            uint32_t eq = 0;
+            HIP_OK(hipMemcpyAsync(&eq, d_eq, 4, hipMemcpyDeviceToHost, e->stream));
+            HIP_OK(hipStreamSynchronize(e->stream));
+            e->tfree(d_eq);
+            e->tfree(d_b);
+            boundary_hit = eq != 0;
+        }
+        bool limit_exit = !boundary_hit &&
+                          (count >= (int64_t)max_kv_count ||
                            iteration_count >= max_iteration_count || size >= max_iter_size);
         uint64_t countable = w - (skipped_first ? 1 : 0) - lo_skip;
         bool consumed_all = (iteration_count >= countable);

@@ -2692,11 +2692,45 @@ __device__ static void mg_core(const DevRun *runs, int R, const MgFusedArgs &a,
         }
         /* completion (on_multi_get:777-788): kIncomplete iff the iterator is
          * still Valid() after a limit exit — even when the remaining records
-         * lie past the range */
+         * lie past the range.  Exception: the post-append c==0
+         * break (:668-672/:739-744) completes when the LAST iterated record
+         * equals the inclusive far boundary, regardless of limits. */
         {
-            int limit_exit = (count >= a.max_kv_count || iteration >= a.max_iteration_count ||
-                              size >= a.max_iteration_size);
             int consumed_all = (iteration >= countable - (uint64_t)skipped_first);
+            int boundary_hit = 0;
+            if (consumed_all && iteration > 0 &&
+                ((!a.reverse && a.stop_inclusive) || (a.reverse && a.start_inclusive))) {
+                /* far-end iterated record: forward = highest non-invisible t,
+                 * reverse = lowest at/above t_floor */
+                int64_t t_far = -1;
+                if (!a.reverse) {
+                    for (int64_t t = (int64_t)total - 1; t >= (int64_t)t_floor; t--)
+                        if (s_state[t] != 2) {
+                            t_far = t;
+                            break;
+                        }
+                } else {
+                    for (uint64_t t = t_floor; t < total; t++)
+                        if (s_state[t] != 2) {
+                            t_far = (int64_t)t;
+                            break;
+                        }
+                }
+                if (t_far >= 0) {
+                    uint64_t id = s_id[t_far];
+                    const DevRun &r = runs[id >> 40];
+                    uint64_t i = id & 0xFFFFFFFFFFull, kl;
+                    const uint8_t *k = run_key(r, i, &kl);
+                    /* original stop = mg_stop minus the '\0' the host
+                     * appended for the inclusive exclusive-bound */
+                    const uint8_t *bnd = a.reverse ? mg_start : mg_stop;
+                    uint64_t blen = a.reverse ? mg_start_len : mg_stop_len - 1;
+                    boundary_hit = dev_key_cmp(k, kl, bnd, blen) == 0;
+                }
+            }
+            int limit_exit = !boundary_hit &&
+                             (count >= a.max_kv_count || iteration >= a.max_iteration_count ||
+                              size >= a.max_iteration_size);
             if (!consumed_all)
                 complete = 0; /* stopped mid-range */
             else if (!limit_exit)

@@ -1545,6 +1545,8 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
             lo_skip = 1;
     }
     uint64_t steps = n;
+    int boundary_hit = 0; /* post-append c==0 break (on_multi_get:668-672 /
+                             :739-744): completes REGARDLESS of limits */
     for (uint64_t s = 0; s < steps; s++) {
         uint64_t vi = q->reverse ? (steps - 1 - s) : s;
         if (q->reverse && lo_skip && vi == 0)
@@ -1562,6 +1564,13 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
         if (q->reverse && !stop_inclusive && key_cmp(k, kl, stop, stop_len) == 0)
             continue;
         iteration_count++;
+        if (!q->reverse) {
+            if (stop_inclusive && key_cmp(k, kl, stop, stop_len) == 0)
+                boundary_hit = 1;
+        } else {
+            if (start_inclusive && key_cmp(k, kl, start, start_len) == 0)
+                boundary_hit = 1;
+        }
         /* append_key_value_for_multi_get (:2462-2504) */
         if (ts_expired(epoch_now, extract_expire_ts(e->data_version, v, vl)))
             continue; /* kExpired */
@@ -1591,7 +1600,8 @@ int32_t rrdb_multi_get(void *h, const rrdb_multi_get_request *q, uint32_t epoch_
      * still Valid() after a limit exit — even when the remaining records lie
      * past the range */
     {
-        int limit_exit = (count >= (int64_t)max_kv_count ||
+        int limit_exit = !boundary_hit &&
+                         (count >= (int64_t)max_kv_count ||
                           iteration_count >= max_iteration_count ||
                           size >= max_iteration_size);
         uint64_t skipped_first = 0;

@@ -116,6 +116,102 @@ class Model:
         items = self.visible_items(lo, hi, True, False)
         return sum(1 for _, v in items if not expired(now, expire_of(v, self.data_version)))
 
+    def _iter_visible(self, lo, hi, lo_incl, hi_incl):
+        """iterator-level view: newest version per key is a PUT (expired
+        records stay visible at this level — expiry is app-level)."""
+        return self.visible_items(lo, hi, lo_incl, hi_incl)
+
+    def _valid_beyond(self, bound: bytes, reverse: bool) -> bool:
+        """it->Valid() after a limit exit: any iterator-visible record
+        beyond the boundary anywhere in the DB (on_multi_get:777-788)."""
+        if not reverse:
+            return len(self.visible_items(bound, None, True, False)) > 0
+        return len(self.visible_items(None, bound, True, False)) > 0
+
+    def multi_get(self, hash_key: bytes, now: int, *, start_sortkey=b"",
+                  stop_sortkey=b"", start_inclusive=True, stop_inclusive=False,
+                  max_kv_count=-1, reverse=False, engine_max_iter=3000):
+        """Range variant of on_multi_get (pegasus_server_impl.cpp:540-799),
+        restated independently: the forward/reverse iterator walk with the
+        range_read_limiter count cap and the iterator-validity completion
+        (kIncomplete iff a limit exit leaves the iterator Valid anywhere)."""
+        pre = struct.pack(">H", len(hash_key)) + hash_key
+        start = pre + start_sortkey
+        if stop_sortkey == b"":
+            hi = bytearray(pre)
+            p = len(hi) - 1
+            while hi[p] == 0xFF:
+                p -= 1
+            hi[p] += 1
+            stop = bytes(hi[:p + 1])
+            stop_incl = False
+        else:
+            stop = pre + stop_sortkey
+            stop_incl = stop_inclusive
+        c = (start > stop) - (start < stop)
+        if c > 0 or (c == 0 and (not start_inclusive or not stop_incl)):
+            return 0, []  # empty range
+        max_count = engine_max_iter if max_kv_count <= 0 else min(max_kv_count,
+                                                                  engine_max_iter)
+        rows = self._iter_visible(None, None, True, True)  # whole DB in key order
+        if reverse:
+            rows = list(reversed(rows))
+        out = []
+        count = 0
+        iteration = 0
+        complete = False
+        first_exclusive = (not start_inclusive) if not reverse else (not stop_incl)
+        it_valid = False
+        i = 0
+        # Seek: first row >= start (fwd) / last row <= stop (rev)
+        while i < len(rows):
+            k = rows[i][0]
+            if not reverse and k >= start:
+                break
+            if reverse and (k < stop or (k == stop and stop_incl)):
+                break
+            i += 1
+        while True:
+            if not (count < max_count and iteration < engine_max_iter):
+                it_valid = i < len(rows)
+                break
+            if i >= len(rows):
+                it_valid = False
+                break
+            k, v = rows[i]
+            if not reverse:
+                cc = (k > stop) - (k < stop)
+                if cc > 0 or (cc == 0 and not stop_incl):
+                    complete = True
+                    break
+                if first_exclusive:
+                    first_exclusive = False
+                    if k == start:
+                        i += 1
+                        continue
+            else:
+                cc = (k > start) - (k < start)
+                if cc < 0 or (cc == 0 and not start_inclusive):
+                    complete = True
+                    break
+                if first_exclusive:
+                    first_exclusive = False
+                    if k == stop:
+                        i += 1
+                        continue
+            iteration += 1
+            if not expired(now, expire_of(v, self.data_version)):
+                out.append((k[len(pre):], v[hdr_len(self.data_version):]))
+                count += 1
+            if cc == 0:
+                complete = True
+                break
+            i += 1
+        if reverse:
+            out.reverse()
+        error = 0 if (complete or not it_valid) else 7  # kIncomplete
+        return error, out
+
     def full_scan(self, now: int, validate_hash_req=True):
         """All visible non-expired rows (hash-valid) — count_data equivalent."""
         out = []

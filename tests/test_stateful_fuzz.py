@@ -71,6 +71,17 @@ class OracleVsModel(RuleBasedStateMachine):
         assert st_ == 0
         assert cnt == self.model.sortkey_count(hk, NOW)
 
+    @rule(hk=HK, start=SK, stop=SK, si=st.booleans(), pi=st.booleans(),
+          rev=st.booleans(), cap=st.sampled_from([-1, 1, 2, 1000]))
+    def check_multi_get(self, hk, start, stop, si, pi, rev, cap):
+        got = self.part.multi_get(hk, NOW, start_sortkey=start, stop_sortkey=stop,
+                                  start_inclusive=si, stop_inclusive=pi,
+                                  reverse=rev, max_kv_count=cap)
+        want = self.model.multi_get(hk, NOW, start_sortkey=start, stop_sortkey=stop,
+                                    start_inclusive=si, stop_inclusive=pi,
+                                    reverse=rev, max_kv_count=cap)
+        assert got == want, (hk, start, stop, si, pi, rev, cap)
+
     @rule()
     def check_scan_count(self):
         # the pipelined count API must agree with the model's full scan
