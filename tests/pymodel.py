@@ -130,7 +130,9 @@ class Model:
 
     def multi_get(self, hash_key: bytes, now: int, *, start_sortkey=b"",
                   stop_sortkey=b"", start_inclusive=True, stop_inclusive=False,
-                  max_kv_count=-1, reverse=False, engine_max_iter=3000):
+                  max_kv_count=-1, max_kv_size=-1, reverse=False,
+                  sort_key_filter_type=0, sort_key_filter_pattern=b"",
+                  engine_max_iter=3000):
         """Range variant of on_multi_get (pegasus_server_impl.cpp:540-799),
         restated independently: the forward/reverse iterator walk with the
         range_read_limiter count cap and the iterator-validity completion
@@ -148,16 +150,45 @@ class Model:
         else:
             stop = pre + stop_sortkey
             stop_incl = stop_inclusive
+        # prefix filter clamps the range (on_multi_get:558-578)
+        if sort_key_filter_type == 2 and len(sort_key_filter_pattern) > 0:
+            ps = pre + sort_key_filter_pattern
+            pe = bytearray(ps)
+            q = len(pe) - 1
+            while pe[q] == 0xFF:
+                q -= 1
+            pe[q] += 1
+            pe = bytes(pe[:q + 1])
+            if ps > start:
+                start = ps
+                start_inclusive = True
+            if pe <= stop:
+                stop = pe
+                stop_incl = False
         c = (start > stop) - (start < stop)
         if c > 0 or (c == 0 and (not start_inclusive or not stop_incl)):
             return 0, []  # empty range
         max_count = engine_max_iter if max_kv_count <= 0 else min(max_kv_count,
                                                                   engine_max_iter)
+        max_size = max_kv_size if max_kv_size > 0 else (1 << 31) - 1
+
+        def sk_filter_ok(sk: bytes) -> bool:
+            ft, pat = sort_key_filter_type, sort_key_filter_pattern
+            if ft == 0 or len(pat) == 0:
+                return True
+            if len(sk) < len(pat):
+                return False
+            if ft == 1:
+                return pat in sk
+            if ft == 2:
+                return sk.startswith(pat)
+            return sk.endswith(pat)
         rows = self._iter_visible(None, None, True, True)  # whole DB in key order
         if reverse:
             rows = list(reversed(rows))
         out = []
         count = 0
+        size = 0
         iteration = 0
         complete = False
         first_exclusive = (not start_inclusive) if not reverse else (not stop_incl)
@@ -172,7 +203,8 @@ class Model:
                 break
             i += 1
         while True:
-            if not (count < max_count and iteration < engine_max_iter):
+            if not (count < max_count and iteration < engine_max_iter and
+                    size < max_size):
                 it_valid = i < len(rows)
                 break
             if i >= len(rows):
@@ -200,9 +232,12 @@ class Model:
                         i += 1
                         continue
             iteration += 1
-            if not expired(now, expire_of(v, self.data_version)):
-                out.append((k[len(pre):], v[hdr_len(self.data_version):]))
+            sk = k[len(pre):]
+            if not expired(now, expire_of(v, self.data_version)) and sk_filter_ok(sk):
+                uv = v[hdr_len(self.data_version):]
+                out.append((sk, uv))
                 count += 1
+                size += len(sk) + len(uv)
             if cc == 0:
                 complete = True
                 break

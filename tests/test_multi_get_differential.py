@@ -47,7 +47,10 @@ def _cases(rnd, n):
         yield rnd.choice(HKS + [b"miss"]), dict(
             start_sortkey=rnd.choice(SKS), stop_sortkey=rnd.choice(SKS),
             start_inclusive=rnd.random() < 0.5, stop_inclusive=rnd.random() < 0.5,
-            reverse=rnd.random() < 0.5, max_kv_count=rnd.choice([-1, 1, 2, 3, 100]))
+            reverse=rnd.random() < 0.5, max_kv_count=rnd.choice([-1, 1, 2, 3, 100]),
+            max_kv_size=rnd.choice([-1, -1, 1, 7, 40]),
+            sort_key_filter_type=rnd.choice([0, 0, 1, 2, 3]),
+            sort_key_filter_pattern=rnd.choice([b"", b"s", b"1", b"zz"]))
 
 
 @pytest.mark.parametrize("seed", range(8))
