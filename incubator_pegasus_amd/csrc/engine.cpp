@@ -1066,6 +1066,21 @@ struct HipEngine {
         return d;
     }
 
+    /* ingest-path variant: a full device returns null (the caller unwinds
+     * and reports kIOError, like rocksdb's background write errors) instead
+     * of aborting the process */
+    uint8_t *upload_bytes_try(const void *p, uint64_t n)
+    {
+        uint8_t *d = nullptr;
+        if (hipMalloc(&d, n ? n : 1) != hipSuccess)
+            return nullptr;
+        if (n && hipMemcpy(d, p, n, hipMemcpyHostToDevice) != hipSuccess) {
+            (void)hipFree(d);
+            return nullptr;
+        }
+        return d;
+    }
+
     /* engine-owned persistent scratch arena for per-operation temporaries.
      * Bump-allocated; reset at the start of each C-ABI operation (no live
      * scratch crosses operations — persistent data uses plain hipMalloc).
@@ -1347,18 +1362,24 @@ std::once_flag g_init_once;
 
 static int32_t engine_flush(HipEngine *e); /* write path, defined below */
 
-/* internal: upload a prepared sorted run (validation already established) */
-static void ingest_prepared(HipEngine *e, const std::string &keys,
-                            const std::vector<uint64_t> &koff, const std::string &vals,
-                            const std::vector<uint64_t> &voff, const std::vector<uint64_t> &sk)
+/* internal: upload a prepared sorted run (validation already established);
+ * kIOError (with the partial run unwound) when the device is full */
+static int32_t ingest_prepared(HipEngine *e, const std::string &keys,
+                               const std::vector<uint64_t> &koff, const std::string &vals,
+                               const std::vector<uint64_t> &voff,
+                               const std::vector<uint64_t> &sk)
 {
     RunBuf r;
     r.n = sk.size();
-    r.keys = e->upload_bytes(keys.data(), keys.size());
-    r.koff = (uint64_t *)e->upload_bytes(koff.data(), koff.size() * 8);
-    r.vals = e->upload_bytes(vals.data(), vals.size());
-    r.voff = (uint64_t *)e->upload_bytes(voff.data(), voff.size() * 8);
-    r.sk = (uint64_t *)e->upload_bytes(sk.data(), sk.size() * 8);
+    r.keys = e->upload_bytes_try(keys.data(), keys.size());
+    r.koff = (uint64_t *)e->upload_bytes_try(koff.data(), koff.size() * 8);
+    r.vals = e->upload_bytes_try(vals.data(), vals.size());
+    r.voff = (uint64_t *)e->upload_bytes_try(voff.data(), voff.size() * 8);
+    r.sk = (uint64_t *)e->upload_bytes_try(sk.data(), sk.size() * 8);
+    if (!r.keys || !r.koff || !r.vals || !r.voff || !r.sk) {
+        e->free_run(r);
+        return RRDB_IO_ERROR;
+    }
     r.fixed_klen = detect_fixed_klen(koff.data(), r.n);
     r.fixed_vlen = detect_fixed_vlen(voff.data(), r.n);
     r.fixed_vlen_put = detect_fixed_vlen_put(voff.data(), sk.data(), r.n);
@@ -1372,6 +1393,7 @@ static void ingest_prepared(HipEngine *e, const std::string &keys,
     e->runs.push_back(r);
     e->d_runs_dirty = true;
     e->ldst_elig_cache = -1;
+    return RRDB_OK;
 }
 
 extern "C" {
@@ -1594,11 +1616,15 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
     e->activate();
     RunBuf r;
     r.n = n;
-    r.keys = e->upload_bytes(keys, key_offs[n]);
-    r.koff = (uint64_t *)e->upload_bytes(key_offs, (n + 1) * 8);
-    r.vals = e->upload_bytes(values, val_offs[n]);
-    r.voff = (uint64_t *)e->upload_bytes(val_offs, (n + 1) * 8);
-    r.sk = (uint64_t *)e->upload_bytes(seq_kind, n * 8);
+    r.keys = e->upload_bytes_try(keys, key_offs[n]);
+    r.koff = (uint64_t *)e->upload_bytes_try(key_offs, (n + 1) * 8);
+    r.vals = e->upload_bytes_try(values, val_offs[n]);
+    r.voff = (uint64_t *)e->upload_bytes_try(val_offs, (n + 1) * 8);
+    r.sk = (uint64_t *)e->upload_bytes_try(seq_kind, n * 8);
+    if (!r.keys || !r.koff || !r.vals || !r.voff || !r.sk) {
+        e->free_run(r);
+        return RRDB_IO_ERROR;
+    }
     r.fixed_klen = detect_fixed_klen(key_offs, n);
     r.fixed_vlen = detect_fixed_vlen(val_offs, n);
     r.fixed_vlen_put = detect_fixed_vlen_put(val_offs, seq_kind, n);
@@ -1978,7 +2004,11 @@ int32_t rrdb_scan_open(void *h, const rrdb_scan_request *q, uint32_t epoch_now, 
 {
     auto *e = (HipEngine *)h;
     std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
-    engine_flush((HipEngine *)h); /* memtable visible to reads */
+    if (engine_flush((HipEngine *)h) != RRDB_OK) {
+        result_init(out);
+        out->error = RRDB_IO_ERROR; /* device full; writes retained */
+        return out->error;
+    }
     ((HipEngine *)h)->scratch_reset();
     Arena *a = result_init(out);
     if (q->hash_key_filter_type < 0 || q->hash_key_filter_type > 3 ||
@@ -2267,7 +2297,11 @@ int32_t rrdb_sortkey_count(void *h, const uint8_t *hash_key, uint64_t hklen, uin
 {
     auto *e = (HipEngine *)h;
     std::lock_guard<std::mutex> g(((HipEngine *)h)->mu);
-    engine_flush((HipEngine *)h); /* memtable visible to reads */
+    if (engine_flush((HipEngine *)h) != RRDB_OK) {
+        result_init(out);
+        out->error = RRDB_IO_ERROR;
+        return out->error;
+    }
     ((HipEngine *)h)->scratch_reset();
     result_init(out);
     e->activate();
@@ -2367,7 +2401,11 @@ static int32_t multi_get_locked(void *h, const rrdb_multi_get_request *q, uint32
                                 rrdb_result *out)
 {
     auto *e = (HipEngine *)h;
-    engine_flush(e); /* memtable visible to reads */
+    if (engine_flush(e) != RRDB_OK) {
+        result_init(out);
+        out->error = RRDB_IO_ERROR;
+        return out->error;
+    }
     e->scratch_reset();
     Arena *a = result_init(out);
     e->activate();
@@ -3271,7 +3309,9 @@ static int32_t engine_flush(HipEngine *e)
         voff.push_back(vals.size());
         sk.push_back((std::get<1>(kv.second) << 1) | (uint64_t)std::get<2>(kv.second));
     }
-    ingest_prepared(e, keys, koff, vals, voff, sk);
+    int32_t rc = ingest_prepared(e, keys, koff, vals, voff, sk);
+    if (rc != RRDB_OK)
+        return rc; /* memtable retained: the write is not lost */
     e->memtable.clear();
     return RRDB_OK;
 }
