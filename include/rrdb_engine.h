@@ -115,6 +115,14 @@ typedef struct {
     uint64_t n_sort_keys;
     const uint8_t *sort_keys;      /* packed */
     const uint64_t *sort_key_offs; /* [n_sort_keys+1] */
+    /* engine extension (batched path): leave the packed result blobs in
+     * HBM — rrdb_result.dev_vals = the packed blob (one region per request,
+     * layout [koff (n+1)*8][voff (n+1)*8][keys][vals]), dev_val_offs = a
+     * DEVICE array of per-request blob offsets [n_req+1], group_counts /
+     * group_errors host-side as usual.  Best-effort: when any request fell
+     * back to the general path the call returns the normal host-marshalled
+     * slices and dev_vals stays null. */
+    uint8_t on_device_out;
 } rrdb_multi_get_request;
 
 /* on_get_scanner request — thrift get_scanner_request (idl/rrdb.thrift:313-329) */

@@ -76,6 +76,7 @@ class _MultiGetRequest(C.Structure):
         ("n_sort_keys", C.c_uint64),
         ("sort_keys", C.POINTER(C.c_uint8)),
         ("sort_key_offs", C.POINTER(C.c_uint64)),
+        ("on_device_out", C.c_uint8),
     ]
 
 

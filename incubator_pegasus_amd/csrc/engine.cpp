@@ -3609,6 +3609,36 @@ extern "C" int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *
             total += used[i];
         }
         pack_off[n_req] = total;
+        bool all_fused = true;
+        for (uint64_t i = 0; i < n_req; i++)
+            if (hdrs[i * 4] < 0)
+                all_fused = false;
+        if (total && shared->on_device_out && all_fused) {
+            /* device-out: pack into caller-owned device memory, no host
+             * marshal (the batched e2e limiter on slow-host boxes) */
+            uint8_t *d_packed = nullptr;
+            uint64_t *d_poff_out = nullptr;
+            HIP_OK(hipMalloc(&d_packed, total));
+            HIP_OK(hipMalloc(&d_poff_out, (n_req + 1) * 8));
+            HIP_OK(hipMemcpyAsync(d_poff_out, pack_off.data(), (n_req + 1) * 8,
+                                  hipMemcpyHostToDevice, e->stream));
+            uint64_t *d_used = (uint64_t *)e->upload_tmp(used.data(), n_req * 8);
+            uint64_t *d_poff = (uint64_t *)e->upload_tmp(pack_off.data(), (n_req + 1) * 8);
+            launch_pack_blobs(d_blobs, BLOB_STRIDE, n_req, d_used, d_poff, d_packed, e->stream);
+            HIP_OK(hipStreamSynchronize(e->stream));
+            uint64_t rows = 0;
+            for (uint64_t i = 0; i < n_req; i++) {
+                out->group_counts[i] = (uint64_t)hdrs[i * 4];
+                out->group_errors[i] = hdrs[i * 4 + 1] ? RRDB_OK : RRDB_INCOMPLETE;
+                rows += (uint64_t)hdrs[i * 4];
+            }
+            out->dev_vals = d_packed;
+            out->dev_val_offs = d_poff_out;
+            a->dev_ptrs.insert(a->dev_ptrs.end(), {(void *)d_packed, (void *)d_poff_out});
+            out->count = rows;
+            out->error = RRDB_OK;
+            return RRDB_OK;
+        }
         if (total) {
             uint64_t *d_used = (uint64_t *)e->upload_tmp(used.data(), n_req * 8);
             uint64_t *d_poff = (uint64_t *)e->upload_tmp(pack_off.data(), (n_req + 1) * 8);

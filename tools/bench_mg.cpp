@@ -151,12 +151,45 @@ int main(int argc, char **argv)
         rrdb_free(&res);
     }
     auto bel = std::chrono::duration<double>(std::chrono::steady_clock::now() - bt0).count();
+    /* device-out batched: results stay in HBM (the serving shim would
+     * serialize from a pinned mirror); measures the ABI rate without the
+     * host slice marshal */
+    uint64_t drows = 0, dcalls = 0;
+    auto dt0 = std::chrono::steady_clock::now();
+    for (uint64_t pass = 0; pass < NB; pass++) {
+        bhks.clear();
+        boffs.resize(1);
+        for (uint64_t i = 0; i < B; i++) {
+            uint64_t id = splitmix64(pass * B + i + 555) % n_hash;
+            snprintf(hk, sizeof(hk), "u:%014llu", (unsigned long long)id);
+            bhks.append(hk, 16);
+            boffs.push_back(bhks.size());
+        }
+        rrdb_multi_get_request shared{};
+        shared.start_inclusive = 1;
+        shared.max_kv_count = -1;
+        shared.max_kv_size = -1;
+        shared.on_device_out = 1;
+        rrdb_result res;
+        st = rrdb_mgb(h, B, (const uint8_t *)bhks.data(), boffs.data(), &shared, 1000000,
+                      &res);
+        if (st != 0) {
+            printf("mg_batch devout failed %d\n", st);
+            return 1;
+        }
+        drows += res.count;
+        dcalls += B;
+        rrdb_free(&res);
+    }
+    auto del = std::chrono::duration<double>(std::chrono::steady_clock::now() - dt0).count();
     printf("{\"multi_get_ops_per_s\": %.1f, \"rows_per_s\": %.1f, \"rows\": %llu, "
            "\"us_per_call\": %.1f, \"calls\": %llu, "
            "\"batched_ops_per_s\": %.1f, \"batched_rows_per_s\": %.1f, "
-           "\"batched_rows\": %llu, \"host\": \"c++\"}\n",
+           "\"batched_rows\": %llu, "
+           "\"batched_devout_ops_per_s\": %.1f, \"batched_devout_rows_per_s\": %.1f, "
+           "\"host\": \"c++\"}\n",
            calls / el, rows / el, (unsigned long long)rows, el * 1e6 / calls,
            (unsigned long long)calls, bcalls / bel, brows / bel,
-           (unsigned long long)brows);
+           (unsigned long long)brows, dcalls / del, drows / del);
     return 0;
 }
