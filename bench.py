@@ -343,6 +343,10 @@ def main():
     from incubator_pegasus_amd import data as D2
 
     hk_pp = args.ycsb_hashkeys_per_part
+    if world > 1:
+        # configs[3] is a single-GPU config; keep the per-rank side tables
+        # small on scaling runs so N=8 setup stays minutes, not tens
+        hk_pp = min(hk_pp, 100_000)
     e_parts = []
     for p_ in range(args.partitions):
         ep = hip.open(2, p_, local_rank)
