@@ -490,3 +490,44 @@ def test_batched_device_out_counts(oracle_lib, hip_lib):
             g._L.rrdb_free_result(C.byref(res))
     finally:
         g.close()
+
+
+def test_batched_device_out_fallback_is_host_marshalled(hip_lib):
+    """Best-effort contract: when any request of a device-out batch falls
+    back to the general path, the call returns normal host slices and
+    dev_vals stays null."""
+    import ctypes as C
+    import numpy as np
+
+    from incubator_pegasus_amd.capi import _MultiGetRequest, _Result, _cslice
+
+    g = hip_lib.open(1, 0, 0)
+    try:
+        recs = [(D.generate_key(b"fbk%02d" % i, b"s"),
+                 D.encode_value(b"v%d" % i, 0, i + 1, 1), i + 1, 0) for i in range(20)]
+        recs.sort(key=lambda r: r[0])
+        g.ingest_run(recs)
+        hks = [b"fbk%02d" % i for i in range(3)] + [b"L" * 5000]  # 5000B hk -> fallback
+        blob = b"".join(hks)
+        offs = np.zeros(len(hks) + 1, dtype=np.uint64)
+        np.cumsum([len(h) for h in hks], out=offs[1:])
+        keep = []
+        req = _MultiGetRequest()
+        req.hash_key = _cslice(b"", keep)
+        req.start_inclusive = 1
+        req.max_kv_count = -1
+        req.max_kv_size = -1
+        req.on_device_out = 1
+        res = _Result()
+        barr = np.frombuffer(blob, dtype=np.uint8)
+        g._L.rrdb_multi_get_batch(
+            g._h, len(hks), np.ascontiguousarray(barr).ctypes.data_as(C.c_void_p),
+            offs.ctypes.data_as(C.c_void_p), C.byref(req), NOW, C.byref(res))
+        try:
+            assert res.error == OK
+            assert not res.dev_vals  # host-marshalled best-effort path
+            assert [res.group_counts[i] for i in range(4)] == [1, 1, 1, 0]
+        finally:
+            g._L.rrdb_free_result(C.byref(res))
+    finally:
+        g.close()
