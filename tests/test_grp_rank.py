@@ -531,3 +531,37 @@ def test_batched_device_out_fallback_is_host_marshalled(hip_lib):
             g._L.rrdb_free_result(C.byref(res))
     finally:
         g.close()
+
+
+def test_fused_count_hash_validation(oracle_lib, hip_lib):
+    """Fused count with partition-hash validation (validate_key_value_for_scan
+    crc64 branch, pegasus_server_impl.cpp:2382-2432): counts only keys whose
+    crc64(hashkey) & partition_version == pidx, matching the oracle and the
+    paged path."""
+    o = oracle_lib.open(1, 2, -1)  # pidx 2
+    g = hip_lib.open(1, 2, 0)
+    try:
+        envs = {"replica.split.validate_partition_hash": "true",
+                "rocksdb.max_iteration_count": str(2**31 - 1)}
+        for p in (o, g):
+            p.set_envs(envs)
+            p.set_partition_version(3)  # 4-way mask
+        recs = []
+        for i in range(2000):
+            recs.append((D.generate_key(b"hv%05d" % i, b""),
+                         D.encode_value(b"v", 0, i + 1, 1), i + 1, 0))
+        recs.sort(key=lambda r: r[0])
+        o.ingest_run(recs)
+        g.ingest_run(recs)
+        co, fo = _count_fused_or_fallback(o, NOW, validate_partition_hash=True)
+        cg, fg = _count_fused_or_fallback(g, NOW, validate_partition_hash=True)
+        assert fg, "expected the fused path"
+        assert co == cg
+        assert 0 < cg < 2000  # the mask really filtered
+        res = g.scan_open(b"\x00\x00", b"\xff\xff", NOW, only_return_count=True,
+                          full_scan=True, batch_size=2**31 - 1,
+                          validate_partition_hash=True)
+        assert res.kv_count == cg
+    finally:
+        o.close()
+        g.close()
