@@ -89,6 +89,8 @@ void launch_rank_grp_view(const DevRun *, int, const uint64_t *, const uint64_t 
                           uint64_t *, uint8_t *, hipStream_t);
 void launch_rank_grp_count(const DevRun *, int, const uint64_t *, const uint64_t *, uint64_t,
                            const ScanParams &, CompactStatsDev *, hipStream_t);
+void launch_count_single(const DevRun *, const uint64_t *, const uint64_t *,
+                         const ScanParams &, CompactStatsDev *, uint64_t, hipStream_t);
 void launch_rank_compact_ldst(const DevRun *, int, const uint64_t *, const uint64_t *,
                               const uint64_t *, uint64_t, const CompactParams &, uint64_t *,
                               uint64_t *, uint8_t *, uint32_t *, uint64_t *, uint64_t *,
@@ -2141,8 +2143,16 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     for (auto &r : e->runs)
         total += r.n;
     uint64_t batch_cap = q->batch_size > 0 ? (uint64_t)q->batch_size : (uint64_t)INT32_MAX;
-    if (!e->grp_eligible() || batch_cap < total || e->max_iter_count < total)
+    int R0 = (int)e->runs.size();
+    bool single = R0 == 1; /* no merge/shadow needed: plain count kernel */
+    if ((R0 > 1 && !e->grp_eligible()) || batch_cap < total || e->max_iter_count < total)
         return RRDB_INVALID_ARGUMENT;
+    if (R0 == 0) {
+        e->pend_scan.active = true;
+        e->pend_scan.trivial = true;
+        e->pend_scan.trivial_count = 0;
+        return RRDB_OK;
+    }
 
     std::string start((const char *)q->start_key.data, q->start_key.len);
     std::string stop((const char *)q->stop_key.data, q->stop_key.len);
@@ -2200,28 +2210,31 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     uint8_t *d_stop_b = upl((const uint8_t *)stop_excl.data(), stop_excl.size());
     launch_bounds(dr, R, d_start_b, start.size(), ps.d_lo, 0, e->stream);
     launch_bounds(dr, R, d_stop_b, stop_excl.size(), ps.d_hi, 0, e->stream);
-    /* anchors from the largest run's FULL size (the window is device-side
-     * only); rows past the window collapse to sentinels in-kernel */
-    int q0 = 0;
-    uint64_t nmax = 0;
-    for (int r = 0; r < R; r++)
-        if (e->runs[r].n > nmax) {
-            nmax = e->runs[r].n;
-            q0 = r;
+    uint64_t n_groups = 0;
+    if (!single) {
+        /* anchors from the largest run's FULL size (the window is device-
+         * side only); rows past the window collapse to sentinels in-kernel */
+        int q0 = 0;
+        uint64_t nmax = 0;
+        for (int r = 0; r < R; r++)
+            if (e->runs[r].n > nmax) {
+                nmax = e->runs[r].n;
+                q0 = r;
+            }
+        int gs = 4;
+        while ((1ull << (gs + 1)) * (uint64_t)R <= GRP_TARGET && gs < 8)
+            gs++;
+        n_groups = (nmax + (1ull << gs) - 1) >> gs;
+        if (n_groups == 0)
+            n_groups = 1;
+        if ((n_groups + 1) * R * 8 > ps.anch_cap) {
+            if (ps.d_anch)
+                (void)hipFree(ps.d_anch);
+            HIP_OK(hipMalloc(&ps.d_anch, (n_groups + 1) * R * 8));
+            ps.anch_cap = (n_groups + 1) * R * 8;
         }
-    int gs = 4;
-    while ((1ull << (gs + 1)) * (uint64_t)R <= GRP_TARGET && gs < 8)
-        gs++;
-    uint64_t n_groups = (nmax + (1ull << gs) - 1) >> gs;
-    if (n_groups == 0)
-        n_groups = 1;
-    if ((n_groups + 1) * R * 8 > ps.anch_cap) {
-        if (ps.d_anch)
-            (void)hipFree(ps.d_anch);
-        HIP_OK(hipMalloc(&ps.d_anch, (n_groups + 1) * R * 8));
-        ps.anch_cap = (n_groups + 1) * R * 8;
+        launch_anchor_rows(dr, R, q0, ps.d_lo, ps.d_hi, gs, n_groups, ps.d_anch, e->stream);
     }
-    launch_anchor_rows(dr, R, q0, ps.d_lo, ps.d_hi, gs, n_groups, ps.d_anch, e->stream);
 
     ScanParams sp{};
     sp.epoch_now = epoch_now;
@@ -2241,7 +2254,10 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     if (!ps.d_stats)
         HIP_OK(hipMalloc(&ps.d_stats, 8 * sizeof(CompactStatsDev)));
     HIP_OK(hipMemsetAsync(ps.d_stats, 0, 8 * sizeof(CompactStatsDev), e->stream));
-    launch_rank_grp_count(dr, R, ps.d_lo, ps.d_anch, n_groups, sp, ps.d_stats, e->stream);
+    if (single)
+        launch_count_single(dr, ps.d_lo, ps.d_hi, sp, ps.d_stats, e->runs[0].n, e->stream);
+    else
+        launch_rank_grp_count(dr, R, ps.d_lo, ps.d_anch, n_groups, sp, ps.d_stats, e->stream);
     ps.active = true;
     return RRDB_OK;
 }

@@ -1897,6 +1897,26 @@ void launch_rank_grp_view(const DevRun *d_runs, int R, const uint64_t *d_lo,
         nullptr, nullptr, d_shadow, nullptr);
 }
 
+/* single-run count: no merge, no shadow — a grid-stride pass of
+ * dev_count_ok over the window (the common post-compaction state) */
+__global__ void k_count_single(const DevRun *runs, const uint64_t *lo, const uint64_t *hi,
+                               ScanParams sp, CompactStatsDev *stats)
+{
+    GrpTally tally;
+    uint64_t l = lo[0], h = hi[0];
+    for (uint64_t i = l + blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; i < h;
+         i += gridDim.x * (uint64_t)blockDim.x)
+        tally.add(dev_count_ok(runs[0], i, sp) ? D_KEEP : D_NONE);
+    tally.flush(stats);
+}
+
+void launch_count_single(const DevRun *d_runs, const uint64_t *d_lo, const uint64_t *d_hi,
+                         const ScanParams &sp, CompactStatsDev *d_stats, uint64_t n_max,
+                         hipStream_t s)
+{
+    k_count_single<<<grid_for(n_max, BLOCK), BLOCK, 0, s>>>(d_runs, d_lo, d_hi, sp, d_stats);
+}
+
 /* fused count scan: shadow + countability evaluated in-kernel, the count
  * lands in stats->output_records (banked); zero intermediate arrays */
 void launch_rank_grp_count(const DevRun *d_runs, int R, const uint64_t *d_lo,
