@@ -836,8 +836,7 @@ struct HipEngine {
         int64_t trivial_count = 0;
         CompactStatsDev *d_stats = nullptr;
         uint64_t *d_anch = nullptr, *d_lo = nullptr, *d_hi = nullptr;
-        uint8_t *d_start = nullptr, *d_stop = nullptr, *d_hk_pat = nullptr,
-                *d_sk_pat = nullptr;
+        uint8_t *d_start = nullptr; /* pooled request-bytes slab */
         /* pooled capacities: buffers persist across calls, realloc only on
          * growth (8 hipMallocs per begin measurably taxed the 16-partition
          * fan-out on slow-host boxes) */
@@ -1459,8 +1458,7 @@ void rrdb_close(void *h)
     {
         auto &ps = e->pend_scan;
         for (void *p : {(void *)ps.d_stats, (void *)ps.d_anch, (void *)ps.d_lo,
-                        (void *)ps.d_hi, (void *)ps.d_start, (void *)ps.d_stop,
-                        (void *)ps.d_hk_pat, (void *)ps.d_sk_pat})
+                        (void *)ps.d_hi, (void *)ps.d_start})
             if (p)
                 (void)hipFree(p);
     }
