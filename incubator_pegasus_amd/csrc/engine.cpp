@@ -2131,7 +2131,8 @@ int32_t rrdb_scan_count_begin(void *h, const rrdb_scan_request *q, uint32_t epoc
     std::lock_guard<std::mutex> g(e->mu);
     if (e->pend_scan.active)
         return RRDB_INVALID_ARGUMENT;
-    engine_flush(e);
+    if (engine_flush(e) != RRDB_OK)
+        return RRDB_IO_ERROR;
     e->activate();
     if (!q->only_return_count || !q->start_inclusive || q->hash_key_filter_type < 0 ||
         q->hash_key_filter_type > 3 || q->sort_key_filter_type < 0 ||
@@ -2892,7 +2893,8 @@ static int32_t compact_begin(HipEngine *e, const rrdb_compact_options *opts, uin
 {
     if (e->pend.active)
         return RRDB_INVALID_ARGUMENT;
-    engine_flush(e); /* memtable visible to reads */
+    if (engine_flush(e) != RRDB_OK) /* memtable visible to reads */
+        return RRDB_IO_ERROR;
     e->scratch_reset();
     rrdb_compact_stats st{};
     if (e->manual_compact_disabled)
@@ -3555,7 +3557,11 @@ extern "C" int32_t rrdb_multi_get_batch(void *h, uint64_t n_req, const uint8_t *
 {
     auto *e = (HipEngine *)h;
     std::lock_guard<std::mutex> g(e->mu);
-    engine_flush(e);
+    if (engine_flush(e) != RRDB_OK) {
+        result_init(out);
+        out->error = RRDB_IO_ERROR;
+        return out->error;
+    }
     e->activate();
     /* bulk path: retire the resident serving kernel so its poll windows do
      * not share hardware queues with the 4096-workgroup batch launches */
