@@ -7,12 +7,14 @@
  * is greater than every seqno in run r (enforced at ingest), which makes
  * "newest version wins" a run-index comparison for equal keys.
  *
- * Merge strategy (MI355X-native, no sequential merging iterator):
- *   rank(record) = own index + sum over other runs of bounded binary-search
- *   counts with (key asc, run desc) tie-break; scatter by rank; a record is
- *   shadowed iff its predecessor in rank order carries the same key.  This is
- *   fully data-parallel and replaces rocksdb's merging-iterator heap
- *   (reference behavior described in SURVEY.md §3.2/§8(c)).
+ * Merge strategy (MI355X-native, no sequential merging iterator): fully
+ * data-parallel rank merging with (key asc, run desc) ordering; a record is
+ * shadowed iff its predecessor in rank order carries the same key.  The
+ * round-2 production form is the group-streaming merge-path rank
+ * (kernels.hip k_rank_grp): anchor keys cut the merged order into groups
+ * whose per-run segments are disjoint, so each group's workgroup streams
+ * its tail words through LDS exactly once and position-in-group IS local
+ * rank.  Replaces rocksdb's merging-iterator heap.
  */
 #pragma once
 #include <stdint.h>

@@ -6,12 +6,27 @@
  * pegasus_server_impl drives (call sites: pegasus_server_impl.cpp:441,618,
  * 804,948,1243,3389).
  *
- * Design notes (MI355X):
- *  - wave64; blocks of 256 threads; grids capped ~2048 blocks + grid-stride
- *  - merge is rank-based and fully data-parallel (engine_common.h header
- *    comment); no sequential heap anywhere
- *  - emit copies are wave-per-record with lane-strided bytes: consecutive
- *    lanes touch consecutive bytes -> coalesced 64B segments
+ * Design notes (MI355X, round-2 state):
+ *  - wave64; 256-thread workgroups; capped grids with grid-stride loops
+ *  - the production merge is the GROUP-STREAMING rank (k_rank_grp): anchor
+ *    keys partition every run into disjoint per-group segments; one
+ *    workgroup streams its group's packed tail words through LDS once and
+ *    merges them with pairwise merge-path rounds — global traffic is one
+ *    8B tail read per record and every output write lands in a contiguous
+ *    rank range.  MODE 0 = compaction (fused filter via the per-run
+ *    (expire<<32)|kind meta column), MODE 1 = scan view build, MODE 2 =
+ *    fused count scan.  The r01 probe-based kernels (k_rank_compact[_ldst],
+ *    bound tables) remain as fallbacks for ineligible run sets and A/B.
+ *  - disposition tallies accumulate in per-thread registers and flush once
+ *    per block into 8 stat banks (shared-counter atomics serialized at
+ *    ~88 adds/us and floored every fused kernel before)
+ *  - emits: direct-indexed 16B-chunk copies for 16B-aligned fixed strides,
+ *    row-per-thread u64 copies for other fixed strides, chunk-anchor
+ *    binary-search copies for variable layouts
+ *  - serving: cooperative p-ary bounds + hashkey-prefix blooms in the fused
+ *    multi_get; a hipGraph-captured lane and a persistent mailbox-polled
+ *    1-workgroup server (bounded 2ms idle exit) remove the per-call
+ *    dispatch cost
  *  - crc64 table lives in __device__ memory (L1/L2-resident after first use)
  *  - integer/byte work only: HBM-bound, no MFMA by design
  */
