@@ -35,6 +35,7 @@ def main():
     o.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
     checks = 0
     tmp = tempfile.mkdtemp(prefix="stress_ckpt_")
+    tmp_o = tempfile.mkdtemp(prefix="stress_ckpt_o_")
     decree = 0
     for step in range(steps):
         op = rnd.randrange(10)
@@ -94,15 +95,22 @@ def main():
             assert g.manual_compact(now) == o.manual_compact(now)
             checks += 1
         elif op == 8 and g.num_records() > 0:  # checkpoint + cross restore
+            # checkpoint BOTH handles: checkpoint flushes the memtable, and
+            # flush-point-dependent compact stats (input_records/shadowed)
+            # only compare when both sides saw the same flush boundaries
             decree += 1
             assert g.checkpoint(tmp, decree) == OK
+            assert o.checkpoint(tmp_o, decree) == OK
             r = hip.open(1, 1, 0)
+            ro_ = orc.open(1, 1, -1)
             try:
                 assert r.restore(tmp, decree) == OK
+                assert ro_.restore(tmp_o, decree) == OK
                 k = D.generate_key(b"hk%03d" % rnd.randrange(60), b"s00")
-                assert r.get(k, now) == g.get(k, now)
+                assert r.get(k, now) == g.get(k, now) == ro_.get(k, now)
             finally:
                 r.close()
+                ro_.close()
             checks += 1
         else:  # sortkey_count
             hk = b"hk%03d" % rnd.randrange(70)
