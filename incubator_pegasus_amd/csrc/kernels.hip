@@ -1648,7 +1648,8 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
     __shared__ uint64_t s_meta[MCAP];
     __shared__ uint64_t s_a0[LDST_MAXR], s_seglen[LDST_MAXR], s_segoff[LDST_MAXR + 1];
     __shared__ uint64_t s_btail[LDST_MAXR];
-    __shared__ uint64_t s_loff[LDST_MAXR + 1]; /* current merge-list offsets */
+    __shared__ uint64_t s_loff[LDST_MAXR + 1]; /* merge-list offsets (ping) */
+    __shared__ uint64_t s_loff2[LDST_MAXR + 1]; /* merge-list offsets (pong) */
     __shared__ uint64_t s_base, s_gsize;
     __shared__ uint32_t s_bmask;
     __shared__ int s_nl, s_allmeta;
@@ -1709,15 +1710,26 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
                 }
             }
             __syncthreads();
-            /* pairwise merge rounds; ping-pong a->b->a->... */
+            /* pairwise merge rounds; ping-pong a->b->a->...  The list-offset
+             * table ping-pongs too: the next round's offsets are a pure
+             * function of this round's (loff'[x] = loff[2x]), so a few lanes
+             * write them into the pong table DURING the merge loop and each
+             * round needs ONE barrier (the former thread-0 rebuild +
+             * second barrier measured as part of the ~88%-parked SQ time) */
             uint64_t E = (gsize + blockDim.x - 1) / blockDim.x;
             int cur = 0;
-            while (s_nl > 1) {
+            int nl = s_nl;
+            while (nl > 1) {
                 const uint64_t *st = cur ? s_tb : s_ta;
                 const uint16_t *so = cur ? s_ob : s_oa;
                 uint64_t *dt = cur ? s_ta : s_tb;
                 uint16_t *do_ = cur ? s_oa : s_ob;
-                int nl = s_nl;
+                const uint64_t *lofc = cur ? s_loff2 : s_loff;
+                uint64_t *lofn = cur ? s_loff : s_loff2;
+                int nn = (nl + 1) / 2;
+                if (threadIdx.x <= (unsigned)nn)
+                    lofn[threadIdx.x] =
+                        (2 * (int)threadIdx.x <= nl) ? lofc[2 * threadIdx.x] : gsize;
                 uint64_t slot = (uint64_t)threadIdx.x * E;
                 uint64_t send = slot + E;
                 if (send > gsize)
@@ -1725,11 +1737,11 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
                 while (slot < send) {
                     /* pair j covering this slot */
                     int j = 0;
-                    while (2 * j + 2 < nl + 1 && s_loff[2 * j + 2] <= slot)
+                    while (2 * j + 2 < nl + 1 && lofc[2 * j + 2] <= slot)
                         j++;
-                    uint64_t ka = s_loff[2 * j];
-                    uint64_t kb = (2 * j + 1 <= nl) ? s_loff[2 * j + 1] : gsize;
-                    uint64_t ke = (2 * j + 2 <= nl) ? s_loff[2 * j + 2] : gsize;
+                    uint64_t ka = lofc[2 * j];
+                    uint64_t kb = (2 * j + 1 <= nl) ? lofc[2 * j + 1] : gsize;
+                    uint64_t ke = (2 * j + 2 <= nl) ? lofc[2 * j + 2] : gsize;
                     uint64_t la = kb - ka, lb = ke - kb;
                     uint64_t k = slot - ka;
                     /* diagonal: i = #A elements among the first k of the pair */
@@ -1757,14 +1769,8 @@ __global__ void __launch_bounds__(GRP_BLOCK, 6) k_rank_grp(
                     }
                 }
                 __syncthreads();
-                if (threadIdx.x == 0) {
-                    int nn = (nl + 1) / 2;
-                    for (int x = 0; x <= nn; x++)
-                        s_loff[x] = (2 * x <= nl) ? s_loff[2 * x] : gsize;
-                    s_nl = nn;
-                }
                 cur ^= 1;
-                __syncthreads();
+                nl = nn;
             }
             const uint64_t *ft = cur ? s_tb : s_ta;
             const uint16_t *fo = cur ? s_ob : s_oa;
