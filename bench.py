@@ -347,6 +347,11 @@ def main():
         # configs[3] is a single-GPU config; keep the per-rank side tables
         # small on scaling runs so N=8 setup stays minutes, not tens
         hk_pp = min(hk_pp, 100_000)
+    # configs[3]'s scale is 100M rows TOTAL (16 parts x 625K hashkeys x 10);
+    # scale hashkeys down when --partitions grows so oversized compaction
+    # configs (e.g. 400M keys / 64 parts) don't also inflate the side table
+    # past the HBM capacity envelope
+    hk_pp = min(hk_pp, max(1, 100_000_000 // (10 * args.partitions)))
     e_parts = []
     for p_ in range(args.partitions):
         ep = hip.open(2, p_, local_rank)
