@@ -122,6 +122,11 @@ def main():
             else:
                 d = os.path.join(ckdir, f"ck{step}")
                 assert o.checkpoint(d, step) == 0, step
+                # checkpoint the engine too: checkpoint flushes the memtable,
+                # and flush-point-dependent compact stats (input_records /
+                # shadowed) only compare when both sides saw the same flush
+                # boundaries (see tools/stress_mixed.py, DESIGN parity notes)
+                assert g.checkpoint(os.path.join(ckdir, f"ckg{step}"), step) == 0, step
                 o2 = RrdbLib(os.path.join(REPO, "oracle", "liboracle.so")).open(1, 0, -1)
                 g2 = glib.open(1, 0, 0)
                 try:
