@@ -585,6 +585,10 @@ def test_mixed_op_soak(oracle_lib, hip_lib, tmp_path):
             else:  # checkpoint round-trip spot check
                 d = str(tmp_path / f"ck{step}")
                 assert o.checkpoint(d, step) == 0
+                # checkpoint flushes the memtable: do it on BOTH handles so
+                # flush-point-dependent compact stats keep comparing equal
+                # (see tools/soak.py / tools/stress_mixed.py)
+                assert g.checkpoint(str(tmp_path / f"ckg{step}"), step) == 0
                 o2 = oracle_lib.open(1, 0, -1)
                 g2 = hip_lib.open(1, 0, 0)
                 try:
