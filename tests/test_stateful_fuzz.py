@@ -94,6 +94,41 @@ class OracleVsModel(RuleBasedStateMachine):
         assert cnt == len(self.model.full_scan(NOW, validate_hash_req=False))
         self.part.set_envs({"rocksdb.max_iteration_count": "1000"})
 
+    @rule(hks=st.lists(HK, min_size=1, max_size=5), rev=st.booleans())
+    def check_multi_get_batch(self, hks, rev):
+        err, groups = self.part.multi_get_batch(hks, NOW, reverse=rev)
+        assert err == 0
+        assert len(groups) == len(hks)
+        for hk, got in zip(hks, groups):
+            assert got == self.model.multi_get(hk, NOW, reverse=rev), hk
+
+    @rule()
+    def checkpoint_roundtrip(self):
+        import tempfile
+
+        from incubator_pegasus_amd.capi import RrdbLib
+
+        self._decree = getattr(self, "_decree", 0) + 1
+        d = tempfile.mkdtemp(prefix="fuzz_ck_")
+        assert self.part.checkpoint(d, self._decree) == 0
+        r = RrdbLib(_ensure_oracle()).open(1, 0, -1)
+        try:
+            assert r.restore(d, self._decree) == 0
+            rows = []
+            res = r.scan_open(b"\x00\x00", b"\xff\xff", NOW,
+                              validate_partition_hash=False, batch_size=1000)
+            from incubator_pegasus_amd.capi import SCAN_COMPLETED
+
+            assert res.error == 0
+            rows.extend(res.kvs)
+            while res.context_id != SCAN_COMPLETED:
+                res = r.scan_next(res.context_id, NOW)
+                assert res.error == 0
+                rows.extend(res.kvs)
+            assert rows == self.model.full_scan(NOW, validate_hash_req=False)
+        finally:
+            r.close()
+
     @rule()
     def compact(self):
         err, st_ = self.part.manual_compact(NOW)
