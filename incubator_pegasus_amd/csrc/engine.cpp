@@ -1845,8 +1845,17 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
     if (w == 0) {
         out->error = RRDB_OK;
         out->context_id = RRDB_SCAN_CONTEXT_ID_COMPLETED;
-        if (c->only_return_count)
+        if (c->only_return_count) {
             out->i64 = 0;
+        } else {
+            /* a normal zero-row batch returns allocated (empty) arrays —
+             * only the empty-RANGE early return leaves them null (matches
+             * the oracle's scan_batch, observable via expire_ts) */
+            out->keys = (rrdb_slice *)a->alloc(sizeof(rrdb_slice));
+            out->values = (rrdb_slice *)a->alloc(sizeof(rrdb_slice));
+            if (c->return_expire_ts)
+                out->expire_ts = (int32_t *)a->alloc(4);
+        }
         return;
     }
     DevRun *dr = e->dev_runs();
@@ -1891,6 +1900,10 @@ static void scan_batch_gpu(HipEngine *e, HipScanCtx *c, uint32_t epoch_now, rrdb
         out->count = 0;
     } else if (n_out == 0) {
         out->count = 0;
+        out->keys = (rrdb_slice *)a->alloc(sizeof(rrdb_slice));
+        out->values = (rrdb_slice *)a->alloc(sizeof(rrdb_slice));
+        if (c->return_expire_ts)
+            out->expire_ts = (int32_t *)a->alloc(4);
     } else {
         launch_cut_sizes(d_state, w, consumed, d_ksz, d_vsz, e->stream);
         uint64_t *d_koffs = e->talloc<uint64_t>(w * 8);

@@ -263,6 +263,90 @@ class Model:
             out.append((k, v[hdr_len(self.data_version):]))
         return out
 
+    def scan(self, now: int, *, start_key=b"", stop_key=b"", start_inclusive=True,
+             stop_inclusive=False, batch_size=-1, max_iteration_count=1000,
+             no_value=False, hash_key_filter_type=0, hash_key_filter_pattern=b"",
+             sort_key_filter_type=0, sort_key_filter_pattern=b"",
+             validate_hash_req=True, return_expire_ts=False,
+             only_return_count=False):
+        """on_get_scanner + on_scan paging (pegasus_server_impl.cpp:1151-1397):
+        returns (error, batches); each batch is (kvs, expire_ts_or_None,
+        count_or_None).  Mirrors the handler's filter-type validation, the
+        hashkey MATCH_PREFIX start clamp (:1206-1224), the empty-range check
+        (:1227-1243), the first-exclusive skip (:1277-1283), and per-batch
+        paging capped by min(batch_size, rocksdb.max_iteration_count) where
+        expiry / hash-mismatch / filter rejects consume iterations but emit
+        nothing (validate_for_scan equivalent)."""
+        if not (0 <= hash_key_filter_type <= 3 and 0 <= sort_key_filter_type <= 3) \
+                or (hash_key_filter_type == 2 and len(hash_key_filter_pattern) >= 0xFFFF):
+            return 4, []  # kInvalidArgument
+        start, s_incl = start_key, start_inclusive
+        if hash_key_filter_type == 2 and len(hash_key_filter_pattern) > 0:
+            pstart = struct.pack(">H", len(hash_key_filter_pattern)) + hash_key_filter_pattern
+            if pstart > start:
+                start, s_incl = pstart, True
+        if start > stop_key or (start == stop_key and not (s_incl and stop_inclusive)):
+            # early return (:1227-1243): no arrays allocated — expire_ts is
+            # null even when requested, unlike a normal zero-row batch
+            return 0, [([], None, 0 if only_return_count else None)]
+        view = self.visible_items(start, stop_key, True, stop_inclusive)
+        cursor = 0
+        if not s_incl and view and view[0][0] == start:
+            cursor = 1
+
+        def match(ft, pat, v):
+            if ft == 0 or len(pat) == 0:
+                return True
+            if len(v) < len(pat):
+                return False
+            if ft == 1:
+                return pat in v
+            if ft == 2:
+                return v.startswith(pat)
+            return v.endswith(pat)
+
+        hdr = hdr_len(self.data_version)
+        batch_count = max_iteration_count
+        if batch_size > 0:
+            batch_count = min(batch_size, batch_count)
+        batches = []
+        while True:
+            kvs, ets, count, iteration = [], [], 0, 0
+            while cursor < len(view):
+                if count >= batch_count or iteration >= max_iteration_count:
+                    break
+                k, v = view[cursor]
+                cursor += 1
+                iteration += 1
+                if expired(now, expire_of(v, self.data_version)):
+                    continue
+                if validate_hash_req and self.validate_hash:
+                    if self.partition_version < 0 or self.pidx > self.partition_version:
+                        continue
+                    hklen = struct.unpack(">H", k[:2])[0]
+                    h = crc64(k[2:2 + hklen] if hklen else k[2:])
+                    if (h & self.partition_version) != self.pidx:
+                        continue
+                if hash_key_filter_type or sort_key_filter_type:
+                    hklen = struct.unpack(">H", k[:2])[0]
+                    if not match(hash_key_filter_type, hash_key_filter_pattern,
+                                 k[2:2 + hklen]):
+                        continue
+                    if not match(sort_key_filter_type, sort_key_filter_pattern,
+                                 k[2 + hklen:]):
+                        continue
+                count += 1
+                if not only_return_count:
+                    e = expire_of(v, self.data_version)
+                    ets.append(e - (1 << 32) if e >= (1 << 31) else e)
+                    kvs.append((k, b"" if no_value else v[hdr:]))
+            batches.append((kvs,
+                            ets if (return_expire_ts and not only_return_count)
+                            else None,
+                            count if only_return_count else None))
+            if cursor >= len(view):
+                return 0, batches
+
     def compact(self, now: int):
         """Manual compaction result: surviving {key: value} after newest-wins,
         tombstone drop, TTL filter, default-ttl rewrite and user ops."""
