@@ -82,6 +82,30 @@ class OracleVsModel(RuleBasedStateMachine):
                                     reverse=rev, max_kv_count=cap)
         assert got == want, (hk, start, stop, si, pi, rev, cap)
 
+    @rule(si=st.booleans(), pi=st.booleans(), bs=st.sampled_from([-1, 1, 3, 1000]),
+          sft=st.sampled_from([0, 1, 2, 3]), pat=st.sampled_from([b"", b"s", b"1"]),
+          ets=st.booleans())
+    def check_paged_scan(self, si, pi, bs, sft, pat, ets):
+        kw = dict(start_key=b"\x00\x00", stop_key=b"\xff\xff",
+                  start_inclusive=si, stop_inclusive=pi, batch_size=bs,
+                  sort_key_filter_type=sft, sort_key_filter_pattern=pat,
+                  return_expire_ts=ets)
+        from incubator_pegasus_amd.capi import OK, SCAN_COMPLETED
+
+        res = self.part.scan_open(kw["start_key"], kw["stop_key"], NOW,
+                                  validate_partition_hash=False,
+                                  **{k: v for k, v in kw.items()
+                                     if k not in ("start_key", "stop_key")})
+        assert res.error == OK
+        got = [(res.kvs, res.expire_ts, res.kv_count)]
+        while res.context_id != SCAN_COMPLETED:
+            res = self.part.scan_next(res.context_id, NOW)
+            assert res.error == OK
+            got.append((res.kvs, res.expire_ts, res.kv_count))
+        err, want = self.model.scan(NOW, max_iteration_count=1000,
+                                    validate_hash_req=False, **kw)
+        assert (OK, got) == (err, want), kw
+
     @rule()
     def check_scan_count(self):
         # the pipelined count API must agree with the model's full scan
