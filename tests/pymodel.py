@@ -105,6 +105,19 @@ class Model:
             return 1, None
         return 0, (e - now) if e > 0 else -1
 
+    def batch_get(self, raw_keys, now: int):
+        """on_batch_get (pegasus_server_impl.cpp:906-1016): multi-point
+        lookup in request order; NotFound/expired keys are skipped, an
+        empty request is kInvalidArgument."""
+        if not raw_keys:
+            return 4, []
+        out = []
+        for k in raw_keys:
+            st, v = self.get(k, now)
+            if st == 0:
+                out.append((k, v))
+        return 0, out
+
     def sortkey_count(self, hash_key: bytes, now: int):
         lo = struct.pack(">H", len(hash_key)) + hash_key
         hi = bytearray(lo)

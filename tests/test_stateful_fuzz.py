@@ -126,6 +126,15 @@ class OracleVsModel(RuleBasedStateMachine):
         for hk, got in zip(hks, groups):
             assert got == self.model.multi_get(hk, NOW, reverse=rev), hk
 
+    @rule(pairs=st.lists(st.tuples(HK, SK), min_size=0, max_size=6))
+    def check_batch_get(self, pairs):
+        keys = [D.generate_key(hk, sk) for hk, sk in pairs]
+        if not keys:
+            st_, kvs = self.part.batch_get(keys, NOW)
+            assert (st_, kvs) == (4, [])  # kInvalidArgument on empty request
+            return
+        assert self.part.batch_get(keys, NOW) == self.model.batch_get(keys, NOW)
+
     @rule()
     def checkpoint_roundtrip(self):
         import tempfile
