@@ -363,28 +363,50 @@ class Model:
     def compact(self, now: int):
         """Manual compaction result: surviving {key: value} after newest-wins,
         tombstone drop, TTL filter, default-ttl rewrite and user ops."""
-        ver = self.data_version
+        surviving, _ = self.compact_full(now)
+        return surviving
+
+    def compact_full(self, now: int):
+        """compact() plus the full CompactStats accounting the C-ABI reports:
+        input_records (all physical records), shadowed (non-newest versions),
+        tombstones (newest-version DELETEs dropped at the bottommost level),
+        filtered (user-rule deletes + stale-split drops), expired (TTL drops
+        on the post-default-ttl / pre-user-op expire), output_bytes (kept
+        key+value bytes)."""
         surviving = {}
+        stats = dict(input_records=0, output_records=0, expired=0, filtered=0,
+                     tombstones=0, shadowed=0, output_bytes=0)
         newest = {}
+        counts = {}
         for run in self.runs:
+            stats["input_records"] += len(run)
             for k, (v, s, kd) in run.items():
+                counts[k] = counts.get(k, 0) + 1
                 if k not in newest or s > newest[k][1]:
                     newest[k] = (v, s, kd)
+        stats["shadowed"] = sum(c - 1 for c in counts.values())
         for k in sorted(newest):
             v, s, kd = newest[k]
             if kd == 1:
+                stats["tombstones"] += 1
                 continue
-            drop, newv = self._filter(k, v, now)
-            if not drop:
+            reason, newv = self._filter(k, v, now)
+            if reason is None:
                 surviving[k] = (newv if newv is not None else v, s)
+                stats["output_records"] += 1
+                stats["output_bytes"] += len(k) + len(v)
+            else:
+                stats[reason] += 1
         self.runs = [{k: (v, s, 0) for k, (v, s) in surviving.items()}]
-        return surviving
+        return surviving, stats
 
     def _filter(self, key: bytes, value: bytes, now: int):
-        """KeyWithTTLCompactionFilter::Filter (:55-92). Returns (drop, new_value)."""
+        """KeyWithTTLCompactionFilter::Filter (:55-92).  Returns
+        (drop_reason, new_value): reason None = keep, "filtered" = user-rule
+        delete or stale-split drop, "expired" = TTL drop."""
         ver = self.data_version
         if len(key) < 2:
-            return False, None
+            return None, None
         expire_ts = expire_of(value, ver)
         new_value = None
         if self.default_ttl != 0 and expire_ts == 0:
@@ -399,7 +421,7 @@ class Model:
                 if not self._all_rules_match(op, hk, sk, vv, now):
                     continue
                 if op["type"] == "delete":
-                    return True, None
+                    return "filtered", None
                 # update_ttl
                 cur = expire_of(vv, ver)
                 t = op["ut_type"]
@@ -416,13 +438,13 @@ class Model:
                 off = 1 if ver == 2 else 0
                 new_value = vv[:off] + struct.pack(">I", new_ts) + vv[off + 4:]
         if expired(now, expire_ts):
-            return True, None
+            return "expired", None
         if self.validate_hash and self.partition_version >= 0 and self.pidx <= self.partition_version:
             hklen = struct.unpack(">H", key[:2])[0]
             h = crc64(key[2:2 + hklen] if hklen else key[2:])
             if (h & self.partition_version) != self.pidx:
-                return True, None
-        return False, new_value
+                return "filtered", None
+        return None, new_value
 
     def _all_rules_match(self, op, hk, sk, value, now):
         if not op["rules"]:

@@ -165,8 +165,13 @@ class OracleVsModel(RuleBasedStateMachine):
     @rule()
     def compact(self):
         err, st_ = self.part.manual_compact(NOW)
-        surviving = self.model.compact(NOW)
+        surviving, want = self.model.compact_full(NOW)
         assert err == 0
+        got = dict(input_records=st_.input_records, output_records=st_.output_records,
+                   expired=st_.expired, filtered=st_.filtered,
+                   tombstones=st_.tombstones, shadowed=st_.shadowed,
+                   output_bytes=st_.output_bytes)
+        assert got == want
         assert st_.output_records == len(surviving)
 
     @invariant()
