@@ -411,9 +411,14 @@ bool parse_user_ops(const std::string &text, std::vector<HostOp> &out)
         const Json *jt = jop.get("type");
         const Json *jp = jop.get("params");
         const Json *jr = jop.get("rules");
-        if (!jt || jt->type != Json::STR || !jp || jp->type != Json::STR || !jr ||
+        /* a missing "params" decodes as "" (json_helper.h:136-143
+         * JSON_TRY_DECODE_ENTRY tolerates absent members): delete_key
+         * ignores params, update_ttl's own params decode then fails and
+         * drops the op */
+        if (!jt || jt->type != Json::STR || (jp && jp->type != Json::STR) || !jr ||
             jr->type != Json::ARR)
             continue;
+        std::string op_params = jp ? jp->str : std::string();
         HostOp op;
         op.type = enum_cot(jt->str);
         for (const Json &jrule : jr->arr) {
@@ -451,7 +456,7 @@ bool parse_user_ops(const std::string &text, std::vector<HostOp> &out)
             continue;
         if (op.type == DOP_UPDATE_TTL) {
             Json oparams;
-            if (!json_parse(jp->str, oparams) || oparams.type != Json::OBJ)
+            if (!json_parse(op_params, oparams) || oparams.type != Json::OBJ)
                 continue;
             const Json *t = oparams.get("type");
             const Json *v = oparams.get("value");

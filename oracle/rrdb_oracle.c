@@ -718,9 +718,14 @@ static int parse_user_ops(orc_ops *out, const char *json, uint64_t len)
                             break;
                         }
                     }
-                    /* op valid? rules non-empty + known type + params decode */
-                    int op_ok = p.ok && op.n_rules > 0 && op.type != COT_INVALID && op_params &&
-                                parse_op_params(&op, op_params, op_params_len);
+                    /* op valid? rules non-empty + known type + params decode.
+                     * A missing "params" decodes as "" (json_helper.h:136-143
+                     * JSON_TRY_DECODE_ENTRY tolerates absent members):
+                     * delete_key ignores params; update_ttl's decode of ""
+                     * fails and drops the op */
+                    int op_ok = p.ok && op.n_rules > 0 && op.type != COT_INVALID &&
+                                parse_op_params(&op, op_params ? op_params : "",
+                                                op_params ? op_params_len : 0);
                     if (op_ok) {
                         if (out->n_ops == cap) {
                             cap *= 2;

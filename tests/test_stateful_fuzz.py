@@ -54,6 +54,36 @@ class OracleVsModel(RuleBasedStateMachine):
         self.part.ingest_run(recs)
         self.model.ingest(recs)
 
+    @rule(pat=st.sampled_from([b"s1", b"s2", b"sort"]),
+          mt=st.sampled_from(["prefix", "postfix", "anywhere"]),
+          action=st.sampled_from(["delete", "update_ttl", "off"]))
+    def set_user_rules(self, pat, mt, action):
+        import json
+
+        if action == "off":
+            self.part.set_envs({"user_specified_compaction": ""})
+            self.model.user_ops = []
+            return
+        smt = {"prefix": "SMT_MATCH_PREFIX", "postfix": "SMT_MATCH_POSTFIX",
+               "anywhere": "SMT_MATCH_ANYWHERE"}[mt]
+        rules = [{"type": "FRT_SORTKEY_PATTERN",
+                  "params": json.dumps({"pattern": pat.decode(),
+                                        "match_type": smt})}]
+        if action == "delete":
+            ops = [{"type": "COT_DELETE", "rules": rules}]
+            self.model.user_ops = [dict(type="delete",
+                                        rules=[dict(type="sortkey", pattern=pat,
+                                                    match_type=mt)])]
+        else:
+            ops = [{"type": "COT_UPDATE_TTL",
+                    "params": json.dumps({"type": "UTOT_FROM_NOW", "value": 500}),
+                    "rules": rules}]
+            self.model.user_ops = [dict(type="update_ttl", ut_type="from_now",
+                                        value=500,
+                                        rules=[dict(type="sortkey", pattern=pat,
+                                                    match_type=mt)])]
+        self.part.set_envs({"user_specified_compaction": json.dumps({"ops": ops})})
+
     @rule(ttl=st.sampled_from(["0", "77", "3600"]))
     def set_default_ttl(self, ttl):
         self.part.set_envs({"default_ttl": ttl})
