@@ -122,3 +122,32 @@ def test_scan_engine_vs_oracle(oracle_lib, hip_lib, seed):
     finally:
         o.close()
         g.close()
+
+
+@pytest.mark.gpu
+def test_paramsless_delete_op_engine_vs_oracle(oracle_lib, hip_lib):
+    """An op object without a "params" key decodes as params="" (the
+    reference's JSON_TRY_DECODE_ENTRY tolerates absent members,
+    json_helper.h:136-143) — delete_key then applies.  Found by the
+    stateful fuzz; pins engine == oracle == 'record deleted'."""
+    import json
+
+    o = oracle_lib.open(1, 0, -1)
+    g = hip_lib.open(1, 0, 0)
+    try:
+        ops = {"ops": [{"type": "COT_DELETE", "rules": [
+            {"type": "FRT_SORTKEY_PATTERN",
+             "params": json.dumps({"pattern": "s1",
+                                   "match_type": "SMT_MATCH_PREFIX"})}]}]}
+        k = D.generate_key(b"a", b"s1")
+        v = D.encode_value(b"", 0, 1, 1)
+        for p_ in (o, g):
+            p_.set_envs({"user_specified_compaction": json.dumps(ops)})
+            p_.ingest_run([(k, v, 1, 0)])
+        so = o.manual_compact(NOW)
+        sg = g.manual_compact(NOW)
+        assert so == sg
+        assert so[1].filtered == 1 and so[1].output_records == 0
+    finally:
+        o.close()
+        g.close()
