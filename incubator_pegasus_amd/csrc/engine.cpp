@@ -460,6 +460,10 @@ bool parse_user_ops(const std::string &text, std::vector<HostOp> &out)
                 continue;
             const Json *t = oparams.get("type");
             const Json *v = oparams.get("value");
+            /* both fields required: the reference's tolerant decode would
+             * accept one missing and read the uninitialized member
+             * (compaction_operation.cpp:75 initializes neither) — UB with
+             * no semantics to restate, so the op is dropped instead */
             if (!t || t->type != Json::STR || !v || v->type != Json::NUM)
                 continue;
             op.ut_type = enum_utot(t->str);

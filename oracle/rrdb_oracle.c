@@ -535,6 +535,12 @@ static int parse_rule_params(orc_rule *r, const char *json, uint64_t len)
     return have_pattern && have_mt;
 }
 
+/* update_ttl params decode.  Note: the reference's tolerant decode would
+ * ACCEPT {"type":...} without "value" (or vice versa) and then read the
+ * UNINITIALIZED member (update_ttl's ctor, compaction_operation.cpp:75,
+ * initializes neither field) — undefined behavior with no semantics to
+ * restate; we deterministically require both fields and drop the op
+ * otherwise. */
 static int parse_op_params(orc_op *op, const char *json, uint64_t len)
 {
     if (op->type == COT_DELETE)
