@@ -145,7 +145,7 @@ class Model:
                   stop_sortkey=b"", start_inclusive=True, stop_inclusive=False,
                   max_kv_count=-1, max_kv_size=-1, reverse=False,
                   sort_key_filter_type=0, sort_key_filter_pattern=b"",
-                  engine_max_iter=3000):
+                  no_value=False, engine_max_iter=3000):
         """Range variant of on_multi_get (pegasus_server_impl.cpp:540-799),
         restated independently: the forward/reverse iterator walk with the
         range_read_limiter count cap and the iterator-validity completion
@@ -247,7 +247,10 @@ class Model:
             iteration += 1
             sk = k[len(pre):]
             if not expired(now, expire_of(v, self.data_version)) and sk_filter_ok(sk):
-                uv = v[hdr_len(self.data_version):]
+                # no_value leaves kv.value empty, so the size budget counts
+                # only the sortkey bytes (append_key_value_for_multi_get
+                # :2498-2502 + the kv_size accounting :726-728)
+                uv = b"" if no_value else v[hdr_len(self.data_version):]
                 out.append((sk, uv))
                 count += 1
                 size += len(sk) + len(uv)

@@ -49,6 +49,7 @@ def _cases(rnd, n):
             start_inclusive=rnd.random() < 0.5, stop_inclusive=rnd.random() < 0.5,
             reverse=rnd.random() < 0.5, max_kv_count=rnd.choice([-1, 1, 2, 3, 100]),
             max_kv_size=rnd.choice([-1, -1, 1, 7, 40]),
+            no_value=rnd.random() < 0.25,
             sort_key_filter_type=rnd.choice([0, 0, 1, 2, 3]),
             sort_key_filter_pattern=rnd.choice([b"", b"s", b"1", b"zz"]))
 

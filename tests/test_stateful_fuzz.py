@@ -102,15 +102,15 @@ class OracleVsModel(RuleBasedStateMachine):
         assert cnt == self.model.sortkey_count(hk, NOW)
 
     @rule(hk=HK, start=SK, stop=SK, si=st.booleans(), pi=st.booleans(),
-          rev=st.booleans(), cap=st.sampled_from([-1, 1, 2, 1000]))
-    def check_multi_get(self, hk, start, stop, si, pi, rev, cap):
-        got = self.part.multi_get(hk, NOW, start_sortkey=start, stop_sortkey=stop,
-                                  start_inclusive=si, stop_inclusive=pi,
-                                  reverse=rev, max_kv_count=cap)
-        want = self.model.multi_get(hk, NOW, start_sortkey=start, stop_sortkey=stop,
-                                    start_inclusive=si, stop_inclusive=pi,
-                                    reverse=rev, max_kv_count=cap)
-        assert got == want, (hk, start, stop, si, pi, rev, cap)
+          rev=st.booleans(), cap=st.sampled_from([-1, 1, 2, 1000]),
+          nv=st.booleans(), msz=st.sampled_from([-1, -1, 3, 20]))
+    def check_multi_get(self, hk, start, stop, si, pi, rev, cap, nv, msz):
+        kw = dict(start_sortkey=start, stop_sortkey=stop, start_inclusive=si,
+                  stop_inclusive=pi, reverse=rev, max_kv_count=cap,
+                  no_value=nv, max_kv_size=msz)
+        got = self.part.multi_get(hk, NOW, **kw)
+        want = self.model.multi_get(hk, NOW, **kw)
+        assert got == want, (hk, kw)
 
     @rule(si=st.booleans(), pi=st.booleans(), bs=st.sampled_from([-1, 1, 3, 1000]),
           sft=st.sampled_from([0, 1, 2, 3]), pat=st.sampled_from([b"", b"s", b"1"]),
