@@ -151,3 +151,79 @@ def test_paramsless_delete_op_engine_vs_oracle(oracle_lib, hip_lib):
     finally:
         o.close()
         g.close()
+
+
+@pytest.mark.parametrize("pidx", [0, 2])
+def test_scan_hash_validation_oracle_vs_model(oracle_lib, pidx):
+    """Stale-split read filtering (validate_partition_hash): rejected rows
+    consume iterations but emit nothing — pinned against the model's crc64
+    restatement across batch boundaries."""
+    p = oracle_lib.open(1, pidx, -1)
+    try:
+        model = Model(pidx=pidx)
+        model.validate_hash = True
+        model.partition_version = 3
+        p.set_envs({"replica.split.validate_partition_hash": "true"})
+        p.set_partition_version(3)
+        recs = [(D.generate_key(b"h%03d" % i, b""),
+                 D.encode_value(b"v", 0, i + 1, 1), i + 1, 0) for i in range(64)]
+        p.ingest_run(recs)
+        model.ingest(recs)
+        for bs in (-1, 3, 7):
+            for mi in (5, 1000):
+                kw = dict(start_key=b"\x00\x00", stop_key=b"\xff\xff",
+                          batch_size=bs, return_expire_ts=True)
+                got = _drain_batches_hv(p, kw, mi)
+                want = model.scan(NOW, max_iteration_count=mi,
+                                  validate_hash_req=True, **kw)
+                assert got == want, (bs, mi)
+                assert sum(len(b[0]) for b in want[1]) > 0
+    finally:
+        p.close()
+
+
+def _drain_batches_hv(part, kw, max_iter):
+    """Like _drain_batches but with validate_partition_hash left ON."""
+    part.set_envs({"rocksdb.max_iteration_count": str(max_iter)})
+    res = part.scan_open(kw["start_key"], kw["stop_key"], NOW,
+                         **{k: v for k, v in kw.items()
+                            if k not in ("start_key", "stop_key")})
+    if res.error != OK:
+        return res.error, []
+    batches = [(res.kvs, res.expire_ts, res.kv_count)]
+    while res.context_id != SCAN_COMPLETED:
+        res = part.scan_next(res.context_id, NOW)
+        assert res.error == OK
+        batches.append((res.kvs, res.expire_ts, res.kv_count))
+    return OK, batches
+
+
+def test_paramsless_delete_op_oracle_vs_model(oracle_lib):
+    """CPU twin of the GPU pin: a params-less COT_DELETE op applies
+    (json_helper.h:136-143 tolerant decode)."""
+    import json
+
+    o = oracle_lib.open(1, 0, -1)
+    try:
+        model = Model()
+        ops = {"ops": [{"type": "COT_DELETE", "rules": [
+            {"type": "FRT_SORTKEY_PATTERN",
+             "params": json.dumps({"pattern": "s1",
+                                   "match_type": "SMT_MATCH_PREFIX"})}]}]}
+        o.set_envs({"user_specified_compaction": json.dumps(ops)})
+        model.user_ops = [dict(type="delete",
+                               rules=[dict(type="sortkey", pattern=b"s1",
+                                           match_type="prefix")])]
+        k = D.generate_key(b"a", b"s1")
+        v = D.encode_value(b"", 0, 1, 1)
+        o.ingest_run([(k, v, 1, 0)])
+        model.ingest([(k, v, 1, 0)])
+        err, st = o.manual_compact(NOW)
+        _, want = model.compact_full(NOW)
+        assert err == OK
+        assert dict(input_records=st.input_records, output_records=st.output_records,
+                    expired=st.expired, filtered=st.filtered, tombstones=st.tombstones,
+                    shadowed=st.shadowed, output_bytes=st.output_bytes) == want
+        assert st.filtered == 1 and st.output_records == 0
+    finally:
+        o.close()
