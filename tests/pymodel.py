@@ -118,6 +118,34 @@ class Model:
                 out.append((k, v))
         return 0, out
 
+    def multi_get_sortkeys(self, hash_key: bytes, sort_keys, now: int, *,
+                           max_kv_count=-1, max_kv_size=-1, no_value=False,
+                           engine_max_iter=3000):
+        """Point-list variant of on_multi_get (pegasus_server_impl.cpp:779-860):
+        per requested sortkey in order, NotFound/expired skip; the cap check
+        runs BEFORE the append (:841-845), so the NEXT surviving record after
+        the caps fill returns kIncomplete; kv.key echoes the REQUESTED
+        sortkey."""
+        max_count = engine_max_iter if max_kv_count <= 0 \
+            else min(max_kv_count, engine_max_iter)
+        max_size = max_kv_size if max_kv_size > 0 else (1 << 31) - 1
+        out = []
+        count = size = 0
+        exceed = False
+        for sk in sort_keys:
+            key = struct.pack(">H", len(hash_key)) + hash_key + sk
+            st, body = self.get(key, now)
+            if st != 0:
+                continue
+            if count >= max_count or size >= max_size:
+                exceed = True
+                break
+            val = b"" if no_value else body
+            out.append((sk, val))
+            count += 1
+            size += len(sk) + len(val)
+        return (7 if exceed else 0), out
+
     def sortkey_count(self, hash_key: bytes, now: int):
         lo = struct.pack(">H", len(hash_key)) + hash_key
         hi = bytearray(lo)

@@ -83,3 +83,24 @@ def test_multi_get_engine_vs_oracle(oracle_lib, hip_lib, seed):
     finally:
         o.close()
         g.close()
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_multi_get_sortkeys_oracle_vs_model(oracle_lib, seed):
+    """Point-list variant: requested-order echo, miss/expired skips, and the
+    check-before-append cap semantics (:841-845)."""
+    rnd = random.Random(3000 + seed)
+    p = oracle_lib.open(1, 0, -1)
+    try:
+        model = _build(rnd, [p])
+        for _ in range(60):
+            hk = rnd.choice(HKS + [b"miss"])
+            sks = [rnd.choice(SKS + [b"nope"]) for _ in range(rnd.randrange(1, 8))]
+            kw = dict(max_kv_count=rnd.choice([-1, 1, 2, 100]),
+                      max_kv_size=rnd.choice([-1, -1, 1, 5, 30]),
+                      no_value=rnd.random() < 0.3)
+            got = p.multi_get(hk, NOW, sort_keys=sks, **kw)
+            want = model.multi_get_sortkeys(hk, sks, NOW, **kw)
+            assert got == want, (hk, sks, kw)
+    finally:
+        p.close()
