@@ -26,19 +26,28 @@ TTL = st.sampled_from([0, 1, NOW, NOW + 50])
 
 
 class OracleVsModel(RuleBasedStateMachine):
+    DATA_VERSION = 1
+
     def __init__(self):
         super().__init__()
         from incubator_pegasus_amd.capi import RrdbLib
 
         self.part = RrdbLib(_ensure_oracle()).open(1, 0, -1)
-        self.model = pymodel.Model()
+        if self.DATA_VERSION != 1:
+            self.part.set_envs({"pegasus.data_version": str(self.DATA_VERSION)})
+        self.model = pymodel.Model(data_version=self.DATA_VERSION)
         self.seq = 1
         self.pending = {}
 
     @rule(hk=HK, sk=SK, body=BODY, ttl=TTL, kind=st.sampled_from([0, 0, 0, 1]))
     def stage_record(self, hk, sk, body, ttl, kind):
         key = D.generate_key(hk, sk)
-        val = D.encode_value(body, ttl, self.seq, 1) if kind == 0 else b"\x00" * 12
+        if kind == 0:
+            val = D.encode_value(body, ttl, self.seq, self.DATA_VERSION)
+        else:
+            val = b"\x00" * (13 if self.DATA_VERSION == 2 else 12)
+            if self.DATA_VERSION == 2:
+                val = b"\x82" + val[1:]
         self.pending[key] = (val, kind)
 
     @rule()
@@ -223,6 +232,15 @@ class OracleVsModel(RuleBasedStateMachine):
         self.part.close()
 
 
+class OracleVsModelV2(OracleVsModel):
+    """Same machine over value schema v2 ([u8 meta][u32 expire][data],
+    value_schema_v2.cpp) — the codec dimension the v1 machine cannot see."""
+    DATA_VERSION = 2
+
+
 OracleVsModel = settings(max_examples=40, stateful_step_count=30,
                          deadline=None)(OracleVsModel)
 TestOracleVsModel = OracleVsModel.TestCase
+TestOracleVsModelV2 = OracleVsModelV2.TestCase
+TestOracleVsModelV2.settings = settings(max_examples=25, stateful_step_count=30,
+                                        deadline=None)
