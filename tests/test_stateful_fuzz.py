@@ -45,7 +45,7 @@ class OracleVsModel(RuleBasedStateMachine):
         if kind == 0:
             val = D.encode_value(body, ttl, self.seq, self.DATA_VERSION)
         else:
-            val = b"\x00" * (13 if self.DATA_VERSION == 2 else 12)
+            val = b"\x00" * {0: 4, 1: 12, 2: 13}[self.DATA_VERSION]
             if self.DATA_VERSION == 2:
                 val = b"\x82" + val[1:]
         self.pending[key] = (val, kind)
@@ -233,9 +233,15 @@ class OracleVsModel(RuleBasedStateMachine):
 
 
 class OracleVsModelV2(OracleVsModel):
-    """Same machine over value schema v2 ([u8 meta][u32 expire][data],
-    value_schema_v2.cpp) — the codec dimension the v1 machine cannot see."""
+    """Same machine over value schema v2 ([u8 meta][u32 expire][u64 timetag]
+    [data], value_schema_v2.cpp) — the codec dimension the v1 machine
+    cannot see."""
     DATA_VERSION = 2
+
+
+class OracleVsModelV0(OracleVsModel):
+    """And over v0 ([u32 expire][data], value_schema_v0.cpp)."""
+    DATA_VERSION = 0
 
 
 OracleVsModel = settings(max_examples=40, stateful_step_count=30,
@@ -243,4 +249,7 @@ OracleVsModel = settings(max_examples=40, stateful_step_count=30,
 TestOracleVsModel = OracleVsModel.TestCase
 TestOracleVsModelV2 = OracleVsModelV2.TestCase
 TestOracleVsModelV2.settings = settings(max_examples=25, stateful_step_count=30,
+                                        deadline=None)
+TestOracleVsModelV0 = OracleVsModelV0.TestCase
+TestOracleVsModelV0.settings = settings(max_examples=25, stateful_step_count=30,
                                         deadline=None)
