@@ -227,3 +227,41 @@ def test_paramsless_delete_op_oracle_vs_model(oracle_lib):
         assert st.filtered == 1 and st.output_records == 0
     finally:
         o.close()
+
+
+def test_compact_stats_shadowed_tombstones_oracle_vs_model(oracle_lib):
+    """Deterministic full-stats pin with every counter NON-ZERO (a
+    mutation-sensitivity check showed zeroing `shadowed` slipped past the
+    randomized suites): two runs with an overwritten key (shadowed), a
+    newest-version tombstone, an expired record and a live survivor."""
+    o = oracle_lib.open(1, 0, -1)
+    try:
+        model = Model()
+        k_over = D.generate_key(b"a", b"s0")   # overwritten across runs
+        k_tomb = D.generate_key(b"a", b"s1")   # newest version is a delete
+        k_exp = D.generate_key(b"a", b"s2")    # expired at compact time
+        k_live = D.generate_key(b"a", b"s3")
+        r1 = [(k_over, D.encode_value(b"old", 0, 1, 1), 1, 0),
+              (k_tomb, D.encode_value(b"x", 0, 2, 1), 2, 0)]
+        r2 = [(k_over, D.encode_value(b"new", 0, 3, 1), 3, 0),
+              (k_tomb, b"\x00" * 12, 4, 1),
+              (k_exp, D.encode_value(b"e", 5, 5, 1), 5, 0),
+              (k_live, D.encode_value(b"v", 0, 6, 1), 6, 0)]
+        for recs in (r1, r2):
+            o.ingest_run(recs)
+            model.ingest(recs)
+        err, st = o.manual_compact(NOW)
+        _, want = model.compact_full(NOW)
+        assert err == OK
+        got = dict(input_records=st.input_records, output_records=st.output_records,
+                   expired=st.expired, filtered=st.filtered, tombstones=st.tombstones,
+                   shadowed=st.shadowed, output_bytes=st.output_bytes)
+        assert got == want
+        ob = (len(k_over) + len(k_live) +
+              len(D.encode_value(b"new", 0, 3, 1)) +
+              len(D.encode_value(b"v", 0, 6, 1)))
+        assert got == dict(input_records=6, output_records=2, expired=1,
+                           filtered=0, tombstones=1, shadowed=2,
+                           output_bytes=ob)
+    finally:
+        o.close()
