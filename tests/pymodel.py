@@ -54,10 +54,47 @@ class Model:
         self.data_version = data_version
         self.default_ttl = 0
         self.user_ops = []  # parsed op dicts
+        self.next_seq_floor = 0  # write-path seqno source / ingest floor
+        self.mem = []  # memtable log: [(key, value, seq, kind)]
 
     def ingest(self, records):
         """records: [(raw_key, raw_value, seq, kind)]"""
         self.runs.append({k: (v, s, kd) for k, v, s, kd in records})
+        if records:
+            self.next_seq_floor = max(self.next_seq_floor,
+                                      max(s for _, _, s, _ in records) + 1)
+
+    # -- write path (pegasus_write_service put/remove + memtable flush,
+    #    reference pegasus_write_service.h:119-207, rocksdb_wrapper.cpp:
+    #    121-247; mirrors the oracle/engine host memtable) --
+    def put(self, hash_key: bytes, sort_key: bytes, body: bytes, expire_ts=0):
+        key = struct.pack(">H", len(hash_key)) + hash_key + sort_key
+        ver = self.data_version
+        if ver == 0:
+            val = struct.pack(">I", expire_ts & 0xFFFFFFFF) + body
+        elif ver == 1:
+            val = struct.pack(">IQ", expire_ts & 0xFFFFFFFF, 0) + body
+        else:
+            val = b"\x82" + struct.pack(">IQ", expire_ts & 0xFFFFFFFF, 0) + body
+        self.mem.append((key, val, self.next_seq_floor, 0))
+        self.next_seq_floor += 1
+
+    def remove(self, hash_key: bytes, sort_key: bytes):
+        key = struct.pack(">H", len(hash_key)) + hash_key + sort_key
+        self.mem.append((key, b"", self.next_seq_floor, 1))
+        self.next_seq_floor += 1
+
+    def flush(self):
+        """memtable -> sorted run: newest version per key survives
+        (memtable upsert semantics)."""
+        if not self.mem:
+            return
+        newest = {}
+        for k, v, s, kd in self.mem:
+            if k not in newest or s > newest[k][1]:
+                newest[k] = (v, s, kd)
+        self.runs.append(newest)
+        self.mem = []
 
     # -- visibility --
     def newest(self, key: bytes):

@@ -55,13 +55,29 @@ class OracleVsModel(RuleBasedStateMachine):
         if not self.pending:
             return
         recs = []
-        for key in sorted(self.pending):
+        base = self.model.next_seq_floor  # ingest seqs must clear the floor
+        for i, key in enumerate(sorted(self.pending)):
             val, kind = self.pending[key]
-            recs.append((key, val, self.seq, kind))
-            self.seq += 1
+            recs.append((key, val, base + i, kind))
         self.pending.clear()
         self.part.ingest_run(recs)
         self.model.ingest(recs)
+
+    # -- write path through the C-ABI (memtable put/remove/flush) --
+    @rule(hk=HK, sk=SK, body=BODY, ttl=TTL)
+    def api_put(self, hk, sk, body, ttl):
+        assert self.part.put(hk, sk, body, ttl) == 0
+        self.model.put(hk, sk, body, ttl)
+
+    @rule(hk=HK, sk=SK)
+    def api_remove(self, hk, sk):
+        assert self.part.remove(hk, sk) == 0
+        self.model.remove(hk, sk)
+
+    @rule()
+    def api_flush(self):
+        assert self.part.flush() == 0
+        self.model.flush()
 
     @rule(pat=st.sampled_from([b"s1", b"s2", b"sort"]),
           mt=st.sampled_from(["prefix", "postfix", "anywhere"]),
@@ -100,12 +116,14 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @rule(hk=HK, sk=SK)
     def check_get_ttl(self, hk, sk):
+        self.model.flush()  # read entries flush the memtable
         key = D.generate_key(hk, sk)
         assert self.part.get(key, NOW) == self.model.get(key, NOW)
         assert self.part.ttl(key, NOW) == self.model.ttl(key, NOW)
 
     @rule(hk=HK)
     def check_sortkey_count(self, hk):
+        self.model.flush()  # read entries flush the memtable
         st_, cnt = self.part.sortkey_count(hk, NOW)
         assert st_ == 0
         assert cnt == self.model.sortkey_count(hk, NOW)
@@ -114,6 +132,7 @@ class OracleVsModel(RuleBasedStateMachine):
           rev=st.booleans(), cap=st.sampled_from([-1, 1, 2, 1000]),
           nv=st.booleans(), msz=st.sampled_from([-1, -1, 3, 20]))
     def check_multi_get(self, hk, start, stop, si, pi, rev, cap, nv, msz):
+        self.model.flush()  # read entries flush the memtable
         kw = dict(start_sortkey=start, stop_sortkey=stop, start_inclusive=si,
                   stop_inclusive=pi, reverse=rev, max_kv_count=cap,
                   no_value=nv, max_kv_size=msz)
@@ -125,6 +144,7 @@ class OracleVsModel(RuleBasedStateMachine):
           sft=st.sampled_from([0, 1, 2, 3]), pat=st.sampled_from([b"", b"s", b"1"]),
           ets=st.booleans())
     def check_paged_scan(self, si, pi, bs, sft, pat, ets):
+        self.model.flush()  # read entries flush the memtable
         kw = dict(start_key=b"\x00\x00", stop_key=b"\xff\xff",
                   start_inclusive=si, stop_inclusive=pi, batch_size=bs,
                   sort_key_filter_type=sft, sort_key_filter_pattern=pat,
@@ -147,6 +167,7 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @rule()
     def check_scan_count(self):
+        self.model.flush()  # read entries flush the memtable
         # the pipelined count API must agree with the model's full scan
         self.part.set_envs({"rocksdb.max_iteration_count": str(2**31 - 1)})
         rc = self.part.scan_count_begin(b"\x00\x00", b"\xff\xff", NOW,
@@ -159,6 +180,7 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @rule(hks=st.lists(HK, min_size=1, max_size=5), rev=st.booleans())
     def check_multi_get_batch(self, hks, rev):
+        self.model.flush()  # read entries flush the memtable
         err, groups = self.part.multi_get_batch(hks, NOW, reverse=rev)
         assert err == 0
         assert len(groups) == len(hks)
@@ -167,6 +189,7 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @rule(pairs=st.lists(st.tuples(HK, SK), min_size=0, max_size=6))
     def check_batch_get(self, pairs):
+        self.model.flush()  # read entries flush the memtable
         keys = [D.generate_key(hk, sk) for hk, sk in pairs]
         if not keys:
             st_, kvs = self.part.batch_get(keys, NOW)
@@ -176,6 +199,7 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @rule()
     def checkpoint_roundtrip(self):
+        self.model.flush()  # read entries flush the memtable
         import tempfile
 
         from incubator_pegasus_amd.capi import RrdbLib
@@ -203,6 +227,7 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @rule()
     def compact(self):
+        self.model.flush()  # read entries flush the memtable
         err, st_ = self.part.manual_compact(NOW)
         surviving, want = self.model.compact_full(NOW)
         assert err == 0
@@ -215,6 +240,7 @@ class OracleVsModel(RuleBasedStateMachine):
 
     @invariant()
     def full_scan_matches(self):
+        self.model.flush()  # read entries flush the memtable
         rows = []
         res = self.part.scan_open(b"\x00\x00", b"\xff\xff", NOW,
                                   validate_partition_hash=False, batch_size=1000)
