@@ -109,6 +109,48 @@ class OracleVsModel(RuleBasedStateMachine):
                                                     match_type=mt)])]
         self.part.set_envs({"user_specified_compaction": json.dumps({"ops": ops})})
 
+    @rule(pv=st.sampled_from([-1, 0, 1, 3]))
+    def set_partition_version(self, pv):
+        self.part.set_envs({"replica.split.validate_partition_hash":
+                            "true" if pv >= 0 else "false"})
+        if pv >= 0:
+            self.part.set_partition_version(pv)
+        self.model.validate_hash = pv >= 0
+        self.model.partition_version = pv if pv >= 0 else -1
+
+    @rule(bs=st.sampled_from([-1, 2, 1000]))
+    def check_hash_validated_scan(self, bs):
+        self.model.flush()
+        from incubator_pegasus_amd.capi import OK, SCAN_COMPLETED
+
+        res = self.part.scan_open(b"\x00\x00", b"\xff\xff", NOW, batch_size=bs)
+        assert res.error == OK
+        got = [(res.kvs, res.expire_ts, res.kv_count)]
+        while res.context_id != SCAN_COMPLETED:
+            res = self.part.scan_next(res.context_id, NOW)
+            assert res.error == OK
+            got.append((res.kvs, res.expire_ts, res.kv_count))
+        err, want = self.model.scan(NOW, start_key=b"\x00\x00",
+                                    stop_key=b"\xff\xff", batch_size=bs,
+                                    max_iteration_count=1000,
+                                    validate_hash_req=True)
+        assert (OK, got) == (err, want), bs
+
+    @rule()
+    def ingest_rejections(self):
+        # below-floor seqs and unsorted keys are kInvalidArgument and leave
+        # the store untouched (mirrors test_ingest_validation, in-stream)
+        self.model.flush()
+        floor = self.model.next_seq_floor
+        k1 = D.generate_key(b"a", b"s1")
+        k2 = D.generate_key(b"bb", b"s2")
+        v = D.encode_value(b"x", 0, 0, self.DATA_VERSION)
+        if floor > 0:
+            with pytest.raises(RuntimeError, match="status 4"):
+                self.part.ingest_run([(k1, v, floor - 1, 0)])
+        with pytest.raises(RuntimeError, match="status 4"):
+            self.part.ingest_run([(k2, v, floor + 1, 0), (k1, v, floor + 2, 0)])
+
     @rule(ttl=st.sampled_from(["0", "77", "3600"]))
     def set_default_ttl(self, ttl):
         self.part.set_envs({"default_ttl": ttl})
