@@ -214,9 +214,14 @@ int32_t rrdb_ingest_run(void *h, const uint8_t *keys, const uint64_t *key_offs,
  * newest write to a key wins (memtable upsert).  rrdb_flush turns the
  * memtable into a sorted run (the memtable-flush -> L0 SST step); reads
  * flush lazily so committed writes are immediately visible, as they are
- * through rocksdb's memtable read path. */
+ * through rocksdb's memtable read path.
+ * epoch_now is the write's wall clock: when expire_ts == 0 and the table
+ * carries a default_ttl env, the stored expire becomes
+ * epoch_now + default_ttl AT WRITE TIME (rocksdb_wrapper::db_expire_ts,
+ * rocksdb_wrapper.cpp:280-286) — not deferred to the compaction filter. */
 int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
-                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts);
+                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts,
+                 uint32_t epoch_now);
 int32_t rrdb_remove(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
                     uint64_t sklen);
 uint64_t rrdb_memtable_entries(void *h);

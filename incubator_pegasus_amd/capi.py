@@ -206,7 +206,7 @@ class RrdbLib:
         L.rrdb_phase_ms.argtypes = [C.c_void_p, C.c_char_p]
         L.rrdb_put.restype = C.c_int32
         L.rrdb_put.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64, C.c_char_p, C.c_uint64,
-                               C.c_char_p, C.c_uint64, C.c_uint32]
+                               C.c_char_p, C.c_uint64, C.c_uint32, C.c_uint32]
         L.rrdb_remove.restype = C.c_int32
         L.rrdb_remove.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64, C.c_char_p, C.c_uint64]
         L.rrdb_flush.restype = C.c_int32
@@ -536,9 +536,13 @@ class RrdbPartition:
         return self._L.rrdb_phase_ms(self._h, phase.encode())
 
     # ---- write path (§8(f)1) ----
-    def put(self, hash_key: bytes, sort_key: bytes, value: bytes, expire_ts: int = 0):
+    def put(self, hash_key: bytes, sort_key: bytes, value: bytes, expire_ts: int = 0,
+            epoch_now: int = 0):
+        """epoch_now = the write's clock; with a default_ttl table env and
+        expire_ts 0, the stored expire becomes epoch_now + default_ttl at
+        WRITE time (rocksdb_wrapper.cpp:280-286)."""
         return self._L.rrdb_put(self._h, hash_key, len(hash_key), sort_key, len(sort_key),
-                                value, len(value), expire_ts)
+                                value, len(value), expire_ts, epoch_now)
 
     def remove(self, hash_key: bytes, sort_key: bytes):
         return self._L.rrdb_remove(self._h, hash_key, len(hash_key), sort_key, len(sort_key))

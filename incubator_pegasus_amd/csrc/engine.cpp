@@ -3297,12 +3297,16 @@ int32_t rrdb_manual_compact_finish(void *h, rrdb_compact_stats *stats)
 extern "C" {
 
 int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
-                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts)
+                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts,
+                 uint32_t epoch_now)
 {
     auto *e = (HipEngine *)h;
     std::lock_guard<std::mutex> g(e->mu);
     if (!hklen_ok(hklen))
         return RRDB_INVALID_ARGUMENT;
+    /* write-time default_ttl (rocksdb_wrapper::db_expire_ts, :280-286) */
+    if (expire_ts == 0 && e->default_ttl != 0)
+        expire_ts = epoch_now + e->default_ttl;
     std::string key = make_key(hash_key, hklen, sort_key, sklen);
     uint32_t hdr = hdr_len(e->data_version);
     std::string val(hdr + vlen, '\0');

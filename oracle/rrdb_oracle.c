@@ -2305,12 +2305,16 @@ static void mem_append(Engine *e, uint8_t *key, uint64_t klen, uint8_t *val, uin
 }
 
 int32_t rrdb_put(void *h, const uint8_t *hash_key, uint64_t hklen, const uint8_t *sort_key,
-                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts)
+                 uint64_t sklen, const uint8_t *value, uint64_t vlen, uint32_t expire_ts,
+                 uint32_t epoch_now)
 {
     Engine *e = (Engine *)h;
     uint64_t klen;
     if (hklen >= 0xFFFFull)
         return RRDB_INVALID_ARGUMENT; /* 2-byte length prefix cap (key_schema.h:43) */
+    /* write-time default_ttl (rocksdb_wrapper::db_expire_ts, :280-286) */
+    if (expire_ts == 0 && e->default_ttl != 0)
+        expire_ts = epoch_now + e->default_ttl;
     uint8_t *key = make_key(hash_key, hklen, sort_key, sklen, &klen);
     uint32_t hdr = value_hdr_len(e->data_version);
     uint8_t *val = (uint8_t *)calloc(1, hdr + vlen);

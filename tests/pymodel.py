@@ -67,8 +67,12 @@ class Model:
     # -- write path (pegasus_write_service put/remove + memtable flush,
     #    reference pegasus_write_service.h:119-207, rocksdb_wrapper.cpp:
     #    121-247; mirrors the oracle/engine host memtable) --
-    def put(self, hash_key: bytes, sort_key: bytes, body: bytes, expire_ts=0):
+    def put(self, hash_key: bytes, sort_key: bytes, body: bytes, expire_ts=0,
+            epoch_now=0):
         key = struct.pack(">H", len(hash_key)) + hash_key + sort_key
+        # write-time default_ttl (rocksdb_wrapper::db_expire_ts, :280-286)
+        if expire_ts == 0 and self.default_ttl != 0:
+            expire_ts = (epoch_now + self.default_ttl) & 0xFFFFFFFF
         ver = self.data_version
         if ver == 0:
             val = struct.pack(">I", expire_ts & 0xFFFFFFFF) + body

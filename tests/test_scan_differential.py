@@ -265,3 +265,30 @@ def test_compact_stats_shadowed_tombstones_oracle_vs_model(oracle_lib):
                            output_bytes=ob)
     finally:
         o.close()
+
+
+def test_write_time_default_ttl(oracle_lib):
+    """default_ttl applies AT WRITE TIME when the put carries no TTL
+    (rocksdb_wrapper::db_expire_ts, rocksdb_wrapper.cpp:280-286; scenario of
+    test_ttl.cpp set_with_default_ttl): ttl() shows the countdown BEFORE any
+    compaction, and an explicit write TTL wins over the default."""
+    o = oracle_lib.open(1, 0, -1)
+    try:
+        model = Model()
+        o.set_envs({"default_ttl": "500"})
+        model.default_ttl = 500
+        for part_put, model_put in ((o.put, model.put),):
+            part_put(b"hk", b"s0", b"v", 0, 1000)
+            model_put(b"hk", b"s0", b"v", 0, 1000)
+            part_put(b"hk", b"s1", b"v", 1000 + 77, 1000)
+            model_put(b"hk", b"s1", b"v", 1000 + 77, 1000)
+        model.flush()  # oracle reads flush the memtable; mirror on the model
+        for sk, want_ttl in ((b"s0", 500), (b"s1", 77)):
+            k = D.generate_key(b"hk", sk)
+            assert o.ttl(k, 1000) == model.ttl(k, 1000) == (OK, want_ttl), sk
+        # past write-time expiry the record is gone without any compaction
+        k0 = D.generate_key(b"hk", b"s0")
+        assert o.get(k0, 1501) == model.get(k0, 1501)
+        assert o.get(k0, 1501)[0] == 1  # kNotFound
+    finally:
+        o.close()

@@ -66,8 +66,8 @@ class OracleVsModel(RuleBasedStateMachine):
     # -- write path through the C-ABI (memtable put/remove/flush) --
     @rule(hk=HK, sk=SK, body=BODY, ttl=TTL)
     def api_put(self, hk, sk, body, ttl):
-        assert self.part.put(hk, sk, body, ttl) == 0
-        self.model.put(hk, sk, body, ttl)
+        assert self.part.put(hk, sk, body, ttl, epoch_now=NOW) == 0
+        self.model.put(hk, sk, body, ttl, epoch_now=NOW)
 
     @rule(hk=HK, sk=SK)
     def api_remove(self, hk, sk):

@@ -69,7 +69,7 @@ def main():
                 v = f"v{step}".encode() * rnd.randrange(1, 6)
                 exp = 0 if rnd.random() < 0.9 else now + rnd.randrange(1, 50)
                 for p in (o, g):
-                    p.put(h, s, v, exp)
+                    p.put(h, s, v, exp, now)
             elif op < 45:
                 h, s = hk(rnd.randrange(args.scale)), sk(rnd.randrange(8))
                 for p in (o, g):

@@ -49,8 +49,8 @@ def main():
                 else:
                     v = b"v%d" % step
                     ttl = rnd.choice([0, 0, 0, now + 50])
-                    g.put(hk, sk, v, ttl)
-                    o.put(hk, sk, v, ttl)
+                    g.put(hk, sk, v, ttl, now)
+                    o.put(hk, sk, v, ttl, now)
         elif op == 2:  # point gets
             for _ in range(8):
                 k = D.generate_key(b"hk%03d" % rnd.randrange(70), b"s%02d" % rnd.randrange(7))
